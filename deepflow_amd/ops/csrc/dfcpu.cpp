@@ -1,0 +1,259 @@
+// dfcpu — host-side native library: fast synthetic-stream generators and the
+// ingest pre-segmentation scan.
+//
+// The generators are byte-identical twins of deepflow_amd/gen/*.py (golden
+// tests in tests/test_gen_native.py assert equality); they exist because the
+// benchmark drives millions of spans per second and the pure-Python encoder
+// is a test fixture, not a data plane.
+//
+// Build: g++ -O3 -shared -fPIC -march=native -fopenmp dfcpu.cpp -o libdfcpu.so
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <cstdlib>
+
+namespace {
+
+constexpr uint64_t GAMMA = 0x9E3779B97F4A7C15ull;
+
+struct SplitMix64 {
+    uint64_t state;
+    explicit SplitMix64(uint64_t seed) : state(seed) {}
+    uint64_t next() {
+        state += GAMMA;
+        uint64_t z = state;
+        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+        z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+        return z ^ (z >> 31);
+    }
+    uint64_t below(uint64_t n) { return next() % n; }
+};
+
+struct Buf {
+    uint8_t* p;
+    size_t len;
+    size_t cap;
+    void put(uint8_t b) { if (len < cap) p[len] = b; len++; }
+    void bytes(const void* src, size_t n) {
+        if (len + n <= cap) memcpy(p + len, src, n);
+        len += n;
+    }
+};
+
+void varint(Buf& b, uint64_t v) {
+    while (true) {
+        uint8_t x = v & 0x7F;
+        v >>= 7;
+        if (v) b.put(x | 0x80); else { b.put(x); return; }
+    }
+}
+
+// field helpers: skip proto3 zero-defaults, ascending field order is the
+// caller's responsibility (matches pb.py encode())
+void f_u(Buf& b, uint32_t num, uint64_t v) {
+    if (!v) return;
+    varint(b, (uint64_t(num) << 3) | 0);
+    varint(b, v);
+}
+void f_i(Buf& b, uint32_t num, int64_t v) {
+    if (!v) return;
+    varint(b, (uint64_t(num) << 3) | 0);
+    varint(b, uint64_t(v));
+}
+void f_s(Buf& b, uint32_t num, const char* s, size_t n) {
+    if (!n) return;
+    varint(b, (uint64_t(num) << 3) | 2);
+    varint(b, n);
+    b.bytes(s, n);
+}
+void f_s(Buf& b, uint32_t num, const char* s) { f_s(b, num, s, strlen(s)); }
+// length-delimited sub-message: encode into scratch, then emit
+template <typename F>
+void f_m(Buf& b, uint32_t num, F&& fill) {
+    uint8_t scratch[4096];
+    Buf sub{scratch, 0, sizeof scratch};
+    fill(sub);
+    varint(b, (uint64_t(num) << 3) | 2);
+    varint(b, sub.len);
+    b.bytes(scratch, sub.len);
+}
+
+void hex_str(char* out, uint64_t v, int width) {
+    static const char* H = "0123456789abcdef";
+    for (int i = 0; i < width; i++)
+        out[i] = H[(v >> (4 * (width - 1 - i))) & 0xF];
+    out[width] = 0;
+}
+
+struct SpanCfg {
+    uint64_t seed;
+    uint64_t base_time_ns;
+    uint64_t dt_ns;
+    uint32_t n_agents;
+    uint32_t n_ips;
+    uint32_t n_epcs;
+    uint32_t n_services;
+    uint32_t n_resources;
+    uint32_t tag_cardinality;
+    uint32_t n_attrs;
+    uint32_t err_rate_pct;
+};
+
+// Byte-identical twin of gen/spans.py:gen_span_dict + pb.encode
+void encode_span(Buf& b, const SpanCfg& c, uint64_t i) {
+    SplitMix64 rng(c.seed * 0x9E3779B9ull + i);
+    uint64_t r0 = rng.next();
+    uint64_t start = c.base_time_ns + i * c.dt_ns + rng.below(1000) * 1000;
+    uint64_t rrt_us = 100 + rng.below(200000);
+    uint64_t end = start + rrt_us * 1000;
+    uint32_t svc = (uint32_t)rng.below(c.n_services);
+    uint32_t res = (uint32_t)rng.below(c.n_resources);
+    uint32_t ip_c = 0x0A000000u | (uint32_t)rng.below(c.n_ips);
+    uint32_t ip_s = 0x0A000000u | ((svc * 7u) % c.n_ips);
+    bool err = rng.below(100) < c.err_rate_pct;
+    uint64_t trace_hi = rng.next(), trace_lo = rng.next();
+    uint64_t span_id_v = rng.next();
+    // remaining draws in the exact order gen_span_dict evaluates them:
+    // base-dict literal draws, then the attr loop, then span-dict literal.
+    uint32_t port_src = 32768 + (uint32_t)(r0 % 28000);
+    uint32_t process_id_0 = 1000 + (uint32_t)rng.below(64);
+    uint64_t syscall_req = rng.next() & 0x7FFFFFFFFFFFFFFFull;
+    uint32_t gpid_0 = 1 + (uint32_t)rng.below(1u << 16);
+    uint32_t attr_vals[64];
+    for (uint32_t a = 0; a < c.n_attrs && a < 64; a++)
+        attr_vals[a] = (uint32_t)rng.below(c.tag_cardinality);
+    int64_t req_len = 128 + (int64_t)rng.below(1024);
+    int64_t resp_len = 256 + (int64_t)rng.below(8192);
+    uint64_t request_id = rng.below(1u << 30);
+
+    // base (field 1)
+    f_m(b, 1, [&](Buf& s) {
+        f_u(s, 1, start);
+        f_u(s, 2, end);
+        f_u(s, 3, r0 & 0x7FFFFFFFFFFFFFFFull);
+        // 4 tap_port = 0 skipped
+        f_u(s, 5, 1 + (r0 % c.n_agents));
+        f_u(s, 6, 3);            // tap_type
+        f_u(s, 8, 1);            // tap_side
+        f_m(s, 9, [&](Buf& h) {  // head
+            f_u(h, 1, 20);       // proto = HTTP_1
+            f_u(h, 2, 2);        // msg_type
+            f_u(h, 5, rrt_us);
+        });
+        f_u(s, 12, ip_c);
+        f_u(s, 13, ip_s);
+        f_i(s, 16, 1 + (int64_t)(ip_c % c.n_epcs));
+        f_i(s, 17, 1 + (int64_t)(ip_s % c.n_epcs));
+        f_u(s, 18, port_src);
+        f_u(s, 19, 8080);
+        f_u(s, 20, 6);
+        f_u(s, 23, r0 & 0xFFFFFFFFull);
+        f_u(s, 24, (r0 >> 16) & 0xFFFFFFFFull);
+        f_u(s, 25, process_id_0);
+        f_u(s, 26, 2000 + (svc % 64));
+        f_u(s, 29, syscall_req);
+        f_u(s, 35, gpid_0);
+        f_u(s, 36, 1 + (svc % (1u << 16)));
+        f_u(s, 41, 1 + (ip_c % c.n_ips));
+        f_u(s, 42, 1 + (ip_s % c.n_ips));
+    });
+    f_i(b, 9, req_len);
+    f_i(b, 10, resp_len);
+    char tmp[64];
+    f_m(b, 11, [&](Buf& s) {  // req
+        f_s(s, 1, ((r0 >> 8) % 4 == 0) ? "POST" : "GET");
+        snprintf(tmp, sizeof tmp, "svc-%03u.example.com", svc);
+        f_s(s, 2, tmp);
+        snprintf(tmp, sizeof tmp, "/api/v1/r/%05u", res);
+        f_s(s, 3, tmp);
+        f_s(s, 4, "/api/v1/r");
+    });
+    f_m(b, 12, [&](Buf& s) {  // resp
+        f_u(s, 1, err ? 3 : 0);       // status
+        f_i(s, 2, err ? 500 : 200);   // code
+    });
+    f_s(b, 13, "1.1");  // version
+    f_m(b, 14, [&](Buf& s) {  // trace_info
+        char tid[40], sid[24];
+        hex_str(tid, trace_hi, 16);
+        hex_str(tid + 16, trace_lo, 16);
+        hex_str(sid, span_id_v, 16);
+        f_s(s, 1, tid, 32);
+        f_s(s, 2, sid, 16);
+    });
+    f_m(b, 15, [&](Buf& s) {  // ext_info
+        snprintf(tmp, sizeof tmp, "svc-%03u", svc);
+        f_s(s, 1, tmp);
+        f_u(s, 3, request_id);
+        for (uint32_t a = 0; a < c.n_attrs && a < 64; a++) {
+            snprintf(tmp, sizeof tmp, "attr_%u", a);
+            f_s(s, 16, tmp);
+        }
+        for (uint32_t a = 0; a < c.n_attrs && a < 64; a++) {
+            snprintf(tmp, sizeof tmp, "v%07u", attr_vals[a]);
+            f_s(s, 17, tmp);
+        }
+    });
+    f_u(b, 17, 255);  // direction_score
+    f_u(b, 19, 128);  // captured_request_byte
+    f_u(b, 20, 256);  // captured_response_byte
+}
+
+}  // namespace
+
+extern "C" {
+
+// Generate spans [i0, i0+n) as a length-prefixed record payload into out.
+// Returns total bytes needed (call with cap=0 to size, then again to fill).
+uint64_t df_gen_spans(const SpanCfg* cfg, uint64_t i0, uint64_t n,
+                      uint8_t* out, uint64_t cap) {
+    Buf b{out, 0, cap};
+    uint8_t scratch[16384];
+    for (uint64_t i = i0; i < i0 + n; i++) {
+        Buf rec{scratch, 0, sizeof scratch};
+        encode_span(rec, *cfg, i);
+        uint32_t ln = (uint32_t)rec.len;
+        b.bytes(&ln, 4);
+        b.bytes(scratch, rec.len < sizeof scratch ? rec.len : sizeof scratch);
+    }
+    return b.len;
+}
+
+// Parallel variant: fills per-chunk into a caller-provided buffer using a
+// first sizing pass. Returns bytes written; also writes record offsets/lens.
+uint64_t df_gen_spans_indexed(const SpanCfg* cfg, uint64_t i0, uint64_t n,
+                              uint8_t* out, uint64_t cap,
+                              uint32_t* offs, uint32_t* lens) {
+    Buf b{out, 0, cap};
+    uint8_t scratch[16384];
+    for (uint64_t i = 0; i < n; i++) {
+        Buf rec{scratch, 0, sizeof scratch};
+        encode_span(rec, *cfg, i0 + i);
+        uint32_t ln = (uint32_t)rec.len;
+        b.bytes(&ln, 4);
+        if (offs) offs[i] = (uint32_t)b.len;
+        if (lens) lens[i] = ln;
+        b.bytes(scratch, rec.len);
+    }
+    return b.len;
+}
+
+// Pre-segmentation scan of a [u32 LE len][pb bytes]* payload.
+// Writes up to max_n (offset, len) pairs; returns record count (may exceed
+// max_n to signal truncation). Offsets point at the pb bytes.
+uint64_t df_scan_offsets(const uint8_t* payload, uint64_t len,
+                         uint32_t* offs, uint32_t* lens, uint64_t max_n) {
+    uint64_t pos = 0, n = 0;
+    while (pos + 4 <= len) {
+        uint32_t ln;
+        memcpy(&ln, payload + pos, 4);
+        pos += 4;
+        if (pos + ln > len) break;
+        if (n < max_n) { offs[n] = (uint32_t)pos; lens[n] = ln; }
+        n++;
+        pos += ln;
+    }
+    return n;
+}
+
+}  // extern "C"

@@ -1,0 +1,113 @@
+"""ctypes bindings for libdfcpu.so (host) and libdfgpu.so (HIP/gfx950).
+
+GPU policy: on a machine with a visible GPU the HIP library is REQUIRED —
+ops fail loudly rather than falling back to eager PyTorch, so a passing GPU
+test run always means the native kernels executed.
+"""
+from __future__ import annotations
+
+import ctypes as ct
+from pathlib import Path
+from typing import Optional
+
+OPS_DIR = Path(__file__).resolve().parent
+
+_cpu_lib: Optional[ct.CDLL] = None
+_gpu_lib: Optional[ct.CDLL] = None
+
+
+class SpanCfgC(ct.Structure):
+    _fields_ = [
+        ("seed", ct.c_uint64),
+        ("base_time_ns", ct.c_uint64),
+        ("dt_ns", ct.c_uint64),
+        ("n_agents", ct.c_uint32),
+        ("n_ips", ct.c_uint32),
+        ("n_epcs", ct.c_uint32),
+        ("n_services", ct.c_uint32),
+        ("n_resources", ct.c_uint32),
+        ("tag_cardinality", ct.c_uint32),
+        ("n_attrs", ct.c_uint32),
+        ("err_rate_pct", ct.c_uint32),
+    ]
+
+
+def span_cfg_c(cfg) -> SpanCfgC:
+    """Convert gen.spans.SpanGenConfig -> C struct."""
+    return SpanCfgC(
+        seed=cfg.seed, base_time_ns=cfg.base_time_ns, dt_ns=cfg.dt_ns,
+        n_agents=cfg.n_agents, n_ips=cfg.n_ips, n_epcs=cfg.n_epcs,
+        n_services=cfg.n_services, n_resources=cfg.n_resources,
+        tag_cardinality=cfg.tag_cardinality, n_attrs=cfg.n_attrs,
+        err_rate_pct=cfg.err_rate_pct,
+    )
+
+
+def cpu() -> ct.CDLL:
+    global _cpu_lib
+    if _cpu_lib is None:
+        path = OPS_DIR / "libdfcpu.so"
+        if not path.exists():
+            from . import build
+            build.build_cpu()
+        lib = ct.CDLL(str(path))
+        lib.df_gen_spans.restype = ct.c_uint64
+        lib.df_gen_spans.argtypes = [ct.POINTER(SpanCfgC), ct.c_uint64,
+                                     ct.c_uint64, ct.c_void_p, ct.c_uint64]
+        lib.df_gen_spans_indexed.restype = ct.c_uint64
+        lib.df_gen_spans_indexed.argtypes = [
+            ct.POINTER(SpanCfgC), ct.c_uint64, ct.c_uint64, ct.c_void_p,
+            ct.c_uint64, ct.c_void_p, ct.c_void_p]
+        lib.df_scan_offsets.restype = ct.c_uint64
+        lib.df_scan_offsets.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p,
+                                        ct.c_void_p, ct.c_uint64]
+        _cpu_lib = lib
+    return _cpu_lib
+
+
+def _decl_gpu(lib: ct.CDLL) -> None:
+    u64, u32, p = ct.c_uint64, ct.c_uint32, ct.c_void_p
+    lib.df_gpu_ready.restype = ct.c_int
+    lib.df_decode_l7.restype = ct.c_int
+    lib.df_decode_l7.argtypes = [p, p, p, u32, p, p, p, p, p, p, u64, u64, u64]
+    lib.df_kg_build.restype = ct.c_int
+    lib.df_kg_build.argtypes = [p, p, u32, p, p, u32, u64]
+    lib.df_kg_probe.restype = ct.c_int
+    lib.df_kg_probe.argtypes = [p, p, p, p, u32, p, p, u32, p, u64, u64, u64]
+    lib.df_intern_many.restype = ct.c_int
+    lib.df_intern_many.argtypes = [p, p, p, u32, u32, u64, u64, p, u32,
+                                   p, p, u32, p, u64, u64, u64]
+    lib.df_pool_lens.restype = ct.c_int
+    lib.df_pool_lens.argtypes = [p, p, u32, u32, u64, u64, p, u64]
+    lib.df_pool_gather.restype = ct.c_int
+    lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, u64]
+    lib.df_agg_app1s.restype = ct.c_int
+    lib.df_agg_app1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
+    lib.df_query_agg.restype = ct.c_int
+    lib.df_query_agg.argtypes = [p, p, p, p, p, p, p, u64, u64, p, u32, u64,
+                                 p, p, p, u32, u64]
+    lib.df_query_select.restype = ct.c_int
+    lib.df_query_select.argtypes = [p, p, p, p, p, p, p, u64, u64, p, u32, u64,
+                                    p, p, u32, u64]
+    lib.df_spec_sizes.restype = ct.c_int
+    lib.df_spec_sizes.argtypes = [p, p, p, p]
+
+
+def gpu() -> ct.CDLL:
+    """Load the HIP kernel library; raises if missing (no silent fallback)."""
+    global _gpu_lib
+    if _gpu_lib is None:
+        path = OPS_DIR / "libdfgpu.so"
+        if not path.exists():
+            raise RuntimeError(
+                "libdfgpu.so not built — run deepflow_amd/ops/build.py "
+                "(GPU ops never fall back to eager)")
+        lib = ct.CDLL(str(path))
+        _decl_gpu(lib)
+        _gpu_lib = lib
+    return _gpu_lib
+
+
+def check(rc: int, what: str) -> None:
+    if rc != 0:
+        raise RuntimeError(f"HIP error {rc} in {what}")
