@@ -1,0 +1,163 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: spans/sec ingested (whole node), BASELINE config #2/#3.
+
+Each step ingests one synthetic AppProtoLogsData batch per rank through the
+full GPU pipeline (H2D copy -> K1 pb decode -> K2 KnowledgeGraph join ->
+K3 SmartEncoding intern -> K4 string-pool gather -> K5 1s metric rollup),
+plus (world_size > 1) the RCCL dictionary-delta all-gather. Weak scaling:
+per-GPU work is fixed as N grows; every rank ingests a disjoint span range.
+
+Usage: python bench.py --gpus N --steps K --warmup W
+Launched multi-GPU by the driver via torch.distributed.run (one rank/GPU).
+"""
+from __future__ import annotations
+
+import argparse
+import ctypes as ct
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from deepflow_amd.gen import SpanGenConfig
+from deepflow_amd.ingest import L7IngestPipeline
+from deepflow_amd.ops import native
+from deepflow_amd.store.kg import KnowledgeGraphTable, default_platform
+
+
+def gen_batches(cfg: SpanGenConfig, rank: int, n_batches: int, batch: int):
+    """Pre-generate n_batches distinct payloads (disjoint span index ranges
+    per rank) with the native generator; returns [(payload, offs, lens)]."""
+    lib = native.cpu()
+    c = native.span_cfg_c(cfg)
+    out = []
+    for b in range(n_batches):
+        i0 = (rank * n_batches + b) * batch
+        need = lib.df_gen_spans(ct.byref(c), i0, batch, None, 0)
+        buf = np.zeros(int(need), dtype=np.uint8)
+        offs = np.zeros(batch, dtype=np.uint32)
+        lens = np.zeros(batch, dtype=np.uint32)
+        lib.df_gen_spans_indexed(ct.byref(c), i0, batch,
+                                 buf.ctypes.data_as(ct.c_void_p), need,
+                                 offs.ctypes.data_as(ct.c_void_p),
+                                 lens.ctypes.data_as(ct.c_void_p))
+        out.append((buf, offs, lens))
+    return out
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=1_000_000,
+                    help="spans per step per rank")
+    ap.add_argument("--tag-card", type=int, default=100_000)
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    have_gpu = torch.cuda.is_available()
+    device = args.device or ("cuda" if have_gpu else "cpu")
+    if device == "cpu" and args.batch > 20000:
+        args.batch = 2000  # CPU reference path is a test fixture, keep tiny
+
+    if world > 1:
+        backend = "nccl" if device == "cuda" else "gloo"
+        torch.distributed.init_process_group(backend=backend)
+    if device == "cuda":
+        torch.cuda.set_device(local_rank)
+
+    cfg = SpanGenConfig(n=args.batch, seed=1234,
+                        tag_cardinality=args.tag_card,
+                        n_ips=4096, n_services=256, n_resources=4096,
+                        n_attrs=4)
+    n_distinct = min(args.steps + args.warmup, 4)
+    batches = gen_batches(cfg, rank, n_distinct, args.batch)
+
+    kg = KnowledgeGraphTable(capacity_pow2=1 << 14, device=device)
+    kg.update(default_platform(cfg))
+    pipe = L7IngestPipeline(device=device, segment_rows=1 << 23,
+                            kg=kg, dict_capacity=1 << 23,
+                            time_base_s=cfg.base_time_ns // 10**9)
+
+    dict_sync = None
+    if world > 1:
+        from deepflow_amd.parallel.dict_sync import DictSync
+        dict_sync = DictSync(pipe.dict)
+
+    def step(i: int) -> None:
+        payload, offs, lens = batches[i % n_distinct]
+        pipe.ingest(payload, offs, lens)
+        if dict_sync is not None:
+            dict_sync.sync_step()
+
+    def barrier_sync() -> None:
+        if world > 1:
+            torch.distributed.barrier()
+        if device == "cuda":
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    barrier_sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device == "cuda" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_spans = args.steps * args.batch * world
+    spans_per_sec = total_spans / elapsed
+    seg = pipe.segments.segments[0]
+    bytes_per_span = seg.stored_bytes_per_row()
+
+    if rank == 0:
+        out = {
+            "metric": "spans_per_sec_ingested",
+            "value": round(spans_per_sec, 1),
+            "unit": "spans/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "uint8/int32 (columnar ints; no FP model)",
+            "data": "synthetic",
+            "config": {
+                "model": "l7_span_ingest (AppProtoLogsData wire decode + "
+                         "SmartEncoding + KG join + 1s rollup)",
+                "global_batch": args.batch * world,
+                "seq_len": None,
+                "parallelism": f"shard{world} (hash-sharded span streams)",
+                "tag_cardinality": args.tag_card,
+                "bytes_per_span_stored": round(bytes_per_span, 1),
+                "dict_entries": pipe.dict.n_entries(),
+                "device": device,
+            },
+        }
+        print(json.dumps(out))
+
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

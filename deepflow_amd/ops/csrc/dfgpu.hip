@@ -369,7 +369,8 @@ __global__ void k_kg_probe(const uint32_t* __restrict__ epc0,
 // ----------------------------------------------------------------------
 
 __global__ void k_intern_many(const uint8_t* __restrict__ payload,
-                              const uint64_t* __restrict__ refs,  // [C, stride]
+                              const uint64_t* __restrict__ refs,  // [*, stride]
+                              const uint16_t* __restrict__ ref_rows,  // [C] row in refs per column
                               const uint8_t* __restrict__ domains,  // [C]
                               uint32_t C, uint32_t n,
                               uint64_t ref_stride, uint64_t ref_base_row,
@@ -383,7 +384,7 @@ __global__ void k_intern_many(const uint8_t* __restrict__ payload,
     if (gid >= total) return;
     uint32_t c = (uint32_t)(gid / n);
     uint32_t i = (uint32_t)(gid % n);
-    uint64_t ref = refs[c * ref_stride + ref_base_row + i];
+    uint64_t ref = refs[(uint64_t)ref_rows[c] * ref_stride + ref_base_row + i];
     uint32_t len = STR_REF_LEN(ref);
     uint64_t off = STR_REF_OFF(ref);
     uint32_t* out = &out_ids[c * out_stride + out_base_row + i];
@@ -683,7 +684,8 @@ int df_kg_probe(const void* epc0, const void* ip0, const void* epc1, const void*
     return (int)hipGetLastError();
 }
 
-int df_intern_many(const void* payload, const void* refs, const void* domains,
+int df_intern_many(const void* payload, const void* refs, const void* ref_rows,
+                   const void* domains,
                    uint32_t C, uint32_t n, uint64_t ref_stride, uint64_t ref_base_row,
                    void* tkeys, uint32_t cap,
                    void* emit, void* emit_ctr, uint32_t emit_cap,
@@ -692,6 +694,7 @@ int df_intern_many(const void* payload, const void* refs, const void* domains,
     uint64_t total = (uint64_t)C * n;
     hipLaunchKernelGGL(k_intern_many, dim3(grid_for(total)), dim3(BLOCK), 0, STREAM(stream),
                        (const uint8_t*)payload, (const uint64_t*)refs,
+                       (const uint16_t*)ref_rows,
                        (const uint8_t*)domains, C, n, ref_stride, ref_base_row,
                        (uint64_t*)tkeys, cap - 1,
                        (uint64_t*)emit, (uint32_t*)emit_ctr, emit_cap,

@@ -1,0 +1,127 @@
+"""Tensor-level wrappers over the HIP C API (libdfgpu.so).
+
+Every function takes torch tensors already resident on the GPU, launches
+asynchronously on the current torch CUDA stream, and raises on HIP errors.
+There is deliberately NO eager/PyTorch fallback here: on a GPU box these are
+the only code paths (see ops/ref.py for the CPU reference used by tests and
+by the device='cpu' pipeline).
+"""
+from __future__ import annotations
+
+import torch
+
+from . import native
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def decode_l7(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
+              seg, base_row: int) -> None:
+    n = offs.numel()
+    lib = native.gpu()
+    native.check(lib.df_decode_l7(
+        payload.data_ptr(), offs.data_ptr(), lens.data_ptr(), n,
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        seg.strref.data_ptr(), seg.attr_ref.data_ptr(), seg.attr_cnt.data_ptr(),
+        seg.capacity, base_row, _stream()), "df_decode_l7")
+
+
+def kg_build(keys: torch.Tensor, vals: torch.Tensor, tkeys: torch.Tensor,
+             tvals: torch.Tensor) -> None:
+    lib = native.gpu()
+    native.check(lib.df_kg_build(
+        keys.data_ptr(), vals.data_ptr(), keys.numel(),
+        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(), _stream()),
+        "df_kg_build")
+
+
+def kg_probe(seg, base_row: int, n: int, tkeys: torch.Tensor,
+             tvals: torch.Tensor) -> None:
+    from ..store import l7_schema as S
+    lib = native.gpu()
+    u32 = seg.u32
+    cap = seg.capacity
+    epc0 = u32[S.U32_COLS.index("l3_epc_id_0")].data_ptr() + 4 * base_row
+    ip0 = u32[S.U32_COLS.index("ip4_0")].data_ptr() + 4 * base_row
+    epc1 = u32[S.U32_COLS.index("l3_epc_id_1")].data_ptr() + 4 * base_row
+    ip1 = u32[S.U32_COLS.index("ip4_1")].data_ptr() + 4 * base_row
+    native.check(lib.df_kg_probe(
+        epc0, ip0, epc1, ip1, n, tkeys.data_ptr(), tvals.data_ptr(),
+        tkeys.numel(), seg.kg.data_ptr(), cap, base_row, _stream()),
+        "df_kg_probe")
+
+
+def intern_many(payload: torch.Tensor, refs: torch.Tensor,
+                ref_rows: torch.Tensor, domains: torch.Tensor,
+                ref_base_row: int, n: int,
+                tkeys: torch.Tensor, emit: torch.Tensor,
+                emit_ctr: torch.Tensor, out_ids: torch.Tensor,
+                out_base_row: int) -> None:
+    C = domains.numel()
+    lib = native.gpu()
+    native.check(lib.df_intern_many(
+        payload.data_ptr(), refs.data_ptr(), ref_rows.data_ptr(),
+        domains.data_ptr(), C, n, refs.shape[1], ref_base_row,
+        tkeys.data_ptr(), tkeys.numel(),
+        emit.data_ptr(), emit_ctr.data_ptr(), emit.shape[0],
+        out_ids.data_ptr(), out_ids.shape[1], out_base_row, _stream()),
+        "df_intern_many")
+
+
+def pool_lens(seg, pool_cols: torch.Tensor, base_row: int, n: int,
+              row_len: torch.Tensor) -> None:
+    lib = native.gpu()
+    native.check(lib.df_pool_lens(
+        seg.strref.data_ptr(), pool_cols.data_ptr(), pool_cols.numel(), n,
+        seg.capacity, base_row, row_len.data_ptr(), _stream()), "df_pool_lens")
+
+
+def pool_gather(payload: torch.Tensor, seg, pool_cols: torch.Tensor,
+                base_row: int, n: int, row_start: torch.Tensor,
+                pool: torch.Tensor, pool_base: int) -> None:
+    lib = native.gpu()
+    native.check(lib.df_pool_gather(
+        payload.data_ptr(), seg.strref.data_ptr(), pool_cols.data_ptr(),
+        pool_cols.numel(), n, seg.capacity, base_row, row_start.data_ptr(),
+        pool.data_ptr(), pool_base, _stream()), "df_pool_gather")
+
+
+def agg_app1s(seg, base_row: int, n: int, time_base_s: int,
+              tkeys: torch.Tensor, tvals: torch.Tensor) -> None:
+    lib = native.gpu()
+    native.check(lib.df_agg_app1s(
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        seg.capacity, base_row, n, time_base_s,
+        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(), _stream()),
+        "df_agg_app1s")
+
+
+def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
+              gkeys: torch.Tensor, graw: torch.Tensor,
+              gvals: torch.Tensor) -> None:
+    import ctypes
+    lib = native.gpu()
+    buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
+    native.check(lib.df_query_agg(
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        seg.did.data_ptr(), seg.kg.data_ptr(), seg.attr_id.data_ptr(),
+        seg.attr_cnt.data_ptr(), seg.capacity, seg.n_rows,
+        ctypes.addressof(buf), n, base_row,
+        gkeys.data_ptr(), graw.data_ptr(), gvals.data_ptr(), gkeys.numel(),
+        _stream()), "df_query_agg")
+
+
+def query_select(seg, spec_bytes: bytes, base_row: int, n: int,
+                 out_rows: torch.Tensor, out_ctr: torch.Tensor) -> None:
+    import ctypes
+    lib = native.gpu()
+    buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
+    native.check(lib.df_query_select(
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        seg.did.data_ptr(), seg.kg.data_ptr(), seg.attr_id.data_ptr(),
+        seg.attr_cnt.data_ptr(), seg.capacity, seg.n_rows,
+        ctypes.addressof(buf), n, base_row,
+        out_rows.data_ptr(), out_ctr.data_ptr(), out_rows.numel(), _stream()),
+        "df_query_select")

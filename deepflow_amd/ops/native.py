@@ -75,7 +75,7 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_kg_probe.restype = ct.c_int
     lib.df_kg_probe.argtypes = [p, p, p, p, u32, p, p, u32, p, u64, u64, u64]
     lib.df_intern_many.restype = ct.c_int
-    lib.df_intern_many.argtypes = [p, p, p, u32, u32, u64, u64, p, u32,
+    lib.df_intern_many.argtypes = [p, p, p, p, u32, u32, u64, u64, p, u32,
                                    p, p, u32, p, u64, u64, u64]
     lib.df_pool_lens.restype = ct.c_int
     lib.df_pool_lens.argtypes = [p, p, u32, u32, u64, u64, p, u64]

@@ -1,0 +1,341 @@
+"""CPU reference implementations of the HIP kernels.
+
+Used (a) by tests on GPU-less machines, (b) as the numerics oracle the GPU
+kernels are compared against (tests/test_gpu_pipeline.py), (c) by the
+device='cpu' pipeline (multi-process gloo tests). Semantics mirror
+ops/csrc/dfgpu.hip exactly; dictionary/table slot ids may differ from a
+concurrent GPU run under hash collisions, so cross-checks compare hydrated
+strings, not raw slot ids.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Tuple
+
+import numpy as np
+import torch
+
+from ..store import l7_schema as S
+from ..store.dictionary import str_hash_py, domain_seed
+from ..wire.pb import read_varint
+
+M64 = (1 << 64) - 1
+
+
+def mix64(z: int) -> int:
+    z &= M64
+    z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & M64
+    z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & M64
+    return (z ^ (z >> 31)) & M64
+
+
+# ---------------------------------------------------------------- K1 decode
+
+_U64_IDX = {c: i for i, c in enumerate(S.U64_COLS)}
+_U32_IDX = {c: i for i, c in enumerate(S.U32_COLS)}
+_U8_IDX = {c: i for i, c in enumerate(S.U8_COLS)}
+_STR_IDX = {c: i for i, c in enumerate(S.STR_COLS)}
+
+_BASE_U = {
+    1: ("u64", "start_time"), 2: ("u64", "end_time"), 3: ("u64", "flow_id"),
+    5: ("u32", "vtap_id"), 6: ("u8", "tap_type"), 7: ("u8", "is_ipv6"),
+    8: ("u8", "tap_side"), 12: ("u32", "ip4_0"), 13: ("u32", "ip4_1"),
+    16: ("u32", "l3_epc_id_0"), 17: ("u32", "l3_epc_id_1"),
+    18: ("u32", "client_port"), 19: ("u32", "server_port"),
+    20: ("u8", "protocol"), 23: ("u32", "req_tcp_seq"),
+    24: ("u32", "resp_tcp_seq"), 25: ("u32", "process_id_0"),
+    26: ("u32", "process_id_1"), 29: ("u64", "syscall_trace_id_request"),
+    30: ("u64", "syscall_trace_id_response"), 35: ("u32", "gprocess_id_0"),
+    36: ("u32", "gprocess_id_1"), 41: ("u32", "pod_id_0"),
+    42: ("u32", "pod_id_1"), 43: ("u32", "biz_type"),
+}
+
+
+def _w(seg, fam, col, row, v):
+    t = {"u64": seg.u64, "u32": seg.u32, "u8": seg.u8}[fam]
+    idx = {"u64": _U64_IDX, "u32": _U32_IDX, "u8": _U8_IDX}[fam][col]
+    if fam == "u64":
+        t[idx, row] = v & M64 if v < (1 << 63) else (v & M64) - (1 << 64)
+    elif fam == "u32":
+        v32 = v & 0xFFFFFFFF
+        t[idx, row] = v32 if v32 < (1 << 31) else v32 - (1 << 32)
+    else:
+        t[idx, row] = v & 0xFF
+
+
+def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
+    mv = memoryview(payload)
+    for rid in range(len(offs)):
+        row = base_row + rid
+        pos = int(offs[rid])
+        end = pos + int(lens[rid])
+        n_names = n_vals = 0
+        while pos < end:
+            key, pos = read_varint(mv, pos)
+            num, wt = key >> 3, key & 7
+            if wt == 0:
+                v, pos = read_varint(mv, pos)
+                if num == 9:
+                    _w(seg, "u32", "request_length", row, v)
+                elif num == 10:
+                    _w(seg, "u32", "response_length", row, v)
+                elif num == 17:
+                    _w(seg, "u8", "direction_score", row, v)
+                elif num == 18:
+                    _w(seg, "u32", "flags", row, v)
+                elif num == 19:
+                    _w(seg, "u32", "captured_request_byte", row, v)
+                elif num == 20:
+                    _w(seg, "u32", "captured_response_byte", row, v)
+            elif wt == 2:
+                ln, pos = read_varint(mv, pos)
+                sub, send = pos, pos + ln
+                pos = send
+                if num == 1:  # base
+                    p2 = sub
+                    while p2 < send:
+                        k2, p2 = read_varint(mv, p2)
+                        n2, w2 = k2 >> 3, k2 & 7
+                        if w2 == 0:
+                            v, p2 = read_varint(mv, p2)
+                            if n2 in _BASE_U:
+                                fam, col = _BASE_U[n2]
+                                _w(seg, fam, col, row, v)
+                        elif w2 == 2:
+                            l3, p2 = read_varint(mv, p2)
+                            s3, e3 = p2, p2 + l3
+                            p2 = e3
+                            if n2 == 9:  # head
+                                p3 = s3
+                                while p3 < e3:
+                                    k3, p3 = read_varint(mv, p3)
+                                    if (k3 & 7) == 0:
+                                        v, p3 = read_varint(mv, p3)
+                                        if k3 >> 3 == 1:
+                                            _w(seg, "u8", "l7_protocol", row, v)
+                                        elif k3 >> 3 == 2:
+                                            _w(seg, "u8", "msg_type", row, v)
+                                        elif k3 >> 3 == 5:
+                                            _w(seg, "u64", "rrt", row, v)
+                                    else:
+                                        p3 = e3
+                            elif n2 == 27:
+                                seg.strref[_STR_IDX["process_kname_0"], row] = \
+                                    S.str_ref_pack(s3, l3)
+                            elif n2 == 28:
+                                seg.strref[_STR_IDX["process_kname_1"], row] = \
+                                    S.str_ref_pack(s3, l3)
+                        elif w2 == 1:
+                            p2 += 8
+                        elif w2 == 5:
+                            p2 += 4
+                elif num in (11, 12, 14, 15):
+                    strmap = {
+                        11: {1: "request_type", 2: "request_domain",
+                             3: "request_resource", 4: "endpoint"},
+                        12: {3: "exception_desc", 4: "response_result"},
+                        14: {1: "trace_id", 2: "span_id", 3: "parent_span_id"},
+                        15: {1: "service_name", 4: "x_request_id_0",
+                             6: "http_user_agent", 7: "http_referer",
+                             10: "x_request_id_1"},
+                    }[num]
+                    p2 = sub
+                    while p2 < send:
+                        k2, p2 = read_varint(mv, p2)
+                        n2, w2 = k2 >> 3, k2 & 7
+                        if w2 == 0:
+                            v, p2 = read_varint(mv, p2)
+                            if num == 12 and n2 == 1:
+                                _w(seg, "u8", "response_status", row, v)
+                            elif num == 12 and n2 == 2:
+                                _w(seg, "u32", "response_code", row, v)
+                            elif num == 15 and n2 == 3:
+                                _w(seg, "u32", "request_id", row, v)
+                        elif w2 == 2:
+                            l3, p2 = read_varint(mv, p2)
+                            if num == 15 and n2 == 16:
+                                if n_names < S.MAX_ATTRS:
+                                    seg.attr_ref[n_names, row] = \
+                                        S.str_ref_pack(p2, l3)
+                                n_names += 1
+                            elif num == 15 and n2 == 17:
+                                if n_vals < S.MAX_ATTRS:
+                                    seg.attr_ref[S.MAX_ATTRS + n_vals, row] = \
+                                        S.str_ref_pack(p2, l3)
+                                n_vals += 1
+                            elif n2 in strmap:
+                                seg.strref[_STR_IDX[strmap[n2]], row] = \
+                                    S.str_ref_pack(p2, l3)
+                            p2 += l3
+                        elif w2 == 1:
+                            p2 += 8
+                        elif w2 == 5:
+                            p2 += 4
+                elif num == 13:
+                    seg.strref[_STR_IDX["version"], row] = S.str_ref_pack(sub, ln)
+                elif num == 21:
+                    seg.strref[_STR_IDX["biz_code"], row] = S.str_ref_pack(sub, ln)
+            elif wt == 1:
+                pos += 8
+            elif wt == 5:
+                pos += 4
+        na = min(n_names, n_vals, S.MAX_ATTRS)
+        seg.attr_cnt[row] = na
+
+
+# ------------------------------------------------------------ K2 kg tables
+
+def kg_build_ref(keys: torch.Tensor, vals: torch.Tensor,
+                 tkeys: torch.Tensor, tvals: torch.Tensor) -> None:
+    cap_mask = tkeys.numel() - 1
+    tk = tkeys.numpy()
+    tv = tvals.numpy()
+    kk = keys.numpy().view(np.uint64)
+    vv = vals.numpy()
+    for i in range(len(kk)):
+        k = int(kk[i])
+        if k == 0:
+            continue
+        slot = mix64(k) & cap_mask
+        for _ in range(cap_mask + 1):
+            cur = int(tk[slot]) & M64
+            if cur == 0 or cur == k:
+                tk[slot] = np.int64(np.uint64(k).astype(np.int64)) \
+                    if k < (1 << 63) else np.int64(k - (1 << 64))
+                tv[slot] = vv[i]
+                break
+            slot = (slot + 1) & cap_mask
+
+
+def kg_probe_ref(seg, base_row: int, n: int, tkeys: torch.Tensor,
+                 tvals: torch.Tensor) -> None:
+    cap_mask = tkeys.numel() - 1
+    tk = tkeys.numpy()
+    tv = tvals.numpy()
+    epc0 = seg.u32[_U32_IDX["l3_epc_id_0"]].numpy()
+    ip0 = seg.u32[_U32_IDX["ip4_0"]].numpy()
+    epc1 = seg.u32[_U32_IDX["l3_epc_id_1"]].numpy()
+    ip1 = seg.u32[_U32_IDX["ip4_1"]].numpy()
+    kg = seg.kg.numpy()
+    for i in range(n):
+        row = base_row + i
+        for side in range(2):
+            epc = int(epc0[row] if side == 0 else epc1[row]) & 0xFFFFFFFF
+            ip = int(ip0[row] if side == 0 else ip1[row]) & 0xFFFFFFFF
+            k = (epc << 32) | ip
+            slot = mix64(k) & cap_mask
+            found = False
+            for _ in range(cap_mask + 1):
+                cur = int(tk[slot]) & M64
+                if cur == k:
+                    found = True
+                    break
+                if cur == 0:
+                    break
+                slot = (slot + 1) & cap_mask
+            for j in range(S.N_KG):
+                kg[side * S.N_KG + j, row] = tv[slot, j] if found else 0
+
+
+# ------------------------------------------------------------- K3 intern
+
+def intern_ref(payload: bytes, refs: torch.Tensor, ref_rows, domains,
+               ref_base_row: int, n: int, tkeys: torch.Tensor,
+               out_ids: torch.Tensor, out_base_row: int,
+               dictionary=None) -> List[Tuple[int, int, bytes]]:
+    """Sequential-deterministic intern; returns new (domain, slot, bytes)."""
+    cap_mask = tkeys.numel() - 1
+    tk = tkeys.numpy()
+    new_entries: List[Tuple[int, int, bytes]] = []
+    for ci, (rrow, dom) in enumerate(zip(ref_rows, domains)):
+        for i in range(n):
+            ref = int(refs[rrow, ref_base_row + i].item()) & M64
+            ln = ref & 0xFFFF
+            off = ref >> 16
+            if ln == 0:
+                out_ids[ci, out_base_row + i] = -1  # DICT_ID_INVALID as i32
+                continue
+            sbytes = payload[off:off + ln]
+            h = str_hash_py(bytes(sbytes), domain_seed(dom))
+            slot = h & cap_mask
+            placed = False
+            for _ in range(cap_mask + 1):
+                cur = int(tk[slot]) & M64
+                if cur == 0:
+                    tk[slot] = np.int64(h) if h < (1 << 63) \
+                        else np.int64(h - (1 << 64))
+                    new_entries.append((dom, slot, bytes(sbytes)))
+                    placed = True
+                    break
+                if cur == h:
+                    placed = True
+                    break
+                slot = (slot + 1) & cap_mask
+            v = slot if placed else S.DICT_ID_INVALID
+            out_ids[ci, out_base_row + i] = v if v < (1 << 31) else v - (1 << 32)
+    if dictionary is not None:
+        for dom, slot, sbytes in new_entries:
+            dictionary.id_to_str[(dom, slot)] = sbytes
+            dictionary.str_to_id[(dom, sbytes)] = slot
+            dictionary.pending_sync.append((dom, slot, sbytes))
+    return new_entries
+
+
+# -------------------------------------------------------------- K4 pool
+
+def pool_lens_ref(seg, pool_cols, base_row: int, n: int) -> torch.Tensor:
+    out = torch.zeros(n, dtype=torch.int32)
+    for i in range(n):
+        total = 0
+        for c in pool_cols:
+            total += int(seg.strref[c, base_row + i].item()) & 0xFFFF
+        out[i] = total
+    return out
+
+
+def pool_gather_ref(payload: bytes, seg, pool_cols, base_row: int, n: int,
+                    row_start: torch.Tensor, pool_base: int) -> None:
+    pool = seg.pool.numpy()
+    for i in range(n):
+        dst = pool_base + int(row_start[i].item())
+        for c in pool_cols:
+            r = int(seg.strref[c, base_row + i].item()) & M64
+            ln = r & 0xFFFF
+            off = r >> 16
+            pool[dst:dst + ln] = np.frombuffer(payload[off:off + ln],
+                                               dtype=np.uint8)
+            seg.strref[c, base_row + i] = S.str_ref_pack(dst, ln)
+            dst += ln
+
+
+# -------------------------------------------------------------- K5 agg
+
+AGG_NVALS = 7  # req, resp, err_c, err_s, rrt_sum, rrt_cnt, rrt_max
+
+
+def agg_app1s_ref(seg, base_row: int, n: int, time_base_s: int,
+                  table: Dict[int, List[int]]) -> None:
+    for i in range(n):
+        row = base_row + i
+        t_s = (int(seg.u64[_U64_IDX["start_time"], row].item()) & M64) // 10**9
+        rel = max(t_s - time_base_s, 0)
+        vtap = int(seg.u32[_U32_IDX["vtap_id"], row].item()) & 0xFFFFFFFF
+        port = int(seg.u32[_U32_IDX["server_port"], row].item()) & 0xFFFFFFFF
+        l7p = int(seg.u8[_U8_IDX["l7_protocol"], row].item())
+        status = int(seg.u8[_U8_IDX["response_status"], row].item())
+        mtype = int(seg.u8[_U8_IDX["msg_type"], row].item())
+        rrt = int(seg.u64[_U64_IDX["rrt"], row].item()) & M64
+        key = (rel << 42) | ((vtap & 0xFFF) << 30) | (l7p << 22) | \
+              ((status & 0xF) << 18) | ((port & 0xFFFF) << 2) | 1
+        acc = table.setdefault(key, [0] * AGG_NVALS)
+        if mtype in (0, 2):
+            acc[0] += 1
+        if mtype in (1, 2):
+            acc[1] += 1
+        if status == 4:
+            acc[2] += 1
+        if status == 3:
+            acc[3] += 1
+        if rrt:
+            acc[4] += rrt
+            acc[5] += 1
+            acc[6] = max(acc[6], rrt)

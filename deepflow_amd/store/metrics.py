@@ -1,0 +1,63 @@
+"""application.1s metric rollup store (K5 output).
+
+GPU mode: open-addressing key/accumulator tensors written by k_agg_app1s.
+CPU mode: plain dict (ops/ref.agg_app1s_ref). Key packing (must match
+dfgpu.hip k_agg_app1s):
+  key = rel_s<<42 | (vtap&0xFFF)<<30 | l7proto<<22 | (status&0xF)<<18
+        | (server_port&0xFFFF)<<2 | 1
+"""
+from __future__ import annotations
+
+from typing import Dict, List
+
+import torch
+
+AGG_FIELDS = ["request", "response", "client_error", "server_error",
+              "rrt_sum", "rrt_count", "rrt_max"]
+AGG_NVALS = len(AGG_FIELDS)
+
+
+def unpack_key(key: int, time_base_s: int) -> Dict[str, int]:
+    return {
+        "time": time_base_s + (key >> 42),
+        "vtap_id": (key >> 30) & 0xFFF,
+        "l7_protocol": (key >> 22) & 0xFF,
+        "response_status": (key >> 18) & 0xF,
+        "server_port": (key >> 2) & 0xFFFF,
+    }
+
+
+class App1sMetrics:
+    def __init__(self, time_base_s: int, capacity_pow2: int = 1 << 20,
+                 device: str = "cpu"):
+        assert capacity_pow2 & (capacity_pow2 - 1) == 0
+        self.time_base_s = time_base_s
+        self.device = device
+        self.capacity = capacity_pow2
+        if device == "cpu":
+            self.table: Dict[int, List[int]] = {}
+        else:
+            dev = torch.device(device)
+            self.tkeys = torch.zeros(capacity_pow2, dtype=torch.int64, device=dev)
+            self.tvals = torch.zeros((capacity_pow2, AGG_NVALS),
+                                     dtype=torch.int64, device=dev)
+
+    def rows(self) -> List[Dict[str, int]]:
+        out = []
+        if self.device == "cpu":
+            items = self.table.items()
+            for key, acc in items:
+                row = unpack_key(key, self.time_base_s)
+                row.update(dict(zip(AGG_FIELDS, acc)))
+                out.append(row)
+        else:
+            mask = self.tkeys != 0
+            keys = self.tkeys[mask].cpu().numpy()
+            vals = self.tvals[mask].cpu().numpy()
+            for key, acc in zip(keys, vals):
+                row = unpack_key(int(key) & ((1 << 64) - 1), self.time_base_s)
+                row.update({f: int(a) for f, a in zip(AGG_FIELDS, acc)})
+                out.append(row)
+        out.sort(key=lambda r: (r["time"], r["vtap_id"], r["server_port"],
+                                r["l7_protocol"], r["response_status"]))
+        return out

@@ -1,0 +1,91 @@
+"""HBM-resident columnar segment for l7_flow_log.
+
+One `L7Segment` is a fixed-capacity SoA block; the decode/join/intern kernels
+write rows in place (no row->column transpose pass: the decoder IS the
+transpose, writing straight into column slices). A `SegmentSet` chains
+segments into the shard's hot window, sized for 288 GB HBM per GPU.
+
+Device-agnostic: tensors live on 'cuda' on a GPU box (written by HIP
+kernels) or 'cpu' in tests (written by ops/ref.py reference ops).
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from . import l7_schema as S
+
+
+class L7Segment:
+    def __init__(self, capacity: int, device: str = "cpu",
+                 pool_capacity: Optional[int] = None):
+        self.capacity = capacity
+        self.device = device
+        dev = torch.device(device)
+        z = lambda shape, dt: torch.zeros(shape, dtype=dt, device=dev)
+        self.u64 = z((S.N_U64, capacity), torch.int64)
+        self.u32 = z((S.N_U32, capacity), torch.int32)
+        self.u8 = z((S.N_U8, capacity), torch.uint8)
+        self.strref = z((S.N_STR, capacity), torch.int64)
+        self.did = torch.full((S.N_DID, capacity), -1, dtype=torch.int32, device=dev)
+        self.kg = z((2 * S.N_KG, capacity), torch.int32)
+        self.attr_ref = z((2 * S.MAX_ATTRS, capacity), torch.int64)
+        self.attr_id = torch.full((2 * S.MAX_ATTRS, capacity), -1,
+                                  dtype=torch.int32, device=dev)
+        self.attr_cnt = z((capacity,), torch.uint8)
+        self.pool = z((pool_capacity or capacity * 96,), torch.uint8)
+        self.pool_len = 0
+        self.n_rows = 0
+
+    def free_rows(self) -> int:
+        return self.capacity - self.n_rows
+
+    def ensure_pool(self, extra: int) -> None:
+        need = self.pool_len + extra
+        if need > self.pool.numel():
+            new_cap = max(need, self.pool.numel() * 2)
+            new_pool = torch.zeros(new_cap, dtype=torch.uint8,
+                                   device=self.pool.device)
+            new_pool[: self.pool_len] = self.pool[: self.pool_len]
+            self.pool = new_pool
+
+    def stored_bytes_per_row(self) -> float:
+        """Bytes/span actually resident (SmartEncoding accounting).
+
+        Counts the fixed-width blocks that survive past ingest (u64/u32/u8,
+        dict ids, KG ids, attr ids + count) plus the variable pool. The
+        transient strref/attr_ref blocks are working state for the batch in
+        flight, not storage, but we count strref's pooled refs (needed to
+        address the pool).
+        """
+        if self.n_rows == 0:
+            return 0.0
+        fixed = (S.N_U64 * 8 + S.N_U32 * 4 + S.N_U8 * 1 + S.N_DID * 4 +
+                 2 * S.N_KG * 4 + 2 * S.MAX_ATTRS * 4 + 1 +
+                 len(S.POOL_COLS) * 8)
+        return fixed + self.pool_len / self.n_rows
+
+
+class SegmentSet:
+    """The shard-local hot window: ordered list of segments."""
+
+    def __init__(self, segment_rows: int, device: str = "cpu"):
+        self.segment_rows = segment_rows
+        self.device = device
+        self.segments: List[L7Segment] = []
+
+    def tail(self, min_free: int) -> L7Segment:
+        if not self.segments or self.segments[-1].free_rows() < min_free:
+            self.segments.append(L7Segment(self.segment_rows, self.device))
+        return self.segments[-1]
+
+    @property
+    def n_rows(self) -> int:
+        return sum(s.n_rows for s in self.segments)
+
+    def total_stored_bytes(self) -> int:
+        total = 0
+        for s in self.segments:
+            total += int(s.stored_bytes_per_row() * s.n_rows)
+        return total
