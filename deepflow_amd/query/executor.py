@@ -1,0 +1,194 @@
+"""Query executor: runs a Plan over the shard's segments.
+
+GPU: k_query_agg / k_query_select (one launch per segment, shared group
+table). CPU: vectorized numpy oracle with identical semantics — used by
+tests as the numerics reference for the GPU kernels and by device='cpu'
+deployments.
+"""
+from __future__ import annotations
+
+from typing import Dict, List
+
+import numpy as np
+import torch
+
+from ..store import l7_schema as S
+from .spec import (Plan, SRC_U64, SRC_U32, SRC_U8, SRC_DID, SRC_KG,
+                   SRC_ATTR_VAL, SRC_TIME_BUCKET, SRC_CONST0,
+                   OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN,
+                   AGGOP_COUNT, AGGOP_SUM, AGGOP_MIN, AGGOP_MAX,
+                   QMAX_KEYS, QMAX_AGGS)
+
+U64MAX = (1 << 64) - 1
+GROUP_CAP = 1 << 20
+
+
+def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
+            n: int) -> np.ndarray:
+    """Vectorized src_value over rows [0, n) as uint64 (CPU oracle)."""
+    def u(t):
+        return t[:, :n][idx].numpy()
+    if family == SRC_U64:
+        return seg.u64[idx, :n].numpy().view(np.uint64)
+    if family == SRC_U32:
+        return seg.u32[idx, :n].numpy().view(np.uint32).astype(np.uint64)
+    if family == SRC_U8:
+        return seg.u8[idx, :n].numpy().astype(np.uint64)
+    if family == SRC_DID:
+        return seg.did[idx, :n].numpy().view(np.uint32).astype(np.uint64)
+    if family == SRC_KG:
+        return seg.kg[idx, :n].numpy().view(np.uint32).astype(np.uint64)
+    if family == SRC_ATTR_VAL:
+        return seg.attr_id[S.MAX_ATTRS + idx, :n].numpy().view(
+            np.uint32).astype(np.uint64)
+    if family == SRC_TIME_BUCKET:
+        t_s = seg.u64[0, :n].numpy().view(np.uint64) // np.uint64(10**9)
+        rel = np.maximum(t_s.astype(np.int64) - time_base_s, 0).astype(np.uint64)
+        if bucket:
+            rel = (rel // np.uint64(bucket)) * np.uint64(bucket)
+        return rel
+    return np.zeros(n, dtype=np.uint64)
+
+
+def _mask_np(seg, plan: Plan, n: int) -> np.ndarray:
+    mask = np.ones(n, dtype=bool)
+    for t in plan.terms:
+        v = _src_np(seg, t.family, t.idx, 0, plan.time_base_s, n)
+        v0 = np.uint64(t.v0 & U64MAX)
+        v1 = np.uint64(t.v1 & U64MAX)
+        if t.op == OP_EQ:
+            mask &= v == v0
+        elif t.op == OP_NE:
+            mask &= v != v0
+        elif t.op == OP_LT:
+            mask &= v < v0
+        elif t.op == OP_LE:
+            mask &= v <= v0
+        elif t.op == OP_GT:
+            mask &= v > v0
+        elif t.op == OP_GE:
+            mask &= v >= v0
+        elif t.op == OP_BETWEEN:
+            mask &= (v >= v0) & (v <= v1)
+    return mask
+
+
+def execute_agg_cpu(plan: Plan, segments) -> List[Dict]:
+    groups: Dict[tuple, List[int]] = {}
+    for seg in segments:
+        n = seg.n_rows
+        if n == 0:
+            continue
+        mask = _mask_np(seg, plan, n)
+        if not mask.any():
+            continue
+        keys = [_src_np(seg, k.family, k.idx, k.bucket, plan.time_base_s, n)[mask]
+                for k in plan.keys]
+        aggvals = []
+        for a in plan.aggs:
+            if a.op == AGGOP_COUNT:
+                aggvals.append(np.ones(int(mask.sum()), dtype=np.uint64))
+            else:
+                aggvals.append(_src_np(seg, a.family, a.idx, 0,
+                                       plan.time_base_s, n)[mask])
+        nk = len(keys)
+        rows = int(mask.sum())
+        key_tup = np.empty((rows, nk), dtype=np.uint64)
+        for j, kv in enumerate(keys):
+            key_tup[:, j] = kv
+        for r in range(rows):
+            kt = tuple(int(x) for x in key_tup[r])
+            acc = groups.get(kt)
+            if acc is None:
+                acc = []
+                for a in plan.aggs:
+                    acc.append(0 if a.op in (AGGOP_COUNT, AGGOP_SUM)
+                               else (U64MAX if a.op == AGGOP_MIN else 0))
+                groups[kt] = acc
+            for ai, a in enumerate(plan.aggs):
+                v = int(aggvals[ai][r])
+                if a.op in (AGGOP_COUNT, AGGOP_SUM):
+                    acc[ai] += v
+                elif a.op == AGGOP_MIN:
+                    acc[ai] = min(acc[ai], v)
+                else:
+                    acc[ai] = max(acc[ai], v)
+    return [{"key": list(k), "agg": list(v)} for k, v in groups.items()]
+
+
+def execute_agg_gpu(plan: Plan, segments, device="cuda") -> List[Dict]:
+    from ..ops import gpu_ops
+    dev = torch.device(device)
+    gkeys = torch.zeros(GROUP_CAP, dtype=torch.int64, device=dev)
+    graw = torch.zeros((GROUP_CAP, QMAX_KEYS), dtype=torch.int64, device=dev)
+    gvals = torch.zeros((GROUP_CAP, QMAX_AGGS), dtype=torch.int64, device=dev)
+    for ai, a in enumerate(plan.aggs):
+        if a.op == AGGOP_MIN:
+            gvals[:, ai] = -1  # 0xFFFF.. as int64
+    spec = plan.to_bytes()
+    for seg in segments:
+        if seg.n_rows == 0:
+            continue
+        gpu_ops.query_agg(seg, spec, 0, seg.n_rows, gkeys, graw, gvals)
+    torch.cuda.synchronize()
+    mask = gkeys != 0
+    raw = graw[mask].cpu().numpy().view(np.uint64)
+    vals = gvals[mask].cpu().numpy().view(np.uint64)
+    out = []
+    nk, na = len(plan.keys), len(plan.aggs)
+    for r in range(raw.shape[0]):
+        out.append({"key": [int(x) for x in raw[r, :nk]],
+                    "agg": [int(x) for x in vals[r, :na]]})
+    return out
+
+
+def execute_select_cpu(plan: Plan, segments, limit: int) -> List[int]:
+    """Returns (segment_idx, row) pairs encoded as global row ids."""
+    out = []
+    for si, seg in enumerate(segments):
+        n = seg.n_rows
+        if n == 0:
+            continue
+        mask = _mask_np(seg, plan, n)
+        rows = np.nonzero(mask)[0]
+        for r in rows:
+            out.append((si, int(r)))
+            if len(out) >= limit:
+                return out
+    return out
+
+
+def execute_select_gpu(plan: Plan, segments, limit: int,
+                       device="cuda") -> List[int]:
+    from ..ops import gpu_ops
+    dev = torch.device(device)
+    out = []
+    spec = plan.to_bytes()
+    for si, seg in enumerate(segments):
+        if seg.n_rows == 0:
+            continue
+        cap = min(limit * 4 + 1024, 1 << 22)
+        out_rows = torch.zeros(cap, dtype=torch.int64, device=dev)
+        out_ctr = torch.zeros(1, dtype=torch.int32, device=dev)
+        gpu_ops.query_select(seg, spec, 0, seg.n_rows, out_rows, out_ctr)
+        torch.cuda.synchronize()
+        cnt = min(int(out_ctr.item()), cap)
+        rows = sorted(out_rows[:cnt].cpu().tolist())
+        for r in rows:
+            out.append((si, int(r)))
+            if len(out) >= limit:
+                return out
+    return out
+
+
+def execute(plan: Plan, segments, device: str = "cpu"):
+    if plan.impossible:
+        return []
+    if plan.select_rows:
+        limit = plan.limit or 100
+        if device == "cpu":
+            return execute_select_cpu(plan, segments, limit)
+        return execute_select_gpu(plan, segments, limit, device)
+    if device == "cpu":
+        return execute_agg_cpu(plan, segments)
+    return execute_agg_gpu(plan, segments, device)

@@ -1,0 +1,107 @@
+"""Tag resolution: DF-SQL tag names -> column sources + hydration metadata.
+
+The moral equivalent of the reference querier's tag translation layer
+(server/querier/engine/clickhouse/tag/translation.go): each queryable tag
+maps to a (family, idx) column source; dict-encoded tags carry their
+SmartEncoding domain so filters can compile string literals to IDs and
+results can be hydrated back to names.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, Optional
+
+from ..store import l7_schema as S
+from . import spec as Q
+
+
+@dataclass
+class TagDef:
+    name: str
+    family: int
+    idx: int
+    # hydration: 'int' | 'dict:<domain>' | 'ip' | 'status' | 'l7proto'
+    hydrate: str = "int"
+    description: str = ""
+
+
+def _u64(name, col, **kw):
+    return TagDef(name, Q.SRC_U64, S.U64_COLS.index(col), **kw)
+
+
+def _u32(name, col, **kw):
+    return TagDef(name, Q.SRC_U32, S.U32_COLS.index(col), **kw)
+
+
+def _u8(name, col, **kw):
+    return TagDef(name, Q.SRC_U8, S.U8_COLS.index(col), **kw)
+
+
+def build_l7_tags() -> Dict[str, TagDef]:
+    tags: Dict[str, TagDef] = {}
+
+    def add(t: TagDef):
+        tags[t.name] = t
+
+    add(_u64("start_time", "start_time"))
+    add(_u64("end_time", "end_time"))
+    add(_u64("flow_id", "flow_id"))
+    add(_u64("response_duration", "rrt"))
+    add(_u64("syscall_trace_id_request", "syscall_trace_id_request"))
+    add(_u64("syscall_trace_id_response", "syscall_trace_id_response"))
+    for name, col in [
+        ("agent_id", "vtap_id"), ("vtap_id", "vtap_id"),
+        ("request_id", "request_id"),
+        ("response_code", "response_code"),
+        ("request_length", "request_length"),
+        ("response_length", "response_length"),
+        ("client_port", "client_port"), ("server_port", "server_port"),
+        ("req_tcp_seq", "req_tcp_seq"), ("resp_tcp_seq", "resp_tcp_seq"),
+        ("captured_request_byte", "captured_request_byte"),
+        ("captured_response_byte", "captured_response_byte"),
+        ("biz_type", "biz_type"),
+    ]:
+        add(_u32(name, col))
+    for side in (0, 1):
+        add(_u32(f"ip4_{side}", f"ip4_{side}", hydrate="ip"))
+        add(_u32(f"l3_epc_id_{side}", f"l3_epc_id_{side}"))
+        add(_u32(f"process_id_{side}", f"process_id_{side}"))
+    add(_u8("tap_side", "tap_side"))
+    add(_u8("tap_type", "tap_type"))
+    add(_u8("protocol", "protocol"))
+    add(_u8("l7_protocol", "l7_protocol", hydrate="l7proto"))
+    add(_u8("type", "msg_type"))
+    add(_u8("response_status", "response_status", hydrate="status"))
+    add(_u8("is_ipv6", "is_ipv6"))
+    # dict-encoded tags
+    for did_idx, (name, _, dom) in enumerate(S.DID_COLS):
+        add(TagDef(name, Q.SRC_DID, did_idx, hydrate=f"dict:{dom}"))
+    # aliases matching reference tag names
+    tags["l7_protocol_str"] = TagDef("l7_protocol_str", Q.SRC_U8,
+                                     S.U8_COLS.index("l7_protocol"),
+                                     hydrate="l7proto")
+    tags["app_service"] = tags["service_name"]
+    # KnowledgeGraph universal tags, client (_0) / server (_1)
+    for side in (0, 1):
+        for j, kname in enumerate(S.KG_COLS):
+            nm = f"{kname}_{side}"
+            add(TagDef(nm, Q.SRC_KG, side * S.N_KG + j))
+    # un-suffixed aliases -> server side (matches reference default for
+    # single-ended metrics tables)
+    for j, kname in enumerate(S.KG_COLS):
+        tags.setdefault(kname, TagDef(kname, Q.SRC_KG, S.N_KG + j))
+    # time pseudo-tag handled by the parser (SRC_TIME_BUCKET)
+    return tags
+
+
+L7_TAGS = build_l7_tags()
+
+# aggregatable metric fields (reference: db_descriptions metrics)
+L7_METRICS: Dict[str, TagDef] = {
+    "response_duration": L7_TAGS["response_duration"],
+    "request_length": L7_TAGS["request_length"],
+    "response_length": L7_TAGS["response_length"],
+    "captured_request_byte": L7_TAGS["captured_request_byte"],
+    "captured_response_byte": L7_TAGS["captured_response_byte"],
+    "log_count": TagDef("log_count", Q.SRC_CONST0, 0),
+}

@@ -1,0 +1,127 @@
+"""DF-SQL engine tests on the CPU pipeline (oracle checked against direct
+pb-truth recomputation)."""
+import pytest
+
+from deepflow_amd.gen import SpanGenConfig
+from deepflow_amd.gen.spans import gen_span_dict, gen_span_payload
+from deepflow_amd.ingest import L7IngestPipeline
+from deepflow_amd.query import QueryEngine
+from deepflow_amd.store.kg import KnowledgeGraphTable, default_platform
+
+N = 300
+CFG = SpanGenConfig(n=N, seed=101, tag_cardinality=40, n_attrs=2,
+                    n_ips=64, n_services=4, n_resources=10)
+
+
+@pytest.fixture(scope="module")
+def eng():
+    kg = KnowledgeGraphTable(capacity_pow2=1 << 12, device="cpu")
+    kg.update(default_platform(CFG))
+    p = L7IngestPipeline(device="cpu", segment_rows=1 << 10, kg=kg,
+                        dict_capacity=1 << 12,
+                        time_base_s=CFG.base_time_ns // 10**9)
+    p.ingest_frame_payload(gen_span_payload(CFG))
+    return QueryEngine(p, device="cpu")
+
+
+def truth():
+    return [gen_span_dict(CFG, i) for i in range(N)]
+
+
+def test_count_all(eng):
+    r = eng.query("SELECT Count(*) AS cnt FROM l7_flow_log")
+    assert r["values"] == [[N]]
+
+
+def test_group_by_domain(eng):
+    r = eng.query(
+        "SELECT request_domain, Count(*) AS cnt FROM l7_flow_log "
+        "GROUP BY request_domain ORDER BY cnt DESC")
+    got = {row[0]: row[1] for row in r["values"]}
+    want = {}
+    for t in truth():
+        want[t["req"]["domain"]] = want.get(t["req"]["domain"], 0) + 1
+    assert got == want
+
+
+def test_where_string_filter(eng):
+    t0 = truth()
+    dom = t0[0]["req"]["domain"]
+    r = eng.query(
+        f"SELECT Count(*) AS cnt FROM l7_flow_log WHERE request_domain = '{dom}'")
+    want = sum(1 for t in t0 if t["req"]["domain"] == dom)
+    assert r["values"] == [[want]]
+
+
+def test_where_unknown_string(eng):
+    r = eng.query(
+        "SELECT Count(*) AS c FROM l7_flow_log WHERE request_domain = 'nope'")
+    assert r["values"] == []
+
+
+def test_avg_and_max_duration(eng):
+    r = eng.query(
+        "SELECT Avg(response_duration) AS a, Max(response_duration) AS m "
+        "FROM l7_flow_log")
+    rrts = [t["base"]["head"]["rrt"] for t in truth()]
+    a, m = r["values"][0]
+    assert m == max(rrts)
+    assert abs(a - sum(rrts) / len(rrts)) < 1e-6
+
+
+def test_status_filter_numeric_and_group(eng):
+    r = eng.query(
+        "SELECT response_status, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY response_status")
+    got = {row[0]: row[1] for row in r["values"]}
+    errs = sum(1 for t in truth() if t["resp"]["status"] == 3)
+    assert got.get("Server Error", 0) == errs
+    assert got.get("Success", 0) == N - errs
+
+
+def test_time_bucket_group(eng):
+    r = eng.query(
+        "SELECT time(60), Count(*) AS c FROM l7_flow_log GROUP BY time(60)")
+    total = sum(row[1] for row in r["values"])
+    assert total == N
+    base = CFG.base_time_ns // 10**9
+    for row in r["values"]:
+        assert (row[0] - base) % 60 == 0
+
+
+def test_select_rows(eng):
+    t0 = truth()
+    fid = t0[5]["base"]["flow_id"]
+    r = eng.query(
+        f"SELECT trace_id, request_domain, response_code FROM l7_flow_log "
+        f"WHERE flow_id = {fid} LIMIT 10")
+    assert len(r["values"]) == 1
+    assert r["values"][0][0] == t0[5]["trace_info"]["trace_id"]
+    assert r["values"][0][1] == t0[5]["req"]["domain"]
+    assert r["values"][0][2] == t0[5]["resp"]["code"]
+
+
+def test_kg_group(eng):
+    r = eng.query(
+        "SELECT service_id_1, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY service_id_1")
+    total = sum(row[1] for row in r["values"])
+    assert total == N
+    # service ids come from the platform table (1 + ip % 256)
+    assert all(row[0] > 0 for row in r["values"])
+
+
+def test_show_tags(eng):
+    r = eng.query("show tags from l7_flow_log")
+    names = {row[0] for row in r["values"]}
+    assert {"request_domain", "pod_id_0", "response_status",
+            "l7_protocol"} <= names
+
+
+def test_time_range_filter(eng):
+    base_s = CFG.base_time_ns // 10**9
+    r = eng.query(
+        f"SELECT Count(*) AS c FROM l7_flow_log WHERE time >= {base_s} "
+        f"AND time <= {base_s + 10}")
+    # spans spaced 1ms apart -> ~all within 1s window plus jitter
+    assert 0 < r["values"][0][0] <= N

@@ -1,0 +1,78 @@
+"""GPU query kernels vs CPU oracle on identical ingested data (-m gpu)."""
+import ctypes as ct
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from deepflow_amd.gen import SpanGenConfig
+from deepflow_amd.gen.spans import gen_span_payload
+from deepflow_amd.ingest import L7IngestPipeline
+from deepflow_amd.query import QueryEngine
+from deepflow_amd.query.spec import QuerySpecC, QTermC, QKeyC, QAggC
+from deepflow_amd.store.kg import KnowledgeGraphTable, default_platform
+
+N = 20000
+CFG = SpanGenConfig(n=N, seed=55, tag_cardinality=300, n_attrs=4,
+                    n_ips=128, n_services=12, n_resources=60)
+
+QUERIES = [
+    "SELECT Count(*) AS cnt FROM l7_flow_log",
+    "SELECT request_domain, Count(*) AS cnt FROM l7_flow_log "
+    "GROUP BY request_domain",
+    "SELECT response_status, Avg(response_duration) AS a, "
+    "Max(response_duration) AS m FROM l7_flow_log GROUP BY response_status",
+    "SELECT time(60), Count(*) AS c FROM l7_flow_log GROUP BY time(60)",
+    "SELECT service_id_1, Sum(response_length) AS b FROM l7_flow_log "
+    "WHERE response_status = 0 GROUP BY service_id_1",
+    "SELECT l7_protocol, server_port, Count(*) AS c FROM l7_flow_log "
+    "WHERE server_port = 8080 GROUP BY l7_protocol, server_port",
+]
+
+
+@pytest.fixture(scope="module")
+def engines():
+    assert torch.cuda.is_available()
+    payload = gen_span_payload(CFG)
+    out = {}
+    for dev in ("cpu", "cuda"):
+        kg = KnowledgeGraphTable(capacity_pow2=1 << 12, device=dev)
+        kg.update(default_platform(CFG))
+        p = L7IngestPipeline(device=dev, segment_rows=1 << 16, kg=kg,
+                            dict_capacity=1 << 16,
+                            time_base_s=CFG.base_time_ns // 10**9)
+        p.ingest_frame_payload(payload)
+        out[dev] = QueryEngine(p, device=dev)
+    torch.cuda.synchronize()
+    return out
+
+
+def test_spec_sizes_match():
+    from deepflow_amd.ops import native
+    lib = native.gpu()
+    a = (ct.c_uint32 * 4)()
+    lib.df_spec_sizes(ct.byref(a, 0), ct.byref(a, 4), ct.byref(a, 8),
+                      ct.byref(a, 12))
+    assert ct.sizeof(QTermC) == a[0]
+    assert ct.sizeof(QKeyC) == a[1]
+    assert ct.sizeof(QAggC) == a[2]
+    assert ct.sizeof(QuerySpecC) == a[3]
+
+
+@pytest.mark.parametrize("sql", QUERIES)
+def test_agg_queries_match(engines, sql):
+    rc = engines["cpu"].query(sql)
+    rg = engines["cuda"].query(sql)
+    assert rc["columns"] == rg["columns"]
+    assert rc["values"] == rg["values"], sql
+
+
+def test_select_rows_match(engines):
+    # limit must exceed the match count so CPU (ordered scan) and GPU
+    # (unordered emit) see the same set
+    sql = ("SELECT trace_id, request_domain, response_code FROM l7_flow_log "
+           "WHERE response_status = 3 LIMIT 2000")
+    rc = engines["cpu"].query(sql)
+    rg = engines["cuda"].query(sql)
+    assert sorted(map(tuple, rc["values"])) == sorted(map(tuple, rg["values"]))
