@@ -38,14 +38,15 @@ def gen_batches(cfg: SpanGenConfig, rank: int, n_batches: int, batch: int):
     out = []
     for b in range(n_batches):
         i0 = (rank * n_batches + b) * batch
-        need = lib.df_gen_spans(ct.byref(c), i0, batch, None, 0)
+        need = lib.df_gen_spans_parallel(ct.byref(c), i0, batch, None, 0,
+                                         None, None)
         buf = np.zeros(int(need), dtype=np.uint8)
         offs = np.zeros(batch, dtype=np.uint32)
         lens = np.zeros(batch, dtype=np.uint32)
-        lib.df_gen_spans_indexed(ct.byref(c), i0, batch,
-                                 buf.ctypes.data_as(ct.c_void_p), need,
-                                 offs.ctypes.data_as(ct.c_void_p),
-                                 lens.ctypes.data_as(ct.c_void_p))
+        lib.df_gen_spans_parallel(ct.byref(c), i0, batch,
+                                  buf.ctypes.data_as(ct.c_void_p), need,
+                                  offs.ctypes.data_as(ct.c_void_p),
+                                  lens.ctypes.data_as(ct.c_void_p))
         out.append((buf, offs, lens))
     return out
 

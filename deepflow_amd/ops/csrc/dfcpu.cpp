@@ -238,6 +238,43 @@ uint64_t df_gen_spans_indexed(const SpanCfg* cfg, uint64_t i0, uint64_t n,
     return b.len;
 }
 
+// OpenMP-parallel generator: pass 1 sizes every record, prefix-sums, pass 2
+// encodes in place. Byte-identical to df_gen_spans. Call with out=NULL to
+// get the total size, then again with the buffer (offs/lens optional).
+uint64_t df_gen_spans_parallel(const SpanCfg* cfg, uint64_t i0, uint64_t n,
+                               uint8_t* out, uint64_t cap,
+                               uint32_t* offs, uint32_t* lens) {
+    static thread_local int dummy = 0; (void)dummy;
+    uint64_t* starts = (uint64_t*)malloc((n + 1) * sizeof(uint64_t));
+    if (!starts) return 0;
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < (int64_t)n; i++) {
+        Buf b{nullptr, 0, 0};
+        encode_span(b, *cfg, i0 + i);
+        starts[i + 1] = b.len + 4;
+    }
+    starts[0] = 0;
+    for (uint64_t i = 0; i < n; i++) starts[i + 1] += starts[i];
+    uint64_t total = starts[n];
+    if (out) {
+#pragma omp parallel for schedule(static)
+        for (int64_t i = 0; i < (int64_t)n; i++) {
+            uint64_t pos = starts[i];
+            uint64_t rec_len = starts[i + 1] - pos - 4;
+            if (pos + 4 + rec_len <= cap) {
+                uint32_t ln = (uint32_t)rec_len;
+                memcpy(out + pos, &ln, 4);
+                Buf b{out + pos + 4, 0, rec_len};
+                encode_span(b, *cfg, i0 + i);
+            }
+            if (offs) offs[i] = (uint32_t)(pos + 4);
+            if (lens) lens[i] = (uint32_t)rec_len;
+        }
+    }
+    free(starts);
+    return total;
+}
+
 // Pre-segmentation scan of a [u32 LE len][pb bytes]* payload.
 // Writes up to max_n (offset, len) pairs; returns record count (may exceed
 // max_n to signal truncation). Offsets point at the pb bytes.

@@ -58,6 +58,10 @@ def cpu() -> ct.CDLL:
         lib.df_gen_spans_indexed.argtypes = [
             ct.POINTER(SpanCfgC), ct.c_uint64, ct.c_uint64, ct.c_void_p,
             ct.c_uint64, ct.c_void_p, ct.c_void_p]
+        lib.df_gen_spans_parallel.restype = ct.c_uint64
+        lib.df_gen_spans_parallel.argtypes = [
+            ct.POINTER(SpanCfgC), ct.c_uint64, ct.c_uint64, ct.c_void_p,
+            ct.c_uint64, ct.c_void_p, ct.c_void_p]
         lib.df_scan_offsets.restype = ct.c_uint64
         lib.df_scan_offsets.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p,
                                         ct.c_void_p, ct.c_uint64]

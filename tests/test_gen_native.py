@@ -73,3 +73,20 @@ def test_gen_chunked_matches_full(cpu_lib):
     a = _gen_native(cpu_lib, cfg, 0, 17)
     b = _gen_native(cpu_lib, cfg, 17, 23)
     assert a + b == full
+
+
+def test_parallel_generator_identical(cpu_lib):
+    import ctypes as ct
+    cfg = SpanGenConfig(n=500, seed=21, tag_cardinality=100)
+    c = native.span_cfg_c(cfg)
+    need = cpu_lib.df_gen_spans_parallel(ct.byref(c), 0, 500, None, 0, None, None)
+    buf = np.zeros(int(need), dtype=np.uint8)
+    offs = np.zeros(500, dtype=np.uint32)
+    lens = np.zeros(500, dtype=np.uint32)
+    cpu_lib.df_gen_spans_parallel(ct.byref(c), 0, 500,
+                                  buf.ctypes.data_as(ct.c_void_p), need,
+                                  offs.ctypes.data_as(ct.c_void_p),
+                                  lens.ctypes.data_as(ct.c_void_p))
+    assert buf.tobytes() == gen_span_payload(cfg)
+    expected = framing.scan_record_offsets(buf.tobytes())
+    assert [(int(o), int(l)) for o, l in zip(offs, lens)] == expected
