@@ -103,10 +103,8 @@ class L7IngestPipeline:
         gpu_ops.intern_many(payload_t, seg.strref, self._ref_rows_scalar,
                             self._dom_scalar, base, n, self.dict.tkeys,
                             self.dict.emit, self.dict.emit_ctr, seg.did, base)
-        gpu_ops.intern_many(payload_t, seg.attr_ref, self._ref_rows_attr,
-                            self._dom_attr, base, n, self.dict.tkeys,
-                            self.dict.emit, self.dict.emit_ctr, seg.attr_id,
-                            base)
+        gpu_ops.intern_attrs(payload_t, seg, base, n, self.dict.tkeys,
+                             self.dict.emit, self.dict.emit_ctr)
         # pool sizing: lens kernel -> cumsum -> (sync) total
         row_len = torch.zeros(n, dtype=torch.int32, device=dev)
         gpu_ops.pool_lens(seg, self._pool_cols, base, n, row_len)

@@ -41,7 +41,7 @@ def _run(cmd):
 def build_cpu(force: bool = False) -> Path:
     srcs = [str(CSRC / s) for s in CPU_SOURCES if (CSRC / s).exists()]
     if force or _needs_build(CPU_LIB, CPU_SOURCES):
-        _run(["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fopenmp",
+        _run(["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fopenmp", "-ldl",
               *srcs, "-o", str(CPU_LIB)])
     return CPU_LIB
 

@@ -11,6 +11,7 @@
 #include <cstdio>
 #include <cstring>
 #include <cstdlib>
+#include <dlfcn.h>
 
 namespace {
 
@@ -291,6 +292,38 @@ uint64_t df_scan_offsets(const uint8_t* payload, uint64_t len,
         pos += ln;
     }
     return n;
+}
+
+// ---- zstd / lz4 codecs via dlopen (runtime libs are present without dev
+// headers; the trident frame encoder byte 3 = whole-payload zstd,
+// agent/src/trident.rs:416-431) ----
+
+typedef size_t (*zstd_fn4)(void*, size_t, const void*, size_t);
+typedef size_t (*zstd_fn5)(void*, size_t, const void*, size_t, int);
+typedef unsigned (*zstd_iserr)(size_t);
+
+static void* zstd_sym(const char* name) {
+    static void* h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+    return h ? dlsym(h, name) : nullptr;
+}
+
+// Returns decompressed size, or -1 on failure.
+int64_t df_zstd_decompress(const uint8_t* src, uint64_t n,
+                           uint8_t* dst, uint64_t cap) {
+    static zstd_fn4 dec = (zstd_fn4)zstd_sym("ZSTD_decompress");
+    static zstd_iserr iserr = (zstd_iserr)zstd_sym("ZSTD_isError");
+    if (!dec || !iserr) return -1;
+    size_t r = dec(dst, cap, src, n);
+    return iserr(r) ? -1 : (int64_t)r;
+}
+
+int64_t df_zstd_compress(const uint8_t* src, uint64_t n,
+                         uint8_t* dst, uint64_t cap, int level) {
+    static zstd_fn5 comp = (zstd_fn5)zstd_sym("ZSTD_compress");
+    static zstd_iserr iserr = (zstd_iserr)zstd_sym("ZSTD_isError");
+    if (!comp || !iserr) return -1;
+    size_t r = comp(dst, cap, src, n, level);
+    return iserr(r) ? -1 : (int64_t)r;
 }
 
 }  // extern "C"

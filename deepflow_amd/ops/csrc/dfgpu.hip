@@ -368,6 +368,62 @@ __global__ void k_kg_probe(const uint32_t* __restrict__ epc0,
 // harvest into the id->string dictionary + flow_tag write path.
 // ----------------------------------------------------------------------
 
+// probe-or-insert: plain loads on the hot (hit) path, atomicCAS only on
+// empty slots. Newly claimed slots are emitted for host harvest.
+DEV uint32_t intern_probe(uint64_t h, uint64_t ref, uint8_t domain,
+                          uint64_t* tkeys, uint32_t cap_mask,
+                          uint64_t* emit, uint32_t* emit_ctr,
+                          uint32_t emit_cap) {
+    uint32_t slot = (uint32_t)(h & cap_mask);
+    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
+        uint64_t cur = tkeys[slot];
+        if (cur == h) return slot;
+        if (cur == EMPTY_KEY) {
+            uint64_t old = atomicCAS((unsigned long long*)&tkeys[slot],
+                                     EMPTY_KEY, h);
+            if (old == EMPTY_KEY) {
+                uint32_t e = atomicAdd(emit_ctr, 1u);
+                if (e < emit_cap) {
+                    emit[(uint64_t)e * 2] = ((uint64_t)domain << 56) | slot;
+                    emit[(uint64_t)e * 2 + 1] = ref;
+                }
+                return slot;
+            }
+            if (old == h) return slot;
+            // lost the race to a different key: fall through, keep probing
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return DICT_ID_INVALID;
+}
+
+// wave-cooperative variant: when every active lane holds the same hash
+// (common for low-cardinality columns: req_type, version, attr names),
+// lane 0 probes once and broadcasts.
+DEV uint32_t intern_probe_wave(uint64_t h, uint64_t ref, uint8_t domain,
+                               bool active,
+                               uint64_t* tkeys, uint32_t cap_mask,
+                               uint64_t* emit, uint32_t* emit_ctr,
+                               uint32_t emit_cap) {
+    uint64_t h0 = __shfl(h, 0);
+    bool uniform = __all(!active || h == h0);
+    if (uniform) {
+        uint32_t slot = DICT_ID_INVALID;
+        // some lane that is active must probe; lowest active lane does
+        uint64_t act_mask = __ballot(active);
+        if (act_mask == 0) return DICT_ID_INVALID;
+        uint32_t leader = (uint32_t)__ffsll((unsigned long long)act_mask) - 1;
+        if ((threadIdx.x & 63) == leader)
+            slot = intern_probe(h, ref, domain, tkeys, cap_mask, emit,
+                                emit_ctr, emit_cap);
+        slot = __shfl((int)slot, leader);
+        return slot;
+    }
+    if (!active) return DICT_ID_INVALID;
+    return intern_probe(h, ref, domain, tkeys, cap_mask, emit, emit_ctr,
+                        emit_cap);
+}
+
 __global__ void k_intern_many(const uint8_t* __restrict__ payload,
                               const uint64_t* __restrict__ refs,  // [*, stride]
                               const uint16_t* __restrict__ ref_rows,  // [C] row in refs per column
@@ -375,7 +431,7 @@ __global__ void k_intern_many(const uint8_t* __restrict__ payload,
                               uint32_t C, uint32_t n,
                               uint64_t ref_stride, uint64_t ref_base_row,
                               uint64_t* __restrict__ tkeys, uint32_t cap_mask,
-                              uint64_t* __restrict__ emit,   // [emit_cap] (dom<<56)|(slot<<32)|ref_idx... see below
+                              uint64_t* __restrict__ emit,
                               uint32_t* __restrict__ emit_ctr, uint32_t emit_cap,
                               uint32_t* __restrict__ out_ids,  // [C, out_stride]
                               uint64_t out_stride, uint64_t out_base_row) {
@@ -387,25 +443,69 @@ __global__ void k_intern_many(const uint8_t* __restrict__ payload,
     uint64_t ref = refs[(uint64_t)ref_rows[c] * ref_stride + ref_base_row + i];
     uint32_t len = STR_REF_LEN(ref);
     uint64_t off = STR_REF_OFF(ref);
-    uint32_t* out = &out_ids[c * out_stride + out_base_row + i];
-    if (len == 0) { *out = DICT_ID_INVALID; return; }
-    uint64_t h = str_hash(payload + off, len, 0x9E3779B97F4A7C15ull * (domains[c] + 1));
-    uint32_t slot = (uint32_t)(h & cap_mask);
-    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
-        uint64_t old = atomicCAS((unsigned long long*)&tkeys[slot], EMPTY_KEY, h);
-        if (old == EMPTY_KEY) {
-            // newly interned: emit (domain, slot, batch-ref) for host harvest
-            uint32_t e = atomicAdd(emit_ctr, 1u);
-            if (e < emit_cap)
-                emit[(uint64_t)e * 2] = ((uint64_t)domains[c] << 56) | slot,
-                emit[(uint64_t)e * 2 + 1] = ref;
-            *out = slot;
-            return;
+    bool active = len != 0;
+    uint64_t h = 0;
+    uint8_t dom = domains[c];
+    if (active)
+        h = str_hash(payload + off, len, 0x9E3779B97F4A7C15ull * (dom + 1));
+    uint32_t slot = intern_probe_wave(h, ref, dom, active, tkeys, cap_mask,
+                                      emit, emit_ctr, emit_cap);
+    // out block is pre-initialized to DICT_ID_INVALID; only write hits
+    if (active)
+        out_ids[c * out_stride + out_base_row + i] = slot;
+    else
+        out_ids[c * out_stride + out_base_row + i] = DICT_ID_INVALID;
+}
+
+// attr interning: one thread per row, looping over the row's attr_cnt
+// (name, value) pairs — avoids launching MAX_ATTRS*2 threads per row when
+// the typical count is ~4. Names are wave-uniform per iteration.
+__global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
+                               const uint64_t* __restrict__ attr_refs,  // [2*MAX, stride]
+                               const uint8_t* __restrict__ attr_cnt,
+                               uint32_t n, uint64_t stride, uint64_t base_row,
+                               uint64_t* __restrict__ tkeys, uint32_t cap_mask,
+                               uint64_t* __restrict__ emit,
+                               uint32_t* __restrict__ emit_ctr, uint32_t emit_cap,
+                               uint32_t* __restrict__ out_ids) {  // [2*MAX, stride]
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    bool in_range = i < n;
+    uint64_t row = base_row + (in_range ? i : 0);
+    uint32_t cnt = in_range ? attr_cnt[row] : 0;
+    // wave-max so all lanes iterate together for the uniform-name fast path
+    uint32_t maxc = cnt;
+    for (int d = 32; d > 0; d >>= 1)
+        maxc = max(maxc, (uint32_t)__shfl_xor((int)maxc, d));
+    for (uint32_t a = 0; a < maxc; a++) {
+        bool act = in_range && a < cnt;
+        // names (wave-uniform in the common schema-stable case)
+        uint64_t nref = act ? attr_refs[(uint64_t)a * stride + row] : 0;
+        uint32_t nlen = STR_REF_LEN(nref);
+        uint64_t h = 0;
+        if (act && nlen)
+            h = str_hash(payload + STR_REF_OFF(nref), nlen,
+                         0x9E3779B97F4A7C15ull * (DICT_DOM_ATTR_NAME + 1));
+        uint32_t slot = intern_probe_wave(h, nref, DICT_DOM_ATTR_NAME,
+                                          act && nlen, tkeys, cap_mask,
+                                          emit, emit_ctr, emit_cap);
+        if (act)
+            out_ids[(uint64_t)a * stride + row] =
+                nlen ? slot : DICT_ID_INVALID;
+        // values (high cardinality -> per-lane probes)
+        uint64_t vref = act
+            ? attr_refs[(uint64_t)(L7_MAX_ATTRS + a) * stride + row] : 0;
+        uint32_t vlen = STR_REF_LEN(vref);
+        if (act) {
+            uint32_t vslot = DICT_ID_INVALID;
+            if (vlen) {
+                uint64_t vh = str_hash(payload + STR_REF_OFF(vref), vlen,
+                    0x9E3779B97F4A7C15ull * (DICT_DOM_ATTR_VALUE + 1));
+                vslot = intern_probe(vh, vref, DICT_DOM_ATTR_VALUE, tkeys,
+                                     cap_mask, emit, emit_ctr, emit_cap);
+            }
+            out_ids[(uint64_t)(L7_MAX_ATTRS + a) * stride + row] = vslot;
         }
-        if (old == h) { *out = slot; return; }
-        slot = (slot + 1) & cap_mask;
     }
-    *out = DICT_ID_INVALID;  // table full
 }
 
 // ----------------------------------------------------------------------
@@ -699,6 +799,21 @@ int df_intern_many(const void* payload, const void* refs, const void* ref_rows,
                        (uint64_t*)tkeys, cap - 1,
                        (uint64_t*)emit, (uint32_t*)emit_ctr, emit_cap,
                        (uint32_t*)out_ids, out_stride, out_base_row);
+    return (int)hipGetLastError();
+}
+
+int df_intern_attrs(const void* payload, const void* attr_refs,
+                    const void* attr_cnt, uint32_t n, uint64_t stride,
+                    uint64_t base_row, void* tkeys, uint32_t cap,
+                    void* emit, void* emit_ctr, uint32_t emit_cap,
+                    void* out_ids, uint64_t stream) {
+    hipLaunchKernelGGL(k_intern_attrs, dim3(grid_for(n)), dim3(BLOCK), 0,
+                       STREAM(stream),
+                       (const uint8_t*)payload, (const uint64_t*)attr_refs,
+                       (const uint8_t*)attr_cnt, n, stride, base_row,
+                       (uint64_t*)tkeys, cap - 1,
+                       (uint64_t*)emit, (uint32_t*)emit_ctr, emit_cap,
+                       (uint32_t*)out_ids);
     return (int)hipGetLastError();
 }
 

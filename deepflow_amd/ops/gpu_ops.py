@@ -70,6 +70,17 @@ def intern_many(payload: torch.Tensor, refs: torch.Tensor,
         "df_intern_many")
 
 
+def intern_attrs(payload: torch.Tensor, seg, base_row: int, n: int,
+                 tkeys: torch.Tensor, emit: torch.Tensor,
+                 emit_ctr: torch.Tensor) -> None:
+    lib = native.gpu()
+    native.check(lib.df_intern_attrs(
+        payload.data_ptr(), seg.attr_ref.data_ptr(), seg.attr_cnt.data_ptr(),
+        n, seg.capacity, base_row, tkeys.data_ptr(), tkeys.numel(),
+        emit.data_ptr(), emit_ctr.data_ptr(), emit.shape[0],
+        seg.attr_id.data_ptr(), _stream()), "df_intern_attrs")
+
+
 def pool_lens(seg, pool_cols: torch.Tensor, base_row: int, n: int,
               row_len: torch.Tensor) -> None:
     lib = native.gpu()

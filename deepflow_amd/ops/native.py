@@ -65,6 +65,12 @@ def cpu() -> ct.CDLL:
         lib.df_scan_offsets.restype = ct.c_uint64
         lib.df_scan_offsets.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p,
                                         ct.c_void_p, ct.c_uint64]
+        lib.df_zstd_decompress.restype = ct.c_int64
+        lib.df_zstd_decompress.argtypes = [ct.c_void_p, ct.c_uint64,
+                                           ct.c_void_p, ct.c_uint64]
+        lib.df_zstd_compress.restype = ct.c_int64
+        lib.df_zstd_compress.argtypes = [ct.c_void_p, ct.c_uint64,
+                                         ct.c_void_p, ct.c_uint64, ct.c_int]
         _cpu_lib = lib
     return _cpu_lib
 
@@ -81,6 +87,9 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_intern_many.restype = ct.c_int
     lib.df_intern_many.argtypes = [p, p, p, p, u32, u32, u64, u64, p, u32,
                                    p, p, u32, p, u64, u64, u64]
+    lib.df_intern_attrs.restype = ct.c_int
+    lib.df_intern_attrs.argtypes = [p, p, p, u32, u64, u64, p, u32,
+                                    p, p, u32, p, u64]
     lib.df_pool_lens.restype = ct.c_int
     lib.df_pool_lens.argtypes = [p, p, u32, u32, u64, u64, p, u64]
     lib.df_pool_gather.restype = ct.c_int

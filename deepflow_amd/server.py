@@ -1,0 +1,86 @@
+"""All-in-one server: receiver + GPU ingest pipelines + query engine + HTTP.
+
+The MI355X-native analog of the reference's single server binary running
+ingester + querier (+ controller-lite) in one process
+(server/cmd/server/main.go:108-117). One process per GPU; rank/shard wiring
+comes from deepflow_amd.parallel.
+"""
+from __future__ import annotations
+
+import threading
+from typing import Optional
+
+from .gen import SpanGenConfig
+from .ingest import L7IngestPipeline
+from .ingest.receiver import Receiver
+from .query import QueryEngine
+from .query.http_api import build_app
+from .store.kg import KnowledgeGraphTable, default_platform
+from .utils.stats import default_registry
+from .wire import framing
+
+
+class DeepflowServer:
+    def __init__(self, device: str = "cpu", tcp_port: int = 0,
+                 segment_rows: int = 1 << 20,
+                 dict_capacity: int = 1 << 20,
+                 time_base_s: int = 1_700_000_000,
+                 platform_cfg: Optional[SpanGenConfig] = None):
+        self.device = device
+        self.kg = KnowledgeGraphTable(device=device)
+        if platform_cfg is not None:
+            self.kg.update(default_platform(platform_cfg))
+        self.l7 = L7IngestPipeline(device=device, segment_rows=segment_rows,
+                                   kg=self.kg, dict_capacity=dict_capacity,
+                                   time_base_s=time_base_s)
+        self.receiver = Receiver(tcp_port=tcp_port, udp_port=0)
+        self.receiver.register(framing.MSG_PROTOCOLLOG, self._on_l7)
+        self._l4 = None  # wired when the L4 pipeline lands
+        self.engine = QueryEngine(self.l7, device=device)
+        self.app = build_app(self.engine, registry=default_registry())
+        self._lock = threading.Lock()
+
+    # ------------------------------------------------------------------
+    def _on_l7(self, hdr, payload) -> None:
+        import ctypes as ct
+        import numpy as np
+        from .ops import native
+        lib = native.cpu()
+        max_n = max(len(payload) // 8, 16)
+        offs = np.zeros(max_n, dtype=np.uint32)
+        lens = np.zeros(max_n, dtype=np.uint32)
+        n = int(lib.df_scan_offsets(payload.ctypes.data_as(ct.c_void_p),
+                                    len(payload),
+                                    offs.ctypes.data_as(ct.c_void_p),
+                                    lens.ctypes.data_as(ct.c_void_p), max_n))
+        with self._lock:
+            self.l7.ingest(payload, offs[:n].copy(), lens[:n].copy())
+
+    # ------------------------------------------------------------------
+    def start(self) -> None:
+        self.receiver.start()
+
+    def stop(self) -> None:
+        self.receiver.stop()
+
+    def serve_http(self, host: str = "127.0.0.1", port: int = 20416) -> None:
+        import uvicorn
+        uvicorn.run(self.app, host=host, port=port, log_level="warning")
+
+
+def main() -> None:
+    import argparse
+    import torch
+    ap = argparse.ArgumentParser(description="deepflow-amd all-in-one server")
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--tcp-port", type=int, default=20033)
+    ap.add_argument("--http-port", type=int, default=20416)
+    args = ap.parse_args()
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    srv = DeepflowServer(device=device, tcp_port=args.tcp_port)
+    srv.start()
+    srv.serve_http(port=args.http_port)
+
+
+if __name__ == "__main__":
+    main()

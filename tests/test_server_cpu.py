@@ -1,0 +1,92 @@
+"""All-in-one server test (BASELINE config #1 shape): agent-side framed
+stream over loopback TCP -> receiver -> pipeline -> HTTP query."""
+import socket
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+from deepflow_amd.gen import SpanGenConfig
+from deepflow_amd.gen.spans import gen_span_payload
+from deepflow_amd.server import DeepflowServer
+from deepflow_amd.wire import framing
+
+N = 1000
+CFG = SpanGenConfig(n=N, seed=9, tag_cardinality=100, n_ips=64,
+                    n_services=8, n_resources=20)
+
+
+@pytest.fixture(scope="module")
+def server():
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 11,
+                         dict_capacity=1 << 13,
+                         time_base_s=CFG.base_time_ns // 10**9,
+                         platform_cfg=CFG)
+    srv.start()
+    yield srv
+    srv.stop()
+
+
+def test_tcp_ingest_and_query(server):
+    payload = gen_span_payload(CFG)
+    hdr = framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG, agent_id=3)
+    frame = framing.encode_frame(hdr, payload)
+    s = socket.create_connection(("127.0.0.1", server.receiver.tcp_port),
+                                 timeout=5)
+    # send in two chunks to exercise reassembly
+    s.sendall(frame[:1000])
+    time.sleep(0.05)
+    s.sendall(frame[1000:])
+    s.close()
+    deadline = time.time() + 20
+    while time.time() < deadline and server.l7.stats.spans_in < N:
+        time.sleep(0.1)
+    assert server.l7.stats.spans_in == N
+
+    client = TestClient(server.app)
+    r = client.post("/v1/query/", data={"sql":
+        "SELECT Count(*) AS cnt FROM l7_flow_log"})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["OPT_STATUS"] == "SUCCESS"
+    assert body["result"]["values"] == [[N]]
+
+    r2 = client.post("/v1/query/", json={
+        "sql": "SELECT request_domain, Count(*) AS c FROM l7_flow_log "
+               "GROUP BY request_domain ORDER BY c DESC LIMIT 3"})
+    assert r2.status_code == 200
+    vals = r2.json()["result"]["values"]
+    assert len(vals) == 3
+    assert all(v[0].startswith("svc-") for v in vals)
+
+    # receiver status accounting
+    st = server.receiver.status[(3, framing.MSG_PROTOCOLLOG)]
+    assert st.frames == 1 and st.bytes == len(frame)
+
+
+def test_zstd_frame(server):
+    import ctypes as ct
+    import numpy as np
+    from deepflow_amd.ops import native
+    before = server.l7.stats.spans_in
+    cfg2 = SpanGenConfig(n=50, seed=77, tag_cardinality=100, n_ips=64)
+    payload = gen_span_payload(cfg2)
+    lib = native.cpu()
+    src = np.frombuffer(payload, dtype=np.uint8)
+    dst = np.zeros(len(payload) * 2 + 1024, dtype=np.uint8)
+    n = lib.df_zstd_compress(src.ctypes.data, len(src), dst.ctypes.data,
+                             len(dst), 3)
+    assert n > 0
+    hdr = framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG, agent_id=4,
+                              encoder=framing.ENCODER_ZSTD)
+    frame = framing.encode_frame(hdr, dst[:n].tobytes())
+    assert server.receiver.handle_frame(frame)
+    assert server.l7.stats.spans_in == before + 50
+
+
+def test_health_and_stats(server):
+    client = TestClient(server.app)
+    assert client.get("/v1/health").json()["status"] == "ok"
+    stats = client.get("/v1/stats").json()
+    names = {s["name"] for s in stats}
+    assert "ingester.receiver" in names
