@@ -1,0 +1,125 @@
+"""L4 (TaggedFlow) ingest pipeline: decode -> KG join -> pool -> network.1s.
+
+Reference counterpart: flow_log L4 Logger + L4FlowLog.Fill
+(server/ingester/flow_log/log_data/l4_flow_log.go) + network metric rollups.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..store import l4_schema as L4
+from ..store import l7_schema as S
+from ..store.segment import SegmentSet, L4Segment
+from ..store.kg import KnowledgeGraphTable
+from ..store.metrics import Net1sMetrics
+from ..utils.stats import Counter
+
+
+@dataclass
+class L4Stats:
+    flows_in: int = 0
+    batches: int = 0
+
+
+class L4IngestPipeline:
+    def __init__(self, device: str = "cpu", segment_rows: int = 1 << 22,
+                 time_base_s: int = 1_700_000_000,
+                 kg: Optional[KnowledgeGraphTable] = None,
+                 counter: Optional[Counter] = None):
+        self.device = device
+        self.segments = SegmentSet(segment_rows, device, cls=L4Segment)
+        self.kg = kg or KnowledgeGraphTable(device=device)
+        self.metrics = Net1sMetrics(time_base_s, device=device)
+        self.time_base_s = time_base_s
+        self.stats = L4Stats()
+        self.counter = counter or Counter("ingester.l4")
+
+    def ingest(self, payload: np.ndarray, offs: np.ndarray,
+               lens: np.ndarray) -> int:
+        n = len(offs)
+        if n == 0:
+            return 0
+        seg = self.segments.tail(n)
+        base = seg.n_rows
+        if self.device == "cpu":
+            self._ingest_cpu(payload, offs, lens, seg, base, n)
+        else:
+            self._ingest_gpu(payload, offs, lens, seg, base, n)
+        seg.n_rows += n
+        self.stats.flows_in += n
+        self.stats.batches += 1
+        self.counter.add("flows_in", n)
+        return n
+
+    def _ingest_gpu(self, payload, offs, lens, seg, base, n) -> None:
+        from ..ops import gpu_ops
+        dev = torch.device(self.device)
+        payload_t = torch.from_numpy(payload).to(dev, non_blocking=True)
+        offs_t = torch.from_numpy(offs.view(np.int32)).to(dev, non_blocking=True)
+        lens_t = torch.from_numpy(lens.view(np.int32)).to(dev, non_blocking=True)
+        gpu_ops.decode_l4(payload_t, offs_t, lens_t, seg, base)
+        epc0 = seg.u32[L4.U32_COLS.index("l3_epc_id_0"), base:base + n]
+        ip0 = seg.u32[L4.U32_COLS.index("ip4_0"), base:base + n]
+        epc1 = seg.u32[L4.U32_COLS.index("l3_epc_id_1"), base:base + n]
+        ip1 = seg.u32[L4.U32_COLS.index("ip4_1"), base:base + n]
+        gpu_ops.kg_probe_cols(epc0, ip0, epc1, ip1, n, self.kg.tkeys,
+                              self.kg.tvals, seg.kg, seg.capacity, base)
+        # pool the request_domain strings
+        refs = seg.strref[0, base:base + n]
+        lens64 = (refs & 0xFFFF).to(torch.int64)
+        cum = torch.cumsum(lens64, 0)
+        total = int(cum[-1].item())
+        seg.ensure_pool(total)
+        if total:
+            starts = cum - lens64
+            pool_cols = torch.zeros(1, dtype=torch.uint8, device=dev)
+            gpu_ops.pool_gather(payload_t, seg, pool_cols, base, n, starts,
+                                seg.pool, seg.pool_len)
+        gpu_ops.agg_net1s(seg, base, n, self.time_base_s,
+                          self.metrics.tkeys, self.metrics.tvals)
+        seg.pool_len += total
+
+    def _ingest_cpu(self, payload, offs, lens, seg, base, n) -> None:
+        from ..ops import ref_l4, ref
+        pbytes = payload.tobytes()
+        ref_l4.decode_l4_ref(pbytes, offs, lens, seg, base)
+        # KG join via the host mirror (independent oracle for the GPU probe)
+        for i in range(n):
+            row = base + i
+            for side, (ec, ic) in enumerate(
+                    [("l3_epc_id_0", "ip4_0"), ("l3_epc_id_1", "ip4_1")]):
+                epc = int(seg.u32[L4.U32_COLS.index(ec), row]) & 0xFFFFFFFF
+                ip = int(seg.u32[L4.U32_COLS.index(ic), row]) & 0xFFFFFFFF
+                info = self.kg.host.get((epc, ip))
+                vals = info.as_list() if info else [0] * S.N_KG
+                for j, v in enumerate(vals):
+                    seg.kg[side * S.N_KG + j, row] = v
+        row_len = torch.zeros(n, dtype=torch.int64)
+        for i in range(n):
+            row_len[i] = int(seg.strref[0, base + i].item()) & 0xFFFF
+        cum = torch.cumsum(row_len, 0)
+        total = int(cum[-1].item()) if n else 0
+        seg.ensure_pool(total)
+        starts = cum - row_len
+        ref.pool_gather_ref(pbytes, seg, [0], base, n, starts, seg.pool_len)
+        ref_l4.agg_net1s_ref(seg, base, n, self.time_base_s,
+                             self.metrics.table)
+        seg.pool_len += total
+
+    def ingest_frame_payload(self, payload: bytes) -> int:
+        from ..ops import native
+        import ctypes as ct
+        arr = np.frombuffer(payload, dtype=np.uint8)
+        max_n = max(len(payload) // 8, 16)
+        offs = np.zeros(max_n, dtype=np.uint32)
+        lens = np.zeros(max_n, dtype=np.uint32)
+        lib = native.cpu()
+        n = int(lib.df_scan_offsets(arr.ctypes.data_as(ct.c_void_p),
+                                    len(payload),
+                                    offs.ctypes.data_as(ct.c_void_p),
+                                    lens.ctypes.data_as(ct.c_void_p), max_n))
+        return self.ingest(arr, offs[:n].copy(), lens[:n].copy())

@@ -303,6 +303,278 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
 }
 
 // ----------------------------------------------------------------------
+// K1b: TaggedFlow (L4 flow log) protobuf decode, thread-per-record.
+// Wire schema: message/flow_log.proto:14-120; columns: l4_layout.h.
+// ----------------------------------------------------------------------
+#include "l4_layout.h"
+
+struct L4Cols {
+    uint64_t* u64c;
+    uint32_t* u32c;
+    uint8_t* u8c;
+    uint64_t* strc;
+    uint64_t stride;
+    uint64_t base_row;
+};
+
+#define L4W64(c, v) cols.u64c[(uint64_t)(c) * cols.stride + row] = (v)
+#define L4W32(c, v) cols.u32c[(uint64_t)(c) * cols.stride + row] = (uint32_t)(v)
+#define L4W8(c, v)  cols.u8c[(uint64_t)(c) * cols.stride + row] = (uint8_t)(v)
+
+__global__ void k_decode_l4(const uint8_t* __restrict__ payload,
+                            const uint32_t* __restrict__ offs,
+                            const uint32_t* __restrict__ lens,
+                            uint32_t n, L4Cols cols) {
+    uint32_t rid = blockIdx.x * blockDim.x + threadIdx.x;
+    if (rid >= n) return;
+    uint64_t row = cols.base_row + rid;
+    uint32_t pos = offs[rid];
+    uint32_t end = pos + lens[rid];
+    // locate flow submessage (TaggedFlow field 1)
+    while (pos < end) {
+        uint64_t key = rd_varint(payload, pos, end);
+        uint32_t num = (uint32_t)(key >> 3), wt = (uint32_t)(key & 7);
+        if (num == 1 && wt == 2) {
+            uint32_t ln = (uint32_t)rd_varint(payload, pos, end);
+            end = pos + ln;  // narrow to Flow
+            break;
+        }
+        skip_field(payload, pos, end, wt);
+    }
+    while (pos < end) {
+        uint64_t key = rd_varint(payload, pos, end);
+        uint32_t num = (uint32_t)(key >> 3), wt = (uint32_t)(key & 7);
+        if (wt == 0) {
+            uint64_t v = rd_varint(payload, pos, end);
+            switch (num) {
+                case 5: L4W64(L4_U64_FLOW_ID, v); break;
+                case 6: L4W64(L4_U64_START_TIME, v); break;
+                case 7: L4W64(L4_U64_END_TIME, v); break;
+                case 8: L4W64(L4_U64_DURATION, v); break;
+                case 10: L4W32(L4_U32_VLAN, v); break;
+                case 11: L4W32(L4_U32_ETH_TYPE, v); break;
+                case 14: L4W8(L4_U8_CLOSE_TYPE, v); break;
+                case 15: L4W8(L4_U8_SIGNAL_SOURCE, v); break;
+                case 16: L4W8(L4_U8_IS_ACTIVE_SERVICE, v); break;
+                case 18: L4W8(L4_U8_IS_NEW_FLOW, v); break;
+                case 19: L4W8(L4_U8_TAP_SIDE, v); break;
+                case 25: L4W8(L4_U8_DIRECTION_SCORE, v); break;
+                default: break;
+            }
+        } else if (wt == 2) {
+            uint32_t ln = (uint32_t)rd_varint(payload, pos, end);
+            uint32_t sub = pos, send = pos + ln;
+            pos = send;
+            switch (num) {
+                case 1: {  // FlowKey
+                    uint32_t p2 = sub;
+                    while (p2 < send) {
+                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
+                        if (w2 == 0) {
+                            uint64_t v = rd_varint(payload, p2, send);
+                            switch (n2) {
+                                case 1: L4W32(L4_U32_VTAP_ID, v); break;
+                                case 2: L4W8(L4_U8_TAP_TYPE, v); break;
+                                case 4: L4W64(L4_U64_MAC_SRC, v); break;
+                                case 5: L4W64(L4_U64_MAC_DST, v); break;
+                                case 6: L4W32(L4_U32_IP4_0, v); break;
+                                case 7: L4W32(L4_U32_IP4_1, v); break;
+                                case 10: L4W32(L4_U32_PORT_SRC, v); break;
+                                case 11: L4W32(L4_U32_PORT_DST, v); break;
+                                case 12: L4W8(L4_U8_PROTOCOL, v); break;
+                                default: break;
+                            }
+                        } else {
+                            skip_field(payload, p2, send, w2);
+                        }
+                    }
+                    break;
+                }
+                case 2: case 3: {  // FlowMetricsPeer src/dst
+                    bool tx = num == 2;
+                    uint32_t p2 = sub;
+                    while (p2 < send) {
+                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
+                        if (w2 == 0) {
+                            uint64_t v = rd_varint(payload, p2, send);
+                            switch (n2) {
+                                case 1: L4W64(tx ? L4_U64_BYTE_TX : L4_U64_BYTE_RX, v); break;
+                                case 2: L4W64(tx ? L4_U64_L3_BYTE_TX : L4_U64_L3_BYTE_RX, v); break;
+                                case 3: L4W64(tx ? L4_U64_L4_BYTE_TX : L4_U64_L4_BYTE_RX, v); break;
+                                case 4: L4W64(tx ? L4_U64_PACKET_TX : L4_U64_PACKET_RX, v); break;
+                                case 5: L4W64(tx ? L4_U64_TOTAL_BYTE_TX : L4_U64_TOTAL_BYTE_RX, v); break;
+                                case 6: L4W64(tx ? L4_U64_TOTAL_PACKET_TX : L4_U64_TOTAL_PACKET_RX, v); break;
+                                case 9: L4W32(tx ? L4_U32_TCP_FLAGS_SRC : L4_U32_TCP_FLAGS_DST, v); break;
+                                case 10: L4W32(tx ? L4_U32_EPC_0 : L4_U32_EPC_1, v); break;
+                                case 20: L4W32(tx ? L4_U32_NAT_REAL_IP_0 : L4_U32_NAT_REAL_IP_1, v); break;
+                                case 21: L4W32(tx ? L4_U32_NAT_REAL_PORT_0 : L4_U32_NAT_REAL_PORT_1, v); break;
+                                case 22: L4W32(tx ? L4_U32_GPID_0 : L4_U32_GPID_1, v); break;
+                                default: break;
+                            }
+                        } else {
+                            skip_field(payload, p2, send, w2);
+                        }
+                    }
+                    break;
+                }
+                case 13: {  // FlowPerfStats
+                    uint32_t p2 = sub;
+                    while (p2 < send) {
+                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
+                        if (w2 == 0) {
+                            uint64_t v = rd_varint(payload, p2, send);
+                            if (n2 == 3) L4W8(L4_U8_L4_PROTOCOL, v);
+                            else if (n2 == 4) L4W8(L4_U8_L7_PROTOCOL, v);
+                        } else if (w2 == 2) {
+                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t s3 = p2, e3 = p2 + l3;
+                            p2 = e3;
+                            if (n2 == 1) {  // TCPPerfStats
+                                uint32_t p3 = s3;
+                                while (p3 < e3) {
+                                    uint64_t k3 = rd_varint(payload, p3, e3);
+                                    uint32_t n3 = (uint32_t)(k3 >> 3), w3 = (uint32_t)(k3 & 7);
+                                    if (w3 == 0) {
+                                        uint64_t v = rd_varint(payload, p3, e3);
+                                        switch (n3) {
+                                            case 3: L4W32(L4_U32_SRT_MAX, v); break;
+                                            case 4: L4W32(L4_U32_ART_MAX, v); break;
+                                            case 5: L4W32(L4_U32_RTT, v); break;
+                                            case 8: L4W32(L4_U32_SRT_SUM, v); break;
+                                            case 9: L4W32(L4_U32_ART_SUM, v); break;
+                                            case 12: L4W32(L4_U32_SRT_COUNT, v); break;
+                                            case 13: L4W32(L4_U32_ART_COUNT, v); break;
+                                            case 16: L4W32(L4_U32_RETRANS_TOTAL, v); break;
+                                            case 17: L4W32(L4_U32_SYN_COUNT, v); break;
+                                            case 18: L4W32(L4_U32_SYNACK_COUNT, v); break;
+                                            case 19: L4W32(L4_U32_CIT_MAX, v); break;
+                                            case 20: L4W32(L4_U32_CIT_SUM, v); break;
+                                            case 21: L4W32(L4_U32_CIT_COUNT, v); break;
+                                            default: break;
+                                        }
+                                    } else if (w3 == 2) {
+                                        uint32_t l4b = (uint32_t)rd_varint(payload, p3, e3);
+                                        uint32_t s4 = p3, e4 = p3 + l4b;
+                                        p3 = e4;
+                                        if (n3 == 14 || n3 == 15) {  // TcpPerfCountsPeer
+                                            bool ptx = n3 == 14;
+                                            uint32_t p4 = s4;
+                                            while (p4 < e4) {
+                                                uint64_t k4 = rd_varint(payload, p4, e4);
+                                                if ((k4 & 7) == 0) {
+                                                    uint64_t v = rd_varint(payload, p4, e4);
+                                                    uint32_t n4 = (uint32_t)(k4 >> 3);
+                                                    if (n4 == 1) L4W32(ptx ? L4_U32_RETRANS_TX : L4_U32_RETRANS_RX, v);
+                                                    else if (n4 == 2) L4W32(ptx ? L4_U32_ZERO_WIN_TX : L4_U32_ZERO_WIN_RX, v);
+                                                    else if (n4 == 3) L4W32(ptx ? L4_U32_OOO_TX : L4_U32_OOO_RX, v);
+                                                } else {
+                                                    uint32_t w4 = (uint32_t)(k4 & 7);
+                                                    skip_field(payload, p4, e4, w4);
+                                                }
+                                            }
+                                        }
+                                    } else {
+                                        skip_field(payload, p3, e3, w3);
+                                    }
+                                }
+                            } else if (n2 == 2) {  // L7PerfStats
+                                uint32_t p3 = s3;
+                                while (p3 < e3) {
+                                    uint64_t k3 = rd_varint(payload, p3, e3);
+                                    if ((k3 & 7) == 0) {
+                                        uint64_t v = rd_varint(payload, p3, e3);
+                                        switch ((uint32_t)(k3 >> 3)) {
+                                            case 1: L4W32(L4_U32_L7_REQUEST, v); break;
+                                            case 2: L4W32(L4_U32_L7_RESPONSE, v); break;
+                                            case 3: L4W32(L4_U32_L7_ERR_CLIENT, v); break;
+                                            case 4: L4W32(L4_U32_L7_ERR_SERVER, v); break;
+                                            case 5: L4W32(L4_U32_L7_ERR_TIMEOUT, v); break;
+                                            case 6: L4W32(L4_U32_L7_RRT_COUNT, v); break;
+                                            case 7: L4W64(L4_U64_L7_RRT_SUM, v); break;
+                                            case 8: L4W32(L4_U32_L7_RRT_MAX, v); break;
+                                            default: break;
+                                        }
+                                    } else {
+                                        uint32_t w3 = (uint32_t)(k3 & 7);
+                                        skip_field(payload, p3, e3, w3);
+                                    }
+                                }
+                            }
+                        } else {
+                            skip_field(payload, p2, send, w2);
+                        }
+                    }
+                    break;
+                }
+                case 26:
+                    cols.strc[(uint64_t)L4_STR_REQUEST_DOMAIN * cols.stride + row] =
+                        STR_REF_PACK(sub, ln);
+                    break;
+                default: break;
+            }
+        } else {
+            skip_field(payload, pos, end, wt);
+        }
+    }
+}
+
+// network.1s rollup from L4 rows (reference: flow_metrics network table).
+// Key: (rel_s << 40) | (vtap & 0xFFF) << 28 | (epc_0 & 0xFFFF) << 12
+//      | (proto << 4) | 1
+enum { NAGG_BYTE_TX = 0, NAGG_BYTE_RX, NAGG_PKT_TX, NAGG_PKT_RX,
+       NAGG_NEW_FLOW, NAGG_CLOSED_FLOW, NAGG_RTT_SUM, NAGG_RTT_CNT,
+       NAGG_RTT_MAX, NAGG_RETRANS, NAGG_NVALS };
+
+__global__ void k_agg_net1s(const L4Cols cols, uint32_t n, uint64_t time_base_s,
+                            uint64_t* __restrict__ tkeys,
+                            unsigned long long* __restrict__ tvals,
+                            uint32_t cap_mask) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t row = cols.base_row + i;
+    uint64_t t_s = cols.u64c[L4_U64_START_TIME * cols.stride + row] / 1000000000ull;
+    uint64_t rel = t_s > time_base_s ? t_s - time_base_s : 0;
+    uint32_t vtap = cols.u32c[L4_U32_VTAP_ID * cols.stride + row];
+    uint32_t epc = cols.u32c[L4_U32_EPC_0 * cols.stride + row];
+    uint8_t proto = cols.u8c[L4_U8_PROTOCOL * cols.stride + row];
+    uint64_t key = (rel << 40) | ((uint64_t)(vtap & 0xFFF) << 28) |
+                   ((uint64_t)(epc & 0xFFFF) << 12) |
+                   ((uint64_t)(proto & 0xFF) << 4) | 1ull;
+    uint32_t slot = (uint32_t)(mix64(key) & cap_mask);
+    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
+        uint64_t cur = tkeys[slot];
+        if (cur == key) break;
+        if (cur == EMPTY_KEY) {
+            uint64_t old = atomicCAS((unsigned long long*)&tkeys[slot],
+                                     EMPTY_KEY, key);
+            if (old == EMPTY_KEY || old == key) break;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    unsigned long long* acc = &tvals[(uint64_t)slot * NAGG_NVALS];
+    atomicAdd(&acc[NAGG_BYTE_TX], cols.u64c[L4_U64_BYTE_TX * cols.stride + row]);
+    atomicAdd(&acc[NAGG_BYTE_RX], cols.u64c[L4_U64_BYTE_RX * cols.stride + row]);
+    atomicAdd(&acc[NAGG_PKT_TX], cols.u64c[L4_U64_PACKET_TX * cols.stride + row]);
+    atomicAdd(&acc[NAGG_PKT_RX], cols.u64c[L4_U64_PACKET_RX * cols.stride + row]);
+    if (cols.u8c[L4_U8_IS_NEW_FLOW * cols.stride + row])
+        atomicAdd(&acc[NAGG_NEW_FLOW], 1ull);
+    if (cols.u8c[L4_U8_CLOSE_TYPE * cols.stride + row])
+        atomicAdd(&acc[NAGG_CLOSED_FLOW], 1ull);
+    uint32_t rtt = cols.u32c[L4_U32_RTT * cols.stride + row];
+    if (rtt) {
+        atomicAdd(&acc[NAGG_RTT_SUM], (unsigned long long)rtt);
+        atomicAdd(&acc[NAGG_RTT_CNT], 1ull);
+        atomicMax(&acc[NAGG_RTT_MAX], (unsigned long long)rtt);
+    }
+    uint64_t retrans = cols.u32c[L4_U32_RETRANS_TX * cols.stride + row] +
+                       cols.u32c[L4_U32_RETRANS_RX * cols.stride + row];
+    if (retrans) atomicAdd(&acc[NAGG_RETRANS], retrans);
+}
+
+// ----------------------------------------------------------------------
 // K2: KnowledgeGraph (epc,ip) -> resource-id join.
 //     Open-addressing table: keys u64 ((epc<<32)|ip), vals KG_VALS_N x u32.
 // ----------------------------------------------------------------------
@@ -762,6 +1034,28 @@ int df_decode_l7(const void* payload, const void* offs, const void* lens,
     hipLaunchKernelGGL(k_decode_l7, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
                        (const uint8_t*)payload, (const uint32_t*)offs,
                        (const uint32_t*)lens, n, cols);
+    return (int)hipGetLastError();
+}
+
+int df_decode_l4(const void* payload, const void* offs, const void* lens,
+                 uint32_t n, void* u64c, void* u32c, void* u8c, void* strc,
+                 uint64_t stride, uint64_t base_row, uint64_t stream) {
+    L4Cols cols{(uint64_t*)u64c, (uint32_t*)u32c, (uint8_t*)u8c,
+                (uint64_t*)strc, stride, base_row};
+    hipLaunchKernelGGL(k_decode_l4, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
+                       (const uint8_t*)payload, (const uint32_t*)offs,
+                       (const uint32_t*)lens, n, cols);
+    return (int)hipGetLastError();
+}
+
+int df_agg_net1s(void* u64c, void* u32c, void* u8c, uint64_t stride,
+                 uint64_t base_row, uint32_t n, uint64_t time_base_s,
+                 void* tkeys, void* tvals, uint32_t cap, uint64_t stream) {
+    L4Cols cols{(uint64_t*)u64c, (uint32_t*)u32c, (uint8_t*)u8c,
+                nullptr, stride, base_row};
+    hipLaunchKernelGGL(k_agg_net1s, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
+                       cols, n, time_base_s, (uint64_t*)tkeys,
+                       (unsigned long long*)tvals, cap - 1);
     return (int)hipGetLastError();
 }
 

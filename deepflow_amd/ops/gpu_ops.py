@@ -28,6 +28,38 @@ def decode_l7(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
         seg.capacity, base_row, _stream()), "df_decode_l7")
 
 
+def decode_l4(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
+              seg, base_row: int) -> None:
+    n = offs.numel()
+    lib = native.gpu()
+    native.check(lib.df_decode_l4(
+        payload.data_ptr(), offs.data_ptr(), lens.data_ptr(), n,
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        seg.strref.data_ptr(), seg.capacity, base_row, _stream()),
+        "df_decode_l4")
+
+
+def agg_net1s(seg, base_row: int, n: int, time_base_s: int,
+              tkeys: torch.Tensor, tvals: torch.Tensor) -> None:
+    lib = native.gpu()
+    native.check(lib.df_agg_net1s(
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        seg.capacity, base_row, n, time_base_s,
+        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(), _stream()),
+        "df_agg_net1s")
+
+
+def kg_probe_cols(epc0, ip0, epc1, ip1, n: int, tkeys: torch.Tensor,
+                  tvals: torch.Tensor, out_kg: torch.Tensor,
+                  stride: int, base_row: int) -> None:
+    """Generic KG probe over explicit column row views."""
+    lib = native.gpu()
+    native.check(lib.df_kg_probe(
+        epc0.data_ptr(), ip0.data_ptr(), epc1.data_ptr(), ip1.data_ptr(), n,
+        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(),
+        out_kg.data_ptr(), stride, base_row, _stream()), "df_kg_probe")
+
+
 def kg_build(keys: torch.Tensor, vals: torch.Tensor, tkeys: torch.Tensor,
              tvals: torch.Tensor) -> None:
     lib = native.gpu()

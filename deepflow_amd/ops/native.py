@@ -80,6 +80,10 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_gpu_ready.restype = ct.c_int
     lib.df_decode_l7.restype = ct.c_int
     lib.df_decode_l7.argtypes = [p, p, p, u32, p, p, p, p, p, p, u64, u64, u64]
+    lib.df_decode_l4.restype = ct.c_int
+    lib.df_decode_l4.argtypes = [p, p, p, u32, p, p, p, p, u64, u64, u64]
+    lib.df_agg_net1s.restype = ct.c_int
+    lib.df_agg_net1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
     lib.df_kg_build.restype = ct.c_int
     lib.df_kg_build.argtypes = [p, p, u32, p, p, u32, u64]
     lib.df_kg_probe.restype = ct.c_int

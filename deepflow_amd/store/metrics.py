@@ -27,6 +27,57 @@ def unpack_key(key: int, time_base_s: int) -> Dict[str, int]:
     }
 
 
+NET_FIELDS = ["byte_tx", "byte_rx", "packet_tx", "packet_rx", "new_flow",
+              "closed_flow", "rtt_sum", "rtt_count", "rtt_max", "retrans"]
+
+
+def unpack_net_key(key: int, time_base_s: int):
+    return {
+        "time": time_base_s + (key >> 40),
+        "vtap_id": (key >> 28) & 0xFFF,
+        "l3_epc_id": (key >> 12) & 0xFFFF,
+        "protocol": (key >> 4) & 0xFF,
+    }
+
+
+class Net1sMetrics:
+    """network.1s rollup (K5b output; reference flow_metrics network table)."""
+
+    def __init__(self, time_base_s: int, capacity_pow2: int = 1 << 20,
+                 device: str = "cpu"):
+        assert capacity_pow2 & (capacity_pow2 - 1) == 0
+        self.time_base_s = time_base_s
+        self.device = device
+        self.capacity = capacity_pow2
+        if device == "cpu":
+            self.table: Dict[int, List[int]] = {}
+        else:
+            dev = torch.device(device)
+            self.tkeys = torch.zeros(capacity_pow2, dtype=torch.int64, device=dev)
+            self.tvals = torch.zeros((capacity_pow2, len(NET_FIELDS)),
+                                     dtype=torch.int64, device=dev)
+
+    def rows(self) -> List[Dict[str, int]]:
+        out = []
+        if self.device == "cpu":
+            for key, acc in self.table.items():
+                row = unpack_net_key(key, self.time_base_s)
+                row.update(dict(zip(NET_FIELDS, acc)))
+                out.append(row)
+        else:
+            mask = self.tkeys != 0
+            keys = self.tkeys[mask].cpu().numpy()
+            vals = self.tvals[mask].cpu().numpy()
+            for key, acc in zip(keys, vals):
+                row = unpack_net_key(int(key) & ((1 << 64) - 1),
+                                     self.time_base_s)
+                row.update({f: int(a) for f, a in zip(NET_FIELDS, acc)})
+                out.append(row)
+        out.sort(key=lambda r: (r["time"], r["vtap_id"], r["l3_epc_id"],
+                                r["protocol"]))
+        return out
+
+
 class App1sMetrics:
     def __init__(self, time_base_s: int, capacity_pow2: int = 1 << 20,
                  device: str = "cpu"):

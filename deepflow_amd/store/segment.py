@@ -67,17 +67,58 @@ class L7Segment:
         return fixed + self.pool_len / self.n_rows
 
 
+class L4Segment:
+    """Columnar segment for l4_flow_log (layout: l4_schema / l4_layout.h)."""
+
+    def __init__(self, capacity: int, device: str = "cpu",
+                 pool_capacity: Optional[int] = None):
+        from . import l4_schema as L4
+        self.capacity = capacity
+        self.device = device
+        dev = torch.device(device)
+        z = lambda shape, dt: torch.zeros(shape, dtype=dt, device=dev)
+        self.u64 = z((L4.N_U64, capacity), torch.int64)
+        self.u32 = z((L4.N_U32, capacity), torch.int32)
+        self.u8 = z((L4.N_U8, capacity), torch.uint8)
+        self.strref = z((L4.N_STR, capacity), torch.int64)
+        self.kg = z((2 * S.N_KG, capacity), torch.int32)
+        self.pool = z((pool_capacity or capacity * 24,), torch.uint8)
+        self.pool_len = 0
+        self.n_rows = 0
+
+    def free_rows(self) -> int:
+        return self.capacity - self.n_rows
+
+    def ensure_pool(self, extra: int) -> None:
+        need = self.pool_len + extra
+        if need > self.pool.numel():
+            new_cap = max(need, self.pool.numel() * 2)
+            new_pool = torch.zeros(new_cap, dtype=torch.uint8,
+                                   device=self.pool.device)
+            new_pool[: self.pool_len] = self.pool[: self.pool_len]
+            self.pool = new_pool
+
+    def stored_bytes_per_row(self) -> float:
+        from . import l4_schema as L4
+        if self.n_rows == 0:
+            return 0.0
+        fixed = (L4.N_U64 * 8 + L4.N_U32 * 4 + L4.N_U8 + 2 * S.N_KG * 4 +
+                 L4.N_STR * 8)
+        return fixed + self.pool_len / self.n_rows
+
+
 class SegmentSet:
     """The shard-local hot window: ordered list of segments."""
 
-    def __init__(self, segment_rows: int, device: str = "cpu"):
+    def __init__(self, segment_rows: int, device: str = "cpu", cls=None):
         self.segment_rows = segment_rows
         self.device = device
-        self.segments: List[L7Segment] = []
+        self.cls = cls or L7Segment
+        self.segments: List = []
 
-    def tail(self, min_free: int) -> L7Segment:
+    def tail(self, min_free: int):
         if not self.segments or self.segments[-1].free_rows() < min_free:
-            self.segments.append(L7Segment(self.segment_rows, self.device))
+            self.segments.append(self.cls(self.segment_rows, self.device))
         return self.segments[-1]
 
     @property

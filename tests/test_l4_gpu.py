@@ -1,0 +1,59 @@
+"""GPU L4 decode/rollup vs CPU oracle (-m gpu)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from deepflow_amd.gen import FlowGenConfig
+from deepflow_amd.gen.flows import gen_flow_payload
+from deepflow_amd.ingest.l4_pipeline import L4IngestPipeline
+from deepflow_amd.store.kg import KnowledgeGraphTable, KgInfo
+
+N = 3000
+CFG = FlowGenConfig(n=N, seed=61, n_ips=64, n_epcs=8)
+
+
+def _mk(device):
+    kg = KnowledgeGraphTable(capacity_pow2=1 << 12, device=device)
+    entries = {}
+    for ipl in range(CFG.n_ips):
+        ip = 0x0A000000 | ipl
+        for epc in range(1, CFG.n_epcs + 1):
+            entries[(epc, ip)] = KgInfo(pod_id=100 + ipl, az_id=1 + ipl % 3)
+    kg.update(entries)
+    p = L4IngestPipeline(device=device, segment_rows=1 << 13, kg=kg,
+                        time_base_s=CFG.base_time_ns // 10**9)
+    p.ingest_frame_payload(gen_flow_payload(CFG))
+    return p
+
+
+@pytest.fixture(scope="module")
+def pipes():
+    assert torch.cuda.is_available()
+    g = _mk("cuda")
+    torch.cuda.synchronize()
+    c = _mk("cpu")
+    return c, g
+
+
+def test_l4_columns_match(pipes):
+    c, g = pipes
+    a, b = c.segments.segments[0], g.segments.segments[0]
+    assert torch.equal(a.u64[:, :N], b.u64[:, :N].cpu())
+    assert torch.equal(a.u32[:, :N], b.u32[:, :N].cpu())
+    assert torch.equal(a.u8[:, :N], b.u8[:, :N].cpu())
+    assert torch.equal(a.kg[:, :N], b.kg[:, :N].cpu())
+
+
+def test_l4_pool_match(pipes):
+    c, g = pipes
+    a, b = c.segments.segments[0], g.segments.segments[0]
+    assert a.pool_len == b.pool_len
+    if a.pool_len:
+        assert torch.equal(a.pool[:a.pool_len], b.pool[:b.pool_len].cpu())
+    assert torch.equal(a.strref[:, :N], b.strref[:, :N].cpu())
+
+
+def test_l4_metrics_match(pipes):
+    c, g = pipes
+    assert c.metrics.rows() == g.metrics.rows()
