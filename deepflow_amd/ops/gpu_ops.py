@@ -141,6 +141,11 @@ def agg_app1s(seg, base_row: int, n: int, time_base_s: int,
         "df_agg_app1s")
 
 
+def _opt_ptr(seg, attr: str) -> int:
+    t = getattr(seg, attr, None)
+    return t.data_ptr() if t is not None else 0
+
+
 def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
               gkeys: torch.Tensor, graw: torch.Tensor,
               gvals: torch.Tensor) -> None:
@@ -149,8 +154,8 @@ def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
     native.check(lib.df_query_agg(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.did.data_ptr(), seg.kg.data_ptr(), seg.attr_id.data_ptr(),
-        seg.attr_cnt.data_ptr(), seg.capacity, seg.n_rows,
+        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
+        _opt_ptr(seg, "attr_cnt"), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         gkeys.data_ptr(), graw.data_ptr(), gvals.data_ptr(), gkeys.numel(),
         _stream()), "df_query_agg")
@@ -163,8 +168,8 @@ def query_select(seg, spec_bytes: bytes, base_row: int, n: int,
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
     native.check(lib.df_query_select(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.did.data_ptr(), seg.kg.data_ptr(), seg.attr_id.data_ptr(),
-        seg.attr_cnt.data_ptr(), seg.capacity, seg.n_rows,
+        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
+        _opt_ptr(seg, "attr_cnt"), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         out_rows.data_ptr(), out_ctr.data_ptr(), out_rows.numel(), _stream()),
         "df_query_select")

@@ -4,10 +4,16 @@ The counterpart of the reference querier's CHEngine
 (server/querier/engine/clickhouse/clickhouse.go) with ClickHouse replaced by
 the in-HBM columnar store: parse -> Plan -> GPU group-by/select kernels ->
 host-side hydration (SmartEncoding id -> name), ORDER BY / LIMIT.
+
+Tables:
+  l7_flow_log, l4_flow_log      -> GPU segment scans (flow_log DB)
+  application / application.1s  -> 1s app rollup rows (flow_metrics DB)
+  network / network.1s          -> 1s net rollup rows
 """
 from __future__ import annotations
 
 import ipaddress
+import re
 from typing import Dict, List, Optional
 
 from ..store import l7_schema as S
@@ -15,18 +21,26 @@ from ..wire.const_enums import L7_PROTOCOL_NAMES
 from . import spec as Q
 from .executor import execute
 from .sql import parse_sql, SqlError
-from .tags import L7_TAGS, L7_METRICS
+from .tags import L7_TAGS, L7_METRICS, L4_TAGS, L4_METRICS, TagDef
 
 STATUS_NAMES = {0: "Success", 1: "Not Exist", 2: "Error", 3: "Server Error",
                 4: "Client Error"}
 
+SRC_ROW = 100  # sentinel family: python row-table field
+
+_FROM_RE = re.compile(r"\bfrom\s+`?([\w.]+)`?", re.IGNORECASE)
+
+
+def _row_tags(fields: List[str]) -> Dict[str, TagDef]:
+    return {f: TagDef(f, SRC_ROW, i) for i, f in enumerate(fields)}
+
 
 class QueryEngine:
-    def __init__(self, pipeline, device: str = "cpu",
+    def __init__(self, pipeline, device: str = "cpu", l4_pipeline=None,
                  remote_hydrator=None):
-        """pipeline: ingest.L7IngestPipeline (owns segments/dict/kg/metrics).
-        remote_hydrator: optional parallel.DictSync for cross-shard names."""
+        """pipeline: ingest.L7IngestPipeline; l4_pipeline optional."""
         self.pipe = pipeline
+        self.l4 = l4_pipeline
         self.device = device
         self.remote = remote_hydrator
 
@@ -35,32 +49,59 @@ class QueryEngine:
         stripped = sql.strip().lower()
         if stripped.startswith("show"):
             return self._show(sql)
-        plan = parse_sql(sql, dictionary=self.pipe.dict,
-                         time_base_s=self.pipe.time_base_s)
-        if plan.select_rows:
-            return self._run_select(plan)
-        return self._run_agg(plan)
+        m = _FROM_RE.search(sql)
+        table = m.group(1).lower() if m else "l7_flow_log"
+        if table in ("l7_flow_log", "l7_flow_log.l7_flow_log"):
+            plan = parse_sql(sql, dictionary=self.pipe.dict,
+                             time_base_s=self.pipe.time_base_s,
+                             tags=L7_TAGS, metrics=L7_METRICS)
+            return self._run_segments(plan, self.pipe.segments.segments,
+                                      L7_TAGS, S.STR_COLS)
+        if table == "l4_flow_log":
+            if self.l4 is None:
+                raise SqlError("l4_flow_log table not enabled")
+            plan = parse_sql(sql, dictionary=None,
+                             time_base_s=self.l4.time_base_s,
+                             tags=L4_TAGS, metrics=L4_METRICS)
+            from ..store import l4_schema as L4S
+            return self._run_segments(plan, self.l4.segments.segments,
+                                      L4_TAGS, L4S.STR_COLS)
+        if table.startswith("application"):
+            rows = self.pipe.metrics.rows()
+            return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
+        if table.startswith("network"):
+            if self.l4 is None:
+                raise SqlError("network table not enabled")
+            rows = self.l4.metrics.rows()
+            return self._run_rows(sql, rows, time_base_s=self.l4.time_base_s)
+        raise SqlError(f"unknown table {table!r}")
 
     # ----------------------------------------------------------- show
     def _show(self, sql: str) -> Dict:
         parts = sql.strip().split()
         what = parts[1].lower() if len(parts) > 1 else ""
+        m = _FROM_RE.search(sql)
+        table = m.group(1).lower() if m else "l7_flow_log"
+        tags = L4_TAGS if table == "l4_flow_log" else L7_TAGS
+        mets = L4_METRICS if table == "l4_flow_log" else L7_METRICS
         if what == "tags":
             cols = ["name", "display_name", "type"]
-            vals = [[n, n, t.hydrate] for n, t in sorted(L7_TAGS.items())]
+            vals = [[n, n, t.hydrate] for n, t in sorted(tags.items())]
             return {"columns": cols, "values": vals}
         if what == "metrics":
             cols = ["name", "display_name", "type"]
-            vals = [[n, n, "counter"] for n in sorted(L7_METRICS)]
+            vals = [[n, n, "counter"] for n in sorted(mets)]
             return {"columns": cols, "values": vals}
         if what == "tables":
             return {"columns": ["name"],
-                    "values": [["l7_flow_log"], ["application.1s"]]}
+                    "values": [["l7_flow_log"], ["l4_flow_log"],
+                               ["application.1s"], ["network.1s"]]}
         raise SqlError(f"unsupported show: {sql!r}")
 
-    # ----------------------------------------------------------- agg
-    def _run_agg(self, plan: Q.Plan) -> Dict:
-        segments = self.pipe.segments.segments
+    # ----------------------------------------------------------- segments
+    def _run_segments(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:
+        if plan.select_rows:
+            return self._run_select(plan, segments, tags, str_cols)
         groups = execute(plan, segments, self.device)
         columns = plan.key_names + plan.agg_names
         rows: List[List] = []
@@ -83,26 +124,28 @@ class QueryEngine:
         return {"columns": columns, "values": rows}
 
     # ----------------------------------------------------------- select
-    def _run_select(self, plan: Q.Plan) -> Dict:
-        segments = self.pipe.segments.segments
+    def _run_select(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:
         hits = execute(plan, segments, self.device)
         cols = plan.select_cols
         if cols == ["*"]:
             cols = ["start_time", "end_time", "flow_id", "l7_protocol",
                     "request_domain", "request_resource", "response_status",
                     "response_code", "response_duration", "trace_id",
-                    "span_id", "service_name"]
+                    "span_id", "service_name"] \
+                if "trace_id" in str_cols else \
+                ["start_time", "end_time", "flow_id", "protocol",
+                 "byte_tx", "byte_rx", "rtt", "close_type"]
         rows = []
         for si, r in hits:
             seg = segments[si]
-            rows.append([self._fetch(seg, r, c) for c in cols])
+            rows.append([self._fetch(seg, r, c, tags, str_cols) for c in cols])
         if plan.limit:
             rows = rows[: plan.limit]
         return {"columns": cols, "values": rows}
 
-    def _fetch(self, seg, row: int, col: str):
-        if col in L7_TAGS:
-            td = L7_TAGS[col]
+    def _fetch(self, seg, row: int, col: str, tags, str_cols):
+        td = tags.get(col)
+        if td is not None:
             fam, idx = td.family, td.idx
             if fam == Q.SRC_U64:
                 return int(seg.u64[idx, row])
@@ -116,8 +159,8 @@ class QueryEngine:
                                      int(seg.did[idx, row]) & 0xFFFFFFFF)
             if fam == Q.SRC_KG:
                 return int(seg.kg[idx, row])
-        if col in S.STR_COLS:
-            sidx = S.STR_COLS.index(col)
+        if col in str_cols:
+            sidx = str_cols.index(col)
             r = int(seg.strref[sidx, row]) & ((1 << 64) - 1)
             off, ln = r >> 16, r & 0xFFFF
             if ln == 0:
@@ -128,6 +171,82 @@ class QueryEngine:
             return bytes(seg.pool[off:off + ln].cpu().numpy()).decode(
                 "utf-8", "replace")
         raise SqlError(f"unknown select column {col!r}")
+
+    # ----------------------------------------------------------- row tables
+    def _run_rows(self, sql: str, rows: List[Dict],
+                  time_base_s: int) -> Dict:
+        fields = list(rows[0].keys()) if rows else \
+            ["time", "vtap_id", "request"]
+        tags = _row_tags(fields)
+        mets = {f: tags[f] for f in fields}
+        mets["log_count"] = TagDef("log_count", Q.SRC_CONST0, 0)
+        plan = parse_sql(sql, dictionary=None, time_base_s=time_base_s,
+                         tags=tags, metrics=mets)
+        if plan.impossible:
+            return {"columns": [], "values": []}
+
+        def val(r: Dict, fam: int, idx: int, bucket: int = 0):
+            if fam == SRC_ROW:
+                return r[fields[idx]]
+            if fam == Q.SRC_TIME_BUCKET:
+                rel = r.get("time", 0) - time_base_s
+                if bucket:
+                    rel = (rel // bucket) * bucket
+                return time_base_s + rel
+            if fam == Q.SRC_U64 and idx == 0:  # `time >= n` filter shape
+                return r.get("time", 0) * 10**9
+            return 0
+
+        OPS = {Q.OP_EQ: lambda a, b: a == b, Q.OP_NE: lambda a, b: a != b,
+               Q.OP_LT: lambda a, b: a < b, Q.OP_LE: lambda a, b: a <= b,
+               Q.OP_GT: lambda a, b: a > b, Q.OP_GE: lambda a, b: a >= b}
+        filtered = []
+        for r in rows:
+            ok = True
+            for t in plan.terms:
+                v = val(r, t.family, t.idx)
+                if not OPS[t.op](v, t.v0):
+                    ok = False
+                    break
+            if ok:
+                filtered.append(r)
+        if plan.select_rows:
+            cols = plan.select_cols
+            if cols == ["*"]:
+                cols = fields
+            out = [[r.get(c) for c in cols] for r in filtered]
+            if plan.limit:
+                out = out[: plan.limit]
+            return {"columns": cols, "values": out}
+        groups: Dict[tuple, list] = {}
+        for r in filtered:
+            key = tuple(val(r, k.family, k.idx, k.bucket) for k in plan.keys)
+            acc = groups.setdefault(key, [None] * len(plan.aggs))
+            for ai, a in enumerate(plan.aggs):
+                v = 1 if a.op == Q.AGGOP_COUNT else val(r, a.family, a.idx)
+                cur = acc[ai]
+                if a.op in (Q.AGGOP_COUNT, Q.AGGOP_SUM):
+                    acc[ai] = (cur or 0) + v
+                elif a.op == Q.AGGOP_MIN:
+                    acc[ai] = v if cur is None else min(cur, v)
+                else:
+                    acc[ai] = v if cur is None else max(cur, v)
+        columns = plan.key_names + plan.agg_names
+        out = []
+        for key, acc in groups.items():
+            row = list(key)
+            ai = 0
+            for meta in plan.agg_meta:
+                if meta["op"] == "avg":
+                    s_, c_ = acc[ai] or 0, acc[ai + 1] or 0
+                    ai += 2
+                    row.append(s_ / c_ if c_ else None)
+                else:
+                    row.append(acc[ai])
+                    ai += 1
+            out.append(row)
+        out = self._order_limit(plan, columns, out)
+        return {"columns": columns, "values": out}
 
     # ----------------------------------------------------------- hydrate
     def _hydrate(self, how: str, v: int):
@@ -156,7 +275,8 @@ class QueryEngine:
                               reverse=desc)
         else:
             rows.sort(key=lambda r: tuple(
-                (x is None, x) for x in r[: len(plan.key_names)]))
+                (x is None, str(type(x)), x) for x in
+                r[: len(plan.key_names)]))
         if plan.limit:
             rows = rows[: plan.limit]
         return rows

@@ -79,6 +79,7 @@ class Agg:
 
 @dataclass
 class Plan:
+    table: str = "l7_flow_log"
     terms: List[Term] = field(default_factory=list)
     keys: List[Key] = field(default_factory=list)
     aggs: List[Agg] = field(default_factory=list)

@@ -84,16 +84,20 @@ class Parser:
         return t is not None and t[0] == "id" and t[1].lower() == kw
 
 
-def _resolve_tag(name: str) -> TagDef:
-    t = L7_TAGS.get(name)
+def _resolve_tag(name: str, tags) -> TagDef:
+    t = tags.get(name)
     if t is None:
         raise SqlError(f"unknown tag {name!r}")
     return t
 
 
-def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
-    """Parse DF-SQL for l7_flow_log into a Plan. `dictionary` compiles
-    string literals on dict tags to IDs (None -> impossible filters)."""
+def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
+              tags=None, metrics=None) -> Q.Plan:
+    """Parse DF-SQL into a Plan against a table's tag map (default:
+    l7_flow_log). `dictionary` compiles string literals on dict tags to
+    SmartEncoding IDs (None -> impossible filters)."""
+    tags = tags if tags is not None else L7_TAGS
+    metrics = metrics if metrics is not None else L7_METRICS
     p = Parser(tokenize(sql))
     plan = Q.Plan(time_base_s=time_base_s)
     p.expect_kw("select")
@@ -144,9 +148,7 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
         break
 
     p.expect_kw("from")
-    table = p.next()[1]
-    plan.key_meta.append({"table": table})  # stashed for the engine
-    plan.key_meta.pop()
+    plan.table = p.next()[1]
 
     # WHERE
     if p.kw_is("where"):
@@ -159,7 +161,7 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
             if op_t[0] != "op" or op_t[1] not in Q.OP_BY_NAME:
                 raise SqlError(f"bad operator {op_t!r}")
             lit = p.next()
-            _add_term(plan, name[1], op_t[1], lit, dictionary)
+            _add_term(plan, name[1], op_t[1], lit, dictionary, tags)
             if p.kw_is("and"):
                 p.next()
                 continue
@@ -187,7 +189,7 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
                 plan.key_names.append("time")
                 plan.key_meta.append({"hydrate": "time"})
             else:
-                td = _resolve_tag(t[1])
+                td = _resolve_tag(t[1], tags)
                 plan.keys.append(Q.Key(td.family, td.idx))
                 plan.key_names.append(t[1])
                 plan.key_meta.append({"hydrate": td.hydrate})
@@ -227,7 +229,7 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
             if kind == "tag":
                 if arg not in group_names and arg != "time":
                     # selecting a non-grouped tag in agg query: treat as group
-                    td = _resolve_tag(arg)
+                    td = _resolve_tag(arg, tags)
                     plan.keys.append(Q.Key(td.family, td.idx))
                     plan.key_names.append(arg)
                     plan.key_meta.append({"hydrate": td.hydrate})
@@ -239,13 +241,13 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
                 plan.agg_names.append(alias)
                 plan.agg_meta.append({"op": "count"})
             elif func == "avg":
-                md = L7_METRICS.get(arg) or _resolve_tag(arg)
+                md = metrics.get(arg) or _resolve_tag(arg, tags)
                 plan.aggs.append(Q.Agg(Q.AGGOP_SUM, md.family, md.idx))
                 plan.aggs.append(Q.Agg(Q.AGGOP_COUNT))
                 plan.agg_names.append(alias)
                 plan.agg_meta.append({"op": "avg"})
             else:
-                md = L7_METRICS.get(arg) or _resolve_tag(arg)
+                md = metrics.get(arg) or _resolve_tag(arg, tags)
                 op = AGG_FUNCS[func]
                 plan.aggs.append(Q.Agg(op, md.family, md.idx))
                 plan.agg_names.append(alias)
@@ -259,13 +261,14 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0) -> Q.Plan:
     return plan
 
 
-def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary) -> None:
+def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
+              tags) -> None:
     if name.lower() == "time":
         # time in epoch seconds against start_time (ns)
         v = int(lit[1]) * 10**9
         plan.terms.append(Q.Term(Q.SRC_U64, 0, Q.OP_BY_NAME[op], v))
         return
-    td = _resolve_tag(name)
+    td = _resolve_tag(name, tags)
     if lit[0] == "num":
         v = int(lit[1]) if "." not in lit[1] else int(float(lit[1]))
         plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v))

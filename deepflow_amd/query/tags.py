@@ -105,3 +105,37 @@ L7_METRICS: Dict[str, TagDef] = {
     "captured_response_byte": L7_TAGS["captured_response_byte"],
     "log_count": TagDef("log_count", Q.SRC_CONST0, 0),
 }
+
+
+def build_l4_tags() -> Dict[str, TagDef]:
+    from ..store import l4_schema as L4
+    tags: Dict[str, TagDef] = {}
+    for i, c in enumerate(L4.U64_COLS):
+        tags[c] = TagDef(c, Q.SRC_U64, i)
+    for i, c in enumerate(L4.U32_COLS):
+        hyd = "ip" if c.startswith("ip4") or c.startswith("nat_real_ip") \
+            else "int"
+        tags[c] = TagDef(c, Q.SRC_U32, i, hydrate=hyd)
+    for i, c in enumerate(L4.U8_COLS):
+        hyd = "l7proto" if c == "l7_protocol" else "int"
+        tags[c] = TagDef(c, Q.SRC_U8, i, hydrate=hyd)
+    for side in (0, 1):
+        for j, kname in enumerate(S.KG_COLS):
+            nm = f"{kname}_{side}"
+            tags[nm] = TagDef(nm, Q.SRC_KG, side * S.N_KG + j)
+    tags["agent_id"] = tags["vtap_id"]
+    return tags
+
+
+L4_TAGS = build_l4_tags()
+
+L4_METRICS: Dict[str, TagDef] = {
+    name: L4_TAGS[name] for name in [
+        "byte_tx", "byte_rx", "packet_tx", "packet_rx", "total_byte_tx",
+        "total_byte_rx", "rtt", "srt_sum", "srt_count", "srt_max",
+        "art_sum", "art_count", "art_max", "retrans_tx", "retrans_rx",
+        "l7_request", "l7_response", "l7_rrt_sum", "l7_rrt_count",
+        "duration",
+    ]
+}
+L4_METRICS["log_count"] = TagDef("log_count", Q.SRC_CONST0, 0)

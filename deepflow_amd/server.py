@@ -33,10 +33,13 @@ class DeepflowServer:
         self.l7 = L7IngestPipeline(device=device, segment_rows=segment_rows,
                                    kg=self.kg, dict_capacity=dict_capacity,
                                    time_base_s=time_base_s)
+        from .ingest.l4_pipeline import L4IngestPipeline
+        self.l4 = L4IngestPipeline(device=device, segment_rows=segment_rows,
+                                   kg=self.kg, time_base_s=time_base_s)
         self.receiver = Receiver(tcp_port=tcp_port, udp_port=0)
         self.receiver.register(framing.MSG_PROTOCOLLOG, self._on_l7)
-        self._l4 = None  # wired when the L4 pipeline lands
-        self.engine = QueryEngine(self.l7, device=device)
+        self.receiver.register(framing.MSG_TAGGEDFLOW, self._on_l4)
+        self.engine = QueryEngine(self.l7, device=device, l4_pipeline=self.l4)
         self.app = build_app(self.engine, registry=default_registry())
         self._lock = threading.Lock()
 
@@ -55,6 +58,21 @@ class DeepflowServer:
                                     lens.ctypes.data_as(ct.c_void_p), max_n))
         with self._lock:
             self.l7.ingest(payload, offs[:n].copy(), lens[:n].copy())
+
+    def _on_l4(self, hdr, payload) -> None:
+        import ctypes as ct
+        import numpy as np
+        from .ops import native
+        lib = native.cpu()
+        max_n = max(len(payload) // 8, 16)
+        offs = np.zeros(max_n, dtype=np.uint32)
+        lens = np.zeros(max_n, dtype=np.uint32)
+        n = int(lib.df_scan_offsets(payload.ctypes.data_as(ct.c_void_p),
+                                    len(payload),
+                                    offs.ctypes.data_as(ct.c_void_p),
+                                    lens.ctypes.data_as(ct.c_void_p), max_n))
+        with self._lock:
+            self.l4.ingest(payload, offs[:n].copy(), lens[:n].copy())
 
     # ------------------------------------------------------------------
     def start(self) -> None:

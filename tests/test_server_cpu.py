@@ -90,3 +90,41 @@ def test_health_and_stats(server):
     stats = client.get("/v1/stats").json()
     names = {s["name"] for s in stats}
     assert "ingester.receiver" in names
+
+
+def test_l4_ingest_and_query(server):
+    from deepflow_amd.gen import FlowGenConfig
+    from deepflow_amd.gen.flows import gen_flow_payload, gen_flow_dict
+    fcfg = FlowGenConfig(n=80, seed=19, n_ips=32, n_epcs=8)
+    payload = gen_flow_payload(fcfg)
+    hdr = framing.FrameHeader(msg_type=framing.MSG_TAGGEDFLOW, agent_id=5)
+    assert server.receiver.handle_frame(framing.encode_frame(hdr, payload))
+    assert server.l4.stats.flows_in == 80
+    client = TestClient(server.app)
+    r = client.post("/v1/query/", json={
+        "sql": "SELECT Count(*) AS c, Sum(byte_tx) AS b FROM l4_flow_log"})
+    body = r.json()
+    assert body["OPT_STATUS"] == "SUCCESS", body
+    want_b = sum(gen_flow_dict(fcfg, i)["flow"]["metrics_peer_src"]["byte_count"]
+                 for i in range(80))
+    assert body["result"]["values"] == [[80, want_b]]
+    # select with string column
+    r2 = client.post("/v1/query/", json={
+        "sql": "SELECT flow_id, server_port, close_type FROM l4_flow_log "
+               "WHERE protocol = 6 LIMIT 5"})
+    assert len(r2.json()["result"]["values"]) == 5
+
+
+def test_metrics_tables(server):
+    client = TestClient(server.app)
+    r = client.post("/v1/query/", json={
+        "sql": "SELECT time(60), Sum(request) AS req FROM application "
+               "GROUP BY time(60)"})
+    body = r.json()
+    assert body["OPT_STATUS"] == "SUCCESS", body
+    total = sum(v[1] for v in body["result"]["values"])
+    assert total == server.l7.stats.spans_in
+    r2 = client.post("/v1/query/", json={
+        "sql": "SELECT Sum(byte_tx) AS b FROM network"})
+    assert r2.json()["OPT_STATUS"] == "SUCCESS"
+    assert r2.json()["result"]["values"][0][0] > 0
