@@ -30,24 +30,30 @@ from deepflow_amd.ops import native
 from deepflow_amd.store.kg import KnowledgeGraphTable, default_platform
 
 
-def gen_batches(cfg: SpanGenConfig, rank: int, n_batches: int, batch: int):
+def gen_batches(cfg: SpanGenConfig, rank: int, n_batches: int, batch: int,
+                pinned: bool):
     """Pre-generate n_batches distinct payloads (disjoint span index ranges
-    per rank) with the native generator; returns [(payload, offs, lens)]."""
+    per rank) with the native generator, straight into pinned host memory
+    when a GPU is present; returns [(payload_np, offs_np, lens_np,
+    payload_t, offs_t, lens_t)]."""
     lib = native.cpu()
     c = native.span_cfg_c(cfg)
     out = []
     for b in range(n_batches):
         i0 = (rank * n_batches + b) * batch
-        need = lib.df_gen_spans_parallel(ct.byref(c), i0, batch, None, 0,
-                                         None, None)
-        buf = np.zeros(int(need), dtype=np.uint8)
-        offs = np.zeros(batch, dtype=np.uint32)
-        lens = np.zeros(batch, dtype=np.uint32)
+        need = int(lib.df_gen_spans_parallel(ct.byref(c), i0, batch, None, 0,
+                                             None, None))
+        pay_t = torch.empty(need, dtype=torch.uint8, pin_memory=pinned)
+        offs_t = torch.empty(batch, dtype=torch.int32, pin_memory=pinned)
+        lens_t = torch.empty(batch, dtype=torch.int32, pin_memory=pinned)
+        buf = pay_t.numpy()
+        offs = offs_t.numpy().view(np.uint32)
+        lens = lens_t.numpy().view(np.uint32)
         lib.df_gen_spans_parallel(ct.byref(c), i0, batch,
                                   buf.ctypes.data_as(ct.c_void_p), need,
                                   offs.ctypes.data_as(ct.c_void_p),
                                   lens.ctypes.data_as(ct.c_void_p))
-        out.append((buf, offs, lens))
+        out.append((buf, offs, lens, pay_t, offs_t, lens_t))
     return out
 
 
@@ -81,7 +87,8 @@ def main() -> None:
                         n_ips=4096, n_services=256, n_resources=4096,
                         n_attrs=4)
     n_distinct = min(args.steps + args.warmup, 4)
-    batches = gen_batches(cfg, rank, n_distinct, args.batch)
+    batches = gen_batches(cfg, rank, n_distinct, args.batch,
+                          pinned=device == "cuda")
 
     kg = KnowledgeGraphTable(capacity_pow2=1 << 14, device=device)
     kg.update(default_platform(cfg))
@@ -94,9 +101,32 @@ def main() -> None:
         from deepflow_amd.parallel.dict_sync import DictSync
         dict_sync = DictSync(pipe.dict)
 
+    # H2D prefetch pipeline: copy batch i+1 on a side stream while batch i's
+    # kernels run on the main stream.
+    copy_stream = torch.cuda.Stream() if device == "cuda" else None
+    pending = {}
+
+    def prefetch(i: int) -> None:
+        buf, offs, lens, pay_t, offs_t, lens_t = batches[i % n_distinct]
+        with torch.cuda.stream(copy_stream):
+            dev_batch = (pay_t.to("cuda", non_blocking=True),
+                         offs_t.to("cuda", non_blocking=True),
+                         lens_t.to("cuda", non_blocking=True))
+            ev = torch.cuda.Event()
+            ev.record(copy_stream)
+        pending[i] = (dev_batch, ev, buf)
+
     def step(i: int) -> None:
-        payload, offs, lens = batches[i % n_distinct]
-        pipe.ingest(payload, offs, lens)
+        if device == "cuda":
+            if i not in pending:
+                prefetch(i)
+            dev_batch, ev, host_payload = pending.pop(i)
+            torch.cuda.current_stream().wait_event(ev)
+            prefetch(i + 1)
+            pipe.ingest_device(*dev_batch, host_payload)
+        else:
+            payload, offs, lens = batches[i % n_distinct][:3]
+            pipe.ingest(payload, offs, lens)
         if dict_sync is not None:
             dict_sync.sync_step()
 

@@ -90,14 +90,34 @@ class L7IngestPipeline:
         return n
 
     # ------------------------------------------------------------------
+    def ingest_device(self, payload_t: torch.Tensor, offs_t: torch.Tensor,
+                      lens_t: torch.Tensor, payload_host) -> int:
+        """GPU fast path: batch tensors already on-device (benchmarks /
+        receivers prefetch H2D on a side stream). payload_host is the same
+        bytes on the host (numpy), used only for dictionary harvest."""
+        n = offs_t.numel()
+        if n == 0:
+            return 0
+        seg = self.segments.tail(n)
+        base = seg.n_rows
+        self._run_gpu(payload_t, offs_t, lens_t, payload_host, seg, base, n)
+        seg.n_rows += n
+        self.stats.spans_in += n
+        self.stats.batches += 1
+        self.counter.add("spans_in", n)
+        return n
+
     def _ingest_gpu(self, payload, offs, lens, seg: L7Segment, base: int,
                     n: int) -> None:
-        from ..ops import gpu_ops
         dev = torch.device(self.device)
         payload_t = torch.from_numpy(payload).to(dev, non_blocking=True)
         offs_t = torch.from_numpy(offs.view(np.int32)).to(dev, non_blocking=True)
         lens_t = torch.from_numpy(lens.view(np.int32)).to(dev, non_blocking=True)
+        self._run_gpu(payload_t, offs_t, lens_t, payload, seg, base, n)
 
+    def _run_gpu(self, payload_t, offs_t, lens_t, payload_host,
+                 seg: L7Segment, base: int, n: int) -> None:
+        from ..ops import gpu_ops
         gpu_ops.decode_l7(payload_t, offs_t, lens_t, seg, base)
         gpu_ops.kg_probe(seg, base, n, self.kg.tkeys, self.kg.tvals)
         gpu_ops.intern_many(payload_t, seg.strref, self._ref_rows_scalar,
@@ -116,7 +136,7 @@ class L7IngestPipeline:
                             row_start, seg.pool, seg.pool_len)
         gpu_ops.agg_app1s(seg, base, n, self.time_base_s,
                           self.metrics.tkeys, self.metrics.tvals)
-        new = self.dict.harvest(payload)  # syncs emit buffer
+        new = self.dict.harvest(payload_host)  # syncs emit buffer
         seg.pool_len += total
         self.stats.dict_new += new
         self.stats.pool_bytes += total
