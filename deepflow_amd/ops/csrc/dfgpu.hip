@@ -874,8 +874,10 @@ __global__ void k_agg_app1s(const L7Cols cols, uint32_t n, uint64_t time_base_s,
 // key/agg source families
 enum {
     SRC_U64 = 0, SRC_U32, SRC_U8, SRC_DID, SRC_KG, SRC_ATTR_VAL,
-    SRC_TIME_BUCKET, SRC_CONST0,
+    SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
 };
+// seed for SRC_STR_HASH terms (host twin: store/dictionary.py STR_FILTER_SEED)
+#define STR_FILTER_SEED 0x5157A15E5EEDull
 // filter ops
 enum { OP_EQ = 0, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN };
 // agg ops
@@ -905,6 +907,8 @@ struct SegView {
     const uint32_t* kgc;     // [2*KG_VALS_N, stride]
     const uint32_t* attrid;  // [2*L7_MAX_ATTRS, stride] interned attr name/val ids
     const uint8_t* attr_cnt;
+    const uint64_t* strc;    // [N_STR, stride] pool-relative refs
+    const uint8_t* pool;     // segment string pool
     uint64_t stride;
     uint64_t n_rows;
 };
@@ -926,6 +930,12 @@ DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
             uint64_t t_s = s.u64c[L7_U64_START_TIME * s.stride + row] / 1000000000ull;
             uint64_t rel = t_s > time_base_s ? t_s - time_base_s : 0;
             return bucket ? (rel / bucket) * bucket : rel;
+        }
+        case SRC_STR_HASH: {
+            uint64_t r = s.strc[(uint64_t)idx * s.stride + row];
+            uint32_t len = STR_REF_LEN(r);
+            if (len == 0) return 0;
+            return str_hash(s.pool + STR_REF_OFF(r), len, STR_FILTER_SEED);
         }
         default: return 0;
     }
@@ -1143,14 +1153,16 @@ int df_agg_app1s(void* u64c, void* u32c, void* u8c, uint64_t stride,
 
 int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
                  const void* didc, const void* kgc, const void* attrid,
-                 const void* attr_cnt, uint64_t stride, uint64_t n_rows,
+                 const void* attr_cnt, const void* strc, const void* pool,
+                 uint64_t stride, uint64_t n_rows,
                  const void* spec,  // QuerySpec, host-built bytes
                  uint32_t n, uint64_t base_row,
                  void* gkeys, void* graw, void* gvals, uint32_t cap,
                  uint64_t stream) {
     SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
               (const uint32_t*)didc, (const uint32_t*)kgc, (const uint32_t*)attrid,
-              (const uint8_t*)attr_cnt, stride, n_rows};
+              (const uint8_t*)attr_cnt, (const uint64_t*)strc,
+              (const uint8_t*)pool, stride, n_rows};
     QuerySpec q;
     __builtin_memcpy(&q, spec, sizeof(QuerySpec));
     hipLaunchKernelGGL(k_query_agg, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
@@ -1161,13 +1173,15 @@ int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
 
 int df_query_select(const void* u64c, const void* u32c, const void* u8c,
                     const void* didc, const void* kgc, const void* attrid,
-                    const void* attr_cnt, uint64_t stride, uint64_t n_rows,
+                    const void* attr_cnt, const void* strc, const void* pool,
+                    uint64_t stride, uint64_t n_rows,
                     const void* spec, uint32_t n, uint64_t base_row,
                     void* out_rows, void* out_ctr, uint32_t out_cap,
                     uint64_t stream) {
     SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
               (const uint32_t*)didc, (const uint32_t*)kgc, (const uint32_t*)attrid,
-              (const uint8_t*)attr_cnt, stride, n_rows};
+              (const uint8_t*)attr_cnt, (const uint64_t*)strc,
+              (const uint8_t*)pool, stride, n_rows};
     QuerySpec q;
     __builtin_memcpy(&q, spec, sizeof(QuerySpec));
     hipLaunchKernelGGL(k_query_select, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),

@@ -155,7 +155,8 @@ def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
     native.check(lib.df_query_agg(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
         _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
-        _opt_ptr(seg, "attr_cnt"), seg.capacity, seg.n_rows,
+        _opt_ptr(seg, "attr_cnt"), seg.strref.data_ptr(),
+        seg.pool.data_ptr(), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         gkeys.data_ptr(), graw.data_ptr(), gvals.data_ptr(), gkeys.numel(),
         _stream()), "df_query_agg")
@@ -169,7 +170,8 @@ def query_select(seg, spec_bytes: bytes, base_row: int, n: int,
     native.check(lib.df_query_select(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
         _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
-        _opt_ptr(seg, "attr_cnt"), seg.capacity, seg.n_rows,
+        _opt_ptr(seg, "attr_cnt"), seg.strref.data_ptr(),
+        seg.pool.data_ptr(), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         out_rows.data_ptr(), out_ctr.data_ptr(), out_rows.numel(), _stream()),
         "df_query_select")

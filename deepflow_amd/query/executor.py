@@ -14,7 +14,8 @@ import torch
 
 from ..store import l7_schema as S
 from .spec import (Plan, SRC_U64, SRC_U32, SRC_U8, SRC_DID, SRC_KG,
-                   SRC_ATTR_VAL, SRC_TIME_BUCKET, SRC_CONST0,
+                   SRC_ATTR_VAL, SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
+                   STR_FILTER_SEED,
                    OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN,
                    AGGOP_COUNT, AGGOP_SUM, AGGOP_MIN, AGGOP_MAX,
                    QMAX_KEYS, QMAX_AGGS)
@@ -47,6 +48,18 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
         if bucket:
             rel = (rel // np.uint64(bucket)) * np.uint64(bucket)
         return rel
+    if family == SRC_STR_HASH:
+        from ..store.dictionary import str_hash_py
+        refs = seg.strref[idx, :n].numpy().view(np.uint64)
+        pool = seg.pool.numpy().tobytes()
+        out = np.zeros(n, dtype=np.uint64)
+        for i in range(n):
+            r = int(refs[i])
+            ln = r & 0xFFFF
+            if ln:
+                off = r >> 16
+                out[i] = str_hash_py(pool[off:off + ln], STR_FILTER_SEED)
+        return out
     return np.zeros(n, dtype=np.uint64)
 
 

@@ -14,7 +14,7 @@ from fastapi.responses import JSONResponse
 
 
 def build_app(engine, registry=None, tempo=None, promql=None,
-              profile=None) -> FastAPI:
+              profile=None, tracing=None) -> FastAPI:
     app = FastAPI(title="deepflow-amd querier")
 
     @app.get("/v1/health")
@@ -57,6 +57,8 @@ def build_app(engine, registry=None, tempo=None, promql=None,
 
     if tempo is not None:
         tempo.register(app)
+    if tracing is not None:
+        tracing.register(app)
     if promql is not None:
         promql.register(app)
     if profile is not None:

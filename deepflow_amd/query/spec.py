@@ -22,6 +22,10 @@ SRC_KG = 4
 SRC_ATTR_VAL = 5
 SRC_TIME_BUCKET = 6
 SRC_CONST0 = 7
+SRC_STR_HASH = 8
+
+# seed for pooled-string filter hashing (twin: dfgpu.hip STR_FILTER_SEED)
+STR_FILTER_SEED = 0x5157A15E5EED
 
 OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN = range(7)
 AGGOP_COUNT, AGGOP_SUM, AGGOP_MIN, AGGOP_MAX = range(4)

@@ -274,6 +274,11 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
         plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v))
         return
     # string literal
+    if td.hydrate == "strhash":
+        from ..store.dictionary import str_hash_py
+        v = str_hash_py(lit[1].encode(), Q.STR_FILTER_SEED)
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v))
+        return
     if td.hydrate.startswith("dict:"):
         dom = int(td.hydrate.split(":")[1])
         ident = dictionary.lookup_id(dom, lit[1].encode()) \

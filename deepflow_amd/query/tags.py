@@ -81,6 +81,12 @@ def build_l7_tags() -> Dict[str, TagDef]:
                                      S.U8_COLS.index("l7_protocol"),
                                      hydrate="l7proto")
     tags["app_service"] = tags["service_name"]
+    # pooled string tags: filterable via GPU string-hash compare,
+    # selectable from the segment pool
+    for sname in ["trace_id", "span_id", "parent_span_id", "x_request_id_0",
+                  "x_request_id_1", "http_user_agent", "biz_code"]:
+        add(TagDef(sname, Q.SRC_STR_HASH, S.STR_COLS.index(sname),
+                   hydrate="strhash"))
     # KnowledgeGraph universal tags, client (_0) / server (_1)
     for side in (0, 1):
         for j, kname in enumerate(S.KG_COLS):

@@ -40,7 +40,12 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_PROTOCOLLOG, self._on_l7)
         self.receiver.register(framing.MSG_TAGGEDFLOW, self._on_l4)
         self.engine = QueryEngine(self.l7, device=device, l4_pipeline=self.l4)
-        self.app = build_app(self.engine, registry=default_registry())
+        from .query.tempo import TempoApp
+        from .query.tracing import DistributedTracer
+        self.tempo = TempoApp(self.engine)
+        self.tracer = DistributedTracer(self.engine)
+        self.app = build_app(self.engine, registry=default_registry(),
+                             tempo=self.tempo, tracing=self.tracer)
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
