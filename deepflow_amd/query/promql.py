@@ -1,0 +1,268 @@
+"""PromQL engine over the rollup tables (reference: server/querier/app/
+prometheus — PromQL -> DF-SQL conversion, converters.go).
+
+Supported subset:
+  selector:      metric{label="v",...}
+  range fns:     rate(sel[w]) / increase(sel[w]) / avg_over_time(sel[w])
+  aggregation:   sum|avg|min|max|count [by (l1, l2)] (expr)
+  APIs:          /prom/api/v1/query (instant), /query_range, /series,
+                 /label/<name>/values
+
+Metric namespace (prom name -> table field):
+  application_<field>  -> application.1s rollup (request, response,
+                          client_error, server_error, rrt_sum, rrt_count,
+                          rrt_max)
+  network_<field>      -> network.1s rollup (byte_tx, byte_rx, packet_tx,
+                          packet_rx, new_flow, closed_flow, rtt_sum, ...)
+Buckets are per-second deltas, so rate(x[w]) = sum(window)/w and
+increase(x[w]) = sum(window).
+"""
+from __future__ import annotations
+
+import re
+import time as _time
+from typing import Dict, List, Optional, Tuple
+
+from fastapi import Request
+
+_SEL_RE = re.compile(
+    r"^\s*(?P<name>[a-zA-Z_:][a-zA-Z0-9_:]*)\s*(?:\{(?P<matchers>[^}]*)\})?"
+    r"\s*(?:\[(?P<range>\d+)(?P<runit>[smh])\])?\s*$")
+def _balanced(s: str):
+    """s starts with '('; return (inner, tail-after-close)."""
+    depth = 0
+    for i, ch in enumerate(s):
+        if ch == "(":
+            depth += 1
+        elif ch == ")":
+            depth -= 1
+            if depth == 0:
+                return s[1:i], s[i + 1:]
+    return None, None
+
+
+def _match_agg(expr: str):
+    m = re.match(r"^\s*(sum|avg|min|max|count)\s*", expr)
+    if not m:
+        return None
+    func, rest = m.group(1), expr[m.end():]
+    by = None
+    m2 = re.match(r"^by\s*\(([^)]*)\)\s*", rest)
+    if m2:
+        by = m2.group(1)
+        rest = rest[m2.end():]
+    if not rest.startswith("("):
+        return None
+    inner, tail = _balanced(rest)
+    if inner is None:
+        return None
+    m3 = re.match(r"^\s*by\s*\(([^)]*)\)\s*$", tail)
+    if m3:
+        by = m3.group(1)
+    elif tail.strip():
+        return None
+    return func, by, inner
+_FN_RE = re.compile(
+    r"^\s*(?P<fn>rate|irate|increase|avg_over_time|sum_over_time|"
+    r"max_over_time)\s*\((?P<inner>.*)\)\s*$", re.DOTALL)
+
+_UNIT = {"s": 1, "m": 60, "h": 3600}
+
+
+class PromError(ValueError):
+    pass
+
+
+def _parse_matchers(s: Optional[str]) -> List[Tuple[str, str, str]]:
+    out = []
+    if not s:
+        return out
+    for part in re.split(r",\s*", s.strip()):
+        if not part:
+            continue
+        m = re.match(r'([a-zA-Z_][a-zA-Z0-9_]*)\s*(=~|!=|=)\s*"([^"]*)"', part)
+        if not m:
+            raise PromError(f"bad matcher {part!r}")
+        out.append((m.group(1), m.group(2), m.group(3)))
+    return out
+
+
+class PromQLEngine:
+    DEFAULT_LOOKBACK = 300
+
+    def __init__(self, app_rows_fn, net_rows_fn=None):
+        """rows_fns return the 1s rollup dict-rows."""
+        self.sources = {"application": app_rows_fn}
+        if net_rows_fn is not None:
+            self.sources["network"] = net_rows_fn
+
+    # -------------------------------------------------------- series fetch
+    def _series(self, name: str,
+                matchers: List[Tuple[str, str, str]]) -> List[Dict]:
+        for prefix, fn in self.sources.items():
+            if name.startswith(prefix + "_"):
+                field = name[len(prefix) + 1:]
+                rows = fn()
+                break
+        else:
+            raise PromError(f"unknown metric {name!r}")
+        series: Dict[tuple, Dict] = {}
+        for r in rows:
+            if field not in r:
+                raise PromError(f"unknown field {field!r} for {name}")
+            labels = {k: str(v) for k, v in r.items()
+                      if k not in ("time", field) and not isinstance(v, float)}
+            ok = True
+            for lname, op, lval in matchers:
+                got = labels.get(lname, "")
+                if op == "=" and got != lval:
+                    ok = False
+                elif op == "!=" and got == lval:
+                    ok = False
+                elif op == "=~" and not re.fullmatch(lval, got):
+                    ok = False
+                if not ok:
+                    break
+            if not ok:
+                continue
+            key = tuple(sorted(labels.items()))
+            s = series.setdefault(key, {"metric": dict(labels, __name__=name),
+                                        "samples": {}})
+            t = r["time"]
+            s["samples"][t] = s["samples"].get(t, 0) + r[field]
+        return list(series.values())
+
+    # -------------------------------------------------------- evaluation
+    def _eval_at(self, expr: str, t: int) -> List[Dict]:
+        """Evaluate expr at instant t -> [{metric, value}]."""
+        agg = _match_agg(expr)
+        if agg:
+            func, by, inner_expr = agg
+            inner = self._eval_at(inner_expr, t)
+            by_labels = [x.strip() for x in by.split(",")] if by else []
+            groups: Dict[tuple, List[float]] = {}
+            metas: Dict[tuple, Dict] = {}
+            for s in inner:
+                labels = {k: v for k, v in s["metric"].items()
+                          if k in by_labels}
+                key = tuple(sorted(labels.items()))
+                groups.setdefault(key, []).append(s["value"])
+                metas[key] = labels
+            out = []
+            for key, vals in groups.items():
+                if func == "sum":
+                    v = sum(vals)
+                elif func == "avg":
+                    v = sum(vals) / len(vals)
+                elif func == "min":
+                    v = min(vals)
+                elif func == "max":
+                    v = max(vals)
+                else:
+                    v = len(vals)
+                out.append({"metric": metas[key], "value": v})
+            return out
+        m = _FN_RE.match(expr)
+        if m:
+            fn = m.group("fn")
+            sel = _SEL_RE.match(m.group("inner"))
+            if not sel or not sel.group("range"):
+                raise PromError(f"{fn}() needs a range selector")
+            w = int(sel.group("range")) * _UNIT[sel.group("runit")]
+            series = self._series(sel.group("name"),
+                                  _parse_matchers(sel.group("matchers")))
+            out = []
+            for s in series:
+                window = [v for ts, v in s["samples"].items()
+                          if t - w < ts <= t]
+                if not window:
+                    continue
+                if fn in ("rate", "irate"):
+                    v = sum(window) / w
+                elif fn in ("increase", "sum_over_time"):
+                    v = float(sum(window))
+                elif fn == "avg_over_time":
+                    v = sum(window) / len(window)
+                else:  # max_over_time
+                    v = float(max(window))
+                metric = {k: v2 for k, v2 in s["metric"].items()
+                          if k != "__name__"}
+                out.append({"metric": metric, "value": v})
+            return out
+        sel = _SEL_RE.match(expr)
+        if sel and not sel.group("range"):
+            series = self._series(sel.group("name"),
+                                  _parse_matchers(sel.group("matchers")))
+            out = []
+            for s in series:
+                recent = [(ts, v) for ts, v in s["samples"].items()
+                          if t - self.DEFAULT_LOOKBACK < ts <= t]
+                if not recent:
+                    continue
+                out.append({"metric": s["metric"],
+                            "value": float(max(recent)[1])})
+            return out
+        raise PromError(f"cannot parse {expr!r}")
+
+    # -------------------------------------------------------- public API
+    def instant(self, query: str, t: Optional[int] = None) -> Dict:
+        t = int(t if t is not None else _time.time())
+        res = self._eval_at(query, t)
+        return {"status": "success",
+                "data": {"resultType": "vector",
+                         "result": [{"metric": s["metric"],
+                                     "value": [t, str(s["value"])]}
+                                    for s in res]}}
+
+    def range_query(self, query: str, start: int, end: int,
+                    step: int) -> Dict:
+        series: Dict[tuple, Dict] = {}
+        t = start
+        while t <= end:
+            for s in self._eval_at(query, t):
+                key = tuple(sorted(s["metric"].items()))
+                e = series.setdefault(key, {"metric": s["metric"],
+                                            "values": []})
+                e["values"].append([t, str(s["value"])])
+            t += step
+        return {"status": "success",
+                "data": {"resultType": "matrix",
+                         "result": list(series.values())}}
+
+    def label_values(self, label: str) -> Dict:
+        vals = set()
+        for fn in self.sources.values():
+            for r in fn():
+                if label in r:
+                    vals.add(str(r[label]))
+        return {"status": "success", "data": sorted(vals)}
+
+    def register(self, app) -> None:
+        @app.get("/prom/api/v1/query")
+        @app.post("/prom/api/v1/query")
+        async def prom_query(request: Request):
+            params = dict(request.query_params)
+            q = params.get("query")
+            t = params.get("time")
+            try:
+                return self.instant(q, int(float(t)) if t else None)
+            except PromError as e:
+                return {"status": "error", "errorType": "bad_data",
+                        "error": str(e)}
+
+        @app.get("/prom/api/v1/query_range")
+        @app.post("/prom/api/v1/query_range")
+        async def prom_range(request: Request):
+            params = dict(request.query_params)
+            try:
+                return self.range_query(
+                    params["query"], int(float(params["start"])),
+                    int(float(params["end"])),
+                    int(float(params.get("step", "60"))))
+            except PromError as e:
+                return {"status": "error", "errorType": "bad_data",
+                        "error": str(e)}
+
+        @app.get("/prom/api/v1/label/{label}/values")
+        async def prom_label_values(label: str):
+            return self.label_values(label)

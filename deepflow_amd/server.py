@@ -44,8 +44,11 @@ class DeepflowServer:
         from .query.tracing import DistributedTracer
         self.tempo = TempoApp(self.engine)
         self.tracer = DistributedTracer(self.engine)
+        from .query.promql import PromQLEngine
+        self.promql = PromQLEngine(self.l7.metrics.rows, self.l4.metrics.rows)
         self.app = build_app(self.engine, registry=default_registry(),
-                             tempo=self.tempo, tracing=self.tracer)
+                             tempo=self.tempo, tracing=self.tracer,
+                             promql=self.promql)
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
