@@ -1,0 +1,55 @@
+"""Continuous GPU profiler: HIP kernel activity -> Profile records.
+
+The reference has no GPU profiler (its ProfileEventType reserves
+EbpfHbmAlloc/InUse slots only, metric.proto:203-204); this is the net-new
+MI355X piece from BASELINE config #5. Kernel activity is captured through
+torch.profiler, whose ROCm backend is kineto over roctracer — i.e. the
+roctracer callback stream — and folded into `gpu;<pipeline-op>;<kernel>`
+stacks that land in the same profile store/flame-graph service as eBPF
+OnCPU profiles.
+
+Usage:
+    gp = GpuProfiler(profile_pipeline)
+    with gp.capture():
+        ... GPU work ...
+    # kernel samples are now queryable via /v1/profile/flame
+"""
+from __future__ import annotations
+
+import contextlib
+import time
+from typing import Optional
+
+from ..wire.metric import PROFILE_EVENT_EBPF_HBM_IN_USE
+
+
+class GpuProfiler:
+    EVENT_TYPE_ON_GPU = PROFILE_EVENT_EBPF_HBM_IN_USE + 1  # 7: OnGpu (ours)
+
+    def __init__(self, pipeline, process_name: str = "deepflow-gpu"):
+        self.pipe = pipeline
+        self.process_name = process_name
+        self.captures = 0
+
+    @contextlib.contextmanager
+    def capture(self):
+        import torch
+        from torch.profiler import profile, ProfilerActivity
+        with profile(activities=[ProfilerActivity.CUDA],
+                     record_shapes=False) as prof:
+            yield prof
+        self._harvest(prof)
+
+    def _harvest(self, prof) -> None:
+        ts = int(time.time() * 1_000_000)
+        common = dict(event_type=self.EVENT_TYPE_ON_GPU, pid=0, tid=0,
+                      pod_id=0, process_name=self.process_name,
+                      app_service="gpu", profile_language_type="roctracer")
+        for evt in prof.key_averages():
+            if getattr(evt, "device_time_total", 0) <= 0:
+                continue
+            name = evt.key.replace(";", "_")
+            stack = f"gpu;{name}".encode()
+            # value: total device-side microseconds for this kernel
+            self.pipe._add(ts, stack, int(evt.device_time_total), common)
+        self.captures += 1

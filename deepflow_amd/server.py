@@ -46,9 +46,15 @@ class DeepflowServer:
         self.tracer = DistributedTracer(self.engine)
         from .query.promql import PromQLEngine
         self.promql = PromQLEngine(self.l7.metrics.rows, self.l4.metrics.rows)
+        from .ingest.profile_pipeline import ProfilePipeline, ProfileApp
+        self.profiles = ProfilePipeline()
+        self.receiver.register(framing.MSG_PROFILE,
+                               lambda hdr, payload:
+                               self.profiles.ingest_payload(payload.tobytes()))
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
-                             promql=self.promql)
+                             promql=self.promql,
+                             profile=ProfileApp(self.profiles))
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
