@@ -31,6 +31,36 @@ SRC_ROW = 100  # sentinel family: python row-table field
 _FROM_RE = re.compile(r"\bfrom\s+`?([\w.]+)`?", re.IGNORECASE)
 
 
+def rollup_rows(rows, bucket_s: int):
+    """1s rollup rows -> coarser buckets (reference datasource 1m/1h
+    AggregatingMergeTree MVs, ingester/datasource/handle.go:136-172):
+    sums for additive fields, max for *_max fields."""
+    out = {}
+    for r in rows:
+        t = (r["time"] // bucket_s) * bucket_s
+        key_fields = tuple(sorted(
+            (k, v) for k, v in r.items()
+            if isinstance(v, str) or k in ("vtap_id", "l7_protocol",
+                                           "response_status", "server_port",
+                                           "l3_epc_id", "protocol")))
+        key = (t,) + key_fields
+        acc = out.get(key)
+        if acc is None:
+            acc = dict(r)
+            acc["time"] = t
+            out[key] = acc
+        else:
+            for k, v in r.items():
+                if k == "time" or isinstance(v, str):
+                    continue
+                if k.endswith("_max"):
+                    acc[k] = max(acc[k], v)
+                elif k not in ("vtap_id", "l7_protocol", "response_status",
+                               "server_port", "l3_epc_id", "protocol"):
+                    acc[k] = acc[k] + v
+    return sorted(out.values(), key=lambda r: r["time"])
+
+
 def _row_tags(fields: List[str]) -> Dict[str, TagDef]:
     return {f: TagDef(f, SRC_ROW, i) for i, f in enumerate(fields)}
 
@@ -68,11 +98,15 @@ class QueryEngine:
                                       L4_TAGS, L4S.STR_COLS)
         if table.startswith("application"):
             rows = self.pipe.metrics.rows()
+            if table.endswith(".1m"):
+                rows = rollup_rows(rows, 60)
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
         if table.startswith("network"):
             if self.l4 is None:
                 raise SqlError("network table not enabled")
             rows = self.l4.metrics.rows()
+            if table.endswith(".1m"):
+                rows = rollup_rows(rows, 60)
             return self._run_rows(sql, rows, time_base_s=self.l4.time_base_s)
         raise SqlError(f"unknown table {table!r}")
 
@@ -84,6 +118,26 @@ class QueryEngine:
         table = m.group(1).lower() if m else "l7_flow_log"
         tags = L4_TAGS if table == "l4_flow_log" else L7_TAGS
         mets = L4_METRICS if table == "l4_flow_log" else L7_METRICS
+        # "show tag <name> values [from <table>]" — flow_tag discovery
+        # (reference flow_tag custom_field_value tables)
+        if what == "tag" and len(parts) >= 4 and parts[3].lower() == "values":
+            name = parts[2].strip("`")
+            td = tags.get(name)
+            vals = []
+            if td is not None and td.hydrate.startswith("dict:"):
+                dom = int(td.hydrate.split(":")[1])
+                vals = sorted(s.decode("utf-8", "replace")
+                              for (d, s) in self.pipe.dict.str_to_id
+                              if d == dom)
+            elif name == "attribute_names":
+                vals = sorted(s.decode("utf-8", "replace")
+                              for (d, s) in self.pipe.dict.str_to_id
+                              if d == 6)
+            elif name.startswith("attribute."):
+                vals = sorted(s.decode("utf-8", "replace")
+                              for (d, s) in self.pipe.dict.str_to_id
+                              if d == 7)
+            return {"columns": ["value"], "values": [[v] for v in vals]}
         if what == "tags":
             cols = ["name", "display_name", "type"]
             vals = [[n, n, t.hydrate] for n, t in sorted(tags.items())]

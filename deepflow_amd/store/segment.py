@@ -110,15 +110,49 @@ class L4Segment:
 class SegmentSet:
     """The shard-local hot window: ordered list of segments."""
 
-    def __init__(self, segment_rows: int, device: str = "cpu", cls=None):
+    def __init__(self, segment_rows: int, device: str = "cpu", cls=None,
+                 max_bytes: Optional[int] = None):
         self.segment_rows = segment_rows
         self.device = device
         self.cls = cls or L7Segment
         self.segments: List = []
+        # HBM hot-window watermark (ckmonitor analog: reference
+        # ckmonitor/monitor.go:339-432 force-drops oldest partitions)
+        self.max_bytes = max_bytes
+        self.evicted_rows = 0
+        self.evicted_segments = 0
+
+    @staticmethod
+    def seg_alloc_bytes(seg) -> int:
+        total = 0
+        for name in ("u64", "u32", "u8", "strref", "did", "kg", "attr_ref",
+                     "attr_id", "attr_cnt", "pool"):
+            t = getattr(seg, name, None)
+            if t is not None:
+                total += t.numel() * t.element_size()
+        return total
+
+    def total_alloc_bytes(self) -> int:
+        return sum(self.seg_alloc_bytes(s) for s in self.segments)
+
+    def enforce_watermark(self) -> int:
+        """Drop oldest full segments while over the byte watermark;
+        returns segments evicted this call."""
+        if self.max_bytes is None:
+            return 0
+        dropped = 0
+        while (len(self.segments) > 1 and
+               self.total_alloc_bytes() > self.max_bytes):
+            seg = self.segments.pop(0)
+            self.evicted_rows += seg.n_rows
+            self.evicted_segments += 1
+            dropped += 1
+        return dropped
 
     def tail(self, min_free: int):
         if not self.segments or self.segments[-1].free_rows() < min_free:
             self.segments.append(self.cls(self.segment_rows, self.device))
+            self.enforce_watermark()
         return self.segments[-1]
 
     @property
