@@ -41,9 +41,9 @@ def test_rollup_rows():
     rows = [
         {"time": 100, "vtap_id": 1, "l7_protocol": 20, "response_status": 0,
          "server_port": 80, "request": 5, "rrt_max": 9, "rrt_sum": 50},
-        {"time": 130, "vtap_id": 1, "l7_protocol": 20, "response_status": 0,
+        {"time": 110, "vtap_id": 1, "l7_protocol": 20, "response_status": 0,
          "server_port": 80, "request": 7, "rrt_max": 20, "rrt_sum": 70},
-        {"time": 130, "vtap_id": 2, "l7_protocol": 20, "response_status": 0,
+        {"time": 111, "vtap_id": 2, "l7_protocol": 20, "response_status": 0,
          "server_port": 80, "request": 1, "rrt_max": 1, "rrt_sum": 1},
     ]
     out = rollup_rows(rows, 60)
