@@ -118,6 +118,7 @@ class L7IngestPipeline:
     def _run_gpu(self, payload_t, offs_t, lens_t, payload_host,
                  seg: L7Segment, base: int, n: int) -> None:
         from ..ops import gpu_ops
+        dev = payload_t.device
         gpu_ops.decode_l7(payload_t, offs_t, lens_t, seg, base)
         gpu_ops.kg_probe(seg, base, n, self.kg.tkeys, self.kg.tvals)
         gpu_ops.intern_many(payload_t, seg.strref, self._ref_rows_scalar,
