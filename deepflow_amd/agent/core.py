@@ -1,0 +1,114 @@
+"""Python control wrapper for the C++ agent core (ops/csrc/agent_core.cpp)
+plus the uniform sender (framing + TCP to the ingester).
+
+Reference counterparts: agent/src/trident.rs component lifecycle,
+sender/uniform_sender.rs. Packet sources: synthetic frames in tests;
+AF_PACKET capture on a deployment host feeds dfa_packet the same way.
+"""
+from __future__ import annotations
+
+import ctypes as ct
+import socket
+from typing import Dict, Optional
+
+import numpy as np
+
+from ..ops import native
+from ..wire import framing
+
+DRAIN_L4, DRAIN_L7, DRAIN_DOC = 0, 1, 2
+
+_MSG_FOR = {DRAIN_L4: framing.MSG_TAGGEDFLOW,
+            DRAIN_L7: framing.MSG_PROTOCOLLOG,
+            DRAIN_DOC: framing.MSG_METRICS}
+
+
+def _lib():
+    lib = native.cpu()
+    if not hasattr(lib, "_agent_decl"):
+        p, u32, u64 = ct.c_void_p, ct.c_uint32, ct.c_uint64
+        lib.dfa_new.restype = p
+        lib.dfa_new.argtypes = [u32]
+        lib.dfa_free.argtypes = [p]
+        lib.dfa_add_cidr.argtypes = [p, u32, u32, ct.c_int32]
+        lib.dfa_packet.restype = ct.c_int
+        lib.dfa_packet.argtypes = [p, p, u32, u64]
+        lib.dfa_tick.argtypes = [p, u64]
+        lib.dfa_drain.restype = u64
+        lib.dfa_drain.argtypes = [p, ct.c_int, p, u64]
+        lib.dfa_stats.argtypes = [p, p]
+        lib._agent_decl = True
+    return lib
+
+
+class Agent:
+    def __init__(self, vtap_id: int = 1, agent_id: Optional[int] = None,
+                 server: Optional[tuple] = None, team_id: int = 0,
+                 org_id: int = 1):
+        self._lib = _lib()
+        self._h = self._lib.dfa_new(vtap_id)
+        self.vtap_id = vtap_id
+        self.agent_id = agent_id if agent_id is not None else vtap_id
+        self.team_id = team_id
+        self.org_id = org_id
+        self.server = server
+        self._sock: Optional[socket.socket] = None
+
+    def close(self) -> None:
+        if self._h:
+            self._lib.dfa_free(self._h)
+            self._h = None
+        if self._sock:
+            self._sock.close()
+            self._sock = None
+
+    def add_cidr(self, net: int, masklen: int, epc: int) -> None:
+        self._lib.dfa_add_cidr(self._h, net, masklen, epc)
+
+    def packet(self, frame: bytes, ts_ns: int) -> int:
+        buf = np.frombuffer(frame, dtype=np.uint8)
+        return self._lib.dfa_packet(self._h, buf.ctypes.data, len(frame),
+                                    ts_ns)
+
+    def tick(self, now_ns: int) -> None:
+        self._lib.dfa_tick(self._h, now_ns)
+
+    def drain(self, which: int) -> bytes:
+        n = self._lib.dfa_drain(self._h, which, None, 0)
+        if n == 0:
+            return b""
+        out = np.zeros(int(n), dtype=np.uint8)
+        got = self._lib.dfa_drain(self._h, which, out.ctypes.data, n)
+        return out[:got].tobytes()
+
+    def stats(self) -> Dict[str, int]:
+        arr = np.zeros(8, dtype=np.uint64)
+        self._lib.dfa_stats(self._h, arr.ctypes.data)
+        keys = ["flows_active", "flows_emitted", "l7_emitted", "docs_emitted",
+                "packets", "bytes", "parse_errors", "_r"]
+        return dict(zip(keys, (int(x) for x in arr)))
+
+    # ---------------------------------------------------------- sender
+    def frame(self, which: int, payload: bytes) -> bytes:
+        hdr = framing.FrameHeader(msg_type=_MSG_FOR[which],
+                                  team_id=self.team_id, org_id=self.org_id,
+                                  agent_id=self.agent_id)
+        return framing.encode_frame(hdr, payload)
+
+    def flush_to_server(self, now_ns: int) -> int:
+        """tick + drain all types + send framed payloads to the server
+        (uniform-sender analog). Returns frames sent."""
+        self.tick(now_ns)
+        sent = 0
+        for which in (DRAIN_L4, DRAIN_L7, DRAIN_DOC):
+            payload = self.drain(which)
+            if not payload:
+                continue
+            frame = self.frame(which, payload)
+            if self.server is not None:
+                if self._sock is None:
+                    self._sock = socket.create_connection(self.server,
+                                                          timeout=5)
+                self._sock.sendall(frame)
+            sent += 1
+        return sent

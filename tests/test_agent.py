@@ -1,0 +1,131 @@
+"""C++ agent core tests: packet -> flow -> L7 parse -> wire records, plus
+the all-in-one loop (BASELINE config #1: agent -> ingester -> query)."""
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+from deepflow_amd.agent import Agent
+from deepflow_amd.agent.packets import (http_session, dns_session,
+                                        redis_session)
+from deepflow_amd.server import DeepflowServer
+from deepflow_amd.wire import pb, flow_log, metric, framing
+
+CLIENT = 0x0A000001
+SERVER = 0x0A000002
+
+
+@pytest.fixture()
+def agent():
+    a = Agent(vtap_id=7)
+    a.add_cidr(0x0A000000, 8, epc=42)
+    yield a
+    a.close()
+
+
+def _decode(payload, schema):
+    return [pb.decode(r, schema) for r in framing.iter_records(payload)]
+
+
+def test_http_flow(agent):
+    for frame, ts in http_session(CLIENT, SERVER, code=200):
+        assert agent.packet(frame, ts) == 0
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    rec = l7[0]
+    assert rec["base"]["vtap_id"] == 7
+    assert rec["base"]["ip_src"] == CLIENT
+    assert rec["base"]["l3_epc_id_src"] == 42
+    assert rec["base"]["head"]["proto"] == 20
+    assert rec["req"]["req_type"] == "GET"
+    assert rec["req"]["domain"] == "svc.example.com"
+    assert rec["req"]["resource"] == "/api/x"
+    assert rec["resp"]["code"] == 200
+    assert rec["base"]["head"]["rrt"] == 4000  # 5ms->9ms
+
+    l4 = _decode(agent.drain(0), flow_log.TAGGED_FLOW)
+    assert len(l4) == 1
+    f = l4[0]["flow"]
+    assert f["close_type"] == 1  # both FINs
+    assert f["metrics_peer_src"]["packet_count"] == 4
+    assert f["metrics_peer_dst"]["packet_count"] == 3
+    perf = f["perf_stats"]
+    assert perf["l7_protocol"] == 20
+    assert perf["tcp"]["rtt"] == 2000  # SYN->SYNACK 2ms
+    assert perf["l7"]["request_count"] == 1
+    assert perf["l7"]["response_count"] == 1
+
+    docs = _decode(agent.drain(2), metric.DOCUMENT)
+    assert len(docs) >= 1
+    d = docs[0]
+    assert d["meter"]["app"]["traffic"]["request"] == 1
+    assert d["tag"]["field"]["server_port"] == 8080
+
+
+def test_http_error_status(agent):
+    for frame, ts in http_session(CLIENT, SERVER, sport=50000, code=500):
+        agent.packet(frame, ts)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert l7[0]["resp"]["status"] == 3
+    assert l7[0]["resp"]["code"] == 500
+
+
+def test_dns(agent):
+    for frame, ts in dns_session(CLIENT, SERVER, qname="db.svc.local"):
+        agent.packet(frame, ts)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 120
+    assert l7[0]["req"]["domain"] == "db.svc.local"
+
+
+def test_redis(agent):
+    for frame, ts in redis_session(CLIENT, SERVER):
+        agent.packet(frame, ts)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 80
+    assert l7[0]["req"]["req_type"] == "GET"
+    assert l7[0]["req"]["resource"] == "mykey"
+
+
+def test_agent_to_server_end_to_end():
+    """BASELINE config #1: agent -> TCP -> ingester -> SQL query."""
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12, time_base_s=0)
+    srv.start()
+    try:
+        a = Agent(vtap_id=3, server=("127.0.0.1", srv.receiver.tcp_port))
+        a.add_cidr(0x0A000000, 8, epc=5)
+        t0 = 10**9
+        for i in range(20):
+            for frame, ts in http_session(CLIENT + i, SERVER,
+                                          sport=40000 + i,
+                                          path=f"/api/item/{i % 4}",
+                                          t0=t0 + i * 10**7):
+                a.packet(frame, ts)
+        sent = a.flush_to_server(10**9 * 100)
+        assert sent == 3
+        deadline = time.time() + 20
+        while time.time() < deadline and srv.l7.stats.spans_in < 20:
+            time.sleep(0.1)
+        assert srv.l7.stats.spans_in == 20
+        assert srv.l4.stats.flows_in == 20
+        client = TestClient(srv.app)
+        r = client.post("/v1/query/", json={
+            "sql": "SELECT request_resource, Count(*) AS c FROM l7_flow_log "
+                   "GROUP BY request_resource ORDER BY c DESC"})
+        vals = r.json()["result"]["values"]
+        assert sum(v[1] for v in vals) == 20
+        assert len(vals) == 4
+        r2 = client.post("/v1/query/", json={
+            "sql": "SELECT Count(*) AS c FROM l4_flow_log WHERE "
+                   "l3_epc_id_0 = 5"})
+        assert r2.json()["result"]["values"] == [[20]]
+        a.close()
+    finally:
+        srv.stop()
