@@ -92,8 +92,11 @@ def main() -> None:
 
     kg = KnowledgeGraphTable(capacity_pow2=1 << 14, device=device)
     kg.update(default_platform(cfg))
+    # hot-window watermark: evicted segments return to the torch caching
+    # allocator so steady-state segment rolls are cache-hit allocations
     pipe = L7IngestPipeline(device=device, segment_rows=1 << 23,
                             kg=kg, dict_capacity=1 << 23,
+                            window_bytes=48 << 30,
                             time_base_s=cfg.base_time_ns // 10**9)
 
     dict_sync = None
@@ -122,6 +125,8 @@ def main() -> None:
                 prefetch(i)
             dev_batch, ev, host_payload = pending.pop(i)
             torch.cuda.current_stream().wait_event(ev)
+            for t in dev_batch:
+                t.record_stream(torch.cuda.current_stream())
             prefetch(i + 1)
             pipe.ingest_device(*dev_batch, host_payload)
         else:

@@ -46,9 +46,11 @@ class L7IngestPipeline:
                  kg: Optional[KnowledgeGraphTable] = None,
                  dictionary: Optional[TagDictionary] = None,
                  dict_capacity: int = 1 << 22,
+                 window_bytes: Optional[int] = None,
                  counter: Optional[Counter] = None):
         self.device = device
-        self.segments = SegmentSet(segment_rows, device)
+        self.segments = SegmentSet(segment_rows, device,
+                                   max_bytes=window_bytes)
         self.kg = kg or KnowledgeGraphTable(device=device)
         self.dict = dictionary or TagDictionary(dict_capacity, device=device)
         self.metrics = App1sMetrics(time_base_s, device=device)

@@ -121,6 +121,7 @@ class SegmentSet:
         self.max_bytes = max_bytes
         self.evicted_rows = 0
         self.evicted_segments = 0
+        self._free: List = []
 
     @staticmethod
     def seg_alloc_bytes(seg) -> int:
@@ -136,22 +137,48 @@ class SegmentSet:
         return sum(self.seg_alloc_bytes(s) for s in self.segments)
 
     def enforce_watermark(self) -> int:
-        """Drop oldest full segments while over the byte watermark;
-        returns segments evicted this call."""
+        """Drop oldest full segments while over the byte watermark; dropped
+        segments go to a free-list and are zero-reset on reuse (steady-state
+        segment rolls are memsets, not allocations). Returns segments
+        evicted this call."""
         if self.max_bytes is None:
             return 0
         dropped = 0
         while (len(self.segments) > 1 and
-               self.total_alloc_bytes() > self.max_bytes):
+               self.total_alloc_bytes() + sum(
+                   self.seg_alloc_bytes(s) for s in self._free) >
+               self.max_bytes):
             seg = self.segments.pop(0)
             self.evicted_rows += seg.n_rows
             self.evicted_segments += 1
+            self._free.append(seg)
             dropped += 1
         return dropped
 
+    @staticmethod
+    def reset_segment(seg) -> None:
+        for name in ("u64", "u32", "u8", "strref", "attr_ref", "kg"):
+            t = getattr(seg, name, None)
+            if t is not None:
+                t.zero_()
+        for name in ("did", "attr_id"):
+            t = getattr(seg, name, None)
+            if t is not None:
+                t.fill_(-1)
+        t = getattr(seg, "attr_cnt", None)
+        if t is not None:
+            t.zero_()
+        seg.pool_len = 0
+        seg.n_rows = 0
+
     def tail(self, min_free: int):
         if not self.segments or self.segments[-1].free_rows() < min_free:
-            self.segments.append(self.cls(self.segment_rows, self.device))
+            if self._free:
+                seg = self._free.pop(0)
+                self.reset_segment(seg)
+                self.segments.append(seg)
+            else:
+                self.segments.append(self.cls(self.segment_rows, self.device))
             self.enforce_watermark()
         return self.segments[-1]
 
