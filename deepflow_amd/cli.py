@@ -1,0 +1,158 @@
+"""deepflow-ctl — operator CLI (reference: cli/ctl/*.go).
+
+Talks to a running all-in-one server over HTTP.
+
+  python -m deepflow_amd.cli query "SELECT ..."         DF-SQL query
+  python -m deepflow_amd.cli tables|tags|metrics        metadata discovery
+  python -m deepflow_amd.cli agent list                 registered agents
+  python -m deepflow_amd.cli agent rebalance            redistribute agents
+  python -m deepflow_amd.cli stats                      self-telemetry
+  python -m deepflow_amd.cli trace <trace_id>           trace tree
+  python -m deepflow_amd.cli promql '<expr>'            instant PromQL
+"""
+from __future__ import annotations
+
+import json
+import sys
+
+import click
+import requests
+
+
+@click.group()
+@click.option("--server", default="http://127.0.0.1:20416",
+              envvar="DEEPFLOW_SERVER", help="querier base URL")
+@click.pass_context
+def cli(ctx, server):
+    ctx.obj = {"server": server.rstrip("/")}
+
+
+def _print_table(result: dict) -> None:
+    cols = result.get("columns", [])
+    vals = result.get("values", [])
+    if not cols:
+        click.echo("(empty)")
+        return
+    widths = [max(len(str(c)), *(len(str(r[i])) for r in vals)) if vals
+              else len(str(c)) for i, c in enumerate(cols)]
+    click.echo("  ".join(str(c).ljust(w) for c, w in zip(cols, widths)))
+    for r in vals:
+        click.echo("  ".join(str(x).ljust(w) for x, w in zip(r, widths)))
+
+
+@cli.command()
+@click.argument("sql")
+@click.option("--json", "as_json", is_flag=True)
+@click.pass_context
+def query(ctx, sql, as_json):
+    """Run a DF-SQL query."""
+    r = requests.post(f"{ctx.obj['server']}/v1/query/", json={"sql": sql},
+                      timeout=60)
+    body = r.json()
+    if body.get("OPT_STATUS") != "SUCCESS":
+        click.echo(f"error: {body.get('DESCRIPTION')}", err=True)
+        sys.exit(1)
+    if as_json:
+        click.echo(json.dumps(body["result"]))
+    else:
+        _print_table(body["result"])
+
+
+@cli.command()
+@click.pass_context
+def tables(ctx):
+    ctx.invoke(query, sql="show tables", as_json=False)
+
+
+@cli.command()
+@click.option("--table", default="l7_flow_log")
+@click.pass_context
+def tags(ctx, table):
+    ctx.invoke(query, sql=f"show tags from {table}", as_json=False)
+
+
+@cli.command()
+@click.option("--table", default="l7_flow_log")
+@click.pass_context
+def metrics(ctx, table):
+    ctx.invoke(query, sql=f"show metrics from {table}", as_json=False)
+
+
+@cli.group()
+def agent():
+    """Agent management."""
+
+
+@agent.command("list")
+@click.pass_context
+def agent_list(ctx):
+    r = requests.get(f"{ctx.obj['server']}/v1/agents/", timeout=30)
+    rows = r.json()
+    if not rows:
+        click.echo("(no agents)")
+        return
+    _print_table({"columns": list(rows[0].keys()),
+                  "values": [list(a.values()) for a in rows]})
+
+
+@agent.command("rebalance")
+@click.pass_context
+def agent_rebalance(ctx):
+    r = requests.post(f"{ctx.obj['server']}/v1/rebalance/", timeout=30)
+    click.echo(json.dumps(r.json()))
+
+
+@cli.command()
+@click.pass_context
+def stats(ctx):
+    """Self-telemetry counters."""
+    r = requests.get(f"{ctx.obj['server']}/v1/stats", timeout=30)
+    for s in r.json():
+        tags = ",".join(f"{k}={v}" for k, v in
+                        zip(s.get("tag_names", []), s.get("tag_values", [])))
+        for name, val in zip(s["metrics_float_names"],
+                             s["metrics_float_values"]):
+            click.echo(f"{s['name']}{{{tags}}} {name}={val}")
+
+
+@cli.command()
+@click.argument("trace_id")
+@click.pass_context
+def trace(ctx, trace_id):
+    """Distributed trace tree."""
+    r = requests.get(f"{ctx.obj['server']}/v1/tracing/{trace_id}", timeout=60)
+    body = r.json()
+
+    def walk(idx, depth):
+        n = body["spans"][idx]
+        click.echo("  " * depth +
+                   f"{n['service'] or '?'} {n['resource'] or ''} "
+                   f"[{n['duration_ns'] / 1e6:.2f} ms]")
+        for c in n["children"]:
+            walk(c, depth + 1)
+
+    for root in body["roots"]:
+        walk(root, 0)
+    click.echo(f"({body['span_count']} spans)")
+
+
+@cli.command()
+@click.argument("expr")
+@click.option("--time", "t", default=None)
+@click.pass_context
+def promql(ctx, expr, t):
+    """Instant PromQL query."""
+    params = {"query": expr}
+    if t:
+        params["time"] = t
+    r = requests.get(f"{ctx.obj['server']}/prom/api/v1/query", params=params,
+                     timeout=60)
+    click.echo(json.dumps(r.json(), indent=2))
+
+
+def main():
+    cli()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,179 @@
+"""Controller-lite: agent registry, config push, platform data, tagrecorder.
+
+The thin MI355X-native control plane covering the reference controller's
+roles that the data plane depends on:
+  - trisolaris-lite: agent (vtap) registry + versioned config/platform sync
+    (reference controller/trisolaris, grpc/synchronizer/service.go:57)
+  - platform data: (epc, ip) -> resource inventory pushed into the GPU
+    KnowledgeGraph tables (reference PlatformData / AnalyzerSync)
+  - tagrecorder-lite: resource id -> name maps served to the querier
+    (reference controller/tagrecorder ch_* dictionaries)
+  - election: single-process leader flag (k8s election is a deployment
+    concern; the API surface is kept)
+  - monitor: agent liveness from sync timestamps + analyzer rebalance hook
+
+Transport: HTTP/JSON on the server app (the reference speaks gRPC; the
+sync *semantics* — version-gated push of config/platform/groups — are
+preserved; gRPC framing is round-2 work).
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from fastapi import Request
+
+from ..store.kg import KgInfo, KnowledgeGraphTable
+
+
+@dataclass
+class AgentRecord:
+    agent_id: int
+    hostname: str = ""
+    ip: str = ""
+    group: str = "default"
+    first_seen: float = 0.0
+    last_sync: float = 0.0
+    config_version: int = 0
+    platform_version: int = 0
+    exceptions: int = 0
+    analyzer: int = 0   # which shard/GPU ingests this agent's streams
+
+
+DEFAULT_AGENT_CONFIG = {
+    "max_millicpus": 1000,
+    "max_memory": 768,          # MiB
+    "sync_interval": 10,
+    "l7_log_collect_enabled": True,
+    "l4_log_collect_enabled": True,
+    "throttle_per_second": 50000,
+    "l7_protocols": ["HTTP", "DNS", "Redis", "MySQL"],
+}
+
+
+class ControllerLite:
+    def __init__(self, kg: Optional[KnowledgeGraphTable] = None,
+                 n_analyzers: int = 1):
+        self.agents: Dict[int, AgentRecord] = {}
+        self.group_configs: Dict[str, dict] = {"default": dict(DEFAULT_AGENT_CONFIG)}
+        self.config_version = 1
+        self.platform_version = 1
+        self.platform: Dict[Tuple[int, int], KgInfo] = {}
+        self.kg = kg
+        self.n_analyzers = n_analyzers
+        self.is_leader = True
+        # tagrecorder-lite: (map_name, id) -> display name
+        self.name_maps: Dict[str, Dict[int, str]] = {
+            "device_map": {}, "pod_map": {}, "l3_epc_map": {},
+            "vtap_map": {}, "service_map": {}, "az_map": {},
+        }
+
+    # ------------------------------------------------------------ sync
+    def sync(self, agent_id: int, hostname: str = "", ip: str = "",
+             config_version: int = 0, platform_version: int = 0,
+             exceptions: int = 0) -> dict:
+        """Agent sync call (trident.Synchronizer/Sync analog): registers the
+        agent and returns config/platform payloads only when versions moved
+        (the reference's version-gated push)."""
+        now = time.time()
+        rec = self.agents.get(agent_id)
+        if rec is None:
+            rec = AgentRecord(agent_id=agent_id, first_seen=now)
+            rec.analyzer = agent_id % self.n_analyzers
+            self.agents[agent_id] = rec
+        rec.hostname = hostname or rec.hostname
+        rec.ip = ip or rec.ip
+        rec.last_sync = now
+        rec.exceptions = exceptions
+        resp: dict = {
+            "status": "ok",
+            "config_version": self.config_version,
+            "platform_version": self.platform_version,
+            "analyzer": rec.analyzer,
+        }
+        if config_version != self.config_version:
+            resp["config"] = self.group_configs.get(rec.group,
+                                                    DEFAULT_AGENT_CONFIG)
+        if platform_version != self.platform_version:
+            resp["platform"] = [
+                {"epc": epc, "ip": ip_, **vars(info)}
+                for (epc, ip_), info in self.platform.items()]
+        rec.config_version = self.config_version
+        rec.platform_version = self.platform_version
+        return resp
+
+    # ------------------------------------------------------------ admin
+    def set_group_config(self, group: str, config: dict) -> None:
+        base = dict(DEFAULT_AGENT_CONFIG)
+        base.update(config)
+        self.group_configs[group] = base
+        self.config_version += 1
+
+    def update_platform(self, entries: Dict[Tuple[int, int], KgInfo],
+                        names: Optional[Dict[str, Dict[int, str]]] = None
+                        ) -> None:
+        """Cloud/genesis-style inventory update -> bump version, refresh the
+        GPU KnowledgeGraph table and tagrecorder name maps."""
+        self.platform.update(entries)
+        self.platform_version += 1
+        if self.kg is not None:
+            self.kg.update(entries)
+        if names:
+            for m, d in names.items():
+                self.name_maps.setdefault(m, {}).update(d)
+
+    def lookup_name(self, map_name: str, ident: int) -> Optional[str]:
+        return self.name_maps.get(map_name, {}).get(ident)
+
+    # ------------------------------------------------------------ monitor
+    def agent_status(self, stale_after_s: float = 60.0) -> List[dict]:
+        now = time.time()
+        out = []
+        for rec in sorted(self.agents.values(), key=lambda r: r.agent_id):
+            out.append({
+                "agent_id": rec.agent_id,
+                "hostname": rec.hostname,
+                "ip": rec.ip,
+                "group": rec.group,
+                "analyzer": rec.analyzer,
+                "alive": now - rec.last_sync < stale_after_s,
+                "last_sync_age_s": round(now - rec.last_sync, 1),
+                "exceptions": rec.exceptions,
+            })
+        return out
+
+    def rebalance(self) -> Dict[int, int]:
+        """Re-assign agents across analyzers round-robin (reference
+        monitor/vtap/rebalance.go analog). Returns agent -> analyzer."""
+        assignment = {}
+        for i, agent_id in enumerate(sorted(self.agents)):
+            self.agents[agent_id].analyzer = i % self.n_analyzers
+            assignment[agent_id] = self.agents[agent_id].analyzer
+        return assignment
+
+    # ------------------------------------------------------------ http
+    def register(self, app) -> None:
+        @app.post("/v1/sync/")
+        async def sync(request: Request):
+            body = await request.json()
+            return self.sync(
+                agent_id=int(body.get("agent_id", 0)),
+                hostname=body.get("hostname", ""),
+                ip=body.get("ip", ""),
+                config_version=int(body.get("config_version", 0)),
+                platform_version=int(body.get("platform_version", 0)),
+                exceptions=int(body.get("exceptions", 0)))
+
+        @app.get("/v1/agents/")
+        def agents():
+            return self.agent_status()
+
+        @app.post("/v1/agent-group-config/{group}")
+        async def set_config(group: str, request: Request):
+            self.set_group_config(group, await request.json())
+            return {"status": "ok", "config_version": self.config_version}
+
+        @app.post("/v1/rebalance/")
+        def rebalance():
+            return self.rebalance()

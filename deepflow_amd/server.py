@@ -51,10 +51,16 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_PROFILE,
                                lambda hdr, payload:
                                self.profiles.ingest_payload(payload.tobytes()))
+        self.receiver.register(framing.MSG_OPENTELEMETRY, self._on_otel)
+        self.receiver.register(framing.MSG_OPENTELEMETRY_COMPRESSED,
+                               self._on_otel)
+        from .control import ControllerLite
+        self.controller = ControllerLite(kg=self.kg)
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
                              promql=self.promql,
                              profile=ProfileApp(self.profiles))
+        self.controller.register(self.app)
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
@@ -87,6 +93,24 @@ class DeepflowServer:
                                     lens.ctypes.data_as(ct.c_void_p), max_n))
         with self._lock:
             self.l4.ingest(payload, offs[:n].copy(), lens[:n].copy())
+
+    def _on_otel(self, hdr, payload) -> None:
+        """OTLP frames carry one zlib-compressed TracesData blob
+        (reference decoder.go:235-266); convert to AppProtoLogsData and run
+        the normal span pipeline."""
+        import numpy as np
+        from .ingest.otel import otlp_to_l7_payload
+        data = payload.tobytes()
+        try:
+            l7_payload = otlp_to_l7_payload(data, compressed=True)
+        except Exception:
+            try:
+                l7_payload = otlp_to_l7_payload(data, compressed=False)
+            except Exception:
+                self.receiver.counter.add("otel_decode_errors")
+                return
+        arr = np.frombuffer(l7_payload, dtype=np.uint8)
+        self._on_l7(hdr, arr)
 
     # ------------------------------------------------------------------
     def start(self) -> None:

@@ -1,0 +1,59 @@
+"""OTLP trace protobuf schemas (subset of opentelemetry-proto trace/v1 +
+common/v1, wire-compatible field numbers; reference vendors the same protos
+under message/opentelemetry)."""
+
+ANY_VALUE = {
+    1: ("string_value", 's'),
+    2: ("bool_value", 'u'),
+    3: ("int_value", 'i'),
+    4: ("double_value", 'd'),
+}
+
+KEY_VALUE = {
+    1: ("key", 's'),
+    2: ("value", 'm', ANY_VALUE),
+}
+
+RESOURCE = {
+    1: ("attributes", '*m', KEY_VALUE),
+}
+
+STATUS = {
+    2: ("message", 's'),
+    3: ("code", 'u'),  # 0 unset, 1 ok, 2 error
+}
+
+SPAN = {
+    1: ("trace_id", 'b'),
+    2: ("span_id", 'b'),
+    3: ("trace_state", 's'),
+    4: ("parent_span_id", 'b'),
+    5: ("name", 's'),
+    6: ("kind", 'u'),  # 0 unspec, 1 internal, 2 server, 3 client, 4 producer, 5 consumer
+    7: ("start_time_unix_nano", 'x'),
+    8: ("end_time_unix_nano", 'x'),
+    9: ("attributes", '*m', KEY_VALUE),
+    15: ("status", 'm', STATUS),
+}
+
+SCOPE = {
+    1: ("name", 's'),
+    2: ("version", 's'),
+}
+
+SCOPE_SPANS = {
+    1: ("scope", 'm', SCOPE),
+    2: ("spans", '*m', SPAN),
+}
+
+RESOURCE_SPANS = {
+    1: ("resource", 'm', RESOURCE),
+    2: ("scope_spans", '*m', SCOPE_SPANS),
+}
+
+TRACES_DATA = {
+    1: ("resource_spans", '*m', RESOURCE_SPANS),
+}
+
+SPAN_KIND_SERVER = 2
+SPAN_KIND_CLIENT = 3

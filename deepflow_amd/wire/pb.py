@@ -11,6 +11,7 @@ Kinds:
   'u'   unsigned varint (uint32/uint64/bool/enum)
   'i'   signed varint, two's-complement 64-bit (int32/int64)
   'd'   double (wire type 1, little-endian f64)
+  'x'   fixed64 unsigned (wire type 1; OTLP time fields)
   's'   utf-8 string (wire type 2)
   'b'   bytes (wire type 2)
   'm'   nested message (wire type 2), third tuple element = sub-schema
@@ -71,6 +72,11 @@ def _encode_field(out: bytearray, num: int, kind: str, val: Any, sub=None) -> No
             return
         write_varint(out, (num << 3) | 1)
         out += struct.pack('<d', float(val))
+    elif kind == 'x':
+        if val == 0:
+            return
+        write_varint(out, (num << 3) | 1)
+        out += struct.pack('<Q', int(val))
     elif kind == 's':
         data = val.encode('utf-8') if isinstance(val, str) else bytes(val)
         if not data:
@@ -156,7 +162,11 @@ def decode(data, schema: Dict[int, tuple], *, _mv=None) -> Dict[str, Any]:
             pos += 8
             if spec:
                 name, kind = spec[0], spec[1]
-                val = struct.unpack('<d', raw)[0]
+                base = kind.lstrip('*')
+                if base == 'x':
+                    val = struct.unpack('<Q', raw)[0]
+                else:
+                    val = struct.unpack('<d', raw)[0]
                 if kind.startswith('*'):
                     out.setdefault(name, []).append(val)
                 else:

@@ -1,0 +1,69 @@
+"""CLI smoke test against a live HTTP server (uvicorn in a thread)."""
+import socket
+import threading
+import time
+
+import pytest
+from click.testing import CliRunner
+
+from deepflow_amd.cli import cli
+from deepflow_amd.gen import SpanGenConfig
+from deepflow_amd.gen.spans import gen_span_payload
+from deepflow_amd.server import DeepflowServer
+from deepflow_amd.wire import framing
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture(scope="module")
+def live_server():
+    import uvicorn
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12)
+    cfg = SpanGenConfig(n=50, seed=4, tag_cardinality=10, n_ips=16)
+    srv.receiver.handle_frame(framing.encode_frame(
+        framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
+        gen_span_payload(cfg)))
+    port = _free_port()
+    config = uvicorn.Config(srv.app, host="127.0.0.1", port=port,
+                            log_level="error")
+    server = uvicorn.Server(config)
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    deadline = time.time() + 15
+    while time.time() < deadline and not server.started:
+        time.sleep(0.05)
+    yield f"http://127.0.0.1:{port}"
+    server.should_exit = True
+    th.join(timeout=5)
+
+
+def test_cli_query(live_server):
+    runner = CliRunner()
+    r = runner.invoke(cli, ["--server", live_server, "query",
+                            "SELECT Count(*) AS cnt FROM l7_flow_log"])
+    assert r.exit_code == 0, r.output
+    assert "50" in r.output
+
+
+def test_cli_tables_and_stats(live_server):
+    runner = CliRunner()
+    r = runner.invoke(cli, ["--server", live_server, "tables"])
+    assert "l7_flow_log" in r.output
+    r2 = runner.invoke(cli, ["--server", live_server, "stats"])
+    assert r2.exit_code == 0
+
+
+def test_cli_agent_list(live_server):
+    import requests
+    requests.post(f"{live_server}/v1/sync/", json={"agent_id": 2})
+    runner = CliRunner()
+    r = runner.invoke(cli, ["--server", live_server, "agent", "list"])
+    assert r.exit_code == 0
+    assert "2" in r.output
