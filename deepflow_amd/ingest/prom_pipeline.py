@@ -1,0 +1,136 @@
+"""Prometheus remote-write ingest with SmartEncoding label interning.
+
+Reference counterpart: server/ingester/prometheus (grpc_label_ids.go caches +
+fully ID-encoded prometheus.samples rows, prometheus_sample.go:106-122).
+Metric names, label names and label values are interned once; samples are
+stored with zero strings: (metric_id, series_id, ts, value) plus the
+series -> [(label_name_id, label_value_id)] layout.
+
+This is the highest-cardinality dictionary stress case; the store is
+host-side (external metric volumes are far below span volumes), the
+encoding discipline is identical to the GPU span path.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+from ..utils.stats import Counter
+from ..wire import pb, prompb
+
+
+class Interner:
+    def __init__(self):
+        self.to_id: Dict[str, int] = {}
+        self.from_id: List[str] = []
+
+    def intern(self, s: str) -> int:
+        i = self.to_id.get(s)
+        if i is None:
+            i = len(self.from_id)
+            self.to_id[s] = i
+            self.from_id.append(s)
+        return i
+
+
+class PromPipeline:
+    def __init__(self, counter: Optional[Counter] = None):
+        self.metric_names = Interner()
+        self.label_names = Interner()
+        self.label_values = Interner()
+        # series: (metric_id, ((lname_id, lval_id), ...)) -> series_id
+        self.series: Dict[Tuple, int] = {}
+        self.series_labels: List[Tuple] = []
+        self.series_metric: List[int] = []
+        # columnar samples
+        self.s_series: List[int] = []
+        self.s_ts: List[int] = []      # ms
+        self.s_value: List[float] = []
+        self.counter = counter or Counter("ingester.prometheus")
+
+    def ingest_write_request(self, data: bytes) -> int:
+        wr = pb.decode(data, prompb.WRITE_REQUEST)
+        n = 0
+        for ts in wr.get("timeseries", []):
+            metric = ""
+            lab_ids = []
+            for lb in ts.get("labels", []):
+                name = lb.get("name", "")
+                value = lb.get("value", "")
+                if name == "__name__":
+                    metric = value
+                    continue
+                lab_ids.append((self.label_names.intern(name),
+                                self.label_values.intern(value)))
+            mid = self.metric_names.intern(metric)
+            key = (mid, tuple(sorted(lab_ids)))
+            sid = self.series.get(key)
+            if sid is None:
+                sid = len(self.series_labels)
+                self.series[key] = sid
+                self.series_labels.append(key[1])
+                self.series_metric.append(mid)
+            for sm in ts.get("samples", []):
+                self.s_series.append(sid)
+                self.s_ts.append(int(sm.get("timestamp", 0)))
+                self.s_value.append(float(sm.get("value", 0.0)))
+                n += 1
+        self.counter.add("samples_in", n)
+        return n
+
+    # ------------------------------------------------------- query side
+    def series_for(self, metric: str,
+                   matchers: List[Tuple[str, str, str]] = ()) -> List[Dict]:
+        """PromQL source: [{metric: labels, samples: {t_s: v}}]."""
+        import re as _re
+        mid = self.metric_names.to_id.get(metric)
+        if mid is None:
+            return []
+        out = []
+        for sid, smid in enumerate(self.series_metric):
+            if smid != mid:
+                continue
+            labels = {self.label_names.from_id[ln]:
+                      self.label_values.from_id[lv]
+                      for ln, lv in self.series_labels[sid]}
+            ok = True
+            for lname, op, lval in matchers:
+                got = labels.get(lname, "")
+                if op == "=" and got != lval:
+                    ok = False
+                elif op == "!=" and got == lval:
+                    ok = False
+                elif op == "=~" and not _re.fullmatch(lval, got):
+                    ok = False
+                if not ok:
+                    break
+            if not ok:
+                continue
+            samples: Dict[int, float] = {}
+            for i, s in enumerate(self.s_series):
+                if s == sid:
+                    samples[self.s_ts[i] // 1000] = self.s_value[i]
+            out.append({"metric": dict(labels, __name__=metric),
+                        "samples": samples})
+        return out
+
+    def stored_bytes(self) -> int:
+        """SmartEncoding accounting: ID-encoded samples + dictionaries."""
+        samples = len(self.s_series) * (4 + 8 + 8)
+        dicts = sum(len(s) for s in self.metric_names.from_id) + \
+            sum(len(s) for s in self.label_names.from_id) + \
+            sum(len(s) for s in self.label_values.from_id)
+        layout = sum(2 * 4 * len(t) for t in self.series_labels)
+        return samples + dicts + layout
+
+    def naive_bytes(self) -> int:
+        """What the same samples cost with string labels per row."""
+        total = 0
+        for i, sid in enumerate(self.s_series):
+            row = 8 + 8
+            mid = self.series_metric[sid]
+            row += len(self.metric_names.from_id[mid])
+            for ln, lv in self.series_labels[sid]:
+                row += len(self.label_names.from_id[ln]) + \
+                    len(self.label_values.from_id[lv])
+            total += row
+        return total

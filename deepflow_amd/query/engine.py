@@ -101,6 +101,10 @@ class QueryEngine:
             if table.endswith(".1m"):
                 rows = rollup_rows(rows, 60)
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
+        if table.startswith("deepflow_system") or \
+                table.startswith("deepflow_tenant"):
+            rows = getattr(self, "system_rows", [])
+            return self._run_rows(sql, rows, time_base_s=0)
         if table.startswith("network"):
             if self.l4 is None:
                 raise SqlError("network table not enabled")

@@ -90,11 +90,14 @@ def _parse_matchers(s: Optional[str]) -> List[Tuple[str, str, str]]:
 class PromQLEngine:
     DEFAULT_LOOKBACK = 300
 
-    def __init__(self, app_rows_fn, net_rows_fn=None):
-        """rows_fns return the 1s rollup dict-rows."""
+    def __init__(self, app_rows_fn, net_rows_fn=None, raw_sources=None):
+        """rows_fns return the 1s rollup dict-rows; raw_sources are
+        callables (metric_name, matchers) -> series list (e.g. the
+        prometheus remote-write store)."""
         self.sources = {"application": app_rows_fn}
         if net_rows_fn is not None:
             self.sources["network"] = net_rows_fn
+        self.raw_sources = list(raw_sources or [])
 
     # -------------------------------------------------------- series fetch
     def _series(self, name: str,
@@ -105,6 +108,10 @@ class PromQLEngine:
                 rows = fn()
                 break
         else:
+            for raw in self.raw_sources:
+                series = raw(name, matchers)
+                if series:
+                    return series
             raise PromError(f"unknown metric {name!r}")
         series: Dict[tuple, Dict] = {}
         for r in rows:

@@ -45,7 +45,14 @@ class DeepflowServer:
         self.tempo = TempoApp(self.engine)
         self.tracer = DistributedTracer(self.engine)
         from .query.promql import PromQLEngine
-        self.promql = PromQLEngine(self.l7.metrics.rows, self.l4.metrics.rows)
+        from .ingest.prom_pipeline import PromPipeline
+        self.prom = PromPipeline()
+        self.receiver.register(framing.MSG_PROMETHEUS,
+                               lambda hdr, payload:
+                               self.prom.ingest_write_request(
+                                   payload.tobytes()))
+        self.promql = PromQLEngine(self.l7.metrics.rows, self.l4.metrics.rows,
+                                   raw_sources=[self.prom.series_for])
         from .ingest.profile_pipeline import ProfilePipeline, ProfileApp
         self.profiles = ProfilePipeline()
         self.receiver.register(framing.MSG_PROFILE,
@@ -56,6 +63,9 @@ class DeepflowServer:
                                self._on_otel)
         from .control import ControllerLite
         self.controller = ControllerLite(kg=self.kg)
+        self.system_rows = []  # deepflow_system self-metrics store
+        self.receiver.register(framing.MSG_DFSTATS, self._on_dfstats)
+        self.engine.system_rows = self.system_rows
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
                              promql=self.promql,
@@ -113,6 +123,30 @@ class DeepflowServer:
         self._on_l7(hdr, arr)
 
     # ------------------------------------------------------------------
+    def ingest_self_stats(self) -> int:
+        """Self-telemetry loop: snapshot all Countables as dfstats records
+        and feed them back through the receiver (reference: libs/stats
+        shipping MESSAGE_TYPE_DFSTATS to its own ingester,
+        SURVEY.md §5.H)."""
+        from .utils.stats import default_registry
+        payload = default_registry().encode_dfstats()
+        hdr = framing.FrameHeader(msg_type=framing.MSG_DFSTATS)
+        return 1 if self.receiver.handle_frame(
+            framing.encode_frame(hdr, payload)) else 0
+
+    def _on_dfstats(self, hdr, payload) -> None:
+        from .wire import pb, metric
+        for rec in framing.iter_records(payload.tobytes()):
+            d = pb.decode(rec, metric.STATS)
+            row = {"time": d.get("timestamp", 0),
+                   "table": d.get("name", "")}
+            for k, v in zip(d.get("tag_names", []), d.get("tag_values", [])):
+                row[k] = v
+            for k, v in zip(d.get("metrics_float_names", []),
+                            d.get("metrics_float_values", [])):
+                row[k] = v
+            self.system_rows.append(row)
+
     def start(self) -> None:
         self.receiver.start()
 
