@@ -99,6 +99,10 @@ def main() -> None:
                             window_bytes=48 << 30,
                             time_base_s=cfg.base_time_ns // 10**9)
 
+    # provision the hot window: enough segments for the whole run up front
+    total_rows = (args.steps + args.warmup) * args.batch
+    pipe.segments.reserve(total_rows // (1 << 23) + 2)
+
     dict_sync = None
     if world > 1:
         from deepflow_amd.parallel.dict_sync import DictSync

@@ -171,6 +171,13 @@ class SegmentSet:
         seg.pool_len = 0
         seg.n_rows = 0
 
+    def reserve(self, n_segments: int) -> None:
+        """Provision the hot window up front: pre-allocate n_segments into
+        the free list so steady-state segment rolls never hit the device
+        allocator (size-for-288GB-HBM startup provisioning)."""
+        while len(self.segments) + len(self._free) < n_segments:
+            self._free.append(self.cls(self.segment_rows, self.device))
+
     def tail(self, min_free: int):
         if not self.segments or self.segments[-1].free_rows() < min_free:
             if self._free:
