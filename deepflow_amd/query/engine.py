@@ -96,6 +96,12 @@ class QueryEngine:
             from ..store import l4_schema as L4S
             return self._run_segments(plan, self.l4.segments.segments,
                                       L4_TAGS, L4S.STR_COLS)
+        if table == "application.agent":
+            rows = getattr(self, "agent_app_rows", lambda: [])()
+            return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
+        if table == "network.agent":
+            rows = getattr(self, "agent_net_rows", lambda: [])()
+            return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
         if table.startswith("application"):
             rows = self.pipe.metrics.rows()
             if table.endswith(".1m"):

@@ -46,6 +46,11 @@ class DeepflowServer:
         self.tracer = DistributedTracer(self.engine)
         from .query.promql import PromQLEngine
         from .ingest.prom_pipeline import PromPipeline
+        from .ingest.doc_pipeline import DocPipeline
+        self.docs = DocPipeline()
+        self.receiver.register(framing.MSG_METRICS,
+                               lambda hdr, payload:
+                               self.docs.ingest_payload(payload.tobytes()))
         self.prom = PromPipeline()
         self.receiver.register(framing.MSG_PROMETHEUS,
                                lambda hdr, payload:
@@ -66,6 +71,8 @@ class DeepflowServer:
         self.system_rows = []  # deepflow_system self-metrics store
         self.receiver.register(framing.MSG_DFSTATS, self._on_dfstats)
         self.engine.system_rows = self.system_rows
+        self.engine.agent_app_rows = lambda: self.docs.app_rows
+        self.engine.agent_net_rows = lambda: self.docs.net_rows
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
                              promql=self.promql,

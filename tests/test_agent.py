@@ -129,3 +129,22 @@ def test_agent_to_server_end_to_end():
         a.close()
     finally:
         srv.stop()
+
+
+def test_agent_documents_queryable():
+    """Agent Document stream -> flow_metrics pipeline -> SQL."""
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 9,
+                         dict_capacity=1 << 10, time_base_s=0)
+    a = Agent(vtap_id=4)
+    for frame, ts in http_session(CLIENT, SERVER):
+        a.packet(frame, ts)
+    a.tick(10**9 * 100)
+    payload = a.drain(2)
+    hdr = framing.FrameHeader(msg_type=framing.MSG_METRICS, agent_id=4)
+    assert srv.receiver.handle_frame(framing.encode_frame(hdr, payload))
+    assert len(srv.docs.app_rows) >= 1
+    r = srv.engine.query(
+        "SELECT Sum(request) AS req, Max(rrt_max) AS m FROM application.agent")
+    assert r["values"][0][0] == 1
+    assert r["values"][0][1] == 4000
+    a.close()
