@@ -246,6 +246,63 @@ bool parse_mysql_request(const uint8_t* p, uint32_t n, std::string& stmt) {
     return false;
 }
 
+// PostgreSQL simple protocol: client 'Q' + int32 len + sql\0
+bool parse_pgsql_request(const uint8_t* p, uint32_t n, std::string& stmt) {
+    if (n < 6 || p[0] != 'Q') return false;
+    uint32_t mlen = (p[1] << 24) | (p[2] << 16) | (p[3] << 8) | p[4];
+    if (mlen < 5) return false;
+    uint32_t sl = mlen - 5;  // excludes type byte, includes \0
+    if (5 + sl > n) sl = n - 5;
+    while (sl && p[5 + sl - 1] == 0) sl--;
+    stmt.assign((const char*)p + 5, sl);
+    return true;
+}
+
+// Kafka request: int32 size, int16 api_key, int16 api_ver, int32 corr,
+// int16 client_id_len + client_id
+static const char* KAFKA_APIS[] = {"Produce", "Fetch", "ListOffsets",
+                                   "Metadata", "LeaderAndIsr", "StopReplica",
+                                   "UpdateMetadata", "ControlledShutdown",
+                                   "OffsetCommit", "OffsetFetch",
+                                   "FindCoordinator", "JoinGroup",
+                                   "Heartbeat", "LeaveGroup", "SyncGroup",
+                                   "DescribeGroups", "ListGroups",
+                                   "SaslHandshake", "ApiVersions"};
+
+bool parse_kafka_request(const uint8_t* p, uint32_t n, std::string& api,
+                         uint32_t& corr, std::string& client_id) {
+    if (n < 14) return false;
+    uint32_t size = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
+    if (size < 8 || size > (64u << 20)) return false;
+    uint16_t key = (p[4] << 8) | p[5];
+    if (key > 67) return false;
+    corr = (p[8] << 24) | (p[9] << 16) | (p[10] << 8) | p[11];
+    int16_t cl = (int16_t)((p[12] << 8) | p[13]);
+    if (cl > 0 && 14 + (uint32_t)cl <= n)
+        client_id.assign((const char*)p + 14, cl);
+    api = key < sizeof(KAFKA_APIS) / sizeof(char*)
+        ? KAFKA_APIS[key] : std::to_string(key);
+    return true;
+}
+
+// MongoDB wire header: int32 len, int32 reqid, int32 responseTo, int32 op
+bool parse_mongo_request(const uint8_t* p, uint32_t n, std::string& op,
+                         uint32_t& reqid) {
+    if (n < 16) return false;
+    uint32_t mlen = p[0] | (p[1] << 8) | (p[2] << 16) | (p[3] << 24);
+    if (mlen < 16 || mlen > (48u << 20)) return false;
+    uint32_t opc = p[12] | (p[13] << 8) | (p[14] << 16) | (p[15] << 24);
+    reqid = p[4] | (p[5] << 8) | (p[6] << 16) | (p[7] << 24);
+    switch (opc) {
+        case 2013: op = "OP_MSG"; break;
+        case 2004: op = "OP_QUERY"; break;
+        case 2010: op = "OP_COMMAND"; break;
+        case 1: op = "OP_REPLY"; return false;  // response
+        default: return false;
+    }
+    return true;
+}
+
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
 uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
@@ -255,6 +312,10 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     if (n >= 1 && (p[0] == '*' || p[0] == '+' || p[0] == '-' || p[0] == '$' ||
                    p[0] == ':') && (server_port == 6379)) return 80;
     if (server_port == 3306 && n >= 5) return 60;
+    if (server_port == 5432 && n >= 6 && (p[0] == 'Q' || p[0] == 'P'))
+        return 61;
+    if (server_port == 9092 && n >= 14) return 100;
+    if (server_port == 27017 && n >= 16) return 81;
     return 0;
 }
 
@@ -529,6 +590,72 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             bool err = p[0] == '-';
             encode_l7_record(a, f, f.l7.req_ts, ts, 0, err ? 3 : 0, f.l7, "");
             f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 61) {  // PostgreSQL
+        if (dir == 0) {
+            std::string stmt;
+            if (parse_pgsql_request(p, n, stmt)) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = n;
+                f.l7.req_type = "Query";
+                f.l7.resource = stmt;
+                f.l7.endpoint = "";
+                f.l7.domain = "";
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            }
+        } else if (dir == 1 && f.l7.active && n >= 1) {
+            bool err = p[0] == 'E';
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, err ? 3 : 0, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 100) {  // Kafka
+        if (dir == 0) {
+            std::string api, client_id;
+            uint32_t corr;
+            if (parse_kafka_request(p, n, api, corr, client_id)) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = n;
+                f.l7.req_type = api;
+                f.l7.resource = api;
+                f.l7.endpoint = api;
+                f.l7.domain = client_id;
+                f.l7.dns_id = corr;
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            }
+        } else if (dir == 1 && f.l7.active && n >= 8) {
+            uint32_t corr = (p[4] << 24) | (p[5] << 16) | (p[6] << 8) | p[7];
+            if (corr == f.l7.dns_id) {
+                encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+                f.l7.active = false;
+            }
+        }
+    } else if (f.l7_protocol == 81) {  // MongoDB
+        if (dir == 0) {
+            std::string op;
+            uint32_t reqid;
+            if (parse_mongo_request(p, n, op, reqid)) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = n;
+                f.l7.req_type = op;
+                f.l7.resource = op;
+                f.l7.endpoint = op;
+                f.l7.domain = "";
+                f.l7.dns_id = reqid;
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            }
+        } else if (dir == 1 && f.l7.active && n >= 16) {
+            uint32_t resp_to = p[8] | (p[9] << 8) | (p[10] << 16) |
+                               (p[11] << 24);
+            if (resp_to == f.l7.dns_id) {
+                encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+                f.l7.active = false;
+            }
         }
     } else if (f.l7_protocol == 60) {  // MySQL
         if (dir == 0) {

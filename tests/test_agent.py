@@ -148,3 +148,61 @@ def test_agent_documents_queryable():
     assert r["values"][0][0] == 1
     assert r["values"][0][1] == 4000
     a.close()
+
+
+def _tcp_exchange(agent_obj, sport, dport, req, resp, t0=10**9):
+    from deepflow_amd.agent.packets import eth_ipv4_tcp, SYN, SYNACK, PSH_ACK
+    pkts = [
+        (eth_ipv4_tcp(CLIENT, SERVER, sport, dport, SYN, 1), t0),
+        (eth_ipv4_tcp(SERVER, CLIENT, dport, sport, SYNACK, 2, 2),
+         t0 + 1_000_000),
+        (eth_ipv4_tcp(CLIENT, SERVER, sport, dport, PSH_ACK, 2, 3, req),
+         t0 + 2_000_000),
+        (eth_ipv4_tcp(SERVER, CLIENT, dport, sport, PSH_ACK, 3, 2 + len(req),
+                      resp), t0 + 5_000_000),
+    ]
+    for frame, ts in pkts:
+        agent_obj.packet(frame, ts)
+
+
+def test_pgsql(agent):
+    import struct
+    sql = b"SELECT 1"
+    req = b"Q" + struct.pack(">I", 4 + len(sql) + 1) + sql + b"\x00"
+    resp = b"T\x00\x00\x00\x06..."
+    _tcp_exchange(agent, 51000, 5432, req, resp)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 61
+    assert l7[0]["req"]["resource"] == "SELECT 1"
+
+
+def test_kafka(agent):
+    import struct
+    cid = b"myclient"
+    body = struct.pack(">hhI", 1, 11, 0xBEEF) + \
+        struct.pack(">h", len(cid)) + cid + b"restoffetch"
+    req = struct.pack(">I", len(body)) + body
+    resp_body = struct.pack(">I", 0xBEEF) + b"resp"
+    resp = struct.pack(">I", len(resp_body)) + resp_body
+    _tcp_exchange(agent, 52000, 9092, req, resp)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 100
+    assert l7[0]["req"]["req_type"] == "Fetch"
+    assert l7[0]["req"]["domain"] == "myclient"
+
+
+def test_mongodb(agent):
+    import struct
+    body = b"\x00" * 10
+    req = struct.pack("<iiii", 16 + len(body), 777, 0, 2013) + body
+    resp = struct.pack("<iiii", 16 + len(body), 1, 777, 1) + body
+    _tcp_exchange(agent, 53000, 27017, req, resp)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 81
+    assert l7[0]["req"]["req_type"] == "OP_MSG"
