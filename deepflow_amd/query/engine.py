@@ -162,6 +162,81 @@ class QueryEngine:
                                ["application.1s"], ["network.1s"]]}
         raise SqlError(f"unsupported show: {sql!r}")
 
+    # --------------------------------------------------- distributed hook
+    def query_partial(self, sql: str) -> Dict:
+        """Shard-local partial for distributed merge: aggregate queries
+        return hydrated group keys + RAW agg vectors (avg still split as
+        sum+count so cross-shard merge is exact); everything else returns
+        the finished local result under kind='rows'."""
+        stripped = sql.strip().lower()
+        m = _FROM_RE.search(sql)
+        table = m.group(1).lower() if m else "l7_flow_log"
+        if stripped.startswith("show") or table not in (
+                "l7_flow_log", "l4_flow_log"):
+            return {"kind": "rows", "result": self.query(sql)}
+        if table == "l4_flow_log" and self.l4 is None:
+            raise SqlError("l4_flow_log table not enabled")
+        if table == "l7_flow_log":
+            plan = parse_sql(sql, dictionary=self.pipe.dict,
+                             time_base_s=self.pipe.time_base_s,
+                             tags=L7_TAGS, metrics=L7_METRICS)
+            segments, tags, str_cols = (self.pipe.segments.segments, L7_TAGS,
+                                        S.STR_COLS)
+        else:
+            from ..store import l4_schema as L4S
+            plan = parse_sql(sql, dictionary=None,
+                             time_base_s=self.l4.time_base_s,
+                             tags=L4_TAGS, metrics=L4_METRICS)
+            segments, tags, str_cols = (self.l4.segments.segments, L4_TAGS,
+                                        L4S.STR_COLS)
+        if plan.select_rows:
+            return {"kind": "rows",
+                    "result": self._run_select(plan, segments, tags,
+                                               str_cols)}
+        groups = execute(plan, segments, self.device)
+        key_rows = []
+        aggs = []
+        for g in groups:
+            key_rows.append([self._hydrate(meta["hydrate"], g["key"][ki])
+                             for ki, meta in enumerate(plan.key_meta)])
+            aggs.append(g["agg"][: len(plan.aggs)])
+        return {
+            "kind": "agg",
+            "key_rows": key_rows,
+            "aggs": aggs,
+            "agg_ops": [a.op for a in plan.aggs],
+        }
+
+    def finalize_groups(self, sql: str, key_rows, aggs) -> Dict:
+        """Turn merged (hydrated keys, raw aggs) back into a result table
+        using the local plan (column names, avg division, order/limit)."""
+        m = _FROM_RE.search(sql)
+        table = m.group(1).lower() if m else "l7_flow_log"
+        if table == "l4_flow_log":
+            plan = parse_sql(sql, dictionary=None,
+                             time_base_s=self.l4.time_base_s,
+                             tags=L4_TAGS, metrics=L4_METRICS)
+        else:
+            plan = parse_sql(sql, dictionary=self.pipe.dict,
+                             time_base_s=self.pipe.time_base_s,
+                             tags=L7_TAGS, metrics=L7_METRICS)
+        columns = plan.key_names + plan.agg_names
+        rows = []
+        for key, agg in zip(key_rows, aggs):
+            row = list(key)
+            ai = 0
+            for meta in plan.agg_meta:
+                if meta["op"] == "avg":
+                    ssum, cnt = agg[ai], agg[ai + 1]
+                    ai += 2
+                    row.append(ssum / cnt if cnt else None)
+                else:
+                    row.append(agg[ai])
+                    ai += 1
+            rows.append(row)
+        rows = self._order_limit(plan, columns, rows)
+        return {"columns": columns, "values": rows}
+
     # ----------------------------------------------------------- segments
     def _run_segments(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:
         if plan.select_rows:
