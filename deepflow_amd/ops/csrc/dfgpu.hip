@@ -963,34 +963,92 @@ DEV bool eval_terms(const SegView& s, uint64_t row, const QuerySpec& q) {
 
 // group table: gkeys u64 hash (claim word), graw [cap, QMAX_KEYS] raw key
 // values, gvals [cap, QMAX_AGGS] u64 accumulators (min encoded as ~v).
+// group claim: load-first probe, CAS on empty (shared by both agg paths)
+DEV uint32_t group_claim(uint64_t h, const uint64_t* kraw, uint32_t n_keys,
+                         uint64_t* gkeys, uint64_t* graw,
+                         uint32_t cap_mask) {
+    uint32_t slot = (uint32_t)(h & cap_mask);
+    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
+        uint64_t cur = gkeys[slot];
+        if (cur == h) return slot;
+        if (cur == EMPTY_KEY) {
+            uint64_t old = atomicCAS((unsigned long long*)&gkeys[slot],
+                                     EMPTY_KEY, h);
+            if (old == EMPTY_KEY) {
+                for (uint32_t k = 0; k < n_keys; k++)
+                    graw[(uint64_t)slot * QMAX_KEYS + k] = kraw[k];
+                return slot;
+            }
+            if (old == h) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return 0;
+}
+
 __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_row,
                             uint64_t* __restrict__ gkeys,
                             uint64_t* __restrict__ graw,
                             unsigned long long* __restrict__ gvals,
                             uint32_t cap_mask) {
     uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
-    uint64_t row = base_row + i;
-    if (!eval_terms(s, row, q)) return;
+    // no early return: the whole wave stays converged so low-cardinality
+    // group keys take the wave-reduced path (one atomic per wave, not 64 —
+    // a lone accumulator otherwise serializes the whole scan)
+    uint64_t row = base_row + (i < n ? i : 0);
+    bool ok = i < n && eval_terms(s, row, q);
     uint64_t kraw[QMAX_KEYS];
     uint64_t h = 0x243F6A8885A308D3ull;
-    for (uint32_t k = 0; k < q.n_keys; k++) {
-        kraw[k] = src_value(s, row, q.keys[k].family, q.keys[k].idx,
-                            q.keys[k].bucket, q.time_base_s);
-        h = mix64(h ^ kraw[k] ^ ((uint64_t)k << 56));
-    }
-    if (h == EMPTY_KEY) h = 1;
-    uint32_t slot = (uint32_t)(h & cap_mask);
-    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
-        uint64_t old = atomicCAS((unsigned long long*)&gkeys[slot], EMPTY_KEY, h);
-        if (old == EMPTY_KEY) {
-            for (uint32_t k = 0; k < q.n_keys; k++)
-                graw[(uint64_t)slot * QMAX_KEYS + k] = kraw[k];
-            break;
+    if (ok) {
+        for (uint32_t k = 0; k < q.n_keys; k++) {
+            kraw[k] = src_value(s, row, q.keys[k].family, q.keys[k].idx,
+                                q.keys[k].bucket, q.time_base_s);
+            h = mix64(h ^ kraw[k] ^ ((uint64_t)k << 56));
         }
-        if (old == h) break;
-        slot = (slot + 1) & cap_mask;
+        if (h == EMPTY_KEY) h = 1;
     }
+    uint64_t act = __ballot(ok);
+    if (act == 0) return;
+    uint32_t leader = (uint32_t)__ffsll((unsigned long long)act) - 1;
+    uint64_t h0 = __shfl(h, leader);
+    bool uniform = __all(!ok || h == h0);
+    if (uniform) {
+        // one claim + one atomic per agg for the whole wave
+        uint32_t slot = 0;
+        if ((threadIdx.x & 63) == leader)
+            slot = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
+        slot = (uint32_t)__shfl((int)slot, leader);
+        unsigned long long* acc = &gvals[(uint64_t)slot * QMAX_AGGS];
+        for (uint32_t a = 0; a < q.n_aggs; a++) {
+            const QAgg& ag = q.aggs[a];
+            bool is_sum = ag.op == AGGOP_COUNT || ag.op == AGGOP_SUM;
+            uint64_t ident = ag.op == AGGOP_MIN ? ~0ull : 0ull;
+            uint64_t v = ident;
+            if (ok)
+                v = ag.op == AGGOP_COUNT ? 1
+                    : src_value(s, row, ag.family, ag.idx, 0, q.time_base_s);
+            // full-wave reduction (all 64 lanes converged)
+            for (int d = 32; d > 0; d >>= 1) {
+                uint64_t o = __shfl_xor((long long)v, d);
+                if (is_sum) v += o;
+                else if (ag.op == AGGOP_MIN) v = o < v ? o : v;
+                else v = o > v ? o : v;
+            }
+            if ((threadIdx.x & 63) == leader) {
+                if (is_sum) {
+                    // subtract the identity contributions? none: ident 0
+                    atomicAdd(&acc[a], (unsigned long long)v);
+                } else if (ag.op == AGGOP_MIN) {
+                    atomicMin(&acc[a], (unsigned long long)v);
+                } else {
+                    atomicMax(&acc[a], (unsigned long long)v);
+                }
+            }
+        }
+        return;
+    }
+    if (!ok) return;
+    uint32_t slot = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
     unsigned long long* acc = &gvals[(uint64_t)slot * QMAX_AGGS];
     for (uint32_t a = 0; a < q.n_aggs; a++) {
         const QAgg& ag = q.aggs[a];
