@@ -54,7 +54,8 @@ DEFAULT_AGENT_CONFIG = {
 
 class ControllerLite:
     def __init__(self, kg: Optional[KnowledgeGraphTable] = None,
-                 n_analyzers: int = 1):
+                 n_analyzers: int = 1, event_sink=None):
+        self.event_sink = event_sink
         self.agents: Dict[int, AgentRecord] = {}
         self.group_configs: Dict[str, dict] = {"default": dict(DEFAULT_AGENT_CONFIG)}
         self.config_version = 1
@@ -115,10 +116,17 @@ class ControllerLite:
                         ) -> None:
         """Cloud/genesis-style inventory update -> bump version, refresh the
         GPU KnowledgeGraph table and tagrecorder name maps."""
+        new_keys = [k for k in entries if k not in self.platform]
         self.platform.update(entries)
         self.platform_version += 1
         if self.kg is not None:
             self.kg.update(entries)
+        if self.event_sink is not None:
+            for (epc, ip_) in new_keys:
+                info = entries[(epc, ip_)]
+                self.event_sink("create", "pod" if info.pod_id else "device",
+                                info.pod_id or info.l3_device_id,
+                                resource_name="", description=f"epc={epc}")
         if names:
             for m, d in names.items():
                 self.name_maps.setdefault(m, {}).update(d)

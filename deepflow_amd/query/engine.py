@@ -96,6 +96,14 @@ class QueryEngine:
             from ..store import l4_schema as L4S
             return self._run_segments(plan, self.l4.segments.segments,
                                       L4_TAGS, L4S.STR_COLS)
+        row_tables = {
+            "event": "event_rows", "perf_event": "perf_event_rows",
+            "alert_event": "alert_event_rows",
+            "application_log": "app_log_rows", "log": "app_log_rows",
+        }
+        if table in row_tables:
+            rows = getattr(self, row_tables[table], lambda: [])()
+            return self._run_rows(sql, rows, time_base_s=0)
         if table == "application.agent":
             rows = getattr(self, "agent_app_rows", lambda: [])()
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)

@@ -66,18 +66,50 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_OPENTELEMETRY, self._on_otel)
         self.receiver.register(framing.MSG_OPENTELEMETRY_COMPRESSED,
                                self._on_otel)
+        from .ingest.event_pipeline import EventPipeline
+        from .ingest.applog_pipeline import AppLogPipeline
+        self.events = EventPipeline()
+        self.applogs = AppLogPipeline()
+        self.receiver.register(framing.MSG_PROC_EVENT,
+                               lambda hdr, payload:
+                               self.events.ingest_proc_events(
+                                   payload.tobytes()))
+        for mt in (framing.MSG_APPLICATION_LOG, framing.MSG_SYSLOG,
+                   framing.MSG_AGENT_LOG):
+            self.receiver.register(
+                mt, lambda hdr, payload:
+                self.applogs.ingest_lines(payload.tobytes(),
+                                          agent_id=hdr.agent_id))
         from .control import ControllerLite
-        self.controller = ControllerLite(kg=self.kg)
+        self.controller = ControllerLite(
+            kg=self.kg, event_sink=self.events.add_resource_event)
         self.system_rows = []  # deepflow_system self-metrics store
         self.receiver.register(framing.MSG_DFSTATS, self._on_dfstats)
         self.engine.system_rows = self.system_rows
         self.engine.agent_app_rows = lambda: self.docs.app_rows
         self.engine.agent_net_rows = lambda: self.docs.net_rows
+        self.engine.event_rows = lambda: self.events.resource_events
+        self.engine.perf_event_rows = lambda: self.events.perf_events
+        self.engine.alert_event_rows = lambda: self.events.alert_events
+        self.engine.app_log_rows = lambda: self.applogs.rows
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
                              promql=self.promql,
                              profile=ProfileApp(self.profiles))
         self.controller.register(self.app)
+        from .export import OtlpExporter
+        self.exporter = OtlpExporter(self.engine)
+        from .query.mcp import McpServer
+        self.mcp = McpServer(self.engine, self.profiles)
+        self.mcp.register(self.app)
+
+        @self.app.get("/v1/export/otlp")
+        def export_otlp(where: str = "", limit: int = 10000):
+            from fastapi.responses import Response
+            blob = self.exporter.export_where(where, limit)
+            return Response(content=blob,
+                            media_type="application/x-protobuf")
+
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
