@@ -55,13 +55,23 @@ class L4IngestPipeline:
         self.counter.add("flows_in", n)
         return n
 
+    def _scratch(self, n: int, dev):
+        if getattr(self, "_scratch_str", None) is None or \
+                self._scratch_str.shape[1] < n:
+            self._scratch_str = torch.zeros((L4.N_STR, n), dtype=torch.int64,
+                                            device=dev)
+        else:
+            self._scratch_str.zero_()
+        return self._scratch_str
+
     def _ingest_gpu(self, payload, offs, lens, seg, base, n) -> None:
         from ..ops import gpu_ops
         dev = torch.device(self.device)
         payload_t = torch.from_numpy(payload).to(dev, non_blocking=True)
         offs_t = torch.from_numpy(offs.view(np.int32)).to(dev, non_blocking=True)
         lens_t = torch.from_numpy(lens.view(np.int32)).to(dev, non_blocking=True)
-        gpu_ops.decode_l4(payload_t, offs_t, lens_t, seg, base)
+        sstr = self._scratch(n, dev)
+        gpu_ops.decode_l4(payload_t, offs_t, lens_t, seg, base, sstr)
         epc0 = seg.u32[L4.U32_COLS.index("l3_epc_id_0"), base:base + n]
         ip0 = seg.u32[L4.U32_COLS.index("ip4_0"), base:base + n]
         epc1 = seg.u32[L4.U32_COLS.index("l3_epc_id_1"), base:base + n]
@@ -69,7 +79,7 @@ class L4IngestPipeline:
         gpu_ops.kg_probe_cols(epc0, ip0, epc1, ip1, n, self.kg.tkeys,
                               self.kg.tvals, seg.kg, seg.capacity, base)
         # pool the request_domain strings
-        refs = seg.strref[0, base:base + n]
+        refs = sstr[0, :n]
         lens64 = (refs & 0xFFFF).to(torch.int64)
         cum = torch.cumsum(lens64, 0)
         total = int(cum[-1].item())
@@ -78,7 +88,7 @@ class L4IngestPipeline:
             starts = cum - lens64
             pool_cols = torch.zeros(1, dtype=torch.uint8, device=dev)
             gpu_ops.pool_gather(payload_t, seg, pool_cols, base, n, starts,
-                                seg.pool, seg.pool_len)
+                                seg.pool, seg.pool_len, sstr)
         gpu_ops.agg_net1s(seg, base, n, self.time_base_s,
                           self.metrics.tkeys, self.metrics.tvals)
         seg.pool_len += total
@@ -86,7 +96,8 @@ class L4IngestPipeline:
     def _ingest_cpu(self, payload, offs, lens, seg, base, n) -> None:
         from ..ops import ref_l4, ref
         pbytes = payload.tobytes()
-        ref_l4.decode_l4_ref(pbytes, offs, lens, seg, base)
+        sstr = self._scratch(n, torch.device("cpu"))
+        ref_l4.decode_l4_ref(pbytes, offs, lens, seg, base, sstr)
         # KG join via the host mirror (independent oracle for the GPU probe)
         for i in range(n):
             row = base + i
@@ -100,12 +111,13 @@ class L4IngestPipeline:
                     seg.kg[side * S.N_KG + j, row] = v
         row_len = torch.zeros(n, dtype=torch.int64)
         for i in range(n):
-            row_len[i] = int(seg.strref[0, base + i].item()) & 0xFFFF
+            row_len[i] = int(sstr[0, i].item()) & 0xFFFF
         cum = torch.cumsum(row_len, 0)
         total = int(cum[-1].item()) if n else 0
         seg.ensure_pool(total)
         starts = cum - row_len
-        ref.pool_gather_ref(pbytes, seg, [0], base, n, starts, seg.pool_len)
+        ref.pool_gather_ref(pbytes, seg, [0], base, n, starts, seg.pool_len,
+                            sstr)
         ref_l4.agg_net1s_ref(seg, base, n, self.time_base_s,
                              self.metrics.table)
         seg.pool_len += total

@@ -70,6 +70,10 @@ class L7IngestPipeline:
             dtype=torch.uint8, device=dev)
         self._pool_cols = torch.tensor(S.POOL_COLS, dtype=torch.uint8,
                                        device=dev)
+        # per-batch scratch (string refs are transient; only pooled refs
+        # persist into segments)
+        self._scratch_str = None
+        self._scratch_attr = None
 
     # ------------------------------------------------------------------
     def ingest(self, payload: np.ndarray, offs: np.ndarray,
@@ -117,26 +121,38 @@ class L7IngestPipeline:
         lens_t = torch.from_numpy(lens.view(np.int32)).to(dev, non_blocking=True)
         self._run_gpu(payload_t, offs_t, lens_t, payload, seg, base, n)
 
+    def _scratch(self, n: int, dev):
+        if self._scratch_str is None or self._scratch_str.shape[1] < n:
+            self._scratch_str = torch.zeros((S.N_STR, n), dtype=torch.int64,
+                                            device=dev)
+            self._scratch_attr = torch.zeros((2 * S.MAX_ATTRS, n),
+                                             dtype=torch.int64, device=dev)
+        else:
+            self._scratch_str.zero_()
+            self._scratch_attr.zero_()
+        return self._scratch_str, self._scratch_attr
+
     def _run_gpu(self, payload_t, offs_t, lens_t, payload_host,
                  seg: L7Segment, base: int, n: int) -> None:
         from ..ops import gpu_ops
         dev = payload_t.device
-        gpu_ops.decode_l7(payload_t, offs_t, lens_t, seg, base)
+        sstr, sattr = self._scratch(n, dev)
+        gpu_ops.decode_l7(payload_t, offs_t, lens_t, seg, base, sstr, sattr)
         gpu_ops.kg_probe(seg, base, n, self.kg.tkeys, self.kg.tvals)
-        gpu_ops.intern_many(payload_t, seg.strref, self._ref_rows_scalar,
-                            self._dom_scalar, base, n, self.dict.tkeys,
+        gpu_ops.intern_many(payload_t, sstr, self._ref_rows_scalar,
+                            self._dom_scalar, 0, n, self.dict.tkeys,
                             self.dict.emit, self.dict.emit_ctr, seg.did, base)
         gpu_ops.intern_attrs(payload_t, seg, base, n, self.dict.tkeys,
-                             self.dict.emit, self.dict.emit_ctr)
+                             self.dict.emit, self.dict.emit_ctr, sattr)
         # pool sizing: lens kernel -> cumsum -> (sync) total
         row_len = torch.zeros(n, dtype=torch.int32, device=dev)
-        gpu_ops.pool_lens(seg, self._pool_cols, base, n, row_len)
+        gpu_ops.pool_lens(sstr, self._pool_cols, n, row_len)
         cum = torch.cumsum(row_len.to(torch.int64), 0)
         total = int(cum[-1].item())
         row_start = cum - row_len.to(torch.int64)
         seg.ensure_pool(total)
         gpu_ops.pool_gather(payload_t, seg, self._pool_cols, base, n,
-                            row_start, seg.pool, seg.pool_len)
+                            row_start, seg.pool, seg.pool_len, sstr)
         gpu_ops.agg_app1s(seg, base, n, self.time_base_s,
                           self.metrics.tkeys, self.metrics.tvals)
         new = self.dict.harvest(payload_host)  # syncs emit buffer
@@ -149,23 +165,24 @@ class L7IngestPipeline:
                     n: int) -> None:
         from ..ops import ref
         pb = payload.tobytes()
-        ref.decode_l7_ref(pb, offs, lens, seg, base)
+        sstr, sattr = self._scratch(n, torch.device("cpu"))
+        ref.decode_l7_ref(pb, offs, lens, seg, base, sstr, sattr)
         ref.kg_probe_ref(seg, base, n, self.kg.tkeys, self.kg.tvals)
-        new = ref.intern_ref(pb, seg.strref, _SCALAR_DICT_REF_ROWS,
-                             _SCALAR_DICT_DOMAINS, base, n, self.dict.tkeys,
+        new = ref.intern_ref(pb, sstr, _SCALAR_DICT_REF_ROWS,
+                             _SCALAR_DICT_DOMAINS, 0, n, self.dict.tkeys,
                              seg.did, base, dictionary=self.dict)
         new += ref.intern_ref(
-            pb, seg.attr_ref, list(range(2 * S.MAX_ATTRS)),
+            pb, sattr, list(range(2 * S.MAX_ATTRS)),
             [S.DICT_DOM_ATTR_NAME] * S.MAX_ATTRS +
             [S.DICT_DOM_ATTR_VALUE] * S.MAX_ATTRS,
-            base, n, self.dict.tkeys, seg.attr_id, base, dictionary=self.dict)
-        row_len = ref.pool_lens_ref(seg, S.POOL_COLS, base, n)
+            0, n, self.dict.tkeys, seg.attr_id, base, dictionary=self.dict)
+        row_len = ref.pool_lens_ref(sstr, S.POOL_COLS, n)
         cum = torch.cumsum(row_len.to(torch.int64), 0)
         total = int(cum[-1].item()) if n else 0
         row_start = cum - row_len.to(torch.int64)
         seg.ensure_pool(total)
         ref.pool_gather_ref(pb, seg, S.POOL_COLS, base, n, row_start,
-                            seg.pool_len)
+                            seg.pool_len, sstr)
         ref.agg_app1s_ref(seg, base, n, self.time_base_s, self.metrics.table)
         seg.pool_len += total
         self.stats.dict_new += len(new)

@@ -85,17 +85,18 @@ struct L7Cols {
     uint64_t* u64c;       // [L7_U64_N, stride]
     uint32_t* u32c;       // [L7_U32_N, stride]
     uint8_t* u8c;         // [L7_U8_N, stride]
-    uint64_t* strc;       // [L7_STR_N, stride]  batch-relative packed refs
-    uint64_t* attrc;      // [2*L7_MAX_ATTRS, stride] name refs then value refs
-    uint8_t* attr_cnt;    // [stride]
+    uint64_t* strc;       // SCRATCH [L7_STR_N, scratch_stride], row = rid
+    uint64_t* attrc;      // SCRATCH [2*L7_MAX_ATTRS, scratch_stride]
+    uint8_t* attr_cnt;    // [stride] (segment)
     uint64_t stride;
     uint64_t base_row;
+    uint64_t scratch_stride;
 };
 
 #define W64(c, v) cols.u64c[(uint64_t)(c) * cols.stride + row] = (v)
 #define W32(c, v) cols.u32c[(uint64_t)(c) * cols.stride + row] = (uint32_t)(v)
 #define W8(c, v)  cols.u8c[(uint64_t)(c) * cols.stride + row] = (uint8_t)(v)
-#define WSTR(c, off, len) cols.strc[(uint64_t)(c) * cols.stride + row] = STR_REF_PACK(off, len)
+#define WSTR(c, off, len) cols.strc[(uint64_t)(c) * cols.scratch_stride + rid] = STR_REF_PACK(off, len)
 
 __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                             const uint32_t* __restrict__ offs,
@@ -272,13 +273,13 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                                 case 10: WSTR(L7_STR_XREQ_1, p2, l3); break;
                                 case 16:
                                     if (n_names < L7_MAX_ATTRS)
-                                        cols.attrc[(uint64_t)n_names * cols.stride + row] =
+                                        cols.attrc[(uint64_t)n_names * cols.scratch_stride + rid] =
                                             STR_REF_PACK(p2, l3);
                                     n_names++;
                                     break;
                                 case 17:
                                     if (n_vals < L7_MAX_ATTRS)
-                                        cols.attrc[(uint64_t)(L7_MAX_ATTRS + n_vals) * cols.stride + row] =
+                                        cols.attrc[(uint64_t)(L7_MAX_ATTRS + n_vals) * cols.scratch_stride + rid] =
                                             STR_REF_PACK(p2, l3);
                                     n_vals++;
                                     break;
@@ -312,9 +313,10 @@ struct L4Cols {
     uint64_t* u64c;
     uint32_t* u32c;
     uint8_t* u8c;
-    uint64_t* strc;
+    uint64_t* strc;   // SCRATCH [L4_STR_N, scratch_stride], row = rid
     uint64_t stride;
     uint64_t base_row;
+    uint64_t scratch_stride;
 };
 
 #define L4W64(c, v) cols.u64c[(uint64_t)(c) * cols.stride + row] = (v)
@@ -510,7 +512,7 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                     break;
                 }
                 case 26:
-                    cols.strc[(uint64_t)L4_STR_REQUEST_DOMAIN * cols.stride + row] =
+                    cols.strc[(uint64_t)L4_STR_REQUEST_DOMAIN * cols.scratch_stride + rid] =
                         STR_REF_PACK(sub, ln);
                     break;
                 default: break;
@@ -733,9 +735,10 @@ __global__ void k_intern_many(const uint8_t* __restrict__ payload,
 // (name, value) pairs — avoids launching MAX_ATTRS*2 threads per row when
 // the typical count is ~4. Names are wave-uniform per iteration.
 __global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
-                               const uint64_t* __restrict__ attr_refs,  // [2*MAX, stride]
+                               const uint64_t* __restrict__ attr_refs,  // SCRATCH [2*MAX, ref_stride]
                                const uint8_t* __restrict__ attr_cnt,
                                uint32_t n, uint64_t stride, uint64_t base_row,
+                               uint64_t ref_stride,
                                uint64_t* __restrict__ tkeys, uint32_t cap_mask,
                                uint64_t* __restrict__ emit,
                                uint32_t* __restrict__ emit_ctr, uint32_t emit_cap,
@@ -751,7 +754,7 @@ __global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
     for (uint32_t a = 0; a < maxc; a++) {
         bool act = in_range && a < cnt;
         // names (wave-uniform in the common schema-stable case)
-        uint64_t nref = act ? attr_refs[(uint64_t)a * stride + row] : 0;
+        uint64_t nref = act ? attr_refs[(uint64_t)a * ref_stride + i] : 0;
         uint32_t nlen = STR_REF_LEN(nref);
         uint64_t h = 0;
         if (act && nlen)
@@ -765,7 +768,7 @@ __global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
                 nlen ? slot : DICT_ID_INVALID;
         // values (high cardinality -> per-lane probes)
         uint64_t vref = act
-            ? attr_refs[(uint64_t)(L7_MAX_ATTRS + a) * stride + row] : 0;
+            ? attr_refs[(uint64_t)(L7_MAX_ATTRS + a) * ref_stride + i] : 0;
         uint32_t vlen = STR_REF_LEN(vref);
         if (act) {
             uint32_t vslot = DICT_ID_INVALID;
@@ -800,22 +803,24 @@ __global__ void k_pool_lens(const uint64_t* __restrict__ strc,  // [L7_STR_N, st
 }
 
 __global__ void k_pool_gather(const uint8_t* __restrict__ payload,
-                              uint64_t* __restrict__ strc,
+                              const uint64_t* __restrict__ strc,  // scratch
                               const uint8_t* __restrict__ pool_cols,
                               uint32_t npc, uint32_t n,
                               uint64_t stride, uint64_t base_row,
                               const uint64_t* __restrict__ row_start,  // exclusive cumsum
-                              uint8_t* __restrict__ pool, uint64_t pool_base) {
+                              uint8_t* __restrict__ pool, uint64_t pool_base,
+                              uint64_t* __restrict__ out_refs,  // [npc, out_stride] segment
+                              uint64_t out_stride, uint64_t out_base_row) {
     uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t dst = pool_base + row_start[i];
     for (uint32_t c = 0; c < npc; c++) {
-        uint64_t* ref = &strc[(uint64_t)pool_cols[c] * stride + base_row + i];
-        uint64_t r = *ref;
+        uint64_t r = strc[(uint64_t)pool_cols[c] * stride + base_row + i];
         uint32_t len = STR_REF_LEN(r);
         uint64_t src = STR_REF_OFF(r);
         for (uint32_t b = 0; b < len; b++) pool[dst + b] = payload[src + b];
-        *ref = STR_REF_PACK(dst, len);
+        out_refs[(uint64_t)c * out_stride + out_base_row + i] =
+            STR_REF_PACK(dst, len);
         dst += len;
     }
 }
@@ -907,7 +912,7 @@ struct SegView {
     const uint32_t* kgc;     // [2*KG_VALS_N, stride]
     const uint32_t* attrid;  // [2*L7_MAX_ATTRS, stride] interned attr name/val ids
     const uint8_t* attr_cnt;
-    const uint64_t* strc;    // [N_STR, stride] pool-relative refs
+    const uint64_t* strc;    // [n_pool_cols, stride] pool-relative refs (pool position = idx)
     const uint8_t* pool;     // segment string pool
     uint64_t stride;
     uint64_t n_rows;
@@ -1096,9 +1101,11 @@ int df_decode_l7(const void* payload, const void* offs, const void* lens,
                  uint32_t n,
                  void* u64c, void* u32c, void* u8c, void* strc,
                  void* attrc, void* attr_cnt,
-                 uint64_t stride, uint64_t base_row, uint64_t stream) {
+                 uint64_t stride, uint64_t base_row, uint64_t scratch_stride,
+                 uint64_t stream) {
     L7Cols cols{(uint64_t*)u64c, (uint32_t*)u32c, (uint8_t*)u8c, (uint64_t*)strc,
-                (uint64_t*)attrc, (uint8_t*)attr_cnt, stride, base_row};
+                (uint64_t*)attrc, (uint8_t*)attr_cnt, stride, base_row,
+                scratch_stride};
     hipLaunchKernelGGL(k_decode_l7, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
                        (const uint8_t*)payload, (const uint32_t*)offs,
                        (const uint32_t*)lens, n, cols);
@@ -1107,9 +1114,10 @@ int df_decode_l7(const void* payload, const void* offs, const void* lens,
 
 int df_decode_l4(const void* payload, const void* offs, const void* lens,
                  uint32_t n, void* u64c, void* u32c, void* u8c, void* strc,
-                 uint64_t stride, uint64_t base_row, uint64_t stream) {
+                 uint64_t stride, uint64_t base_row, uint64_t scratch_stride,
+                 uint64_t stream) {
     L4Cols cols{(uint64_t*)u64c, (uint32_t*)u32c, (uint8_t*)u8c,
-                (uint64_t*)strc, stride, base_row};
+                (uint64_t*)strc, stride, base_row, scratch_stride};
     hipLaunchKernelGGL(k_decode_l4, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
                        (const uint8_t*)payload, (const uint32_t*)offs,
                        (const uint32_t*)lens, n, cols);
@@ -1166,14 +1174,15 @@ int df_intern_many(const void* payload, const void* refs, const void* ref_rows,
 
 int df_intern_attrs(const void* payload, const void* attr_refs,
                     const void* attr_cnt, uint32_t n, uint64_t stride,
-                    uint64_t base_row, void* tkeys, uint32_t cap,
+                    uint64_t base_row, uint64_t ref_stride,
+                    void* tkeys, uint32_t cap,
                     void* emit, void* emit_ctr, uint32_t emit_cap,
                     void* out_ids, uint64_t stream) {
     hipLaunchKernelGGL(k_intern_attrs, dim3(grid_for(n)), dim3(BLOCK), 0,
                        STREAM(stream),
                        (const uint8_t*)payload, (const uint64_t*)attr_refs,
                        (const uint8_t*)attr_cnt, n, stride, base_row,
-                       (uint64_t*)tkeys, cap - 1,
+                       ref_stride, (uint64_t*)tkeys, cap - 1,
                        (uint64_t*)emit, (uint32_t*)emit_ctr, emit_cap,
                        (uint32_t*)out_ids);
     return (int)hipGetLastError();
@@ -1187,14 +1196,16 @@ int df_pool_lens(const void* strc, const void* pool_cols, uint32_t npc, uint32_t
     return (int)hipGetLastError();
 }
 
-int df_pool_gather(const void* payload, void* strc, const void* pool_cols,
+int df_pool_gather(const void* payload, const void* strc, const void* pool_cols,
                    uint32_t npc, uint32_t n, uint64_t stride, uint64_t base_row,
                    const void* row_start, void* pool, uint64_t pool_base,
+                   void* out_refs, uint64_t out_stride, uint64_t out_base_row,
                    uint64_t stream) {
     hipLaunchKernelGGL(k_pool_gather, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
-                       (const uint8_t*)payload, (uint64_t*)strc,
+                       (const uint8_t*)payload, (const uint64_t*)strc,
                        (const uint8_t*)pool_cols, npc, n, stride, base_row,
-                       (const uint64_t*)row_start, (uint8_t*)pool, pool_base);
+                       (const uint64_t*)row_start, (uint8_t*)pool, pool_base,
+                       (uint64_t*)out_refs, out_stride, out_base_row);
     return (int)hipGetLastError();
 }
 

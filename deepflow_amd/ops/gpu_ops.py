@@ -18,25 +18,28 @@ def _stream() -> int:
 
 
 def decode_l7(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
-              seg, base_row: int) -> None:
+              seg, base_row: int, scratch_str: torch.Tensor,
+              scratch_attr: torch.Tensor) -> None:
     n = offs.numel()
     lib = native.gpu()
     native.check(lib.df_decode_l7(
         payload.data_ptr(), offs.data_ptr(), lens.data_ptr(), n,
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.strref.data_ptr(), seg.attr_ref.data_ptr(), seg.attr_cnt.data_ptr(),
-        seg.capacity, base_row, _stream()), "df_decode_l7")
+        scratch_str.data_ptr(), scratch_attr.data_ptr(),
+        seg.attr_cnt.data_ptr(),
+        seg.capacity, base_row, scratch_str.shape[1], _stream()),
+        "df_decode_l7")
 
 
 def decode_l4(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
-              seg, base_row: int) -> None:
+              seg, base_row: int, scratch_str: torch.Tensor) -> None:
     n = offs.numel()
     lib = native.gpu()
     native.check(lib.df_decode_l4(
         payload.data_ptr(), offs.data_ptr(), lens.data_ptr(), n,
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.strref.data_ptr(), seg.capacity, base_row, _stream()),
-        "df_decode_l4")
+        scratch_str.data_ptr(), seg.capacity, base_row,
+        scratch_str.shape[1], _stream()), "df_decode_l4")
 
 
 def agg_net1s(seg, base_row: int, n: int, time_base_s: int,
@@ -104,31 +107,36 @@ def intern_many(payload: torch.Tensor, refs: torch.Tensor,
 
 def intern_attrs(payload: torch.Tensor, seg, base_row: int, n: int,
                  tkeys: torch.Tensor, emit: torch.Tensor,
-                 emit_ctr: torch.Tensor) -> None:
+                 emit_ctr: torch.Tensor, scratch_attr: torch.Tensor) -> None:
     lib = native.gpu()
     native.check(lib.df_intern_attrs(
-        payload.data_ptr(), seg.attr_ref.data_ptr(), seg.attr_cnt.data_ptr(),
-        n, seg.capacity, base_row, tkeys.data_ptr(), tkeys.numel(),
+        payload.data_ptr(), scratch_attr.data_ptr(), seg.attr_cnt.data_ptr(),
+        n, seg.capacity, base_row, scratch_attr.shape[1],
+        tkeys.data_ptr(), tkeys.numel(),
         emit.data_ptr(), emit_ctr.data_ptr(), emit.shape[0],
         seg.attr_id.data_ptr(), _stream()), "df_intern_attrs")
 
 
-def pool_lens(seg, pool_cols: torch.Tensor, base_row: int, n: int,
+def pool_lens(scratch_str: torch.Tensor, pool_cols: torch.Tensor, n: int,
               row_len: torch.Tensor) -> None:
     lib = native.gpu()
     native.check(lib.df_pool_lens(
-        seg.strref.data_ptr(), pool_cols.data_ptr(), pool_cols.numel(), n,
-        seg.capacity, base_row, row_len.data_ptr(), _stream()), "df_pool_lens")
+        scratch_str.data_ptr(), pool_cols.data_ptr(), pool_cols.numel(), n,
+        scratch_str.shape[1], 0, row_len.data_ptr(), _stream()),
+        "df_pool_lens")
 
 
 def pool_gather(payload: torch.Tensor, seg, pool_cols: torch.Tensor,
                 base_row: int, n: int, row_start: torch.Tensor,
-                pool: torch.Tensor, pool_base: int) -> None:
+                pool: torch.Tensor, pool_base: int,
+                scratch_str: torch.Tensor) -> None:
     lib = native.gpu()
     native.check(lib.df_pool_gather(
-        payload.data_ptr(), seg.strref.data_ptr(), pool_cols.data_ptr(),
-        pool_cols.numel(), n, seg.capacity, base_row, row_start.data_ptr(),
-        pool.data_ptr(), pool_base, _stream()), "df_pool_gather")
+        payload.data_ptr(), scratch_str.data_ptr(), pool_cols.data_ptr(),
+        pool_cols.numel(), n, scratch_str.shape[1], 0, row_start.data_ptr(),
+        pool.data_ptr(), pool_base,
+        seg.poolref.data_ptr(), seg.capacity, base_row, _stream()),
+        "df_pool_gather")
 
 
 def agg_app1s(seg, base_row: int, n: int, time_base_s: int,
@@ -155,7 +163,7 @@ def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
     native.check(lib.df_query_agg(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
         _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
-        _opt_ptr(seg, "attr_cnt"), seg.strref.data_ptr(),
+        _opt_ptr(seg, "attr_cnt"), seg.poolref.data_ptr(),
         seg.pool.data_ptr(), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         gkeys.data_ptr(), graw.data_ptr(), gvals.data_ptr(), gkeys.numel(),
@@ -170,7 +178,7 @@ def query_select(seg, spec_bytes: bytes, base_row: int, n: int,
     native.check(lib.df_query_select(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
         _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
-        _opt_ptr(seg, "attr_cnt"), seg.strref.data_ptr(),
+        _opt_ptr(seg, "attr_cnt"), seg.poolref.data_ptr(),
         seg.pool.data_ptr(), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         out_rows.data_ptr(), out_ctr.data_ptr(), out_rows.numel(), _stream()),

@@ -79,9 +79,9 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     u64, u32, p = ct.c_uint64, ct.c_uint32, ct.c_void_p
     lib.df_gpu_ready.restype = ct.c_int
     lib.df_decode_l7.restype = ct.c_int
-    lib.df_decode_l7.argtypes = [p, p, p, u32, p, p, p, p, p, p, u64, u64, u64]
+    lib.df_decode_l7.argtypes = [p, p, p, u32, p, p, p, p, p, p, u64, u64, u64, u64]
     lib.df_decode_l4.restype = ct.c_int
-    lib.df_decode_l4.argtypes = [p, p, p, u32, p, p, p, p, u64, u64, u64]
+    lib.df_decode_l4.argtypes = [p, p, p, u32, p, p, p, p, u64, u64, u64, u64]
     lib.df_agg_net1s.restype = ct.c_int
     lib.df_agg_net1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
     lib.df_kg_build.restype = ct.c_int
@@ -92,12 +92,12 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_intern_many.argtypes = [p, p, p, p, u32, u32, u64, u64, p, u32,
                                    p, p, u32, p, u64, u64, u64]
     lib.df_intern_attrs.restype = ct.c_int
-    lib.df_intern_attrs.argtypes = [p, p, p, u32, u64, u64, p, u32,
+    lib.df_intern_attrs.argtypes = [p, p, p, u32, u64, u64, u64, p, u32,
                                     p, p, u32, p, u64]
     lib.df_pool_lens.restype = ct.c_int
     lib.df_pool_lens.argtypes = [p, p, u32, u32, u64, u64, p, u64]
     lib.df_pool_gather.restype = ct.c_int
-    lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, u64]
+    lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, p, u64, u64, u64]
     lib.df_agg_app1s.restype = ct.c_int
     lib.df_agg_app1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
     lib.df_query_agg.restype = ct.c_int

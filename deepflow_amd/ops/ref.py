@@ -62,8 +62,14 @@ def _w(seg, fam, col, row, v):
         t[idx, row] = v & 0xFF
 
 
-def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
+def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int,
+                  sstr=None, sattr=None) -> None:
     mv = memoryview(payload)
+    if sstr is None:
+        import torch
+        sstr = torch.zeros((S.N_STR, len(offs)), dtype=torch.int64)
+        sattr = torch.zeros((2 * S.MAX_ATTRS, len(offs)), dtype=torch.int64)
+    seg._last_sstr, seg._last_sattr = sstr, sattr
     for rid in range(len(offs)):
         row = base_row + rid
         pos = int(offs[rid])
@@ -119,10 +125,10 @@ def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
                                     else:
                                         p3 = e3
                             elif n2 == 27:
-                                seg.strref[_STR_IDX["process_kname_0"], row] = \
+                                sstr[_STR_IDX["process_kname_0"], rid] = \
                                     S.str_ref_pack(s3, l3)
                             elif n2 == 28:
-                                seg.strref[_STR_IDX["process_kname_1"], row] = \
+                                sstr[_STR_IDX["process_kname_1"], rid] = \
                                     S.str_ref_pack(s3, l3)
                         elif w2 == 1:
                             p2 += 8
@@ -154,16 +160,16 @@ def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
                             l3, p2 = read_varint(mv, p2)
                             if num == 15 and n2 == 16:
                                 if n_names < S.MAX_ATTRS:
-                                    seg.attr_ref[n_names, row] = \
+                                    sattr[n_names, rid] = \
                                         S.str_ref_pack(p2, l3)
                                 n_names += 1
                             elif num == 15 and n2 == 17:
                                 if n_vals < S.MAX_ATTRS:
-                                    seg.attr_ref[S.MAX_ATTRS + n_vals, row] = \
+                                    sattr[S.MAX_ATTRS + n_vals, rid] = \
                                         S.str_ref_pack(p2, l3)
                                 n_vals += 1
                             elif n2 in strmap:
-                                seg.strref[_STR_IDX[strmap[n2]], row] = \
+                                sstr[_STR_IDX[strmap[n2]], rid] = \
                                     S.str_ref_pack(p2, l3)
                             p2 += l3
                         elif w2 == 1:
@@ -171,9 +177,9 @@ def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
                         elif w2 == 5:
                             p2 += 4
                 elif num == 13:
-                    seg.strref[_STR_IDX["version"], row] = S.str_ref_pack(sub, ln)
+                    sstr[_STR_IDX["version"], rid] = S.str_ref_pack(sub, ln)
                 elif num == 21:
-                    seg.strref[_STR_IDX["biz_code"], row] = S.str_ref_pack(sub, ln)
+                    sstr[_STR_IDX["biz_code"], rid] = S.str_ref_pack(sub, ln)
             elif wt == 1:
                 pos += 8
             elif wt == 5:
@@ -282,28 +288,28 @@ def intern_ref(payload: bytes, refs: torch.Tensor, ref_rows, domains,
 
 # -------------------------------------------------------------- K4 pool
 
-def pool_lens_ref(seg, pool_cols, base_row: int, n: int) -> torch.Tensor:
+def pool_lens_ref(sstr, pool_cols, n: int) -> torch.Tensor:
     out = torch.zeros(n, dtype=torch.int32)
     for i in range(n):
         total = 0
         for c in pool_cols:
-            total += int(seg.strref[c, base_row + i].item()) & 0xFFFF
+            total += int(sstr[c, i].item()) & 0xFFFF
         out[i] = total
     return out
 
 
 def pool_gather_ref(payload: bytes, seg, pool_cols, base_row: int, n: int,
-                    row_start: torch.Tensor, pool_base: int) -> None:
+                    row_start: torch.Tensor, pool_base: int, sstr) -> None:
     pool = seg.pool.numpy()
     for i in range(n):
         dst = pool_base + int(row_start[i].item())
-        for c in pool_cols:
-            r = int(seg.strref[c, base_row + i].item()) & M64
+        for ci, c in enumerate(pool_cols):
+            r = int(sstr[c, i].item()) & M64
             ln = r & 0xFFFF
             off = r >> 16
             pool[dst:dst + ln] = np.frombuffer(payload[off:off + ln],
                                                dtype=np.uint8)
-            seg.strref[c, base_row + i] = S.str_ref_pack(dst, ln)
+            seg.poolref[ci, base_row + i] = S.str_ref_pack(dst, ln)
             dst += ln
 
 

@@ -55,8 +55,12 @@ def _find_request_domain_ref(mv: memoryview, pos: int, end: int):
     return None
 
 
-def decode_l4_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
+def decode_l4_ref(payload: bytes, offs, lens, seg, base_row: int,
+                  sstr=None) -> None:
     mv = memoryview(payload)
+    if sstr is None:
+        import torch
+        sstr = torch.zeros((L4.N_STR, len(offs)), dtype=torch.int64)
     for rid in range(len(offs)):
         row = base_row + rid
         off, ln = int(offs[rid]), int(lens[rid])
@@ -146,7 +150,7 @@ def decode_l4_ref(payload: bytes, offs, lens, seg, base_row: int) -> None:
         w8("direction_score", f.get("direction_score", 0))
         ref = _find_request_domain_ref(mv, off, off + ln)
         if ref is not None:
-            seg.strref[0, row] = S.str_ref_pack(ref[0], ref[1])
+            sstr[0, rid] = S.str_ref_pack(ref[0], ref[1])
 
 
 NAGG_FIELDS = ["byte_tx", "byte_rx", "packet_tx", "packet_rx", "new_flow",

@@ -307,8 +307,16 @@ class QueryEngine:
             if fam == Q.SRC_KG:
                 return int(seg.kg[idx, row])
         if col in str_cols:
-            sidx = str_cols.index(col)
-            r = int(seg.strref[sidx, row]) & ((1 << 64) - 1)
+            # pooled string columns live in poolref (dict-encoded string
+            # tags were already handled via SRC_DID above)
+            if hasattr(seg, "attr_id"):  # l7 segment: map via POOL_POS
+                if col not in S.POOL_POS:
+                    raise SqlError(f"column {col!r} is dict-encoded; "
+                                   f"select it via its tag")
+                sidx = S.POOL_POS[col]
+            else:
+                sidx = str_cols.index(col)
+            r = int(seg.poolref[sidx, row]) & ((1 << 64) - 1)
             off, ln = r >> 16, r & 0xFFFF
             if ln == 0:
                 return ""

@@ -50,7 +50,7 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
         return rel
     if family == SRC_STR_HASH:
         from ..store.dictionary import str_hash_py
-        refs = seg.strref[idx, :n].numpy().view(np.uint64)
+        refs = seg.poolref[idx, :n].numpy().view(np.uint64)
         pool = seg.pool.numpy().tobytes()
         out = np.zeros(n, dtype=np.uint64)
         for i in range(n):

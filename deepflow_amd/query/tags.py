@@ -85,7 +85,7 @@ def build_l7_tags() -> Dict[str, TagDef]:
     # selectable from the segment pool
     for sname in ["trace_id", "span_id", "parent_span_id", "x_request_id_0",
                   "x_request_id_1", "http_user_agent", "biz_code"]:
-        add(TagDef(sname, Q.SRC_STR_HASH, S.STR_COLS.index(sname),
+        add(TagDef(sname, Q.SRC_STR_HASH, S.POOL_POS[sname],
                    hydrate="strhash"))
     # KnowledgeGraph universal tags, client (_0) / server (_1)
     for side in (0, 1):

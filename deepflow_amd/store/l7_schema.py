@@ -65,6 +65,10 @@ POOL_COLS = [
     ]
 ]
 
+N_POOL = len(POOL_COLS)
+# str col name -> pool position (for SRC_STR_HASH filters + select fetch)
+POOL_POS = {STR_COLS[sc]: i for i, sc in enumerate(POOL_COLS)}
+
 DICT_ID_INVALID = 0xFFFFFFFF
 
 N_U64 = len(U64_COLS)

@@ -27,10 +27,9 @@ class L7Segment:
         self.u64 = z((S.N_U64, capacity), torch.int64)
         self.u32 = z((S.N_U32, capacity), torch.int32)
         self.u8 = z((S.N_U8, capacity), torch.uint8)
-        self.strref = z((S.N_STR, capacity), torch.int64)
+        self.poolref = z((S.N_POOL, capacity), torch.int64)
         self.did = torch.full((S.N_DID, capacity), -1, dtype=torch.int32, device=dev)
         self.kg = z((2 * S.N_KG, capacity), torch.int32)
-        self.attr_ref = z((2 * S.MAX_ATTRS, capacity), torch.int64)
         self.attr_id = torch.full((2 * S.MAX_ATTRS, capacity), -1,
                                   dtype=torch.int32, device=dev)
         self.attr_cnt = z((capacity,), torch.uint8)
@@ -62,8 +61,7 @@ class L7Segment:
         if self.n_rows == 0:
             return 0.0
         fixed = (S.N_U64 * 8 + S.N_U32 * 4 + S.N_U8 * 1 + S.N_DID * 4 +
-                 2 * S.N_KG * 4 + 2 * S.MAX_ATTRS * 4 + 1 +
-                 len(S.POOL_COLS) * 8)
+                 2 * S.N_KG * 4 + 2 * S.MAX_ATTRS * 4 + 1 + S.N_POOL * 8)
         return fixed + self.pool_len / self.n_rows
 
 
@@ -80,7 +78,7 @@ class L4Segment:
         self.u64 = z((L4.N_U64, capacity), torch.int64)
         self.u32 = z((L4.N_U32, capacity), torch.int32)
         self.u8 = z((L4.N_U8, capacity), torch.uint8)
-        self.strref = z((L4.N_STR, capacity), torch.int64)
+        self.poolref = z((L4.N_STR, capacity), torch.int64)
         self.kg = z((2 * S.N_KG, capacity), torch.int32)
         self.pool = z((pool_capacity or capacity * 24,), torch.uint8)
         self.pool_len = 0
@@ -103,7 +101,7 @@ class L4Segment:
         if self.n_rows == 0:
             return 0.0
         fixed = (L4.N_U64 * 8 + L4.N_U32 * 4 + L4.N_U8 + 2 * S.N_KG * 4 +
-                 L4.N_STR * 8)
+                 L4.N_STR * 8)  # N_STR == pooled refs for l4
         return fixed + self.pool_len / self.n_rows
 
 
@@ -126,7 +124,7 @@ class SegmentSet:
     @staticmethod
     def seg_alloc_bytes(seg) -> int:
         total = 0
-        for name in ("u64", "u32", "u8", "strref", "did", "kg", "attr_ref",
+        for name in ("u64", "u32", "u8", "poolref", "did", "kg",
                      "attr_id", "attr_cnt", "pool"):
             t = getattr(seg, name, None)
             if t is not None:
@@ -157,7 +155,7 @@ class SegmentSet:
 
     @staticmethod
     def reset_segment(seg) -> None:
-        for name in ("u64", "u32", "u8", "strref", "attr_ref", "kg"):
+        for name in ("u64", "u32", "u8", "poolref", "kg"):
             t = getattr(seg, name, None)
             if t is not None:
                 t.zero_()

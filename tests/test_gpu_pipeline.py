@@ -47,7 +47,7 @@ def test_fixed_columns_match(pipes):
     assert torch.equal(a.u32[:, :n], b.u32[:, :n].cpu())
     assert torch.equal(a.u8[:, :n], b.u8[:, :n].cpu())
     assert torch.equal(a.attr_cnt[:n], b.attr_cnt[:n].cpu())
-    assert torch.equal(a.strref[:, :n] & 0xFFFF, b.strref[:, :n].cpu() & 0xFFFF)
+    assert torch.equal(a.poolref[:, :n], b.poolref[:, :n].cpu())
 
 
 def test_kg_columns_match(pipes):
@@ -81,11 +81,11 @@ def test_pool_contents_match(pipes):
     cpu, gpu = pipes
     a, b = cpu.segments.segments[0], gpu.segments.segments[0]
     assert a.pool_len == b.pool_len
-    tid = S.STR_COLS.index("trace_id")
+    tid = S.POOL_POS["trace_id"]
     pa = a.pool.numpy().tobytes()
     pb_ = b.pool.cpu().numpy().tobytes()
-    ra = a.strref[tid, :N].tolist()
-    rb = b.strref[tid, :N].cpu().tolist()
+    ra = a.poolref[tid, :N].tolist()
+    rb = b.poolref[tid, :N].cpu().tolist()
     for i in range(0, N, 97):
         oa, la = ra[i] >> 16, ra[i] & 0xFFFF
         ob, lb = rb[i] >> 16, rb[i] & 0xFFFF

@@ -51,7 +51,7 @@ def test_l4_pool_match(pipes):
     assert a.pool_len == b.pool_len
     if a.pool_len:
         assert torch.equal(a.pool[:a.pool_len], b.pool[:b.pool_len].cpu())
-    assert torch.equal(a.strref[:, :N], b.strref[:, :N].cpu())
+    assert torch.equal(a.poolref[:, :N], b.poolref[:, :N].cpu())
 
 
 def test_l4_metrics_match(pipes):

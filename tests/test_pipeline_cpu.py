@@ -118,10 +118,10 @@ def test_kg_join(pipe):
 def test_pool_strings(pipe):
     seg = pipe.segments.segments[0]
     pool = seg.pool.numpy().tobytes()
-    tid_col = S.STR_COLS.index("trace_id")
+    tid_col = S.POOL_POS["trace_id"]
     for i in range(0, N, 23):
         t = _truth(i)
-        r = int(seg.strref[tid_col, i]) & ((1 << 64) - 1)
+        r = int(seg.poolref[tid_col, i]) & ((1 << 64) - 1)
         off, ln = r >> 16, r & 0xFFFF
         assert pool[off:off + ln].decode() == t["trace_info"]["trace_id"]
 
