@@ -166,6 +166,27 @@ def main() -> None:
     spans_per_sec = total_spans / elapsed
     seg = pipe.segments.segments[0]
     bytes_per_span = seg.stored_bytes_per_row()
+    # SmartEncoding ratio on the tag side: what the same tags cost as
+    # verbatim string columns (measured string bytes + universal tags
+    # hydrated to synthetic k8s-style names of fixed lengths) vs the
+    # ID-encoded layout (dict ids + attr-id pool + pooled strings + refs
+    # + KG id columns). Ratio is workload-dependent; lengths noted here
+    # are the synthetic inventory's name sizes.
+    n_rows = max(pipe.stats.spans_in, 1)
+    from deepflow_amd.store import l7_schema as S_
+    KG_NAME_LEN = {  # per-side synthetic resource-name lengths
+        "pod_id": 24, "pod_node_id": 16, "pod_ns_id": 12,
+        "pod_group_id": 20, "pod_cluster_id": 10, "l3_device_type": 8,
+        "l3_device_id": 16, "subnet_id": 12, "host_id": 16, "az_id": 10,
+        "service_id": 18, "gprocess_id": 22,
+    }
+    naive_kg = 2 * sum(KG_NAME_LEN.values())
+    naive_str = pipe.stats.naive_str_bytes / n_rows + naive_kg
+    smart_str = (S_.N_DID * 4 + 4 + 1 + 8 + S_.N_POOL * 2 +
+                 (pipe.stats.pool_bytes +
+                  4 * sum(getattr(s, "attr_pool_len", 0)
+                          for s in pipe.segments.segments)) / n_rows +
+                 2 * S_.N_KG * 4)
 
     if rank == 0:
         out = {
@@ -189,6 +210,10 @@ def main() -> None:
                 "parallelism": f"shard{world} (hash-sharded span streams)",
                 "tag_cardinality": args.tag_card,
                 "bytes_per_span_stored": round(bytes_per_span, 1),
+                "tag_bytes_per_span_naive_strings": round(naive_str, 1),
+                "tag_bytes_per_span_smart": round(smart_str, 1),
+                "smart_encoding_ratio": round(naive_str / max(smart_str, 1),
+                                              2),
                 "dict_entries": pipe.dict.n_entries(),
                 "device": device,
             },

@@ -38,6 +38,10 @@ class PipelineStats:
     batches: int = 0
     dict_new: int = 0
     pool_bytes: int = 0
+    # bytes the batch's tag/trace strings would cost stored verbatim
+    # (the ClickHouse String-column baseline SmartEncoding is measured
+    # against; numeric columns excluded on both sides)
+    naive_str_bytes: int = 0
 
 
 class L7IngestPipeline:
@@ -142,8 +146,17 @@ class L7IngestPipeline:
         gpu_ops.intern_many(payload_t, sstr, self._ref_rows_scalar,
                             self._dom_scalar, 0, n, self.dict.tkeys,
                             self.dict.emit, self.dict.emit_ctr, seg.did, base)
+        # variable attr-id pool: per-row block offsets from attr_cnt cumsum
+        cnts = seg.attr_cnt[base:base + n].to(torch.int64) * 2
+        acum = torch.cumsum(cnts, 0)
+        attr_total = int(acum[-1].item())
+        seg.ensure_attr_pool(attr_total)
+        starts = (acum - cnts + seg.attr_pool_len).to(torch.int32)
+        seg.attr_start[base:base + n] = starts
         gpu_ops.intern_attrs(payload_t, seg, base, n, self.dict.tkeys,
-                             self.dict.emit, self.dict.emit_ctr, sattr)
+                             self.dict.emit, self.dict.emit_ctr, sattr,
+                             starts)
+        seg.attr_pool_len += attr_total
         # pool sizing: lens kernel -> cumsum -> (sync) total
         row_len = torch.zeros(n, dtype=torch.int32, device=dev)
         gpu_ops.pool_lens(sstr, self._pool_cols, n, row_len)
@@ -156,9 +169,12 @@ class L7IngestPipeline:
         gpu_ops.agg_app1s(seg, base, n, self.time_base_s,
                           self.metrics.tkeys, self.metrics.tvals)
         new = self.dict.harvest(payload_host)  # syncs emit buffer
+        naive = int((sstr[:, :n] & 0xFFFF).sum()) + \
+            int((sattr[:, :n] & 0xFFFF).sum())
         seg.pool_len += total
         self.stats.dict_new += new
         self.stats.pool_bytes += total
+        self.stats.naive_str_bytes += naive
 
     # ------------------------------------------------------------------
     def _ingest_cpu(self, payload, offs, lens, seg: L7Segment, base: int,
@@ -171,11 +187,9 @@ class L7IngestPipeline:
         new = ref.intern_ref(pb, sstr, _SCALAR_DICT_REF_ROWS,
                              _SCALAR_DICT_DOMAINS, 0, n, self.dict.tkeys,
                              seg.did, base, dictionary=self.dict)
-        new += ref.intern_ref(
-            pb, sattr, list(range(2 * S.MAX_ATTRS)),
-            [S.DICT_DOM_ATTR_NAME] * S.MAX_ATTRS +
-            [S.DICT_DOM_ATTR_VALUE] * S.MAX_ATTRS,
-            0, n, self.dict.tkeys, seg.attr_id, base, dictionary=self.dict)
+        new += ref.intern_attrs_pool_ref(pb, sattr, seg, base, n,
+                                         self.dict.tkeys,
+                                         dictionary=self.dict)
         row_len = ref.pool_lens_ref(sstr, S.POOL_COLS, n)
         cum = torch.cumsum(row_len.to(torch.int64), 0)
         total = int(cum[-1].item()) if n else 0
@@ -184,6 +198,9 @@ class L7IngestPipeline:
         ref.pool_gather_ref(pb, seg, S.POOL_COLS, base, n, row_start,
                             seg.pool_len, sstr)
         ref.agg_app1s_ref(seg, base, n, self.time_base_s, self.metrics.table)
+        naive = int((sstr[:, :n] & 0xFFFF).sum()) + \
+            int((sattr[:, :n] & 0xFFFF).sum())
+        self.stats.naive_str_bytes += naive
         seg.pool_len += total
         self.stats.dict_new += len(new)
         self.stats.pool_bytes += total

@@ -742,11 +742,13 @@ __global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
                                uint64_t* __restrict__ tkeys, uint32_t cap_mask,
                                uint64_t* __restrict__ emit,
                                uint32_t* __restrict__ emit_ctr, uint32_t emit_cap,
-                               uint32_t* __restrict__ out_ids) {  // [2*MAX, stride]
+                               const uint32_t* __restrict__ attr_start,  // [n] row block offsets (global)
+                               int32_t* __restrict__ attr_pool) {  // segment attr-id pool
     uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
     bool in_range = i < n;
     uint64_t row = base_row + (in_range ? i : 0);
     uint32_t cnt = in_range ? attr_cnt[row] : 0;
+    uint32_t pbase = in_range ? attr_start[i] : 0;
     // wave-max so all lanes iterate together for the uniform-name fast path
     uint32_t maxc = cnt;
     for (int d = 32; d > 0; d >>= 1)
@@ -764,8 +766,7 @@ __global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
                                           act && nlen, tkeys, cap_mask,
                                           emit, emit_ctr, emit_cap);
         if (act)
-            out_ids[(uint64_t)a * stride + row] =
-                nlen ? slot : DICT_ID_INVALID;
+            attr_pool[pbase + a] = nlen ? (int32_t)slot : -1;
         // values (high cardinality -> per-lane probes)
         uint64_t vref = act
             ? attr_refs[(uint64_t)(L7_MAX_ATTRS + a) * ref_stride + i] : 0;
@@ -778,7 +779,7 @@ __global__ void k_intern_attrs(const uint8_t* __restrict__ payload,
                 vslot = intern_probe(vh, vref, DICT_DOM_ATTR_VALUE, tkeys,
                                      cap_mask, emit, emit_ctr, emit_cap);
             }
-            out_ids[(uint64_t)(L7_MAX_ATTRS + a) * stride + row] = vslot;
+            attr_pool[pbase + cnt + a] = (int32_t)vslot;
         }
     }
 }
@@ -809,20 +810,22 @@ __global__ void k_pool_gather(const uint8_t* __restrict__ payload,
                               uint64_t stride, uint64_t base_row,
                               const uint64_t* __restrict__ row_start,  // exclusive cumsum
                               uint8_t* __restrict__ pool, uint64_t pool_base,
-                              uint64_t* __restrict__ out_refs,  // [npc, out_stride] segment
+                              uint64_t* __restrict__ out_rowref,  // [out_stride] segment
+                              int16_t* __restrict__ out_lens,     // [npc, out_stride]
                               uint64_t out_stride, uint64_t out_base_row) {
     uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    uint64_t dst = pool_base + row_start[i];
+    uint64_t dst0 = pool_base + row_start[i];
+    uint64_t dst = dst0;
     for (uint32_t c = 0; c < npc; c++) {
         uint64_t r = strc[(uint64_t)pool_cols[c] * stride + base_row + i];
         uint32_t len = STR_REF_LEN(r);
         uint64_t src = STR_REF_OFF(r);
         for (uint32_t b = 0; b < len; b++) pool[dst + b] = payload[src + b];
-        out_refs[(uint64_t)c * out_stride + out_base_row + i] =
-            STR_REF_PACK(dst, len);
+        out_lens[(uint64_t)c * out_stride + out_base_row + i] = (int16_t)len;
         dst += len;
     }
+    out_rowref[out_base_row + i] = STR_REF_PACK(dst0, dst - dst0);
 }
 
 // ----------------------------------------------------------------------
@@ -908,12 +911,14 @@ struct SegView {
     const uint64_t* u64c;
     const uint32_t* u32c;
     const uint8_t* u8c;
-    const uint32_t* didc;    // [L7_DID_N, stride]
-    const uint32_t* kgc;     // [2*KG_VALS_N, stride]
-    const uint32_t* attrid;  // [2*L7_MAX_ATTRS, stride] interned attr name/val ids
-    const uint8_t* attr_cnt;
-    const uint64_t* strc;    // [n_pool_cols, stride] pool-relative refs (pool position = idx)
-    const uint8_t* pool;     // segment string pool
+    const uint32_t* didc;      // [L7_DID_N, stride]
+    const uint32_t* kgc;       // [2*KG_VALS_N, stride]
+    const int32_t* attr_pool;  // variable attr-id pool
+    const uint32_t* attr_start;  // [stride] row block offsets into attr_pool
+    const uint8_t* attr_cnt;   // [stride]
+    const uint64_t* str_rowref;  // [stride] (pool_off<<16 | total_len)
+    const int16_t* str_lens;   // [n_pool_cols, stride]
+    const uint8_t* pool;       // segment string pool
     uint64_t stride;
     uint64_t n_rows;
 };
@@ -927,9 +932,11 @@ DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
         case SRC_DID: return s.didc[(uint64_t)idx * s.stride + row];
         case SRC_KG:  return s.kgc[(uint64_t)idx * s.stride + row];
         case SRC_ATTR_VAL: {
-            // value id of attr whose interned name id == idx is impractical
-            // per-row without a scan; idx here is the attr slot (0..MAX-1).
-            return s.attrid[(uint64_t)(L7_MAX_ATTRS + idx) * s.stride + row];
+            // idx is the attr slot (0..cnt-1); value ids follow name ids in
+            // the row's attr-pool block
+            uint32_t cnt = s.attr_cnt[row];
+            if (idx >= cnt) return DICT_ID_INVALID;
+            return (uint32_t)s.attr_pool[s.attr_start[row] + cnt + idx];
         }
         case SRC_TIME_BUCKET: {
             uint64_t t_s = s.u64c[L7_U64_START_TIME * s.stride + row] / 1000000000ull;
@@ -937,10 +944,13 @@ DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
             return bucket ? (rel / bucket) * bucket : rel;
         }
         case SRC_STR_HASH: {
-            uint64_t r = s.strc[(uint64_t)idx * s.stride + row];
-            uint32_t len = STR_REF_LEN(r);
+            uint64_t rr = s.str_rowref[row];
+            uint64_t off = STR_REF_OFF(rr);
+            for (uint16_t c = 0; c < (uint16_t)idx; c++)
+                off += (uint16_t)s.str_lens[(uint64_t)c * s.stride + row];
+            uint32_t len = (uint16_t)s.str_lens[(uint64_t)idx * s.stride + row];
             if (len == 0) return 0;
-            return str_hash(s.pool + STR_REF_OFF(r), len, STR_FILTER_SEED);
+            return str_hash(s.pool + off, len, STR_FILTER_SEED);
         }
         default: return 0;
     }
@@ -1177,14 +1187,15 @@ int df_intern_attrs(const void* payload, const void* attr_refs,
                     uint64_t base_row, uint64_t ref_stride,
                     void* tkeys, uint32_t cap,
                     void* emit, void* emit_ctr, uint32_t emit_cap,
-                    void* out_ids, uint64_t stream) {
+                    const void* attr_start, void* attr_pool,
+                    uint64_t stream) {
     hipLaunchKernelGGL(k_intern_attrs, dim3(grid_for(n)), dim3(BLOCK), 0,
                        STREAM(stream),
                        (const uint8_t*)payload, (const uint64_t*)attr_refs,
                        (const uint8_t*)attr_cnt, n, stride, base_row,
                        ref_stride, (uint64_t*)tkeys, cap - 1,
                        (uint64_t*)emit, (uint32_t*)emit_ctr, emit_cap,
-                       (uint32_t*)out_ids);
+                       (const uint32_t*)attr_start, (int32_t*)attr_pool);
     return (int)hipGetLastError();
 }
 
@@ -1199,13 +1210,15 @@ int df_pool_lens(const void* strc, const void* pool_cols, uint32_t npc, uint32_t
 int df_pool_gather(const void* payload, const void* strc, const void* pool_cols,
                    uint32_t npc, uint32_t n, uint64_t stride, uint64_t base_row,
                    const void* row_start, void* pool, uint64_t pool_base,
-                   void* out_refs, uint64_t out_stride, uint64_t out_base_row,
+                   void* out_rowref, void* out_lens,
+                   uint64_t out_stride, uint64_t out_base_row,
                    uint64_t stream) {
     hipLaunchKernelGGL(k_pool_gather, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
                        (const uint8_t*)payload, (const uint64_t*)strc,
                        (const uint8_t*)pool_cols, npc, n, stride, base_row,
                        (const uint64_t*)row_start, (uint8_t*)pool, pool_base,
-                       (uint64_t*)out_refs, out_stride, out_base_row);
+                       (uint64_t*)out_rowref, (int16_t*)out_lens,
+                       out_stride, out_base_row);
     return (int)hipGetLastError();
 }
 
@@ -1221,16 +1234,20 @@ int df_agg_app1s(void* u64c, void* u32c, void* u8c, uint64_t stride,
 }
 
 int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
-                 const void* didc, const void* kgc, const void* attrid,
-                 const void* attr_cnt, const void* strc, const void* pool,
+                 const void* didc, const void* kgc, const void* attr_pool,
+                 const void* attr_start, const void* attr_cnt,
+                 const void* str_rowref, const void* str_lens,
+                 const void* pool,
                  uint64_t stride, uint64_t n_rows,
                  const void* spec,  // QuerySpec, host-built bytes
                  uint32_t n, uint64_t base_row,
                  void* gkeys, void* graw, void* gvals, uint32_t cap,
                  uint64_t stream) {
     SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
-              (const uint32_t*)didc, (const uint32_t*)kgc, (const uint32_t*)attrid,
-              (const uint8_t*)attr_cnt, (const uint64_t*)strc,
+              (const uint32_t*)didc, (const uint32_t*)kgc,
+              (const int32_t*)attr_pool, (const uint32_t*)attr_start,
+              (const uint8_t*)attr_cnt, (const uint64_t*)str_rowref,
+              (const int16_t*)str_lens,
               (const uint8_t*)pool, stride, n_rows};
     QuerySpec q;
     __builtin_memcpy(&q, spec, sizeof(QuerySpec));
@@ -1241,15 +1258,19 @@ int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
 }
 
 int df_query_select(const void* u64c, const void* u32c, const void* u8c,
-                    const void* didc, const void* kgc, const void* attrid,
-                    const void* attr_cnt, const void* strc, const void* pool,
+                    const void* didc, const void* kgc, const void* attr_pool,
+                    const void* attr_start, const void* attr_cnt,
+                    const void* str_rowref, const void* str_lens,
+                    const void* pool,
                     uint64_t stride, uint64_t n_rows,
                     const void* spec, uint32_t n, uint64_t base_row,
                     void* out_rows, void* out_ctr, uint32_t out_cap,
                     uint64_t stream) {
     SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
-              (const uint32_t*)didc, (const uint32_t*)kgc, (const uint32_t*)attrid,
-              (const uint8_t*)attr_cnt, (const uint64_t*)strc,
+              (const uint32_t*)didc, (const uint32_t*)kgc,
+              (const int32_t*)attr_pool, (const uint32_t*)attr_start,
+              (const uint8_t*)attr_cnt, (const uint64_t*)str_rowref,
+              (const int16_t*)str_lens,
               (const uint8_t*)pool, stride, n_rows};
     QuerySpec q;
     __builtin_memcpy(&q, spec, sizeof(QuerySpec));

@@ -107,14 +107,16 @@ def intern_many(payload: torch.Tensor, refs: torch.Tensor,
 
 def intern_attrs(payload: torch.Tensor, seg, base_row: int, n: int,
                  tkeys: torch.Tensor, emit: torch.Tensor,
-                 emit_ctr: torch.Tensor, scratch_attr: torch.Tensor) -> None:
+                 emit_ctr: torch.Tensor, scratch_attr: torch.Tensor,
+                 attr_start: torch.Tensor) -> None:
     lib = native.gpu()
     native.check(lib.df_intern_attrs(
         payload.data_ptr(), scratch_attr.data_ptr(), seg.attr_cnt.data_ptr(),
         n, seg.capacity, base_row, scratch_attr.shape[1],
         tkeys.data_ptr(), tkeys.numel(),
         emit.data_ptr(), emit_ctr.data_ptr(), emit.shape[0],
-        seg.attr_id.data_ptr(), _stream()), "df_intern_attrs")
+        attr_start.data_ptr(), seg.attr_pool.data_ptr(), _stream()),
+        "df_intern_attrs")
 
 
 def pool_lens(scratch_str: torch.Tensor, pool_cols: torch.Tensor, n: int,
@@ -135,7 +137,8 @@ def pool_gather(payload: torch.Tensor, seg, pool_cols: torch.Tensor,
         payload.data_ptr(), scratch_str.data_ptr(), pool_cols.data_ptr(),
         pool_cols.numel(), n, scratch_str.shape[1], 0, row_start.data_ptr(),
         pool.data_ptr(), pool_base,
-        seg.poolref.data_ptr(), seg.capacity, base_row, _stream()),
+        seg.str_rowref.data_ptr(), seg.str_lens.data_ptr(),
+        seg.capacity, base_row, _stream()),
         "df_pool_gather")
 
 
@@ -162,8 +165,9 @@ def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
     native.check(lib.df_query_agg(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
-        _opt_ptr(seg, "attr_cnt"), seg.poolref.data_ptr(),
+        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_pool"),
+        _opt_ptr(seg, "attr_start"), _opt_ptr(seg, "attr_cnt"),
+        seg.str_rowref.data_ptr(), seg.str_lens.data_ptr(),
         seg.pool.data_ptr(), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         gkeys.data_ptr(), graw.data_ptr(), gvals.data_ptr(), gkeys.numel(),
@@ -177,8 +181,9 @@ def query_select(seg, spec_bytes: bytes, base_row: int, n: int,
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
     native.check(lib.df_query_select(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_id"),
-        _opt_ptr(seg, "attr_cnt"), seg.poolref.data_ptr(),
+        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_pool"),
+        _opt_ptr(seg, "attr_start"), _opt_ptr(seg, "attr_cnt"),
+        seg.str_rowref.data_ptr(), seg.str_lens.data_ptr(),
         seg.pool.data_ptr(), seg.capacity, seg.n_rows,
         ctypes.addressof(buf), n, base_row,
         out_rows.data_ptr(), out_ctr.data_ptr(), out_rows.numel(), _stream()),

@@ -93,19 +93,19 @@ def _decl_gpu(lib: ct.CDLL) -> None:
                                    p, p, u32, p, u64, u64, u64]
     lib.df_intern_attrs.restype = ct.c_int
     lib.df_intern_attrs.argtypes = [p, p, p, u32, u64, u64, u64, p, u32,
-                                    p, p, u32, p, u64]
+                                    p, p, u32, p, p, u64]
     lib.df_pool_lens.restype = ct.c_int
     lib.df_pool_lens.argtypes = [p, p, u32, u32, u64, u64, p, u64]
     lib.df_pool_gather.restype = ct.c_int
-    lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, p, u64, u64, u64]
+    lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, p, p, u64, u64, u64]
     lib.df_agg_app1s.restype = ct.c_int
     lib.df_agg_app1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
     lib.df_query_agg.restype = ct.c_int
-    lib.df_query_agg.argtypes = [p, p, p, p, p, p, p, p, p, u64, u64, p, u32,
-                                 u64, p, p, p, u32, u64]
+    lib.df_query_agg.argtypes = [p, p, p, p, p, p, p, p, p, p, p, u64, u64,
+                                 p, u32, u64, p, p, p, u32, u64]
     lib.df_query_select.restype = ct.c_int
-    lib.df_query_select.argtypes = [p, p, p, p, p, p, p, p, p, u64, u64, p,
-                                    u32, u64, p, p, u32, u64]
+    lib.df_query_select.argtypes = [p, p, p, p, p, p, p, p, p, p, p, u64,
+                                    u64, p, u32, u64, p, p, u32, u64]
     lib.df_spec_sizes.restype = ct.c_int
     lib.df_spec_sizes.argtypes = [p, p, p, p]
 

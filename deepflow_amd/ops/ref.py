@@ -286,6 +286,50 @@ def intern_ref(payload: bytes, refs: torch.Tensor, ref_rows, domains,
     return new_entries
 
 
+def intern_attrs_pool_ref(payload: bytes, sattr, seg, base_row: int, n: int,
+                          tkeys, dictionary=None):
+    """CPU twin of k_intern_attrs with the variable attr-id pool."""
+    new_entries = []
+    for i in range(n):
+        row = base_row + i
+        cnt = int(seg.attr_cnt[row].item())
+        start = seg.attr_pool_len
+        seg.ensure_attr_pool(2 * cnt)
+        seg.attr_start[row] = start
+        for a in range(cnt):
+            for half, dom in ((0, S.DICT_DOM_ATTR_NAME),
+                              (1, S.DICT_DOM_ATTR_VALUE)):
+                ref = int(sattr[half * S.MAX_ATTRS + a, i].item()) & M64
+                ln = ref & 0xFFFF
+                off = ref >> 16
+                ident = -1
+                if ln:
+                    sbytes = payload[off:off + ln]
+                    h = str_hash_py(bytes(sbytes), domain_seed(dom))
+                    cap_mask = tkeys.numel() - 1
+                    tk = tkeys.numpy()
+                    slot = h & cap_mask
+                    for _ in range(cap_mask + 1):
+                        cur = int(tk[slot]) & M64
+                        if cur == 0:
+                            tk[slot] = np.int64(h) if h < (1 << 63) \
+                                else np.int64(h - (1 << 64))
+                            new_entries.append((dom, slot, bytes(sbytes)))
+                            break
+                        if cur == h:
+                            break
+                        slot = (slot + 1) & cap_mask
+                    ident = slot if slot < (1 << 31) else slot - (1 << 32)
+                seg.attr_pool[start + half * cnt + a] = ident
+        seg.attr_pool_len += 2 * cnt
+    if dictionary is not None:
+        for dom, slot, sbytes in new_entries:
+            dictionary.id_to_str[(dom, slot)] = sbytes
+            dictionary.str_to_id[(dom, sbytes)] = slot
+            dictionary.pending_sync.append((dom, slot, sbytes))
+    return new_entries
+
+
 # -------------------------------------------------------------- K4 pool
 
 def pool_lens_ref(sstr, pool_cols, n: int) -> torch.Tensor:
@@ -302,15 +346,17 @@ def pool_gather_ref(payload: bytes, seg, pool_cols, base_row: int, n: int,
                     row_start: torch.Tensor, pool_base: int, sstr) -> None:
     pool = seg.pool.numpy()
     for i in range(n):
-        dst = pool_base + int(row_start[i].item())
+        dst0 = pool_base + int(row_start[i].item())
+        dst = dst0
         for ci, c in enumerate(pool_cols):
             r = int(sstr[c, i].item()) & M64
             ln = r & 0xFFFF
             off = r >> 16
             pool[dst:dst + ln] = np.frombuffer(payload[off:off + ln],
                                                dtype=np.uint8)
-            seg.poolref[ci, base_row + i] = S.str_ref_pack(dst, ln)
+            seg.str_lens[ci, base_row + i] = ln
             dst += ln
+        seg.str_rowref[base_row + i] = S.str_ref_pack(dst0, dst - dst0)
 
 
 # -------------------------------------------------------------- K5 agg

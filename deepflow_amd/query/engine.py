@@ -307,19 +307,21 @@ class QueryEngine:
             if fam == Q.SRC_KG:
                 return int(seg.kg[idx, row])
         if col in str_cols:
-            # pooled string columns live in poolref (dict-encoded string
-            # tags were already handled via SRC_DID above)
-            if hasattr(seg, "attr_id"):  # l7 segment: map via POOL_POS
+            # pooled string columns: row block ref + per-col u16 lens
+            # (dict-encoded string tags were already handled via SRC_DID)
+            if hasattr(seg, "attr_pool"):  # l7 segment: map via POOL_POS
                 if col not in S.POOL_POS:
                     raise SqlError(f"column {col!r} is dict-encoded; "
                                    f"select it via its tag")
                 sidx = S.POOL_POS[col]
             else:
                 sidx = str_cols.index(col)
-            r = int(seg.poolref[sidx, row]) & ((1 << 64) - 1)
-            off, ln = r >> 16, r & 0xFFFF
+            rr = int(seg.str_rowref[row]) & ((1 << 64) - 1)
+            ln = int(seg.str_lens[sidx, row]) & 0xFFFF
             if ln == 0:
                 return ""
+            off = (rr >> 16) + sum(
+                int(seg.str_lens[c, row]) & 0xFFFF for c in range(sidx))
             if seg.pool.device.type == "cpu":
                 return bytes(seg.pool[off:off + ln].numpy()).decode(
                     "utf-8", "replace")

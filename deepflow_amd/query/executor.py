@@ -40,8 +40,14 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
     if family == SRC_KG:
         return seg.kg[idx, :n].numpy().view(np.uint32).astype(np.uint64)
     if family == SRC_ATTR_VAL:
-        return seg.attr_id[S.MAX_ATTRS + idx, :n].numpy().view(
-            np.uint32).astype(np.uint64)
+        starts = seg.attr_start[:n].numpy()
+        cnts = seg.attr_cnt[:n].numpy()
+        pool = seg.attr_pool.numpy()
+        out = np.full(n, 0xFFFFFFFF, dtype=np.uint64)
+        for i in range(n):
+            if idx < cnts[i]:
+                out[i] = np.uint32(pool[starts[i] + cnts[i] + idx])
+        return out
     if family == SRC_TIME_BUCKET:
         t_s = seg.u64[0, :n].numpy().view(np.uint64) // np.uint64(10**9)
         rel = np.maximum(t_s.astype(np.int64) - time_base_s, 0).astype(np.uint64)
@@ -50,14 +56,14 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
         return rel
     if family == SRC_STR_HASH:
         from ..store.dictionary import str_hash_py
-        refs = seg.poolref[idx, :n].numpy().view(np.uint64)
+        rowref = seg.str_rowref[:n].numpy().view(np.uint64)
+        lens = seg.str_lens[:, :n].numpy().astype(np.uint16)
         pool = seg.pool.numpy().tobytes()
         out = np.zeros(n, dtype=np.uint64)
         for i in range(n):
-            r = int(refs[i])
-            ln = r & 0xFFFF
+            ln = int(lens[idx, i])
             if ln:
-                off = r >> 16
+                off = (int(rowref[i]) >> 16) + int(lens[:idx, i].sum())
                 out[i] = str_hash_py(pool[off:off + ln], STR_FILTER_SEED)
         return out
     return np.zeros(n, dtype=np.uint64)

@@ -27,15 +27,27 @@ class L7Segment:
         self.u64 = z((S.N_U64, capacity), torch.int64)
         self.u32 = z((S.N_U32, capacity), torch.int32)
         self.u8 = z((S.N_U8, capacity), torch.uint8)
-        self.poolref = z((S.N_POOL, capacity), torch.int64)
+        self.str_rowref = z((capacity,), torch.int64)
+        self.str_lens = z((S.N_POOL, capacity), torch.int16)
         self.did = torch.full((S.N_DID, capacity), -1, dtype=torch.int32, device=dev)
         self.kg = z((2 * S.N_KG, capacity), torch.int32)
-        self.attr_id = torch.full((2 * S.MAX_ATTRS, capacity), -1,
-                                  dtype=torch.int32, device=dev)
+        self.attr_start = z((capacity,), torch.int32)
+        self.attr_pool = torch.full((capacity * 8,), -1, dtype=torch.int32,
+                                    device=dev)
+        self.attr_pool_len = 0
         self.attr_cnt = z((capacity,), torch.uint8)
         self.pool = z((pool_capacity or capacity * 96,), torch.uint8)
         self.pool_len = 0
         self.n_rows = 0
+
+    def ensure_attr_pool(self, extra: int) -> None:
+        need = self.attr_pool_len + extra
+        if need > self.attr_pool.numel():
+            new_cap = max(need, self.attr_pool.numel() * 2)
+            newp = torch.full((new_cap,), -1, dtype=torch.int32,
+                              device=self.attr_pool.device)
+            newp[: self.attr_pool_len] = self.attr_pool[: self.attr_pool_len]
+            self.attr_pool = newp
 
     def free_rows(self) -> int:
         return self.capacity - self.n_rows
@@ -61,8 +73,8 @@ class L7Segment:
         if self.n_rows == 0:
             return 0.0
         fixed = (S.N_U64 * 8 + S.N_U32 * 4 + S.N_U8 * 1 + S.N_DID * 4 +
-                 2 * S.N_KG * 4 + 2 * S.MAX_ATTRS * 4 + 1 + S.N_POOL * 8)
-        return fixed + self.pool_len / self.n_rows
+                 2 * S.N_KG * 4 + 4 + 1 + 8 + S.N_POOL * 2)
+        return fixed + (self.pool_len + 4 * self.attr_pool_len) / self.n_rows
 
 
 class L4Segment:
@@ -78,7 +90,8 @@ class L4Segment:
         self.u64 = z((L4.N_U64, capacity), torch.int64)
         self.u32 = z((L4.N_U32, capacity), torch.int32)
         self.u8 = z((L4.N_U8, capacity), torch.uint8)
-        self.poolref = z((L4.N_STR, capacity), torch.int64)
+        self.str_rowref = z((capacity,), torch.int64)
+        self.str_lens = z((L4.N_STR, capacity), torch.int16)
         self.kg = z((2 * S.N_KG, capacity), torch.int32)
         self.pool = z((pool_capacity or capacity * 24,), torch.uint8)
         self.pool_len = 0
@@ -101,7 +114,7 @@ class L4Segment:
         if self.n_rows == 0:
             return 0.0
         fixed = (L4.N_U64 * 8 + L4.N_U32 * 4 + L4.N_U8 + 2 * S.N_KG * 4 +
-                 L4.N_STR * 8)  # N_STR == pooled refs for l4
+                 8 + L4.N_STR * 2)
         return fixed + self.pool_len / self.n_rows
 
 
@@ -124,8 +137,8 @@ class SegmentSet:
     @staticmethod
     def seg_alloc_bytes(seg) -> int:
         total = 0
-        for name in ("u64", "u32", "u8", "poolref", "did", "kg",
-                     "attr_id", "attr_cnt", "pool"):
+        for name in ("u64", "u32", "u8", "str_rowref", "str_lens", "did",
+                     "kg", "attr_start", "attr_pool", "attr_cnt", "pool"):
             t = getattr(seg, name, None)
             if t is not None:
                 total += t.numel() * t.element_size()
@@ -155,11 +168,12 @@ class SegmentSet:
 
     @staticmethod
     def reset_segment(seg) -> None:
-        for name in ("u64", "u32", "u8", "poolref", "kg"):
+        for name in ("u64", "u32", "u8", "str_rowref", "str_lens", "kg",
+                     "attr_start"):
             t = getattr(seg, name, None)
             if t is not None:
                 t.zero_()
-        for name in ("did", "attr_id"):
+        for name in ("did", "attr_pool"):
             t = getattr(seg, name, None)
             if t is not None:
                 t.fill_(-1)
@@ -167,6 +181,8 @@ class SegmentSet:
         if t is not None:
             t.zero_()
         seg.pool_len = 0
+        if hasattr(seg, "attr_pool_len"):
+            seg.attr_pool_len = 0
         seg.n_rows = 0
 
     def reserve(self, n_segments: int) -> None:

@@ -47,7 +47,9 @@ def test_fixed_columns_match(pipes):
     assert torch.equal(a.u32[:, :n], b.u32[:, :n].cpu())
     assert torch.equal(a.u8[:, :n], b.u8[:, :n].cpu())
     assert torch.equal(a.attr_cnt[:n], b.attr_cnt[:n].cpu())
-    assert torch.equal(a.poolref[:, :n], b.poolref[:, :n].cpu())
+    assert torch.equal(a.str_rowref[:n], b.str_rowref[:n].cpu())
+    assert torch.equal(a.str_lens[:, :n], b.str_lens[:, :n].cpu())
+    assert torch.equal(a.attr_start[:n], b.attr_start[:n].cpu())
 
 
 def test_kg_columns_match(pipes):
@@ -64,10 +66,15 @@ def test_dict_hydration_matches(pipes):
         hb = gpu.dict.hydrate(dom, b.did[did_idx, :N].cpu().tolist())
         assert ha == hb
     # attrs
-    ha = cpu.dict.hydrate(S.DICT_DOM_ATTR_VALUE,
-                          a.attr_id[S.MAX_ATTRS, :N].tolist())
-    hb = gpu.dict.hydrate(S.DICT_DOM_ATTR_VALUE,
-                          b.attr_id[S.MAX_ATTRS, :N].cpu().tolist())
+    sa = a.attr_start[:N].tolist()
+    sb = b.attr_start[:N].cpu().tolist()
+    ca = a.attr_cnt[:N].tolist()
+    pa_pool = a.attr_pool.tolist()
+    pb_pool = b.attr_pool.cpu().tolist()
+    va = [pa_pool[sa[i] + ca[i]] if ca[i] else -1 for i in range(N)]
+    vb = [pb_pool[sb[i] + ca[i]] if ca[i] else -1 for i in range(N)]
+    ha = cpu.dict.hydrate(S.DICT_DOM_ATTR_VALUE, va)
+    hb = gpu.dict.hydrate(S.DICT_DOM_ATTR_VALUE, vb)
     assert ha == hb
 
 
@@ -84,11 +91,13 @@ def test_pool_contents_match(pipes):
     tid = S.POOL_POS["trace_id"]
     pa = a.pool.numpy().tobytes()
     pb_ = b.pool.cpu().numpy().tobytes()
-    ra = a.poolref[tid, :N].tolist()
-    rb = b.poolref[tid, :N].cpu().tolist()
     for i in range(0, N, 97):
-        oa, la = ra[i] >> 16, ra[i] & 0xFFFF
-        ob, lb = rb[i] >> 16, rb[i] & 0xFFFF
+        oa = (int(a.str_rowref[i]) >> 16) + \
+            sum(int(a.str_lens[c, i]) for c in range(tid))
+        la = int(a.str_lens[tid, i])
+        ob = (int(b.str_rowref[i]) >> 16) + \
+            sum(int(b.str_lens[c, i]) for c in range(tid))
+        lb = int(b.str_lens[tid, i])
         assert pa[oa:oa + la] == pb_[ob:ob + lb]
 
 

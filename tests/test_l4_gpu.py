@@ -51,7 +51,8 @@ def test_l4_pool_match(pipes):
     assert a.pool_len == b.pool_len
     if a.pool_len:
         assert torch.equal(a.pool[:a.pool_len], b.pool[:b.pool_len].cpu())
-    assert torch.equal(a.poolref[:, :N], b.poolref[:, :N].cpu())
+    assert torch.equal(a.str_rowref[:N], b.str_rowref[:N].cpu())
+    assert torch.equal(a.str_lens[:, :N], b.str_lens[:, :N].cpu())
 
 
 def test_l4_metrics_match(pipes):

@@ -87,10 +87,10 @@ def test_attrs_hydrate(pipe):
         t = _truth(i)
         cnt = int(seg.attr_cnt[i])
         assert cnt == CFG.n_attrs
-        names = pipe.dict.hydrate(S.DICT_DOM_ATTR_NAME,
-                                  seg.attr_id[:cnt, i].tolist())
-        vals = pipe.dict.hydrate(S.DICT_DOM_ATTR_VALUE,
-                                 seg.attr_id[S.MAX_ATTRS:S.MAX_ATTRS + cnt, i].tolist())
+        start = int(seg.attr_start[i])
+        block = seg.attr_pool[start:start + 2 * cnt].tolist()
+        names = pipe.dict.hydrate(S.DICT_DOM_ATTR_NAME, block[:cnt])
+        vals = pipe.dict.hydrate(S.DICT_DOM_ATTR_VALUE, block[cnt:])
         assert names == t["ext_info"]["attribute_names"]
         assert vals == t["ext_info"]["attribute_values"]
 
@@ -121,8 +121,10 @@ def test_pool_strings(pipe):
     tid_col = S.POOL_POS["trace_id"]
     for i in range(0, N, 23):
         t = _truth(i)
-        r = int(seg.poolref[tid_col, i]) & ((1 << 64) - 1)
-        off, ln = r >> 16, r & 0xFFFF
+        rr = int(seg.str_rowref[i]) & ((1 << 64) - 1)
+        ln = int(seg.str_lens[tid_col, i])
+        off = (rr >> 16) + sum(int(seg.str_lens[c, i])
+                               for c in range(tid_col))
         assert pool[off:off + ln].decode() == t["trace_info"]["trace_id"]
 
 
