@@ -66,6 +66,14 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_OPENTELEMETRY, self._on_otel)
         self.receiver.register(framing.MSG_OPENTELEMETRY_COMPRESSED,
                                self._on_otel)
+        from .ingest.pcap_pipeline import PcapPipeline
+        self.pcap = PcapPipeline()
+        self.receiver.register(framing.MSG_RAW_PCAP,
+                               lambda hdr, payload:
+                               self.pcap.ingest_payload(payload.tobytes()))
+        self.receiver.register(framing.MSG_PACKETSEQUENCE,
+                               lambda hdr, payload:
+                               self.pcap.ingest_payload(payload.tobytes()))
         from .ingest.event_pipeline import EventPipeline
         from .ingest.applog_pipeline import AppLogPipeline
         self.events = EventPipeline()
@@ -102,6 +110,15 @@ class DeepflowServer:
         from .query.mcp import McpServer
         self.mcp = McpServer(self.engine, self.profiles)
         self.mcp.register(self.app)
+
+        @self.app.get("/v1/pcap/{flow_id}")
+        def pcap_export(flow_id: int):
+            from fastapi.responses import Response
+            blob = self.pcap.export_pcap(flow_id)
+            if blob is None:
+                return Response(status_code=404)
+            return Response(content=blob,
+                            media_type="application/vnd.tcpdump.pcap")
 
         @self.app.get("/v1/export/otlp")
         def export_otlp(where: str = "", limit: int = 10000):
