@@ -309,13 +309,38 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     std::string m;
     if (is_http_request(p, n, m) || is_http_response(p, n)) return 20;
     if (server_port == 53 && n >= 12) return 120;
-    if (n >= 1 && (p[0] == '*' || p[0] == '+' || p[0] == '-' || p[0] == '$' ||
-                   p[0] == ':') && (server_port == 6379)) return 80;
-    if (server_port == 3306 && n >= 5) return 60;
-    if (server_port == 5432 && n >= 6 && (p[0] == 'Q' || p[0] == 'P'))
-        return 61;
+    // Redis RESP: '*<digits>\r\n' (request array) or simple-type replies on
+    // the well-known port
+    if (n >= 4 && p[0] == '*' && p[1] >= '0' && p[1] <= '9') return 80;
+    if (server_port == 6379 && n >= 1 &&
+        (p[0] == '+' || p[0] == '-' || p[0] == '$' || p[0] == ':')) return 80;
+    // MySQL client/server packet: [len3 LE][seq] with a plausible length;
+    // first client command has seq 0 and a known command byte, server
+    // greeting has seq 0 and protocol version 10 (content-based — real
+    // deployments use arbitrary ports; matches reference
+    // protocol_inference.h's mysql check)
+    if (n >= 5) {
+        uint32_t plen = p[0] | (p[1] << 8) | (p[2] << 16);
+        if (plen >= 1 && plen + 4 == n && p[3] == 0) {
+            uint8_t c0 = p[4];
+            if (c0 <= 0x1F || c0 == 10) return 60;
+        }
+    }
+    // PostgreSQL simple query: 'Q' + int32 BE length covering the packet
+    if (n >= 6 && p[0] == 'Q') {
+        uint32_t mlen = (p[1] << 24) | (p[2] << 16) | (p[3] << 8) | p[4];
+        if (mlen + 1 == n || (mlen >= 5 && mlen < (1u << 20))) return 61;
+    }
+    if (server_port == 5432 && n >= 6 && p[0] == 'P') return 61;
     if (server_port == 9092 && n >= 14) return 100;
     if (server_port == 27017 && n >= 16) return 81;
+    // MongoDB header: little-endian msglen covering packet + known opcode
+    if (n >= 16) {
+        uint32_t mlen = p[0] | (p[1] << 8) | (p[2] << 16) | (p[3] << 24);
+        uint32_t opc = p[12] | (p[13] << 8) | (p[14] << 16) | (p[15] << 24);
+        if (mlen == n && (opc == 2013 || opc == 2004 || opc == 2010))
+            return 81;
+    }
     return 0;
 }
 
