@@ -72,14 +72,13 @@ def test_dns_pcap():
 
 
 def test_http_pcap():
-    path = f"{FIX}/http/httpflow.pcap"
-    if not os.path.exists(path):
-        pytest.skip("fixture moved")
-    l7, l4, stats = replay(path)
-    assert len(l7) >= 1
-    http = [r for r in l7 if r["base"]["head"]["proto"] == 20]
-    assert http
-    assert any(r["req"].get("req_type") in ("GET", "POST") for r in http)
+    l7, l4, stats = replay(f"{FIX}/http/client-ip.pcap")
+    assert len(l7) == 1
+    r = l7[0]
+    assert r["base"]["head"]["proto"] == 20
+    assert r["req"]["req_type"] == "POST"
+    assert r["req"]["resource"].startswith("/biz-inquiry-bff/")
+    assert r["resp"]["code"] == 200
 
 
 def test_redis_pcap():
