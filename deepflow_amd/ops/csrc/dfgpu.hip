@@ -891,11 +891,15 @@ enum { OP_EQ = 0, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN };
 // agg ops
 enum { AGGOP_COUNT = 0, AGGOP_SUM, AGGOP_MIN, AGGOP_MAX };
 
-struct QTerm { uint8_t family; uint8_t op; uint16_t idx; uint64_t v0, v1; };
+// group 0 terms AND together; groups >= 1 are OR-clauses: the row must
+// satisfy at least one term of every present group (CNF — covers
+// `(a=1 OR a=2)` and `x IN (...)`)
+struct QTerm { uint8_t family; uint8_t op; uint16_t idx; uint8_t group;
+               uint64_t v0, v1; };
 struct QKey  { uint8_t family; uint16_t idx; uint32_t bucket; };  // bucket: seconds per bucket for SRC_TIME_BUCKET
 struct QAgg  { uint8_t op; uint8_t family; uint16_t idx; };
 
-#define QMAX_TERMS 8
+#define QMAX_TERMS 16
 #define QMAX_KEYS 4
 #define QMAX_AGGS 8
 
@@ -957,6 +961,7 @@ DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
 }
 
 DEV bool eval_terms(const SegView& s, uint64_t row, const QuerySpec& q) {
+    uint32_t need = 0, have = 0;
     for (uint32_t t = 0; t < q.n_terms; t++) {
         const QTerm& term = q.terms[t];
         uint64_t v = src_value(s, row, term.family, term.idx, 0, q.time_base_s);
@@ -971,9 +976,15 @@ DEV bool eval_terms(const SegView& s, uint64_t row, const QuerySpec& q) {
             case OP_BETWEEN: ok = v >= term.v0 && v <= term.v1; break;
             default: ok = true;
         }
-        if (!ok) return false;
+        if (term.group == 0) {
+            if (!ok) return false;
+        } else {
+            uint32_t bit = 1u << (term.group & 31);
+            need |= bit;
+            if (ok) have |= bit;
+        }
     }
-    return true;
+    return (have & need) == need;
 }
 
 // group table: gkeys u64 hash (claim word), graw [cap, QMAX_KEYS] raw key

@@ -69,26 +69,39 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
     return np.zeros(n, dtype=np.uint64)
 
 
+def _term_mask(seg, t, plan: Plan, n: int) -> np.ndarray:
+    v = _src_np(seg, t.family, t.idx, 0, plan.time_base_s, n)
+    v0 = np.uint64(t.v0 & U64MAX)
+    v1 = np.uint64(t.v1 & U64MAX)
+    if t.op == OP_EQ:
+        return v == v0
+    if t.op == OP_NE:
+        return v != v0
+    if t.op == OP_LT:
+        return v < v0
+    if t.op == OP_LE:
+        return v <= v0
+    if t.op == OP_GT:
+        return v > v0
+    if t.op == OP_GE:
+        return v >= v0
+    if t.op == OP_BETWEEN:
+        return (v >= v0) & (v <= v1)
+    return np.ones(n, dtype=bool)
+
+
 def _mask_np(seg, plan: Plan, n: int) -> np.ndarray:
     mask = np.ones(n, dtype=bool)
+    groups = {}
     for t in plan.terms:
-        v = _src_np(seg, t.family, t.idx, 0, plan.time_base_s, n)
-        v0 = np.uint64(t.v0 & U64MAX)
-        v1 = np.uint64(t.v1 & U64MAX)
-        if t.op == OP_EQ:
-            mask &= v == v0
-        elif t.op == OP_NE:
-            mask &= v != v0
-        elif t.op == OP_LT:
-            mask &= v < v0
-        elif t.op == OP_LE:
-            mask &= v <= v0
-        elif t.op == OP_GT:
-            mask &= v > v0
-        elif t.op == OP_GE:
-            mask &= v >= v0
-        elif t.op == OP_BETWEEN:
-            mask &= (v >= v0) & (v <= v1)
+        tm = _term_mask(seg, t, plan, n)
+        g = getattr(t, "group", 0)
+        if g == 0:
+            mask &= tm
+        else:
+            groups[g] = groups.get(g, np.zeros(n, dtype=bool)) | tm
+    for gm in groups.values():
+        mask &= gm
     return mask
 
 

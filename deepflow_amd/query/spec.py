@@ -9,7 +9,7 @@ import ctypes as ct
 from dataclasses import dataclass, field
 from typing import List, Optional
 
-QMAX_TERMS = 8
+QMAX_TERMS = 16
 QMAX_KEYS = 4
 QMAX_AGGS = 8
 
@@ -36,7 +36,8 @@ OP_BY_NAME = {"=": OP_EQ, "==": OP_EQ, "!=": OP_NE, "<>": OP_NE, "<": OP_LT,
 
 class QTermC(ct.Structure):
     _fields_ = [("family", ct.c_uint8), ("op", ct.c_uint8),
-                ("idx", ct.c_uint16), ("v0", ct.c_uint64), ("v1", ct.c_uint64)]
+                ("idx", ct.c_uint16), ("group", ct.c_uint8),
+                ("v0", ct.c_uint64), ("v1", ct.c_uint64)]
 
 
 class QKeyC(ct.Structure):
@@ -65,6 +66,7 @@ class Term:
     op: int
     v0: int
     v1: int = 0
+    group: int = 0  # 0 = AND; >=1 = member of that OR-clause
 
 
 @dataclass
@@ -106,6 +108,7 @@ class Plan:
         assert len(self.aggs) <= QMAX_AGGS
         for i, t in enumerate(self.terms):
             c.terms[i] = QTermC(family=t.family, op=t.op, idx=t.idx,
+                                group=getattr(t, "group", 0),
                                 v0=t.v0 & (2**64 - 1), v1=t.v1 & (2**64 - 1))
         for i, k in enumerate(self.keys):
             c.keys[i] = QKeyC(family=k.family, idx=k.idx, bucket=k.bucket)

@@ -150,18 +150,53 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
     p.expect_kw("from")
     plan.table = p.next()[1]
 
-    # WHERE
+    # WHERE: conjunction of simple terms, parenthesized OR-clauses and
+    # IN lists (each OR/IN becomes a CNF group)
     if p.kw_is("where"):
         p.next()
+        group_counter = 0
         while True:
-            name = p.next()
-            if name[0] != "id":
-                raise SqlError("expected tag in where")
-            op_t = p.next()
-            if op_t[0] != "op" or op_t[1] not in Q.OP_BY_NAME:
-                raise SqlError(f"bad operator {op_t!r}")
-            lit = p.next()
-            _add_term(plan, name[1], op_t[1], lit, dictionary, tags)
+            if p.peek() == ("op", "("):
+                p.next()
+                group_counter += 1
+                while True:
+                    name = p.next()
+                    op_t = p.next()
+                    lit = p.next()
+                    _add_term(plan, name[1], op_t[1], lit, dictionary, tags,
+                              group=group_counter)
+                    if p.kw_is("or"):
+                        p.next()
+                        continue
+                    if p.peek() == ("op", ")"):
+                        p.next()
+                        break
+                    raise SqlError("expected OR or ) in clause")
+            else:
+                name = p.next()
+                if name[0] != "id":
+                    raise SqlError("expected tag in where")
+                if p.kw_is("in"):
+                    p.next()
+                    if p.next() != ("op", "("):
+                        raise SqlError("expected ( after IN")
+                    group_counter += 1
+                    while True:
+                        lit = p.next()
+                        _add_term(plan, name[1], "=", lit, dictionary, tags,
+                                  group=group_counter)
+                        if p.peek() == ("op", ","):
+                            p.next()
+                            continue
+                        if p.next() == ("op", ")"):
+                            break
+                        raise SqlError("expected , or ) in IN list")
+                else:
+                    op_t = p.next()
+                    if op_t[0] != "op" or op_t[1] not in Q.OP_BY_NAME:
+                        raise SqlError(f"bad operator {op_t!r}")
+                    lit = p.next()
+                    _add_term(plan, name[1], op_t[1], lit, dictionary, tags)
             if p.kw_is("and"):
                 p.next()
                 continue
@@ -261,23 +296,32 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
     return plan
 
 
+def _never(plan: Q.Plan, group: int) -> None:
+    """A term that can never match (unknown dict string inside an OR
+    group: the branch just contributes nothing)."""
+    plan.terms.append(Q.Term(Q.SRC_CONST0, 0, Q.OP_EQ, 1, group=group))
+
+
 def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
-              tags) -> None:
+              tags, group: int = 0) -> None:
     if name.lower() == "time":
         # time in epoch seconds against start_time (ns)
         v = int(lit[1]) * 10**9
-        plan.terms.append(Q.Term(Q.SRC_U64, 0, Q.OP_BY_NAME[op], v))
+        plan.terms.append(Q.Term(Q.SRC_U64, 0, Q.OP_BY_NAME[op], v,
+                                 group=group))
         return
     td = _resolve_tag(name, tags)
     if lit[0] == "num":
         v = int(lit[1]) if "." not in lit[1] else int(float(lit[1]))
-        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v))
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
+                                 group=group))
         return
     # string literal
     if td.hydrate == "strhash":
         from ..store.dictionary import str_hash_py
         v = str_hash_py(lit[1].encode(), Q.STR_FILTER_SEED)
-        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v))
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
+                                 group=group))
         return
     if td.hydrate.startswith("dict:"):
         dom = int(td.hydrate.split(":")[1])
@@ -286,20 +330,29 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
         if ident is None:
             if Q.OP_BY_NAME[op] == Q.OP_NE:
                 return  # != unknown-string matches everything
-            plan.impossible = True
+            if group:
+                _never(plan, group)
+            else:
+                plan.impossible = True
             return
-        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident))
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident,
+                                 group=group))
     elif td.hydrate == "l7proto":
         from ..wire.const_enums import L7_PROTOCOL_NAMES
         rev = {v.lower(): k for k, v in L7_PROTOCOL_NAMES.items()}
         ident = rev.get(lit[1].lower())
         if ident is None:
-            plan.impossible = True
+            if group:
+                _never(plan, group)
+            else:
+                plan.impossible = True
             return
-        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident))
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident,
+                                 group=group))
     elif td.hydrate == "ip":
         import ipaddress
         v = int(ipaddress.IPv4Address(lit[1]))
-        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v))
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
+                                 group=group))
     else:
         raise SqlError(f"tag {name} does not accept string literal")

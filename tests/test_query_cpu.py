@@ -125,3 +125,34 @@ def test_time_range_filter(eng):
         f"AND time <= {base_s + 10}")
     # spans spaced 1ms apart -> ~all within 1s window plus jitter
     assert 0 < r["values"][0][0] <= N
+
+
+def test_or_clause(eng):
+    t0 = truth()
+    d0, d1 = t0[0]["req"]["domain"], t0[1]["req"]["domain"]
+    r = eng.query(
+        f"SELECT Count(*) AS c FROM l7_flow_log WHERE "
+        f"(request_domain = '{d0}' OR request_domain = '{d1}')")
+    want = sum(1 for t in t0 if t["req"]["domain"] in (d0, d1))
+    assert r["values"] == [[want]]
+
+
+def test_in_list(eng):
+    t0 = truth()
+    r = eng.query(
+        "SELECT Count(*) AS c FROM l7_flow_log WHERE "
+        "response_code IN (200, 500)")
+    assert r["values"] == [[len(t0)]]
+    r2 = eng.query(
+        "SELECT Count(*) AS c FROM l7_flow_log WHERE response_code IN (404)")
+    assert r2["values"] == [] or r2["values"] == [[0]]
+
+
+def test_or_with_and(eng):
+    t0 = truth()
+    d0 = t0[0]["req"]["domain"]
+    r = eng.query(
+        f"SELECT Count(*) AS c FROM l7_flow_log WHERE server_port = 8080 "
+        f"AND (request_domain = '{d0}' OR request_domain = 'nope')")
+    want = sum(1 for t in t0 if t["req"]["domain"] == d0)
+    assert r["values"] == [[want]]

@@ -76,3 +76,18 @@ def test_select_rows_match(engines):
     rc = engines["cpu"].query(sql)
     rg = engines["cuda"].query(sql)
     assert sorted(map(tuple, rc["values"])) == sorted(map(tuple, rg["values"]))
+
+
+QUERIES_OR = [
+    "SELECT Count(*) AS c FROM l7_flow_log WHERE "
+    "(response_status = 0 OR response_status = 3)",
+    "SELECT request_domain, Count(*) AS c FROM l7_flow_log WHERE "
+    "response_code IN (200, 500) GROUP BY request_domain",
+]
+
+
+@pytest.mark.parametrize("sql", QUERIES_OR)
+def test_or_in_queries_match(engines, sql):
+    rc = engines["cpu"].query(sql)
+    rg = engines["cuda"].query(sql)
+    assert rc["values"] == rg["values"], sql
