@@ -69,6 +69,11 @@ class ControllerLite:
             "device_map": {}, "pod_map": {}, "l3_epc_map": {},
             "vtap_map": {}, "service_map": {}, "az_map": {},
         }
+        # genesis: agent-reported process/socket inventory + GPID allocation
+        # (reference controller/genesis + gpid semantics)
+        self.genesis_inventory: Dict[int, dict] = {}
+        self._gpid_by_key: Dict[Tuple[int, int], int] = {}
+        self._next_gpid = 1
 
     # ------------------------------------------------------------ sync
     def sync(self, agent_id: int, hostname: str = "", ip: str = "",
@@ -134,6 +139,29 @@ class ControllerLite:
     def lookup_name(self, map_name: str, ident: int) -> Optional[str]:
         return self.name_maps.get(map_name, {}).get(ident)
 
+    # ------------------------------------------------------------ genesis
+    def genesis_report(self, agent_id: int, processes: List[dict],
+                       sockets: List[dict]) -> dict:
+        """Store the agent's inventory; allocate stable GPIDs per
+        (agent, pid)."""
+        self.genesis_inventory[agent_id] = {
+            "processes": processes, "sockets": sockets,
+            "reported_at": time.time(),
+        }
+        gpids = {}
+        for proc in processes:
+            key = (agent_id, proc["pid"])
+            gpid = self._gpid_by_key.get(key)
+            if gpid is None:
+                gpid = self._next_gpid
+                self._next_gpid += 1
+                self._gpid_by_key[key] = gpid
+            gpids[proc["pid"]] = gpid
+        return {"status": "ok", "gpids": gpids}
+
+    def lookup_gpid(self, agent_id: int, pid: int) -> int:
+        return self._gpid_by_key.get((agent_id, pid), 0)
+
     # ------------------------------------------------------------ monitor
     def agent_status(self, stale_after_s: float = 60.0) -> List[dict]:
         now = time.time()
@@ -185,3 +213,14 @@ class ControllerLite:
         @app.post("/v1/rebalance/")
         def rebalance():
             return self.rebalance()
+
+        @app.post("/v1/genesis/")
+        async def genesis(request: Request):
+            body = await request.json()
+            return self.genesis_report(int(body.get("agent_id", 0)),
+                                       body.get("processes", []),
+                                       body.get("sockets", []))
+
+        @app.get("/v1/genesis/{agent_id}")
+        def genesis_get(agent_id: int):
+            return self.genesis_inventory.get(agent_id, {})
