@@ -250,6 +250,10 @@ class QueryEngine:
         if plan.select_rows:
             return self._run_select(plan, segments, tags, str_cols)
         groups = execute(plan, segments, self.device)
+        needs_second = any(m["op"] in ("percentile", "apdex")
+                           for m in plan.agg_meta)
+        if needs_second and len(groups) > 256:
+            raise SqlError("Percentile/Apdex limited to <=256 groups")
         columns = plan.key_names + plan.agg_names
         rows: List[List] = []
         for g in groups:
@@ -263,12 +267,85 @@ class QueryEngine:
                     cnt = g["agg"][ai + 1]
                     ai += 2
                     row.append(ssum / cnt if cnt else None)
+                elif meta["op"] in ("percentile", "apdex"):
+                    vals = self._group_values(plan, segments, g["key"], meta)
+                    ai += 1
+                    row.append(self._finish_quantile(meta, vals))
                 else:
                     row.append(g["agg"][ai])
                     ai += 1
             rows.append(row)
+        if plan.slimit:
+            rows = self._apply_slimit(plan, rows)
         rows = self._order_limit(plan, columns, rows)
         return {"columns": columns, "values": rows}
+
+    def _group_values(self, plan: Q.Plan, segments, key_raw, meta):
+        """Second pass for Percentile/Apdex: gather the metric's values for
+        one group (original filters + key equality) on the GPU/CPU."""
+        import copy
+        import torch
+        sub = copy.deepcopy(plan)
+        sub.select_rows = True
+        sub.limit = 1 << 22
+        for ki, k in enumerate(plan.keys):
+            v = key_raw[ki]
+            if k.family == Q.SRC_TIME_BUCKET and k.bucket > 1:
+                sub.terms.append(Q.Term(Q.SRC_TIME_BUCKET, 0, Q.OP_BETWEEN,
+                                        v, v + k.bucket - 1))
+            else:
+                sub.terms.append(Q.Term(k.family, k.idx, Q.OP_EQ, v))
+        hits = execute(sub, segments, self.device)
+        fam, idx = meta["family"], meta["idx"]
+        out = []
+        by_seg: Dict[int, List[int]] = {}
+        for si, r in hits:
+            by_seg.setdefault(si, []).append(r)
+        for si, rws in by_seg.items():
+            seg = segments[si]
+            idx_t = torch.tensor(rws, dtype=torch.long,
+                                 device=seg.u64.device)
+            if fam == Q.SRC_U64:
+                vals = seg.u64[idx].index_select(0, idx_t)
+            elif fam == Q.SRC_U32:
+                vals = seg.u32[idx].index_select(0, idx_t)
+            elif fam == Q.SRC_U8:
+                vals = seg.u8[idx].index_select(0, idx_t)
+            else:
+                raise SqlError("Percentile/Apdex needs a numeric metric")
+            out.append(vals.to(torch.float64))
+        if not out:
+            return None
+        return torch.cat([v.cpu() for v in out])
+
+    @staticmethod
+    def _finish_quantile(meta, vals):
+        import torch
+        if vals is None or vals.numel() == 0:
+            return None
+        if meta["op"] == "percentile":
+            q = float(meta.get("param", 95)) / 100.0
+            return float(torch.quantile(vals, q))
+        # Apdex: param = satisfied threshold (us); tolerated = 4x
+        t = float(meta.get("param", 100000))
+        sat = float((vals <= t).sum())
+        tol = float(((vals > t) & (vals <= 4 * t)).sum())
+        return (sat + tol / 2) / vals.numel()
+
+    def _apply_slimit(self, plan: Q.Plan, rows: List[List]) -> List[List]:
+        """Keep only the top-N series (distinct non-time key combos), ranked
+        by the first aggregate (reference SLIMIT semantics)."""
+        non_time = [i for i, n in enumerate(plan.key_names) if n != "time"]
+        first_agg = len(plan.key_names)
+        series: Dict[tuple, float] = {}
+        for r in rows:
+            k = tuple(r[i] for i in non_time)
+            v = r[first_agg] if len(r) > first_agg and \
+                isinstance(r[first_agg], (int, float)) else 0
+            series[k] = series.get(k, 0) + (v or 0)
+        top = sorted(series, key=lambda k: -series[k])[: plan.slimit]
+        keep = set(top)
+        return [r for r in rows if tuple(r[i] for i in non_time) in keep]
 
     # ----------------------------------------------------------- select
     def _run_select(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:

@@ -97,6 +97,7 @@ class Plan:
     agg_meta: List[dict] = field(default_factory=list)
     order_by: Optional[List] = None
     limit: Optional[int] = None
+    slimit: Optional[int] = None
     select_rows: bool = False  # non-aggregated SELECT
     select_cols: List[str] = field(default_factory=list)
     impossible: bool = False   # filter references unknown dict string

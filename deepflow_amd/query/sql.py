@@ -28,7 +28,8 @@ _TOKEN_RE = re.compile(
     r"|(?P<op><=|>=|!=|<>|=|<|>|\(|\)|,|\*))")
 
 AGG_FUNCS = {"count": Q.AGGOP_COUNT, "sum": Q.AGGOP_SUM, "avg": None,
-             "max": Q.AGGOP_MAX, "min": Q.AGGOP_MIN}
+             "max": Q.AGGOP_MAX, "min": Q.AGGOP_MIN,
+             "percentile": None, "apdex": None}
 
 
 class SqlError(ValueError):
@@ -115,6 +116,10 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
             p.next()  # (
             arg_t = p.next()
             arg = arg_t[1] if arg_t[1] != "*" else "*"
+            param = None
+            if p.peek() == ("op", ","):
+                p.next()
+                param = float(p.next()[1])
             nxt = p.next()
             if nxt != ("op", ")"):
                 raise SqlError("expected )")
@@ -122,7 +127,7 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
             if p.kw_is("as"):
                 p.next()
                 alias = p.next()[1]
-            select_items.append(("agg", func, arg, alias))
+            select_items.append(("agg", func, (arg, param), alias))
         elif t[0] == "id" and t[1].lower() == "time" and p.peek() == ("op", "("):
             p.next()
             p.next()  # bucket value (the GROUP BY clause carries it)
@@ -253,6 +258,11 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
                 continue
             break
         plan.order_by = order
+    if p.kw_is("slimit"):
+        # two-phase series limit: keep only the top-N groups by the first
+        # aggregate (reference CHEngine QuerySlimitSql, clickhouse.go:627)
+        p.next()
+        plan.slimit = int(p.next()[1])
     if p.kw_is("limit"):
         p.next()
         plan.limit = int(p.next()[1])
@@ -271,18 +281,30 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
                 continue
             if kind == "star":
                 raise SqlError("SELECT * not valid in aggregated query")
+            arg_name, arg_param = arg if isinstance(arg, tuple) else (arg, None)
             if func == "count":
                 plan.aggs.append(Q.Agg(Q.AGGOP_COUNT))
                 plan.agg_names.append(alias)
                 plan.agg_meta.append({"op": "count"})
             elif func == "avg":
-                md = metrics.get(arg) or _resolve_tag(arg, tags)
+                md = metrics.get(arg_name) or _resolve_tag(arg_name, tags)
                 plan.aggs.append(Q.Agg(Q.AGGOP_SUM, md.family, md.idx))
                 plan.aggs.append(Q.Agg(Q.AGGOP_COUNT))
                 plan.agg_names.append(alias)
                 plan.agg_meta.append({"op": "avg"})
+            elif func in ("percentile", "apdex"):
+                # executed as a second pass over gathered values
+                md = metrics.get(arg_name) or _resolve_tag(arg_name, tags)
+                plan.aggs.append(Q.Agg(Q.AGGOP_COUNT))
+                plan.agg_names.append(alias)
+                plan.agg_meta.append({"op": func, "family": md.family,
+                                      "idx": md.idx,
+                                      "param": arg_param if arg_param
+                                      is not None else
+                                      (95 if func == "percentile" else
+                                       100000)})
             else:
-                md = metrics.get(arg) or _resolve_tag(arg, tags)
+                md = metrics.get(arg_name) or _resolve_tag(arg_name, tags)
                 op = AGG_FUNCS[func]
                 plan.aggs.append(Q.Agg(op, md.family, md.idx))
                 plan.agg_names.append(alias)

@@ -156,3 +156,46 @@ def test_or_with_and(eng):
         f"AND (request_domain = '{d0}' OR request_domain = 'nope')")
     want = sum(1 for t in t0 if t["req"]["domain"] == d0)
     assert r["values"] == [[want]]
+
+
+def test_percentile(eng):
+    import numpy as np
+    rrts = sorted(t["base"]["head"]["rrt"] for t in truth())
+    r = eng.query(
+        "SELECT Percentile(response_duration, 50) AS p50 FROM l7_flow_log")
+    got = r["values"][0][0]
+    med = float(np.quantile(np.array(rrts, dtype=float), 0.5))
+    assert abs(got - med) < max(2.0, med * 0.01)
+
+
+def test_percentile_grouped(eng):
+    r = eng.query(
+        "SELECT response_status, Percentile(response_duration, 90) AS p "
+        "FROM l7_flow_log GROUP BY response_status")
+    assert len(r["values"]) == 2
+    assert all(row[1] > 0 for row in r["values"])
+
+
+def test_apdex(eng):
+    t0 = truth()
+    T = 100000.0
+    sat = sum(1 for t in t0 if t["base"]["head"]["rrt"] <= T)
+    tol = sum(1 for t in t0 if T < t["base"]["head"]["rrt"] <= 4 * T)
+    want = (sat + tol / 2) / len(t0)
+    r = eng.query(
+        "SELECT Apdex(response_duration, 100000) AS a FROM l7_flow_log")
+    assert abs(r["values"][0][0] - want) < 1e-9
+
+
+def test_slimit(eng):
+    r = eng.query(
+        "SELECT request_domain, time(60), Count(*) AS c FROM l7_flow_log "
+        "GROUP BY request_domain, time(60) SLIMIT 2")
+    domains = {row[0] for row in r["values"]}
+    assert len(domains) == 2
+    # the two kept series are the two biggest domains
+    full = eng.query(
+        "SELECT request_domain, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY request_domain ORDER BY c DESC")
+    top2 = {row[0] for row in full["values"][:2]}
+    assert domains == top2

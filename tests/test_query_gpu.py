@@ -91,3 +91,19 @@ def test_or_in_queries_match(engines, sql):
     rc = engines["cpu"].query(sql)
     rg = engines["cuda"].query(sql)
     assert rc["values"] == rg["values"], sql
+
+
+def test_percentile_apdex_match(engines):
+    for sql in [
+        "SELECT Percentile(response_duration, 75) AS p FROM l7_flow_log",
+        "SELECT response_status, Apdex(response_duration, 100000) AS a "
+        "FROM l7_flow_log GROUP BY response_status",
+    ]:
+        rc = engines["cpu"].query(sql)
+        rg = engines["cuda"].query(sql)
+        for a, b in zip(rc["values"], rg["values"]):
+            for x, y in zip(a, b):
+                if isinstance(x, float):
+                    assert abs(x - y) < max(1e-6, abs(x) * 0.01), sql
+                else:
+                    assert x == y, sql
