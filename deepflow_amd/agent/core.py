@@ -95,6 +95,53 @@ class Agent:
                                   agent_id=self.agent_id)
         return framing.encode_frame(hdr, payload)
 
+    # ---------------------------------------------------------- guard
+    # exception bits (reference agent/src/exception.rs bitmask idea)
+    EXC_MEM_LIMIT = 1 << 0
+    EXC_FLOW_TABLE_FULL = 1 << 1
+    EXC_SERVER_UNREACHABLE = 1 << 2
+
+    def guard_check(self, max_memory_mb: int = 768,
+                    max_flows: int = 1 << 20) -> int:
+        """Melt-down circuit breaker (reference utils/guard.rs:770-790):
+        returns the exception bitmask; callers stop feeding packets when
+        non-zero (the reference disables dispatchers)."""
+        exc = 0
+        try:
+            with open("/proc/self/statm") as f:
+                rss_pages = int(f.read().split()[1])
+            rss_mb = rss_pages * 4096 // (1 << 20)
+            if rss_mb > max_memory_mb:
+                exc |= self.EXC_MEM_LIMIT
+        except OSError:
+            pass
+        if self.stats()["flows_active"] > max_flows:
+            exc |= self.EXC_FLOW_TABLE_FULL
+        self.exceptions = exc
+        return exc
+
+    # ---------------------------------------------------------- sync
+    def sync_with_controller(self, post_fn) -> dict:
+        """One trident-Synchronizer cycle: report versions + exceptions,
+        apply pushed config and platform data (CIDR->EPC for the labeler).
+        post_fn(dict) -> dict (HTTP client or in-process)."""
+        resp = post_fn({
+            "agent_id": self.agent_id,
+            "hostname": socket.gethostname(),
+            "config_version": getattr(self, "config_version", 0),
+            "platform_version": getattr(self, "platform_version", 0),
+            "exceptions": getattr(self, "exceptions", 0),
+        })
+        if "config" in resp:
+            self.config = resp["config"]
+            self.config_version = resp["config_version"]
+        if "platform" in resp:
+            for entry in resp["platform"]:
+                # epc labeling for every known endpoint (/32)
+                self.add_cidr(entry["ip"], 32, entry["epc"])
+            self.platform_version = resp["platform_version"]
+        return resp
+
     def flush_to_server(self, now_ns: int) -> int:
         """tick + drain all types + send framed payloads to the server
         (uniform-sender analog). Returns frames sent."""

@@ -206,3 +206,37 @@ def test_mongodb(agent):
     assert len(l7) == 1
     assert l7[0]["base"]["head"]["proto"] == 81
     assert l7[0]["req"]["req_type"] == "OP_MSG"
+
+
+def test_guard_and_sync():
+    from fastapi.testclient import TestClient
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 9,
+                         dict_capacity=1 << 10)
+    from deepflow_amd.store.kg import KgInfo
+    srv.controller.update_platform({(7, CLIENT): KgInfo(pod_id=1)})
+    client = TestClient(srv.app)
+    a = Agent(vtap_id=9)
+    assert a.guard_check(max_memory_mb=10**6) == 0
+    assert a.guard_check(max_memory_mb=1) & Agent.EXC_MEM_LIMIT
+
+    def post(payload):
+        return client.post("/v1/sync/", json=payload).json()
+
+    resp = a.sync_with_controller(post)
+    assert "config" in resp and "platform" in resp
+    # second sync: versions caught up -> nothing pushed
+    resp2 = a.sync_with_controller(post)
+    assert "config" not in resp2 and "platform" not in resp2
+    # pushed platform applied to the labeler: client ip now labels epc=7
+    for frame, ts in http_session(CLIENT, SERVER, sport=47000):
+        a.packet(frame, ts)
+    a.tick(10**9 * 100)
+    l7 = _decode(a.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert l7[0]["base"]["l3_epc_id_src"] == 7
+    # exception visible to the controller monitor
+    a.guard_check(max_memory_mb=1)
+    a.sync_with_controller(post)
+    agents = client.get("/v1/agents/").json()
+    me = next(x for x in agents if x["agent_id"] == 9)
+    assert me["exceptions"] & Agent.EXC_MEM_LIMIT
+    a.close()
