@@ -31,6 +31,7 @@ def _lib():
         lib.dfa_new.argtypes = [u32]
         lib.dfa_free.argtypes = [p]
         lib.dfa_add_cidr.argtypes = [p, u32, u32, ct.c_int32]
+        lib.dfa_add_custom_port.argtypes = [p, u32]
         lib.dfa_packet.restype = ct.c_int
         lib.dfa_packet.argtypes = [p, p, u32, u64]
         lib.dfa_tick.argtypes = [p, u64]
@@ -65,6 +66,17 @@ class Agent:
     def add_cidr(self, net: int, masklen: int, epc: int) -> None:
         self._lib.dfa_add_cidr(self._h, net, masklen, epc)
 
+    def add_custom_protocol_port(self, port: int) -> None:
+        """Port-rule custom protocol (L7 proto 127); raw sessions are
+        captured and re-parsed by plugins (agent/plugins.py)."""
+        self._lib.dfa_add_custom_port(self._h, port)
+
+    def register_plugin(self, parse_fn) -> None:
+        from .plugins import PluginHost
+        if not hasattr(self, "plugin_host"):
+            self.plugin_host = PluginHost()
+        self.plugin_host.register(parse_fn)
+
     def packet(self, frame: bytes, ts_ns: int) -> int:
         buf = np.frombuffer(frame, dtype=np.uint8)
         return self._lib.dfa_packet(self._h, buf.ctypes.data, len(frame),
@@ -79,7 +91,10 @@ class Agent:
             return b""
         out = np.zeros(int(n), dtype=np.uint8)
         got = self._lib.dfa_drain(self._h, which, out.ctypes.data, n)
-        return out[:got].tobytes()
+        payload = out[:got].tobytes()
+        if which == DRAIN_L7 and getattr(self, "plugin_host", None):
+            payload = self.plugin_host.process_l7_payload(payload)
+        return payload
 
     def stats(self) -> Dict[str, int]:
         arr = np.zeros(8, dtype=np.uint64)

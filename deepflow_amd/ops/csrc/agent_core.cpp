@@ -122,6 +122,7 @@ struct Agent {
     uint64_t next_flow_id = 1;
     std::unordered_map<FlowKeyC, FlowNode, FlowKeyHash> flows;
     std::vector<Cidr> cidrs;
+    std::vector<uint16_t> custom_ports;  // port-rule custom protocols (127)
     std::map<MeterKey, AppMeterAcc> meters;
     std::vector<uint8_t> out_l4, out_l7, out_doc;
     // stats
@@ -305,6 +306,12 @@ bool parse_mongo_request(const uint8_t* p, uint32_t n, std::string& op,
 
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
+uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
+    for (uint16_t cp : a.custom_ports)
+        if (cp == server_port) return 127;
+    return 0;
+}
+
 uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     std::string m;
     if (is_http_request(p, n, m) || is_http_response(p, n)) return 20;
@@ -556,9 +563,30 @@ void encode_documents(Agent& a) {
 
 void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                        uint32_t n, uint64_t ts) {
-    if (f.l7_protocol == 0)
-        f.l7_protocol = infer_l7(p, n, f.port[1]);
+    if (f.l7_protocol == 0) {
+        f.l7_protocol = infer_l7_custom(a, f.port[1]);
+        if (f.l7_protocol == 0)
+            f.l7_protocol = infer_l7(p, n, f.port[1]);
+    }
     if (f.l7_protocol == 0) return;
+    if (f.l7_protocol == 127) {  // custom protocol: generic session capture;
+        // host-side plugins re-parse the raw prefix (wasm-plugin analog)
+        if (dir == 0 && !f.l7.active) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = "";
+            f.l7.resource.assign((const char*)p, n > 64 ? 64 : n);
+            f.l7.endpoint = "";
+            f.l7.domain = "";
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (dir == 1 && f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+            f.l7.active = false;
+        }
+        return;
+    }
     if (f.l7_protocol == 20) {  // HTTP/1
         std::string method;
         if (dir == 0 && is_http_request(p, n, method)) {
@@ -716,6 +744,10 @@ void* dfa_new(uint32_t vtap_id) {
 }
 
 void dfa_free(void* h) { delete (Agent*)h; }
+
+void dfa_add_custom_port(void* h, uint32_t port) {
+    ((Agent*)h)->custom_ports.push_back((uint16_t)port);
+}
 
 void dfa_add_cidr(void* h, uint32_t net, uint32_t masklen, int32_t epc) {
     Agent* a = (Agent*)h;
