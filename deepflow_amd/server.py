@@ -66,6 +66,14 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_OPENTELEMETRY, self._on_otel)
         self.receiver.register(framing.MSG_OPENTELEMETRY_COMPRESSED,
                                self._on_otel)
+        self.receiver.register(
+            framing.MSG_SKYWALKING,
+            lambda hdr, payload: self._on_third_party(hdr, payload,
+                                                      "skywalking"))
+        self.receiver.register(
+            framing.MSG_DATADOG,
+            lambda hdr, payload: self._on_third_party(hdr, payload,
+                                                      "datadog"))
         from .ingest.pcap_pipeline import PcapPipeline
         self.pcap = PcapPipeline()
         self.receiver.register(framing.MSG_RAW_PCAP,
@@ -177,6 +185,17 @@ class DeepflowServer:
                 return
         arr = np.frombuffer(l7_payload, dtype=np.uint8)
         self._on_l7(hdr, arr)
+
+    def _on_third_party(self, hdr, payload, kind: str) -> None:
+        import numpy as np
+        from .ingest.thirdparty import third_party_frame_to_l7_payload
+        try:
+            l7_payload = third_party_frame_to_l7_payload(payload.tobytes(),
+                                                         kind)
+        except Exception:
+            self.receiver.counter.add(f"{kind}_decode_errors")
+            return
+        self._on_l7(hdr, np.frombuffer(l7_payload, dtype=np.uint8))
 
     # ------------------------------------------------------------------
     def ingest_self_stats(self) -> int:
