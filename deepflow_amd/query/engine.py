@@ -112,8 +112,9 @@ class QueryEngine:
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
         if table.startswith("application"):
             rows = self.pipe.metrics.rows()
-            if table.endswith(".1m"):
-                rows = rollup_rows(rows, 60)
+            iv = self._table_interval(table)
+            if iv > 1:
+                rows = rollup_rows(rows, iv)
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
         if table.startswith("deepflow_system") or \
                 table.startswith("deepflow_tenant"):
@@ -123,10 +124,31 @@ class QueryEngine:
             if self.l4 is None:
                 raise SqlError("network table not enabled")
             rows = self.l4.metrics.rows()
-            if table.endswith(".1m"):
-                rows = rollup_rows(rows, 60)
+            iv = self._table_interval(table)
+            if iv > 1:
+                rows = rollup_rows(rows, iv)
             return self._run_rows(sql, rows, time_base_s=self.l4.time_base_s)
         raise SqlError(f"unknown table {table!r}")
+
+    # datasource intervals (reference ingester/datasource REST: 1h/1d MVs)
+    DATASOURCE_INTERVALS = {"1s": 1, "1m": 60, "1h": 3600, "1d": 86400}
+
+    def _table_interval(self, table: str) -> int:
+        if "." not in table:
+            return 1
+        suffix = table.split(".", 1)[1]
+        extra = getattr(self, "extra_datasources", {})
+        if suffix in extra:
+            return extra[suffix]
+        iv = self.DATASOURCE_INTERVALS.get(suffix)
+        if iv is None:
+            raise SqlError(f"unknown datasource interval {suffix!r}")
+        return iv
+
+    def add_datasource(self, name: str, interval_s: int) -> None:
+        if not hasattr(self, "extra_datasources"):
+            self.extra_datasources = {}
+        self.extra_datasources[name] = interval_s
 
     # ----------------------------------------------------------- show
     def _show(self, sql: str) -> Dict:

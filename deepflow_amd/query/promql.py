@@ -273,3 +273,19 @@ class PromQLEngine:
         @app.get("/prom/api/v1/label/{label}/values")
         async def prom_label_values(label: str):
             return self.label_values(label)
+
+        @app.get("/prom/api/v1/series")
+        async def prom_series(request: Request):
+            m = request.query_params.get("match[]", "")
+            sel = _SEL_RE.match(m or "")
+            if not sel:
+                return {"status": "error", "errorType": "bad_data",
+                        "error": "bad match[]"}
+            try:
+                series = self._series(sel.group("name"),
+                                      _parse_matchers(sel.group("matchers")))
+            except PromError as e:
+                return {"status": "error", "errorType": "bad_data",
+                        "error": str(e)}
+            return {"status": "success",
+                    "data": [s["metric"] for s in series]}

@@ -10,6 +10,8 @@ from __future__ import annotations
 import threading
 from typing import Optional
 
+from fastapi import Request
+
 from .gen import SpanGenConfig
 from .ingest import L7IngestPipeline
 from .ingest.receiver import Receiver
@@ -118,6 +120,13 @@ class DeepflowServer:
         from .query.mcp import McpServer
         self.mcp = McpServer(self.engine, self.profiles)
         self.mcp.register(self.app)
+
+        @self.app.post("/v1/datasources/")
+        async def add_datasource(request: Request):
+            body = await request.json()
+            self.engine.add_datasource(body["name"],
+                                       int(body["interval"]))
+            return {"status": "ok"}
 
         @self.app.get("/v1/pcap/{flow_id}")
         def pcap_export(flow_id: int):

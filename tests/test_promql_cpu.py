@@ -81,3 +81,27 @@ def test_label_values(server):
     r = client.get("/prom/api/v1/label/vtap_id/values")
     vals = r.json()["data"]
     assert "1" in vals
+
+
+def test_series_endpoint(server):
+    client = TestClient(server.app)
+    r = client.get("/prom/api/v1/series",
+                   params={"match[]": "application_request"})
+    body = r.json()
+    assert body["status"] == "success"
+    assert len(body["data"]) >= 1
+
+
+def test_datasource_1h(server):
+    client = TestClient(server.app)
+    r = client.post("/v1/query/", json={
+        "sql": "SELECT time(3600), Sum(request) AS r FROM application.1h "
+               "GROUP BY time(3600)"})
+    body = r.json()
+    assert body["OPT_STATUS"] == "SUCCESS", body
+    assert sum(v[1] for v in body["result"]["values"]) == 400
+    # custom interval via datasource API
+    client.post("/v1/datasources/", json={"name": "10m", "interval": 600})
+    r2 = client.post("/v1/query/", json={
+        "sql": "SELECT Sum(request) AS r FROM application.10m"})
+    assert r2.json()["result"]["values"][0][0] == 400
