@@ -19,6 +19,7 @@
 #include <unordered_map>
 #include <vector>
 
+#include "http2.h"
 #include "pbenc.h"
 
 using dfpb::Buf;
@@ -38,7 +39,7 @@ struct L7Pending {
     bool active = false;
     uint64_t req_ts = 0;
     uint32_t req_len = 0;
-    std::string req_type, domain, resource, endpoint;
+    std::string req_type, domain, resource, endpoint, service;
     uint32_t dns_id = 0;
 };
 
@@ -73,6 +74,7 @@ struct FlowNode {
     bool emitted_new = false;
     L7Pending l7;
     L7Counters l7c;
+    h2::DynTable h2dyn[2];  // HPACK dynamic tables (per direction)
 };
 
 struct FlowKeyC {
@@ -315,6 +317,7 @@ uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
 uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     std::string m;
     if (is_http_request(p, n, m) || is_http_response(p, n)) return 20;
+    if (h2::looks_like_http2(p, n)) return 21;
     if (server_port == 53 && n >= 12) return 120;
     // Redis RESP: '*<digits>\r\n' (request array) or simple-type replies on
     // the well-known port
@@ -401,6 +404,11 @@ void encode_l7_record(Agent& a, FlowNode& f, uint64_t req_ts, uint64_t resp_ts,
         dfpb::f_i(s, 2, (int64_t)status_code);
     });
     if (!version.empty()) dfpb::f_s(b, 13, version.c_str(), version.size());
+    if (!pend.service.empty()) {
+        dfpb::f_m<512>(b, 15, [&](Buf& s) {  // ext_info
+            dfpb::f_s(s, 1, pend.service.c_str(), pend.service.size());
+        });
+    }
     dfpb::f_u(b, 17, 255);  // direction_score
     emit_record(a.out_l7, buf, b.len);
     a.l7_emitted++;
@@ -584,6 +592,66 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
         } else if (dir == 1 && f.l7.active) {
             encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
             f.l7.active = false;
+        }
+        return;
+    }
+    if (f.l7_protocol == 21 || f.l7_protocol == 41) {  // HTTP/2 / gRPC
+        uint32_t pos = 0;
+        if (n >= 24 && memcmp(p, h2::PREFACE, 24) == 0) pos = 24;
+        h2::FrameView fr;
+        while (h2::next_frame(p, n, pos, fr)) {
+            if (fr.type != h2::F_HEADERS) continue;
+            uint32_t off = 0, pad = 0;
+            if (fr.flags & 0x08) {  // PADDED
+                if (fr.len < 1) continue;
+                pad = fr.payload[0];
+                off += 1;
+            }
+            if (fr.flags & 0x20) off += 5;  // PRIORITY
+            if (off + pad > fr.len) continue;
+            uint32_t blen = fr.len - off - pad;
+            std::vector<h2::Header> hs;
+            if (!h2::hpack_decode(fr.payload + off, blen, f.h2dyn[dir], hs))
+                continue;
+            std::string method, path, authority, status, ctype, grpc_status;
+            for (auto& h : hs) {
+                if (h.first == ":method") method = h.second;
+                else if (h.first == ":path") path = h.second;
+                else if (h.first == ":authority" || h.first == "host")
+                    authority = h.second;
+                else if (h.first == ":status") status = h.second;
+                else if (h.first == "content-type") ctype = h.second;
+                else if (h.first == "grpc-status") grpc_status = h.second;
+            }
+            if (!method.empty() && dir == 0) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = n;
+                f.l7.req_type = method;
+                f.l7.resource = path;
+                f.l7.endpoint = path;
+                f.l7.domain = authority;
+                f.l7.service.clear();
+                if (ctype.rfind("application/grpc", 0) == 0) {
+                    f.l7_protocol = 41;
+                    size_t slash = path.rfind('/');
+                    if (slash != std::string::npos && slash > 1)
+                        f.l7.service = path.substr(1, slash - 1);
+                }
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            } else if (!status.empty() && dir == 1 && f.l7.active) {
+                int code = atoi(status.c_str());
+                uint8_t st = code >= 500 ? 3 : (code >= 400 ? 4 : 0);
+                if (!grpc_status.empty() && grpc_status != "0") st = 3;
+                encode_l7_record(a, f, f.l7.req_ts, ts, code, st, f.l7, "2");
+                f.l7.active = false;
+            } else if (!grpc_status.empty() && dir == 1 && f.l7.active) {
+                // trailers-only completion
+                uint8_t st = grpc_status == "0" ? 0 : 3;
+                encode_l7_record(a, f, f.l7.req_ts, ts, 0, st, f.l7, "2");
+                f.l7.active = false;
+            }
         }
         return;
     }

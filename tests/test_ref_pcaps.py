@@ -106,3 +106,67 @@ def test_l4_flow_metrics_from_pcap():
     assert f["metrics_peer_src"]["packet_count"] > 0
     assert f["flow_key"]["port_dst"] == 6379 or \
         f["flow_key"]["port_src"] == 6379
+
+
+def test_grpc_pcaps():
+    """HTTP/2 + HPACK + gRPC against the reference's own golden captures
+    (expected values from the matching .result files)."""
+    l7, _, _ = replay(f"{FIX}/http/grpc-unary.pcap")
+    assert len(l7) == 1
+    r = l7[0]
+    assert r["base"]["head"]["proto"] == 41  # gRPC
+    assert r["req"]["req_type"] == "POST"
+    assert r["req"]["resource"] == "/agent.Synchronizer/Sync"
+    assert r["ext_info"]["service_name"] == "agent.Synchronizer"
+    assert r["resp"]["code"] == 200
+
+    l7b, _, _ = replay(f"{FIX}/http/grpc-server-stream.pcap")
+    assert any(x["req"]["resource"] ==
+               "/timeseriesquery.TimeSeriesQueryService/ServerStreamQuery"
+               for x in l7b)
+
+
+def test_grpc_in_agent_session():
+    """Synthetic HTTP/2 exchange through the normal packet path."""
+    import struct
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import (eth_ipv4_tcp, SYN, SYNACK,
+                                            PSH_ACK)
+
+    def hb(s):  # literal-without-indexing header (name+value raw strings)
+        n, v = s
+        return bytes([0x00, len(n)]) + n.encode() + \
+            bytes([len(v)]) + v.encode()
+
+    req_block = (b"\x83"  # :method POST (indexed 3)
+                 + hb((":path", "/svc.Api/Do"))
+                 + hb((":authority", "api.local"))
+                 + hb(("content-type", "application/grpc")))
+    req_frame = struct.pack(">I", len(req_block))[1:] + \
+        bytes([1, 0x04]) + struct.pack(">I", 1) + req_block
+    resp_block = b"\x88" + hb(("content-type", "application/grpc"))
+    resp_frame = struct.pack(">I", len(resp_block))[1:] + \
+        bytes([1, 0x04]) + struct.pack(">I", 1) + resp_block
+    a = Agent(vtap_id=5)
+    t0 = 10**9
+    pkts = [
+        (eth_ipv4_tcp(0x0A000001, 0x0A000002, 43000, 50051, SYN, 1), t0),
+        (eth_ipv4_tcp(0x0A000002, 0x0A000001, 50051, 43000, SYNACK, 2, 2),
+         t0 + 10**6),
+        (eth_ipv4_tcp(0x0A000001, 0x0A000002, 43000, 50051, PSH_ACK, 2, 3,
+                      req_frame), t0 + 2 * 10**6),
+        (eth_ipv4_tcp(0x0A000002, 0x0A000001, 50051, 43000, PSH_ACK, 3,
+                      2 + len(req_frame), resp_frame), t0 + 6 * 10**6),
+    ]
+    for frame, ts in pkts:
+        a.packet(frame, ts)
+    a.tick(10**9 * 100)
+    from deepflow_amd.wire import pb, flow_log, framing
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(a.drain(1))]
+    assert len(recs) == 1
+    assert recs[0]["base"]["head"]["proto"] == 41
+    assert recs[0]["req"]["resource"] == "/svc.Api/Do"
+    assert recs[0]["req"]["domain"] == "api.local"
+    assert recs[0]["ext_info"]["service_name"] == "svc.Api"
+    a.close()
