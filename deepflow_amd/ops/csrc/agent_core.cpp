@@ -306,6 +306,94 @@ bool parse_mongo_request(const uint8_t* p, uint32_t n, std::string& op,
     return true;
 }
 
+// MQTT fixed header: type nibble + varint remaining length; CONNECT
+// carries protocol name "MQTT"/"MQIsdp"
+static const char* MQTT_TYPES[] = {"", "CONNECT", "CONNACK", "PUBLISH",
+    "PUBACK", "PUBREC", "PUBREL", "PUBCOMP", "SUBSCRIBE", "SUBACK",
+    "UNSUBSCRIBE", "UNSUBACK", "PINGREQ", "PINGRESP", "DISCONNECT", "AUTH"};
+
+bool mqtt_varint(const uint8_t* p, uint32_t n, uint32_t& pos, uint32_t& out) {
+    out = 0;
+    int m = 0;
+    while (pos < n && m <= 21) {
+        uint8_t b = p[pos++];
+        out |= (uint32_t)(b & 0x7F) << m;
+        if (!(b & 0x80)) return true;
+        m += 7;
+    }
+    return false;
+}
+
+bool parse_mqtt(const uint8_t* p, uint32_t n, std::string& type,
+                std::string& topic, bool& is_connect) {
+    if (n < 2) return false;
+    uint8_t t = p[0] >> 4;
+    if (t == 0 || t > 15) return false;
+    uint32_t pos = 1, rem;
+    if (!mqtt_varint(p, n, pos, rem)) return false;
+    if (pos + rem > n + 4) return false;  // allow slight trailing slack
+    type = MQTT_TYPES[t];
+    is_connect = t == 1;
+    if (t == 1) {  // CONNECT: u16 len + protocol name
+        if (pos + 2 > n) return false;
+        uint16_t pl = (p[pos] << 8) | p[pos + 1];
+        if (pl > 8 || pos + 2 + pl > n) return false;
+        std::string proto((const char*)p + pos + 2, pl);
+        if (proto != "MQTT" && proto != "MQIsdp") return false;
+        topic = proto;
+    } else if (t == 3) {  // PUBLISH: u16 topic len + topic
+        if (pos + 2 > n) return false;
+        uint16_t tl = (p[pos] << 8) | p[pos + 1];
+        if (pos + 2 + tl > n) return false;
+        topic.assign((const char*)p + pos + 2, tl);
+    }
+    return true;
+}
+
+// AMQP 0-9-1: protocol header "AMQP\x00\x00\x09\x01" or frames
+// [type u8][chan u16][size u32][payload][0xCE]; method frame payload
+// starts with class-id u16 + method-id u16
+const char* amqp_method_name(uint16_t cls, uint16_t mth) {
+    if (cls == 10) return mth == 10 ? "Connection.Start" :
+                          mth == 11 ? "Connection.StartOk" :
+                          mth == 30 ? "Connection.Tune" :
+                          mth == 40 ? "Connection.Open" : "Connection";
+    if (cls == 20) return mth == 10 ? "Channel.Open" : "Channel";
+    if (cls == 40) return "Exchange.Declare";
+    if (cls == 50) return "Queue.Declare";
+    if (cls == 60) return mth == 40 ? "Basic.Publish" :
+                          mth == 20 ? "Basic.Consume" :
+                          mth == 60 ? "Basic.Deliver" :
+                          mth == 70 ? "Basic.Get" :
+                          mth == 80 ? "Basic.Ack" : "Basic";
+    return "Method";
+}
+
+bool parse_amqp(const uint8_t* p, uint32_t n, std::string& method,
+                bool& is_header) {
+    if (n >= 8 && memcmp(p, "AMQP", 4) == 0) {
+        method = "ProtocolHeader";
+        is_header = true;
+        return true;
+    }
+    is_header = false;
+    if (n < 12) return false;
+    uint8_t t = p[0];
+    uint32_t size = (p[3] << 24) | (p[4] << 16) | (p[5] << 8) | p[6];
+    if (t < 1 || t > 8 || size + 8 > n + 4 || size > (16u << 20))
+        return false;
+    if (size + 7 < n && p[7 + size] != 0xCE) return false;
+    if (t == 1 && size >= 4) {
+        uint16_t cls = (p[7] << 8) | p[8];
+        uint16_t mth = (p[9] << 8) | p[10];
+        method = amqp_method_name(cls, mth);
+    } else {
+        method = t == 2 ? "ContentHeader" : t == 3 ? "ContentBody"
+                                                    : "Heartbeat";
+    }
+    return true;
+}
+
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
 uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
@@ -344,6 +432,20 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     if (server_port == 5432 && n >= 6 && p[0] == 'P') return 61;
     if (server_port == 9092 && n >= 14) return 100;
     if (server_port == 27017 && n >= 16) return 81;
+    // AMQP protocol header (content-based)
+    if (n >= 8 && memcmp(p, "AMQP", 4) == 0) return 102;
+    if (server_port == 5672 && n >= 12) return 102;
+    // MQTT on its well-known ports (content check is weak alone)
+    if ((server_port == 1883 || server_port == 8883) && n >= 2) {
+        std::string ty, topic;
+        bool isc;
+        if (parse_mqtt(p, n, ty, topic, isc)) return 101;
+    }
+    if (n >= 10 && (p[0] >> 4) == 1 && p[4] == 'M') {
+        std::string ty, topic;
+        bool isc;
+        if (parse_mqtt(p, n, ty, topic, isc) && isc) return 101;
+    }
     // MongoDB header: little-endian msglen covering packet + known opcode
     if (n >= 16) {
         uint32_t mlen = p[0] | (p[1] << 8) | (p[2] << 16) | (p[3] << 24);
@@ -777,6 +879,44 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
                 f.l7.active = false;
             }
+        }
+    } else if (f.l7_protocol == 101) {  // MQTT
+        std::string ty, topic;
+        bool isc;
+        if (!parse_mqtt(p, n, ty, topic, isc)) return;
+        if (dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = ty;
+            f.l7.resource = topic;
+            f.l7.endpoint = topic;
+            f.l7.domain = "";
+            f.l7.service.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 102) {  // AMQP
+        std::string method;
+        bool is_hdr;
+        if (!parse_amqp(p, n, method, is_hdr)) return;
+        if (dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = method;
+            f.l7.resource = method;
+            f.l7.endpoint = method;
+            f.l7.domain = "";
+            f.l7.service.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+            f.l7.active = false;
         }
     } else if (f.l7_protocol == 60) {  // MySQL
         if (dir == 0) {

@@ -170,3 +170,40 @@ def test_grpc_in_agent_session():
     assert recs[0]["req"]["domain"] == "api.local"
     assert recs[0]["ext_info"]["service_name"] == "svc.Api"
     a.close()
+
+
+def test_mqtt_pcaps():
+    l7, _, _ = replay(f"{FIX}/mqtt/mqtt_pub.pcap")
+    types = [r["req"]["req_type"] for r in l7]
+    assert "CONNECT" in types and "PUBLISH" in types
+    pub = next(r for r in l7 if r["req"]["req_type"] == "PUBLISH")
+    assert pub["req"]["resource"] == "bench"
+    assert pub["base"]["head"]["proto"] == 101
+
+
+def test_amqp_pcaps():
+    l7, _, _ = replay(f"{FIX}/amqp/amqp1.pcap")
+    types = [r["req"]["req_type"] for r in l7]
+    assert "ProtocolHeader" in types
+    assert any(t.startswith("Connection.") for t in types)
+    assert all(r["base"]["head"]["proto"] == 102 for r in l7)
+
+
+def test_kafka_pcap():
+    l7, _, _ = replay(f"{FIX}/kafka/00-produce-v2.pcap")
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 100
+    assert l7[0]["req"]["req_type"] == "Produce"
+
+
+def test_mongo_pcap():
+    l7, _, _ = replay(f"{FIX}/mongo/mongo.pcap")
+    assert len(l7) >= 10
+    assert all(r["base"]["head"]["proto"] == 81 for r in l7)
+
+
+def test_pgsql_pcaps():
+    l7, _, _ = replay(f"{FIX}/postgre/simple_query.pcap")
+    assert l7[0]["req"]["resource"].strip().startswith("delete")
+    l7e, _, _ = replay(f"{FIX}/postgre/error.pcap")
+    assert l7e[0]["resp"]["status"] == 3
