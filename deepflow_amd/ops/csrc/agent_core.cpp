@@ -394,6 +394,92 @@ bool parse_amqp(const uint8_t* p, uint32_t n, std::string& method,
     return true;
 }
 
+// Memcached text protocol
+bool parse_memcached_request(const uint8_t* p, uint32_t n, std::string& cmd,
+                             std::string& key) {
+    static const char* cmds[] = {"get ", "gets ", "set ", "add ", "replace ",
+                                 "append ", "prepend ", "cas ", "delete ",
+                                 "incr ", "decr ", "touch ", "stats"};
+    for (const char* c : cmds) {
+        size_t cl = strlen(c);
+        if (n >= cl && memcmp(p, c, cl) == 0) {
+            cmd.assign(c, cl - (c[cl - 1] == ' ' ? 1 : 0));
+            uint32_t i = (uint32_t)cl, e = i;
+            while (e < n && p[e] != ' ' && p[e] != '\r') e++;
+            key.assign((const char*)p + i, e - i);
+            return true;
+        }
+    }
+    return false;
+}
+
+// Dubbo: 16-byte header (magic 0xdabb, flag, status, id, len) + hessian2
+// body whose first strings are dubbo-version, service, version, method
+bool parse_dubbo(const uint8_t* p, uint32_t n, bool& is_req, int& status,
+                 std::string& service, std::string& method) {
+    if (n < 16 || p[0] != 0xda || p[1] != 0xbb) return false;
+    is_req = (p[2] & 0x80) != 0;
+    status = p[3];
+    if (!is_req) return true;
+    uint32_t pos = 16;
+    std::string strs[4];
+    for (int s = 0; s < 4 && pos < n; s++) {
+        uint8_t l = p[pos];
+        if (l >= 0x20) return s >= 2;  // non-short-string: stop
+        pos++;
+        if (pos + l > n) return false;
+        strs[s].assign((const char*)p + pos, l);
+        pos += l;
+    }
+    service = strs[1];
+    method = strs[3];
+    return true;
+}
+
+// FastCGI records: [ver][type][reqId u16][clen u16][plen][rsvd] + content
+bool parse_fastcgi(const uint8_t* p, uint32_t n, bool& is_req,
+                   std::string& method, std::string& uri, int& code) {
+    if (n < 8 || p[0] != 1) return false;
+    is_req = false;
+    uint32_t pos = 0;
+    bool any = false;
+    while (pos + 8 <= n) {
+        uint8_t type = p[pos + 1];
+        uint16_t clen = (p[pos + 4] << 8) | p[pos + 5];
+        uint8_t plen = p[pos + 6];
+        const uint8_t* c = p + pos + 8;
+        uint32_t avail = n - pos - 8;
+        if (clen > avail) clen = (uint16_t)avail;
+        any = true;
+        if (type == 1) is_req = true;           // BEGIN_REQUEST
+        else if (type == 4 && clen) {           // PARAMS
+            is_req = true;
+            uint32_t i = 0;
+            while (i + 2 <= clen) {
+                uint32_t nl = c[i], vl;
+                if (nl > 127) break;  // long-form lengths: rare, skip
+                vl = c[i + 1];
+                if (vl > 127) break;
+                i += 2;
+                if (i + nl + vl > clen) break;
+                std::string name((const char*)c + i, nl);
+                std::string val((const char*)c + i + nl, vl);
+                if (name == "REQUEST_URI" || name == "SCRIPT_NAME" ||
+                    (uri.empty() && name == "SCRIPT_FILENAME"))
+                    uri = val;
+                if (name == "REQUEST_METHOD") method = val;
+                i += nl + vl;
+            }
+        } else if (type == 6 && clen) {         // STDOUT (response)
+            const char* st = (const char*)memmem(c, clen, "Status:", 7);
+            code = 200;
+            if (st) code = atoi(st + 7);
+        }
+        pos += 8 + clen + plen;
+    }
+    return any;
+}
+
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
 uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
@@ -432,6 +518,18 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     if (server_port == 5432 && n >= 6 && p[0] == 'P') return 61;
     if (server_port == 9092 && n >= 14) return 100;
     if (server_port == 27017 && n >= 16) return 81;
+    // Dubbo magic
+    if (n >= 16 && p[0] == 0xda && p[1] == 0xbb) return 40;
+    // Memcached text commands on the well-known port
+    if (server_port == 11211 && n >= 4) {
+        std::string c, k;
+        if (parse_memcached_request(p, n, c, k)) return 82;
+        if (n >= 5 && (memcmp(p, "VALUE", 5) == 0 ||
+                       memcmp(p, "STORE", 5) == 0)) return 82;
+    }
+    // FastCGI version-1 records on port 9000
+    if (server_port == 9000 && n >= 8 && p[0] == 1 && p[1] >= 1 && p[1] <= 11)
+        return 44;
     // AMQP protocol header (content-based)
     if (n >= 8 && memcmp(p, "AMQP", 4) == 0) return 102;
     if (server_port == 5672 && n >= 12) return 102;
@@ -879,6 +977,71 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
                 f.l7.active = false;
             }
+        }
+    } else if (f.l7_protocol == 82) {  // Memcached
+        if (dir == 0) {
+            std::string cmd, key;
+            if (parse_memcached_request(p, n, cmd, key)) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = n;
+                f.l7.req_type = cmd;
+                f.l7.resource = key;
+                f.l7.endpoint = cmd;
+                f.l7.domain = "";
+                f.l7.service.clear();
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            }
+        } else if (f.l7.active && n >= 3) {
+            bool err = memcmp(p, "ERROR", n < 5 ? n : 5) == 0 ||
+                       (n >= 12 && memcmp(p, "CLIENT_ERROR", 12) == 0) ||
+                       (n >= 12 && memcmp(p, "SERVER_ERROR", 12) == 0);
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, err ? 3 : 0, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 40) {  // Dubbo
+        bool is_req;
+        int status;
+        std::string service, method;
+        if (!parse_dubbo(p, n, is_req, status, service, method)) return;
+        if (is_req && dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = method;
+            f.l7.resource = service + "/" + method;
+            f.l7.endpoint = method;
+            f.l7.domain = service;
+            f.l7.service = service;
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (!is_req && dir == 1 && f.l7.active) {
+            // dubbo status 20 == OK
+            encode_l7_record(a, f, f.l7.req_ts, ts, status,
+                             status == 20 ? 0 : 3, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 44) {  // FastCGI
+        bool is_req;
+        std::string method, uri;
+        int code = 0;
+        if (!parse_fastcgi(p, n, is_req, method, uri, code)) return;
+        if (is_req && dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = method.empty() ? "FCGI" : method;
+            f.l7.resource = uri;
+            f.l7.endpoint = uri;
+            f.l7.domain = "";
+            f.l7.service.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (!is_req && dir == 1 && f.l7.active) {
+            uint8_t st = code >= 500 ? 3 : (code >= 400 ? 4 : 0);
+            encode_l7_record(a, f, f.l7.req_ts, ts, code, st, f.l7, "");
+            f.l7.active = false;
         }
     } else if (f.l7_protocol == 101) {  // MQTT
         std::string ty, topic;

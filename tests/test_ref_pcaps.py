@@ -207,3 +207,28 @@ def test_pgsql_pcaps():
     assert l7[0]["req"]["resource"].strip().startswith("delete")
     l7e, _, _ = replay(f"{FIX}/postgre/error.pcap")
     assert l7e[0]["resp"]["status"] == 3
+
+
+def test_memcached_pcap():
+    l7, _, _ = replay(f"{FIX}/memcached/memcached.pcap")
+    assert [(r["req"]["req_type"], r["req"]["resource"]) for r in l7] == \
+        [("set", "foo"), ("get", "foo")]
+
+
+def test_dubbo_pcap():
+    l7, _, _ = replay(f"{FIX}/dubbo/dubbo_hessian2.pcap")
+    assert len(l7) == 1
+    r = l7[0]
+    assert r["base"]["head"]["proto"] == 40
+    assert r["req"]["resource"] == "my.demo.service.UserService/login"
+    assert r["ext_info"]["service_name"] == "my.demo.service.UserService"
+    # dubbo status 20 == OK
+    assert r["resp"].get("status", 0) == 0
+
+
+def test_fastcgi_pcap():
+    l7, _, _ = replay(f"{FIX}/fastcgi/fastcgi.pcap")
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 44
+    assert l7[0]["req"]["req_type"] == "GET"
+    assert l7[0]["resp"]["code"] == 200
