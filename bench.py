@@ -65,6 +65,7 @@ def main() -> None:
     ap.add_argument("--batch", type=int, default=1_000_000,
                     help="spans per step per rank")
     ap.add_argument("--tag-card", type=int, default=100_000)
+    ap.add_argument("--n-attrs", type=int, default=4)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -85,7 +86,7 @@ def main() -> None:
     cfg = SpanGenConfig(n=args.batch, seed=1234,
                         tag_cardinality=args.tag_card,
                         n_ips=4096, n_services=256, n_resources=4096,
-                        n_attrs=4)
+                        n_attrs=args.n_attrs)
     n_distinct = min(args.steps + args.warmup, 4)
     batches = gen_batches(cfg, rank, n_distinct, args.batch,
                           pinned=device == "cuda")
