@@ -121,6 +121,47 @@ class DeepflowServer:
         self.mcp = McpServer(self.engine, self.profiles)
         self.mcp.register(self.app)
 
+        @self.app.get("/v1/debug/threads")
+        def debug_threads():
+            # self-tracing aid (reference: server self-pprof,
+            # ingester/droplet/profiler)
+            import sys
+            import traceback
+            out = {}
+            for tid, frame in sys._current_frames().items():
+                out[str(tid)] = traceback.format_stack(frame)[-4:]
+            return out
+
+        @self.app.get("/v1/debug/store")
+        def debug_store():
+            segs = self.l7.segments
+            return {
+                "l7_rows": segs.n_rows,
+                "l7_segments": len(segs.segments),
+                "l7_alloc_bytes": segs.total_alloc_bytes(),
+                "l7_evicted_rows": segs.evicted_rows,
+                "layout_version": getattr(segs.segments[0],
+                                          "layout_version", 0)
+                if segs.segments else 0,
+                "dict_entries": self.l7.dict.n_entries(),
+                "l4_rows": self.l4.segments.n_rows,
+            }
+
+        @self.app.post("/v1/debug/self-profile")
+        def self_profile(seconds: float = 1.0):
+            # continuous self-profiling hook: capture our own GPU kernel
+            # activity into the profile store (reference: server pushes its
+            # own pprof into DeepFlow, cmd/server/config.go:47)
+            if self.device != "cuda":
+                return {"status": "skipped", "reason": "cpu device"}
+            import time as _t
+            from .profiler import GpuProfiler
+            gp = GpuProfiler(self.profiles, process_name="deepflow-server")
+            with gp.capture():
+                _t.sleep(min(seconds, 10.0))
+            return {"status": "ok",
+                    "rows": len(self.profiles.store.rows)}
+
         @self.app.post("/v1/datasources/")
         async def add_datasource(request: Request):
             body = await request.json()

@@ -2,6 +2,12 @@
 columnar layout. tests/test_layout_sync.py parses the header and asserts
 these stay in sync."""
 
+# Storage layout generation (ckissu analog): bump when any column block's
+# shape or packing changes; segments record it so mixed-layout windows are
+# rejected instead of misread (migration = drain + reingest, matching the
+# reference's at-most-once durability posture).
+LAYOUT_VERSION = 2  # v2 = packed attr pool + row string block
+
 U64_COLS = [
     "start_time", "end_time", "flow_id", "rrt", "syscall_trace_id_request",
     "syscall_trace_id_response",

@@ -39,6 +39,7 @@ class L7Segment:
         self.pool = z((pool_capacity or capacity * 96,), torch.uint8)
         self.pool_len = 0
         self.n_rows = 0
+        self.layout_version = S.LAYOUT_VERSION
 
     def ensure_attr_pool(self, extra: int) -> None:
         need = self.attr_pool_len + extra

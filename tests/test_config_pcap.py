@@ -48,3 +48,21 @@ def test_pcap_http():
     r = client.get("/v1/pcap/7")
     assert r.status_code == 200 and len(r.content) == 24 + 16 + 40
     assert client.get("/v1/pcap/99").status_code == 404
+
+
+def test_debug_endpoints():
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 9,
+                         dict_capacity=1 << 10)
+    srv.receiver.handle_frame(framing.encode_frame(
+        framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
+        gen_span_payload(SpanGenConfig(n=20, seed=2, tag_cardinality=5))))
+    client = TestClient(srv.app)
+    st = client.get("/v1/debug/store").json()
+    assert st["l7_rows"] == 20
+    assert st["layout_version"] >= 2
+    th = client.get("/v1/debug/threads").json()
+    assert len(th) >= 1
+    sp = client.post("/v1/debug/self-profile").json()
+    assert sp["status"] == "skipped"
