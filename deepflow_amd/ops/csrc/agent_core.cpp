@@ -480,6 +480,219 @@ bool parse_fastcgi(const uint8_t* p, uint32_t n, bool& is_req,
     return any;
 }
 
+// NATS text protocol (verbs; server greets with INFO)
+bool parse_nats(const uint8_t* p, uint32_t n, std::string& verb,
+                std::string& subject, bool& is_client_msg) {
+    static const char* client_verbs[] = {"CONNECT", "PUB ", "HPUB ", "SUB ",
+                                         "UNSUB ", "PING"};
+    static const char* server_verbs[] = {"INFO ", "MSG ", "HMSG ", "+OK",
+                                         "-ERR", "PONG"};
+    for (const char* v : client_verbs) {
+        size_t l = strlen(v);
+        if (n >= l && memcmp(p, v, l) == 0) {
+            is_client_msg = true;
+            verb.assign(v, v[l - 1] == ' ' ? l - 1 : l);
+            if (verb == "PUB" || verb == "HPUB" || verb == "SUB") {
+                uint32_t i = (uint32_t)l, e = i;
+                while (e < n && p[e] != ' ' && p[e] != '\r') e++;
+                subject.assign((const char*)p + i, e - i);
+            }
+            return true;
+        }
+    }
+    for (const char* v : server_verbs) {
+        size_t l = strlen(v);
+        if (n >= l && memcmp(p, v, l) == 0) {
+            is_client_msg = false;
+            verb.assign(v, v[l - 1] == ' ' ? l - 1 : l);
+            return true;
+        }
+    }
+    return false;
+}
+
+// RocketMQ remoting: [total len u32][header len u32 (low 24b) | serializer]
+// + JSON header {"code":N, "flag":F, "extFields":{"topic":...}}
+bool parse_rocketmq(const uint8_t* p, uint32_t n, int& code, bool& is_resp,
+                    std::string& topic) {
+    if (n < 9) return false;
+    uint32_t total = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
+    uint32_t hlen = (p[5] << 16) | (p[6] << 8) | p[7];
+    if (total < 4 || total > (32u << 20) || hlen + 8 > n + 8 ||
+        p[8] != '{')
+        return false;
+    const char* j = (const char*)p + 8;
+    uint32_t jl = hlen > n - 8 ? n - 8 : hlen;
+    std::string js(j, jl);
+    code = -1;
+    size_t cpos = js.find("\"code\":");
+    if (cpos != std::string::npos) code = atoi(js.c_str() + cpos + 7);
+    size_t fpos = js.find("\"flag\":");
+    int flag = fpos != std::string::npos ? atoi(js.c_str() + fpos + 7) : 0;
+    is_resp = (flag & 1) != 0;
+    size_t tpos = js.find("\"topic\":\"");
+    if (tpos != std::string::npos) {
+        size_t s = tpos + 9, e = js.find('"', s);
+        if (e != std::string::npos) topic = js.substr(s, e - s);
+    }
+    return code >= 0;
+}
+
+static const char* rocketmq_code_name(int code) {
+    switch (code) {
+        case 10: return "SendMessage";
+        case 11: return "PullMessage";
+        case 310: return "SendMessageV2";
+        case 105: return "GetRouteInfo";
+        case 34: return "HeartBeat";
+        default: return "RemotingCommand";
+    }
+}
+
+// SofaRPC bolt v1: proto(1)=1, type(1), cmdcode(2), ver(1), reqId(4),
+// codec(1), then request: timeout(4), classLen(2), headerLen(2),
+// contentLen(4), className | response: respStatus(2), classLen(2), ...
+bool parse_sofarpc(const uint8_t* p, uint32_t n, bool& is_req, int& status,
+                   std::string& class_name) {
+    if (n < 20 || p[0] != 1) return false;
+    uint8_t type = p[1];
+    uint16_t cmd = (p[2] << 8) | p[3];
+    if (cmd == 0 || cmd > 2) return false;
+    if (type == 1 && cmd == 1) {  // request
+        is_req = true;
+        uint16_t clen = (p[14] << 8) | p[15];
+        if (22u + clen <= n)
+            class_name.assign((const char*)p + 22, clen);
+        return true;
+    }
+    if (type == 0 || cmd == 2) {  // response
+        is_req = false;
+        status = (p[10] << 8) | p[11];
+        return true;
+    }
+    return false;
+}
+
+// bRPC: "PRPC" + body_size u32 + meta_size u32 + pb meta
+// (request meta field 1 { service=1, method=2 }; response field 2)
+static bool brpc_pb_string(const uint8_t* p, uint32_t n, uint32_t& pos,
+                           std::string& out) {
+    uint32_t len = 0;
+    int sh = 0;
+    while (pos < n) {
+        uint8_t b = p[pos++];
+        len |= (b & 0x7F) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+    }
+    if (pos + len > n) return false;
+    out.assign((const char*)p + pos, len);
+    pos += len;
+    return true;
+}
+
+bool parse_brpc(const uint8_t* p, uint32_t n, bool& is_req,
+                std::string& service, std::string& method) {
+    if (n < 16 || memcmp(p, "PRPC", 4) != 0) return false;
+    uint32_t meta_size = (p[8] << 24) | (p[9] << 16) | (p[10] << 8) | p[11];
+    const uint8_t* m = p + 12;
+    uint32_t mn = meta_size > n - 12 ? n - 12 : meta_size;
+    uint32_t pos = 0;
+    is_req = false;
+    while (pos < mn) {
+        uint8_t key = m[pos++];
+        uint32_t num = key >> 3, wt = key & 7;
+        if (wt != 2) {  // varint field: skip
+            while (pos < mn && (m[pos] & 0x80)) pos++;
+            pos++;
+            continue;
+        }
+        uint32_t len = 0;
+        int sh = 0;
+        while (pos < mn) {
+            uint8_t b = m[pos++];
+            len |= (b & 0x7F) << sh;
+            if (!(b & 0x80)) break;
+            sh += 7;
+        }
+        if (num == 1) {  // request meta
+            is_req = true;
+            uint32_t p2 = pos, e2 = pos + len;
+            while (p2 < e2) {
+                uint8_t k2 = m[p2++];
+                if ((k2 & 7) == 2) {
+                    std::string s;
+                    if (!brpc_pb_string(m, e2, p2, s)) break;
+                    if ((k2 >> 3) == 1) service = s;
+                    else if ((k2 >> 3) == 2) method = s;
+                } else {
+                    while (p2 < e2 && (m[p2] & 0x80)) p2++;
+                    p2++;
+                }
+            }
+        }
+        pos += len;
+    }
+    return true;
+}
+
+// Tars (jce header): 0x10 ver, 0x2c ptype, 0x3c mtype, 0x40 reqid,
+// 0x56 <len> servant, 0x66 <len> func
+bool parse_tars(const uint8_t* p, uint32_t n, bool& is_req,
+                std::string& servant, std::string& func) {
+    if (n < 10) return false;
+    uint32_t total = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
+    if (total < 10 || total > (16u << 20)) return false;
+    if (p[4] != 0x10) return false;  // tag1 type0 (iVersion)
+    is_req = false;
+    for (uint32_t i = 4; i + 2 < n && i < 64; i++) {
+        if (p[i] == 0x56) {  // tag5 string1 = servant
+            uint8_t l = p[i + 1];
+            if (i + 2 + l <= n) {
+                servant.assign((const char*)p + i + 2, l);
+                uint32_t j = i + 2 + l;
+                if (j + 2 <= n && p[j] == 0x66) {
+                    uint8_t fl = p[j + 1];
+                    if (j + 2 + fl <= n)
+                        func.assign((const char*)p + j + 2, fl);
+                }
+                is_req = true;
+                return true;
+            }
+        }
+    }
+    return true;  // header-only (response)
+}
+
+// TLS ClientHello: record type 22, handshake type 1; SNI in extension 0
+bool parse_tls_client_hello(const uint8_t* p, uint32_t n, std::string& sni) {
+    if (n < 46 || p[0] != 0x16 || p[1] != 3 || p[5] != 1) return false;
+    uint32_t pos = 43;  // record(5) + hs(4) + ver(2) + random(32)
+    if (pos >= n) return false;
+    pos += 1 + p[pos];                       // session id
+    if (pos + 2 > n) return false;
+    pos += 2 + ((p[pos] << 8) | p[pos + 1]);  // cipher suites
+    if (pos + 1 > n) return false;
+    pos += 1 + p[pos];                       // compression
+    if (pos + 2 > n) return true;            // no extensions
+    uint32_t ext_end = pos + 2 + ((p[pos] << 8) | p[pos + 1]);
+    pos += 2;
+    if (ext_end > n) ext_end = n;
+    while (pos + 4 <= ext_end) {
+        uint16_t et = (p[pos] << 8) | p[pos + 1];
+        uint16_t el = (p[pos + 2] << 8) | p[pos + 3];
+        pos += 4;
+        if (et == 0 && pos + 5 <= ext_end) {  // server_name
+            uint16_t nl = (p[pos + 3] << 8) | p[pos + 4];
+            if (pos + 5 + nl <= ext_end)
+                sni.assign((const char*)p + pos + 5, nl);
+            return true;
+        }
+        pos += el;
+    }
+    return true;
+}
+
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
 uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
@@ -520,6 +733,46 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
     if (server_port == 27017 && n >= 16) return 81;
     // Dubbo magic
     if (n >= 16 && p[0] == 0xda && p[1] == 0xbb) return 40;
+    // bRPC magic
+    if (n >= 16 && memcmp(p, "PRPC", 4) == 0) return 45;
+    // TLS handshake record
+    if (n >= 6 && p[0] == 0x16 && p[1] == 3 && p[5] == 1) return 121;
+    // NATS verbs
+    {
+        std::string v, s;
+        bool c;
+        if ((server_port == 4222 && n >= 4) && parse_nats(p, n, v, s, c))
+            return 104;
+        if (n >= 6 && (memcmp(p, "INFO {", 6) == 0 ||
+                       memcmp(p, "CONNECT ", 8 > n ? n : 8) == 0))
+            return 104;
+    }
+    // RocketMQ: length-prefixed JSON header
+    {
+        int code;
+        bool r;
+        std::string t;
+        if (n >= 12 && p[8] == '{' && parse_rocketmq(p, n, code, r, t))
+            return 107;
+    }
+    // SofaRPC bolt
+    if (server_port == 12200 && n >= 20 && p[0] == 1) return 43;
+    if (n >= 24 && p[0] == 1 && (p[1] == 0 || p[1] == 1) &&
+        ((p[2] << 8) | p[3]) <= 2 && n > 22) {
+        // require the request class-name to look like a java class
+        if (p[1] == 1 && n >= 32 && memcmp(p + 22, "com.", 4) == 0) return 43;
+    }
+    // Tars
+    if (server_port == 18993 || server_port == 18913) {
+        std::string sv, fn;
+        bool r;
+        if (parse_tars(p, n, r, sv, fn)) return 46;
+    }
+    if (n >= 10 && p[4] == 0x10 && p[5] == 0x01 && p[6] == 0x2c) {
+        std::string sv, fn;
+        bool r;
+        if (parse_tars(p, n, r, sv, fn) && !sv.empty()) return 46;
+    }
     // Memcached text commands on the well-known port
     if (server_port == 11211 && n >= 4) {
         std::string c, k;
@@ -977,6 +1230,129 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
                 f.l7.active = false;
             }
+        }
+    } else if (f.l7_protocol == 104) {  // NATS (dir-agnostic verbs)
+        std::string verb, subject;
+        bool is_client;
+        if (!parse_nats(p, n, verb, subject, is_client)) return;
+        if (is_client && verb != "CONNECT") {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = verb;
+            f.l7.resource = subject;
+            f.l7.endpoint = subject;
+            f.l7.domain = "";
+            f.l7.service.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+            if (verb == "PUB" || verb == "HPUB") {
+                // fire-and-forget publish: emit immediately as session
+                encode_l7_record(a, f, ts, ts, 0, 0, f.l7, "");
+                f.l7.active = false;
+            }
+        } else if (!is_client && f.l7.active) {
+            bool err = verb == "-ERR";
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, err ? 3 : 0, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 107) {  // RocketMQ
+        int code;
+        bool is_resp;
+        std::string topic;
+        if (!parse_rocketmq(p, n, code, is_resp, topic)) return;
+        if (!is_resp) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = rocketmq_code_name(code);
+            f.l7.resource = topic;
+            f.l7.endpoint = f.l7.req_type;
+            f.l7.domain = "";
+            f.l7.service.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, code,
+                             code == 0 ? 0 : 3, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 43) {  // SofaRPC (bolt)
+        bool is_req;
+        int status = 0;
+        std::string cls;
+        if (!parse_sofarpc(p, n, is_req, status, cls)) return;
+        if (is_req && dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = "SofaRequest";
+            f.l7.resource = cls;
+            f.l7.endpoint = cls;
+            f.l7.domain = "";
+            f.l7.service = cls;
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (!is_req && dir == 1 && f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, status,
+                             status == 0 ? 0 : 3, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 45) {  // bRPC
+        bool is_req;
+        std::string service, method;
+        if (!parse_brpc(p, n, is_req, service, method)) return;
+        if (is_req && dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = method;
+            f.l7.resource = service + "/" + method;
+            f.l7.endpoint = method;
+            f.l7.domain = service;
+            f.l7.service = service;
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (!is_req && dir == 1 && f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 46) {  // Tars
+        bool is_req;
+        std::string servant, func;
+        if (!parse_tars(p, n, is_req, servant, func)) return;
+        if (is_req && dir == 0 && !servant.empty()) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = func;
+            f.l7.resource = servant + "/" + func;
+            f.l7.endpoint = func;
+            f.l7.domain = servant;
+            f.l7.service = servant;
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (dir == 1 && f.l7.active) {
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+            f.l7.active = false;
+        }
+    } else if (f.l7_protocol == 121) {  // TLS: ClientHello SNI only
+        std::string sni;
+        if (dir == 0 && parse_tls_client_hello(p, n, sni)) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = "ClientHello";
+            f.l7.resource = sni;
+            f.l7.endpoint = sni;
+            f.l7.domain = sni;
+            f.l7.service.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (dir == 1 && f.l7.active && n >= 6 && p[0] == 0x16 &&
+                   p[5] == 2) {  // ServerHello
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+            f.l7.active = false;
         }
     } else if (f.l7_protocol == 82) {  // Memcached
         if (dir == 0) {

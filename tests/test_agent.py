@@ -240,3 +240,35 @@ def test_guard_and_sync():
     me = next(x for x in agents if x["agent_id"] == 9)
     assert me["exceptions"] & Agent.EXC_MEM_LIMIT
     a.close()
+
+
+def test_tls_sni(agent):
+    import struct
+    sni = b"api.internal.example"
+    ext = struct.pack(">HHHBH", 0, len(sni) + 5, len(sni) + 3, 0,
+                      len(sni)) + sni
+    body = (b"\x03\x03" + b"\x00" * 32      # version + random
+            + b"\x00"                        # session id len
+            + struct.pack(">H", 2) + b"\x13\x01"  # cipher suites
+            + b"\x01\x00"                    # compression
+            + struct.pack(">H", len(ext)) + ext)
+    hs = b"\x01" + struct.pack(">I", len(body))[1:] + body
+    rec = b"\x16\x03\x01" + struct.pack(">H", len(hs)) + hs
+    server_hello = b"\x16\x03\x03\x00\x06\x02\x00\x00\x02\x03\x03"
+    from deepflow_amd.agent.packets import eth_ipv4_tcp, SYN, SYNACK, PSH_ACK
+    t0 = 10**9
+    pkts = [
+        (eth_ipv4_tcp(CLIENT, SERVER, 48000, 443, SYN, 1), t0),
+        (eth_ipv4_tcp(SERVER, CLIENT, 443, 48000, SYNACK, 2, 2), t0 + 10**6),
+        (eth_ipv4_tcp(CLIENT, SERVER, 48000, 443, PSH_ACK, 2, 3, rec),
+         t0 + 2 * 10**6),
+        (eth_ipv4_tcp(SERVER, CLIENT, 443, 48000, PSH_ACK, 3, 2 + len(rec),
+                      server_hello), t0 + 4 * 10**6),
+    ]
+    for frame, ts in pkts:
+        agent.packet(frame, ts)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 1
+    assert l7[0]["base"]["head"]["proto"] == 121
+    assert l7[0]["req"]["domain"] == "api.internal.example"

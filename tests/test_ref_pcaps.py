@@ -232,3 +232,25 @@ def test_fastcgi_pcap():
     assert l7[0]["base"]["head"]["proto"] == 44
     assert l7[0]["req"]["req_type"] == "GET"
     assert l7[0]["resp"]["code"] == 200
+
+
+def test_rpc_mq_pcaps():
+    l7, _, _ = replay(f"{FIX}/brpc/brpc-echo.pcap")
+    assert all(r["base"]["head"]["proto"] == 45 for r in l7)
+    assert l7[0]["req"]["resource"] == "example.EchoService/Echo"
+    assert l7[0]["ext_info"]["service_name"] == "example.EchoService"
+
+    l7, _, _ = replay(f"{FIX}/tars/tars-echo.pcap")
+    assert any(r["req"]["resource"] == "tars.tarslog.LogObj/tars_ping"
+               for r in l7)
+
+    l7, _, _ = replay(f"{FIX}/sofarpc/sofa-old.pcap")
+    assert len(l7) == 1 and l7[0]["base"]["head"]["proto"] == 43
+    assert l7[0]["req"]["resource"].startswith("com.alipay.sofa.rpc")
+
+    l7, _, _ = replay(f"{FIX}/rocketmq/rocketmq-send-message-v2.pcap")
+    assert len(l7) == 2
+    assert all(r["req"]["req_type"] == "SendMessageV2" for r in l7)
+
+    l7, _, _ = replay(f"{FIX}/nats/nats-skywalking.pcap")
+    assert any(r["req"]["req_type"] in ("PUB", "HPUB") for r in l7)
