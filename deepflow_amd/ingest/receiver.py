@@ -8,6 +8,7 @@ Reference counterpart: server/libs/receiver/receiver.go.
 from __future__ import annotations
 
 import ctypes as ct
+import queue as _queue
 import socket
 import struct
 import threading
@@ -54,7 +55,7 @@ class AgentStatus:
 
 class Receiver:
     def __init__(self, tcp_port: int = 20033, udp_port: int = 20033,
-                 host: str = "127.0.0.1"):
+                 host: str = "127.0.0.1", queue_depth: int = 256):
         self.host = host
         self.tcp_port = tcp_port
         self.udp_port = udp_port
@@ -65,6 +66,11 @@ class Receiver:
         self._threads = []
         self._tcp_sock: Optional[socket.socket] = None
         self._udp_sock: Optional[socket.socket] = None
+        # decode queue decoupling socket reads from pipeline work
+        # (reference: receiver hashes frames to N decoder queues with
+        # overwrite-on-full drop accounting, receiver.go:519-566)
+        self._queue: "_queue.Queue" = _queue.Queue(maxsize=queue_depth)
+        self._dispatcher: Optional[threading.Thread] = None
 
     def register(self, msg_type: int, handler: Handler) -> None:
         self.handlers[msg_type] = handler
@@ -95,8 +101,36 @@ class Receiver:
         handler(hdr, data)
         return True
 
+    def enqueue_frame(self, frame: bytes) -> bool:
+        """Queue a frame for the dispatcher thread; drops (with counter)
+        when the pipeline is backed up — at-most-once like the reference's
+        overwrite queues."""
+        if self._dispatcher is None:
+            return self.handle_frame(frame)
+        try:
+            self._queue.put_nowait(frame)
+            return True
+        except _queue.Full:
+            self.counter.add("queue_drops")
+            return False
+
+    def _dispatch_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                frame = self._queue.get(timeout=0.5)
+            except _queue.Empty:
+                continue
+            try:
+                self.handle_frame(frame)
+            except Exception:  # noqa: BLE001 — a bad frame must not kill
+                self.counter.add("handler_errors")
+
     # ------------------------------------------------------------- servers
     def start(self) -> None:
+        self._dispatcher = threading.Thread(target=self._dispatch_loop,
+                                            daemon=True)
+        self._dispatcher.start()
+        self._threads.append(self._dispatcher)
         self._tcp_sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
         self._tcp_sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
         self._tcp_sock.bind((self.host, self.tcp_port))
@@ -158,7 +192,7 @@ class Receiver:
                     break
                 if len(buf) < size:
                     break
-                self.handle_frame(buf[:size])
+                self.enqueue_frame(buf[:size])
                 buf = buf[size:]
         conn.close()
 
@@ -171,4 +205,4 @@ class Receiver:
             except OSError:
                 return
             if len(data) >= framing.HEADER_LEN:
-                self.handle_frame(data)
+                self.enqueue_frame(data)
