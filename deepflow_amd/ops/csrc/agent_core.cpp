@@ -75,6 +75,7 @@ struct FlowNode {
     L7Pending l7;
     L7Counters l7c;
     h2::DynTable h2dyn[2];  // HPACK dynamic tables (per direction)
+    std::map<uint32_t, L7Pending> h2_pending;  // per-stream outstanding reqs
 };
 
 struct FlowKeyC {
@@ -1076,34 +1077,39 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 else if (h.first == "content-type") ctype = h.second;
                 else if (h.first == "grpc-status") grpc_status = h.second;
             }
+            // stream-id-aware matching: gRPC/h2 multiplexes many
+            // concurrent requests per connection
             if (!method.empty() && dir == 0) {
-                f.l7.active = true;
-                f.l7.req_ts = ts;
-                f.l7.req_len = n;
-                f.l7.req_type = method;
-                f.l7.resource = path;
-                f.l7.endpoint = path;
-                f.l7.domain = authority;
-                f.l7.service.clear();
+                if (f.h2_pending.size() >= 64)
+                    f.h2_pending.erase(f.h2_pending.begin());
+                L7Pending pend;
+                pend.active = true;
+                pend.req_ts = ts;
+                pend.req_len = n;
+                pend.req_type = method;
+                pend.resource = path;
+                pend.endpoint = path;
+                pend.domain = authority;
                 if (ctype.rfind("application/grpc", 0) == 0) {
                     f.l7_protocol = 41;
                     size_t slash = path.rfind('/');
                     if (slash != std::string::npos && slash > 1)
-                        f.l7.service = path.substr(1, slash - 1);
+                        pend.service = path.substr(1, slash - 1);
                 }
+                f.h2_pending[fr.stream_id] = std::move(pend);
                 f.l7c.request_count++;
                 f.last_req_pkt_ts = ts;
-            } else if (!status.empty() && dir == 1 && f.l7.active) {
-                int code = atoi(status.c_str());
-                uint8_t st = code >= 500 ? 3 : (code >= 400 ? 4 : 0);
-                if (!grpc_status.empty() && grpc_status != "0") st = 3;
-                encode_l7_record(a, f, f.l7.req_ts, ts, code, st, f.l7, "2");
-                f.l7.active = false;
-            } else if (!grpc_status.empty() && dir == 1 && f.l7.active) {
-                // trailers-only completion
-                uint8_t st = grpc_status == "0" ? 0 : 3;
-                encode_l7_record(a, f, f.l7.req_ts, ts, 0, st, f.l7, "2");
-                f.l7.active = false;
+            } else if (dir == 1 &&
+                       (!status.empty() || !grpc_status.empty())) {
+                auto it = f.h2_pending.find(fr.stream_id);
+                if (it != f.h2_pending.end()) {
+                    int code = status.empty() ? 0 : atoi(status.c_str());
+                    uint8_t st = code >= 500 ? 3 : (code >= 400 ? 4 : 0);
+                    if (!grpc_status.empty() && grpc_status != "0") st = 3;
+                    encode_l7_record(a, f, it->second.req_ts, ts, code, st,
+                                     it->second, "2");
+                    f.h2_pending.erase(it);
+                }
             }
         }
         return;
