@@ -104,6 +104,19 @@ class QueryEngine:
         if table in row_tables:
             rows = getattr(self, row_tables[table], lambda: [])()
             return self._run_rows(sql, rows, time_base_s=0)
+        if table.startswith("application_map") or \
+                table.startswith("network_map"):
+            # _map series (per client/server pair): derived on demand from
+            # the columnar store by the GPU group-by kernel — at B-rows/s
+            # scan rates a persisted map rollup is unnecessary; the table
+            # surface matches the reference's flow_metrics *_map tables.
+            rows = self._map_rows("app" if table.startswith("application")
+                                  else "net",
+                                  self._table_interval(table)
+                                  if "." in table else 1)
+            return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s
+                                  if table.startswith("application")
+                                  else self.l4.time_base_s)
         if table == "application.agent":
             rows = getattr(self, "agent_app_rows", lambda: [])()
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
@@ -149,6 +162,49 @@ class QueryEngine:
         if not hasattr(self, "extra_datasources"):
             self.extra_datasources = {}
         self.extra_datasources[name] = interval_s
+
+    def _map_rows(self, kind: str, interval_s: int = 1) -> List[Dict]:
+        """(time, ip pair, server_port) series derived from the segment
+        store via the group-by kernel."""
+        plan = Q.Plan(time_base_s=self.pipe.time_base_s if kind == "app"
+                      else (self.l4.time_base_s if self.l4 else 0))
+        if kind == "app":
+            segs = self.pipe.segments.segments
+            from .tags import L7_TAGS as T
+            dur = T["response_duration"]
+            blen = T["response_length"]
+        else:
+            if self.l4 is None:
+                raise SqlError("network_map needs the l4 pipeline")
+            segs = self.l4.segments.segments
+            from .tags import L4_TAGS as T
+            dur = T["rtt"]
+            blen = T["byte_rx"]
+        plan.keys = [Q.Key(Q.SRC_TIME_BUCKET, 0, interval_s),
+                     Q.Key(T["ip4_0"].family, T["ip4_0"].idx),
+                     Q.Key(T["ip4_1"].family, T["ip4_1"].idx),
+                     Q.Key(T["server_port"].family, T["server_port"].idx)]
+        plan.aggs = [Q.Agg(Q.AGGOP_COUNT),
+                     Q.Agg(Q.AGGOP_SUM, dur.family, dur.idx),
+                     Q.Agg(Q.AGGOP_MAX, dur.family, dur.idx),
+                     Q.Agg(Q.AGGOP_SUM, blen.family, blen.idx)]
+        groups = execute(plan, segs, self.device)
+        base = plan.time_base_s
+        rows = []
+        import ipaddress
+        for g in groups:
+            t, ip0, ip1, port = g["key"]
+            rows.append({
+                "time": base + t,
+                "ip_0": str(ipaddress.IPv4Address(ip0 & 0xFFFFFFFF)),
+                "ip_1": str(ipaddress.IPv4Address(ip1 & 0xFFFFFFFF)),
+                "server_port": port,
+                "request": g["agg"][0],
+                "rrt_sum": g["agg"][1],
+                "rrt_max": g["agg"][2],
+                "byte": g["agg"][3],
+            })
+        return rows
 
     # ----------------------------------------------------------- show
     def _show(self, sql: str) -> Dict:

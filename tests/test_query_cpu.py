@@ -199,3 +199,13 @@ def test_slimit(eng):
         "GROUP BY request_domain ORDER BY c DESC")
     top2 = {row[0] for row in full["values"][:2]}
     assert domains == top2
+
+
+def test_application_map_table(eng):
+    r = eng.query(
+        "SELECT ip_1, Sum(request) AS req FROM application_map "
+        "GROUP BY ip_1 ORDER BY req DESC LIMIT 5")
+    assert sum(row[1] for row in r["values"]) <= N
+    assert len(r["values"]) >= 1
+    full = eng.query("SELECT Sum(request) AS r FROM application_map")
+    assert full["values"][0][0] == N
