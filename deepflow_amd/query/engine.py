@@ -100,6 +100,7 @@ class QueryEngine:
             "event": "event_rows", "perf_event": "perf_event_rows",
             "alert_event": "alert_event_rows",
             "application_log": "app_log_rows", "log": "app_log_rows",
+            "trace_tree": "trace_tree_rows",
         }
         if table in row_tables:
             rows = getattr(self, row_tables[table], lambda: [])()

@@ -14,6 +14,9 @@ from typing import Dict, List, Optional
 class DistributedTracer:
     def __init__(self, engine):
         self.engine = engine
+        # assembled-tree summaries (reference: libs/tracetree rows written
+        # back by the ingester; ours caches on assembly)
+        self.tree_rows = []
 
     def fetch_spans(self, trace_id: str) -> List[Dict]:
         r = self.engine.query(
@@ -76,8 +79,22 @@ class DistributedTracer:
                 nodes[n["parent_index"]]["children"].append(n["index"])
             else:
                 roots.append(n["index"])
-        return {"trace_id": trace_id, "spans": nodes, "roots": roots,
-                "span_count": len(nodes)}
+        result = {"trace_id": trace_id, "spans": nodes, "roots": roots,
+                  "span_count": len(nodes)}
+        if nodes:
+            def depth(i, d=0):
+                n = nodes[i]
+                return max([d] + [depth(c, d + 1) for c in n["children"]])
+            root = nodes[roots[0]] if roots else nodes[0]
+            self.tree_rows.append({
+                "time": (root["start_time"] or 0) // 10**9,
+                "trace_id": trace_id,
+                "span_count": len(nodes),
+                "max_depth": max(depth(r) for r in roots) if roots else 0,
+                "root_service": root["service"] or "",
+                "duration_us": root["duration_ns"] // 1000,
+            })
+        return result
 
     def register(self, app) -> None:
         @app.get("/v1/tracing/{trace_id}")

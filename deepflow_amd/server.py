@@ -110,6 +110,7 @@ class DeepflowServer:
         self.engine.perf_event_rows = lambda: self.events.perf_events
         self.engine.alert_event_rows = lambda: self.events.alert_events
         self.engine.app_log_rows = lambda: self.applogs.rows
+        self.engine.trace_tree_rows = lambda: self.tracer.tree_rows
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
                              promql=self.promql,

@@ -90,3 +90,14 @@ def test_trace_tree(server):
     # leaf linked via syscall_trace_id join (no parent_span_id on wire)
     assert body["spans"][nodes["s-leaf"]["parent_index"]]["span_id"] == "s-mid"
     assert len(body["roots"]) == 1
+
+
+def test_trace_tree_table(server):
+    from fastapi.testclient import TestClient
+    client = TestClient(server.app)
+    client.get(f"/v1/tracing/{TRACE}")
+    r = client.post("/v1/query/", json={
+        "sql": "SELECT trace_id, span_count, max_depth FROM trace_tree "
+               "LIMIT 10"})
+    vals = r.json()["result"]["values"]
+    assert [TRACE, 3, 2] in vals
