@@ -27,6 +27,52 @@ namespace {
 constexpr uint64_t EMPTY_KEY = 0ull;       // hash tables reserve 0 = empty
 constexpr uint32_t BLOCK = 256;
 
+// Word-buffered byte stream: protobuf parsing is byte-granular and
+// sequential per record; buffering an aligned u64 per 8 bytes turns 8
+// byte-loads into one dwordx2 load (the payload tensor base is
+// 256B-aligned, so aligned word indexing is safe).
+struct ByteStream {
+    const uint64_t* w;
+    uint64_t cur;
+    uint32_t wpos;
+    DEV void init(const uint8_t* base) {
+        w = (const uint64_t*)base;
+        wpos = 0xFFFFFFFFu;
+        cur = 0;
+    }
+    DEV uint8_t get(uint32_t i) {
+        uint32_t wp = i >> 3;
+        if (wp != wpos) {
+            wpos = wp;
+            cur = w[wp];
+        }
+        return (uint8_t)(cur >> ((i & 7) * 8));
+    }
+};
+
+DEV uint64_t rd_varint(ByteStream& bs, uint32_t& pos, uint32_t end) {
+    uint64_t v = 0;
+    int sh = 0;
+    while (pos < end) {
+        uint8_t b = bs.get(pos++);
+        v |= (uint64_t)(b & 0x7F) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+        if (sh >= 70) break;
+    }
+    return v;
+}
+
+DEV void skip_field(ByteStream& bs, uint32_t& pos, uint32_t end, uint32_t wt) {
+    switch (wt) {
+        case 0: rd_varint(bs, pos, end); break;
+        case 1: pos += 8; break;
+        case 2: { uint64_t ln = rd_varint(bs, pos, end); pos += (uint32_t)ln; } break;
+        case 5: pos += 4; break;
+        default: pos = end; break;  // malformed
+    }
+}
+
 DEV uint64_t rd_varint(const uint8_t* p, uint32_t& pos, uint32_t end) {
     uint64_t v = 0;
     int sh = 0;
@@ -106,14 +152,16 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
     if (rid >= n) return;
     uint64_t row = cols.base_row + rid;
     uint32_t pos = offs[rid];
+    ByteStream bs;
+    bs.init(payload);
     uint32_t end = pos + lens[rid];
     uint32_t n_names = 0, n_vals = 0;
 
     while (pos < end) {
-        uint64_t key = rd_varint(payload, pos, end);
+        uint64_t key = rd_varint(bs, pos, end);
         uint32_t num = (uint32_t)(key >> 3), wt = (uint32_t)(key & 7);
         if (wt == 0) {
-            uint64_t v = rd_varint(payload, pos, end);
+            uint64_t v = rd_varint(bs, pos, end);
             switch (num) {
                 case 9: W32(L7_U32_REQ_LEN, v); break;
                 case 10: W32(L7_U32_RESP_LEN, v); break;
@@ -124,17 +172,17 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                 default: break;
             }
         } else if (wt == 2) {
-            uint32_t ln = (uint32_t)rd_varint(payload, pos, end);
+            uint32_t ln = (uint32_t)rd_varint(bs, pos, end);
             uint32_t sub = pos, send = pos + ln;
             pos = send;
             switch (num) {
                 case 1: {  // AppProtoLogsBaseInfo
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 0) {
-                            uint64_t v = rd_varint(payload, p2, send);
+                            uint64_t v = rd_varint(bs, p2, send);
                             switch (n2) {
                                 case 1: W64(L7_U64_START_TIME, v); break;
                                 case 2: W64(L7_U64_END_TIME, v); break;
@@ -164,15 +212,15 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                                 default: break;
                             }
                         } else if (w2 == 2) {
-                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
                             uint32_t s3 = p2, e3 = p2 + l3;
                             p2 = e3;
                             if (n2 == 9) {  // AppProtoHead
                                 uint32_t p3 = s3;
                                 while (p3 < e3) {
-                                    uint64_t k3 = rd_varint(payload, p3, e3);
+                                    uint64_t k3 = rd_varint(bs, p3, e3);
                                     if ((k3 & 7) == 0) {
-                                        uint64_t v = rd_varint(payload, p3, e3);
+                                        uint64_t v = rd_varint(bs, p3, e3);
                                         switch ((uint32_t)(k3 >> 3)) {
                                             case 1: W8(L7_U8_L7_PROTOCOL, v); break;
                                             case 2: W8(L7_U8_MSG_TYPE, v); break;
@@ -181,7 +229,7 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                                         }
                                     } else {
                                         uint32_t w3 = (uint32_t)(k3 & 7);
-                                        skip_field(payload, p3, e3, w3);
+                                        skip_field(bs, p3, e3, w3);
                                     }
                                 }
                             } else if (n2 == 27) {
@@ -191,7 +239,7 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                             }
                             // ip6_src/dst (14/15) skipped: ipv4 hot path v1
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -199,10 +247,10 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                 case 11: {  // L7Request
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 2) {
-                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
                             switch (n2) {
                                 case 1: WSTR(L7_STR_REQ_TYPE, p2, l3); break;
                                 case 2: WSTR(L7_STR_DOMAIN, p2, l3); break;
@@ -212,7 +260,7 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                             }
                             p2 += l3;
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -220,19 +268,19 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                 case 12: {  // L7Response
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 0) {
-                            uint64_t v = rd_varint(payload, p2, send);
+                            uint64_t v = rd_varint(bs, p2, send);
                             if (n2 == 1) W8(L7_U8_STATUS, v);
                             else if (n2 == 2) W32(L7_U32_CODE, v);
                         } else if (w2 == 2) {
-                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
                             if (n2 == 3) WSTR(L7_STR_EXCEPTION, p2, l3);
                             else if (n2 == 4) WSTR(L7_STR_RESULT, p2, l3);
                             p2 += l3;
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -241,16 +289,16 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                 case 14: {  // TraceInfo
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 2) {
-                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
                             if (n2 == 1) WSTR(L7_STR_TRACE_ID, p2, l3);
                             else if (n2 == 2) WSTR(L7_STR_SPAN_ID, p2, l3);
                             else if (n2 == 3) WSTR(L7_STR_PARENT_SPAN_ID, p2, l3);
                             p2 += l3;
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -258,13 +306,13 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                 case 15: {  // ExtendedInfo
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 0) {
-                            uint64_t v = rd_varint(payload, p2, send);
+                            uint64_t v = rd_varint(bs, p2, send);
                             if (n2 == 3) W32(L7_U32_REQUEST_ID, v);
                         } else if (w2 == 2) {
-                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
                             switch (n2) {
                                 case 1: WSTR(L7_STR_SERVICE_NAME, p2, l3); break;
                                 case 4: WSTR(L7_STR_XREQ_0, p2, l3); break;
@@ -287,7 +335,7 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                             }
                             p2 += l3;
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -296,7 +344,7 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                 default: break;
             }
         } else {
-            skip_field(payload, pos, end, wt);
+            skip_field(bs, pos, end, wt);
         }
     }
     uint32_t na = n_names < n_vals ? n_names : n_vals;
@@ -331,23 +379,25 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
     if (rid >= n) return;
     uint64_t row = cols.base_row + rid;
     uint32_t pos = offs[rid];
+    ByteStream bs;
+    bs.init(payload);
     uint32_t end = pos + lens[rid];
     // locate flow submessage (TaggedFlow field 1)
     while (pos < end) {
-        uint64_t key = rd_varint(payload, pos, end);
+        uint64_t key = rd_varint(bs, pos, end);
         uint32_t num = (uint32_t)(key >> 3), wt = (uint32_t)(key & 7);
         if (num == 1 && wt == 2) {
-            uint32_t ln = (uint32_t)rd_varint(payload, pos, end);
+            uint32_t ln = (uint32_t)rd_varint(bs, pos, end);
             end = pos + ln;  // narrow to Flow
             break;
         }
-        skip_field(payload, pos, end, wt);
+        skip_field(bs, pos, end, wt);
     }
     while (pos < end) {
-        uint64_t key = rd_varint(payload, pos, end);
+        uint64_t key = rd_varint(bs, pos, end);
         uint32_t num = (uint32_t)(key >> 3), wt = (uint32_t)(key & 7);
         if (wt == 0) {
-            uint64_t v = rd_varint(payload, pos, end);
+            uint64_t v = rd_varint(bs, pos, end);
             switch (num) {
                 case 5: L4W64(L4_U64_FLOW_ID, v); break;
                 case 6: L4W64(L4_U64_START_TIME, v); break;
@@ -364,17 +414,17 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                 default: break;
             }
         } else if (wt == 2) {
-            uint32_t ln = (uint32_t)rd_varint(payload, pos, end);
+            uint32_t ln = (uint32_t)rd_varint(bs, pos, end);
             uint32_t sub = pos, send = pos + ln;
             pos = send;
             switch (num) {
                 case 1: {  // FlowKey
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 0) {
-                            uint64_t v = rd_varint(payload, p2, send);
+                            uint64_t v = rd_varint(bs, p2, send);
                             switch (n2) {
                                 case 1: L4W32(L4_U32_VTAP_ID, v); break;
                                 case 2: L4W8(L4_U8_TAP_TYPE, v); break;
@@ -388,7 +438,7 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                                 default: break;
                             }
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -397,10 +447,10 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                     bool tx = num == 2;
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 0) {
-                            uint64_t v = rd_varint(payload, p2, send);
+                            uint64_t v = rd_varint(bs, p2, send);
                             switch (n2) {
                                 case 1: L4W64(tx ? L4_U64_BYTE_TX : L4_U64_BYTE_RX, v); break;
                                 case 2: L4W64(tx ? L4_U64_L3_BYTE_TX : L4_U64_L3_BYTE_RX, v); break;
@@ -416,7 +466,7 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                                 default: break;
                             }
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -424,23 +474,23 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                 case 13: {  // FlowPerfStats
                     uint32_t p2 = sub;
                     while (p2 < send) {
-                        uint64_t k2 = rd_varint(payload, p2, send);
+                        uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 0) {
-                            uint64_t v = rd_varint(payload, p2, send);
+                            uint64_t v = rd_varint(bs, p2, send);
                             if (n2 == 3) L4W8(L4_U8_L4_PROTOCOL, v);
                             else if (n2 == 4) L4W8(L4_U8_L7_PROTOCOL, v);
                         } else if (w2 == 2) {
-                            uint32_t l3 = (uint32_t)rd_varint(payload, p2, send);
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
                             uint32_t s3 = p2, e3 = p2 + l3;
                             p2 = e3;
                             if (n2 == 1) {  // TCPPerfStats
                                 uint32_t p3 = s3;
                                 while (p3 < e3) {
-                                    uint64_t k3 = rd_varint(payload, p3, e3);
+                                    uint64_t k3 = rd_varint(bs, p3, e3);
                                     uint32_t n3 = (uint32_t)(k3 >> 3), w3 = (uint32_t)(k3 & 7);
                                     if (w3 == 0) {
-                                        uint64_t v = rd_varint(payload, p3, e3);
+                                        uint64_t v = rd_varint(bs, p3, e3);
                                         switch (n3) {
                                             case 3: L4W32(L4_U32_SRT_MAX, v); break;
                                             case 4: L4W32(L4_U32_ART_MAX, v); break;
@@ -458,36 +508,36 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                                             default: break;
                                         }
                                     } else if (w3 == 2) {
-                                        uint32_t l4b = (uint32_t)rd_varint(payload, p3, e3);
+                                        uint32_t l4b = (uint32_t)rd_varint(bs, p3, e3);
                                         uint32_t s4 = p3, e4 = p3 + l4b;
                                         p3 = e4;
                                         if (n3 == 14 || n3 == 15) {  // TcpPerfCountsPeer
                                             bool ptx = n3 == 14;
                                             uint32_t p4 = s4;
                                             while (p4 < e4) {
-                                                uint64_t k4 = rd_varint(payload, p4, e4);
+                                                uint64_t k4 = rd_varint(bs, p4, e4);
                                                 if ((k4 & 7) == 0) {
-                                                    uint64_t v = rd_varint(payload, p4, e4);
+                                                    uint64_t v = rd_varint(bs, p4, e4);
                                                     uint32_t n4 = (uint32_t)(k4 >> 3);
                                                     if (n4 == 1) L4W32(ptx ? L4_U32_RETRANS_TX : L4_U32_RETRANS_RX, v);
                                                     else if (n4 == 2) L4W32(ptx ? L4_U32_ZERO_WIN_TX : L4_U32_ZERO_WIN_RX, v);
                                                     else if (n4 == 3) L4W32(ptx ? L4_U32_OOO_TX : L4_U32_OOO_RX, v);
                                                 } else {
                                                     uint32_t w4 = (uint32_t)(k4 & 7);
-                                                    skip_field(payload, p4, e4, w4);
+                                                    skip_field(bs, p4, e4, w4);
                                                 }
                                             }
                                         }
                                     } else {
-                                        skip_field(payload, p3, e3, w3);
+                                        skip_field(bs, p3, e3, w3);
                                     }
                                 }
                             } else if (n2 == 2) {  // L7PerfStats
                                 uint32_t p3 = s3;
                                 while (p3 < e3) {
-                                    uint64_t k3 = rd_varint(payload, p3, e3);
+                                    uint64_t k3 = rd_varint(bs, p3, e3);
                                     if ((k3 & 7) == 0) {
-                                        uint64_t v = rd_varint(payload, p3, e3);
+                                        uint64_t v = rd_varint(bs, p3, e3);
                                         switch ((uint32_t)(k3 >> 3)) {
                                             case 1: L4W32(L4_U32_L7_REQUEST, v); break;
                                             case 2: L4W32(L4_U32_L7_RESPONSE, v); break;
@@ -501,12 +551,12 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                                         }
                                     } else {
                                         uint32_t w3 = (uint32_t)(k3 & 7);
-                                        skip_field(payload, p3, e3, w3);
+                                        skip_field(bs, p3, e3, w3);
                                     }
                                 }
                             }
                         } else {
-                            skip_field(payload, p2, send, w2);
+                            skip_field(bs, p2, send, w2);
                         }
                     }
                     break;
@@ -518,7 +568,7 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                 default: break;
             }
         } else {
-            skip_field(payload, pos, end, wt);
+            skip_field(bs, pos, end, wt);
         }
     }
 }
