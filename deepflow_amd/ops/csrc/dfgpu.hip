@@ -933,6 +933,8 @@ __global__ void k_agg_app1s(const L7Cols cols, uint32_t n, uint64_t time_base_s,
 enum {
     SRC_U64 = 0, SRC_U32, SRC_U8, SRC_DID, SRC_KG, SRC_ATTR_VAL,
     SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
+    // filter-only: exists(slot): name_id == v0 && value_id == v1
+    SRC_ATTR_MATCH,
 };
 // seed for SRC_STR_HASH terms (host twin: store/dictionary.py STR_FILTER_SEED)
 #define STR_FILTER_SEED 0x5157A15E5EEDull
@@ -1014,6 +1016,24 @@ DEV bool eval_terms(const SegView& s, uint64_t row, const QuerySpec& q) {
     uint32_t need = 0, have = 0;
     for (uint32_t t = 0; t < q.n_terms; t++) {
         const QTerm& term = q.terms[t];
+        if (term.family == SRC_ATTR_MATCH) {
+            uint32_t cnt = s.attr_cnt[row];
+            uint32_t start = s.attr_start[row];
+            bool okm = false;
+            for (uint32_t a = 0; a < cnt && !okm; a++)
+                okm = (uint32_t)s.attr_pool[start + a] == (uint32_t)term.v0 &&
+                      (uint32_t)s.attr_pool[start + cnt + a] ==
+                          (uint32_t)term.v1;
+            if (term.op == OP_NE) okm = !okm;
+            if (term.group == 0) {
+                if (!okm) return false;
+            } else {
+                uint32_t bit = 1u << (term.group & 31);
+                need |= bit;
+                if (okm) have |= bit;
+            }
+            continue;
+        }
         uint64_t v = src_value(s, row, term.family, term.idx, 0, q.time_base_s);
         bool ok;
         switch (term.op) {

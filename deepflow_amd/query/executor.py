@@ -15,7 +15,7 @@ import torch
 from ..store import l7_schema as S
 from .spec import (Plan, SRC_U64, SRC_U32, SRC_U8, SRC_DID, SRC_KG,
                    SRC_ATTR_VAL, SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
-                   STR_FILTER_SEED,
+                   SRC_ATTR_MATCH, STR_FILTER_SEED,
                    OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN,
                    AGGOP_COUNT, AGGOP_SUM, AGGOP_MIN, AGGOP_MAX,
                    QMAX_KEYS, QMAX_AGGS)
@@ -70,6 +70,22 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
 
 
 def _term_mask(seg, t, plan: Plan, n: int) -> np.ndarray:
+    if t.family == SRC_ATTR_MATCH:
+        starts = seg.attr_start[:n].numpy()
+        cnts = seg.attr_cnt[:n].numpy()
+        pool = seg.attr_pool.numpy()
+        out = np.zeros(n, dtype=bool)
+        for i in range(n):
+            c = int(cnts[i])
+            s0 = int(starts[i])
+            for a in range(c):
+                if (pool[s0 + a] & 0xFFFFFFFF) == t.v0 and \
+                        (pool[s0 + c + a] & 0xFFFFFFFF) == t.v1:
+                    out[i] = True
+                    break
+        if t.op == OP_NE:
+            out = ~out
+        return out
     v = _src_np(seg, t.family, t.idx, 0, plan.time_base_s, n)
     v0 = np.uint64(t.v0 & U64MAX)
     v1 = np.uint64(t.v1 & U64MAX)

@@ -23,6 +23,7 @@ SRC_ATTR_VAL = 5
 SRC_TIME_BUCKET = 6
 SRC_CONST0 = 7
 SRC_STR_HASH = 8
+SRC_ATTR_MATCH = 9  # filter-only: attr name_id == v0 AND value_id == v1
 
 # seed for pooled-string filter hashing (twin: dfgpu.hip STR_FILTER_SEED)
 STR_FILTER_SEED = 0x5157A15E5EED

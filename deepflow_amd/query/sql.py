@@ -326,6 +326,30 @@ def _never(plan: Q.Plan, group: int) -> None:
 
 def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
               tags, group: int = 0) -> None:
+    if name.startswith("attribute."):
+        # custom-tag filter: both sides resolved to SmartEncoding ids
+        # (reference: flow_tag custom_field_value filters)
+        from ..store.l7_schema import DICT_DOM_ATTR_NAME, DICT_DOM_ATTR_VALUE
+        if op not in ("=", "==", "!=", "<>"):
+            raise SqlError("attribute.* supports = / != only")
+        if lit[0] != "str":
+            raise SqlError("attribute.* compares against a string")
+        nid = dictionary.lookup_id(DICT_DOM_ATTR_NAME,
+                                   name[len("attribute."):].encode()) \
+            if dictionary is not None else None
+        vid = dictionary.lookup_id(DICT_DOM_ATTR_VALUE, lit[1].encode()) \
+            if dictionary is not None else None
+        if nid is None or vid is None:
+            if Q.OP_BY_NAME[op] == Q.OP_NE:
+                return
+            if group:
+                _never(plan, group)
+            else:
+                plan.impossible = True
+            return
+        plan.terms.append(Q.Term(Q.SRC_ATTR_MATCH, 0, Q.OP_BY_NAME[op],
+                                 nid, vid, group=group))
+        return
     if name.lower() == "time":
         # time in epoch seconds against start_time (ns)
         v = int(lit[1]) * 10**9

@@ -107,3 +107,15 @@ def test_percentile_apdex_match(engines):
                     assert abs(x - y) < max(1e-6, abs(x) * 0.01), sql
                 else:
                     assert x == y, sql
+
+
+def test_attribute_filter_match(engines):
+    # find a real attr value from the cpu dict
+    from deepflow_amd.store.l7_schema import DICT_DOM_ATTR_VALUE
+    val = next(s for (d, s) in engines["cpu"].pipe.dict.str_to_id
+               if d == DICT_DOM_ATTR_VALUE).decode()
+    sql = (f"SELECT Count(*) AS c FROM l7_flow_log WHERE "
+           f"attribute.attr_0 = '{val}'")
+    rc = engines["cpu"].query(sql)
+    rg = engines["cuda"].query(sql)
+    assert rc["values"] == rg["values"]
