@@ -213,18 +213,18 @@ def test_application_map_table(eng):
 
 def test_attribute_filter(eng):
     t0 = truth()
-    val = t0[0]["ext_info"]["attribute_values"][2]
-    want = sum(1 for t in t0 if t["ext_info"]["attribute_values"][2] == val)
+    val = t0[0]["ext_info"]["attribute_values"][1]
+    want = sum(1 for t in t0 if t["ext_info"]["attribute_values"][1] == val)
     r = eng.query(
         f"SELECT Count(*) AS c FROM l7_flow_log WHERE "
-        f"attribute.attr_2 = '{val}'")
+        f"attribute.attr_1 = '{val}'")
     assert r["values"] == [[want]]
     # unknown value -> empty; != unknown -> everything
     r2 = eng.query(
         "SELECT Count(*) AS c FROM l7_flow_log WHERE "
-        "attribute.attr_2 = 'v9999999'")
+        "attribute.attr_1 = 'v9999999'")
     assert r2["values"] == []
     r3 = eng.query(
         f"SELECT Count(*) AS c FROM l7_flow_log WHERE "
-        f"attribute.attr_2 != '{val}'")
+        f"attribute.attr_1 != '{val}'")
     assert r3["values"] == [[N - want]]
