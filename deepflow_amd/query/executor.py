@@ -79,8 +79,8 @@ def _term_mask(seg, t, plan: Plan, n: int) -> np.ndarray:
             c = int(cnts[i])
             s0 = int(starts[i])
             for a in range(c):
-                if (pool[s0 + a] & 0xFFFFFFFF) == t.v0 and \
-                        (pool[s0 + c + a] & 0xFFFFFFFF) == t.v1:
+                if (int(pool[s0 + a]) & 0xFFFFFFFF) == t.v0 and \
+                        (int(pool[s0 + c + a]) & 0xFFFFFFFF) == t.v1:
                     out[i] = True
                     break
         if t.op == OP_NE:
