@@ -104,11 +104,25 @@ class Agent:
         return dict(zip(keys, (int(x) for x in arr)))
 
     # ---------------------------------------------------------- sender
-    def frame(self, which: int, payload: bytes) -> bytes:
+    def frame(self, which: int, payload: bytes,
+              compress: bool = False) -> bytes:
+        encoder = framing.ENCODER_RAW
+        if compress and len(payload) > 128:
+            payload_z = self._zstd(payload)
+            if payload_z is not None and len(payload_z) < len(payload):
+                payload = payload_z
+                encoder = framing.ENCODER_ZSTD
         hdr = framing.FrameHeader(msg_type=_MSG_FOR[which],
                                   team_id=self.team_id, org_id=self.org_id,
-                                  agent_id=self.agent_id)
+                                  agent_id=self.agent_id, encoder=encoder)
         return framing.encode_frame(hdr, payload)
+
+    def _zstd(self, payload: bytes):
+        src = np.frombuffer(payload, dtype=np.uint8)
+        dst = np.zeros(len(payload) + 1024, dtype=np.uint8)
+        n = self._lib.df_zstd_compress(src.ctypes.data, len(src),
+                                       dst.ctypes.data, len(dst), 3)
+        return dst[:n].tobytes() if n > 0 else None
 
     # ---------------------------------------------------------- guard
     # exception bits (reference agent/src/exception.rs bitmask idea)
@@ -157,16 +171,17 @@ class Agent:
             self.platform_version = resp["platform_version"]
         return resp
 
-    def flush_to_server(self, now_ns: int) -> int:
+    def flush_to_server(self, now_ns: int, compress: bool = False) -> int:
         """tick + drain all types + send framed payloads to the server
-        (uniform-sender analog). Returns frames sent."""
+        (uniform-sender analog; compress=True uses whole-payload zstd like
+        the reference's SenderEncoder::Zstd). Returns frames sent."""
         self.tick(now_ns)
         sent = 0
         for which in (DRAIN_L4, DRAIN_L7, DRAIN_DOC):
             payload = self.drain(which)
             if not payload:
                 continue
-            frame = self.frame(which, payload)
+            frame = self.frame(which, payload, compress=compress)
             if self.server is not None:
                 if self._sock is None:
                     self._sock = socket.create_connection(self.server,

@@ -272,3 +272,21 @@ def test_tls_sni(agent):
     assert len(l7) == 1
     assert l7[0]["base"]["head"]["proto"] == 121
     assert l7[0]["req"]["domain"] == "api.internal.example"
+
+
+def test_compressed_sender():
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 9,
+                         dict_capacity=1 << 10, time_base_s=0)
+    a = Agent(vtap_id=6)
+    for i in range(10):
+        for frame, ts in http_session(CLIENT + i, SERVER, sport=45000 + i):
+            a.packet(frame, ts)
+    a.tick(10**9 * 100)
+    payload = a.drain(1)
+    framed = a.frame(1, payload, compress=True)
+    # encoder byte says zstd and the server decompresses transparently
+    assert framed[7] == framing.ENCODER_ZSTD
+    assert len(framed) < len(payload)
+    assert srv.receiver.handle_frame(framed)
+    assert srv.l7.stats.spans_in == 10
+    a.close()
