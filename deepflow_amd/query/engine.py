@@ -62,7 +62,61 @@ def rollup_rows(rows, bucket_s: int):
 
 
 def _row_tags(fields: List[str]) -> Dict[str, TagDef]:
-    return {f: TagDef(f, SRC_ROW, i) for i, f in enumerate(fields)}
+    return {f: TagDef(f, SRC_ROW, i, hydrate="raw")
+            for i, f in enumerate(fields)}
+
+
+def _match_paren(sql: str, i: int) -> int:
+    """`i` points at '('; return the index of the matching ')'.
+    Skips over single-quoted string literals."""
+    depth = 0
+    in_str = False
+    for j in range(i, len(sql)):
+        c = sql[j]
+        if in_str:
+            if c == "'":
+                in_str = False
+        elif c == "'":
+            in_str = True
+        elif c == "(":
+            depth += 1
+        elif c == ")":
+            depth -= 1
+            if depth == 0:
+                return j
+    raise SqlError("unbalanced parentheses")
+
+
+_STOP_KWS = {"where", "group", "order", "limit", "slimit", "having", "union"}
+
+
+def split_with(sql: str):
+    """'WITH a AS (...), b AS (...) SELECT ...' -> ([(name, inner)], main).
+    Reference counterpart: CHEngine's WITH handling
+    (server/querier/engine/clickhouse — TransWhere/with clauses)."""
+    m = re.match(r"\s*with\s+", sql, re.IGNORECASE)
+    pos = m.end()
+    ctes = []
+    while True:
+        m = re.match(r"\s*`?(\w+)`?\s+as\s*", sql[pos:], re.IGNORECASE)
+        if not m:
+            raise SqlError("WITH: expected `name AS (...)`")
+        name = m.group(1).lower()
+        i = pos + m.end()
+        if i >= len(sql) or sql[i] != "(":
+            raise SqlError("WITH: expected '(' after AS")
+        j = _match_paren(sql, i)
+        ctes.append((name, sql[i + 1:j]))
+        pos = j + 1
+        m = re.match(r"\s*,", sql[pos:])
+        if not m:
+            break
+        pos += m.end()
+    return ctes, sql[pos:]
+
+
+def _result_rows(res: Dict) -> List[Dict]:
+    return [dict(zip(res["columns"], row)) for row in res["values"]]
 
 
 class QueryEngine:
@@ -75,12 +129,33 @@ class QueryEngine:
         self.remote = remote_hydrator
 
     # ----------------------------------------------------------- dispatch
-    def query(self, sql: str) -> Dict:
+    def query(self, sql: str, _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
         stripped = sql.strip().lower()
         if stripped.startswith("show"):
             return self._show(sql)
+        if stripped.startswith("with"):
+            ctes, main = split_with(sql)
+            env = dict(_ctes or {})
+            for name, inner in ctes:
+                env[name] = self.query(inner, _ctes=env)
+            return self.query(main, _ctes=env)
+        md = re.search(r"\bfrom\s*\(", sql, re.IGNORECASE)
+        if md:
+            # derived table: FROM ( SELECT ... ) [AS alias]
+            i = md.end() - 1
+            j = _match_paren(sql, i)
+            res = self.query(sql[i + 1:j], _ctes=_ctes)
+            rest = sql[j + 1:]
+            ma = re.match(r"\s*(?:as\s+)?`?(\w+)`?", rest, re.IGNORECASE)
+            if ma and ma.group(1).lower() not in _STOP_KWS:
+                rest = rest[ma.end():]
+            outer = sql[: md.start()] + " FROM __sub__ " + rest
+            return self._run_rows(outer, _result_rows(res), time_base_s=0)
         m = _FROM_RE.search(sql)
         table = m.group(1).lower() if m else "l7_flow_log"
+        if _ctes and table in _ctes:
+            return self._run_rows(sql, _result_rows(_ctes[table]),
+                                  time_base_s=0)
         if table in ("l7_flow_log", "l7_flow_log.l7_flow_log"):
             plan = parse_sql(sql, dictionary=self.pipe.dict,
                              time_base_s=self.pipe.time_base_s,
@@ -258,8 +333,13 @@ class QueryEngine:
         stripped = sql.strip().lower()
         m = _FROM_RE.search(sql)
         table = m.group(1).lower() if m else "l7_flow_log"
-        if stripped.startswith("show") or table not in (
-                "l7_flow_log", "l4_flow_log"):
+        if stripped.startswith(("show", "with")) or \
+                re.search(r"\bfrom\s*\(", sql, re.IGNORECASE) or \
+                table not in ("l7_flow_log", "l4_flow_log"):
+            # WITH/derived queries run shard-locally and merge as rows:
+            # correct when the inner query is per-row (filters/select);
+            # a cross-shard inner aggregate needs the dist engine's agg
+            # path (round-2: rewrite WITH into partial+finalize).
             return {"kind": "rows", "result": self.query(sql)}
         if table == "l4_flow_log" and self.l4 is None:
             raise SqlError("l4_flow_log table not enabled")
@@ -512,15 +592,30 @@ class QueryEngine:
 
         OPS = {Q.OP_EQ: lambda a, b: a == b, Q.OP_NE: lambda a, b: a != b,
                Q.OP_LT: lambda a, b: a < b, Q.OP_LE: lambda a, b: a <= b,
-               Q.OP_GT: lambda a, b: a > b, Q.OP_GE: lambda a, b: a >= b}
+               Q.OP_GT: lambda a, b: a > b, Q.OP_GE: lambda a, b: a >= b,
+               Q.OP_BETWEEN: lambda a, b: True}
+
+        def term_ok(r, t):
+            if t.family == Q.SRC_CONST0:
+                return OPS[t.op](0, t.v0)
+            v = val(r, t.family, t.idx)
+            if t.op == Q.OP_BETWEEN:
+                return t.v0 <= v <= t.v1
+            try:
+                return OPS[t.op](v, t.v0)
+            except TypeError:
+                return False
+        # CNF: group 0 terms are ANDed; each group>=1 is an OR clause
         filtered = []
         for r in rows:
-            ok = True
-            for t in plan.terms:
-                v = val(r, t.family, t.idx)
-                if not OPS[t.op](v, t.v0):
-                    ok = False
-                    break
+            ok = all(term_ok(r, t) for t in plan.terms if t.group == 0)
+            if ok:
+                groups_seen: Dict[int, bool] = {}
+                for t in plan.terms:
+                    if t.group:
+                        groups_seen[t.group] = groups_seen.get(
+                            t.group, False) or term_ok(r, t)
+                ok = all(groups_seen.values())
             if ok:
                 filtered.append(r)
         if plan.select_rows:

@@ -363,6 +363,11 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
                                  group=group))
         return
     # string literal
+    if td.hydrate == "raw":
+        # row-table / CTE column: compare the Python value directly
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op],
+                                 lit[1], group=group))
+        return
     if td.hydrate == "strhash":
         from ..store.dictionary import str_hash_py
         v = str_hash_py(lit[1].encode(), Q.STR_FILTER_SEED)
