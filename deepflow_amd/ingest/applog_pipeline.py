@@ -61,6 +61,15 @@ class AppLogPipeline:
         self.counter.add("logs_in", n)
         return n
 
+    def ingest_rows(self, rows: List[Dict]) -> int:
+        """Append pre-converted rows (OTLP logs path)."""
+        for r in rows:
+            app = r.get("app_service", "")
+            r.setdefault("app_id", self._intern(app))
+            self.rows.append(r)
+        self.counter.add("logs_in", len(rows))
+        return len(rows)
+
     def search(self, substr: str = "", severity_max: int = 7,
                limit: int = 100) -> List[Dict]:
         out = []

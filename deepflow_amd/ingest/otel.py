@@ -120,3 +120,87 @@ def otlp_to_l7_payload(data: bytes, compressed: bool = False) -> bytes:
                 rec = span_to_l7(span, res_attrs)
                 records.append(pb.encode(rec, flow_log.APP_PROTO_LOGS_DATA))
     return framing.pack_records(records)
+
+
+def otlp_logs_to_rows(data: bytes, agent_id: int = 0) -> List[Dict]:
+    """OTLP LogsData bytes -> application_log rows (reference:
+    server/ingester/app_log otel log import)."""
+    ld = pb.decode(data, otlp.LOGS_DATA)
+    rows: List[Dict] = []
+    for rl in ld.get("resource_logs", []):
+        res_attrs = {kv.get("key", ""): _attr_val(kv.get("value", {}))
+                     for kv in rl.get("resource", {}).get("attributes", [])}
+        svc = res_attrs.get("service.name", "")
+        for sl in rl.get("scope_logs", []):
+            for rec in sl.get("log_records", []):
+                ts = rec.get("time_unix_nano", 0) or \
+                    rec.get("observed_time_unix_nano", 0)
+                attrs = {kv.get("key", ""): _attr_val(kv.get("value", {}))
+                         for kv in rec.get("attributes", [])}
+                rows.append({
+                    "time": ts // 10**9,
+                    "agent_id": agent_id,
+                    "log_type": "otlp",
+                    # OTLP severity bands (1-4 trace .. 21-24 fatal)
+                    # -> syslog levels
+                    "severity": {0: 7, 1: 7, 2: 6, 3: 4, 4: 3, 5: 2}[
+                        min(max(rec.get("severity_number", 9) - 1, 0) // 4,
+                            5)],
+                    "severity_text": rec.get("severity_text", ""),
+                    "app_service": svc,
+                    "trace_id": rec.get("trace_id", b"").hex(),
+                    "span_id": rec.get("span_id", b"").hex(),
+                    "body": _attr_val(rec.get("body", {})),
+                    **{f"attr.{k}": v for k, v in attrs.items()},
+                })
+    return rows
+
+
+def _dp_labels(res_attrs: Dict[str, str], dp: Dict) -> Dict[str, str]:
+    labels = dict(res_attrs)
+    for kv in dp.get("attributes", []):
+        labels[kv.get("key", "")] = _attr_val(kv.get("value", {}))
+    return labels
+
+
+def otlp_metrics_to_samples(data: bytes) -> List[tuple]:
+    """OTLP MetricsData bytes -> [(metric_name, labels, ts_ms, value)].
+
+    Gauge/Sum map 1:1; Histogram expands to the Prometheus convention
+    (<name>_count, <name>_sum, <name>_bucket{le=...}) so PromQL sees the
+    same series a prometheus remote-write would produce."""
+    md = pb.decode(data, otlp.METRICS_DATA)
+    out: List[tuple] = []
+    for rm in md.get("resource_metrics", []):
+        res_attrs = {kv.get("key", ""): _attr_val(kv.get("value", {}))
+                     for kv in rm.get("resource", {}).get("attributes", [])}
+        res_attrs = {k: v for k, v in res_attrs.items()
+                     if k in ("service.name", "host.name")}
+        for sm in rm.get("scope_metrics", []):
+            for m in sm.get("metrics", []):
+                name = m.get("name", "")
+                for dp in m.get("gauge", {}).get("data_points", []) + \
+                        m.get("sum", {}).get("data_points", []):
+                    v = dp.get("as_double", 0.0) or float(
+                        _sfixed64(dp.get("as_int", 0)))
+                    out.append((name, _dp_labels(res_attrs, dp),
+                                dp.get("time_unix_nano", 0) // 10**6, v))
+                for dp in m.get("histogram", {}).get("data_points", []):
+                    ts = dp.get("time_unix_nano", 0) // 10**6
+                    labels = _dp_labels(res_attrs, dp)
+                    out.append((name + "_count", labels, ts,
+                                float(dp.get("count", 0))))
+                    out.append((name + "_sum", labels, ts,
+                                float(dp.get("sum", 0.0))))
+                    bounds = dp.get("explicit_bounds", [])
+                    cum = 0
+                    for bi, c in enumerate(dp.get("bucket_counts", [])):
+                        cum += c
+                        le = str(bounds[bi]) if bi < len(bounds) else "+Inf"
+                        out.append((name + "_bucket",
+                                    {**labels, "le": le}, ts, float(cum)))
+    return out
+
+
+def _sfixed64(v: int) -> int:
+    return v - (1 << 64) if v >= (1 << 63) else v

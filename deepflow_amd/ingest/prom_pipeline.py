@@ -47,6 +47,34 @@ class PromPipeline:
         self.s_value: List[float] = []
         self.counter = counter or Counter("ingester.prometheus")
 
+    def _series_id(self, metric: str, lab_ids) -> int:
+        mid = self.metric_names.intern(metric)
+        key = (mid, tuple(sorted(lab_ids)))
+        sid = self.series.get(key)
+        if sid is None:
+            sid = len(self.series_labels)
+            self.series[key] = sid
+            self.series_labels.append(key[1])
+            self.series_metric.append(mid)
+        return sid
+
+    def ingest_labeled_samples(self, samples) -> int:
+        """[(metric, labels_dict, ts_ms, value)] -> ID-encoded store.
+        Entry point for OTLP metrics / telegraf (same SmartEncoding
+        discipline as remote-write)."""
+        n = 0
+        for metric, labels, ts, value in samples:
+            lab_ids = [(self.label_names.intern(k),
+                        self.label_values.intern(str(v)))
+                       for k, v in labels.items()]
+            sid = self._series_id(metric, lab_ids)
+            self.s_series.append(sid)
+            self.s_ts.append(int(ts))
+            self.s_value.append(float(value))
+            n += 1
+        self.counter.add("samples_in", n)
+        return n
+
     def ingest_write_request(self, data: bytes) -> int:
         wr = pb.decode(data, prompb.WRITE_REQUEST)
         n = 0
@@ -61,14 +89,7 @@ class PromPipeline:
                     continue
                 lab_ids.append((self.label_names.intern(name),
                                 self.label_values.intern(value)))
-            mid = self.metric_names.intern(metric)
-            key = (mid, tuple(sorted(lab_ids)))
-            sid = self.series.get(key)
-            if sid is None:
-                sid = len(self.series_labels)
-                self.series[key] = sid
-                self.series_labels.append(key[1])
-                self.series_metric.append(mid)
+            sid = self._series_id(metric, lab_ids)
             for sm in ts.get("samples", []):
                 self.s_series.append(sid)
                 self.s_ts.append(int(sm.get("timestamp", 0)))
@@ -110,7 +131,10 @@ class PromPipeline:
                 if s == sid:
                     samples[self.s_ts[i] // 1000] = self.s_value[i]
             out.append({"metric": dict(labels, __name__=metric),
-                        "samples": samples})
+                        "samples": samples,
+                        # raw scraped series are cumulative counters/gauges,
+                        # unlike the engine's per-second rollup deltas
+                        "kind": "counter"})
         return out
 
     def stored_bytes(self) -> int:

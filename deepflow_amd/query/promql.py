@@ -180,11 +180,26 @@ class PromQLEngine:
                                   _parse_matchers(sel.group("matchers")))
             out = []
             for s in series:
-                window = [v for ts, v in s["samples"].items()
-                          if t - w < ts <= t]
+                pts = sorted((ts, v) for ts, v in s["samples"].items()
+                             if t - w < ts <= t)
+                window = [v for _, v in pts]
                 if not window:
                     continue
-                if fn in ("rate", "irate"):
+                if s.get("kind") == "counter" and fn in (
+                        "rate", "irate", "increase"):
+                    # cumulative counter semantics: last-first over the
+                    # window (reset -> restart from the last value)
+                    if len(pts) < 2:
+                        continue
+                    if fn == "irate":
+                        (t0, v0), (t1, v1) = pts[-2], pts[-1]
+                    else:
+                        (t0, v0), (t1, v1) = pts[0], pts[-1]
+                    delta = v1 - v0 if v1 >= v0 else v1
+                    span = max(t1 - t0, 1)
+                    v = delta / span if fn in ("rate", "irate") \
+                        else float(delta)
+                elif fn in ("rate", "irate"):
                     v = sum(window) / w
                 elif fn in ("increase", "sum_over_time"):
                     v = float(sum(window))

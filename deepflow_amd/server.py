@@ -186,6 +186,46 @@ class DeepflowServer:
             return Response(content=blob,
                             media_type="application/x-protobuf")
 
+        # OTLP/HTTP collector endpoints (standard /v1/{traces,logs,metrics}
+        # paths, protobuf bodies; reference accepts the same three
+        # signals through its otel integration port)
+        from fastapi import Request
+
+        @self.app.post("/otlp/v1/traces")
+        async def otlp_traces(request: Request):
+            import numpy as np
+            from .ingest.otel import otlp_to_l7_payload
+            body = await request.body()
+            try:
+                l7_payload = otlp_to_l7_payload(body, compressed=False)
+            except Exception:
+                return {"error": "bad TracesData"}
+            self._on_l7(framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
+                        np.frombuffer(l7_payload, dtype=np.uint8))
+            return {}
+
+        @self.app.post("/otlp/v1/logs")
+        async def otlp_logs(request: Request):
+            from .ingest.otel import otlp_logs_to_rows
+            body = await request.body()
+            try:
+                rows = otlp_logs_to_rows(body)
+            except Exception:
+                return {"error": "bad LogsData"}
+            self.applogs.ingest_rows(rows)
+            return {"accepted": len(rows)}
+
+        @self.app.post("/otlp/v1/metrics")
+        async def otlp_metrics(request: Request):
+            from .ingest.otel import otlp_metrics_to_samples
+            body = await request.body()
+            try:
+                samples = otlp_metrics_to_samples(body)
+            except Exception:
+                return {"error": "bad MetricsData"}
+            n = self.prom.ingest_labeled_samples(samples)
+            return {"accepted": n}
+
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
