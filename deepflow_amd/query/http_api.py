@@ -14,7 +14,7 @@ from fastapi.responses import JSONResponse
 
 
 def build_app(engine, registry=None, tempo=None, promql=None,
-              profile=None, tracing=None) -> FastAPI:
+              profile=None, tracing=None, engine_for=None) -> FastAPI:
     app = FastAPI(title="deepflow-amd querier")
 
     @app.get("/v1/health")
@@ -24,6 +24,14 @@ def build_app(engine, registry=None, tempo=None, promql=None,
     @app.post("/v1/query/")
     async def query(request: Request):
         ctype = request.headers.get("content-type", "")
+        # org isolation: X-Org-Id header routes to that org's engine
+        eng = engine
+        org_hdr = request.headers.get("x-org-id")
+        if org_hdr and engine_for is not None:
+            try:
+                eng = engine_for(int(org_hdr))
+            except ValueError:
+                pass
         sql: Optional[str] = None
         db = "flow_log"
         if "json" in ctype:
@@ -41,7 +49,7 @@ def build_app(engine, registry=None, tempo=None, promql=None,
             return JSONResponse({"OPT_STATUS": "INVALID_PARAMETERS",
                                  "DESCRIPTION": "missing sql"}, status_code=400)
         try:
-            result = engine.query(sql)
+            result = eng.query(sql)
         except Exception as e:  # noqa: BLE001
             return JSONResponse({"OPT_STATUS": "FAILED",
                                  "DESCRIPTION": str(e)}, status_code=400)

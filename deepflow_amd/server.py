@@ -42,6 +42,12 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_PROTOCOLLOG, self._on_l7)
         self.receiver.register(framing.MSG_TAGGEDFLOW, self._on_l4)
         self.engine = QueryEngine(self.l7, device=device, l4_pipeline=self.l4)
+        # multi-org isolation: each non-default org gets its own KG,
+        # dictionary, segment sets and engine (reference: per-org
+        # ClickHouse databases, org_id from the frame header / ORG_ID
+        # HTTP header). Aux pipelines (logs/events/profiles) stay global.
+        self.default_org = 1
+        self._org_ctx: dict = {}
         from .query.tempo import TempoApp
         from .query.tracing import DistributedTracer
         self.tempo = TempoApp(self.engine)
@@ -114,7 +120,8 @@ class DeepflowServer:
         self.app = build_app(self.engine, registry=default_registry(),
                              tempo=self.tempo, tracing=self.tracer,
                              promql=self.promql,
-                             profile=ProfileApp(self.profiles))
+                             profile=ProfileApp(self.profiles),
+                             engine_for=self.engine_for_org)
         self.controller.register(self.app)
         from .export import OtlpExporter
         self.exporter = OtlpExporter(self.engine)
@@ -229,7 +236,33 @@ class DeepflowServer:
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
-    def _on_l7(self, hdr, payload) -> None:
+    def org_context(self, org_id: int):
+        """Lazily-created isolated (l7, l4, engine) triple for a
+        non-default org."""
+        if org_id in (0, self.default_org):
+            return self
+        ctx = self._org_ctx.get(org_id)
+        if ctx is None:
+            from types import SimpleNamespace
+            from .ingest.l4_pipeline import L4IngestPipeline
+            kg = KnowledgeGraphTable(device=self.device)
+            l7 = L7IngestPipeline(device=self.device,
+                                  segment_rows=self.l7.segments.segment_rows,
+                                  kg=kg, dict_capacity=self.l7.dict.capacity,
+                                  time_base_s=self.l7.time_base_s)
+            l4 = L4IngestPipeline(device=self.device,
+                                  segment_rows=self.l4.segments.segment_rows,
+                                  kg=kg, time_base_s=self.l4.time_base_s)
+            eng = QueryEngine(l7, device=self.device, l4_pipeline=l4)
+            ctx = SimpleNamespace(l7=l7, l4=l4, engine=eng, kg=kg)
+            self._org_ctx[org_id] = ctx
+        return ctx
+
+    def engine_for_org(self, org_id: int):
+        return self.org_context(org_id).engine
+
+    @staticmethod
+    def _scan_records(payload):
         import ctypes as ct
         import numpy as np
         from .ops import native
@@ -241,23 +274,19 @@ class DeepflowServer:
                                     len(payload),
                                     offs.ctypes.data_as(ct.c_void_p),
                                     lens.ctypes.data_as(ct.c_void_p), max_n))
+        return offs[:n].copy(), lens[:n].copy()
+
+    def _on_l7(self, hdr, payload) -> None:
+        offs, lens = self._scan_records(payload)
+        pipe = self.org_context(hdr.org_id).l7
         with self._lock:
-            self.l7.ingest(payload, offs[:n].copy(), lens[:n].copy())
+            pipe.ingest(payload, offs, lens)
 
     def _on_l4(self, hdr, payload) -> None:
-        import ctypes as ct
-        import numpy as np
-        from .ops import native
-        lib = native.cpu()
-        max_n = max(len(payload) // 8, 16)
-        offs = np.zeros(max_n, dtype=np.uint32)
-        lens = np.zeros(max_n, dtype=np.uint32)
-        n = int(lib.df_scan_offsets(payload.ctypes.data_as(ct.c_void_p),
-                                    len(payload),
-                                    offs.ctypes.data_as(ct.c_void_p),
-                                    lens.ctypes.data_as(ct.c_void_p), max_n))
+        offs, lens = self._scan_records(payload)
+        pipe = self.org_context(hdr.org_id).l4
         with self._lock:
-            self.l4.ingest(payload, offs[:n].copy(), lens[:n].copy())
+            pipe.ingest(payload, offs, lens)
 
     def _on_otel(self, hdr, payload) -> None:
         """OTLP frames carry one zlib-compressed TracesData blob
