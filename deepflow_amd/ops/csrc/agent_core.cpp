@@ -76,15 +76,23 @@ struct FlowNode {
     L7Counters l7c;
     h2::DynTable h2dyn[2];  // HPACK dynamic tables (per direction)
     std::map<uint32_t, L7Pending> h2_pending;  // per-stream outstanding reqs
+    uint32_t mq_seq = 0;  // FIFO key generator (pulsar pending queue)
+    uint8_t zmtp_greet_left[2] = {64, 64};  // greeting bytes to consume
+    bool is_v6 = false;
+    uint8_t ip6[2][16] = {{0}, {0}};  // client/server IPv6 addresses
 };
 
 struct FlowKeyC {
     uint32_t ip_a, ip_b;
     uint16_t port_a, port_b;
     uint8_t proto;
+    // folded 128->64-bit IPv6 addresses (0 for IPv4) disambiguate
+    // v6 flows whose 32-bit folds collide
+    uint64_t v6_a = 0, v6_b = 0;
     bool operator==(const FlowKeyC& o) const {
         return ip_a == o.ip_a && ip_b == o.ip_b && port_a == o.port_a &&
-               port_b == o.port_b && proto == o.proto;
+               port_b == o.port_b && proto == o.proto &&
+               v6_a == o.v6_a && v6_b == o.v6_b;
     }
 };
 
@@ -92,6 +100,7 @@ struct FlowKeyHash {
     size_t operator()(const FlowKeyC& k) const {
         uint64_t x = (uint64_t)k.ip_a << 32 | k.ip_b;
         uint64_t y = (uint64_t)k.port_a << 17 | (uint64_t)k.port_b << 1 | k.proto;
+        x ^= k.v6_a * 0x2545F4914F6CDD1Dull ^ k.v6_b;
         x ^= y + 0x9e3779b97f4a7c15ull + (x << 6) + (x >> 2);
         x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
         return (size_t)(x ^ (x >> 31));
@@ -694,6 +703,202 @@ bool parse_tls_client_hello(const uint8_t* p, uint32_t n, std::string& sni) {
     return true;
 }
 
+// SOME/IP (AUTOSAR): 16-byte header
+// [service u16][method u16][length u32][client u16][session u16]
+// [proto_ver=1][iface_ver][msg_type][return_code]; length covers bytes
+// from client_id on (reference parser: agent/src/flow_generator/
+// protocol_logs/rpc/some_ip.rs; validated against some_ip.pcap/.result)
+bool parse_someip(const uint8_t* p, uint32_t n, uint16_t& service,
+                  uint16_t& method, uint16_t& client, uint16_t& session,
+                  uint8_t& msg_type, uint8_t& ret_code, uint32_t& msg_len) {
+    if (n < 16) return false;
+    uint32_t len = (p[4] << 24) | (p[5] << 16) | (p[6] << 8) | p[7];
+    if (len < 8 || len + 8 > n) return false;
+    if (p[12] != 1) return false;  // protocol_version is always 1
+    uint8_t mt = p[14];
+    uint8_t mb = mt & ~0x20;  // 0x20 = TP segmentation flag
+    if (!(mb <= 2 || mb == 0x80 || mb == 0x81)) return false;
+    service = (p[0] << 8) | p[1];
+    method = (p[2] << 8) | p[3];
+    client = (p[8] << 8) | p[9];
+    session = (p[10] << 8) | p[11];
+    msg_type = mt;
+    ret_code = p[15];
+    msg_len = len + 8;
+    return true;
+}
+
+// ZMTP v3 (ZeroMQ transport, rfc.zeromq.org/spec/23): greeting
+// signature ff ..x8 7f, then version+mechanism; frames are
+// [flags][size(1|8 BE)][body], flags bit2=command bit1=long bit0=more.
+// Reference parser: protocol_logs/mq/zmtp.rs (zmtp_*.pcap corpus).
+struct ZmtpFrame {
+    bool is_command = false, more = false;
+    const uint8_t* body = nullptr;
+    uint64_t body_len = 0;
+    uint32_t frame_len = 0;
+};
+bool zmtp_next_frame(const uint8_t* p, uint32_t n, ZmtpFrame& fr) {
+    if (n < 2) return false;
+    uint8_t flags = p[0];
+    if (flags & 0xF8) return false;  // reserved bits must be 0
+    uint64_t size;
+    uint32_t hdr;
+    if (flags & 0x02) {  // long
+        if (n < 9) return false;
+        size = 0;
+        for (int i = 0; i < 8; i++) size = (size << 8) | p[1 + i];
+        hdr = 9;
+    } else {
+        size = p[1];
+        hdr = 2;
+    }
+    if (size > n - hdr) size = n - hdr;  // tolerate truncated capture
+    fr.is_command = flags & 0x04;
+    fr.more = flags & 0x01;
+    fr.body = p + hdr;
+    fr.body_len = size;
+    fr.frame_len = hdr + (uint32_t)size;
+    return true;
+}
+
+// Pulsar binary protocol: [totalSize u32BE][commandSize u32BE]
+// [BaseCommand protobuf]; BaseCommand field 1 = type enum.
+// Reference parser: protocol_logs/mq/pulsar.rs + PulsarApi.proto.
+bool parse_pulsar(const uint8_t* p, uint32_t n, uint32_t& type,
+                  std::string& topic, uint32_t& frame_len) {
+    if (n < 10) return false;
+    uint32_t total = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
+    uint32_t csize = (p[4] << 24) | (p[5] << 16) | (p[6] << 8) | p[7];
+    if (csize < 2 || total < csize + 4 || csize > n - 8) return false;
+    if (p[8] != 0x08) return false;  // BaseCommand field 1 (type) varint
+    // varint type (single or two bytes is enough for the enum range)
+    uint32_t t = p[9] & 0x7F;
+    uint32_t pos = 10;
+    if (p[9] & 0x80) {
+        if (csize < 3) return false;
+        t |= (uint32_t)(p[10] & 0x7F) << 7;
+        pos = 11;
+    }
+    if (t > 64) return false;
+    type = t;
+    // the per-type submessage follows; its field 1 is the topic string for
+    // PRODUCER/SUBSCRIBE/LOOKUP/PARTITIONED_METADATA (tag may be 2 bytes
+    // for field numbers >= 16)
+    uint32_t cmd_end = 8 + csize;
+    auto rd_varint = [&](uint32_t& q, uint64_t& out) {
+        out = 0;
+        int shift = 0;
+        while (q < cmd_end && shift < 35) {
+            uint8_t byt = p[q++];
+            out |= (uint64_t)(byt & 0x7F) << shift;
+            if (!(byt & 0x80)) return true;
+            shift += 7;
+        }
+        return false;
+    };
+    uint32_t q = pos;
+    uint64_t tag, sub_len;
+    if (rd_varint(q, tag) && (tag & 0x07) == 2 && rd_varint(q, sub_len) &&
+        q + sub_len <= cmd_end) {
+        uint32_t sp = q;
+        if (sp < cmd_end && p[sp] == 0x0A) {  // inner field 1 string
+            uint32_t tq = sp + 1;
+            uint64_t tl;
+            if (rd_varint(tq, tl) && tq + tl <= cmd_end && tl > 0) {
+                bool printable = true;
+                for (uint32_t i = 0; i < tl; i++)
+                    if (p[tq + i] < 0x20 || p[tq + i] > 0x7E)
+                        printable = false;
+                if (printable)
+                    topic.assign((const char*)p + tq, tl);
+            }
+        }
+    }
+    frame_len = total + 4 > n ? n : total + 4;
+    return true;
+}
+
+const char* pulsar_cmd_name(uint32_t t) {
+    switch (t) {
+        case 2: return "CONNECT";
+        case 3: return "CONNECTED";
+        case 4: return "SUBSCRIBE";
+        case 5: return "PRODUCER";
+        case 6: return "SEND";
+        case 7: return "SEND_RECEIPT";
+        case 8: return "SEND_ERROR";
+        case 9: return "MESSAGE";
+        case 10: return "ACK";
+        case 11: return "FLOW";
+        case 12: return "UNSUBSCRIBE";
+        case 13: return "SUCCESS";
+        case 14: return "ERROR";
+        case 15: return "CLOSE_PRODUCER";
+        case 16: return "CLOSE_CONSUMER";
+        case 17: return "PRODUCER_SUCCESS";
+        case 18: return "PING";
+        case 19: return "PONG";
+        case 21: return "PARTITIONED_METADATA";
+        case 22: return "PARTITIONED_METADATA_RESPONSE";
+        case 23: return "LOOKUP";
+        case 24: return "LOOKUP_RESPONSE";
+        default: return "COMMAND";
+    }
+}
+// request types await a response; others are responses or one-way
+inline bool pulsar_is_request(uint32_t t) {
+    return t == 2 || t == 4 || t == 5 || t == 6 || t == 12 || t == 18 ||
+           t == 21 || t == 23 || t == 15 || t == 16;
+}
+inline bool pulsar_is_response(uint32_t t) {
+    return t == 3 || t == 7 || t == 8 || t == 13 || t == 14 || t == 17 ||
+           t == 19 || t == 22 || t == 24;
+}
+
+// OpenWire (ActiveMQ): [length u32BE][data-type u8][...]. The
+// WireFormatInfo negotiation frame (type 1) carries the "ActiveMQ"
+// magic; loose-encoded commands then carry
+// [commandId u32][responseRequired u8], RESPONSE(30)/EXCEPTION_
+// RESPONSE(31) add [correlationId u32]. Tight encoding bit-packs these
+// (command names still recovered from the type byte).
+// Reference parser: protocol_logs/mq/openwire.rs (openwire_* corpus).
+const char* openwire_cmd_name(uint8_t t) {
+    switch (t) {
+        case 1: return "WIREFORMAT_INFO";
+        case 2: return "BROKER_INFO";
+        case 3: return "CONNECTION_INFO";
+        case 4: return "SESSION_INFO";
+        case 5: return "CONSUMER_INFO";
+        case 6: return "PRODUCER_INFO";
+        case 7: return "TRANSACTION_INFO";
+        case 8: return "DESTINATION_INFO";
+        case 9: return "REMOVE_SUBSCRIPTION_INFO";
+        case 10: return "KEEPALIVE_INFO";
+        case 11: return "SHUTDOWN_INFO";
+        case 12: return "REMOVE_INFO";
+        case 14: return "CONTROL_COMMAND";
+        case 15: return "FLUSH_COMMAND";
+        case 16: return "CONNECTION_ERROR";
+        case 17: return "CONSUMER_CONTROL";
+        case 18: return "CONNECTION_CONTROL";
+        case 21: return "MESSAGE_DISPATCH";
+        case 22: return "MESSAGE_ACK";
+        case 23: return "ACTIVEMQ_MESSAGE";
+        case 24: return "ACTIVEMQ_BYTES_MESSAGE";
+        case 25: return "ACTIVEMQ_MAP_MESSAGE";
+        case 26: return "ACTIVEMQ_OBJECT_MESSAGE";
+        case 27: return "ACTIVEMQ_STREAM_MESSAGE";
+        case 28: return "ACTIVEMQ_TEXT_MESSAGE";
+        case 30: return "RESPONSE";
+        case 31: return "EXCEPTION_RESPONSE";
+        case 32: return "DATA_RESPONSE";
+        case 33: return "DATA_ARRAY_RESPONSE";
+        case 34: return "INTEGER_RESPONSE";
+        default: return "COMMAND";
+    }
+}
+
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
 uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
@@ -805,6 +1010,32 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
         if (mlen == n && (opc == 2013 || opc == 2004 || opc == 2010))
             return 81;
     }
+    // ZMTP greeting signature (always the first bytes on the wire)
+    if (n >= 10 && p[0] == 0xFF && p[9] == 0x7F) return 106;
+    // OpenWire WireFormatInfo magic
+    if (n >= 13 && p[4] == 1 && memcmp(p + 5, "ActiveMQ", 8) == 0) {
+        uint32_t flen = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
+        if (flen + 4 <= n + 4096) return 103;
+    }
+    // Pulsar framed BaseCommand (strict: frame length matches packet)
+    {
+        uint32_t t, flen;
+        std::string topic;
+        if (parse_pulsar(p, n, t, topic, flen)) {
+            uint32_t total = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) |
+                             p[3];
+            if (total + 4 == n || server_port == 6650) return 105;
+        }
+    }
+    // SOME/IP: 16-byte header with proto_ver 1 and a known msg_type
+    {
+        uint16_t sv, me, cl, se;
+        uint8_t mt, rc;
+        uint32_t ml;
+        if (parse_someip(p, n, sv, me, cl, se, mt, rc, ml) &&
+            (ml == n || server_port == 30490 || server_port == 30501))
+            return 47;
+    }
     return 0;
 }
 
@@ -838,10 +1069,16 @@ void encode_l7_record(Agent& a, FlowNode& f, uint64_t req_ts, uint64_t resp_ts,
         });
         dfpb::f_u(s, 10, f.mac[0]);
         dfpb::f_u(s, 11, f.mac[1]);
-        dfpb::f_u(s, 12, f.ip[0]);
-        dfpb::f_u(s, 13, f.ip[1]);
-        dfpb::f_i(s, 16, lookup_epc(a, f.ip[0]));
-        dfpb::f_i(s, 17, lookup_epc(a, f.ip[1]));
+        if (f.is_v6) {
+            dfpb::f_u(s, 7, 1);  // is_ipv6
+            dfpb::f_s(s, 14, (const char*)f.ip6[0], 16);
+            dfpb::f_s(s, 15, (const char*)f.ip6[1], 16);
+        } else {
+            dfpb::f_u(s, 12, f.ip[0]);
+            dfpb::f_u(s, 13, f.ip[1]);
+        }
+        dfpb::f_i(s, 16, f.is_v6 ? 0 : lookup_epc(a, f.ip[0]));
+        dfpb::f_i(s, 17, f.is_v6 ? 0 : lookup_epc(a, f.ip[1]));
         dfpb::f_u(s, 18, f.port[0]);
         dfpb::f_u(s, 19, f.port[1]);
         dfpb::f_u(s, 20, f.proto);
@@ -898,8 +1135,13 @@ void encode_l4_record(Agent& a, FlowNode& f) {
             dfpb::f_u(k, 2, 3);
             dfpb::f_u(k, 4, f.mac[0]);
             dfpb::f_u(k, 5, f.mac[1]);
-            dfpb::f_u(k, 6, f.ip[0]);
-            dfpb::f_u(k, 7, f.ip[1]);
+            if (f.is_v6) {
+                dfpb::f_s(k, 8, (const char*)f.ip6[0], 16);
+                dfpb::f_s(k, 9, (const char*)f.ip6[1], 16);
+            } else {
+                dfpb::f_u(k, 6, f.ip[0]);
+                dfpb::f_u(k, 7, f.ip[1]);
+            }
             dfpb::f_u(k, 10, f.port[0]);
             dfpb::f_u(k, 11, f.port[1]);
             dfpb::f_u(k, 12, f.proto);
@@ -925,7 +1167,7 @@ void encode_l4_record(Agent& a, FlowNode& f) {
         dfpb::f_u(fl, 6, f.start_ns);
         dfpb::f_u(fl, 7, f.last_ns);
         dfpb::f_u(fl, 8, f.last_ns - f.start_ns);
-        dfpb::f_u(fl, 11, 0x0800);
+        dfpb::f_u(fl, 11, f.is_v6 ? 0x86DD : 0x0800);
         bool has_perf = f.rtt_us || f.srt_cnt || f.l7c.response_count;
         dfpb::f_u(fl, 12, has_perf ? 1 : 0);
         if (has_perf) {
@@ -1342,6 +1584,228 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
             f.l7.active = false;
         }
+    } else if (f.l7_protocol == 47) {  // SOME/IP (possibly batched)
+        uint32_t off = 0;
+        while (off + 16 <= n) {
+            uint16_t sv, me, cl, se;
+            uint8_t mt, rc;
+            uint32_t ml;
+            if (!parse_someip(p + off, n - off, sv, me, cl, se, mt, rc, ml))
+                break;
+            char svc[16], mth[16];
+            snprintf(svc, sizeof svc, "%u", sv);
+            snprintf(mth, sizeof mth, "%u", me);
+            uint8_t base = mt & ~0x20;  // strip TP segmentation flag
+            if (base <= 2) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = ml;
+                f.l7.req_type = base == 2 ? "Notification"
+                                : (base == 1 ? "RequestNoReturn" : "Request");
+                f.l7.service = svc;
+                f.l7.domain = svc;
+                f.l7.endpoint = mth;
+                f.l7.resource = std::string(svc) + "/" + mth;
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+                if (base == 1 || base == 2) {  // fire-and-forget
+                    encode_l7_record(a, f, ts, ts, 0, 0, f.l7,
+                                     "SOME/IP 1");
+                    f.l7.active = false;
+                }
+            } else if ((base == 0x80 || base == 0x81) && f.l7.active) {
+                uint8_t st = rc == 0 ? 0 : 3;
+                encode_l7_record(a, f, f.l7.req_ts, ts, rc, st, f.l7,
+                                 "SOME/IP 1");
+                f.l7.active = false;
+            }
+            off += ml;
+        }
+    } else if (f.l7_protocol == 106) {  // ZMTP
+        uint32_t off = 0;
+        // the first 64 bytes per direction are the greeting
+        // (signature + version + mechanism + as-server + filler)
+        if (f.zmtp_greet_left[dir]) {
+            uint32_t g = f.zmtp_greet_left[dir] < n ? f.zmtp_greet_left[dir]
+                                                    : n;
+            for (const char* m : {"PLAIN", "CURVE", "NULL"}) {
+                uint32_t ml = (uint32_t)strlen(m);
+                for (uint32_t i = 0; i + ml <= g; i++)
+                    if (memcmp(p + i, m, ml) == 0) {
+                        f.l7.domain = m;  // mechanism, carried on records
+                        i = g;
+                        break;
+                    }
+                if (!f.l7.domain.empty()) break;
+            }
+            f.zmtp_greet_left[dir] -= (uint8_t)g;
+            off = g;
+        }
+        while (off + 2 <= n) {
+            ZmtpFrame fr;
+            if (!zmtp_next_frame(p + off, n - off, fr)) break;
+            if (fr.is_command && fr.body_len >= 1) {
+                uint32_t nl = fr.body[0];
+                std::string name;
+                const uint8_t* data = nullptr;
+                uint64_t dlen = 0;
+                // libzmq quirk: "\x05ERROR" can appear merged as
+                // "\x5eRROR" on the wire; recognized like the reference
+                if (nl == 0x5E && fr.body_len >= 5 &&
+                    memcmp(fr.body + 1, "RROR", 4) == 0) {
+                    name = "ERROR";
+                    data = fr.body + 5;
+                    dlen = fr.body_len - 5;
+                } else if (1 + nl <= fr.body_len) {
+                    name.assign((const char*)fr.body + 1, nl);
+                    data = fr.body + 1 + nl;
+                    dlen = fr.body_len - 1 - nl;
+                }
+                if (!name.empty()) {
+                    L7Pending c;
+                    c.req_type = name;
+                    c.domain = f.l7.domain;
+                    c.req_len = fr.frame_len;
+                    uint8_t st = 0;
+                    if (name == "ERROR" && dlen >= 1 &&
+                        1u + data[0] <= dlen) {
+                        c.resource.assign((const char*)data + 1, data[0]);
+                        st = 3;
+                    } else if (name == "SUBSCRIBE" && dlen > 0) {
+                        c.resource.assign((const char*)data,
+                                          dlen > 255 ? 255 : dlen);
+                    } else if (name == "READY") {
+                        // metadata: [nlen][name][vlen u32 BE][value]...
+                        uint64_t mp = 0;
+                        while (mp + 1 < dlen) {
+                            uint32_t pn = data[mp];
+                            if (mp + 1 + pn + 4 > dlen) break;
+                            std::string prop((const char*)data + mp + 1, pn);
+                            uint32_t vl = (data[mp + 1 + pn] << 24) |
+                                          (data[mp + 2 + pn] << 16) |
+                                          (data[mp + 3 + pn] << 8) |
+                                          data[mp + 4 + pn];
+                            if (mp + 5 + pn + vl > dlen) break;
+                            if (prop == "Socket-Type")
+                                c.resource.assign(
+                                    (const char*)data + mp + 5 + pn, vl);
+                            mp += 5 + pn + vl;
+                        }
+                    }
+                    encode_l7_record(a, f, ts, ts, 0, st, c, "3");
+                }
+            } else if (!fr.is_command) {
+                if (dir == 0) {
+                    if (!f.l7.active) {
+                        f.l7.active = true;
+                        f.l7.req_ts = ts;
+                        f.l7.req_len = 0;
+                        f.l7.req_type = "Message";
+                        f.l7.resource.clear();
+                        f.l7.endpoint.clear();
+                        f.l7c.request_count++;
+                        f.last_req_pkt_ts = ts;
+                    }
+                    f.l7.req_len += (uint32_t)fr.body_len;
+                } else if (!fr.more) {  // final frame of a reply
+                    if (f.l7.active) {
+                        encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7,
+                                         "3");
+                        f.l7.active = false;
+                    } else if (fr.body_len) {  // pub/sub push: one-way
+                        L7Pending m;
+                        m.req_type = "Message";
+                        m.domain = f.l7.domain;
+                        m.req_len = (uint32_t)fr.body_len;
+                        encode_l7_record(a, f, ts, ts, 0, 0, m, "3");
+                    }
+                }
+            }
+            off += fr.frame_len;
+        }
+    } else if (f.l7_protocol == 105) {  // Pulsar
+        uint32_t off = 0;
+        while (off + 10 <= n) {
+            uint32_t t, flen;
+            std::string topic;
+            if (!parse_pulsar(p + off, n - off, t, topic, flen)) break;
+            const char* nm = pulsar_cmd_name(t);
+            if (pulsar_is_request(t)) {
+                L7Pending pend;
+                pend.active = true;
+                pend.req_ts = ts;
+                pend.req_len = flen;
+                pend.req_type = nm;
+                pend.resource = topic;
+                pend.endpoint = topic;
+                if (f.h2_pending.size() >= 64)
+                    f.h2_pending.erase(f.h2_pending.begin());
+                f.h2_pending[f.mq_seq++] = std::move(pend);
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            } else if (pulsar_is_response(t) && !f.h2_pending.empty()) {
+                auto it = f.h2_pending.begin();  // FIFO match
+                uint8_t st = (t == 8 || t == 14) ? 3 : 0;
+                encode_l7_record(a, f, it->second.req_ts, ts, 0, st,
+                                 it->second, "");
+                f.h2_pending.erase(it);
+            }
+            // one-way types (MESSAGE/ACK/FLOW) tracked via counters only
+            off += flen;
+        }
+    } else if (f.l7_protocol == 103) {  // OpenWire
+        uint32_t off = 0;
+        while (off + 5 <= n) {
+            uint32_t flen = (p[off] << 24) | (p[off + 1] << 16) |
+                            (p[off + 2] << 8) | p[off + 3];
+            if (flen < 1) break;
+            if (flen + 4 > n - off) flen = n - off - 4;  // truncated capture
+            uint32_t fl = flen + 4;
+            uint8_t t = p[off + 4];
+            if (t >= 30 && t <= 34) {
+                // loose response: [len][type][cmdId][respReq][corrId]
+                if (fl >= 14 && p[off + 9] <= 1) {
+                    uint32_t corr = (p[off + 10] << 24) |
+                                    (p[off + 11] << 16) |
+                                    (p[off + 12] << 8) | p[off + 13];
+                    auto it = f.h2_pending.find(corr);
+                    if (it != f.h2_pending.end()) {
+                        encode_l7_record(a, f, it->second.req_ts, ts, 0,
+                                         t == 31 ? 3 : 0, it->second, "");
+                        f.h2_pending.erase(it);
+                    }
+                }
+            } else if (t == 1) {
+                L7Pending w;
+                w.req_type = "WIREFORMAT_INFO";
+                w.domain = "ActiveMQ";
+                w.req_len = fl;
+                encode_l7_record(a, f, ts, ts, 0, 0, w, "");
+            } else if (fl >= 10 && p[off + 9] == 1) {
+                // loose command awaiting a RESPONSE (keyed by commandId)
+                uint32_t cid = (p[off + 5] << 24) | (p[off + 6] << 16) |
+                               (p[off + 7] << 8) | p[off + 8];
+                L7Pending pend;
+                pend.active = true;
+                pend.req_ts = ts;
+                pend.req_len = fl;
+                pend.req_type = openwire_cmd_name(t);
+                if (f.h2_pending.size() >= 64)
+                    f.h2_pending.erase(f.h2_pending.begin());
+                f.h2_pending[cid] = std::move(pend);
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            } else if (t >= 21 && t <= 28) {
+                // fire-and-forget message/dispatch (or tight encoding,
+                // where only the type byte is recoverable)
+                L7Pending m;
+                m.req_type = openwire_cmd_name(t);
+                m.req_len = fl;
+                encode_l7_record(a, f, ts, ts, 0, 0, m, "");
+                f.l7c.request_count++;
+            }
+            off += fl;
+        }
     } else if (f.l7_protocol == 121) {  // TLS: ClientHello SNI only
         std::string sni;
         if (dir == 0 && parse_tls_client_hello(p, n, sni)) {
@@ -1527,14 +1991,51 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
         eth = (pkt[off + 2] << 8) | pkt[off + 3];
         off += 4;
     }
-    if (eth != 0x0800) return 0;  // non-IPv4 ignored
+    bool is_v6 = false;
+    const uint8_t *v6src = nullptr, *v6dst = nullptr;
+    uint32_t ihl;
+    uint32_t tot;
+    uint8_t proto;
+    uint32_t src, dst;
     const uint8_t* ip = pkt + off;
-    if (len < off + 20) { a.parse_errors++; return -1; }
-    uint32_t ihl = (ip[0] & 0x0F) * 4;
-    uint16_t tot = (ip[2] << 8) | ip[3];
-    uint8_t proto = ip[9];
-    uint32_t src = (ip[12] << 24) | (ip[13] << 16) | (ip[14] << 8) | ip[15];
-    uint32_t dst = (ip[16] << 24) | (ip[17] << 16) | (ip[18] << 8) | ip[19];
+    if (eth == 0x0800) {
+        if (len < off + 20) { a.parse_errors++; return -1; }
+        ihl = (ip[0] & 0x0F) * 4;
+        tot = (ip[2] << 8) | ip[3];
+        proto = ip[9];
+        src = (ip[12] << 24) | (ip[13] << 16) | (ip[14] << 8) | ip[15];
+        dst = (ip[16] << 24) | (ip[17] << 16) | (ip[18] << 8) | ip[19];
+    } else if (eth == 0x86DD) {  // IPv6: fixed 40B header + ext chain
+        if (len < off + 40) { a.parse_errors++; return -1; }
+        uint16_t plen = (ip[4] << 8) | ip[5];
+        uint8_t next = ip[6];
+        uint32_t hl = 40;
+        while (next == 0 || next == 43 || next == 60) {  // hop/route/dst
+            if (len < off + hl + 8) { a.parse_errors++; return -1; }
+            const uint8_t* eh = ip + hl;
+            next = eh[0];
+            hl += (eh[1] + 1) * 8;
+        }
+        if (next != 6 && next != 17) return 0;
+        proto = next;
+        is_v6 = true;
+        v6src = ip + 8;
+        v6dst = ip + 24;
+        // folded 32-bit forms feed the flow key tie-break and EPC lookup
+        auto fold = [](const uint8_t* q) {
+            uint32_t v = 0;
+            for (int i = 0; i < 16; i += 4)
+                v ^= (q[i] << 24) | (q[i + 1] << 16) | (q[i + 2] << 8) |
+                     q[i + 3];
+            return v;
+        };
+        src = fold(v6src);
+        dst = fold(v6dst);
+        ihl = hl;
+        tot = 40 + plen;  // so tot - ihl = L4 bytes, as in the v4 path
+    } else {
+        return 0;  // other ethertypes ignored
+    }
     if (proto != 6 && proto != 17) return 0;
     const uint8_t* l4 = ip + ihl;
     if (len < off + ihl + (proto == 6 ? 20 : 8)) { a.parse_errors++; return -1; }
@@ -1553,6 +2054,16 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
     FlowKeyC key{a_first ? src : dst, a_first ? dst : src,
                  (uint16_t)(a_first ? sport : dport),
                  (uint16_t)(a_first ? dport : sport), proto};
+    if (is_v6) {
+        auto fold64 = [](const uint8_t* q) {
+            uint64_t v = 0;
+            for (int i = 0; i < 16; i++) v = v * 131 + q[i];
+            return v;
+        };
+        uint64_t va = fold64(v6src), vb = fold64(v6dst);
+        key.v6_a = a_first ? va : vb;
+        key.v6_b = a_first ? vb : va;
+    }
     auto it = a.flows.find(key);
     int dir;
     if (it == a.flows.end()) {
@@ -1564,6 +2075,11 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
         f.port[0] = sport; f.port[1] = dport;
         f.mac[0] = mac_src; f.mac[1] = mac_dst;
         f.proto = proto;
+        if (is_v6) {
+            f.is_v6 = true;
+            memcpy(f.ip6[0], v6src, 16);
+            memcpy(f.ip6[1], v6dst, 16);
+        }
         it = a.flows.emplace(key, std::move(f)).first;
     }
     FlowNode& f = it->second;

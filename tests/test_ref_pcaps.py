@@ -254,3 +254,77 @@ def test_rpc_mq_pcaps():
 
     l7, _, _ = replay(f"{FIX}/nats/nats-skywalking.pcap")
     assert any(r["req"]["req_type"] in ("PUB", "HPUB") for r in l7)
+
+
+def test_zmtp_pcaps():
+    # REQ/REP NULL-mechanism session: READY handshake + request/response
+    l7, _, stats = replay(f"{FIX}/zmtp/zmtp_null.pcap")
+    assert all(r["base"]["head"]["proto"] == 106 for r in l7)
+    types = [r["req"]["req_type"] for r in l7]
+    assert "READY" in types and "Message" in types
+    ready = [r for r in l7 if r["req"]["req_type"] == "READY"]
+    assert {r["req"]["resource"] for r in ready} == {"REQ", "REP"}
+    msg = [r for r in l7 if r["req"]["req_type"] == "Message"]
+    assert msg and msg[0]["base"]["head"].get("rrt", 0) > 0
+
+    # SUBSCRIBE command carries the topic
+    l7, _, _ = replay(f"{FIX}/zmtp/zmtp_subscribe_one.pcap")
+    subs = [r["req"]["resource"] for r in l7
+            if r["req"]["req_type"] == "SUBSCRIBE"]
+    assert subs == ["sports.general"], \
+        [r["req"]["req_type"] for r in l7]
+
+    # ERROR command (incl. the libzmq \x5eRROR merge quirk) -> server error
+    l7, _, _ = replay(f"{FIX}/zmtp/zmtp_error.pcap")
+    errs = [r for r in l7 if r["req"]["req_type"] == "ERROR"]
+    assert errs and all(e["resp"].get("status") == 3 for e in errs)
+    assert errs[0]["req"]["resource"] == "400"
+
+
+def test_someip_pcap():
+    l7, _, stats = replay(f"{FIX}/some_ip/some_ip.pcap")
+    assert l7, "no SOME/IP records"
+    assert all(r["base"]["head"]["proto"] == 47 for r in l7)
+    # reference .result: request_resource 41 (service), endpoint 20484
+    # (method), matched request/response with return code E_OK
+    r = l7[0]
+    assert r["req"]["req_type"] == "Request"
+    assert r["req"]["resource"] == "41/20484"
+    assert r["resp"].get("code", 0) == 0
+    assert r["base"]["head"].get("rrt", 0) > 0
+
+
+def test_pulsar_pcaps():
+    # IPv6 capture: also exercises the agent's v6 L3 path
+    l7, l4, _ = replay(f"{FIX}/pulsar/pulsar-producer.pcap")
+    assert all(r["base"]["head"]["proto"] == 105 for r in l7)
+    types = {r["req"]["req_type"] for r in l7}
+    assert "CONNECT" in types and "SEND" in types
+    prod = [r for r in l7 if r["req"]["req_type"] == "PRODUCER"]
+    assert prod and prod[0]["req"]["resource"] == \
+        "persistent://public/default/my-topic"
+    sends = [r for r in l7 if r["req"]["req_type"] == "SEND"]
+    assert len(sends) == 10  # ten send/send_receipt pairs in the fixture
+    # rrt values line up with the reference .result for the same pairs
+    assert sends[0]["base"]["head"]["rrt"] == 11240
+    # the L4 flow carries the IPv6 addresses
+    assert l4 and l4[0]["flow"]["flow_key"].get("ip6_src")
+
+    l7, _, _ = replay(f"{FIX}/pulsar/pulsar-consumer.pcap")
+    subs = [r for r in l7 if r["req"]["req_type"] == "SUBSCRIBE"]
+    assert subs and subs[0]["req"]["resource"] == \
+        "persistent://public/default/my-topic"
+
+
+def test_openwire_pcaps():
+    l7, _, _ = replay(f"{FIX}/openwire/openwire_loose_producer.pcap")
+    assert all(r["base"]["head"]["proto"] == 103 for r in l7)
+    types = [r["req"]["req_type"] for r in l7]
+    assert "WIREFORMAT_INFO" in types
+    assert "CONNECTION_INFO" in types  # loose command matched by corrId
+    conn = [r for r in l7 if r["req"]["req_type"] == "CONNECTION_INFO"]
+    assert conn[0]["base"]["head"].get("rrt", 0) > 0
+
+    l7, _, _ = replay(f"{FIX}/openwire/openwire_exception.pcap")
+    errs = [r for r in l7 if r["resp"].get("status") == 3]
+    assert errs, [r["req"]["req_type"] for r in l7]
