@@ -80,6 +80,7 @@ struct FlowNode {
     uint8_t zmtp_greet_left[2] = {64, 64};  // greeting bytes to consume
     bool is_v6 = false;
     uint8_t ip6[2][16] = {{0}, {0}};  // client/server IPv6 addresses
+    std::vector<uint8_t> h2_carry[2];  // cross-segment frame reassembly
 };
 
 struct FlowKeyC {
@@ -1292,10 +1293,22 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
         return;
     }
     if (f.l7_protocol == 21 || f.l7_protocol == 41) {  // HTTP/2 / gRPC
+        // TCP reassembly: a frame split across segments is carried over
+        // and re-parsed with the next payload of the same direction
+        // (reference: flow_node.rs tcp-segment merge; grpc-segmented.pcap)
+        std::vector<uint8_t> merged;
+        if (!f.h2_carry[dir].empty()) {
+            merged.swap(f.h2_carry[dir]);
+            merged.insert(merged.end(), p, p + n);
+            p = merged.data();
+            n = (uint32_t)merged.size();
+        }
         uint32_t pos = 0;
         if (n >= 24 && memcmp(p, h2::PREFACE, 24) == 0) pos = 24;
         h2::FrameView fr;
+        uint32_t consumed = pos;
         while (h2::next_frame(p, n, pos, fr)) {
+            consumed = pos;
             if (fr.type != h2::F_HEADERS) continue;
             uint32_t off = 0, pad = 0;
             if (fr.flags & 0x08) {  // PADDED
@@ -1354,6 +1367,8 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 }
             }
         }
+        if (consumed < n && n - consumed <= (128u << 10))
+            f.h2_carry[dir].assign(p + consumed, p + n);
         return;
     }
     if (f.l7_protocol == 20) {  // HTTP/1
@@ -2128,6 +2143,19 @@ void dfa_tick(void* h, uint64_t now_ns) {
         bool closed = f.close_type != 0;
         bool idle = now_ns > f.last_ns && now_ns - f.last_ns > FLOW_TIMEOUT_NS;
         if (closed || idle) {
+            // flush unanswered requests (reference emits request-only
+            // records on flow close; msg_type stays session, rrt 0)
+            if (f.l7.active) {
+                encode_l7_record(a, f, f.l7.req_ts, f.l7.req_ts, 0, 0,
+                                 f.l7, "");
+                f.l7.active = false;
+            }
+            for (auto& kv : f.h2_pending)
+                encode_l7_record(a, f, kv.second.req_ts, kv.second.req_ts,
+                                 0, 0, kv.second,
+                                 f.l7_protocol == 21 ||
+                                 f.l7_protocol == 41 ? "2" : "");
+            f.h2_pending.clear();
             encode_l4_record(a, f);
             it = a.flows.erase(it);
         } else {

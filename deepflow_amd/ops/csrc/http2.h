@@ -210,14 +210,24 @@ inline bool looks_like_http2(const uint8_t* p, uint32_t n) {
     uint32_t pos = 0;
     FrameView f;
     int frames = 0;
+    uint8_t first_type = 0xFF;
     while (next_frame(p, n, pos, f)) {
         if (f.type > 9) return false;
+        if (frames == 0) first_type = f.type;
         frames++;
         if (frames >= 2) break;
     }
-    // one full HEADERS/SETTINGS frame covering the payload also qualifies
-    return frames >= 2 || (frames == 1 && pos == n &&
-                           (f.type == F_HEADERS || f.type == F_SETTINGS));
+    // one full HEADERS/SETTINGS frame also qualifies, covering the
+    // payload exactly or followed by a truncated frame (TCP-segmented
+    // captures end mid-frame; the tail must still look like a frame
+    // header with a sane type). NB: next_frame fills f before its
+    // length check, so use the remembered first_type.
+    if (frames >= 2) return true;
+    if (frames != 1 || (first_type != F_HEADERS && first_type != F_SETTINGS))
+        return false;
+    if (pos == n) return true;
+    if (n - pos >= 9) return p[pos + 3] <= 9;
+    return true;  // < 9 tail bytes: indeterminate, accept
 }
 
 }  // namespace h2

@@ -328,3 +328,17 @@ def test_openwire_pcaps():
     l7, _, _ = replay(f"{FIX}/openwire/openwire_exception.pcap")
     errs = [r for r in l7 if r["resp"].get("status") == 3]
     assert errs, [r["req"]["req_type"] for r in l7]
+
+
+def test_grpc_segmented_pcap():
+    """HEADERS frame followed by a DATA frame split across TCP segments:
+    exercises the per-direction carry-over reassembly + the request-flush
+    on flow close (.result expects one request record, no response)."""
+    l7, _, _ = replay(f"{FIX}/http/grpc-segmented.pcap")
+    assert len(l7) == 1
+    assert l7[0]["req"]["resource"] == "/agent.Synchronizer/Push"
+    assert l7[0]["ext_info"]["service_name"] == "agent.Synchronizer"
+
+    # pure bulk TCP (iperf) must yield no L7 records
+    l7, _, stats = replay(f"{FIX}/tcp-segment.pcap")
+    assert l7 == [] and stats["parse_errors"] == 0
