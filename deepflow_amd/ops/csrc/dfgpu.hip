@@ -1171,6 +1171,51 @@ __global__ void k_query_select(SegView s, QuerySpec q, uint32_t n, uint64_t base
     if (e < out_cap) out_rows[e] = row;
 }
 
+// ----------------------------------------------------------------------
+// cold-segment bit packing: fixed-width pack of (value - base) streams.
+// Word-centric pack (each thread owns one output u32 — no atomics, fully
+// coalesced writes); value-centric unpack. A constant column packs to
+// bits==0 (no payload at all). Extends the HBM hot window several-fold
+// for demoted segments (reference analog: ClickHouse column codecs
+// T64/DoubleDelta on cold parts).
+// ----------------------------------------------------------------------
+__global__ void k_pack_bits(const uint32_t* __restrict__ src, uint32_t n,
+                            uint32_t base, uint32_t bits,
+                            uint32_t* __restrict__ out,
+                            uint32_t out_words) {
+    uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
+    if (w >= out_words) return;
+    uint64_t bit0 = (uint64_t)w * 32u;
+    uint32_t i = (uint32_t)(bit0 / bits);
+    uint32_t acc = 0;
+    for (; i < n; i++) {
+        uint64_t vb = (uint64_t)i * bits;
+        if (vb >= bit0 + 32u) break;
+        if (vb + bits <= bit0) continue;  // first value may start earlier
+        uint64_t v = (uint64_t)(src[i] - base);
+        int64_t sh = (int64_t)vb - (int64_t)bit0;
+        if (sh >= 0)
+            acc |= (uint32_t)(v << sh);
+        else
+            acc |= (uint32_t)(v >> (uint32_t)(-sh));
+    }
+    out[w] = acc;
+}
+
+__global__ void k_unpack_bits(const uint32_t* __restrict__ packed,
+                              uint32_t n, uint32_t base, uint32_t bits,
+                              uint32_t* __restrict__ out) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t vb = (uint64_t)i * bits;
+    uint32_t w = (uint32_t)(vb >> 5), off = (uint32_t)(vb & 31u);
+    uint64_t lo = packed[w];
+    uint64_t hi = (off + bits > 32u) ? packed[w + 1] : 0ull;
+    uint64_t v = ((hi << 32) | lo) >> off;
+    uint32_t mask = bits >= 32u ? 0xFFFFFFFFu : ((1u << bits) - 1u);
+    out[i] = base + (uint32_t)(v & mask);
+}
+
 inline uint32_t grid_for(uint64_t total) {
     return (uint32_t)((total + BLOCK - 1) / BLOCK);
 }
@@ -1365,6 +1410,22 @@ int df_spec_sizes(uint32_t* qterm, uint32_t* qkey, uint32_t* qagg, uint32_t* qsp
     *qterm = sizeof(QTerm); *qkey = sizeof(QKey); *qagg = sizeof(QAgg);
     *qspec = sizeof(QuerySpec);
     return 0;
+}
+
+int df_pack_bits(const void* src, uint32_t n, uint32_t base, uint32_t bits,
+                 void* out, uint32_t out_words, uint64_t stream) {
+    hipLaunchKernelGGL(k_pack_bits, dim3(grid_for(out_words)), dim3(BLOCK),
+                       0, STREAM(stream), (const uint32_t*)src, n, base,
+                       bits, (uint32_t*)out, out_words);
+    return (int)hipGetLastError();
+}
+
+int df_unpack_bits(const void* packed, uint32_t n, uint32_t base,
+                   uint32_t bits, void* out, uint64_t stream) {
+    hipLaunchKernelGGL(k_unpack_bits, dim3(grid_for(n)), dim3(BLOCK), 0,
+                       STREAM(stream), (const uint32_t*)packed, n, base,
+                       bits, (uint32_t*)out);
+    return (int)hipGetLastError();
 }
 
 }  // extern "C"

@@ -108,6 +108,10 @@ def _decl_gpu(lib: ct.CDLL) -> None:
                                     u64, p, u32, u64, p, p, u32, u64]
     lib.df_spec_sizes.restype = ct.c_int
     lib.df_spec_sizes.argtypes = [p, p, p, p]
+    lib.df_pack_bits.restype = ct.c_int
+    lib.df_pack_bits.argtypes = [p, u32, u32, u32, p, u32, u64]
+    lib.df_unpack_bits.restype = ct.c_int
+    lib.df_unpack_bits.argtypes = [p, u32, u32, u32, p, u64]
 
 
 def gpu() -> ct.CDLL:

@@ -160,7 +160,7 @@ class QueryEngine:
             plan = parse_sql(sql, dictionary=self.pipe.dict,
                              time_base_s=self.pipe.time_base_s,
                              tags=L7_TAGS, metrics=L7_METRICS)
-            return self._run_segments(plan, self.pipe.segments.segments,
+            return self._run_segments(plan, self.pipe.segments.scan_list(),
                                       L7_TAGS, S.STR_COLS)
         if table == "l4_flow_log":
             if self.l4 is None:
@@ -169,7 +169,7 @@ class QueryEngine:
                              time_base_s=self.l4.time_base_s,
                              tags=L4_TAGS, metrics=L4_METRICS)
             from ..store import l4_schema as L4S
-            return self._run_segments(plan, self.l4.segments.segments,
+            return self._run_segments(plan, self.l4.segments.scan_list(),
                                       L4_TAGS, L4S.STR_COLS)
         row_tables = {
             "event": "event_rows", "perf_event": "perf_event_rows",
@@ -245,14 +245,14 @@ class QueryEngine:
         plan = Q.Plan(time_base_s=self.pipe.time_base_s if kind == "app"
                       else (self.l4.time_base_s if self.l4 else 0))
         if kind == "app":
-            segs = self.pipe.segments.segments
+            segs = self.pipe.segments.scan_list()
             from .tags import L7_TAGS as T
             dur = T["response_duration"]
             blen = T["response_length"]
         else:
             if self.l4 is None:
                 raise SqlError("network_map needs the l4 pipeline")
-            segs = self.l4.segments.segments
+            segs = self.l4.segments.scan_list()
             from .tags import L4_TAGS as T
             dur = T["rtt"]
             blen = T["byte_rx"]
@@ -347,14 +347,14 @@ class QueryEngine:
             plan = parse_sql(sql, dictionary=self.pipe.dict,
                              time_base_s=self.pipe.time_base_s,
                              tags=L7_TAGS, metrics=L7_METRICS)
-            segments, tags, str_cols = (self.pipe.segments.segments, L7_TAGS,
+            segments, tags, str_cols = (self.pipe.segments.scan_list(), L7_TAGS,
                                         S.STR_COLS)
         else:
             from ..store import l4_schema as L4S
             plan = parse_sql(sql, dictionary=None,
                              time_base_s=self.l4.time_base_s,
                              tags=L4_TAGS, metrics=L4_METRICS)
-            segments, tags, str_cols = (self.l4.segments.segments, L4_TAGS,
+            segments, tags, str_cols = (self.l4.segments.scan_list(), L4_TAGS,
                                         L4S.STR_COLS)
         if plan.select_rows:
             return {"kind": "rows",

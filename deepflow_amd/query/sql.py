@@ -400,6 +400,18 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
             return
         plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident,
                                  group=group))
+    elif td.hydrate == "status":
+        from .engine import STATUS_NAMES
+        rev = {v.lower(): k for k, v in STATUS_NAMES.items()}
+        ident = rev.get(lit[1].lower())
+        if ident is None:
+            if group:
+                _never(plan, group)
+            else:
+                plan.impossible = True
+            return
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident,
+                                 group=group))
     elif td.hydrate == "ip":
         import ipaddress
         v = int(ipaddress.IPv4Address(lit[1]))

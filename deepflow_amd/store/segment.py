@@ -206,10 +206,49 @@ class SegmentSet:
 
     @property
     def n_rows(self) -> int:
-        return sum(s.n_rows for s in self.segments)
+        return sum(s.n_rows for s in self.segments) + \
+            sum(c.n_rows for c in getattr(self, "cold", []))
 
     def total_stored_bytes(self) -> int:
         total = 0
         for s in self.segments:
             total += int(s.stored_bytes_per_row() * s.n_rows)
+        for c in getattr(self, "cold", []):
+            total += c.compressed_bytes()
         return total
+
+    # -------------------------------------------------- cold demotion
+    def demote_oldest(self, stream: int = 0) -> bool:
+        """Compress the oldest full hot segment into the cold tier
+        (store/coldstore.py bit-pack codec); its buffers go to the
+        free-list. Returns False if nothing is demotable."""
+        if len(self.segments) <= 1:
+            return False
+        from .coldstore import CompressedL7Segment
+        if not hasattr(self, "cold"):
+            self.cold = []
+        seg = self.segments.pop(0)
+        self.cold.append(CompressedL7Segment(seg, stream))
+        self._free.append(seg)
+        return True
+
+    def scan_list(self, stream: int = 0) -> List:
+        """All queryable segments: cold ones materialized into recycled
+        scratch segments (one-shot; the scratch returns to the free-list
+        semantics by being reused on the next call)."""
+        cold = getattr(self, "cold", [])
+        if not cold:
+            return self.segments
+        out = []
+        if not hasattr(self, "_scratch"):
+            self._scratch = []
+        while len(self._scratch) < len(cold):
+            if self._free:
+                seg = self._free.pop(0)
+            else:
+                seg = self.cls(self.segment_rows, self.device)
+            self._scratch.append(seg)
+        for c, scratch in zip(cold, self._scratch):
+            self.reset_segment(scratch)
+            out.append(c.materialize(scratch, stream))
+        return out + self.segments
