@@ -156,10 +156,20 @@ class SegmentSet:
         if self.max_bytes is None:
             return 0
         dropped = 0
-        while (len(self.segments) > 1 and
-               self.total_alloc_bytes() + sum(
-                   self.seg_alloc_bytes(s) for s in self._free) >
-               self.max_bytes):
+
+        def over():
+            return self.total_alloc_bytes() + sum(
+                self.seg_alloc_bytes(s) for s in self._free) + sum(
+                c.compressed_bytes() for c in getattr(self, "cold", [])) > \
+                self.max_bytes
+
+        # oldest cold segments go first (they were demoted earliest)
+        while getattr(self, "cold", []) and over():
+            c = self.cold.pop(0)
+            self.evicted_rows += c.n_rows
+            self.evicted_segments += 1
+            dropped += 1
+        while len(self.segments) > 1 and over():
             seg = self.segments.pop(0)
             self.evicted_rows += seg.n_rows
             self.evicted_segments += 1
@@ -201,6 +211,14 @@ class SegmentSet:
                 self.segments.append(seg)
             else:
                 self.segments.append(self.cls(self.segment_rows, self.device))
+            # two-tier policy: over the hot watermark -> demote (compress)
+            # oldest hot segments; over max_bytes -> drop oldest cold
+            hot_max = getattr(self, "hot_max_bytes", None)
+            if hot_max is not None:
+                while (len(self.segments) > 1 and
+                       self.total_alloc_bytes() > hot_max):
+                    if not self.demote_oldest():
+                        break
             self.enforce_watermark()
         return self.segments[-1]
 

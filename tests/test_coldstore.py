@@ -95,3 +95,20 @@ def test_demote_recycles_buffers():
     assert len(pipe.segments.segments) == segs_before - 1
     assert len(pipe.segments._free) == free_before + 1
     assert pipe.segments.n_rows == 2 * CFG.n  # rows still accounted
+
+
+def test_two_tier_watermark():
+    """Over hot_max_bytes -> compress oldest; over max_bytes -> drop
+    oldest cold."""
+    pipe = _pipe()                      # 2 segments of 2000 rows
+    segset = pipe.segments
+    seg_bytes = segset.seg_alloc_bytes(segset.segments[0])
+    segset.hot_max_bytes = int(seg_bytes * 1.5)   # hot window: 1 segment
+    from deepflow_amd.gen.spans import gen_span_payload
+    pipe.ingest_frame_payload(gen_span_payload(CFG))  # rolls a 3rd segment
+    assert len(segset.cold) >= 1        # oldest got compressed, not lost
+    assert segset.n_rows == 3 * CFG.n
+    r_cold = segset.cold[0].compressed_bytes()
+    # packed form is well under the allocated segment footprint (the raw
+    # string pool survives uncompressed, so the bound is conservative)
+    assert r_cold < seg_bytes * 0.7
