@@ -171,21 +171,46 @@ class Agent:
             self.platform_version = resp["platform_version"]
         return resp
 
-    def flush_to_server(self, now_ns: int, compress: bool = False) -> int:
+    def dfstats_payload(self, now_s: int = 0) -> bytes:
+        """Agent self-metrics as a MSG_DFSTATS payload (reference: the
+        agent's stats collector shipping deepflow_system rows)."""
+        from ..wire import pb, metric
+        st = self.stats()
+        rec = {
+            "timestamp": now_s,
+            "name": "deepflow_agent",
+            "tag_names": ["host", "agent_id"],
+            "tag_values": [socket.gethostname(), str(self.agent_id)],
+            "metrics_float_names": [k for k in st if k != "_r"],
+            "metrics_float_values": [float(st[k]) for k in st if k != "_r"],
+            "org_id": self.org_id,
+            "team_id": self.team_id,
+        }
+        return framing.pack_records([pb.encode(rec, metric.STATS)])
+
+    def flush_to_server(self, now_ns: int, compress: bool = False,
+                        with_stats: bool = True) -> int:
         """tick + drain all types + send framed payloads to the server
         (uniform-sender analog; compress=True uses whole-payload zstd like
-        the reference's SenderEncoder::Zstd). Returns frames sent."""
+        the reference's SenderEncoder::Zstd). Also ships a MSG_DFSTATS
+        self-metrics frame. Returns frames sent."""
         self.tick(now_ns)
-        sent = 0
+        frames = []
         for which in (DRAIN_L4, DRAIN_L7, DRAIN_DOC):
             payload = self.drain(which)
-            if not payload:
-                continue
-            frame = self.frame(which, payload, compress=compress)
+            if payload:
+                frames.append(self.frame(which, payload, compress=compress))
+        if with_stats:
+            hdr = framing.FrameHeader(msg_type=framing.MSG_DFSTATS,
+                                      team_id=self.team_id,
+                                      org_id=self.org_id,
+                                      agent_id=self.agent_id)
+            frames.append(framing.encode_frame(
+                hdr, self.dfstats_payload(now_ns // 10**9)))
+        for frame in frames:
             if self.server is not None:
                 if self._sock is None:
                     self._sock = socket.create_connection(self.server,
                                                           timeout=5)
                 self._sock.sendall(frame)
-            sent += 1
-        return sent
+        return len(frames)
