@@ -109,7 +109,7 @@ def test_agent_to_server_end_to_end():
                                           t0=t0 + i * 10**7):
                 a.packet(frame, ts)
         sent = a.flush_to_server(10**9 * 100)
-        assert sent == 3
+        assert sent == 4  # l4 + l7 + doc + dfstats frames
         deadline = time.time() + 20
         while time.time() < deadline and srv.l7.stats.spans_in < 20:
             time.sleep(0.1)
