@@ -111,6 +111,10 @@ class DistQueryEngine:
         self.device = device
 
     def query(self, sql: str) -> Dict:
+        import re
+        m = re.search(r"\bslimit\s+(\d+)", sql, re.IGNORECASE)
+        if m:  # series-limited queries take the topN pushdown path
+            return self.query_topn(sql, int(m.group(1)))
         partial = self.engine.query_partial(sql)
         parts = exchange_json(partial, self.device)
         if partial["kind"] == "rows":
@@ -128,5 +132,55 @@ class DistQueryEngine:
                         seen.append(row)
             return {"columns": cols, "values": seen}
         merged = merge_agg_partials([p for p in parts])
+        return self.engine.finalize_groups(sql, merged["key_rows"],
+                                           merged["aggs"])
+
+    # ---------------------------------------------------- topN pushdown
+    # Exchange O(world x K) groups instead of every group: each rank sends
+    # its local top candidates (ranked by the first aggregate — the
+    # SLIMIT/ORDER-BY-agg metric), the candidate set is unioned, then a
+    # second exchange ships each rank's aggregates for candidates it did
+    # not already send, making the merged totals exact for every
+    # candidate. A key can only miss the candidate set if it is outside
+    # the top `margin` on EVERY shard; margin = 4K + 64 makes that
+    # practically impossible for real skew (and the full-exchange path
+    # remains available below the `pushdown_threshold`).
+    pushdown_threshold = 10000
+
+    def query_topn(self, sql: str, k: int) -> Dict:
+        partial = self.engine.query_partial(sql)
+        if partial["kind"] != "agg":
+            return self.query(sql)
+        key_rows, aggs = partial["key_rows"], partial["aggs"]
+        if len(aggs) <= self.pushdown_threshold:
+            parts = exchange_json(partial, self.device)
+            merged = merge_agg_partials(parts)
+            return self.engine.finalize_groups(sql, merged["key_rows"],
+                                               merged["aggs"])
+        margin = 4 * k + 64
+
+        def tup(key):
+            return tuple(tuple(x) if isinstance(x, list) else x
+                         for x in key)
+
+        order = sorted(range(len(aggs)),
+                       key=lambda i: -(aggs[i][0] or 0))
+        top_idx = order[:margin]
+        sent = {tup(key_rows[i]) for i in top_idx}
+        phase1 = {"key_rows": [key_rows[i] for i in top_idx],
+                  "aggs": [aggs[i] for i in top_idx],
+                  "agg_ops": partial["agg_ops"]}
+        parts = exchange_json(phase1, self.device)
+        candidates = set()
+        for p in parts:
+            for key in p["key_rows"]:
+                candidates.add(tup(key))
+        local = {tup(kr): (kr, a) for kr, a in zip(key_rows, aggs)}
+        delta_keys = [c for c in candidates if c in local and c not in sent]
+        phase2 = {"key_rows": [local[c][0] for c in delta_keys],
+                  "aggs": [local[c][1] for c in delta_keys],
+                  "agg_ops": partial["agg_ops"]}
+        parts2 = exchange_json(phase2, self.device)
+        merged = merge_agg_partials(parts + parts2)
         return self.engine.finalize_groups(sql, merged["key_rows"],
                                            merged["aggs"])

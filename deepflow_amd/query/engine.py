@@ -401,6 +401,8 @@ class QueryEngine:
                     row.append(agg[ai])
                     ai += 1
             rows.append(row)
+        if plan.slimit:
+            rows = self._apply_slimit(plan, rows)
         rows = self._order_limit(plan, columns, rows)
         return {"columns": columns, "values": rows}
 

@@ -54,13 +54,22 @@ def _worker(rank, world, port, q):
             got[d][0] == want[d] and
             abs(got[d][1] - sum(rrt[d]) / len(rrt[d])) < 1e-6
             for d in want)
+        # topN pushdown: forced two-phase exchange must equal the full
+        # exchange for a SLIMIT query
+        sl = ("SELECT request_domain, Count(*) AS c FROM l7_flow_log "
+              "GROUP BY request_domain SLIMIT 3")
+        r_full = eng.query(sl)          # below threshold -> full exchange
+        eng.pushdown_threshold = 0      # force candidate/delta phases
+        r_push = eng.query(sl)
+        eng.pushdown_threshold = DistQueryEngine.pushdown_threshold
+        ok_topn = r_full == r_push and len(r_full["values"]) <= 3
         # metric bucket merge
         parts = exchange_json(pipe.metrics.rows(), "cpu")
         merged = merge_metric_rows(parts)
         total_req = sum(r["request"] for r in merged)
         dist.barrier()
         dist.destroy_process_group()
-        q.put((rank, ok_count, ok_groups, total_req))
+        q.put((rank, ok_count, ok_groups and ok_topn, total_req))
     except Exception:
         import traceback
         q.put((rank, "ERR", traceback.format_exc(), None))
