@@ -1,0 +1,249 @@
+"""gRPC (HTTP/2) framing for trident.Synchronizer.
+
+A minimal threaded HTTP/2 server speaking just enough of RFC 7540 + the
+gRPC wire protocol for the agent Sync path: connection preface, SETTINGS
+exchange, HPACK-decoded HEADERS (via the C decoder shared with the
+agent's h2 parser, ops/csrc/http2.h), DATA frames carrying
+[compressed u8][len u32 BE][protobuf] gRPC messages, and response
+HEADERS + DATA + trailers (grpc-status). Dispatches :path
+"/trident.Synchronizer/Sync" onto ControllerLite's version-gated sync.
+
+Reference counterpart: server/controller/trisolaris/server/grpc
+(the tonic/grpc Synchronizer service).
+"""
+from __future__ import annotations
+
+import ctypes as ct
+import socket
+import struct
+import threading
+from typing import Callable, Dict, List, Optional, Tuple
+
+from ..ops import native
+from ..wire import pb, trident
+
+F_DATA, F_HEADERS, F_SETTINGS, F_PING, F_GOAWAY, F_WINDOW = 0, 1, 4, 6, 7, 8
+FLAG_END_STREAM, FLAG_END_HEADERS, FLAG_ACK = 0x1, 0x4, 0x1
+PREFACE = b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n"
+
+
+class HpackDecoder:
+    """ctypes wrapper over the C HPACK decoder (stateful per connection)."""
+
+    def __init__(self):
+        self._lib = native.cpu()
+        self._lib.dfh2_hpack_new.restype = ct.c_void_p
+        self._lib.dfh2_hpack_new.argtypes = []
+        self._lib.dfh2_hpack_free.restype = None
+        self._lib.dfh2_hpack_free.argtypes = [ct.c_void_p]
+        self._lib.dfh2_hpack_decode.restype = ct.c_int64
+        self._lib.dfh2_hpack_decode.argtypes = [
+            ct.c_void_p, ct.c_char_p, ct.c_uint64, ct.c_void_p, ct.c_uint64]
+        self._h = self._lib.dfh2_hpack_new()
+
+    def decode(self, block: bytes) -> List[Tuple[str, str]]:
+        cap = 1 << 16
+        out = ct.create_string_buffer(cap)
+        n = self._lib.dfh2_hpack_decode(self._h, block, len(block), out, cap)
+        if n < 0:
+            raise ValueError("hpack decode failed")
+        parts = out.raw.split(b"\0")
+        return [(parts[2 * i].decode("utf-8", "replace"),
+                 parts[2 * i + 1].decode("utf-8", "replace"))
+                for i in range(n)]
+
+    def close(self):
+        if self._h:
+            self._lib.dfh2_hpack_free(self._h)
+            self._h = None
+
+
+def hpack_encode(headers: List[Tuple[str, str]]) -> bytes:
+    """Literal-without-indexing HPACK encoding (always legal, stateless)."""
+    out = bytearray()
+    for name, value in headers:
+        nb, vb = name.encode(), value.encode()
+        out.append(0x00)
+        out.append(len(nb))
+        out += nb
+        out.append(len(vb))
+        out += vb
+    return bytes(out)
+
+
+def frame(ftype: int, flags: int, stream_id: int, payload: bytes) -> bytes:
+    return struct.pack(">I", len(payload))[1:] + bytes([ftype, flags]) + \
+        struct.pack(">I", stream_id) + payload
+
+
+def grpc_message(pb_bytes: bytes) -> bytes:
+    return b"\x00" + struct.pack(">I", len(pb_bytes)) + pb_bytes
+
+
+class GrpcSyncServer:
+    """Serves trident.Synchronizer/Sync over real gRPC framing."""
+
+    def __init__(self, controller, host: str = "127.0.0.1", port: int = 0):
+        self.controller = controller
+        self.host = host
+        self.port = port
+        self._sock: Optional[socket.socket] = None
+        self._stop = threading.Event()
+        self._threads: List[threading.Thread] = []
+
+    # ----------------------------------------------------------- service
+    def _handle_sync(self, req: Dict) -> Dict:
+        """SyncRequest dict -> SyncResponse dict via ControllerLite.
+        Identity follows the reference's IP_AND_MAC agent identifier:
+        a stable id is derived from (ctrl_ip, ctrl_mac)."""
+        import zlib
+        ctl = self.controller
+        ident = (req.get("ctrl_ip", "") + "|" +
+                 req.get("ctrl_mac", "")).encode()
+        agent_id = (zlib.crc32(ident) % 64000) or 1
+        r = ctl.sync(agent_id=agent_id,
+                     hostname=req.get("host", ""),
+                     ip=req.get("ctrl_ip", ""),
+                     config_version=0,
+                     platform_version=req.get("version_platform_data", 0),
+                     exceptions=req.get("exception", 0))
+        resp = {
+            "status": trident.STATUS_SUCCESS,
+            "version_platform_data": r.get("platform_version",
+                                           req.get("version_platform_data",
+                                                   0)),
+            "version_acls": req.get("version_acls", 0),
+            "version_groups": req.get("version_groups", 0),
+        }
+        if "config" in r:
+            cfg = r["config"]
+            resp["config"] = {
+                "enabled": 1,
+                "sync_interval": cfg.get("sync_interval", 60),
+                "max_memory": cfg.get("max_memory", 768),
+                "vtap_id": cfg.get("vtap_id", 0),
+            }
+        if "platform" in r:
+            import ipaddress
+            pd = {"cidrs": [
+                {"prefix": f"{ipaddress.IPv4Address(e['ip'])}/32",
+                 "epc_id": e["epc"]} for e in r["platform"]]}
+            resp["platform_data"] = pb.encode(pd, trident.PLATFORM_DATA)
+        return resp
+
+    # ------------------------------------------------------------ server
+    def start(self) -> None:
+        self._sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        self._sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        self._sock.bind((self.host, self.port))
+        self.port = self._sock.getsockname()[1]
+        self._sock.listen(16)
+        self._sock.settimeout(0.5)
+        t = threading.Thread(target=self._accept_loop, daemon=True)
+        t.start()
+        self._threads.append(t)
+
+    def stop(self) -> None:
+        self._stop.set()
+        for t in self._threads:
+            t.join(timeout=2)
+        if self._sock:
+            self._sock.close()
+
+    def _accept_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                conn, _ = self._sock.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                return
+            t = threading.Thread(target=self._conn, args=(conn,),
+                                 daemon=True)
+            t.start()
+            self._threads.append(t)
+
+    def _recv_exact(self, conn, n: int) -> Optional[bytes]:
+        buf = b""
+        while len(buf) < n:
+            try:
+                chunk = conn.recv(n - len(buf))
+            except socket.timeout:
+                if self._stop.is_set():
+                    return None
+                continue
+            except OSError:
+                return None
+            if not chunk:
+                return None
+            buf += chunk
+        return buf
+
+    def _conn(self, conn: socket.socket) -> None:
+        conn.settimeout(1.0)
+        decoder = HpackDecoder()
+        streams: Dict[int, Dict] = {}
+        try:
+            pre = self._recv_exact(conn, len(PREFACE))
+            if pre != PREFACE:
+                return
+            conn.sendall(frame(F_SETTINGS, 0, 0, b""))
+            while not self._stop.is_set():
+                hdr = self._recv_exact(conn, 9)
+                if hdr is None:
+                    return
+                length = (hdr[0] << 16) | (hdr[1] << 8) | hdr[2]
+                ftype, flags = hdr[3], hdr[4]
+                sid = struct.unpack(">I", hdr[5:9])[0] & 0x7FFFFFFF
+                payload = self._recv_exact(conn, length) if length else b""
+                if payload is None:
+                    return
+                if ftype == F_SETTINGS and not flags & FLAG_ACK:
+                    conn.sendall(frame(F_SETTINGS, FLAG_ACK, 0, b""))
+                elif ftype == F_PING and not flags & FLAG_ACK:
+                    conn.sendall(frame(F_PING, FLAG_ACK, 0, payload))
+                elif ftype == F_HEADERS:
+                    off = 0
+                    if flags & 0x08:  # padded
+                        off += 1
+                    if flags & 0x20:  # priority
+                        off += 5
+                    headers = decoder.decode(payload[off:])
+                    st = streams.setdefault(sid, {"data": b""})
+                    st["headers"] = dict(headers)
+                    if flags & FLAG_END_STREAM:
+                        self._dispatch(conn, sid, st)
+                        streams.pop(sid, None)
+                elif ftype == F_DATA:
+                    st = streams.setdefault(sid, {"data": b""})
+                    st["data"] += payload
+                    if flags & FLAG_END_STREAM:
+                        self._dispatch(conn, sid, st)
+                        streams.pop(sid, None)
+                elif ftype == F_GOAWAY:
+                    return
+        finally:
+            decoder.close()
+            conn.close()
+
+    def _dispatch(self, conn, sid: int, st: Dict) -> None:
+        path = st.get("headers", {}).get(":path", "")
+        data = st.get("data", b"")
+        status = "0"
+        body = b""
+        if path == "/trident.Synchronizer/Sync" and len(data) >= 5:
+            mlen = struct.unpack(">I", data[1:5])[0]
+            try:
+                req = pb.decode(data[5:5 + mlen], trident.SYNC_REQUEST)
+                resp = self._handle_sync(req)
+                body = grpc_message(pb.encode(resp, trident.SYNC_RESPONSE))
+            except Exception:  # noqa: BLE001
+                status = "13"  # INTERNAL
+        else:
+            status = "12"  # UNIMPLEMENTED
+        conn.sendall(frame(F_HEADERS, FLAG_END_HEADERS, sid, hpack_encode([
+            (":status", "200"), ("content-type", "application/grpc")])))
+        if body:
+            conn.sendall(frame(F_DATA, 0, sid, body))
+        conn.sendall(frame(F_HEADERS, FLAG_END_HEADERS | FLAG_END_STREAM,
+                           sid, hpack_encode([("grpc-status", status)])))

@@ -2189,4 +2189,32 @@ void dfa_stats(void* h, uint64_t* out8) {
     out8[7] = 0;
 }
 
+// --------------------------------------------------------------- HPACK
+// Stateful HPACK decoder handles for the Python gRPC/HTTP-2 server
+// (control-plane trident.Synchronizer framing reuses the same RFC 7541
+// machinery the agent's h2 parser uses).
+void* dfh2_hpack_new() { return new h2::DynTable(); }
+void dfh2_hpack_free(void* h) { delete (h2::DynTable*)h; }
+
+// Decode one header block; writes "name\0value\0"... into out.
+// Returns the number of headers, or -1 on decode error / overflow.
+int64_t dfh2_hpack_decode(void* h, const uint8_t* p, uint64_t n,
+                          uint8_t* out, uint64_t cap) {
+    h2::DynTable& dyn = *(h2::DynTable*)h;
+    std::vector<h2::Header> hs;
+    if (!h2::hpack_decode(p, (uint32_t)n, dyn, hs)) return -1;
+    uint64_t w = 0;
+    for (auto& hd : hs) {
+        uint64_t need = hd.first.size() + hd.second.size() + 2;
+        if (w + need > cap) return -1;
+        memcpy(out + w, hd.first.data(), hd.first.size());
+        w += hd.first.size();
+        out[w++] = 0;
+        memcpy(out + w, hd.second.data(), hd.second.size());
+        w += hd.second.size();
+        out[w++] = 0;
+    }
+    return (int64_t)hs.size();
+}
+
 }  // extern "C"
