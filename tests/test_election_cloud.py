@@ -1,0 +1,69 @@
+"""Leader election (lease-file takeover) + cloud poller diff-push tests."""
+import ipaddress
+
+from deepflow_amd.control import ControllerLite
+from deepflow_amd.control.cloud import CloudPoller, k8s_snapshot_to_platform
+from deepflow_amd.control.election import LeaderElector
+
+
+def test_election_takeover(tmp_path):
+    lease = str(tmp_path / "leader.lease")
+    a = LeaderElector(lease, "server-a", ttl_s=5)
+    b = LeaderElector(lease, "server-b", ttl_s=5)
+    t = 1000.0
+    assert a.campaign(t) is True
+    assert b.campaign(t + 1) is False       # lease held by a
+    assert b.leader() is None or b._read()["leader"] == "server-a"
+    assert a.campaign(t + 2) is True        # renewal
+    # a dies; lease expires; b takes over
+    assert b.campaign(t + 20) is True
+    assert a.campaign(t + 21) is False      # a follows the new leader
+    b.resign()
+    assert a.campaign(t + 22) is True       # immediate re-election
+
+
+SNAP1 = {
+    "cluster_id": 3, "cluster_name": "prod",
+    "pods": [
+        {"id": 11, "name": "web-0", "ip": "10.0.0.5", "namespace": "shop",
+         "node_id": 2, "epc": 1},
+        {"id": 12, "name": "db-0", "ip": "10.0.0.6", "namespace": "shop",
+         "node_id": 2, "epc": 1},
+    ],
+    "nodes": [{"id": 2, "name": "node-2", "ip": "10.0.1.2", "epc": 1}],
+    "services": [{"id": 5, "name": "web-svc", "cluster_ip": "10.96.0.10",
+                  "epc": 1}],
+}
+
+
+def test_k8s_normalize():
+    entries, names = k8s_snapshot_to_platform(SNAP1)
+    key = (1, int(ipaddress.IPv4Address("10.0.0.5")))
+    assert entries[key].pod_id == 11
+    assert entries[key].pod_cluster_id == 3
+    assert names["pod"][11] == "web-0"
+    assert names["pod_ns"][entries[key].pod_ns_id] == "shop"
+    assert names["service"][5] == "web-svc"
+    svc_key = (1, int(ipaddress.IPv4Address("10.96.0.10")))
+    assert entries[svc_key].service_id == 5
+
+
+def test_cloud_poller_diff_push():
+    ctl = ControllerLite()
+    snap = {"holder": dict(SNAP1)}
+    poller = CloudPoller(ctl, source=lambda: snap["holder"])
+    v0 = ctl.platform_version
+    assert poller.poll_once() is True
+    assert ctl.platform_version == v0 + 1
+    assert poller.poll_once() is False      # unchanged snapshot: no push
+    assert ctl.platform_version == v0 + 1
+    snap2 = dict(SNAP1)
+    snap2["pods"] = SNAP1["pods"] + [
+        {"id": 13, "name": "web-1", "ip": "10.0.0.7", "namespace": "shop",
+         "node_id": 2, "epc": 1}]
+    snap["holder"] = snap2
+    assert poller.poll_once() is True       # pod added -> version bump
+    assert ctl.platform_version == v0 + 2
+    assert poller.polls == 3 and poller.pushes == 2
+    # tagrecorder name maps landed
+    assert ctl.name_maps["pod"][13] == "web-1"
