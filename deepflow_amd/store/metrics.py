@@ -40,7 +40,26 @@ def unpack_net_key(key: int, time_base_s: int):
     }
 
 
-class Net1sMetrics:
+
+
+class _MetricsCkpt:
+    """Checkpoint mixin shared by the 1s rollup tables."""
+
+    def state_dict(self):
+        if self.device == "cpu":
+            return {"table": {k: list(v) for k, v in self.table.items()}}
+        return {"tkeys": self.tkeys.cpu().clone(),
+                "tvals": self.tvals.cpu().clone()}
+
+    def load_state_dict(self, st):
+        if self.device == "cpu":
+            self.table.update(st["table"])
+        else:
+            self.tkeys.copy_(st["tkeys"].to(self.tkeys.device))
+            self.tvals.copy_(st["tvals"].to(self.tvals.device))
+
+
+class Net1sMetrics(_MetricsCkpt):
     """network.1s rollup (K5b output; reference flow_metrics network table)."""
 
     def __init__(self, time_base_s: int, capacity_pow2: int = 1 << 20,
@@ -78,7 +97,7 @@ class Net1sMetrics:
         return out
 
 
-class App1sMetrics:
+class App1sMetrics(_MetricsCkpt):
     def __init__(self, time_base_s: int, capacity_pow2: int = 1 << 20,
                  device: str = "cpu"):
         assert capacity_pow2 & (capacity_pow2 - 1) == 0
