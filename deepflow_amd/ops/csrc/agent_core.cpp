@@ -64,9 +64,13 @@ struct FlowNode {
     uint64_t syn_ts = 0, synack_ts = 0;
     uint32_t rtt_us = 0;
     uint32_t syn_count = 0, synack_count = 0;
-    uint64_t last_req_pkt_ts = 0;  // for SRT
-    uint32_t srt_sum = 0, srt_cnt = 0, srt_max = 0;
-    uint32_t art_sum = 0, art_cnt = 0, art_max = 0;
+    uint64_t last_req_pkt_ts = 0;  // for ART (last request data pkt)
+    uint32_t srt_sum = 0, srt_cnt = 0, srt_max = 0;   // data -> pure ACK
+    uint32_t art_sum = 0, art_cnt = 0, art_max = 0;   // req data -> resp data
+    uint32_t cit_sum = 0, cit_cnt = 0, cit_max = 0;   // resp end -> next req
+    uint32_t ack_wait_seq = 0;      // client data awaiting server ACK
+    uint64_t ack_wait_ts = 0;
+    uint64_t last_resp_pkt_ts = 0;  // for CIT
     bool fin_seen[2] = {false, false};
     bool rst = false;
     uint8_t close_type = 0;
@@ -1175,11 +1179,17 @@ void encode_l4_record(Agent& a, FlowNode& f) {
             dfpb::f_m<512>(fl, 13, [&](Buf& p) {  // FlowPerfStats
                 dfpb::f_m<256>(p, 1, [&](Buf& t) {  // TCP
                     dfpb::f_u(t, 3, f.srt_max);
+                    dfpb::f_u(t, 4, f.art_max);
                     dfpb::f_u(t, 5, f.rtt_us);
                     dfpb::f_u(t, 8, f.srt_sum);
+                    dfpb::f_u(t, 9, f.art_sum);
                     dfpb::f_u(t, 12, f.srt_cnt);
+                    dfpb::f_u(t, 13, f.art_cnt);
                     dfpb::f_u(t, 17, f.syn_count);
                     dfpb::f_u(t, 18, f.synack_count);
+                    dfpb::f_u(t, 19, f.cit_max);
+                    dfpb::f_u(t, 20, f.cit_sum);
+                    dfpb::f_u(t, 21, f.cit_cnt);
                 });
                 if (f.l7c.response_count || f.l7c.request_count) {
                     dfpb::f_m<128>(p, 2, [&](Buf& l) {  // L7PerfStats
@@ -2121,14 +2131,42 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
         if (fin) f.fin_seen[dir] = true;
         if (rst) { f.rst = true; f.close_type = 2; }
         if (f.fin_seen[0] && f.fin_seen[1] && !f.close_type) f.close_type = 1;
-        // SRT: first server packet with payload after a client request
-        if (dir == 1 && paylen > 0 && f.last_req_pkt_ts &&
-            ts_ns > f.last_req_pkt_ts) {
-            uint32_t srt = (uint32_t)((ts_ns - f.last_req_pkt_ts) / 1000);
+        uint32_t ackno = (l4[8] << 24) | (l4[9] << 16) | (l4[10] << 8) |
+                         l4[11];
+        if (dir == 0 && paylen > 0) {
+            // client data: arm the SRT ack-watch, close a CIT interval
+            f.ack_wait_seq = seq + paylen;
+            f.ack_wait_ts = ts_ns;
+            if (f.last_resp_pkt_ts && ts_ns > f.last_resp_pkt_ts) {
+                uint32_t cit = (uint32_t)((ts_ns - f.last_resp_pkt_ts) /
+                                          1000);
+                f.cit_sum += cit;
+                f.cit_cnt++;
+                if (cit > f.cit_max) f.cit_max = cit;
+                f.last_resp_pkt_ts = 0;
+            }
+        }
+        if (dir == 1 && ack && f.ack_wait_seq &&
+            (int32_t)(ackno - f.ack_wait_seq) >= 0 &&
+            ts_ns > f.ack_wait_ts) {
+            // SRT: client data packet -> the server ACK covering it
+            uint32_t srt = (uint32_t)((ts_ns - f.ack_wait_ts) / 1000);
             f.srt_sum += srt;
             f.srt_cnt++;
             if (srt > f.srt_max) f.srt_max = srt;
-            f.last_req_pkt_ts = 0;
+            f.ack_wait_seq = 0;
+        }
+        if (dir == 1 && paylen > 0) {
+            // ART: last request data packet -> first response data packet
+            if (f.last_req_pkt_ts && ts_ns > f.last_req_pkt_ts) {
+                uint32_t art = (uint32_t)((ts_ns - f.last_req_pkt_ts) /
+                                          1000);
+                f.art_sum += art;
+                f.art_cnt++;
+                if (art > f.art_max) f.art_max = art;
+                f.last_req_pkt_ts = 0;
+            }
+            f.last_resp_pkt_ts = ts_ns;
         }
     }
     if (paylen > 0) handle_l7_payload(a, f, dir, payload, paylen, ts_ns);

@@ -29,3 +29,56 @@ def test_agent_dfstats_to_server():
     assert rows[0]["packets"] >= 4
     a.close()
     srv.stop()
+
+
+def test_srt_art_cit_metrics():
+    """TCP perf triad: SRT (data->ACK), ART (req data->resp data),
+    CIT (resp end->next req) on a two-request keep-alive flow."""
+    import struct
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import (eth_ipv4_tcp, SYN, SYNACK,
+                                            PSH_ACK)
+    from deepflow_amd.wire import pb, flow_log, framing
+    ACK = 0x10
+    C, S = 0x0A000001, 0x0A000002
+    a = Agent(vtap_id=2)
+    ms = 10**6
+    t = 10**9
+    req1 = b"GET /a HTTP/1.1\r\nHost: x\r\n\r\n"
+    resp1 = b"HTTP/1.1 200 OK\r\nContent-Length: 0\r\n\r\n"
+    req2 = b"GET /b HTTP/1.1\r\nHost: x\r\n\r\n"
+    resp2 = b"HTTP/1.1 200 OK\r\nContent-Length: 0\r\n\r\n"
+    seq_c, seq_s = 100, 500
+    pkts = [
+        (eth_ipv4_tcp(C, S, 4000, 80, SYN, seq_c), t),
+        (eth_ipv4_tcp(S, C, 80, 4000, SYNACK, seq_s, seq_c + 1), t + ms),
+        # req1 at t+2ms
+        (eth_ipv4_tcp(C, S, 4000, 80, PSH_ACK, seq_c + 1, seq_s + 1, req1),
+         t + 2 * ms),
+        # pure ACK from server 3ms later -> SRT = 3ms
+        (eth_ipv4_tcp(S, C, 80, 4000, ACK, seq_s + 1,
+                      seq_c + 1 + len(req1)), t + 5 * ms),
+        # response data 8ms after req1 -> ART = 8ms
+        (eth_ipv4_tcp(S, C, 80, 4000, PSH_ACK, seq_s + 1,
+                      seq_c + 1 + len(req1), resp1), t + 10 * ms),
+        # second request 30ms after the response -> CIT = 30ms
+        (eth_ipv4_tcp(C, S, 4000, 80, PSH_ACK, seq_c + 1 + len(req1),
+                      seq_s + 1 + len(resp1), req2), t + 40 * ms),
+        (eth_ipv4_tcp(S, C, 80, 4000, PSH_ACK, seq_s + 1 + len(resp1),
+                      seq_c + 1 + len(req1) + len(req2), resp2),
+         t + 44 * ms),
+    ]
+    for frame, ts in pkts:
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    flows = [pb.decode(r, flow_log.TAGGED_FLOW)
+             for r in framing.iter_records(a.drain(0))]
+    tcp = flows[0]["flow"]["perf_stats"]["tcp"]
+    assert tcp["rtt"] == 1000                     # handshake 1ms
+    assert tcp["srt_max"] == 4000  # max(3ms pure-ACK, 4ms piggyback)
+    assert tcp["art_max"] == 8000                 # req data -> resp data
+    assert tcp["cit_max"] == 30000                # idle gap before req2
+    assert tcp["srt_count"] == 2 and tcp["art_count"] == 2
+    assert tcp["srt_sum"] == 7000
+    assert tcp["cit_count"] == 1
+    a.close()

@@ -104,3 +104,31 @@ def test_pool_contents_match(pipes):
 def test_metrics_match(pipes):
     cpu, gpu = pipes
     assert cpu.metrics.rows() == gpu.metrics.rows()
+
+
+@pytest.mark.gpu
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    """Checkpoint a GPU pipeline, restore into a fresh one, queries match."""
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query.engine import QueryEngine
+    from deepflow_amd.store import checkpoint as CK
+    cfg = SpanGenConfig(n=5000, seed=77, tag_cardinality=64, n_attrs=2,
+                        n_ips=64, n_services=4, n_resources=16)
+    pipe = L7IngestPipeline(device="cuda", segment_rows=1 << 13,
+                            dict_capacity=1 << 13,
+                            time_base_s=cfg.base_time_ns // 10**9)
+    pipe.ingest_frame_payload(gen_span_payload(cfg))
+    eng = QueryEngine(pipe, device="cuda")
+    q = ("SELECT request_resource, Count(*) AS c, Avg(response_duration) "
+         "AS a FROM l7_flow_log GROUP BY request_resource ORDER BY c DESC")
+    want = eng.query(q)
+    path = str(tmp_path / "gpu.ckpt")
+    CK.save_l7(pipe, path)
+    fresh = L7IngestPipeline(device="cuda", segment_rows=1 << 13,
+                             dict_capacity=1 << 13,
+                             time_base_s=cfg.base_time_ns // 10**9)
+    assert CK.load_l7(fresh, path) == cfg.n
+    got = QueryEngine(fresh, device="cuda").query(q)
+    assert want == got
