@@ -127,6 +127,8 @@ class QueryEngine:
         self.l4 = l4_pipeline
         self.device = device
         self.remote = remote_hydrator
+        # tagrecorder name maps (id -> display name), set by the server
+        self.name_maps = {}
 
     # ----------------------------------------------------------- dispatch
     def query(self, sql: str, _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
@@ -159,7 +161,8 @@ class QueryEngine:
         if table in ("l7_flow_log", "l7_flow_log.l7_flow_log"):
             plan = parse_sql(sql, dictionary=self.pipe.dict,
                              time_base_s=self.pipe.time_base_s,
-                             tags=L7_TAGS, metrics=L7_METRICS)
+                             tags=L7_TAGS, metrics=L7_METRICS,
+                             name_maps=self.name_maps)
             return self._run_segments(plan, self.pipe.segments.scan_list(),
                                       L7_TAGS, S.STR_COLS)
         if table == "l4_flow_log":
@@ -167,7 +170,8 @@ class QueryEngine:
                 raise SqlError("l4_flow_log table not enabled")
             plan = parse_sql(sql, dictionary=None,
                              time_base_s=self.l4.time_base_s,
-                             tags=L4_TAGS, metrics=L4_METRICS)
+                             tags=L4_TAGS, metrics=L4_METRICS,
+                             name_maps=self.name_maps)
             from ..store import l4_schema as L4S
             return self._run_segments(plan, self.l4.segments.scan_list(),
                                       L4_TAGS, L4S.STR_COLS)
@@ -346,14 +350,16 @@ class QueryEngine:
         if table == "l7_flow_log":
             plan = parse_sql(sql, dictionary=self.pipe.dict,
                              time_base_s=self.pipe.time_base_s,
-                             tags=L7_TAGS, metrics=L7_METRICS)
+                             tags=L7_TAGS, metrics=L7_METRICS,
+                             name_maps=self.name_maps)
             segments, tags, str_cols = (self.pipe.segments.scan_list(), L7_TAGS,
                                         S.STR_COLS)
         else:
             from ..store import l4_schema as L4S
             plan = parse_sql(sql, dictionary=None,
                              time_base_s=self.l4.time_base_s,
-                             tags=L4_TAGS, metrics=L4_METRICS)
+                             tags=L4_TAGS, metrics=L4_METRICS,
+                             name_maps=self.name_maps)
             segments, tags, str_cols = (self.l4.segments.scan_list(), L4_TAGS,
                                         L4S.STR_COLS)
         if plan.select_rows:
@@ -382,11 +388,13 @@ class QueryEngine:
         if table == "l4_flow_log":
             plan = parse_sql(sql, dictionary=None,
                              time_base_s=self.l4.time_base_s,
-                             tags=L4_TAGS, metrics=L4_METRICS)
+                             tags=L4_TAGS, metrics=L4_METRICS,
+                             name_maps=self.name_maps)
         else:
             plan = parse_sql(sql, dictionary=self.pipe.dict,
                              time_base_s=self.pipe.time_base_s,
-                             tags=L7_TAGS, metrics=L7_METRICS)
+                             tags=L7_TAGS, metrics=L7_METRICS,
+                             name_maps=self.name_maps)
         columns = plan.key_names + plan.agg_names
         rows = []
         for key, agg in zip(key_rows, aggs):
@@ -543,7 +551,7 @@ class QueryEngine:
                 return self._hydrate(td.hydrate,
                                      int(seg.did[idx, row]) & 0xFFFFFFFF)
             if fam == Q.SRC_KG:
-                return int(seg.kg[idx, row])
+                return self._hydrate(td.hydrate, int(seg.kg[idx, row]))
         if col in str_cols:
             # pooled string columns: row block ref + per-col u16 lens
             # (dict-encoded string tags were already handled via SRC_DID)
@@ -674,6 +682,9 @@ class QueryEngine:
             return L7_PROTOCOL_NAMES.get(v, str(v))
         if how == "status":
             return STATUS_NAMES.get(v, str(v))
+        if how.startswith("kgname:"):
+            mp = how.split(":", 1)[1]
+            return self.name_maps.get(mp, {}).get(v, str(v))
         return v
 
     def _order_limit(self, plan: Q.Plan, columns, rows):

@@ -93,7 +93,7 @@ def _resolve_tag(name: str, tags) -> TagDef:
 
 
 def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
-              tags=None, metrics=None) -> Q.Plan:
+              tags=None, metrics=None, name_maps=None) -> Q.Plan:
     """Parse DF-SQL into a Plan against a table's tag map (default:
     l7_flow_log). `dictionary` compiles string literals on dict tags to
     SmartEncoding IDs (None -> impossible filters)."""
@@ -169,7 +169,7 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
                     op_t = p.next()
                     lit = p.next()
                     _add_term(plan, name[1], op_t[1], lit, dictionary, tags,
-                              group=group_counter)
+                              group=group_counter, name_maps=name_maps)
                     if p.kw_is("or"):
                         p.next()
                         continue
@@ -189,7 +189,8 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
                     while True:
                         lit = p.next()
                         _add_term(plan, name[1], "=", lit, dictionary, tags,
-                                  group=group_counter)
+                                  group=group_counter,
+                                  name_maps=name_maps)
                         if p.peek() == ("op", ","):
                             p.next()
                             continue
@@ -201,7 +202,8 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
                     if op_t[0] != "op" or op_t[1] not in Q.OP_BY_NAME:
                         raise SqlError(f"bad operator {op_t!r}")
                     lit = p.next()
-                    _add_term(plan, name[1], op_t[1], lit, dictionary, tags)
+                    _add_term(plan, name[1], op_t[1], lit, dictionary, tags,
+                              name_maps=name_maps)
             if p.kw_is("and"):
                 p.next()
                 continue
@@ -325,7 +327,7 @@ def _never(plan: Q.Plan, group: int) -> None:
 
 
 def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
-              tags, group: int = 0) -> None:
+              tags, group: int = 0, name_maps=None) -> None:
     if name.startswith("attribute."):
         # custom-tag filter: both sides resolved to SmartEncoding ids
         # (reference: flow_tag custom_field_value filters)
@@ -392,6 +394,18 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
         from ..wire.const_enums import L7_PROTOCOL_NAMES
         rev = {v.lower(): k for k, v in L7_PROTOCOL_NAMES.items()}
         ident = rev.get(lit[1].lower())
+        if ident is None:
+            if group:
+                _never(plan, group)
+            else:
+                plan.impossible = True
+            return
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], ident,
+                                 group=group))
+    elif td.hydrate.startswith("kgname:"):
+        mp = td.hydrate.split(":", 1)[1]
+        rev = {v: k for k, v in (name_maps or {}).get(mp, {}).items()}
+        ident = rev.get(lit[1])
         if ident is None:
             if group:
                 _never(plan, group)

@@ -96,6 +96,23 @@ def build_l7_tags() -> Dict[str, TagDef]:
     # single-ended metrics tables)
     for j, kname in enumerate(S.KG_COLS):
         tags.setdefault(kname, TagDef(kname, Q.SRC_KG, S.N_KG + j))
+    # id -> display-name tags via tagrecorder name maps (reference:
+    # ClickHouse dictGet on the *_map dictionaries)
+    KG_NAME_MAPS = {"pod_id": "pod", "pod_node_id": "pod_node",
+                    "pod_ns_id": "pod_ns", "pod_group_id": "pod_group",
+                    "pod_cluster_id": "pod_cluster",
+                    "service_id": "service"}
+    for j, kname in enumerate(S.KG_COLS):
+        mp = KG_NAME_MAPS.get(kname)
+        if mp is None:
+            continue
+        base = kname[:-3]  # strip _id
+        for side in (0, 1):
+            add(TagDef(f"{base}_name_{side}", Q.SRC_KG,
+                       side * S.N_KG + j, hydrate=f"kgname:{mp}"))
+        tags.setdefault(f"{base}_name",
+                        TagDef(f"{base}_name", Q.SRC_KG, S.N_KG + j,
+                               hydrate=f"kgname:{mp}"))
     # time pseudo-tag handled by the parser (SRC_TIME_BUCKET)
     return tags
 

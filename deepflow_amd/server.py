@@ -110,6 +110,9 @@ class DeepflowServer:
         self.system_rows = []  # deepflow_system self-metrics store
         self.receiver.register(framing.MSG_DFSTATS, self._on_dfstats)
         self.engine.system_rows = self.system_rows
+        # tagrecorder hydration: the engine resolves KG ids to display
+        # names through the controller's name maps (live reference)
+        self.engine.name_maps = self.controller.name_maps
         self.engine.agent_app_rows = lambda: self.docs.app_rows
         self.engine.agent_net_rows = lambda: self.docs.net_rows
         self.engine.event_rows = lambda: self.events.resource_events

@@ -67,3 +67,44 @@ def test_cloud_poller_diff_push():
     assert poller.polls == 3 and poller.pushes == 2
     # tagrecorder name maps landed
     assert ctl.name_maps["pod"][13] == "web-1"
+
+
+def test_pod_name_hydration_and_filter():
+    """KG ids hydrate to display names in SQL results (dictGet analog),
+    and name literals compile back to id filters."""
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.wire import pb, flow_log, framing
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12)
+    poller = CloudPoller(srv.controller, source=lambda: SNAP1)
+    poller.poll_once()
+
+    def span(dst_ip, res):
+        return {"base": {"start_time": 10**18, "end_time": 10**18 + 10**6,
+                         "flow_id": 1, "vtap_id": 1, "tap_side": 1,
+                         "head": {"proto": 20, "msg_type": 2, "rrt": 10},
+                         "ip_src": 0x0A00000A,
+                         "ip_dst": int(ipaddress.IPv4Address(dst_ip)),
+                         "l3_epc_id_src": 1, "l3_epc_id_dst": 1,
+                         "port_src": 999, "port_dst": 80, "protocol": 6},
+                "req": {"req_type": "GET", "domain": "d", "resource": res,
+                        "endpoint": res},
+                "resp": {"status": 0, "code": 200}}
+
+    spans = [span("10.0.0.5", "/to-web")] * 3 + [span("10.0.0.6", "/to-db")]
+    srv.receiver.handle_frame(framing.encode_frame(
+        framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
+        framing.pack_records([pb.encode(s, flow_log.APP_PROTO_LOGS_DATA)
+                              for s in spans])))
+    r = srv.engine.query(
+        "SELECT pod_name_1, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY pod_name_1 ORDER BY c DESC")
+    assert r["values"] == [["web-0", 3], ["db-0", 1]]
+    # filter by display name -> compiled to the pod id
+    r2 = srv.engine.query(
+        "SELECT Count(*) AS c FROM l7_flow_log WHERE pod_name_1 = 'db-0'")
+    assert r2["values"] == [[1]]
+    r3 = srv.engine.query(
+        "SELECT Count(*) AS c FROM l7_flow_log "
+        "WHERE pod_name_1 = 'no-such-pod'")
+    assert r3["values"] in ([[0]], [])
