@@ -2072,7 +2072,12 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
         ? ((l4[4] << 24) | (l4[5] << 16) | (l4[6] << 8) | l4[7]) : 0;
     const uint8_t* payload = l4 + l4hdr;
     uint32_t paylen = tot > ihl + l4hdr ? tot - ihl - l4hdr : 0;
-    if (payload + paylen > pkt + len) paylen = (uint32_t)(pkt + len - payload);
+    // header fields may lie (fuzzed/corrupt frames): a data offset past
+    // the captured frame must clamp to zero, not wrap negative
+    if (payload >= pkt + len)
+        paylen = 0;
+    else if (payload + paylen > pkt + len)
+        paylen = (uint32_t)(pkt + len - payload);
 
     // canonical key: (lower (ip,port)) first
     bool a_first = (src < dst) || (src == dst && sport <= dport);
