@@ -31,6 +31,8 @@ def _lib():
         lib.dfa_new.argtypes = [u32]
         lib.dfa_free.argtypes = [p]
         lib.dfa_add_cidr.argtypes = [p, u32, u32, ct.c_int32]
+        lib.dfa_add_acl.argtypes = [p, u32, u32, u32, u32, u32, u32, u32,
+                                    u32, u32]
         lib.dfa_add_custom_port.argtypes = [p, u32]
         lib.dfa_packet.restype = ct.c_int
         lib.dfa_packet.argtypes = [p, p, u32, u64]
@@ -65,6 +67,16 @@ class Agent:
 
     def add_cidr(self, net: int, masklen: int, epc: int) -> None:
         self._lib.dfa_add_cidr(self._h, net, masklen, epc)
+
+    def add_acl(self, gid: int, src_net: int = 0, src_masklen: int = 0,
+                dst_net: int = 0, dst_masklen: int = 0, proto: int = 0,
+                port_min: int = 0, port_max: int = 65535,
+                action: int = 0) -> None:
+        """FlowAcl rule (policy/labeler analog): matched once per new
+        flow, cached on the flow, gids emitted in TaggedFlow acl_gids."""
+        self._lib.dfa_add_acl(self._h, gid, src_net, src_masklen, dst_net,
+                              dst_masklen, proto, port_min, port_max,
+                              action)
 
     def add_custom_protocol_port(self, port: int) -> None:
         """Port-rule custom protocol (L7 proto 127); raw sessions are

@@ -85,6 +85,8 @@ struct FlowNode {
     bool is_v6 = false;
     uint8_t ip6[2][16] = {{0}, {0}};  // client/server IPv6 addresses
     std::vector<uint8_t> h2_carry[2];  // cross-segment frame reassembly
+    std::vector<uint32_t> acl_gids;    // fast-path cached ACL matches
+    uint32_t acl_actions = 0;
 };
 
 struct FlowKeyC {
@@ -134,18 +136,44 @@ struct AppMeterAcc {
 
 struct Cidr { uint32_t net; uint32_t mask; int32_t epc; };
 
+// FlowAcl rule (reference policy/labeler: DDBS first path; matched once
+// per flow, result cached on the FlowNode = the LRU fast-path analog)
+struct AclRule {
+    uint32_t gid;            // acl group id, emitted in acl_gids
+    uint32_t src_net = 0, src_mask = 0, dst_net = 0, dst_mask = 0;
+    uint8_t proto = 0;       // 0 = any
+    uint16_t port_min = 0, port_max = 65535;  // server port range
+    uint32_t action = 0;     // bit0: pcap-capture tag
+};
+
 struct Agent {
     uint32_t vtap_id;
     uint64_t next_flow_id = 1;
     std::unordered_map<FlowKeyC, FlowNode, FlowKeyHash> flows;
     std::vector<Cidr> cidrs;
     std::vector<uint16_t> custom_ports;  // port-rule custom protocols (127)
+    std::vector<AclRule> acls;
     std::map<MeterKey, AppMeterAcc> meters;
     std::vector<uint8_t> out_l4, out_l7, out_doc;
     // stats
     uint64_t pkts = 0, bytes = 0, flows_emitted = 0, l7_emitted = 0,
              docs_emitted = 0, parse_errors = 0;
 };
+
+// first-path ACL match for a new flow (client = peer 0)
+void match_acls(const Agent& a, FlowNode& f) {
+    for (const auto& r : a.acls) {
+        bool fwd = (f.ip[0] & r.src_mask) == r.src_net &&
+                   (f.ip[1] & r.dst_mask) == r.dst_net;
+        bool rev = (f.ip[1] & r.src_mask) == r.src_net &&
+                   (f.ip[0] & r.dst_mask) == r.dst_net;
+        if (!fwd && !rev) continue;
+        if (r.proto && r.proto != f.proto) continue;
+        if (f.port[1] < r.port_min || f.port[1] > r.port_max) continue;
+        if (f.acl_gids.size() < 8) f.acl_gids.push_back(r.gid);
+        f.acl_actions |= r.action;
+    }
+}
 
 int32_t lookup_epc(const Agent& a, uint32_t ip) {
     for (const auto& c : a.cidrs)
@@ -1173,6 +1201,7 @@ void encode_l4_record(Agent& a, FlowNode& f) {
         dfpb::f_u(fl, 7, f.last_ns);
         dfpb::f_u(fl, 8, f.last_ns - f.start_ns);
         dfpb::f_u(fl, 11, f.is_v6 ? 0x86DD : 0x0800);
+        for (uint32_t g : f.acl_gids) dfpb::f_u(fl, 24, g);
         bool has_perf = f.rtt_us || f.srt_cnt || f.l7c.response_count;
         dfpb::f_u(fl, 12, has_perf ? 1 : 0);
         if (has_perf) {
@@ -1991,6 +2020,24 @@ void dfa_add_custom_port(void* h, uint32_t port) {
     ((Agent*)h)->custom_ports.push_back((uint16_t)port);
 }
 
+void dfa_add_acl(void* h, uint32_t gid, uint32_t src_net,
+                 uint32_t src_masklen, uint32_t dst_net,
+                 uint32_t dst_masklen, uint32_t proto, uint32_t port_min,
+                 uint32_t port_max, uint32_t action) {
+    Agent* a = (Agent*)h;
+    AclRule r;
+    r.gid = gid;
+    r.src_mask = src_masklen == 0 ? 0 : ~0u << (32 - src_masklen);
+    r.dst_mask = dst_masklen == 0 ? 0 : ~0u << (32 - dst_masklen);
+    r.src_net = src_net & r.src_mask;
+    r.dst_net = dst_net & r.dst_mask;
+    r.proto = (uint8_t)proto;
+    r.port_min = (uint16_t)port_min;
+    r.port_max = (uint16_t)port_max;
+    r.action = action;
+    a->acls.push_back(r);
+}
+
 void dfa_add_cidr(void* h, uint32_t net, uint32_t masklen, int32_t epc) {
     Agent* a = (Agent*)h;
     uint32_t mask = masklen == 0 ? 0 : ~0u << (32 - masklen);
@@ -2110,6 +2157,7 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
             memcpy(f.ip6[0], v6src, 16);
             memcpy(f.ip6[1], v6dst, 16);
         }
+        match_acls(a, f);
         it = a.flows.emplace(key, std::move(f)).first;
     }
     FlowNode& f = it->second;

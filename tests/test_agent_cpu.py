@@ -82,3 +82,30 @@ def test_srt_art_cit_metrics():
     assert tcp["srt_sum"] == 7000
     assert tcp["cit_count"] == 1
     a.close()
+
+
+def test_acl_policy_labeler():
+    """FlowAcl rules: first-path match on flow creation, gids on the
+    TaggedFlow; non-matching flows carry none."""
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import http_session
+    from deepflow_amd.wire import pb, flow_log, framing
+    a = Agent(vtap_id=1)
+    # gid 7: traffic to 10.0.0.2:80/tcp ; gid 9: any traffic from 10.0.0.0/8
+    a.add_acl(7, dst_net=0x0A000002, dst_masklen=32, proto=6,
+              port_min=8080, port_max=8080)
+    a.add_acl(9, src_net=0x0A000000, src_masklen=8)
+    a.add_acl(5, dst_net=0xC0A80000, dst_masklen=16)  # no match expected
+    for frame, ts in http_session(0x0A000001, 0x0A000002, sport=41000,
+                                  path="/x", code=200, t0=10**9):
+        a.packet(frame, ts)
+    for frame, ts in http_session(0x0A000001, 0x0A000003, sport=41001,
+                                  path="/y", code=200, t0=10**9):
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    flows = [pb.decode(r, flow_log.TAGGED_FLOW)
+             for r in framing.iter_records(a.drain(0))]
+    by_dst = {f["flow"]["flow_key"]["ip_dst"]: f["flow"] for f in flows}
+    assert sorted(by_dst[0x0A000002].get("acl_gids", [])) == [7, 9]
+    assert by_dst[0x0A000003].get("acl_gids", []) == [9]
+    a.close()
