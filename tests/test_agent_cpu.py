@@ -109,3 +109,40 @@ def test_acl_policy_labeler():
     assert sorted(by_dst[0x0A000002].get("acl_gids", [])) == [7, 9]
     assert by_dst[0x0A000003].get("acl_gids", []) == [9]
     a.close()
+
+
+def test_agent_config_template_and_diffs(tmp_path):
+    """template.yaml-style defaults + controller-pushed diffs firing
+    registered callbacks for changed keys only."""
+    import yaml as _yaml
+    from deepflow_amd.agent.config import AgentConfig, TEMPLATE
+    path = tmp_path / "agent.yaml"
+    path.write_text(_yaml.safe_dump({"max_memory": 1024,
+                                     "custom_protocol_ports": [9999]}))
+    cfg = AgentConfig.load(str(path))
+    assert cfg.get("max_memory") == 1024          # file overrides template
+    assert cfg.get("sync_interval") == TEMPLATE["sync_interval"]
+
+    fired = []
+    cfg.on_change("max_", lambda k, old, new: fired.append((k, old, new)))
+    changed = cfg.apply({"max_memory": 2048, "sync_interval": 60,
+                         "stats_interval": 5}, version=3)
+    assert sorted(changed) == ["max_memory", "stats_interval"]
+    assert fired == [("max_memory", 1024, 2048)]  # prefix-filtered
+    assert cfg.version == 3
+    assert cfg.apply({"max_memory": 2048}) == []  # no-op push
+
+    # actionable keys flow into a live agent
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import http_session
+    a = Agent(vtap_id=1)
+    cfg.configure_agent(a)
+    for frame, ts in http_session(0x0A000001, 0x0A000002, dport=9999,
+                                  t0=10**9):
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    from deepflow_amd.wire import pb, flow_log, framing
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(a.drain(1))]
+    assert recs and recs[0]["base"]["head"]["proto"] == 127  # custom port
+    a.close()
