@@ -224,3 +224,66 @@ class ControllerLite:
         @app.get("/v1/genesis/{agent_id}")
         def genesis_get(agent_id: int):
             return self.genesis_inventory.get(agent_id, {})
+
+        # -------- CRUD: domains (cloud platforms), vtap groups, orgs
+        # (reference controller http API: /v1/domains, /v1/vtap-groups,
+        # /v1/orgs — backing its MySQL tables; here in-process state)
+        if not hasattr(self, "domains"):
+            self.domains = {}
+        if not hasattr(self, "vtap_groups"):
+            self.vtap_groups = {"default": {"name": "default",
+                                            "agents": []}}
+        if not hasattr(self, "orgs"):
+            self.orgs = {1: {"org_id": 1, "name": "default"}}
+
+        @app.get("/v1/domains/")
+        def list_domains():
+            return list(self.domains.values())
+
+        @app.post("/v1/domains/")
+        async def create_domain(request: Request):
+            body = await request.json()
+            name = body["name"]
+            self.domains[name] = {"name": name,
+                                  "type": body.get("type", "kubernetes"),
+                                  "config": body.get("config", {})}
+            return self.domains[name]
+
+        @app.delete("/v1/domains/{name}")
+        def delete_domain(name: str):
+            return {"deleted": self.domains.pop(name, None) is not None}
+
+        @app.get("/v1/vtap-groups/")
+        def list_groups():
+            return list(self.vtap_groups.values())
+
+        @app.post("/v1/vtap-groups/")
+        async def create_group(request: Request):
+            body = await request.json()
+            name = body["name"]
+            self.vtap_groups[name] = {"name": name, "agents": []}
+            return self.vtap_groups[name]
+
+        @app.post("/v1/vtap-groups/{name}/agents/{agent_id}")
+        def assign_agent(name: str, agent_id: int):
+            g = self.vtap_groups.get(name)
+            if g is None:
+                return {"error": "no such group"}
+            if agent_id not in g["agents"]:
+                g["agents"].append(agent_id)
+            rec = self.agents.get(agent_id)
+            if rec is not None:
+                rec.group = name
+            return g
+
+        @app.get("/v1/orgs/")
+        def list_orgs():
+            return list(self.orgs.values())
+
+        @app.post("/v1/orgs/")
+        async def create_org(request: Request):
+            body = await request.json()
+            oid = int(body["org_id"])
+            self.orgs[oid] = {"org_id": oid,
+                              "name": body.get("name", f"org-{oid}")}
+            return self.orgs[oid]
