@@ -1105,51 +1105,57 @@ __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_ro
     }
     uint64_t act = __ballot(ok);
     if (act == 0) return;
-    uint32_t leader = (uint32_t)__ffsll((unsigned long long)act) - 1;
-    uint64_t h0 = __shfl(h, leader);
-    bool uniform = __all(!ok || h == h0);
-    if (uniform) {
-        // one claim + one atomic per agg for the whole wave
+    // hoist agg inputs once per row
+    uint64_t varr[QMAX_AGGS];
+    if (ok)
+        for (uint32_t a = 0; a < q.n_aggs; a++)
+            varr[a] = q.aggs[a].op == AGGOP_COUNT ? 1
+                : src_value(s, row, q.aggs[a].family, q.aggs[a].idx, 0,
+                            q.time_base_s);
+    // wave-clustered aggregation: iterate the distinct keys present in
+    // this wave; each cluster does ONE atomic per agg instead of one per
+    // lane (low-cardinality group-bys otherwise serialize on hot group
+    // slots across 256 CUs). Whole wave stays converged throughout.
+    uint64_t remaining = act;
+    int iters = 0;
+    while (remaining && iters++ < 16) {
+        uint32_t lead = (uint32_t)__ffsll((unsigned long long)remaining) - 1;
+        uint64_t hl = __shfl((long long)h, lead);
+        bool mine = ok && h == hl;
+        uint64_t grp = __ballot(mine);
+        remaining &= ~grp;
         uint32_t slot = 0;
-        if ((threadIdx.x & 63) == leader)
+        if ((threadIdx.x & 63) == lead)
             slot = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
-        slot = (uint32_t)__shfl((int)slot, leader);
+        slot = (uint32_t)__shfl((int)slot, lead);
         unsigned long long* acc = &gvals[(uint64_t)slot * QMAX_AGGS];
         for (uint32_t a = 0; a < q.n_aggs; a++) {
             const QAgg& ag = q.aggs[a];
             bool is_sum = ag.op == AGGOP_COUNT || ag.op == AGGOP_SUM;
             uint64_t ident = ag.op == AGGOP_MIN ? ~0ull : 0ull;
-            uint64_t v = ident;
-            if (ok)
-                v = ag.op == AGGOP_COUNT ? 1
-                    : src_value(s, row, ag.family, ag.idx, 0, q.time_base_s);
-            // full-wave reduction (all 64 lanes converged)
+            uint64_t v = mine ? varr[a] : ident;
             for (int d = 32; d > 0; d >>= 1) {
                 uint64_t o = __shfl_xor((long long)v, d);
                 if (is_sum) v += o;
                 else if (ag.op == AGGOP_MIN) v = o < v ? o : v;
                 else v = o > v ? o : v;
             }
-            if ((threadIdx.x & 63) == leader) {
-                if (is_sum) {
-                    // subtract the identity contributions? none: ident 0
-                    atomicAdd(&acc[a], (unsigned long long)v);
-                } else if (ag.op == AGGOP_MIN) {
+            if ((threadIdx.x & 63) == lead) {
+                if (is_sum) atomicAdd(&acc[a], (unsigned long long)v);
+                else if (ag.op == AGGOP_MIN)
                     atomicMin(&acc[a], (unsigned long long)v);
-                } else {
-                    atomicMax(&acc[a], (unsigned long long)v);
-                }
+                else atomicMax(&acc[a], (unsigned long long)v);
             }
         }
-        return;
     }
-    if (!ok) return;
+    // high-cardinality tail (>16 distinct keys in one wave): per-lane
+    // atomics — contention is negligible exactly when keys are diverse
+    if (!(ok && (remaining >> (threadIdx.x & 63)) & 1ull)) return;
     uint32_t slot = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
     unsigned long long* acc = &gvals[(uint64_t)slot * QMAX_AGGS];
     for (uint32_t a = 0; a < q.n_aggs; a++) {
         const QAgg& ag = q.aggs[a];
-        uint64_t v = ag.op == AGGOP_COUNT ? 1
-            : src_value(s, row, ag.family, ag.idx, 0, q.time_base_s);
+        uint64_t v = varr[a];
         switch (ag.op) {
             case AGGOP_COUNT:
             case AGGOP_SUM: atomicAdd(&acc[a], (unsigned long long)v); break;
