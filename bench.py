@@ -62,7 +62,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=1_000_000,
+    ap.add_argument("--batch", type=int, default=2_000_000,
                     help="spans per step per rank")
     ap.add_argument("--tag-card", type=int, default=100_000)
     ap.add_argument("--n-attrs", type=int, default=4)
