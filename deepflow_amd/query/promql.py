@@ -73,6 +73,16 @@ class PromError(ValueError):
     pass
 
 
+def _apply_op(a: float, b: float, op: str) -> float:
+    if op == "+":
+        return a + b
+    if op == "-":
+        return a - b
+    if op == "*":
+        return a * b
+    return a / b if b else float("nan")
+
+
 def _parse_matchers(s: Optional[str]) -> List[Tuple[str, str, str]]:
     out = []
     if not s:
@@ -140,8 +150,72 @@ class PromQLEngine:
         return list(series.values())
 
     # -------------------------------------------------------- evaluation
+    @staticmethod
+    def _split_binop(expr: str):
+        """Find a top-level binary operator (lowest precedence first:
+        +,- then *,/). Returns (lhs, op, rhs) or None."""
+        for ops in ("+-", "*/"):
+            depth = 0
+            in_str = False
+            for i in range(len(expr) - 1, 0, -1):
+                c = expr[i]
+                if c == '"':
+                    in_str = not in_str
+                elif in_str:
+                    continue
+                elif c == ")":
+                    depth += 1
+                elif c == "(":
+                    depth -= 1
+                elif depth == 0 and c in ops:
+                    # don't split inside a range selector like [1m]
+                    lhs, rhs = expr[:i], expr[i + 1:]
+                    if not lhs.strip() or not rhs.strip():
+                        continue
+                    return lhs, c, rhs
+        return None
+
     def _eval_at(self, expr: str, t: int) -> List[Dict]:
         """Evaluate expr at instant t -> [{metric, value}]."""
+        expr = expr.strip()
+        # parenthesized sub-expression
+        if expr.startswith("(") :
+            inner, tail = _balanced(expr[expr.index("("):])
+            if inner is not None and not tail.strip():
+                return self._eval_at(inner, t)
+        sp = self._split_binop(expr)
+        if sp is not None:
+            lhs, op, rhs = sp
+            try:
+                rscalar = float(rhs.strip())
+                left = self._eval_at(lhs, t)
+                right = None
+            except ValueError:
+                right = self._eval_at(rhs, t)
+                left = self._eval_at(lhs, t)
+                rscalar = None
+            out = []
+            if rscalar is not None:
+                for s in left:
+                    out.append({"metric": s["metric"],
+                                "value": _apply_op(s["value"], rscalar,
+                                                   op)})
+                return out
+            # vector/vector: match on identical label sets (__name__
+            # dropped, standard PromQL matching)
+            def key(s):
+                return tuple(sorted((k, v) for k, v in s["metric"].items()
+                                    if k != "__name__"))
+            rmap = {key(s): s["value"] for s in right}
+            for s in left:
+                k = key(s)
+                if k in rmap:
+                    out.append({"metric": {kk: vv for kk, vv in
+                                           s["metric"].items()
+                                           if kk != "__name__"},
+                                "value": _apply_op(s["value"], rmap[k],
+                                                   op)})
+            return out
         agg = _match_agg(expr)
         if agg:
             func, by, inner_expr = agg

@@ -105,3 +105,25 @@ def test_datasource_1h(server):
     r2 = client.post("/v1/query/", json={
         "sql": "SELECT Sum(request) AS r FROM application.10m"})
     assert r2.json()["result"]["values"][0][0] == 400
+
+
+def test_binary_ops():
+    """Vector/vector and vector/scalar arithmetic (error-ratio shape)."""
+    from deepflow_amd.query.promql import PromQLEngine
+    rows = [
+        {"time": 10, "vtap_id": 1, "request": 100, "server_error": 5},
+        {"time": 10, "vtap_id": 2, "request": 200, "server_error": 2},
+    ]
+    eng = PromQLEngine(lambda: rows)
+    r = eng.instant(
+        "sum(rate(application_server_error[1m])) by (vtap_id) / "
+        "sum(rate(application_request[1m])) by (vtap_id)", t=10)
+    vals = {s["metric"]["vtap_id"]: float(s["value"][1])
+            for s in r["data"]["result"]}
+    assert abs(vals["1"] - 5 / 100) < 1e-9
+    assert abs(vals["2"] - 2 / 200) < 1e-9
+    r2 = eng.instant(
+        "sum(rate(application_request[1m])) by (vtap_id) * 60", t=10)
+    vals2 = {s["metric"]["vtap_id"]: float(s["value"][1])
+             for s in r2["data"]["result"]}
+    assert abs(vals2["1"] - 100) < 1e-9
