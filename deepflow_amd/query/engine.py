@@ -687,7 +687,22 @@ class QueryEngine:
             return self.name_maps.get(mp, {}).get(v, str(v))
         return v
 
+    _HAVING_OPS = {"=": lambda a, b: a == b, "!=": lambda a, b: a != b,
+                   "<>": lambda a, b: a != b,
+                   "<": lambda a, b: a < b, "<=": lambda a, b: a <= b,
+                   ">": lambda a, b: a > b, ">=": lambda a, b: a >= b}
+
     def _order_limit(self, plan: Q.Plan, columns, rows):
+        if plan.having:
+            for name, op, num in plan.having:
+                if name not in columns:
+                    from .sql import SqlError
+                    raise SqlError(f"HAVING references unknown column "
+                                   f"{name!r}")
+                ci = columns.index(name)
+                fn = self._HAVING_OPS[op]
+                rows = [r for r in rows
+                        if r[ci] is not None and fn(r[ci], num)]
         if plan.order_by:
             for name, desc in reversed(plan.order_by):
                 if name in columns:

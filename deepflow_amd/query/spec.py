@@ -97,6 +97,7 @@ class Plan:
     key_meta: List[dict] = field(default_factory=list)  # hydration info
     agg_meta: List[dict] = field(default_factory=list)
     order_by: Optional[List] = None
+    having: Optional[List] = None  # [(column, op, number)] post-agg
     limit: Optional[int] = None
     slimit: Optional[int] = None
     select_rows: bool = False  # non-aggregated SELECT

@@ -241,6 +241,25 @@ def parse_sql(sql: str, dictionary=None, time_base_s: int = 0,
                 continue
             break
 
+    # HAVING: post-aggregation filter on select aliases
+    if p.kw_is("having"):
+        p.next()
+        having = []
+        while True:
+            name = p.next()[1]
+            op_t = p.next()
+            if op_t[0] != "op" or op_t[1] not in Q.OP_BY_NAME:
+                raise SqlError(f"bad HAVING operator {op_t!r}")
+            lit = p.next()
+            if lit[0] != "num":
+                raise SqlError("HAVING compares against a number")
+            having.append((name, op_t[1], float(lit[1])))
+            if p.kw_is("and"):
+                p.next()
+                continue
+            break
+        plan.having = having
+
     # ORDER BY / LIMIT
     if p.kw_is("order"):
         p.next()

@@ -85,3 +85,17 @@ def test_row_table_or_group(engine):
         "SELECT Count(*) AS c FROM l7_flow_log "
         "WHERE (l7_protocol = 'HTTP' OR l7_protocol = 'DNS')")
     assert r["values"][0][0] == direct["values"][0][0]
+
+
+def test_having(engine):
+    r = engine.query(
+        "SELECT request_resource, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY request_resource HAVING c >= 200 ORDER BY c DESC")
+    assert r["values"], "expected some groups over the threshold"
+    assert all(row[1] >= 200 for row in r["values"])
+    full = engine.query(
+        "SELECT request_resource, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY request_resource")
+    want = sorted([v for _, v in map(tuple, full["values"]) if v >= 200],
+                  reverse=True)
+    assert [row[1] for row in r["values"]] == want
