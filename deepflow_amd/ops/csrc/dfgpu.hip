@@ -1087,80 +1087,108 @@ __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_ro
                             uint64_t* __restrict__ graw,
                             unsigned long long* __restrict__ gvals,
                             uint32_t cap_mask) {
-    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-    // no early return: the whole wave stays converged so low-cardinality
-    // group keys take the wave-reduced path (one atomic per wave, not 64 —
-    // a lone accumulator otherwise serializes the whole scan)
-    uint64_t row = base_row + (i < n ? i : 0);
-    bool ok = i < n && eval_terms(s, row, q);
-    uint64_t kraw[QMAX_KEYS];
-    uint64_t h = 0x243F6A8885A308D3ull;
-    if (ok) {
+    // Grid-strided scan with an LDS-resident group table: each workgroup
+    // aggregates its stripe into shared memory (26 KB of the 160 KB/CU
+    // LDS) and flushes once at the end — global atomics drop from
+    // per-row to per-(block x live slot). Groups are identified by the
+    // 64-bit mixed key hash (same convention as the global table; raw
+    // keys are kept for hydration only). Rows whose key misses the LDS
+    // table (very high per-block cardinality) fall back to per-lane
+    // global accumulation.
+    constexpr uint32_t NSLOT = 256;
+    __shared__ uint64_t lkey[NSLOT];
+    __shared__ uint64_t lraw[NSLOT][QMAX_KEYS];
+    __shared__ unsigned long long lagg[NSLOT][QMAX_AGGS];
+    for (uint32_t sl = threadIdx.x; sl < NSLOT; sl += blockDim.x) {
+        lkey[sl] = EMPTY_KEY;
+        for (uint32_t a = 0; a < q.n_aggs; a++)
+            lagg[sl][a] = q.aggs[a].op == AGGOP_MIN ? ~0ull : 0ull;
+    }
+    __syncthreads();
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (uint64_t)gridDim.x * blockDim.x) {
+        uint64_t row = base_row + i;
+        if (!eval_terms(s, row, q)) continue;
+        uint64_t kraw[QMAX_KEYS];
+        uint64_t h = 0x243F6A8885A308D3ull;
         for (uint32_t k = 0; k < q.n_keys; k++) {
             kraw[k] = src_value(s, row, q.keys[k].family, q.keys[k].idx,
                                 q.keys[k].bucket, q.time_base_s);
             h = mix64(h ^ kraw[k] ^ ((uint64_t)k << 56));
         }
         if (h == EMPTY_KEY) h = 1;
-    }
-    uint64_t act = __ballot(ok);
-    if (act == 0) return;
-    // hoist agg inputs once per row
-    uint64_t varr[QMAX_AGGS];
-    if (ok)
+        uint64_t varr[QMAX_AGGS];
         for (uint32_t a = 0; a < q.n_aggs; a++)
             varr[a] = q.aggs[a].op == AGGOP_COUNT ? 1
                 : src_value(s, row, q.aggs[a].family, q.aggs[a].idx, 0,
                             q.time_base_s);
-    // wave-clustered aggregation: iterate the distinct keys present in
-    // this wave; each cluster does ONE atomic per agg instead of one per
-    // lane (low-cardinality group-bys otherwise serialize on hot group
-    // slots across 256 CUs). Whole wave stays converged throughout.
-    uint64_t remaining = act;
-    int iters = 0;
-    while (remaining && iters++ < 16) {
-        uint32_t lead = (uint32_t)__ffsll((unsigned long long)remaining) - 1;
-        uint64_t hl = __shfl((long long)h, lead);
-        bool mine = ok && h == hl;
-        uint64_t grp = __ballot(mine);
-        remaining &= ~grp;
-        uint32_t slot = 0;
-        if ((threadIdx.x & 63) == lead)
-            slot = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
-        slot = (uint32_t)__shfl((int)slot, lead);
-        unsigned long long* acc = &gvals[(uint64_t)slot * QMAX_AGGS];
-        for (uint32_t a = 0; a < q.n_aggs; a++) {
-            const QAgg& ag = q.aggs[a];
-            bool is_sum = ag.op == AGGOP_COUNT || ag.op == AGGOP_SUM;
-            uint64_t ident = ag.op == AGGOP_MIN ? ~0ull : 0ull;
-            uint64_t v = mine ? varr[a] : ident;
-            for (int d = 32; d > 0; d >>= 1) {
-                uint64_t o = __shfl_xor((long long)v, d);
-                if (is_sum) v += o;
-                else if (ag.op == AGGOP_MIN) v = o < v ? o : v;
-                else v = o > v ? o : v;
+        uint32_t slot = (uint32_t)h & (NSLOT - 1);
+        bool placed = false;
+        for (uint32_t probe = 0; probe < NSLOT / 4; probe++) {
+            uint64_t cur = lkey[slot];
+            if (cur == EMPTY_KEY) {
+                uint64_t old = atomicCAS(
+                    (unsigned long long*)&lkey[slot],
+                    (unsigned long long)EMPTY_KEY, (unsigned long long)h);
+                if (old == EMPTY_KEY) {
+                    for (uint32_t k = 0; k < q.n_keys; k++)
+                        lraw[slot][k] = kraw[k];
+                    cur = h;
+                } else {
+                    cur = old;
+                }
             }
-            if ((threadIdx.x & 63) == lead) {
-                if (is_sum) atomicAdd(&acc[a], (unsigned long long)v);
-                else if (ag.op == AGGOP_MIN)
-                    atomicMin(&acc[a], (unsigned long long)v);
-                else atomicMax(&acc[a], (unsigned long long)v);
+            if (cur == h) {
+                for (uint32_t a = 0; a < q.n_aggs; a++) {
+                    uint32_t op = q.aggs[a].op;
+                    if (op == AGGOP_COUNT || op == AGGOP_SUM)
+                        atomicAdd(&lagg[slot][a],
+                                  (unsigned long long)varr[a]);
+                    else if (op == AGGOP_MIN)
+                        atomicMin(&lagg[slot][a],
+                                  (unsigned long long)varr[a]);
+                    else
+                        atomicMax(&lagg[slot][a],
+                                  (unsigned long long)varr[a]);
+                }
+                placed = true;
+                break;
+            }
+            slot = (slot + 1) & (NSLOT - 1);
+        }
+        if (!placed) {  // LDS table saturated for this key: go global
+            uint32_t g = group_claim(h, kraw, q.n_keys, gkeys, graw,
+                                     cap_mask);
+            unsigned long long* acc = &gvals[(uint64_t)g * QMAX_AGGS];
+            for (uint32_t a = 0; a < q.n_aggs; a++) {
+                uint32_t op = q.aggs[a].op;
+                if (op == AGGOP_COUNT || op == AGGOP_SUM)
+                    atomicAdd(&acc[a], (unsigned long long)varr[a]);
+                else if (op == AGGOP_MIN)
+                    atomicMin(&acc[a], (unsigned long long)varr[a]);
+                else
+                    atomicMax(&acc[a], (unsigned long long)varr[a]);
             }
         }
     }
-    // high-cardinality tail (>16 distinct keys in one wave): per-lane
-    // atomics — contention is negligible exactly when keys are diverse
-    if (!(ok && (remaining >> (threadIdx.x & 63)) & 1ull)) return;
-    uint32_t slot = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
-    unsigned long long* acc = &gvals[(uint64_t)slot * QMAX_AGGS];
-    for (uint32_t a = 0; a < q.n_aggs; a++) {
-        const QAgg& ag = q.aggs[a];
-        uint64_t v = varr[a];
-        switch (ag.op) {
-            case AGGOP_COUNT:
-            case AGGOP_SUM: atomicAdd(&acc[a], (unsigned long long)v); break;
-            case AGGOP_MIN: atomicMin(&acc[a], (unsigned long long)v); break;
-            case AGGOP_MAX: atomicMax(&acc[a], (unsigned long long)v); break;
+    __syncthreads();
+    // flush the block-local table into the global group store
+    for (uint32_t sl = threadIdx.x; sl < NSLOT; sl += blockDim.x) {
+        uint64_t h = lkey[sl];
+        if (h == EMPTY_KEY) continue;
+        uint32_t g = group_claim(h, lraw[sl], q.n_keys, gkeys, graw,
+                                 cap_mask);
+        unsigned long long* acc = &gvals[(uint64_t)g * QMAX_AGGS];
+        for (uint32_t a = 0; a < q.n_aggs; a++) {
+            uint32_t op = q.aggs[a].op;
+            unsigned long long v = lagg[sl][a];
+            if (op == AGGOP_COUNT || op == AGGOP_SUM) {
+                if (v) atomicAdd(&acc[a], v);
+            } else if (op == AGGOP_MIN) {
+                if (v != ~0ull) atomicMin(&acc[a], v);
+            } else {
+                if (v) atomicMax(&acc[a], v);
+            }
         }
     }
 }
@@ -1383,7 +1411,12 @@ int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
               (const uint8_t*)pool, stride, n_rows};
     QuerySpec q;
     __builtin_memcpy(&q, spec, sizeof(QuerySpec));
-    hipLaunchKernelGGL(k_query_agg, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
+    // grid-stride kernel: bounded grid so each workgroup amortizes its
+    // LDS group table over many rows (still >> 256 workgroups: 4096
+    // covers all 8 XCDs with deep occupancy)
+    uint32_t blocks = grid_for(n);
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL(k_query_agg, dim3(blocks), dim3(BLOCK), 0, STREAM(stream),
                        s, q, n, base_row, (uint64_t*)gkeys, (uint64_t*)graw,
                        (unsigned long long*)gvals, cap - 1);
     return (int)hipGetLastError();
