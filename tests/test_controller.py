@@ -28,3 +28,43 @@ def test_crud_routes():
 
     c.post("/v1/orgs/", json={"org_id": 7, "name": "tenant-7"})
     assert {o["org_id"] for o in c.get("/v1/orgs/").json()} == {1, 7}
+
+
+def test_alert_policies():
+    """DF-SQL alert policy fires into alert_event, queryable back."""
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.ingest.event_pipeline import EventPipeline
+    from deepflow_amd.query import QueryEngine
+    from deepflow_amd.control.alerting import (AlertEvaluator, AlertPolicy,
+                                               LEVEL_CRITICAL)
+    cfg = SpanGenConfig(n=500, seed=11, tag_cardinality=20, n_ips=32,
+                        n_services=3, n_resources=6)
+    pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 10,
+                            dict_capacity=1 << 11,
+                            time_base_s=cfg.base_time_ns // 10**9)
+    pipe.ingest_frame_payload(gen_span_payload(cfg))
+    eng = QueryEngine(pipe, device="cpu")
+    events = EventPipeline()
+    ev = AlertEvaluator(eng, events)
+    ev.add_policy(AlertPolicy(
+        "too-many-errors",
+        "SELECT request_resource, Count(*) AS errs FROM l7_flow_log "
+        "WHERE response_status = 'Server Error' GROUP BY request_resource",
+        column="errs", op=">=", threshold=1, level=LEVEL_CRITICAL,
+        target_column="request_resource"))
+    ev.add_policy(AlertPolicy(
+        "never-fires",
+        "SELECT Count(*) AS c FROM l7_flow_log", column="c", op=">",
+        threshold=10**9))
+    fired = ev.evaluate_once()
+    assert fired >= 1
+    assert all(e["policy_name"] == "too-many-errors"
+               for e in events.alert_events)
+    assert events.alert_events[0]["level"] == LEVEL_CRITICAL
+    assert events.alert_events[0]["target"].startswith("/")
+    # and it is queryable through the alert_event table
+    eng.alert_event_rows = lambda: events.alert_events
+    r = eng.query("SELECT policy_name, level FROM alert_event LIMIT 5")
+    assert ["too-many-errors", LEVEL_CRITICAL] in r["values"]
