@@ -196,6 +196,25 @@ class DeepflowServer:
             return Response(content=blob,
                             media_type="application/x-protobuf")
 
+        # alert policies: DF-SQL rules evaluated on a timer -> alert_event
+        from fastapi import Request
+        from .control.alerting import AlertEvaluator, AlertPolicy
+        self.alerts = AlertEvaluator(self.engine, self.events)
+
+        @self.app.post("/v1/alert-policies/")
+        async def add_alert_policy(request: Request):
+            body = await request.json()
+            self.alerts.add_policy(AlertPolicy(
+                body["name"], body["sql"], body["column"],
+                body.get("op", ">="), float(body["threshold"]),
+                level=int(body.get("level", 2)),
+                target_column=body.get("target_column")))
+            return {"policies": len(self.alerts.policies)}
+
+        @self.app.post("/v1/alert-policies/evaluate")
+        def evaluate_alerts():
+            return {"fired": self.alerts.evaluate_once()}
+
         # OTLP/HTTP collector endpoints (standard /v1/{traces,logs,metrics}
         # paths, protobuf bodies; reference accepts the same three
         # signals through its otel integration port)
