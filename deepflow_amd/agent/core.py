@@ -56,6 +56,7 @@ class Agent:
         self.org_id = org_id
         self.server = server
         self._sock: Optional[socket.socket] = None
+        self._server_idx = 0
 
     def close(self) -> None:
         if self._h:
@@ -221,8 +222,30 @@ class Agent:
                 hdr, self.dfstats_payload(now_ns // 10**9)))
         for frame in frames:
             if self.server is not None:
-                if self._sock is None:
-                    self._sock = socket.create_connection(self.server,
-                                                          timeout=5)
-                self._sock.sendall(frame)
+                self._send_with_failover(frame)
         return len(frames)
+
+    def _send_with_failover(self, frame: bytes, retries: int = 2) -> None:
+        """uniform-sender failover: on send failure, reconnect — rotating
+        through the server list when more than one ingester is known."""
+        servers = self.server if isinstance(self.server, list) \
+            else [self.server]
+        last: Optional[Exception] = None
+        for attempt in range(retries + 1):
+            try:
+                if self._sock is None:
+                    target = servers[(self._server_idx + attempt)
+                                     % len(servers)]
+                    self._sock = socket.create_connection(target, timeout=5)
+                    self._server_idx = (self._server_idx + attempt) % \
+                        len(servers)
+                self._sock.sendall(frame)
+                return
+            except OSError as e:
+                last = e
+                if self._sock:
+                    self._sock.close()
+                self._sock = None
+        self.exceptions = getattr(self, "exceptions", 0) | \
+            self.EXC_SERVER_UNREACHABLE
+        raise ConnectionError(f"all ingesters unreachable: {last}")

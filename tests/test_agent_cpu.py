@@ -146,3 +146,33 @@ def test_agent_config_template_and_diffs(tmp_path):
             for r in framing.iter_records(a.drain(1))]
     assert recs and recs[0]["base"]["head"]["proto"] == 127  # custom port
     a.close()
+
+
+def test_sender_failover():
+    """Primary ingester down -> frames land on the secondary."""
+    from fastapi.testclient import TestClient  # noqa: F401 (env parity)
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import http_session
+    from deepflow_amd.server import DeepflowServer
+    backup = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                            dict_capacity=1 << 12)
+    backup.start()
+    # primary: a port nobody listens on
+    import socket as _s
+    dead = _s.socket()
+    dead.bind(("127.0.0.1", 0))
+    dead_port = dead.getsockname()[1]
+    dead.close()  # released: connect will be refused
+    a = Agent(vtap_id=4, server=[("127.0.0.1", dead_port),
+                                 ("127.0.0.1", backup.receiver.tcp_port)])
+    for frame, ts in http_session(0x0A000001, 0x0A000002, t0=10**9):
+        a.packet(frame, ts)
+    sent = a.flush_to_server(10**12)
+    assert sent >= 2
+    import time
+    deadline = time.time() + 10
+    while time.time() < deadline and backup.l7.stats.spans_in < 1:
+        time.sleep(0.05)
+    assert backup.l7.stats.spans_in >= 1  # failed over to the backup
+    a.close()
+    backup.stop()
