@@ -319,8 +319,12 @@ class QueryEngine:
             vals = [[n, n, t.hydrate] for n, t in sorted(tags.items())]
             return {"columns": cols, "values": vals}
         if what == "metrics":
-            cols = ["name", "display_name", "type"]
-            vals = [[n, n, "counter"] for n in sorted(mets)]
+            from .tags import METRIC_UNITS
+            cols = ["name", "display_name", "unit", "type"]
+            vals = []
+            for n in sorted(mets):
+                unit, disp = METRIC_UNITS.get(n, ("count", n))
+                vals.append([n, disp, unit, "counter"])
             return {"columns": cols, "values": vals}
         if what == "tables":
             return {"columns": ["name"],
@@ -578,8 +582,9 @@ class QueryEngine:
     # ----------------------------------------------------------- row tables
     def _run_rows(self, sql: str, rows: List[Dict],
                   time_base_s: int) -> Dict:
-        fields = list(rows[0].keys()) if rows else \
-            ["time", "vtap_id", "request"]
+        if not rows:
+            return {"columns": [], "values": []}
+        fields = list(rows[0].keys())
         tags = _row_tags(fields)
         mets = {f: tags[f] for f in fields}
         mets["log_count"] = TagDef("log_count", Q.SRC_CONST0, 0)

@@ -120,6 +120,24 @@ def build_l7_tags() -> Dict[str, TagDef]:
 L7_TAGS = build_l7_tags()
 
 # aggregatable metric fields (reference: db_descriptions metrics)
+# db_descriptions analog: units + display info for `show metrics`
+METRIC_UNITS: Dict[str, tuple] = {
+    "response_duration": ("us", "Response Duration"),
+    "request_length": ("byte", "Request Length"),
+    "response_length": ("byte", "Response Length"),
+    "captured_request_byte": ("byte", "Captured Request Bytes"),
+    "captured_response_byte": ("byte", "Captured Response Bytes"),
+    "log_count": ("count", "Log Count"),
+    "byte_tx": ("byte", "Bytes TX"), "byte_rx": ("byte", "Bytes RX"),
+    "packet_tx": ("count", "Packets TX"),
+    "packet_rx": ("count", "Packets RX"),
+    "rtt": ("us", "TCP Handshake RTT"),
+    "srt_sum": ("us", "System Response Time Sum"),
+    "art_sum": ("us", "Application Response Time Sum"),
+    "retrans_tx": ("count", "Retransmits TX"),
+    "retrans_rx": ("count", "Retransmits RX"),
+}
+
 L7_METRICS: Dict[str, TagDef] = {
     "response_duration": L7_TAGS["response_duration"],
     "request_length": L7_TAGS["request_length"],
