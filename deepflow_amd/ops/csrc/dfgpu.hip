@@ -1088,16 +1088,17 @@ __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_ro
                             unsigned long long* __restrict__ gvals,
                             uint32_t cap_mask) {
     // Grid-strided scan with an LDS-resident group table: each workgroup
-    // aggregates its stripe into shared memory (26 KB of the 160 KB/CU
-    // LDS) and flushes once at the end — global atomics drop from
-    // per-row to per-(block x live slot). Groups are identified by the
-    // 64-bit mixed key hash (same convention as the global table; raw
-    // keys are kept for hydration only). Rows whose key misses the LDS
-    // table (very high per-block cardinality) fall back to per-lane
-    // global accumulation.
-    constexpr uint32_t NSLOT = 256;
+    // aggregates its stripe into shared memory (~37 KB of the 160 KB/CU
+    // LDS: 512 slots x (key + global-slot + 8 accumulators)) and flushes
+    // once at the end — global atomics drop from per-row to
+    // per-(block x live slot). Groups are identified by the 64-bit mixed
+    // key hash (same convention as the global table); the LDS claimant
+    // registers the group globally immediately, so raw keys never stage
+    // in LDS. Rows whose key misses the table (very high per-block
+    // cardinality) fall back to per-lane global accumulation.
+    constexpr uint32_t NSLOT = 512;
     __shared__ uint64_t lkey[NSLOT];
-    __shared__ uint64_t lraw[NSLOT][QMAX_KEYS];
+    __shared__ uint32_t lgslot[NSLOT];
     __shared__ unsigned long long lagg[NSLOT][QMAX_AGGS];
     for (uint32_t sl = threadIdx.x; sl < NSLOT; sl += blockDim.x) {
         lkey[sl] = EMPTY_KEY;
@@ -1122,41 +1123,46 @@ __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_ro
             varr[a] = q.aggs[a].op == AGGOP_COUNT ? 1
                 : src_value(s, row, q.aggs[a].family, q.aggs[a].idx, 0,
                             q.time_base_s);
+        // LDS probe; the claimant registers the group globally at claim
+        // time, so the block-local slot carries the global slot id and
+        // no raw keys need staging in LDS
         uint32_t slot = (uint32_t)h & (NSLOT - 1);
-        bool placed = false;
-        for (uint32_t probe = 0; probe < NSLOT / 4; probe++) {
+        uint32_t gslot = 0xFFFFFFFFu;
+        for (uint32_t probe = 0; probe < 16; probe++) {
             uint64_t cur = lkey[slot];
             if (cur == EMPTY_KEY) {
                 uint64_t old = atomicCAS(
                     (unsigned long long*)&lkey[slot],
                     (unsigned long long)EMPTY_KEY, (unsigned long long)h);
                 if (old == EMPTY_KEY) {
-                    for (uint32_t k = 0; k < q.n_keys; k++)
-                        lraw[slot][k] = kraw[k];
+                    lgslot[slot] = group_claim(h, kraw, q.n_keys, gkeys,
+                                               graw, cap_mask);
+                    __threadfence_block();
                     cur = h;
                 } else {
                     cur = old;
                 }
             }
             if (cur == h) {
-                for (uint32_t a = 0; a < q.n_aggs; a++) {
-                    uint32_t op = q.aggs[a].op;
-                    if (op == AGGOP_COUNT || op == AGGOP_SUM)
-                        atomicAdd(&lagg[slot][a],
-                                  (unsigned long long)varr[a]);
-                    else if (op == AGGOP_MIN)
-                        atomicMin(&lagg[slot][a],
-                                  (unsigned long long)varr[a]);
-                    else
-                        atomicMax(&lagg[slot][a],
-                                  (unsigned long long)varr[a]);
-                }
-                placed = true;
+                gslot = slot;
                 break;
             }
             slot = (slot + 1) & (NSLOT - 1);
         }
-        if (!placed) {  // LDS table saturated for this key: go global
+        if (gslot != 0xFFFFFFFFu) {
+            for (uint32_t a = 0; a < q.n_aggs; a++) {
+                uint32_t op = q.aggs[a].op;
+                if (op == AGGOP_COUNT || op == AGGOP_SUM)
+                    atomicAdd(&lagg[gslot][a],
+                              (unsigned long long)varr[a]);
+                else if (op == AGGOP_MIN)
+                    atomicMin(&lagg[gslot][a],
+                              (unsigned long long)varr[a]);
+                else
+                    atomicMax(&lagg[gslot][a],
+                              (unsigned long long)varr[a]);
+            }
+        } else {  // LDS table saturated for this key: go global directly
             uint32_t g = group_claim(h, kraw, q.n_keys, gkeys, graw,
                                      cap_mask);
             unsigned long long* acc = &gvals[(uint64_t)g * QMAX_AGGS];
@@ -1172,13 +1178,10 @@ __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_ro
         }
     }
     __syncthreads();
-    // flush the block-local table into the global group store
+    // flush the block-local accumulators into the global group store
     for (uint32_t sl = threadIdx.x; sl < NSLOT; sl += blockDim.x) {
-        uint64_t h = lkey[sl];
-        if (h == EMPTY_KEY) continue;
-        uint32_t g = group_claim(h, lraw[sl], q.n_keys, gkeys, graw,
-                                 cap_mask);
-        unsigned long long* acc = &gvals[(uint64_t)g * QMAX_AGGS];
+        if (lkey[sl] == EMPTY_KEY) continue;
+        unsigned long long* acc = &gvals[(uint64_t)lgslot[sl] * QMAX_AGGS];
         for (uint32_t a = 0; a < q.n_aggs; a++) {
             uint32_t op = q.aggs[a].op;
             unsigned long long v = lagg[sl][a];
