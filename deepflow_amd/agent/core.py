@@ -16,11 +16,12 @@ import numpy as np
 from ..ops import native
 from ..wire import framing
 
-DRAIN_L4, DRAIN_L7, DRAIN_DOC = 0, 1, 2
+DRAIN_L4, DRAIN_L7, DRAIN_DOC, DRAIN_PCAP = 0, 1, 2, 3
 
 _MSG_FOR = {DRAIN_L4: framing.MSG_TAGGEDFLOW,
             DRAIN_L7: framing.MSG_PROTOCOLLOG,
-            DRAIN_DOC: framing.MSG_METRICS}
+            DRAIN_DOC: framing.MSG_METRICS,
+            DRAIN_PCAP: framing.MSG_RAW_PCAP}
 
 
 def _lib():
@@ -209,7 +210,7 @@ class Agent:
         self-metrics frame. Returns frames sent."""
         self.tick(now_ns)
         frames = []
-        for which in (DRAIN_L4, DRAIN_L7, DRAIN_DOC):
+        for which in (DRAIN_L4, DRAIN_L7, DRAIN_DOC, DRAIN_PCAP):
             payload = self.drain(which)
             if payload:
                 frames.append(self.frame(which, payload, compress=compress))

@@ -154,7 +154,7 @@ struct Agent {
     std::vector<uint16_t> custom_ports;  // port-rule custom protocols (127)
     std::vector<AclRule> acls;
     std::map<MeterKey, AppMeterAcc> meters;
-    std::vector<uint8_t> out_l4, out_l7, out_doc;
+    std::vector<uint8_t> out_l4, out_l7, out_doc, out_pcap;
     // stats
     uint64_t pkts = 0, bytes = 0, flows_emitted = 0, l7_emitted = 0,
              docs_emitted = 0, parse_errors = 0;
@@ -2163,6 +2163,19 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
     FlowNode& f = it->second;
     dir = (src == f.ip[0] && sport == f.port[0]) ? 0 : 1;
     f.last_ns = ts_ns;
+    // ACL pcap action: mirror raw frames of matched flows into the
+    // packet store drain ([flow_id u64][ts u64][len u16][frame])
+    if ((f.acl_actions & 1u) && a.out_pcap.size() < (4u << 20)) {
+        size_t base = a.out_pcap.size();
+        a.out_pcap.resize(base + 18 + len);
+        uint8_t* w = a.out_pcap.data() + base;
+        memcpy(w, &f.flow_id, 8);
+        memcpy(w + 8, &ts_ns, 8);
+        uint16_t l16 = (uint16_t)(len > 0xFFFF ? 0xFFFF : len);
+        memcpy(w + 16, &l16, 2);
+        memcpy(w + 18, pkt, l16);
+        if (l16 < len) a.out_pcap.resize(base + 18 + l16);
+    }
     PeerStats& ps = f.peer[dir];
     ps.packets++; ps.total_packets++;
     ps.bytes += len; ps.total_bytes += len;
@@ -2261,7 +2274,8 @@ void dfa_tick(void* h, uint64_t now_ns) {
 uint64_t dfa_drain(void* h, int which, uint8_t* out, uint64_t cap) {
     Agent& a = *(Agent*)h;
     std::vector<uint8_t>& src = which == 0 ? a.out_l4
-                               : which == 1 ? a.out_l7 : a.out_doc;
+                               : which == 1 ? a.out_l7
+                               : which == 2 ? a.out_doc : a.out_pcap;
     uint64_t n = src.size();
     if (out && n <= cap) memcpy(out, src.data(), n);
     if (out) src.clear();

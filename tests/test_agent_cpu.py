@@ -176,3 +176,35 @@ def test_sender_failover():
     assert backup.l7.stats.spans_in >= 1  # failed over to the backup
     a.close()
     backup.stop()
+
+
+def test_acl_pcap_action_end_to_end():
+    """ACL action bit 0 mirrors matched flows' raw frames; they ride
+    MSG_RAW_PCAP to the server's packet store and export as pcap."""
+    import time
+    from fastapi.testclient import TestClient
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import http_session
+    from deepflow_amd.server import DeepflowServer
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12)
+    srv.start()
+    a = Agent(vtap_id=1, server=("127.0.0.1", srv.receiver.tcp_port))
+    a.add_acl(3, dst_net=0x0A000002, dst_masklen=32, action=1)  # pcap
+    for frame, ts in http_session(0x0A000001, 0x0A000002, t0=10**9):
+        a.packet(frame, ts)
+    for frame, ts in http_session(0x0A000001, 0x0A000003, sport=43999,
+                                  t0=10**9):
+        a.packet(frame, ts)  # unmatched flow: NOT mirrored
+    a.flush_to_server(10**12)
+    deadline = time.time() + 10
+    while time.time() < deadline and not srv.pcap.flows:
+        time.sleep(0.05)
+    assert len(srv.pcap.flows) == 1  # only the ACL-matched flow
+    fid = next(iter(srv.pcap.flows))
+    client = TestClient(srv.app)
+    blob = client.get(f"/v1/pcap/{fid}").content
+    assert blob[:4] in (b"\xa1\xb2\xc3\xd4", b"\xd4\xc3\xb2\xa1")
+    assert len(srv.pcap.flows[fid]) >= 7  # handshake+data+teardown frames
+    a.close()
+    srv.stop()
