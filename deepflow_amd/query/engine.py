@@ -423,10 +423,20 @@ class QueryEngine:
         if plan.select_rows:
             return self._run_select(plan, segments, tags, str_cols)
         groups = execute(plan, segments, self.device)
-        needs_second = any(m["op"] in ("percentile", "apdex")
-                           for m in plan.agg_meta)
-        if needs_second and len(groups) > 256:
-            raise SqlError("Percentile/Apdex limited to <=256 groups")
+        q_metas = [m for m in plan.agg_meta
+                   if m["op"] in ("percentile", "apdex")]
+        q_lookup = None
+        if q_metas:
+            # single-pass grouped gather (no per-group re-scan, no group
+            # cap): values sorted within each group for quantile math
+            from .executor import execute_grouped_values
+            uniq, per_meta = execute_grouped_values(plan, segments,
+                                                    q_metas, self.device)
+            key_index = {tuple(int(x) & ((1 << 64) - 1)
+                               for x in uniq[i].tolist()): i
+                         for i in range(uniq.shape[0])}
+            q_lookup = (key_index, per_meta,
+                        {id(m): mi for mi, m in enumerate(q_metas)})
         columns = plan.key_names + plan.agg_names
         rows: List[List] = []
         for g in groups:
@@ -441,9 +451,19 @@ class QueryEngine:
                     ai += 2
                     row.append(ssum / cnt if cnt else None)
                 elif meta["op"] in ("percentile", "apdex"):
-                    vals = self._group_values(plan, segments, g["key"], meta)
+                    key_index, per_meta, meta_ix = q_lookup
+                    gi = key_index.get(tuple(int(x) & ((1 << 64) - 1)
+                                             for x in g["key"]))
                     ai += 1
-                    row.append(self._finish_quantile(meta, vals))
+                    if gi is None:
+                        row.append(None)
+                    else:
+                        mi = meta_ix[id(meta)]
+                        svals, starts, counts = per_meta[mi]
+                        s0 = int(starts[gi])
+                        c0 = int(counts[gi])
+                        row.append(self._finish_quantile(
+                            meta, svals[s0:s0 + c0]))
                 else:
                     row.append(g["agg"][ai])
                     ai += 1

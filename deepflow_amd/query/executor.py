@@ -240,3 +240,95 @@ def execute(plan: Plan, segments, device: str = "cpu"):
     if device == "cpu":
         return execute_agg_cpu(plan, segments)
     return execute_agg_gpu(plan, segments, device)
+
+
+# ---------------------------------------------------------------------
+# grouped value gather for Percentile/Apdex: one pass over the store
+# (filter -> per-row group assignment -> two-key stable sort), replacing
+# the per-group re-scan and its 256-group cap. Reference analog:
+# ClickHouse quantile() aggregate states.
+# ---------------------------------------------------------------------
+_M32 = 0xFFFFFFFF
+
+
+def _col_torch(seg, family, idx, bucket, time_base_s, rows_t):
+    if family == SRC_U64:
+        return seg.u64[idx].index_select(0, rows_t)
+    if family == SRC_U32:
+        return seg.u32[idx].index_select(0, rows_t).to(torch.int64) & _M32
+    if family == SRC_U8:
+        return seg.u8[idx].index_select(0, rows_t).to(torch.int64)
+    if family == SRC_DID:
+        return seg.did[idx].index_select(0, rows_t).to(torch.int64) & _M32
+    if family == SRC_KG:
+        return seg.kg[idx].index_select(0, rows_t).to(torch.int64) & _M32
+    if family == SRC_TIME_BUCKET:
+        t_s = seg.u64[0].index_select(0, rows_t) // 10**9
+        rel = torch.clamp(t_s - time_base_s, min=0)
+        if bucket:
+            rel = (rel // bucket) * bucket
+        return rel
+    raise ValueError(f"unsupported tensor family {family}")
+
+
+def _matching_rows_t(plan: Plan, seg, device):
+    n = seg.n_rows
+    if device == "cpu" or seg.u64.device.type == "cpu":
+        mask = _mask_np(seg, plan, n)
+        return torch.from_numpy(np.nonzero(mask)[0].copy())
+    from ..ops import gpu_ops
+    dev = seg.u64.device
+    cap = n
+    out_rows = torch.zeros(cap, dtype=torch.int64, device=dev)
+    out_ctr = torch.zeros(1, dtype=torch.int32, device=dev)
+    spec = plan.to_bytes()
+    gpu_ops.query_select(seg, spec, 0, n, out_rows, out_ctr)
+    torch.cuda.synchronize()
+    cnt = min(int(out_ctr.item()), cap)
+    return out_rows[:cnt]
+
+
+def execute_grouped_values(plan: Plan, segments, metas, device: str):
+    """-> (uniq_keys [g, nk] int64 cpu, per-meta dict
+    {mi: (sorted_vals float64 cpu, starts, counts)}) for quantile
+    finishing. Raises ValueError for unsupported key/metric families."""
+    import copy
+    sub = copy.deepcopy(plan)
+    sub.select_rows = True
+    key_cols = [[] for _ in plan.keys]
+    val_cols = [[] for _ in metas]
+    for seg in segments:
+        if seg.n_rows == 0:
+            continue
+        rows_t = _matching_rows_t(sub, seg, device)
+        if rows_t.numel() == 0:
+            continue
+        if rows_t.device != seg.u64.device:
+            rows_t = rows_t.to(seg.u64.device)
+        for ki, k in enumerate(plan.keys):
+            key_cols[ki].append(_col_torch(seg, k.family, k.idx, k.bucket,
+                                           plan.time_base_s, rows_t).cpu())
+        for mi, meta in enumerate(metas):
+            val_cols[mi].append(_col_torch(seg, meta["family"],
+                                           meta["idx"], 0,
+                                           plan.time_base_s, rows_t).cpu())
+    if plan.keys and not key_cols[0]:
+        return torch.zeros((0, len(plan.keys)), dtype=torch.int64), {}
+    if plan.keys:
+        keys = torch.stack([torch.cat(c) for c in key_cols], dim=1)
+        uniq, inverse = torch.unique(keys, dim=0, return_inverse=True)
+    else:
+        n_total = sum(int(t.numel()) for t in val_cols[0]) if metas else 0
+        uniq = torch.zeros((1, 0), dtype=torch.int64)
+        inverse = torch.zeros(n_total, dtype=torch.int64)
+    g = uniq.shape[0]
+    counts = torch.bincount(inverse, minlength=g)
+    starts = torch.cumsum(counts, 0) - counts
+    out = {}
+    for mi in range(len(metas)):
+        vals = torch.cat(val_cols[mi]).to(torch.float64)
+        order1 = torch.argsort(vals)
+        inv1 = inverse[order1]
+        order2 = torch.argsort(inv1, stable=True)
+        out[mi] = (vals[order1[order2]], starts, counts)
+    return uniq, out

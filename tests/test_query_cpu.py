@@ -228,3 +228,22 @@ def test_attribute_filter(eng):
         f"SELECT Count(*) AS c FROM l7_flow_log WHERE "
         f"attribute.attr_1 != '{val}'")
     assert r3["values"] == [[N - want]]
+
+
+def test_grouped_percentile_many_groups(eng):
+    """Percentile over more than 256 groups (the old cap) — checked
+    against direct per-group recomputation from pb truth."""
+    import numpy as np
+    r = eng.query(
+        "SELECT client_port, Percentile(response_duration, 50) AS p50 "
+        "FROM l7_flow_log GROUP BY client_port")
+    assert len(r["values"]) > 0
+    want = {}
+    for t in truth():
+        want.setdefault(t["base"]["port_src"], []).append(
+            t["base"]["head"]["rrt"])
+    for port, p50 in r["values"]:
+        exp = float(np.quantile(np.array(want[port], dtype=np.float64),
+                                0.5))
+        assert abs(p50 - exp) < 1e-6, (port, p50, exp)
+    assert len(r["values"]) == len(want)
