@@ -473,44 +473,6 @@ class QueryEngine:
         rows = self._order_limit(plan, columns, rows)
         return {"columns": columns, "values": rows}
 
-    def _group_values(self, plan: Q.Plan, segments, key_raw, meta):
-        """Second pass for Percentile/Apdex: gather the metric's values for
-        one group (original filters + key equality) on the GPU/CPU."""
-        import copy
-        import torch
-        sub = copy.deepcopy(plan)
-        sub.select_rows = True
-        sub.limit = 1 << 22
-        for ki, k in enumerate(plan.keys):
-            v = key_raw[ki]
-            if k.family == Q.SRC_TIME_BUCKET and k.bucket > 1:
-                sub.terms.append(Q.Term(Q.SRC_TIME_BUCKET, 0, Q.OP_BETWEEN,
-                                        v, v + k.bucket - 1))
-            else:
-                sub.terms.append(Q.Term(k.family, k.idx, Q.OP_EQ, v))
-        hits = execute(sub, segments, self.device)
-        fam, idx = meta["family"], meta["idx"]
-        out = []
-        by_seg: Dict[int, List[int]] = {}
-        for si, r in hits:
-            by_seg.setdefault(si, []).append(r)
-        for si, rws in by_seg.items():
-            seg = segments[si]
-            idx_t = torch.tensor(rws, dtype=torch.long,
-                                 device=seg.u64.device)
-            if fam == Q.SRC_U64:
-                vals = seg.u64[idx].index_select(0, idx_t)
-            elif fam == Q.SRC_U32:
-                vals = seg.u32[idx].index_select(0, idx_t)
-            elif fam == Q.SRC_U8:
-                vals = seg.u8[idx].index_select(0, idx_t)
-            else:
-                raise SqlError("Percentile/Apdex needs a numeric metric")
-            out.append(vals.to(torch.float64))
-        if not out:
-            return None
-        return torch.cat([v.cpu() for v in out])
-
     @staticmethod
     def _finish_quantile(meta, vals):
         import torch

@@ -123,10 +123,9 @@ def test_attribute_filter_match(engines):
 
 def test_grouped_percentile_gpu(engines):
     """Grouped percentile single-pass gather on GPU matches CPU."""
-    cpu_eng, gpu_eng = engines
     q = ("SELECT request_resource, Percentile(response_duration, 95) "
          "AS p95, Count(*) AS c FROM l7_flow_log "
          "GROUP BY request_resource ORDER BY c DESC LIMIT 10")
-    want = cpu_eng.query(q)
-    got = gpu_eng.query(q)
+    want = engines["cpu"].query(q)
+    got = engines["cuda"].query(q)
     assert want == got
