@@ -14,7 +14,6 @@ import zlib
 from typing import Optional, Tuple
 
 from fastapi import FastAPI, Request
-from fastapi.responses import JSONResponse
 
 from ..wire import pb, metric, framing
 

@@ -11,7 +11,7 @@ scanner is the /proc-based path that also runs without eBPF.)
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 
 def scan_processes(limit: int = 2000) -> List[Dict]:

@@ -12,7 +12,7 @@ from typing import Dict, List
 import numpy as np
 import torch
 
-from ..store import l7_schema as S
+
 from .spec import (Plan, SRC_U64, SRC_U32, SRC_U8, SRC_DID, SRC_KG,
                    SRC_ATTR_VAL, SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
                    SRC_ATTR_MATCH, STR_FILTER_SEED,
