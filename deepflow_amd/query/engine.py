@@ -370,6 +370,12 @@ class QueryEngine:
             return {"kind": "rows",
                     "result": self._run_select(plan, segments, tags,
                                                str_cols)}
+        if any(m["op"] in ("percentile", "apdex") for m in plan.agg_meta):
+            # raw per-shard values would be needed for an exact merge;
+            # refuse loudly instead of shipping a wrong number
+            # (round-2: per-shard histogram exchange)
+            raise SqlError("Percentile/Apdex is not supported in "
+                           "distributed queries yet")
         groups = execute(plan, segments, self.device)
         key_rows = []
         aggs = []
