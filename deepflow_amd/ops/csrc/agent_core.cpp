@@ -1050,6 +1050,9 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
         uint32_t flen = (p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
         if (flen + 4 <= n + 4096) return 103;
     }
+    // Pulsar well-known port: even a partial first segment pins the
+    // protocol so the handler's reassembly can take over
+    if (server_port == 6650 && n >= 4) return 105;
     // Pulsar framed BaseCommand (strict: frame length matches packet)
     {
         uint32_t t, flen;
@@ -1639,11 +1642,26 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             f.l7.active = false;
         }
     } else if (f.l7_protocol == 47) {  // SOME/IP (possibly batched)
+        std::vector<uint8_t> merged47;
+        if (!f.h2_carry[dir].empty()) {
+            merged47.swap(f.h2_carry[dir]);
+            merged47.insert(merged47.end(), p, p + n);
+            p = merged47.data();
+            n = (uint32_t)merged47.size();
+        }
         uint32_t off = 0;
         while (off + 16 <= n) {
             uint16_t sv, me, cl, se;
             uint8_t mt, rc;
             uint32_t ml;
+            {   // message split across TCP segments: carry the tail
+                uint32_t want = ((p[off + 4] << 24) | (p[off + 5] << 16) |
+                                 (p[off + 6] << 8) | p[off + 7]) + 8;
+                if (want >= 16 && want > n - off && want < (128u << 10)) {
+                    f.h2_carry[dir].assign(p + off, p + n);
+                    break;
+                }
+            }
             if (!parse_someip(p + off, n - off, sv, me, cl, se, mt, rc, ml))
                 break;
             char svc[16], mth[16];
@@ -1675,6 +1693,8 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             }
             off += ml;
         }
+        if (off < n && n - off < (128u << 10))
+            f.h2_carry[dir].assign(p + off, p + n);
     } else if (f.l7_protocol == 106) {  // ZMTP
         uint32_t off = 0;
         // the first 64 bytes per direction are the greeting
@@ -1778,10 +1798,26 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             off += fr.frame_len;
         }
     } else if (f.l7_protocol == 105) {  // Pulsar
+        std::vector<uint8_t> merged105;
+        if (!f.h2_carry[dir].empty()) {
+            merged105.swap(f.h2_carry[dir]);
+            merged105.insert(merged105.end(), p, p + n);
+            p = merged105.data();
+            n = (uint32_t)merged105.size();
+        }
         uint32_t off = 0;
         while (off + 10 <= n) {
             uint32_t t, flen;
             std::string topic;
+            {
+                uint32_t total = (p[off] << 24) | (p[off + 1] << 16) |
+                                 (p[off + 2] << 8) | p[off + 3];
+                if (total >= 6 && total + 4 > n - off &&
+                    total < (128u << 10)) {
+                    f.h2_carry[dir].assign(p + off, p + n);
+                    break;
+                }
+            }
             if (!parse_pulsar(p + off, n - off, t, topic, flen)) break;
             const char* nm = pulsar_cmd_name(t);
             if (pulsar_is_request(t)) {
@@ -1807,13 +1843,27 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             // one-way types (MESSAGE/ACK/FLOW) tracked via counters only
             off += flen;
         }
+        if (off < n && n - off < (128u << 10))
+            f.h2_carry[dir].assign(p + off, p + n);
     } else if (f.l7_protocol == 103) {  // OpenWire
+        std::vector<uint8_t> merged103;
+        if (!f.h2_carry[dir].empty()) {
+            merged103.swap(f.h2_carry[dir]);
+            merged103.insert(merged103.end(), p, p + n);
+            p = merged103.data();
+            n = (uint32_t)merged103.size();
+        }
         uint32_t off = 0;
         while (off + 5 <= n) {
             uint32_t flen = (p[off] << 24) | (p[off + 1] << 16) |
                             (p[off + 2] << 8) | p[off + 3];
             if (flen < 1) break;
-            if (flen + 4 > n - off) flen = n - off - 4;  // truncated capture
+            if (flen + 4 > n - off && flen < (128u << 10)) {
+                // command split across segments: reassemble next payload
+                f.h2_carry[dir].assign(p + off, p + n);
+                break;
+            }
+            if (flen + 4 > n - off) flen = n - off - 4;  // oversized: clamp
             uint32_t fl = flen + 4;
             uint8_t t = p[off + 4];
             if (t >= 30 && t <= 34) {
@@ -1860,6 +1910,8 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             }
             off += fl;
         }
+        if (off < n && n - off < (128u << 10))
+            f.h2_carry[dir].assign(p + off, p + n);
     } else if (f.l7_protocol == 121) {  // TLS: ClientHello SNI only
         std::string sni;
         if (dir == 0 && parse_tls_client_hello(p, n, sni)) {

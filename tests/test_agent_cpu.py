@@ -208,3 +208,46 @@ def test_acl_pcap_action_end_to_end():
     assert len(srv.pcap.flows[fid]) >= 7  # handshake+data+teardown frames
     a.close()
     srv.stop()
+
+
+def test_pulsar_cross_segment_reassembly():
+    """A Pulsar frame split mid-header across TCP segments is carried
+    over and parsed whole (generic length-framed reassembly)."""
+    import struct
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import (eth_ipv4_tcp, SYN, SYNACK,
+                                            PSH_ACK)
+    from deepflow_amd.wire import pb, flow_log, framing
+
+    def pulsar_frame(tb):  # [totalSize][commandSize][BaseCommand pb]
+        return struct.pack(">II", len(tb) + 4, len(tb)) + tb
+
+    # LOOKUP(23) with topic; then LOOKUP_RESPONSE(24)
+    topic = b"persistent://public/default/seg-topic"
+    sub = bytes([0x0A, len(topic)]) + topic + b"\x10\x05"  # topic, req_id
+    lookup = bytes([0x08, 23, 0xBA, 0x01, len(sub)]) + sub
+    resp = bytes([0x08, 24])
+    req = pulsar_frame(lookup)
+    a = Agent(vtap_id=1)
+    C, S = 0x0A000001, 0x0A000002
+    t = 10**9
+    cut = 6  # split inside the 8-byte length header
+    pkts = [
+        (eth_ipv4_tcp(C, S, 41000, 6650, SYN, 1), t),
+        (eth_ipv4_tcp(S, C, 6650, 41000, SYNACK, 1, 2), t + 1),
+        (eth_ipv4_tcp(C, S, 41000, 6650, PSH_ACK, 2, 2, req[:cut]), t + 2),
+        (eth_ipv4_tcp(C, S, 41000, 6650, PSH_ACK, 2 + cut, 2, req[cut:]),
+         t + 3),
+        (eth_ipv4_tcp(S, C, 6650, 41000, PSH_ACK, 2, 2 + len(req),
+                      pulsar_frame(resp)), t + 10**7),
+    ]
+    for frame, ts in pkts:
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(a.drain(1))]
+    lk = [r for r in recs if r["req"].get("req_type") == "LOOKUP"]
+    assert lk, [r["req"].get("req_type") for r in recs]
+    assert lk[0]["req"]["resource"] == topic.decode()
+    assert lk[0]["base"]["head"]["proto"] == 105
+    a.close()
