@@ -961,6 +961,10 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
             uint8_t c0 = p[4];
             if (c0 <= 0x1F || c0 == 10) return 60;
         }
+        // well-known port: allow a message longer than this segment
+        // (reassembly pre-stage buffers the rest)
+        if (server_port == 3306 && plen >= 1 && plen + 4 > n && p[3] == 0)
+            return 60;
     }
     // PostgreSQL simple query: 'Q' + int32 BE length covering the packet
     if (n >= 6 && p[0] == 'Q') {
@@ -1308,6 +1312,22 @@ void encode_documents(Agent& a) {
 
 // ------------------------------------------------------------- packet path
 
+// total message length for single-message length-framed protocols
+// (0 = not length-framed / unknown)
+uint32_t framed_need(uint8_t proto, const uint8_t* p, uint32_t n) {
+    if (n < 4) return 0;
+    switch (proto) {
+        case 60:   // MySQL: [len3 LE][seq]
+            return 4 + (p[0] | (p[1] << 8) | (p[2] << 16));
+        case 81:   // MongoDB: msglen LE covers everything
+            return p[0] | (p[1] << 8) | (p[2] << 16) | (p[3] << 24);
+        case 100:  // Kafka: [len BE] + payload
+        case 107:  // RocketMQ: [len BE] + payload
+            return 4 + ((p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3]);
+    }
+    return 0;
+}
+
 void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                        uint32_t n, uint64_t ts) {
     if (f.l7_protocol == 0) {
@@ -1316,6 +1336,24 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             f.l7_protocol = infer_l7(p, n, f.port[1]);
     }
     if (f.l7_protocol == 0) return;
+    // generic reassembly pre-stage for single-message length-framed
+    // protocols: a message larger than one TCP segment is buffered per
+    // direction until complete (mysql result sets, kafka fetches, ...)
+    std::vector<uint8_t> fmerged;
+    if (f.l7_protocol == 60 || f.l7_protocol == 81 ||
+        f.l7_protocol == 100 || f.l7_protocol == 107) {
+        if (!f.h2_carry[dir].empty()) {
+            fmerged.swap(f.h2_carry[dir]);
+            fmerged.insert(fmerged.end(), p, p + n);
+            p = fmerged.data();
+            n = (uint32_t)fmerged.size();
+        }
+        uint32_t need = framed_need(f.l7_protocol, p, n);
+        if (need > n && need >= 4 && need < (1u << 20)) {
+            f.h2_carry[dir].assign(p, p + n);
+            return;  // wait for the rest of the message
+        }
+    }
     if (f.l7_protocol == 127) {  // custom protocol: generic session capture;
         // host-side plugins re-parse the raw prefix (wasm-plugin analog)
         if (dir == 0 && !f.l7.active) {

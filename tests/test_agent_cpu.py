@@ -251,3 +251,38 @@ def test_pulsar_cross_segment_reassembly():
     assert lk[0]["req"]["resource"] == topic.decode()
     assert lk[0]["base"]["head"]["proto"] == 105
     a.close()
+
+
+def test_mysql_cross_segment_reassembly():
+    """A MySQL query packet split across TCP segments is reassembled."""
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import (eth_ipv4_tcp, SYN, SYNACK,
+                                            PSH_ACK)
+    from deepflow_amd.wire import pb, flow_log, framing
+    stmt = b"SELECT col FROM really_long_table WHERE id = 42"
+    body = b"\x03" + stmt  # COM_QUERY
+    req = bytes([len(body) & 0xFF, (len(body) >> 8) & 0xFF,
+                 (len(body) >> 16) & 0xFF, 0]) + body
+    ok = bytes([7, 0, 0, 1, 0x00, 0, 0, 2, 0, 0, 0])  # OK packet
+    a = Agent(vtap_id=1)
+    C, S = 0x0A000001, 0x0A000002
+    t = 10**9
+    cut = 9
+    pkts = [
+        (eth_ipv4_tcp(C, S, 41010, 3306, SYN, 1), t),
+        (eth_ipv4_tcp(S, C, 3306, 41010, SYNACK, 1, 2), t + 1),
+        (eth_ipv4_tcp(C, S, 41010, 3306, PSH_ACK, 2, 2, req[:cut]), t + 2),
+        (eth_ipv4_tcp(C, S, 41010, 3306, PSH_ACK, 2 + cut, 2, req[cut:]),
+         t + 3),
+        (eth_ipv4_tcp(S, C, 3306, 41010, PSH_ACK, 2, 2 + len(req), ok),
+         t + 10**7),
+    ]
+    for frame, ts in pkts:
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(a.drain(1))]
+    got = [r for r in recs if r["base"]["head"]["proto"] == 60]
+    assert got, [r["base"]["head"].get("proto") for r in recs]
+    assert got[0]["req"]["resource"].startswith("SELECT col FROM")
+    a.close()
