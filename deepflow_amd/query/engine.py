@@ -529,6 +529,25 @@ class QueryEngine:
         return {"columns": cols, "values": rows}
 
     def _fetch(self, seg, row: int, col: str, tags, str_cols):
+        if col.startswith("attribute."):
+            # custom tag by name: scan the row's attr-id slots
+            # (layout: pool[start .. start+cnt) = name ids,
+            #  pool[start+cnt .. start+2cnt) = value ids)
+            from ..store.l7_schema import (DICT_DOM_ATTR_NAME,
+                                           DICT_DOM_ATTR_VALUE)
+            want = self.pipe.dict.lookup_id(DICT_DOM_ATTR_NAME,
+                                            col[len("attribute."):].encode())
+            if want is None:
+                return None
+            cnt = int(seg.attr_cnt[row])
+            start = int(seg.attr_start[row])
+            for i in range(cnt):
+                nid = int(seg.attr_pool[start + i]) & 0xFFFFFFFF
+                if nid == want:
+                    vid = int(seg.attr_pool[start + cnt + i]) & 0xFFFFFFFF
+                    return self.pipe.dict.hydrate(DICT_DOM_ATTR_VALUE,
+                                                  [vid])[0]
+            return None
         td = tags.get(col)
         if td is not None:
             fam, idx = td.family, td.idx

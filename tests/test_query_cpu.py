@@ -247,3 +247,20 @@ def test_grouped_percentile_many_groups(eng):
                                 0.5))
         assert abs(p50 - exp) < 1e-6, (port, p50, exp)
     assert len(r["values"]) == len(want)
+
+
+def test_select_attribute_column(eng):
+    """attribute.<name> as a SELECT column hydrates per-row values."""
+    t0 = gen_span_dict(CFG, 0)
+    names = t0["ext_info"]["attribute_names"]
+    vals = t0["ext_info"]["attribute_values"]
+    name = names[0]
+    r = eng.query(f"SELECT request_resource, attribute.{name} "
+                  f"FROM l7_flow_log LIMIT 400")
+    got = {tuple(row) for row in r["values"]}
+    # every truth row appears with its attribute value
+    for i in range(5):
+        t = gen_span_dict(CFG, i)
+        nm = dict(zip(t["ext_info"]["attribute_names"],
+                      t["ext_info"]["attribute_values"]))
+        assert (t["req"]["resource"], nm[name]) in got
