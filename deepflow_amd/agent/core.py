@@ -37,6 +37,8 @@ def _lib():
         lib.dfa_add_custom_port.argtypes = [p, u32]
         lib.dfa_packet.restype = ct.c_int
         lib.dfa_packet.argtypes = [p, p, u32, u64]
+        lib.dfa_packet_batch.restype = ct.c_int64
+        lib.dfa_packet_batch.argtypes = [p, p, u64]
         lib.dfa_tick.argtypes = [p, u64]
         lib.dfa_drain.restype = u64
         lib.dfa_drain.argtypes = [p, ct.c_int, p, u64]
@@ -95,6 +97,13 @@ class Agent:
         buf = np.frombuffer(frame, dtype=np.uint8)
         return self._lib.dfa_packet(self._h, buf.ctypes.data, len(frame),
                                     ts_ns)
+
+    def packet_batch(self, blob: bytes) -> int:
+        """Batch entry: [u32 len][u64 ts][frame]... packed; one native
+        call for the whole batch (capture rings, pps benches)."""
+        buf = np.frombuffer(blob, dtype=np.uint8)
+        return int(self._lib.dfa_packet_batch(self._h, buf.ctypes.data,
+                                              len(blob)))
 
     def tick(self, now_ns: int) -> None:
         self._lib.dfa_tick(self._h, now_ns)

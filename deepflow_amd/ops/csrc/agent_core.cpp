@@ -2134,6 +2134,27 @@ void dfa_add_cidr(void* h, uint32_t net, uint32_t masklen, int32_t epc) {
     a->cidrs.push_back({net & mask, mask, epc});
 }
 
+int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns);
+
+// Feed a batch of frames packed as [u32 len][u64 ts][frame]... — one
+// ctypes crossing per batch (AF_PACKET rings / pps benches).
+int64_t dfa_packet_batch(void* h, const uint8_t* buf, uint64_t total) {
+    uint64_t pos = 0;
+    int64_t n = 0;
+    while (pos + 12 <= total) {
+        uint32_t len;
+        uint64_t ts;
+        memcpy(&len, buf + pos, 4);
+        memcpy(&ts, buf + pos + 4, 8);
+        pos += 12;
+        if (pos + len > total) break;
+        dfa_packet(h, buf + pos, len, ts);
+        pos += len;
+        n++;
+    }
+    return n;
+}
+
 // Feed one raw Ethernet frame. Returns 0 ok, <0 parse error.
 int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
     Agent& a = *(Agent*)h;
