@@ -17,31 +17,47 @@ cfg = SpanGenConfig(n=2_000_000, seed=5, tag_cardinality=100_000,
 pay = gen_span_payload(cfg)
 
 
-def run(steps, profiler=None):
+from deepflow_amd.profiler.gpu_profiler import ContinuousGpuProfiler
+
+
+def run(steps, mode="off"):
     pipe = L7IngestPipeline(device="cuda", segment_rows=1 << 23,
                             dict_capacity=1 << 22,
                             time_base_s=cfg.base_time_ns // 10**9)
-    pipe.segments.reserve(steps * 2_000_000 // (1 << 23) + 2)
+    pipe.segments.reserve(4)
+    pipe.segments.max_bytes = 10 << 30   # recycle within the soak
     for _ in range(3):
         pipe.ingest_frame_payload(pay)          # warmup
     torch.cuda.synchronize()
+    prof = None
+    if mode == "full":
+        prof = GpuProfiler(ProfilePipeline())
+    elif mode == "sampled":
+        prof = ContinuousGpuProfiler(ProfilePipeline(), period=50)
     t0 = time.perf_counter()
-    if profiler is not None:
-        prof = GpuProfiler(profiler)
-        with prof.capture():
-            for _ in range(steps):
-                pipe.ingest_frame_payload(pay)
-            torch.cuda.synchronize()
-    else:
-        for _ in range(steps):
+    for _ in range(steps):
+        if mode == "off":
             pipe.ingest_frame_payload(pay)
-        torch.cuda.synchronize()
-    return time.perf_counter() - t0
+        elif mode == "full":
+            with prof.capture():
+                pipe.ingest_frame_payload(pay)
+        else:
+            with prof.step():
+                pipe.ingest_frame_payload(pay)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    return dt, prof.captures if prof else 0
 
 
-steps = 15
-base = min(run(steps), run(steps))
-profiled = run(steps, profiler=ProfilePipeline())
-print({"steps": steps, "base_s": round(base, 3),
-       "profiled_s": round(profiled, 3),
-       "overhead_pct": round((profiled - base) / base * 100, 2)})
+steps = 100
+base, _ = min(run(steps), run(steps))
+sampled, caps = run(steps, mode="sampled")
+full, _ = run(20, mode="full")
+base20 = base / steps * 20
+print({"steps": steps,
+       "base_ms_per_step": round(base / steps * 1e3, 2),
+       "sampled_ms_per_step": round(sampled / steps * 1e3, 2),
+       "sampled_overhead_pct": round((sampled - base) / base * 100, 2),
+       "captures": caps,
+       "full_trace_overhead_pct": round((full - base20) / base20 * 100,
+                                        1)})
