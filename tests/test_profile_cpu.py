@@ -69,3 +69,23 @@ def test_zstd_compressed_data():
     pipe.ingest_profile(d)
     assert pipe.store.id_to_loc == [b"x;y;z"]
     assert pipe.store.rows[0].value == 11
+
+
+def test_continuous_profiler_duty_cycle():
+    """period/window duty logic: captures only its share of steps."""
+    import contextlib
+    from deepflow_amd.profiler.gpu_profiler import ContinuousGpuProfiler
+
+    class Probe(ContinuousGpuProfiler):
+        @contextlib.contextmanager
+        def capture(self):
+            self.captures += 1
+            yield None
+
+    prof = Probe(pipeline=None, period=5, window=2)
+    ran = 0
+    for _ in range(20):
+        with prof.step():
+            ran += 1
+    assert ran == 20
+    assert prof.captures == 8  # 2 of every 5 steps
