@@ -16,7 +16,7 @@ CSRC = OPS_DIR / "csrc"
 CPU_LIB = OPS_DIR / "libdfcpu.so"
 GPU_LIB = OPS_DIR / "libdfgpu.so"
 
-CPU_SOURCES = ["dfcpu.cpp", "agent_core.cpp"]
+CPU_SOURCES = ["dfcpu.cpp", "agent_core.cpp", "otlp_conv.cpp"]
 GPU_SOURCES = ["dfgpu.hip"]
 
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")

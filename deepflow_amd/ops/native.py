@@ -62,6 +62,9 @@ def cpu() -> ct.CDLL:
         lib.df_gen_spans_parallel.argtypes = [
             ct.POINTER(SpanCfgC), ct.c_uint64, ct.c_uint64, ct.c_void_p,
             ct.c_uint64, ct.c_void_p, ct.c_void_p]
+        lib.df_otlp_to_l7.restype = ct.c_int64
+        lib.df_otlp_to_l7.argtypes = [ct.c_void_p, ct.c_uint64,
+                                      ct.c_void_p, ct.c_uint64]
         lib.df_scan_offsets.restype = ct.c_uint64
         lib.df_scan_offsets.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p,
                                         ct.c_void_p, ct.c_uint64]

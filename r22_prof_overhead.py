@@ -24,8 +24,8 @@ def run(steps, mode="off"):
     pipe = L7IngestPipeline(device="cuda", segment_rows=1 << 23,
                             dict_capacity=1 << 22,
                             time_base_s=cfg.base_time_ns // 10**9)
-    pipe.segments.reserve(4)
-    pipe.segments.max_bytes = 10 << 30   # recycle within the soak
+    # provision like bench.py: no allocator or watermark work in the loop
+    pipe.segments.reserve(steps * 2_000_000 // (1 << 23) + 2)
     for _ in range(3):
         pipe.ingest_frame_payload(pay)          # warmup
     torch.cuda.synchronize()
@@ -33,7 +33,7 @@ def run(steps, mode="off"):
     if mode == "full":
         prof = GpuProfiler(ProfilePipeline())
     elif mode == "sampled":
-        prof = ContinuousGpuProfiler(ProfilePipeline(), period=50)
+        prof = ContinuousGpuProfiler(ProfilePipeline(), period=100)
     t0 = time.perf_counter()
     for _ in range(steps):
         if mode == "off":
