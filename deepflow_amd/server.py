@@ -313,20 +313,33 @@ class DeepflowServer:
     def _on_otel(self, hdr, payload) -> None:
         """OTLP frames carry one zlib-compressed TracesData blob
         (reference decoder.go:235-266); convert to AppProtoLogsData and run
-        the normal span pipeline."""
+        the normal span pipeline. Conversion runs in C++ (OpenMP over
+        spans, ops/csrc/otlp_conv.cpp — byte-identical to the Python
+        twin) with the Python converter as fallback."""
         import numpy as np
-        from .ingest.otel import otlp_to_l7_payload
+        import zlib
         data = payload.tobytes()
         try:
-            l7_payload = otlp_to_l7_payload(data, compressed=True)
+            data = zlib.decompress(data)
+        except zlib.error:
+            pass  # uncompressed push
+        from .ops import native
+        lib = native.cpu()
+        src = np.frombuffer(data, dtype=np.uint8)
+        need = lib.df_otlp_to_l7(src.ctypes.data, len(src), None, 0)
+        if need > 0:
+            dst = np.zeros(int(need), dtype=np.uint8)
+            lib.df_otlp_to_l7(src.ctypes.data, len(src),
+                              dst.ctypes.data, int(need))
+            self._on_l7(hdr, dst)
+            return
+        from .ingest.otel import otlp_to_l7_payload
+        try:
+            l7_payload = otlp_to_l7_payload(data, compressed=False)
         except Exception:
-            try:
-                l7_payload = otlp_to_l7_payload(data, compressed=False)
-            except Exception:
-                self.receiver.counter.add("otel_decode_errors")
-                return
-        arr = np.frombuffer(l7_payload, dtype=np.uint8)
-        self._on_l7(hdr, arr)
+            self.receiver.counter.add("otel_decode_errors")
+            return
+        self._on_l7(hdr, np.frombuffer(l7_payload, dtype=np.uint8))
 
     def _on_third_party(self, hdr, payload, kind: str) -> None:
         import numpy as np

@@ -109,3 +109,57 @@ def test_config_push_version_bump(server):
                                         "config_version": v0,
                                         "platform_version": 999})
     assert r2.json()["config"]["throttle_per_second"] == 1000
+
+
+def test_native_otlp_converter_matches_python():
+    """C++ df_otlp_to_l7 output is byte-identical to the Python
+    converter across span shapes (http/grpc/db/plain + statuses)."""
+    import numpy as np
+    from deepflow_amd.wire import pb, otlp
+    from deepflow_amd.ingest.otel import otlp_to_l7_payload
+    from deepflow_amd.ops import native
+
+    def kv(k, v):
+        return {"key": k, "value": {"string_value": v}}
+
+    spans = [
+        {"trace_id": bytes.fromhex("aa" * 16), "span_id": b"\x01" * 8,
+         "name": "GET /x", "kind": 3,
+         "start_time_unix_nano": 10**18,
+         "end_time_unix_nano": 10**18 + 5 * 10**6,
+         "attributes": [kv("http.method", "GET"), kv("http.target", "/x"),
+                        kv("http.host", "h"), kv("team", "core"),
+                        {"key": "http.status_code",
+                         "value": {"int_value": 503}},
+                        {"key": "retries", "value": {"int_value": 2}},
+                        {"key": "cached", "value": {"bool_value": True}}],
+         "status": {"code": 2}},
+        {"trace_id": bytes.fromhex("bb" * 16), "span_id": b"\x02" * 8,
+         "parent_span_id": b"\x03" * 8, "name": "svc.Api/Do", "kind": 2,
+         "start_time_unix_nano": 10**18,
+         "end_time_unix_nano": 10**18 + 10**6,
+         "attributes": [kv("rpc.system", "grpc"), kv("rpc.method", "Do"),
+                        kv("rpc.service", "svc.Api")]},
+        {"name": "lonely-span", "kind": 0,
+         "start_time_unix_nano": 5, "end_time_unix_nano": 9},
+        {"name": "q", "kind": 2,
+         "start_time_unix_nano": 1, "end_time_unix_nano": 2,
+         "attributes": [kv("db.system", "redis"), kv("db.operation", "GET"),
+                        kv("db.name", "cache"),
+                        kv("db.statement", "GET k")]},
+    ]
+    blob = pb.encode({"resource_spans": [{
+        "resource": {"attributes": [kv("service.name", "checkout")]},
+        "scope_spans": [{"spans": spans}]}]}, otlp.TRACES_DATA)
+    want = otlp_to_l7_payload(blob)
+    lib = native.cpu()
+    src = np.frombuffer(blob, dtype=np.uint8)
+    need = int(lib.df_otlp_to_l7(src.ctypes.data, len(src), None, 0))
+    dst = np.zeros(need, dtype=np.uint8)
+    got = int(lib.df_otlp_to_l7(src.ctypes.data, len(src),
+                                dst.ctypes.data, need))
+    assert got == need
+    assert dst.tobytes() == want
+    # malformed input refuses cleanly
+    bad = np.frombuffer(b"\xff\xff\xff\xff\x02", dtype=np.uint8)
+    assert lib.df_otlp_to_l7(bad.ctypes.data, len(bad), None, 0) <= 0
