@@ -57,23 +57,33 @@ class GpuProfiler:
 
 class ContinuousGpuProfiler(GpuProfiler):
     """Duty-cycled continuous profiling: capture `window` steps out of
-    every `period` (the reference's <1%-overhead profilers sample the
-    same way — full tracing of every step costs ~50%). Flame graphs
-    accumulate across capture windows."""
+    every `period` steps AND at most one window per `interval_s`
+    seconds. A kineto/roctracer capture window costs ~0.7 s of session
+    setup (measured, profiles/profiler_overhead_r01.txt), so the
+    time cadence — not the step duty cycle — is what keeps steady-state
+    overhead below 1% (0.7 s / 120 s ~= 0.6%); full per-step tracing
+    costs ~50% and exists only for debugging. Flame graphs accumulate
+    across capture windows."""
 
     def __init__(self, pipeline, process_name: str = "deepflow-gpu",
-                 period: int = 100, window: int = 1):
+                 period: int = 100, window: int = 1,
+                 interval_s: float = 120.0):
         super().__init__(pipeline, process_name)
         self.period = max(period, 1)
         self.window = max(window, 1)
+        self.interval_s = interval_s
         self._step = 0
+        self._last_capture = 0.0
 
     @contextlib.contextmanager
     def step(self):
         """Wrap one unit of work; traces only inside the duty window."""
         i = self._step
         self._step += 1
-        if i % self.period < self.window:
+        due = (i % self.period < self.window and
+               time.monotonic() - self._last_capture >= self.interval_s)
+        if due:
+            self._last_capture = time.monotonic()
             with self.capture():
                 yield
         else:

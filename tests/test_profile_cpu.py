@@ -82,7 +82,7 @@ def test_continuous_profiler_duty_cycle():
             self.captures += 1
             yield None
 
-    prof = Probe(pipeline=None, period=5, window=2)
+    prof = Probe(pipeline=None, period=5, window=2, interval_s=0.0)
     ran = 0
     for _ in range(20):
         with prof.step():
