@@ -172,6 +172,17 @@ class Agent:
         self.exceptions = exc
         return exc
 
+    # escape timer: if the controller stays unreachable past escape_s,
+    # the agent self-disables its capture (reference
+    # rpc/synchronizer.rs:1464 escape behavior)
+    ESCAPE_S_DEFAULT = 3600
+
+    def escaped(self, now_s: float, escape_s: Optional[float] = None) -> bool:
+        last = getattr(self, "last_sync_ok", None)
+        if last is None:
+            return False  # never synced: standalone mode
+        return (now_s - last) > (escape_s or self.ESCAPE_S_DEFAULT)
+
     # ---------------------------------------------------------- sync
     def sync_with_controller(self, post_fn) -> dict:
         """One trident-Synchronizer cycle: report versions + exceptions,
@@ -192,6 +203,8 @@ class Agent:
                 # epc labeling for every known endpoint (/32)
                 self.add_cidr(entry["ip"], 32, entry["epc"])
             self.platform_version = resp["platform_version"]
+        import time as _time
+        self.last_sync_ok = _time.time()
         return resp
 
     def dfstats_payload(self, now_s: int = 0) -> bytes:

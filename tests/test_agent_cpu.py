@@ -322,3 +322,19 @@ def test_packet_batch_and_pps():
     pps = len(frames) * 20 / dt
     assert pps > 300_000, f"agent path too slow: {pps:.0f} pps"
     a2.close()
+
+
+def test_escape_timer():
+    """Controller unreachable past the escape window -> agent reports
+    escaped (callers stop feeding, like the reference's self-disable)."""
+    from deepflow_amd.agent import Agent
+    a = Agent(vtap_id=1)
+    assert a.escaped(10**9) is False          # never synced: standalone
+    a.sync_with_controller(lambda body: {"status": "ok",
+                                         "config_version": 0,
+                                         "platform_version": 0})
+    now = a.last_sync_ok
+    assert a.escaped(now + 10) is False
+    assert a.escaped(now + 7200) is True
+    assert a.escaped(now + 7200, escape_s=10**6) is False
+    a.close()
