@@ -255,6 +255,31 @@ class DeepflowServer:
             n = self.prom.ingest_labeled_samples(samples)
             return {"accepted": n}
 
+        # UDP debug command bus (deepflow-ctl ingester analog)
+        from .utils.debug_bus import DebugBus
+        from .utils.stats import default_registry as _dr
+        self.debug_bus = DebugBus()
+        self.debug_bus.register(
+            "stats", lambda req: _dr().snapshot_all())
+        self.debug_bus.register(
+            "store", lambda req: {
+                "l7_rows": self.l7.segments.n_rows,
+                "l4_rows": self.l4.segments.n_rows,
+                "l7_segments": len(self.l7.segments.segments),
+                "l7_cold": len(getattr(self.l7.segments, "cold", [])),
+                "dict_entries": self.l7.dict.n_entries(),
+                "evicted_rows": self.l7.segments.evicted_rows,
+            })
+        self.debug_bus.register(
+            "queues", lambda req: {
+                "decode_queue": self.receiver._queue.qsize(),
+                "drops": self.receiver.counter.snapshot(),
+            })
+        self.debug_bus.register(
+            "agents", lambda req: {
+                f"{aid}/{mt}": {"frames": st.frames, "bytes": st.bytes}
+                for (aid, mt), st in self.receiver.status.items()})
+
         self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
@@ -379,9 +404,11 @@ class DeepflowServer:
 
     def start(self) -> None:
         self.receiver.start()
+        self.debug_bus.start()
 
     def stop(self) -> None:
         self.receiver.stop()
+        self.debug_bus.stop()
 
     def serve_http(self, host: str = "127.0.0.1", port: int = 20416) -> None:
         import uvicorn

@@ -63,3 +63,36 @@ def test_segment_eviction():
     assert len(ss.segments) <= 3
     assert ss.evicted_segments >= 2
     assert ss.evicted_rows == ss.evicted_segments * 64
+
+
+def test_debug_bus():
+    """UDP command bus: stats/store/queues served off the data plane."""
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.utils.debug_bus import debug_call
+    from deepflow_amd.wire import pb, flow_log, framing
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 11)
+    srv.start()
+    try:
+        rec = {"base": {"start_time": 10**18, "end_time": 10**18 + 10**6,
+                        "flow_id": 1, "vtap_id": 1, "tap_side": 1,
+                        "head": {"proto": 20, "msg_type": 2, "rrt": 10},
+                        "ip_src": 1, "ip_dst": 2, "port_src": 9,
+                        "port_dst": 80, "protocol": 6},
+               "req": {"req_type": "GET", "domain": "d", "resource": "/r",
+                       "endpoint": "/r"}}
+        srv.receiver.handle_frame(framing.encode_frame(
+            framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG, agent_id=7),
+            framing.pack_records([pb.encode(rec,
+                                            flow_log.APP_PROTO_LOGS_DATA)])))
+        r = debug_call(srv.debug_bus.port, "store")
+        assert r["result"]["l7_rows"] == 1
+        assert r["result"]["dict_entries"] > 0
+        r2 = debug_call(srv.debug_bus.port, "agents")
+        assert any(k.startswith("7/") for k in r2["result"])
+        r3 = debug_call(srv.debug_bus.port, "nope")
+        assert "error" in r3 and "store" in r3["cmds"]
+        r4 = debug_call(srv.debug_bus.port, "queues")
+        assert "decode_queue" in r4["result"]
+    finally:
+        srv.stop()
