@@ -85,32 +85,37 @@ StrView rd_str(Rd& r) {
     return s;
 }
 
-// AnyValue -> string form (matches _attr_val in otel.py)
-void any_value_str(Rd r, std::string& out) {
+// AnyValue -> string view; non-string scalars render into `scratch`
+// (no heap traffic on the hot path — matches _attr_val in otel.py)
+StrView any_value_view(Rd r, char* scratch, size_t scratch_cap) {
     uint32_t wt;
     while (uint32_t f = r.tag(wt)) {
         if (f == 1 && wt == 2) {           // string_value
-            StrView s = rd_str(r);
-            out.assign(s.p, s.n);
-            return;
+            return rd_str(r);
         } else if (f == 2 && wt == 0) {    // bool
-            out = r.varint() ? "true" : "false";
-            return;
+            const char* b = r.varint() ? "true" : "false";
+            return {b, (uint32_t)strlen(b)};
         } else if (f == 3 && wt == 0) {    // int
-            out = std::to_string((int64_t)r.varint());
-            return;
+            int n = snprintf(scratch, scratch_cap, "%lld",
+                             (long long)(int64_t)r.varint());
+            return {scratch, (uint32_t)n};
         } else if (f == 4 && wt == 1) {    // double
             double d;
             uint64_t v = r.fixed64();
             memcpy(&d, &v, 8);
-            char buf[32];
-            snprintf(buf, sizeof buf, "%g", d);
-            out = buf;
-            return;
+            int n = snprintf(scratch, scratch_cap, "%g", d);
+            return {scratch, (uint32_t)n};
         } else {
             r.skip(wt);
         }
     }
+    return {};
+}
+
+void any_value_str(Rd r, std::string& out) {
+    char scratch[32];
+    StrView v = any_value_view(r, scratch, sizeof scratch);
+    out.assign(v.p ? v.p : "", v.n);
 }
 
 struct Attr {
@@ -144,26 +149,42 @@ bool well_known(const std::string& k) {
     return false;
 }
 
-// convert one OTLP Span submessage -> AppProtoLogsData record bytes
+// convert one OTLP Span submessage -> AppProtoLogsData record bytes.
+// Allocation-free per span: attrs and ids live in stack views/buffers.
+constexpr uint32_t MAX_ATTRS = 48;
+
+struct AttrSlot {
+    StrView key, val;
+    char scratch[32];
+};
+
 void convert_span(const SpanView& sv, std::vector<uint8_t>& out) {
     Rd r{sv.p, sv.n};
-    std::string trace_id, span_id, parent_id, name;
+    char trace_hex[64], span_hex[64], parent_hex[64];
+    uint32_t trace_n = 0, span_n = 0, parent_n = 0;
+    StrView name;
     uint64_t kind = 0, t0 = 0, t1 = 0, status_code = 0;
-    std::vector<Attr> attrs;
+    AttrSlot attrs[MAX_ATTRS];
+    uint32_t n_attrs = 0;
+    auto hex_into = [](StrView s, char* dst, uint32_t cap) -> uint32_t {
+        static const char* H = "0123456789abcdef";
+        uint32_t n = s.n * 2 > cap ? cap / 2 : s.n;
+        for (uint32_t i = 0; i < n; i++) {
+            dst[2 * i] = H[(uint8_t)s.p[i] >> 4];
+            dst[2 * i + 1] = H[(uint8_t)s.p[i] & 0xF];
+        }
+        return n * 2;
+    };
     uint32_t wt;
     while (uint32_t f = r.tag(wt)) {
         if (f == 1 && wt == 2) {
-            StrView s = rd_str(r);
-            hex_of((const uint8_t*)s.p, s.n, trace_id);
+            trace_n = hex_into(rd_str(r), trace_hex, sizeof trace_hex);
         } else if (f == 2 && wt == 2) {
-            StrView s = rd_str(r);
-            hex_of((const uint8_t*)s.p, s.n, span_id);
+            span_n = hex_into(rd_str(r), span_hex, sizeof span_hex);
         } else if (f == 4 && wt == 2) {
-            StrView s = rd_str(r);
-            hex_of((const uint8_t*)s.p, s.n, parent_id);
+            parent_n = hex_into(rd_str(r), parent_hex, sizeof parent_hex);
         } else if (f == 5 && wt == 2) {
-            StrView s = rd_str(r);
-            name.assign(s.p, s.n);
+            name = rd_str(r);
         } else if (f == 6 && wt == 0) {
             kind = r.varint();
         } else if (f == 7 && wt == 1) {
@@ -172,19 +193,22 @@ void convert_span(const SpanView& sv, std::vector<uint8_t>& out) {
             t1 = r.fixed64();
         } else if (f == 9 && wt == 2) {  // KeyValue
             Rd kv = r.sub();
-            Attr a;
+            if (n_attrs >= MAX_ATTRS) continue;
+            AttrSlot& a = attrs[n_attrs];
             uint32_t wt2;
+            bool got = false;
             while (uint32_t f2 = kv.tag(wt2)) {
                 if (f2 == 1 && wt2 == 2) {
-                    StrView s = rd_str(kv);
-                    a.key.assign(s.p, s.n);
+                    a.key = rd_str(kv);
+                    got = true;
                 } else if (f2 == 2 && wt2 == 2) {
-                    any_value_str(kv.sub(), a.val);
+                    a.val = any_value_view(kv.sub(), a.scratch,
+                                           sizeof a.scratch);
                 } else {
                     kv.skip(wt2);
                 }
             }
-            attrs.push_back(std::move(a));
+            if (got) n_attrs++;
         } else if (f == 15 && wt == 2) {  // Status
             Rd st = r.sub();
             uint32_t wt2;
@@ -196,16 +220,24 @@ void convert_span(const SpanView& sv, std::vector<uint8_t>& out) {
             r.skip(wt);
         }
     }
-    auto attr = [&](const char* k) -> const std::string* {
-        for (auto& a : attrs)
-            if (a.key == k) return &a.val;
+    auto attr = [&](const char* k) -> const StrView* {
+        for (uint32_t i = 0; i < n_attrs; i++)
+            if (attrs[i].key.eq(k)) return &attrs[i].val;
         return nullptr;
+    };
+    auto to_i = [](const StrView* v) -> int64_t {
+        if (!v || !v->n) return 0;
+        char buf[24];
+        uint32_t n = v->n < 23 ? v->n : 23;
+        memcpy(buf, v->p, n);
+        buf[n] = 0;
+        return atoll(buf);
     };
     uint32_t tap_side = kind == 3 ? 1 : (kind == 2 ? 2 : 0);
     uint32_t proto = 0;
-    std::string req_type, domain, resource, endpoint;
+    StrView req_type, domain, resource, endpoint;
     int64_t code = 0;
-    const std::string *v, *v2;
+    const StrView *v, *v2;
     if ((v = attr("http.method")) || (v = attr("http.request.method"))) {
         proto = 20;
         req_type = *v;
@@ -217,19 +249,19 @@ void convert_span(const SpanView& sv, std::vector<uint8_t>& out) {
         endpoint = name;
         if ((v2 = attr("http.status_code")) ||
             (v2 = attr("http.response.status_code")))
-            code = atoll(v2->c_str());
+            code = to_i(v2);
     } else if (attr("rpc.system")) {
         proto = 41;
         if ((v2 = attr("rpc.method"))) req_type = *v2;
         if ((v2 = attr("rpc.service"))) domain = *v2;
         resource = name;
         endpoint = name;
-        if ((v2 = attr("rpc.grpc.status_code"))) code = atoll(v2->c_str());
+        if ((v2 = attr("rpc.grpc.status_code"))) code = to_i(v2);
     } else if ((v = attr("db.system"))) {
-        proto = *v == "redis" ? 80 : 60;
+        proto = v->eq("redis") ? 80 : 60;
         if ((v2 = attr("db.operation"))) req_type = *v2;
         if ((v2 = attr("db.name"))) domain = *v2;
-        if ((v2 = attr("db.statement")) && !v2->empty()) resource = *v2;
+        if ((v2 = attr("db.statement")) && v2->n) resource = *v2;
         else resource = name;
         endpoint = req_type;
     } else {
@@ -238,6 +270,17 @@ void convert_span(const SpanView& sv, std::vector<uint8_t>& out) {
     }
     uint32_t status = status_code != 2 ? 0
                       : (code >= 400 && code < 500 ? 4 : 3);
+    auto wk = [](StrView k) {
+        static const char* W[] = {
+            "http.method", "http.request.method", "http.target",
+            "url.path", "http.url", "http.host", "server.address",
+            "http.status_code", "http.response.status_code", "rpc.system",
+            "rpc.method", "rpc.service", "rpc.grpc.status_code",
+            "db.system", "db.operation", "db.name", "db.statement"};
+        for (const char* w : W)
+            if (k.eq(w)) return true;
+        return false;
+    };
     uint8_t buf[16384];
     dfpb::Buf b{buf, 0, sizeof buf};
     dfpb::f_m<512>(b, 1, [&](dfpb::Buf& s) {  // base
@@ -251,28 +294,28 @@ void convert_span(const SpanView& sv, std::vector<uint8_t>& out) {
         });
     });
     dfpb::f_m<4096>(b, 11, [&](dfpb::Buf& s) {  // req
-        dfpb::f_s(s, 1, req_type.data(), req_type.size());
-        dfpb::f_s(s, 2, domain.data(), domain.size());
-        dfpb::f_s(s, 3, resource.data(), resource.size());
-        dfpb::f_s(s, 4, endpoint.data(), endpoint.size());
+        dfpb::f_s(s, 1, req_type.p, req_type.n);
+        dfpb::f_s(s, 2, domain.p, domain.n);
+        dfpb::f_s(s, 3, resource.p, resource.n);
+        dfpb::f_s(s, 4, endpoint.p, endpoint.n);
     });
     dfpb::f_m<64>(b, 12, [&](dfpb::Buf& s) {  // resp
         dfpb::f_u(s, 1, status);
         dfpb::f_i(s, 2, code);
     });
     dfpb::f_m<256>(b, 14, [&](dfpb::Buf& s) {  // trace_info
-        dfpb::f_s(s, 1, trace_id.data(), trace_id.size());
-        dfpb::f_s(s, 2, span_id.data(), span_id.size());
-        dfpb::f_s(s, 3, parent_id.data(), parent_id.size());
+        dfpb::f_s(s, 1, trace_hex, trace_n);
+        dfpb::f_s(s, 2, span_hex, span_n);
+        dfpb::f_s(s, 3, parent_hex, parent_n);
     });
     dfpb::f_m<8192>(b, 15, [&](dfpb::Buf& s) {  // ext_info
         dfpb::f_s(s, 1, sv.service.data(), sv.service.size());
-        for (auto& a : attrs)
-            if (!well_known(a.key))
-                dfpb::f_s(s, 16, a.key.data(), a.key.size());
-        for (auto& a : attrs)
-            if (!well_known(a.key))
-                dfpb::f_s(s, 17, a.val.data(), a.val.size());
+        for (uint32_t i = 0; i < n_attrs; i++)
+            if (!wk(attrs[i].key))
+                dfpb::f_s(s, 16, attrs[i].key.p, attrs[i].key.n);
+        for (uint32_t i = 0; i < n_attrs; i++)
+            if (!wk(attrs[i].key))
+                dfpb::f_s(s, 17, attrs[i].val.p, attrs[i].val.n);
     });
     uint32_t len = (uint32_t)b.len;
     size_t base = out.size();
