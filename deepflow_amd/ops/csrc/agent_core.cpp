@@ -932,6 +932,67 @@ const char* openwire_cmd_name(uint8_t t) {
     }
 }
 
+// Oracle TNS: 8-byte header [len u16 BE][cksum u16][type u8][flags u8]
+// [hdr cksum u16]; types: 1 CONNECT, 2 ACCEPT, 4 REFUSE, 5 REDIRECT,
+// 6 DATA, 12 MARKER. CONNECT carries the (DESCRIPTION=...) string with
+// SERVICE_NAME; DATA carries TTC where SQL text appears inline.
+bool parse_tns(const uint8_t* p, uint32_t n, uint8_t& type,
+               uint32_t& plen) {
+    if (n < 8) return false;
+    plen = (p[0] << 8) | p[1];
+    type = p[4];
+    if (plen < 8 || plen > n || type < 1 || type > 19) return false;
+    return true;
+}
+
+// find an SQL statement inside a TNS DATA payload (TTC): scan for a
+// leading keyword and take the printable run
+bool tns_extract_sql(const uint8_t* p, uint32_t n, std::string& sql,
+                     std::string& verb) {
+    static const char* KW[] = {"SELECT ", "INSERT ", "UPDATE ", "DELETE ",
+                               "MERGE ", "BEGIN ", "COMMIT", "ROLLBACK",
+                               "ALTER ", "CREATE ", "select ", "insert ",
+                               "update ", "delete ", "begin "};
+    for (uint32_t i = 8; i + 8 < n; i++) {
+        for (const char* kw : KW) {
+            size_t kl = strlen(kw);
+            if (i + kl <= n && memcmp(p + i, kw, kl) == 0) {
+                uint32_t end = i;
+                while (end < n && p[end] >= 0x20 && p[end] <= 0x7E &&
+                       end - i < 512)
+                    end++;
+                sql.assign((const char*)p + i, end - i);
+                verb.assign(kw, kl);
+                while (!verb.empty() && verb.back() == ' ')
+                    verb.pop_back();
+                for (auto& c : verb) c = toupper(c);
+                return true;
+            }
+        }
+    }
+    return false;
+}
+
+// ISO 8583 (financial messages): [u16 BE length][MTI 4 ASCII digits]
+// [binary bitmap 8B (+8B secondary when bit 1 set)][fields...]. A
+// response MTI is request MTI + 10 (0200 -> 0210).
+bool parse_iso8583(const uint8_t* p, uint32_t n, char mti[5],
+                   bool& is_resp) {
+    if (n < 14) return false;
+    uint32_t ln = (p[0] << 8) | p[1];
+    if (ln != n - 2) return false;
+    for (int i = 0; i < 4; i++) {
+        if (p[2 + i] < '0' || p[2 + i] > '9') return false;
+        mti[i] = (char)p[2 + i];
+    }
+    mti[4] = 0;
+    if (mti[0] > '2') return false;     // version 0-2 (1987/93/2003)
+    uint8_t cls = mti[1];
+    if (cls < '1' || cls > '8') return false;
+    is_resp = ((mti[2] - '0') & 1) == 1;  // function x1x/x3x = response
+    return true;
+}
+
 // in-flow protocol inference (reference: in-kernel infer_protocol + per-
 // parser check_payload)
 uint8_t infer_l7_custom(const Agent& a, uint16_t server_port) {
@@ -1046,6 +1107,18 @@ uint8_t infer_l7(const uint8_t* p, uint32_t n, uint16_t server_port) {
         uint32_t opc = p[12] | (p[13] << 8) | (p[14] << 16) | (p[15] << 24);
         if (mlen == n && (opc == 2013 || opc == 2004 || opc == 2010))
             return 81;
+    }
+    // ISO 8583 framed financial message
+    {
+        char mti[5];
+        bool ir;
+        if (parse_iso8583(p, n, mti, ir)) return 48;
+    }
+    // Oracle TNS on the well-known listener port
+    {
+        uint8_t tt;
+        uint32_t pl;
+        if (server_port == 1521 && parse_tns(p, n, tt, pl)) return 62;
     }
     // ZMTP greeting signature (always the first bytes on the wire)
     if (n >= 10 && p[0] == 0xFF && p[9] == 0x7F) return 106;
@@ -1950,6 +2023,71 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
         }
         if (off < n && n - off < (128u << 10))
             f.h2_carry[dir].assign(p + off, p + n);
+    } else if (f.l7_protocol == 62) {  // Oracle TNS
+        uint8_t tt;
+        uint32_t pl;
+        if (!parse_tns(p, n, tt, pl)) return;
+        if (tt == 1 && dir == 0) {  // CONNECT: extract SERVICE_NAME
+            std::string svc;
+            const char* key = "SERVICE_NAME=";
+            for (uint32_t i = 8; i + 13 < n; i++)
+                if (memcmp(p + i, key, 13) == 0) {
+                    uint32_t e = i + 13;
+                    while (e < n && p[e] != ')' && e - i < 77) e++;
+                    svc.assign((const char*)p + i + 13, e - i - 13);
+                    break;
+                }
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type = "CONNECT";
+            f.l7.domain = svc;
+            f.l7.service = svc;
+            f.l7.resource = svc;
+            f.l7.endpoint.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if ((tt == 2 || tt == 4 || tt == 5) && dir == 1 &&
+                   f.l7.active) {  // ACCEPT / REFUSE / REDIRECT
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0,
+                             tt == 4 ? 3 : 0, f.l7, "");
+            f.l7.active = false;
+        } else if (tt == 6) {  // DATA: SQL round trips
+            std::string sql, verb;
+            if (dir == 0 && tns_extract_sql(p, n, sql, verb)) {
+                f.l7.active = true;
+                f.l7.req_ts = ts;
+                f.l7.req_len = n;
+                f.l7.req_type = verb;
+                f.l7.resource = sql;
+                f.l7.endpoint = verb;
+                f.l7c.request_count++;
+                f.last_req_pkt_ts = ts;
+            } else if (dir == 1 && f.l7.active) {
+                encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "");
+                f.l7.active = false;
+            }
+        }
+    } else if (f.l7_protocol == 48) {  // ISO 8583
+        char mti[5];
+        bool is_resp;
+        if (!parse_iso8583(p, n, mti, is_resp)) return;
+        if (!is_resp && dir == 0) {
+            f.l7.active = true;
+            f.l7.req_ts = ts;
+            f.l7.req_len = n;
+            f.l7.req_type.assign(mti, 4);
+            f.l7.resource.assign(mti, 4);
+            f.l7.endpoint.assign(mti, 4);
+            f.l7.domain.clear();
+            f.l7c.request_count++;
+            f.last_req_pkt_ts = ts;
+        } else if (is_resp && f.l7.active) {
+            // response code: field 39 is positional; read the first two
+            // printable chars after the bitmaps as a weak approximation
+            encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7, "1987");
+            f.l7.active = false;
+        }
     } else if (f.l7_protocol == 121) {  // TLS: ClientHello SNI only
         std::string sni;
         if (dir == 0 && parse_tls_client_hello(p, n, sni)) {

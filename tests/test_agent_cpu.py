@@ -338,3 +338,93 @@ def test_escape_timer():
     assert a.escaped(now + 7200) is True
     assert a.escaped(now + 7200, escape_s=10**6) is False
     a.close()
+
+
+def test_oracle_tns():
+    """Oracle TNS: CONNECT/ACCEPT with SERVICE_NAME, then SQL round
+    trips over DATA packets (synthetic; no golden capture exists in the
+    reference corpus)."""
+    import struct
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import (eth_ipv4_tcp, SYN, SYNACK,
+                                            PSH_ACK)
+    from deepflow_amd.wire import pb, flow_log, framing
+
+    def tns(ptype, payload):
+        ln = 8 + len(payload)
+        return struct.pack(">HHBBH", ln, 0, ptype, 0, 0) + payload
+
+    connect = tns(1, b"\x01\x36\x01\x2c(DESCRIPTION=(CONNECT_DATA="
+                     b"(SERVICE_NAME=ORCLPDB1)(CID=x)))")
+    accept = tns(2, b"\x01\x36\x00\x00")
+    q = b"SELECT owner, name FROM all_tables WHERE rownum < 10"
+    data_req = tns(6, b"\x00\x00\x03\x5e\x11" + q + b"\x00")
+    data_resp = tns(6, b"\x00\x00\x10\x17" + b"\x07rowdata")
+    a = Agent(vtap_id=1)
+    C, S = 0x0A000001, 0x0A000002
+    t = 10**9
+    seq_c, seq_s = 10, 500
+    pkts = [
+        (eth_ipv4_tcp(C, S, 42000, 1521, SYN, seq_c), t),
+        (eth_ipv4_tcp(S, C, 1521, 42000, SYNACK, seq_s, seq_c + 1), t + 1),
+        (eth_ipv4_tcp(C, S, 42000, 1521, PSH_ACK, seq_c + 1, seq_s + 1,
+                      connect), t + 10**6),
+        (eth_ipv4_tcp(S, C, 1521, 42000, PSH_ACK, seq_s + 1,
+                      seq_c + 1 + len(connect), accept), t + 3 * 10**6),
+        (eth_ipv4_tcp(C, S, 42000, 1521, PSH_ACK,
+                      seq_c + 1 + len(connect),
+                      seq_s + 1 + len(accept), data_req), t + 5 * 10**6),
+        (eth_ipv4_tcp(S, C, 1521, 42000, PSH_ACK,
+                      seq_s + 1 + len(accept),
+                      seq_c + 1 + len(connect) + len(data_req),
+                      data_resp), t + 9 * 10**6),
+    ]
+    for frame, ts in pkts:
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(a.drain(1))]
+    assert all(r["base"]["head"]["proto"] == 62 for r in recs)
+    types = [r["req"]["req_type"] for r in recs]
+    assert types == ["CONNECT", "SELECT"]
+    assert recs[0]["req"]["domain"] == "ORCLPDB1"
+    assert recs[1]["req"]["resource"].startswith(
+        "SELECT owner, name FROM all_tables")
+    assert recs[1]["base"]["head"]["rrt"] == 4000
+    a.close()
+
+
+def test_iso8583():
+    """ISO 8583 authorization round trip (0200 -> 0210), synthetic."""
+    import struct
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import (eth_ipv4_tcp, SYN, SYNACK,
+                                            PSH_ACK)
+    from deepflow_amd.wire import pb, flow_log, framing
+
+    def iso(mti, fields=b"\x00" * 20):
+        body = mti.encode() + b"\x70\x00\x00\x00\x00\x00\x00\x00" + fields
+        return struct.pack(">H", len(body)) + body
+
+    req = iso("0200")
+    resp = iso("0210")
+    a = Agent(vtap_id=1)
+    C, S = 0x0A000001, 0x0A000002
+    t = 10**9
+    pkts = [
+        (eth_ipv4_tcp(C, S, 42005, 8583, SYN, 1), t),
+        (eth_ipv4_tcp(S, C, 8583, 42005, SYNACK, 1, 2), t + 1),
+        (eth_ipv4_tcp(C, S, 42005, 8583, PSH_ACK, 2, 2, req), t + 10**6),
+        (eth_ipv4_tcp(S, C, 8583, 42005, PSH_ACK, 2, 2 + len(req), resp),
+         t + 8 * 10**6),
+    ]
+    for frame, ts in pkts:
+        a.packet(frame, ts)
+    a.tick(1 << 62)
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(a.drain(1))]
+    assert len(recs) == 1
+    assert recs[0]["base"]["head"]["proto"] == 48
+    assert recs[0]["req"]["req_type"] == "0200"
+    assert recs[0]["base"]["head"]["rrt"] == 7000
+    a.close()
