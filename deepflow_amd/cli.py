@@ -156,3 +156,13 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+@cli.command()
+@click.argument("cmd")
+@click.option("--port", type=int, required=True,
+              help="server debug-bus UDP port")
+def debug(cmd, port):
+    """Query the server's UDP debug bus (stats/store/queues/agents)."""
+    from .utils.debug_bus import debug_call
+    click.echo(json.dumps(debug_call(port, cmd), indent=2, default=str))

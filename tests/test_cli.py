@@ -67,3 +67,19 @@ def test_cli_agent_list(live_server):
     r = runner.invoke(cli, ["--server", live_server, "agent", "list"])
     assert r.exit_code == 0
     assert "2" in r.output
+
+
+def test_cli_debug_bus(tmp_path):
+    from click.testing import CliRunner
+    from deepflow_amd.cli import cli
+    from deepflow_amd.utils.debug_bus import DebugBus
+    bus = DebugBus()
+    bus.register("ping", lambda req: {"pong": True})
+    bus.start()
+    try:
+        r = CliRunner().invoke(cli, ["debug", "ping", "--port",
+                                     str(bus.port)])
+        assert r.exit_code == 0, r.output
+        assert '"pong": true' in r.output
+    finally:
+        bus.stop()
