@@ -125,6 +125,37 @@ def main() -> None:
 
     blob = client.get("/v1/export/otlp", params={"limit": 50}).content
     print(f"[export] otlp bytes={len(blob)}")
+
+    # --- alert policy over the ingested spans ---------------------------
+    client.post("/v1/alert-policies/", json={
+        "name": "5xx-spike", "column": "c", "op": ">=", "threshold": 1,
+        "level": 3, "target_column": "request_resource",
+        "sql": "SELECT request_resource, Count(*) AS c FROM l7_flow_log "
+               "WHERE response_status = 'Server Error' "
+               "GROUP BY request_resource"})
+    fired = client.post("/v1/alert-policies/evaluate").json()["fired"]
+    alerts = srv.engine.query(
+        "SELECT policy_name, target, level FROM alert_event LIMIT 3")
+    print(f"[alerts] fired={fired} first={alerts['values'][:1]}")
+
+    # --- org isolation --------------------------------------------------
+    from deepflow_amd.wire import flow_log as _fl
+    org_span = dict(mk_traced_span("org9-span", svc="tenant"),
+                    trace_info={})
+    srv.receiver.handle_frame(framing.encode_frame(
+        framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG, org_id=9),
+        framing.pack_records([pb.encode(org_span,
+                                        _fl.APP_PROTO_LOGS_DATA)])))
+    c9 = client.post("/v1/query/", headers={"X-Org-Id": "9"},
+                     json={"sql": "SELECT Count(*) AS c FROM l7_flow_log"})
+    print(f"[org 9] rows={c9.json()['result']['values'][0][0]} "
+          f"(isolated from default org)")
+
+    # --- UDP debug bus --------------------------------------------------
+    from deepflow_amd.utils.debug_bus import debug_call
+    st = debug_call(srv.debug_bus.port, "store")["result"]
+    print(f"[debug-bus] l7_rows={st['l7_rows']} "
+          f"dict_entries={st['dict_entries']}")
     print("[demo] done")
     a.close()
     srv.stop()
