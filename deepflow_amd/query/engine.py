@@ -119,6 +119,33 @@ def _result_rows(res: Dict) -> List[Dict]:
     return [dict(zip(res["columns"], row)) for row in res["values"]]
 
 
+def _plan_needed(plan) -> Optional[Dict]:
+    """Column set a plan touches, by family — drives lazy cold-segment
+    decompression. None = everything (row-fetch paths)."""
+    if plan.select_rows:
+        return None
+    need: Dict[int, set] = {}
+
+    def add(fam, idx):
+        if fam == Q.SRC_TIME_BUCKET:
+            need.setdefault(Q.SRC_U64, set()).add(0)
+        elif fam in (Q.SRC_STR_HASH, Q.SRC_ATTR_MATCH, Q.SRC_ATTR_VAL):
+            need.setdefault(fam, set())
+        elif fam in (Q.SRC_U64, Q.SRC_U32, Q.SRC_U8, Q.SRC_DID, Q.SRC_KG):
+            need.setdefault(fam, set()).add(idx)
+
+    for t in plan.terms:
+        add(t.family, t.idx)
+    for k in plan.keys:
+        add(k.family, k.idx)
+    for a in plan.aggs:
+        add(a.family, a.idx)
+    for m in plan.agg_meta:
+        if "family" in m:
+            add(m["family"], m["idx"])
+    return need
+
+
 class QueryEngine:
     def __init__(self, pipeline, device: str = "cpu", l4_pipeline=None,
                  remote_hydrator=None):
@@ -163,8 +190,9 @@ class QueryEngine:
                              time_base_s=self.pipe.time_base_s,
                              tags=L7_TAGS, metrics=L7_METRICS,
                              name_maps=self.name_maps)
-            return self._run_segments(plan, self.pipe.segments.scan_list(),
-                                      L7_TAGS, S.STR_COLS)
+            return self._run_segments(
+                plan, self.pipe.segments.scan_list(
+                    needed=_plan_needed(plan)), L7_TAGS, S.STR_COLS)
         if table == "l4_flow_log":
             if self.l4 is None:
                 raise SqlError("l4_flow_log table not enabled")
@@ -173,8 +201,9 @@ class QueryEngine:
                              tags=L4_TAGS, metrics=L4_METRICS,
                              name_maps=self.name_maps)
             from ..store import l4_schema as L4S
-            return self._run_segments(plan, self.l4.segments.scan_list(),
-                                      L4_TAGS, L4S.STR_COLS)
+            return self._run_segments(
+                plan, self.l4.segments.scan_list(
+                    needed=_plan_needed(plan)), L4_TAGS, L4S.STR_COLS)
         row_tables = {
             "event": "event_rows", "perf_event": "perf_event_rows",
             "alert_event": "alert_event_rows",

@@ -143,8 +143,11 @@ def _compress_i32_matrix(mat: torch.Tensor, n: int,
 
 
 def _restore_i32_matrix(cols: List[PackedColumn], mat: torch.Tensor,
-                        n: int, stream: int = 0) -> None:
+                        n: int, stream: int = 0,
+                        only: Optional[set] = None) -> None:
     for c, pc in enumerate(cols):
+        if only is not None and c not in only:
+            continue
         dst = mat[c, :n]
         if pc.raw is not None:
             dst.copy_(pc.raw)
@@ -209,25 +212,52 @@ class CompressedL7Segment:
             total += t.numel() * t.element_size()
         return total
 
-    def materialize(self, seg, stream: int = 0):
-        """Decompress into a recycled L7Segment (capacity >= n_rows)."""
+    def materialize(self, seg, stream: int = 0, needed=None):
+        """Decompress into a recycled L7Segment (capacity >= n_rows).
+
+        `needed` (optional): {family: set(col_idx) | None} from the query
+        plan — only the referenced columns are unpacked (a scan touches a
+        handful of the ~60 columns; unpacking just those makes cold scans
+        ~10x cheaper). None = everything (select-row fetch paths).
+        NB: untouched scratch columns hold stale/zero data, which is safe
+        because the kernel's access pattern is exactly the plan's."""
+        from ..query import spec as Q
         n = self.n_rows
         assert seg.capacity >= n
-        _restore_i32_matrix(self.u64_cols, seg.u64, n, stream)
-        _restore_i32_matrix(self.u32_cols, seg.u32, n, stream)
-        _restore_i32_matrix(self.did_cols, seg.did, n, stream)
-        _restore_i32_matrix(self.kg_cols, seg.kg, n, stream)
-        _restore_i32_matrix(self.rowref_col, seg.str_rowref.view(1, -1), n,
-                            stream)
-        seg.u8[:, :n] = self.u8
-        seg.str_lens[:, :n] = self.str_lens
-        seg.attr_start[:n] = self.attr_start
-        seg.attr_cnt[:n] = self.attr_cnt
-        seg.ensure_attr_pool(self.attr_pool_len)
-        seg.attr_pool[: self.attr_pool_len] = self.attr_pool
-        seg.attr_pool_len = self.attr_pool_len
-        seg.ensure_pool(self.pool_len)
-        seg.pool[: self.pool_len] = self.pool
-        seg.pool_len = self.pool_len
+
+        def want(fam):
+            if needed is None:
+                return None
+            return needed.get(fam, set())
+
+        _restore_i32_matrix(self.u64_cols, seg.u64, n, stream,
+                            only=want(Q.SRC_U64))
+        _restore_i32_matrix(self.u32_cols, seg.u32, n, stream,
+                            only=want(Q.SRC_U32))
+        _restore_i32_matrix(self.did_cols, seg.did, n, stream,
+                            only=want(Q.SRC_DID))
+        _restore_i32_matrix(self.kg_cols, seg.kg, n, stream,
+                            only=want(Q.SRC_KG))
+        u8_only = want(Q.SRC_U8)
+        if u8_only is None:
+            seg.u8[:, :n] = self.u8
+        else:
+            for c in u8_only:
+                seg.u8[c, :n] = self.u8[c]
+        strings_needed = needed is None or Q.SRC_STR_HASH in needed
+        attrs_needed = needed is None or Q.SRC_ATTR_MATCH in needed or             Q.SRC_ATTR_VAL in needed
+        if strings_needed:
+            _restore_i32_matrix(self.rowref_col,
+                                seg.str_rowref.view(1, -1), n, stream)
+            seg.str_lens[:, :n] = self.str_lens
+            seg.ensure_pool(self.pool_len)
+            seg.pool[: self.pool_len] = self.pool
+            seg.pool_len = self.pool_len
+        if attrs_needed:
+            seg.attr_start[:n] = self.attr_start
+            seg.attr_cnt[:n] = self.attr_cnt
+            seg.ensure_attr_pool(self.attr_pool_len)
+            seg.attr_pool[: self.attr_pool_len] = self.attr_pool
+            seg.attr_pool_len = self.attr_pool_len
         seg.n_rows = n
         return seg

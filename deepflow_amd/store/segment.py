@@ -250,10 +250,11 @@ class SegmentSet:
         self._free.append(seg)
         return True
 
-    def scan_list(self, stream: int = 0) -> List:
+    def scan_list(self, stream: int = 0, needed=None) -> List:
         """All queryable segments: cold ones materialized into recycled
         scratch segments (one-shot; the scratch returns to the free-list
-        semantics by being reused on the next call)."""
+        semantics by being reused on the next call). `needed` restricts
+        decompression to the columns a query plan touches."""
         cold = getattr(self, "cold", [])
         if not cold:
             return self.segments
@@ -268,5 +269,5 @@ class SegmentSet:
             self._scratch.append(seg)
         for c, scratch in zip(cold, self._scratch):
             self.reset_segment(scratch)
-            out.append(c.materialize(scratch, stream))
+            out.append(c.materialize(scratch, stream, needed=needed))
         return out + self.segments

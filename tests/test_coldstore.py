@@ -112,3 +112,35 @@ def test_two_tier_watermark():
     # packed form is well under the allocated segment footprint (the raw
     # string pool survives uncompressed, so the bound is conservative)
     assert r_cold < seg_bytes * 0.7
+
+
+def test_lazy_cold_materialization():
+    """Back-to-back queries with different column sets over cold data
+    (lazy per-column decompression must not leak stale columns)."""
+    pipe = _pipe()
+    eng = QueryEngine(pipe, device="cpu")
+    want = [
+        eng.query("SELECT l7_protocol, Count(*) AS c FROM l7_flow_log "
+                  "GROUP BY l7_protocol ORDER BY c DESC"),
+        eng.query("SELECT request_domain, Sum(response_duration) AS s "
+                  "FROM l7_flow_log GROUP BY request_domain "
+                  "ORDER BY s DESC LIMIT 5"),
+        eng.query("SELECT request_resource, trace_id FROM l7_flow_log "
+                  "LIMIT 5"),
+    ]
+    pipe.segments.demote_oldest()
+    got = [
+        eng.query("SELECT l7_protocol, Count(*) AS c FROM l7_flow_log "
+                  "GROUP BY l7_protocol ORDER BY c DESC"),
+        eng.query("SELECT request_domain, Sum(response_duration) AS s "
+                  "FROM l7_flow_log GROUP BY request_domain "
+                  "ORDER BY s DESC LIMIT 5"),
+        eng.query("SELECT request_resource, trace_id FROM l7_flow_log "
+                  "LIMIT 5"),
+    ]
+    assert want == got
+    # explicit needed-set restriction: only one u8 column decompressed
+    from deepflow_amd.query import spec as Q
+    segs = pipe.segments.scan_list(
+        needed={Q.SRC_U8: {0}})
+    assert segs[0].n_rows == CFG.n
