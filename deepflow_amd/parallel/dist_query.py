@@ -47,14 +47,42 @@ def exchange_json(obj, device: str = "cpu") -> List:
     return [json.loads(bytes(p.cpu().numpy()).decode()) for p in parts]
 
 
+def _merge_qrow(a, b):
+    """Merge two per-group quantile rows (histograms sum by bucket,
+    apdex counts sum elementwise)."""
+    if a is None:
+        return b
+    if b is None:
+        return a
+    out = []
+    for x, y in zip(a, b):
+        if x is None:
+            out.append(y)
+        elif y is None:
+            out.append(x)
+        elif "apdex" in x:
+            out.append({"apdex": [i + j
+                                  for i, j in zip(x["apdex"], y["apdex"])]})
+        else:
+            h = {int(k): v for k, v in x["hist"].items()}
+            for k, v in y["hist"].items():
+                h[int(k)] = h.get(int(k), 0) + v
+            out.append({"hist": h})
+    return out
+
+
 def merge_agg_partials(parts: List[Dict]) -> Dict:
-    """Merge per-rank {key_rows, aggs, agg_ops} by agg op."""
+    """Merge per-rank {key_rows, aggs, agg_ops[, qhist]} by agg op."""
     agg_ops = next((p["agg_ops"] for p in parts if p["aggs"]), [])
     merged: Dict[tuple, list] = {}
     keys_of: Dict[tuple, list] = {}
+    qof: Dict[tuple, list] = {}
     for p in parts:
-        for key, agg in zip(p["key_rows"], p["aggs"]):
+        qh = p.get("qhist")
+        for i, (key, agg) in enumerate(zip(p["key_rows"], p["aggs"])):
             k = tuple(tuple(x) if isinstance(x, list) else x for x in key)
+            if qh is not None:
+                qof[k] = _merge_qrow(qof.get(k), qh[i])
             acc = merged.get(k)
             if acc is None:
                 merged[k] = list(agg)
@@ -67,8 +95,11 @@ def merge_agg_partials(parts: List[Dict]) -> Dict:
                     acc[ai] = min(acc[ai], agg[ai])
                 else:
                     acc[ai] = max(acc[ai], agg[ai])
-    return {"key_rows": [keys_of[k] for k in merged],
-            "aggs": list(merged.values())}
+    out = {"key_rows": [keys_of[k] for k in merged],
+           "aggs": list(merged.values())}
+    if qof:
+        out["qdata"] = [qof.get(k) for k in merged]
+    return out
 
 
 def merge_metric_rows(parts: List[List[Dict]], max_fields=("rrt_max",
@@ -133,7 +164,8 @@ class DistQueryEngine:
             return {"columns": cols, "values": seen}
         merged = merge_agg_partials([p for p in parts])
         return self.engine.finalize_groups(sql, merged["key_rows"],
-                                           merged["aggs"])
+                                           merged["aggs"],
+                                           qdata=merged.get("qdata"))
 
     # ---------------------------------------------------- topN pushdown
     # Exchange O(world x K) groups instead of every group: each rank sends

@@ -399,29 +399,76 @@ class QueryEngine:
             return {"kind": "rows",
                     "result": self._run_select(plan, segments, tags,
                                                str_cols)}
-        if any(m["op"] in ("percentile", "apdex") for m in plan.agg_meta):
-            # raw per-shard values would be needed for an exact merge;
-            # refuse loudly instead of shipping a wrong number
-            # (round-2: per-shard histogram exchange)
-            raise SqlError("Percentile/Apdex is not supported in "
-                           "distributed queries yet")
         groups = execute(plan, segments, self.device)
         key_rows = []
         aggs = []
+        raw_keys = []
         for g in groups:
             key_rows.append([self._hydrate(meta["hydrate"], g["key"][ki])
                              for ki, meta in enumerate(plan.key_meta)])
             aggs.append(g["agg"][: len(plan.aggs)])
-        return {
+            raw_keys.append(g["key"])
+        out = {
             "kind": "agg",
             "key_rows": key_rows,
             "aggs": aggs,
             "agg_ops": [a.op for a in plan.aggs],
         }
+        q_metas = [m for m in plan.agg_meta
+                   if m["op"] in ("percentile", "apdex")]
+        if q_metas:
+            # cross-shard quantiles merge as log-bucket histograms
+            # (16 sub-buckets per octave: <=~4.5% relative value error —
+            # the quantileTiming-style tradeoff); apdex stays exact by
+            # shipping (satisfied, tolerated, total) counts per group.
+            out["qhist"] = self._quantile_partial(plan, segments, q_metas,
+                                                  raw_keys)
+        return out
 
-    def finalize_groups(self, sql: str, key_rows, aggs) -> Dict:
+    _QH_SCALE = 16  # histogram sub-buckets per value octave
+
+    def _qbucket_value(self, b: int) -> float:
+        return 2.0 ** (b / self._QH_SCALE) - 1.0
+
+    def _quantile_partial(self, plan, segments, q_metas, raw_keys):
+        """Per-group histograms / apdex counts aligned with key_rows."""
+        import math
+        from .executor import execute_grouped_values
+        uniq, per_meta = execute_grouped_values(plan, segments, q_metas,
+                                                self.device)
+        index = {tuple(int(x) & ((1 << 64) - 1) for x in uniq[i].tolist()):
+                 i for i in range(uniq.shape[0])}
+        out = []
+        for key in raw_keys:
+            gi = index.get(tuple(int(x) & ((1 << 64) - 1) for x in key))
+            row = []
+            for mi, meta in enumerate(q_metas):
+                if gi is None or mi not in per_meta:
+                    row.append(None)
+                    continue
+                svals, starts, counts = per_meta[mi]
+                s0, c0 = int(starts[gi]), int(counts[gi])
+                vals = svals[s0:s0 + c0]
+                if meta["op"] == "apdex":
+                    t = float(meta.get("param", 100000))
+                    sat = int((vals <= t).sum())
+                    tol = int(((vals > t) & (vals <= 4 * t)).sum())
+                    row.append({"apdex": [sat, tol, int(vals.numel())]})
+                else:
+                    hist = {}
+                    for v in vals.tolist():
+                        b = int(round(math.log2(v + 1.0) * self._QH_SCALE))
+                        hist[b] = hist.get(b, 0) + 1
+                    row.append({"hist": hist})
+            out.append(row)
+        return out
+
+    def finalize_groups(self, sql: str, key_rows, aggs,
+                        qdata=None) -> Dict:
         """Turn merged (hydrated keys, raw aggs) back into a result table
-        using the local plan (column names, avg division, order/limit)."""
+        using the local plan (column names, avg division, order/limit).
+        `qdata` carries merged quantile histograms / apdex counts per
+        group (aligned with key_rows), from the distributed merge."""
         m = _FROM_RE.search(sql)
         table = m.group(1).lower() if m else "l7_flow_log"
         if table == "l4_flow_log":
@@ -436,14 +483,20 @@ class QueryEngine:
                              name_maps=self.name_maps)
         columns = plan.key_names + plan.agg_names
         rows = []
-        for key, agg in zip(key_rows, aggs):
+        for gi, (key, agg) in enumerate(zip(key_rows, aggs)):
             row = list(key)
             ai = 0
+            qi = 0
             for meta in plan.agg_meta:
                 if meta["op"] == "avg":
                     ssum, cnt = agg[ai], agg[ai + 1]
                     ai += 2
                     row.append(ssum / cnt if cnt else None)
+                elif meta["op"] in ("percentile", "apdex"):
+                    q = qdata[gi][qi] if qdata and qdata[gi] else None
+                    qi += 1
+                    ai += 1
+                    row.append(self._finish_qdata(meta, q))
                 else:
                     row.append(agg[ai])
                     ai += 1
@@ -452,6 +505,32 @@ class QueryEngine:
             rows = self._apply_slimit(plan, rows)
         rows = self._order_limit(plan, columns, rows)
         return {"columns": columns, "values": rows}
+
+    def _finish_qdata(self, meta, q):
+        """Merged quantile data -> final value (histogram walk with
+        within-bucket interpolation, or exact apdex counts)."""
+        if q is None:
+            return None
+        if "apdex" in q:
+            sat, tol, total = q["apdex"]
+            return (sat + tol / 2) / total if total else None
+        hist = {int(k): v for k, v in q["hist"].items()}
+        total = sum(hist.values())
+        if total == 0:
+            return None
+        qq = float(meta.get("param", 95)) / 100.0
+        target = qq * (total - 1)
+        seen = 0.0
+        for b in sorted(hist):
+            c = hist[b]
+            if seen + c > target:
+                # interpolate inside the bucket's value span
+                lo = self._qbucket_value(b - 1) if b > 0 else 0.0
+                hi = self._qbucket_value(b)
+                frac = (target - seen) / c
+                return lo + (hi - lo) * frac if c > 1 else hi
+            seen += c
+        return self._qbucket_value(max(hist))
 
     # ----------------------------------------------------------- segments
     def _run_segments(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:

@@ -54,6 +54,19 @@ def _worker(rank, world, port, q):
             got[d][0] == want[d] and
             abs(got[d][1] - sum(rrt[d]) / len(rrt[d])) < 1e-6
             for d in want)
+        # distributed percentile: histogram merge vs exact recompute
+        rq = eng.query(
+            "SELECT request_domain, Percentile(response_duration, 90) "
+            "AS p90 FROM l7_flow_log GROUP BY request_domain")
+        import numpy as np
+        ok_pct = True
+        got_p90 = {row[0]: row[1] for row in rq["values"]}
+        for d, vals in rrt.items():
+            exact = float(np.quantile(np.array(vals, dtype=np.float64),
+                                      0.9))
+            approx = got_p90[d]
+            if abs(approx - exact) / max(exact, 1.0) > 0.08:
+                ok_pct = False
         # topN pushdown: forced two-phase exchange must equal the full
         # exchange for a SLIMIT query
         sl = ("SELECT request_domain, Count(*) AS c FROM l7_flow_log "
@@ -69,7 +82,8 @@ def _worker(rank, world, port, q):
         total_req = sum(r["request"] for r in merged)
         dist.barrier()
         dist.destroy_process_group()
-        q.put((rank, ok_count, ok_groups and ok_topn, total_req))
+        q.put((rank, ok_count,
+               ok_groups and ok_topn and ok_pct, total_req))
     except Exception:
         import traceback
         q.put((rank, "ERR", traceback.format_exc(), None))
