@@ -143,6 +143,23 @@ class DistQueryEngine:
 
     def query(self, sql: str) -> Dict:
         import re
+        if sql.strip().lower().startswith("with"):
+            # resolve each CTE through the DISTRIBUTED path (globally
+            # correct aggregates), then run the outer query on the
+            # merged rows — identical on every rank, so no exchange
+            from ..query.engine import split_with, _result_rows
+            ctes, main = split_with(sql)
+            env = {}
+            for name, inner in ctes:
+                env[name] = self.query(inner)
+            m = re.search(r"\bfrom\s+`?(\w+)`?", main, re.IGNORECASE)
+            table = m.group(1).lower() if m else ""
+            if table in env:
+                return self.engine._run_rows(main,
+                                             _result_rows(env[table]),
+                                             time_base_s=0)
+            raise ValueError("distributed WITH: the outer query must "
+                             "select FROM one of its CTEs")
         m = re.search(r"\bslimit\s+(\d+)", sql, re.IGNORECASE)
         if m:  # series-limited queries take the topN pushdown path
             return self.query_topn(sql, int(m.group(1)))

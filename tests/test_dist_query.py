@@ -67,6 +67,13 @@ def _worker(rank, world, port, q):
             approx = got_p90[d]
             if abs(approx - exact) / max(exact, 1.0) > 0.08:
                 ok_pct = False
+        # distributed WITH: inner aggregate resolved globally, outer
+        # re-aggregates the merged CTE rows identically on every rank
+        rw = eng.query(
+            "WITH per AS (SELECT request_domain, Count(*) AS c "
+            "FROM l7_flow_log GROUP BY request_domain) "
+            "SELECT Sum(c) AS total FROM per")
+        ok_with = rw["values"] == [[300]]
         # topN pushdown: forced two-phase exchange must equal the full
         # exchange for a SLIMIT query
         sl = ("SELECT request_domain, Count(*) AS c FROM l7_flow_log "
@@ -83,7 +90,7 @@ def _worker(rank, world, port, q):
         dist.barrier()
         dist.destroy_process_group()
         q.put((rank, ok_count,
-               ok_groups and ok_topn and ok_pct, total_req))
+               ok_groups and ok_topn and ok_pct and ok_with, total_req))
     except Exception:
         import traceback
         q.put((rank, "ERR", traceback.format_exc(), None))
