@@ -178,6 +178,52 @@ class PromQLEngine:
     def _eval_at(self, expr: str, t: int) -> List[Dict]:
         """Evaluate expr at instant t -> [{metric, value}]."""
         expr = expr.strip()
+        # histogram_quantile(q, expr): prometheus-convention le-bucket
+        # interpolation over cumulative counts
+        m = re.match(r"^histogram_quantile\s*\(\s*([0-9.]+)\s*,(.*)\)\s*$",
+                     expr, re.DOTALL)
+        if m:
+            qq = float(m.group(1))
+            inner = self._eval_at(m.group(2), t)
+            groups: Dict[tuple, list] = {}
+            for s_ in inner:
+                le = s_["metric"].get("le")
+                if le is None:
+                    continue
+                key = tuple(sorted((k, v) for k, v in s_["metric"].items()
+                                   if k not in ("le", "__name__")))
+                bound = float("inf") if le in ("+Inf", "Inf") else float(le)
+                groups.setdefault(key, []).append((bound, s_["value"]))
+            out = []
+            for key, buckets in groups.items():
+                buckets.sort()
+                total = buckets[-1][1]
+                if total <= 0:
+                    continue
+                target = qq * total
+                prev_b, prev_c = 0.0, 0.0
+                val = buckets[-1][0]
+                for bound, cum in buckets:
+                    if cum >= target:
+                        if bound == float("inf"):
+                            val = prev_b
+                        else:
+                            frac = (target - prev_c) / max(cum - prev_c,
+                                                           1e-12)
+                            val = prev_b + (bound - prev_b) * frac
+                        break
+                    prev_b, prev_c = bound, cum
+                out.append({"metric": dict(key), "value": val})
+            return out
+        # topk(k, expr) / bottomk(k, expr)
+        m = re.match(r"^(topk|bottomk)\s*\(\s*(\d+)\s*,(.*)\)\s*$", expr,
+                     re.DOTALL)
+        if m:
+            k = int(m.group(2))
+            inner = self._eval_at(m.group(3), t)
+            inner.sort(key=lambda s_: s_["value"],
+                       reverse=m.group(1) == "topk")
+            return inner[:k]
         # parenthesized sub-expression
         if expr.startswith("(") :
             inner, tail = _balanced(expr[expr.index("("):])

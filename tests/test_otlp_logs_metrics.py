@@ -90,3 +90,21 @@ def test_otlp_metrics_to_promql():
     les = {s["metric"].get("le"): float(s["value"][1])
            for s in res2["data"]["result"]}
     assert les == {"0.1": 6.0, "0.5": 9.0, "+Inf": 10.0}
+
+
+def test_histogram_quantile_and_topk():
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12)
+    client = TestClient(srv.app)
+    client.post("/otlp/v1/metrics", content=metrics_blob())
+    r = client.get("/prom/api/v1/query", params={
+        "query": "histogram_quantile(0.5, latency_bucket)",
+        "time": "1"}).json()
+    # buckets: 6 <= 0.1, cum 9 <= 0.5, 10 total; p50 target=5 inside
+    # the first bucket -> 0.1 * 5/6
+    v = float(r["data"]["result"][0]["value"][1])
+    assert abs(v - 0.1 * 5 / 6) < 1e-9, v
+    r2 = client.get("/prom/api/v1/query", params={
+        "query": "topk(1, latency_bucket)", "time": "1"}).json()
+    assert float(r2["data"]["result"][0]["value"][1]) == 10.0
+    assert r2["data"]["result"][0]["metric"]["le"] == "+Inf"
