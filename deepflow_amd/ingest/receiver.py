@@ -169,6 +169,9 @@ class Receiver:
             t = threading.Thread(target=self._conn_loop, args=(conn,),
                                  daemon=True)
             t.start()
+            # daemon connection threads exit with their sockets; keeping
+            # references forever would leak under connection churn
+            self._threads = [th for th in self._threads if th.is_alive()]
             self._threads.append(t)
 
     def _conn_loop(self, conn: socket.socket) -> None:
