@@ -156,9 +156,19 @@ class QueryEngine:
         self.remote = remote_hydrator
         # tagrecorder name maps (id -> display name), set by the server
         self.name_maps = {}
+        # serializes query execution against concurrent ingest (the
+        # server shares this lock with its ingest handlers) and protects
+        # the shared cold-scratch segments from concurrent queries
+        import threading
+        self.lock = threading.RLock()
 
     # ----------------------------------------------------------- dispatch
     def query(self, sql: str, _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
+        with self.lock:
+            return self._query_locked(sql, _ctes)
+
+    def _query_locked(self, sql: str,
+                      _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
         stripped = sql.strip().lower()
         if stripped.startswith("show"):
             return self._show(sql)

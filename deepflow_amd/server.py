@@ -280,7 +280,10 @@ class DeepflowServer:
                 f"{aid}/{mt}": {"frames": st.frames, "bytes": st.bytes}
                 for (aid, mt), st in self.receiver.status.items()})
 
-        self._lock = threading.Lock()
+        # one lock serializes GPU-store mutation (ingest) and scans
+        # (queries): the engine shares it
+        self._lock = threading.RLock()
+        self.engine.lock = self._lock
 
     # ------------------------------------------------------------------
     def org_context(self, org_id: int):
@@ -301,6 +304,7 @@ class DeepflowServer:
                                   segment_rows=self.l4.segments.segment_rows,
                                   kg=kg, time_base_s=self.l4.time_base_s)
             eng = QueryEngine(l7, device=self.device, l4_pipeline=l4)
+            eng.lock = self._lock  # orgs share the ingest lock
             ctx = SimpleNamespace(l7=l7, l4=l4, engine=eng, kg=kg)
             self._org_ctx[org_id] = ctx
         return ctx

@@ -128,3 +128,61 @@ def test_metrics_tables(server):
         "sql": "SELECT Sum(byte_tx) AS b FROM network"})
     assert r2.json()["OPT_STATUS"] == "SUCCESS"
     assert r2.json()["result"]["values"][0][0] > 0
+
+
+def test_concurrent_ingest_and_query():
+    """Queries serialize against ingest via the shared engine lock —
+    hammer both concurrently and check invariants hold."""
+    import threading
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.wire import pb, flow_log, framing
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 9,
+                         dict_capacity=1 << 12)
+
+    def span(i):
+        return {"base": {"start_time": 10**18 + i, "end_time": 10**18 + i,
+                         "flow_id": i, "vtap_id": 1, "tap_side": 1,
+                         "head": {"proto": 20, "msg_type": 2, "rrt": 5},
+                         "ip_src": 1, "ip_dst": 2, "port_src": 9,
+                         "port_dst": 80, "protocol": 6},
+                "req": {"req_type": "GET", "domain": "d",
+                        "resource": f"/r{i % 7}", "endpoint": "e"}}
+
+    stop = threading.Event()
+    errors = []
+
+    def ingester():
+        i = 0
+        while not stop.is_set():
+            recs = [pb.encode(span(i * 50 + j),
+                              flow_log.APP_PROTO_LOGS_DATA)
+                    for j in range(50)]
+            srv.receiver.handle_frame(framing.encode_frame(
+                framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
+                framing.pack_records(recs)))
+            i += 1
+
+    def querier():
+        while not stop.is_set():
+            try:
+                r = srv.engine.query(
+                    "SELECT request_resource, Count(*) AS c "
+                    "FROM l7_flow_log GROUP BY request_resource")
+                total = sum(v[1] for v in r["values"])
+                if total % 50 != 0:  # partial batch visible mid-ingest
+                    errors.append(f"torn read: {total}")
+            except Exception as e:  # noqa: BLE001
+                errors.append(repr(e))
+
+    threads = [threading.Thread(target=ingester),
+               threading.Thread(target=querier),
+               threading.Thread(target=querier)]
+    for t in threads:
+        t.start()
+    import time
+    time.sleep(2.0)
+    stop.set()
+    for t in threads:
+        t.join(timeout=10)
+    assert not errors, errors[:3]
+    assert srv.l7.stats.spans_in > 0
