@@ -66,6 +66,55 @@ def _seg_restore(state: Dict, seg) -> None:
     seg.n_rows = n
 
 
+def _cold_state(c) -> Dict:
+    """Serialize a CompressedL7Segment (packed payloads stay packed)."""
+    def cols(lst):
+        return [{"base": pc.base, "bits": pc.bits, "n": pc.n,
+                 "data": pc.data.cpu().clone() if pc.data is not None
+                 else None,
+                 "raw": pc.raw.cpu().clone() if pc.raw is not None
+                 else None} for pc in lst]
+    return {"n_rows": c.n_rows, "capacity": c.capacity,
+            "layout_version": c.layout_version,
+            "u64_cols": cols(c.u64_cols), "u32_cols": cols(c.u32_cols),
+            "did_cols": cols(c.did_cols), "kg_cols": cols(c.kg_cols),
+            "rowref_col": cols(c.rowref_col),
+            "u8": c.u8.cpu().clone(), "str_lens": c.str_lens.cpu().clone(),
+            "attr_start": c.attr_start.cpu().clone(),
+            "attr_cnt": c.attr_cnt.cpu().clone(),
+            "attr_pool": c.attr_pool.cpu().clone(),
+            "attr_pool_len": c.attr_pool_len,
+            "pool": c.pool.cpu().clone(), "pool_len": c.pool_len}
+
+
+def _cold_restore(state: Dict, device: str):
+    from .coldstore import CompressedL7Segment, PackedColumn
+    c = CompressedL7Segment.__new__(CompressedL7Segment)
+
+    def cols(lst):
+        return [PackedColumn(
+            d["base"], d["bits"],
+            d["data"].to(device) if d["data"] is not None else None,
+            d["n"],
+            raw=d["raw"].to(device) if d["raw"] is not None else None)
+            for d in lst]
+    c.n_rows = state["n_rows"]
+    c.capacity = state["capacity"]
+    c.device = device
+    c.layout_version = state["layout_version"]
+    c.u64_cols = cols(state["u64_cols"])
+    c.u32_cols = cols(state["u32_cols"])
+    c.did_cols = cols(state["did_cols"])
+    c.kg_cols = cols(state["kg_cols"])
+    c.rowref_col = cols(state["rowref_col"])
+    for name in ("u8", "str_lens", "attr_start", "attr_cnt", "attr_pool",
+                 "pool"):
+        setattr(c, name, state[name].to(device))
+    c.attr_pool_len = state["attr_pool_len"]
+    c.pool_len = state["pool_len"]
+    return c
+
+
 def save_l7(pipeline, path: str) -> None:
     segs = pipeline.segments
     payload = {
@@ -73,6 +122,7 @@ def save_l7(pipeline, path: str) -> None:
         "time_base_s": pipeline.time_base_s,
         "segment_rows": segs.segment_rows,
         "segments": [_seg_state(s) for s in segs.segments],
+        "cold": [_cold_state(c) for c in getattr(segs, "cold", [])],
         "dict": {
             "capacity": pipeline.dict.capacity,
             "tkeys": pipeline.dict.tkeys.cpu().clone(),
@@ -106,6 +156,10 @@ def load_l7(pipeline, path: str) -> int:
         v = payload["layout_version"]
     segs = pipeline.segments
     total = 0
+    if payload.get("cold"):
+        segs.cold = [_cold_restore(st, pipeline.device)
+                     for st in payload["cold"]]
+        total += sum(c.n_rows for c in segs.cold)
     for st in payload["segments"]:
         seg = segs.tail(min_free=segs.segment_rows)  # fresh segment
         _seg_restore(st, seg)

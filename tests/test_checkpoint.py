@@ -87,3 +87,30 @@ def test_migration_hook(tmp_path):
                               dict_capacity=1 << 12)
     with pytest.raises(RuntimeError, match="no migration"):
         CK.load_l7(fresh2, path)
+
+
+def test_checkpoint_with_cold_tier(tmp_path):
+    """Demoted (bit-packed) segments survive checkpoint/restore packed."""
+    pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 10,
+                            dict_capacity=1 << 12,
+                            time_base_s=CFG.base_time_ns // 10**9)
+    small = SpanGenConfig(n=900, seed=21, tag_cardinality=40, n_attrs=2,
+                          n_ips=64, n_services=4, n_resources=10)
+    pipe.ingest_frame_payload(gen_span_payload(small))
+    pipe.ingest_frame_payload(gen_span_payload(small))
+    assert pipe.segments.demote_oldest()
+    eng = QueryEngine(pipe, device="cpu")
+    q = ("SELECT request_resource, Count(*) AS c, "
+         "Avg(response_duration) AS a FROM l7_flow_log "
+         "GROUP BY request_resource ORDER BY c DESC")
+    want = eng.query(q)
+    path = str(tmp_path / "cold.ckpt")
+    CK.save_l7(pipe, path)
+    fresh = L7IngestPipeline(device="cpu", segment_rows=1 << 10,
+                             dict_capacity=1 << 12,
+                             time_base_s=CFG.base_time_ns // 10**9)
+    assert CK.load_l7(fresh, path) == 1800
+    assert len(fresh.segments.cold) == 1           # restored still packed
+    assert fresh.segments.cold[0].compressed_bytes() > 0
+    got = QueryEngine(fresh, device="cpu").query(q)
+    assert want == got
