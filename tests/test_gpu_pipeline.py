@@ -132,3 +132,32 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
     assert CK.load_l7(fresh, path) == cfg.n
     got = QueryEngine(fresh, device="cuda").query(q)
     assert want == got
+
+
+@pytest.mark.gpu
+def test_cold_checkpoint_gpu(tmp_path):
+    """Cold (bit-packed) tier checkpoints and restores packed on GPU."""
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query.engine import QueryEngine
+    from deepflow_amd.store import checkpoint as CK
+    cfg = SpanGenConfig(n=4000, seed=3, tag_cardinality=64, n_ips=64,
+                        n_services=4, n_resources=8)
+    p = L7IngestPipeline(device="cuda", segment_rows=1 << 12,
+                         dict_capacity=1 << 13,
+                         time_base_s=cfg.base_time_ns // 10**9)
+    p.ingest_frame_payload(gen_span_payload(cfg))
+    p.ingest_frame_payload(gen_span_payload(cfg))
+    assert p.segments.demote_oldest()
+    q = ("SELECT l7_protocol, Count(*) AS c FROM l7_flow_log "
+         "GROUP BY l7_protocol")
+    want = QueryEngine(p, device="cuda").query(q)
+    path = str(tmp_path / "cold_gpu.ckpt")
+    CK.save_l7(p, path)
+    f = L7IngestPipeline(device="cuda", segment_rows=1 << 12,
+                         dict_capacity=1 << 13,
+                         time_base_s=cfg.base_time_ns // 10**9)
+    assert CK.load_l7(f, path) == 8000
+    assert len(f.segments.cold) == 1
+    assert QueryEngine(f, device="cuda").query(q) == want
