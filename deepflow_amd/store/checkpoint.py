@@ -80,9 +80,12 @@ def _cold_state(c) -> Dict:
             "did_cols": cols(c.did_cols), "kg_cols": cols(c.kg_cols),
             "rowref_col": cols(c.rowref_col),
             "u8": c.u8.cpu().clone(), "str_lens": c.str_lens.cpu().clone(),
-            "attr_start": c.attr_start.cpu().clone(),
-            "attr_cnt": c.attr_cnt.cpu().clone(),
-            "attr_pool": c.attr_pool.cpu().clone(),
+            "attr_start": c.attr_start.cpu().clone()
+            if c.attr_start is not None else None,
+            "attr_cnt": c.attr_cnt.cpu().clone()
+            if c.attr_cnt is not None else None,
+            "attr_pool": c.attr_pool.cpu().clone()
+            if c.attr_pool is not None else None,
             "attr_pool_len": c.attr_pool_len,
             "pool": c.pool.cpu().clone(), "pool_len": c.pool_len}
 
@@ -109,7 +112,8 @@ def _cold_restore(state: Dict, device: str):
     c.rowref_col = cols(state["rowref_col"])
     for name in ("u8", "str_lens", "attr_start", "attr_cnt", "attr_pool",
                  "pool"):
-        setattr(c, name, state[name].to(device))
+        t = state[name]
+        setattr(c, name, t.to(device) if t is not None else None)
     c.attr_pool_len = state["attr_pool_len"]
     c.pool_len = state["pool_len"]
     return c

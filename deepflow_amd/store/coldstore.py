@@ -174,7 +174,9 @@ def _wrap_i32(t: torch.Tensor) -> torch.Tensor:
 
 
 class CompressedL7Segment:
-    """Bit-packed demoted segment; queryable after materialize()."""
+    """Bit-packed demoted segment; queryable after materialize().
+    Handles both L7 and L4 segment layouts (L4 has no dict-id or attr
+    blocks — those compress to empty column lists)."""
 
     is_compressed = True
 
@@ -183,22 +185,29 @@ class CompressedL7Segment:
         self.n_rows = n
         self.capacity = seg.capacity
         self.device = seg.device
-        self.layout_version = seg.layout_version
+        self.layout_version = getattr(seg, "layout_version", 0)
         # u64 columns: 64-bit frame-of-reference (delta almost always
         # fits 32 bits: times share the segment window, rrt/lens are small)
         self.u64_cols = _compress_i32_matrix(seg.u64, n, stream)
         self.u32_cols = _compress_i32_matrix(seg.u32, n, stream)
-        self.did_cols = _compress_i32_matrix(seg.did, n, stream)
+        self.did_cols = _compress_i32_matrix(seg.did, n, stream) \
+            if hasattr(seg, "did") else []
         self.kg_cols = _compress_i32_matrix(seg.kg, n, stream)
         self.rowref_col = _compress_i32_matrix(seg.str_rowref.view(1, -1),
                                                n, stream)
         # small/raw blocks (u8 is already 1 B/row; pools are variable)
         self.u8 = seg.u8[:, :n].clone()
         self.str_lens = seg.str_lens[:, :n].clone()
-        self.attr_start = seg.attr_start[:n].clone()
-        self.attr_cnt = seg.attr_cnt[:n].clone()
-        self.attr_pool = seg.attr_pool[: seg.attr_pool_len].clone()
-        self.attr_pool_len = seg.attr_pool_len
+        if hasattr(seg, "attr_start"):
+            self.attr_start = seg.attr_start[:n].clone()
+            self.attr_cnt = seg.attr_cnt[:n].clone()
+            self.attr_pool = seg.attr_pool[: seg.attr_pool_len].clone()
+            self.attr_pool_len = seg.attr_pool_len
+        else:
+            self.attr_start = None
+            self.attr_cnt = None
+            self.attr_pool = None
+            self.attr_pool_len = 0
         self.pool = seg.pool[: seg.pool_len].clone()
         self.pool_len = seg.pool_len
 
@@ -209,7 +218,8 @@ class CompressedL7Segment:
             total += sum(pc.nbytes() for pc in group)
         for t in (self.u8, self.str_lens, self.attr_start, self.attr_cnt,
                   self.attr_pool, self.pool):
-            total += t.numel() * t.element_size()
+            if t is not None:
+                total += t.numel() * t.element_size()
         return total
 
     def materialize(self, seg, stream: int = 0, needed=None):
@@ -234,8 +244,9 @@ class CompressedL7Segment:
                             only=want(Q.SRC_U64))
         _restore_i32_matrix(self.u32_cols, seg.u32, n, stream,
                             only=want(Q.SRC_U32))
-        _restore_i32_matrix(self.did_cols, seg.did, n, stream,
-                            only=want(Q.SRC_DID))
+        if self.did_cols:
+            _restore_i32_matrix(self.did_cols, seg.did, n, stream,
+                                only=want(Q.SRC_DID))
         _restore_i32_matrix(self.kg_cols, seg.kg, n, stream,
                             only=want(Q.SRC_KG))
         u8_only = want(Q.SRC_U8)
@@ -253,7 +264,7 @@ class CompressedL7Segment:
             seg.ensure_pool(self.pool_len)
             seg.pool[: self.pool_len] = self.pool
             seg.pool_len = self.pool_len
-        if attrs_needed:
+        if attrs_needed and self.attr_start is not None:
             seg.attr_start[:n] = self.attr_start
             seg.attr_cnt[:n] = self.attr_cnt
             seg.ensure_attr_pool(self.attr_pool_len)

@@ -144,3 +144,43 @@ def test_lazy_cold_materialization():
     segs = pipe.segments.scan_list(
         needed={Q.SRC_U8: {0}})
     assert segs[0].n_rows == CFG.n
+
+
+def test_l4_cold_tier():
+    """L4 segments demote/query through the same codec (no dict/attr
+    blocks)."""
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import http_session
+    from deepflow_amd.ingest.l4_pipeline import L4IngestPipeline
+    from deepflow_amd.query.engine import QueryEngine
+    from deepflow_amd.ingest import L7IngestPipeline
+    import numpy as np
+    l4 = L4IngestPipeline(device="cpu", segment_rows=1 << 6, time_base_s=0)
+    a = Agent(vtap_id=1)
+    for i in range(150):
+        for frame, ts in http_session(0x0A000001 + i, 0x0A000002,
+                                      sport=40000 + i, t0=10**9):
+            a.packet(frame, ts)
+    a.tick(1 << 62)
+    payload = a.drain(0)
+    from deepflow_amd.ops import native
+    import ctypes as ct
+    lib = native.cpu()
+    buf = np.frombuffer(payload, dtype=np.uint8)
+    offs = np.zeros(4096, dtype=np.uint32)
+    lens = np.zeros(4096, dtype=np.uint32)
+    n = int(lib.df_scan_offsets(buf.ctypes.data_as(ct.c_void_p), len(buf),
+                                offs.ctypes.data_as(ct.c_void_p),
+                                lens.ctypes.data_as(ct.c_void_p), 4096))
+    # chunk so several segments fill (segment_rows = 64)
+    for i in range(0, n, 60):
+        l4.ingest(buf, offs[i:i + 60].copy(), lens[i:i + 60].copy())
+    assert len(l4.segments.segments) >= 2
+    l7 = L7IngestPipeline(device="cpu", segment_rows=1 << 8,
+                          dict_capacity=1 << 10, time_base_s=0)
+    eng = QueryEngine(l7, device="cpu", l4_pipeline=l4)
+    q = "SELECT Count(*) AS c, Sum(byte_tx) AS tx FROM l4_flow_log"
+    want = eng.query(q)
+    assert l4.segments.demote_oldest()
+    got = eng.query(q)
+    assert want == got
