@@ -162,6 +162,45 @@ class ControllerLite:
     def lookup_gpid(self, agent_id: int, pid: int) -> int:
         return self._gpid_by_key.get((agent_id, pid), 0)
 
+    # ------------------------------------------------------- persistence
+    # (reference keeps this state in MySQL metadb; here the registry,
+    # platform inventory, name maps and GPID allocations serialize into
+    # the checkpoint manifest so agents re-sync to the same ids after a
+    # controller restart)
+    def state_dict(self) -> dict:
+        from dataclasses import asdict
+        return {
+            "agents": {aid: asdict(rec) for aid, rec in self.agents.items()},
+            "group_configs": {g: dict(c)
+                              for g, c in self.group_configs.items()},
+            "config_version": self.config_version,
+            "platform_version": self.platform_version,
+            "platform": {k: asdict(v) for k, v in self.platform.items()},
+            "name_maps": {m: dict(d) for m, d in self.name_maps.items()},
+            "genesis_inventory": dict(self.genesis_inventory),
+            "gpid_by_key": dict(self._gpid_by_key),
+            "next_gpid": self._next_gpid,
+        }
+
+    def load_state_dict(self, state: dict) -> None:
+        self.agents = {aid: AgentRecord(**rec)
+                       for aid, rec in state["agents"].items()}
+        self.group_configs = {g: dict(c)
+                              for g, c in state["group_configs"].items()}
+        self.config_version = state["config_version"]
+        self.platform_version = state["platform_version"]
+        self.platform = {tuple(k): KgInfo(**v)
+                         for k, v in state["platform"].items()}
+        if self.kg is not None and self.platform:
+            self.kg.update(self.platform)
+        # in-place: the query engine holds a live reference to name_maps
+        for m, d in state["name_maps"].items():
+            self.name_maps.setdefault(m, {}).update(d)
+        self.genesis_inventory = dict(state["genesis_inventory"])
+        self._gpid_by_key = {tuple(k): v
+                             for k, v in state["gpid_by_key"].items()}
+        self._next_gpid = state["next_gpid"]
+
     # ------------------------------------------------------------ monitor
     def agent_status(self, stale_after_s: float = 60.0) -> List[dict]:
         now = time.time()

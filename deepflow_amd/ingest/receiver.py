@@ -33,12 +33,16 @@ def _decompress(encoder: int, payload: bytes) -> Optional[np.ndarray]:
         lib = native.cpu()
         src = np.frombuffer(payload, dtype=np.uint8)
         cap = max(len(payload) * 20, 1 << 20)
-        dst = np.zeros(cap, dtype=np.uint8)
-        n = lib.df_zstd_decompress(src.ctypes.data, len(src),
-                                   dst.ctypes.data, cap)
-        if n < 0:
-            return None
-        return dst[:n]
+        while cap <= MAX_FRAME * 64:
+            dst = np.zeros(cap, dtype=np.uint8)
+            n = lib.df_zstd_decompress(src.ctypes.data, len(src),
+                                       dst.ctypes.data, cap)
+            if n >= 0:
+                return dst[:n]
+            # retry with a larger buffer: highly repetitive payloads can
+            # exceed the 20x first-guess ratio
+            cap *= 4
+        return None
     return None
 
 

@@ -165,7 +165,17 @@ class QueryEngine:
     # ----------------------------------------------------------- dispatch
     def query(self, sql: str, _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
         with self.lock:
-            return self._query_locked(sql, _ctes)
+            try:
+                return self._query_locked(sql, _ctes)
+            finally:
+                self._release_scratch()
+
+    def _release_scratch(self) -> None:
+        """Return cold-materialization scratch segments to the store's
+        free-list once the query is finished (bounds query-time HBM)."""
+        for pipe in (self.pipe, self.l4):
+            if pipe is not None:
+                pipe.segments.release_scratch()
 
     def _query_locked(self, sql: str,
                       _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
@@ -377,6 +387,13 @@ class QueryEngine:
         return hydrated group keys + RAW agg vectors (avg still split as
         sum+count so cross-shard merge is exact); everything else returns
         the finished local result under kind='rows'."""
+        with self.lock:
+            try:
+                return self._query_partial_locked(sql)
+            finally:
+                self._release_scratch()
+
+    def _query_partial_locked(self, sql: str) -> Dict:
         stripped = sql.strip().lower()
         m = _FROM_RE.search(sql)
         table = m.group(1).lower() if m else "l7_flow_log"

@@ -285,6 +285,67 @@ class DeepflowServer:
         self._lock = threading.RLock()
         self.engine.lock = self._lock
 
+        # checkpoint wiring (ckissu/metadb analog: restart-persistence
+        # of the hot store + controller registry)
+        self.checkpoint_dir: Optional[str] = None
+        self._ckpt_thread: Optional[threading.Thread] = None
+
+        @self.app.post("/v1/checkpoint/save")
+        def checkpoint_save():
+            if self.checkpoint_dir is None:
+                return {"error": "no --checkpoint-dir configured"}
+            return self.save_checkpoint(self.checkpoint_dir)
+
+    # ------------------------------------------------------- checkpoint
+    def save_checkpoint(self, ckpt_dir: str) -> dict:
+        """Persist the L7 hot store + controller state atomically."""
+        import os
+        from .store import checkpoint as ck
+        os.makedirs(ckpt_dir, exist_ok=True)
+        with self._lock:
+            ck.save_l7(self.l7, os.path.join(ckpt_dir, "l7.ckpt"))
+            import torch as _t
+            tmp = os.path.join(ckpt_dir, "controller.ckpt.tmp")
+            _t.save(self.controller.state_dict(), tmp)
+            os.replace(tmp, os.path.join(ckpt_dir, "controller.ckpt"))
+        return {"status": "ok", "rows": self.l7.segments.n_rows,
+                "agents": len(self.controller.agents)}
+
+    def load_checkpoint(self, ckpt_dir: str) -> dict:
+        import os
+        from .store import checkpoint as ck
+        out = {"l7_rows": 0, "controller": False}
+        p7 = os.path.join(ckpt_dir, "l7.ckpt")
+        if os.path.exists(p7):
+            with self._lock:
+                out["l7_rows"] = ck.load_l7(self.l7, p7)
+        pc = os.path.join(ckpt_dir, "controller.ckpt")
+        if os.path.exists(pc):
+            import torch as _t
+            self.controller.load_state_dict(
+                _t.load(pc, map_location="cpu", weights_only=False))
+            out["controller"] = True
+        return out
+
+    def enable_checkpoints(self, ckpt_dir: str,
+                           interval_s: float = 300.0,
+                           load: bool = True) -> dict:
+        """Startup load + periodic/background save loop."""
+        self.checkpoint_dir = ckpt_dir
+        restored = self.load_checkpoint(ckpt_dir) if load else {}
+        if interval_s > 0:
+            def loop():
+                import time as _t
+                while not self._stop_ckpt.wait(interval_s):
+                    try:
+                        self.save_checkpoint(ckpt_dir)
+                    except Exception:
+                        pass
+            self._stop_ckpt = threading.Event()
+            self._ckpt_thread = threading.Thread(target=loop, daemon=True)
+            self._ckpt_thread.start()
+        return restored
+
     # ------------------------------------------------------------------
     def org_context(self, org_id: int):
         """Lazily-created isolated (l7, l4, engine) triple for a
@@ -413,6 +474,13 @@ class DeepflowServer:
     def stop(self) -> None:
         self.receiver.stop()
         self.debug_bus.stop()
+        if self._ckpt_thread is not None:
+            self._stop_ckpt.set()
+        if self.checkpoint_dir is not None:
+            try:
+                self.save_checkpoint(self.checkpoint_dir)
+            except Exception:
+                pass
 
     def serve_http(self, host: str = "127.0.0.1", port: int = 20416) -> None:
         import uvicorn
@@ -426,11 +494,22 @@ def main() -> None:
     ap.add_argument("--device", default=None)
     ap.add_argument("--tcp-port", type=int, default=20033)
     ap.add_argument("--http-port", type=int, default=20416)
+    ap.add_argument("--checkpoint-dir", default=None,
+                    help="load on start, save every --checkpoint-interval "
+                         "seconds and on shutdown")
+    ap.add_argument("--checkpoint-interval", type=float, default=300.0)
     args = ap.parse_args()
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     srv = DeepflowServer(device=device, tcp_port=args.tcp_port)
+    if args.checkpoint_dir:
+        restored = srv.enable_checkpoints(args.checkpoint_dir,
+                                          args.checkpoint_interval)
+        print(f"checkpoint restore: {restored}")
     srv.start()
-    srv.serve_http(port=args.http_port)
+    try:
+        srv.serve_http(port=args.http_port)
+    finally:
+        srv.stop()
 
 
 if __name__ == "__main__":

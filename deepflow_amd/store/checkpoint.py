@@ -159,6 +159,23 @@ def load_l7(pipeline, path: str) -> int:
         payload = mig(payload)
         v = payload["layout_version"]
     segs = pipeline.segments
+    # restore the time base: rows carry absolute ns timestamps, but query
+    # buckets are relative to time_base_s — a mismatched base silently
+    # shifts every timestamp in results
+    if "time_base_s" in payload:
+        pipeline.time_base_s = payload["time_base_s"]
+    saved_rows = payload.get("segment_rows")
+    if saved_rows is not None and saved_rows > segs.segment_rows:
+        raise RuntimeError(
+            f"checkpoint segment_rows={saved_rows} exceeds pipeline "
+            f"segment_rows={segs.segment_rows}; rebuild the pipeline with "
+            f"segment_rows>={saved_rows} to restore")
+    d_cap = payload["dict"]["tkeys"].numel()
+    if d_cap != pipeline.dict.tkeys.numel():
+        raise RuntimeError(
+            f"checkpoint dictionary capacity {d_cap} != pipeline "
+            f"{pipeline.dict.tkeys.numel()}; slot ids are positional — "
+            f"restore requires the same dict capacity")
     total = 0
     if payload.get("cold"):
         segs.cold = [_cold_restore(st, pipeline.device)

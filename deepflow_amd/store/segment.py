@@ -146,7 +146,9 @@ class SegmentSet:
         return total
 
     def total_alloc_bytes(self) -> int:
-        return sum(self.seg_alloc_bytes(s) for s in self.segments)
+        # scratch segments (cold-query materialization) are live HBM too
+        return sum(self.seg_alloc_bytes(s) for s in self.segments) + \
+            sum(self.seg_alloc_bytes(s) for s in getattr(self, "_scratch", []))
 
     def enforce_watermark(self) -> int:
         """Drop oldest full segments while over the byte watermark; dropped
@@ -252,9 +254,10 @@ class SegmentSet:
 
     def scan_list(self, stream: int = 0, needed=None) -> List:
         """All queryable segments: cold ones materialized into recycled
-        scratch segments (one-shot; the scratch returns to the free-list
-        semantics by being reused on the next call). `needed` restricts
-        decompression to the columns a query plan touches."""
+        scratch segments. `needed` restricts decompression to the columns
+        a query plan touches. Call release_scratch() when the query is
+        done — the scratch buffers go back to the free-list so they are
+        reused by ingest and counted by the watermark."""
         cold = getattr(self, "cold", [])
         if not cold:
             return self.segments
@@ -271,3 +274,11 @@ class SegmentSet:
             self.reset_segment(scratch)
             out.append(c.materialize(scratch, stream, needed=needed))
         return out + self.segments
+
+    def release_scratch(self) -> None:
+        """Return cold-query scratch segments to the free-list (bounds the
+        scratch pool to the lifetime of one query instead of forever)."""
+        scratch = getattr(self, "_scratch", None)
+        if scratch:
+            self._free.extend(scratch)
+            self._scratch = []

@@ -114,3 +114,78 @@ def test_checkpoint_with_cold_tier(tmp_path):
     assert fresh.segments.cold[0].compressed_bytes() > 0
     got = QueryEngine(fresh, device="cpu").query(q)
     assert want == got
+
+
+def test_load_validates_time_base_and_capacity(tmp_path):
+    pipe = _pipe()
+    path = str(tmp_path / "s.ckpt")
+    CK.save_l7(pipe, path)
+    # a pipeline built with a different time base gets the saved one back
+    fresh = L7IngestPipeline(device="cpu", segment_rows=1 << 11,
+                             dict_capacity=1 << 12, time_base_s=0)
+    CK.load_l7(fresh, path)
+    assert fresh.time_base_s == pipe.time_base_s
+    # mismatched dict capacity is a hard error (slot ids are positional)
+    bad = L7IngestPipeline(device="cpu", segment_rows=1 << 11,
+                           dict_capacity=1 << 10,
+                           time_base_s=pipe.time_base_s)
+    with pytest.raises(RuntimeError, match="dictionary capacity"):
+        CK.load_l7(bad, path)
+    # smaller segment_rows is a hard error too
+    small = L7IngestPipeline(device="cpu", segment_rows=1 << 9,
+                             dict_capacity=1 << 12,
+                             time_base_s=pipe.time_base_s)
+    with pytest.raises(RuntimeError, match="segment_rows"):
+        CK.load_l7(small, path)
+
+
+def test_server_checkpoint_wiring(tmp_path):
+    """save_checkpoint/load_checkpoint on the server persist the hot store
+    AND the controller registry (agents keep ids/GPIDs across restart)."""
+    from deepflow_amd.server import DeepflowServer
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 11,
+                         dict_capacity=1 << 12,
+                         time_base_s=CFG.base_time_ns // 10**9)
+    srv.l7.ingest_frame_payload(gen_span_payload(CFG))
+    srv.controller.sync(7, hostname="host-a", ip="10.0.0.7")
+    g = srv.controller.genesis_report(
+        7, [{"pid": 123, "name": "nginx"}], [])
+    gpid = g["gpids"][123]
+    srv.controller.set_group_config("default", {"sync_interval": 30})
+    want = srv.engine.query(QUERIES[1])
+    out = srv.save_checkpoint(str(tmp_path))
+    assert out["rows"] == CFG.n and out["agents"] == 1
+
+    srv2 = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 11,
+                          dict_capacity=1 << 12,
+                          time_base_s=CFG.base_time_ns // 10**9)
+    restored = srv2.load_checkpoint(str(tmp_path))
+    assert restored["l7_rows"] == CFG.n and restored["controller"]
+    assert srv2.engine.query(QUERIES[1]) == want
+    # GPID allocation is stable across restart (metadb role)
+    assert srv2.controller.lookup_gpid(7, 123) == gpid
+    g2 = srv2.controller.genesis_report(
+        7, [{"pid": 123, "name": "nginx"}, {"pid": 456, "name": "redis"}], [])
+    assert g2["gpids"][123] == gpid
+    assert g2["gpids"][456] != gpid
+    assert srv2.controller.group_configs["default"]["sync_interval"] == 30
+
+
+def test_scratch_pool_released_after_cold_query():
+    """Cold-tier query scratch returns to the free-list (bounded), and
+    scratch is counted by the watermark accounting."""
+    pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 8,
+                            dict_capacity=1 << 12,
+                            time_base_s=CFG.base_time_ns // 10**9)
+    pipe.ingest_frame_payload(gen_span_payload(CFG))
+    segs = pipe.segments
+    # force all but the tail segment into the cold tier
+    while segs.demote_oldest():
+        pass
+    assert getattr(segs, "cold", [])
+    eng = QueryEngine(pipe, device="cpu")
+    r = eng.query("SELECT Count(*) AS c FROM l7_flow_log")
+    assert r["values"][0][0] == CFG.n
+    # after the query the scratch pool is empty and its segments are free
+    assert not getattr(segs, "_scratch", [])
+    assert segs._free
