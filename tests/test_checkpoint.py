@@ -177,7 +177,11 @@ def test_scratch_pool_released_after_cold_query():
     pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 8,
                             dict_capacity=1 << 12,
                             time_base_s=CFG.base_time_ns // 10**9)
-    pipe.ingest_frame_payload(gen_span_payload(CFG))
+    import dataclasses
+    small = dataclasses.replace(CFG, n=200)
+    for i in range(6):
+        pipe.ingest_frame_payload(gen_span_payload(
+            dataclasses.replace(small, seed=21 + i)))
     segs = pipe.segments
     # force all but the tail segment into the cold tier
     while segs.demote_oldest():
@@ -185,7 +189,7 @@ def test_scratch_pool_released_after_cold_query():
     assert getattr(segs, "cold", [])
     eng = QueryEngine(pipe, device="cpu")
     r = eng.query("SELECT Count(*) AS c FROM l7_flow_log")
-    assert r["values"][0][0] == CFG.n
+    assert r["values"][0][0] == 6 * 200
     # after the query the scratch pool is empty and its segments are free
     assert not getattr(segs, "_scratch", [])
     assert segs._free
