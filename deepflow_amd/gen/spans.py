@@ -36,6 +36,7 @@ class SpanGenConfig:
     tag_cardinality: int = 100_000  # custom attribute value cardinality
     n_attrs: int = 4                # custom attributes per span
     err_rate_pct: int = 2
+    ip6_rate_pct: int = 0           # % of spans carrying IPv6 addresses
 
 
 HEX = "0123456789abcdef"
@@ -59,6 +60,10 @@ def gen_span_dict(cfg: SpanGenConfig, i: int) -> Dict:
     res = rng.below(cfg.n_resources)
     ip_c = 0x0A000000 | rng.below(cfg.n_ips)
     ip_s = 0x0A000000 | (svc * 7 % cfg.n_ips)
+    # v6 decision draws nothing from the record rng stream so the byte
+    # stream with ip6_rate_pct=0 is unchanged (golden-test stability)
+    v6 = cfg.ip6_rate_pct > 0 and \
+        ((i * 2654435761 + cfg.seed) % 100) < cfg.ip6_rate_pct
     err = rng.below(100) < cfg.err_rate_pct
     trace_hi, trace_lo = rng.next(), rng.next()
     span_id_v = rng.next()
@@ -72,8 +77,8 @@ def gen_span_dict(cfg: SpanGenConfig, i: int) -> Dict:
         "tap_type": 3,
         "tap_side": 1,  # client-side
         "head": {"proto": L7_PROTOCOL_HTTP_1, "msg_type": 2, "rrt": rrt_us},
-        "ip_src": ip_c,
-        "ip_dst": ip_s,
+        "ip_src": 0 if v6 else ip_c,
+        "ip_dst": 0 if v6 else ip_s,
         "l3_epc_id_src": 1 + (ip_c % cfg.n_epcs),
         "l3_epc_id_dst": 1 + (ip_s % cfg.n_epcs),
         "port_src": 32768 + (r0 % 28000),
@@ -89,6 +94,12 @@ def gen_span_dict(cfg: SpanGenConfig, i: int) -> Dict:
         "pod_id_0": 1 + (ip_c % cfg.n_ips),
         "pod_id_1": 1 + (ip_s % cfg.n_ips),
     }
+    if v6:
+        base["is_ipv6"] = 1
+        base["ip6_src"] = (b"\x20\x01\x0d\xb8" + b"\x00" * 8 +
+                           ip_c.to_bytes(4, "big"))
+        base["ip6_dst"] = (b"\x20\x01\x0d\xb8" + b"\x00" * 8 +
+                           ip_s.to_bytes(4, "big"))
     attrs_n: List[str] = []
     attrs_v: List[str] = []
     for a in range(cfg.n_attrs):

@@ -98,6 +98,7 @@ struct SpanCfg {
     uint32_t tag_cardinality;
     uint32_t n_attrs;
     uint32_t err_rate_pct;
+    uint32_t ip6_rate_pct;
 };
 
 // Byte-identical twin of gen/spans.py:gen_span_dict + pb.encode
@@ -111,6 +112,9 @@ void encode_span(Buf& b, const SpanCfg& c, uint64_t i) {
     uint32_t res = (uint32_t)rng.below(c.n_resources);
     uint32_t ip_c = 0x0A000000u | (uint32_t)rng.below(c.n_ips);
     uint32_t ip_s = 0x0A000000u | ((svc * 7u) % c.n_ips);
+    // v6 decision draws nothing from the rng stream (gen/spans.py twin)
+    bool v6 = c.ip6_rate_pct > 0 &&
+              ((i * 2654435761ull + c.seed) % 100) < c.ip6_rate_pct;
     bool err = rng.below(100) < c.err_rate_pct;
     uint64_t trace_hi = rng.next(), trace_lo = rng.next();
     uint64_t span_id_v = rng.next();
@@ -135,14 +139,26 @@ void encode_span(Buf& b, const SpanCfg& c, uint64_t i) {
         // 4 tap_port = 0 skipped
         f_u(s, 5, 1 + (r0 % c.n_agents));
         f_u(s, 6, 3);            // tap_type
+        if (v6) f_u(s, 7, 1);    // is_ipv6
         f_u(s, 8, 1);            // tap_side
         f_m(s, 9, [&](Buf& h) {  // head
             f_u(h, 1, 20);       // proto = HTTP_1
             f_u(h, 2, 2);        // msg_type
             f_u(h, 5, rrt_us);
         });
-        f_u(s, 12, ip_c);
-        f_u(s, 13, ip_s);
+        if (!v6) {
+            f_u(s, 12, ip_c);
+            f_u(s, 13, ip_s);
+        } else {
+            uint8_t a6[16] = {0x20, 0x01, 0x0d, 0xb8, 0, 0, 0, 0,
+                              0, 0, 0, 0, 0, 0, 0, 0};
+            a6[12] = (uint8_t)(ip_c >> 24); a6[13] = (uint8_t)(ip_c >> 16);
+            a6[14] = (uint8_t)(ip_c >> 8);  a6[15] = (uint8_t)ip_c;
+            f_s(s, 14, (const char*)a6, 16);
+            a6[12] = (uint8_t)(ip_s >> 24); a6[13] = (uint8_t)(ip_s >> 16);
+            a6[14] = (uint8_t)(ip_s >> 8);  a6[15] = (uint8_t)ip_s;
+            f_s(s, 15, (const char*)a6, 16);
+        }
         f_i(s, 16, 1 + (int64_t)(ip_c % c.n_epcs));
         f_i(s, 17, 1 + (int64_t)(ip_s % c.n_epcs));
         f_u(s, 18, port_src);

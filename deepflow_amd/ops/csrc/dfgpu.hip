@@ -236,8 +236,11 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                                 WSTR(L7_STR_PKNAME_0, s3, l3);
                             } else if (n2 == 28) {
                                 WSTR(L7_STR_PKNAME_1, s3, l3);
+                            } else if (n2 == 14) {   // ip6_src (16 raw bytes,
+                                WSTR(L7_STR_IP6_0, s3, l3);  // pooled)
+                            } else if (n2 == 15) {   // ip6_dst
+                                WSTR(L7_STR_IP6_1, s3, l3);
                             }
-                            // ip6_src/dst (14/15) skipped: ipv4 hot path v1
                         } else {
                             skip_field(bs, p2, send, w2);
                         }

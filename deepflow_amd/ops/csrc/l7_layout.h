@@ -86,6 +86,8 @@ enum {
     L7_STR_PKNAME_0,
     L7_STR_PKNAME_1,
     L7_STR_BIZ_CODE,
+    L7_STR_IP6_0,         // 16-byte v6 address, pooled (empty for v4 rows)
+    L7_STR_IP6_1,
     L7_STR_N
 };
 

@@ -29,6 +29,7 @@ class SpanCfgC(ct.Structure):
         ("tag_cardinality", ct.c_uint32),
         ("n_attrs", ct.c_uint32),
         ("err_rate_pct", ct.c_uint32),
+        ("ip6_rate_pct", ct.c_uint32),
     ]
 
 
@@ -40,6 +41,7 @@ def span_cfg_c(cfg) -> SpanCfgC:
         n_services=cfg.n_services, n_resources=cfg.n_resources,
         tag_cardinality=cfg.tag_cardinality, n_attrs=cfg.n_attrs,
         err_rate_pct=cfg.err_rate_pct,
+        ip6_rate_pct=getattr(cfg, "ip6_rate_pct", 0),
     )
 
 

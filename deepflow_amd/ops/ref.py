@@ -130,6 +130,12 @@ def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int,
                             elif n2 == 28:
                                 sstr[_STR_IDX["process_kname_1"], rid] = \
                                     S.str_ref_pack(s3, l3)
+                            elif n2 == 14:  # ip6_src (raw 16 bytes)
+                                sstr[_STR_IDX["ip6_0"], rid] = \
+                                    S.str_ref_pack(s3, l3)
+                            elif n2 == 15:  # ip6_dst
+                                sstr[_STR_IDX["ip6_1"], rid] = \
+                                    S.str_ref_pack(s3, l3)
                         elif w2 == 1:
                             p2 += 8
                         elif w2 == 5:

@@ -714,11 +714,13 @@ class QueryEngine:
                 return ""
             off = (rr >> 16) + sum(
                 int(seg.str_lens[c, row]) & 0xFFFF for c in range(sidx))
-            if seg.pool.device.type == "cpu":
-                return bytes(seg.pool[off:off + ln].numpy()).decode(
-                    "utf-8", "replace")
-            return bytes(seg.pool[off:off + ln].cpu().numpy()).decode(
-                "utf-8", "replace")
+            raw = bytes(seg.pool[off:off + ln].numpy()
+                        if seg.pool.device.type == "cpu"
+                        else seg.pool[off:off + ln].cpu().numpy())
+            if col in ("ip6_0", "ip6_1") and ln == 16:
+                import ipaddress
+                return str(ipaddress.IPv6Address(raw))
+            return raw.decode("utf-8", "replace")
         raise SqlError(f"unknown select column {col!r}")
 
     # ----------------------------------------------------------- row tables

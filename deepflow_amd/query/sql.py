@@ -395,6 +395,18 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
         plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
                                  group=group))
         return
+    if td.hydrate == "ip6str":
+        # stored value is the packed 16-byte address
+        import ipaddress
+        from ..store.dictionary import str_hash_py
+        try:
+            packed = ipaddress.IPv6Address(lit[1]).packed
+        except ValueError:
+            raise SqlError(f"bad IPv6 literal {lit[1]!r}")
+        v = str_hash_py(packed, Q.STR_FILTER_SEED)
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
+                                 group=group))
+        return
     if td.hydrate.startswith("dict:"):
         dom = int(td.hydrate.split(":")[1])
         ident = dictionary.lookup_id(dom, lit[1].encode()) \

@@ -87,6 +87,11 @@ def build_l7_tags() -> Dict[str, TagDef]:
                   "x_request_id_1", "http_user_agent", "biz_code"]:
         add(TagDef(sname, Q.SRC_STR_HASH, S.POOL_POS[sname],
                    hydrate="strhash"))
+    # pooled 16-byte v6 addresses: equality filter hashes the packed
+    # address bytes; select formats them back to text
+    for sname in ["ip6_0", "ip6_1"]:
+        add(TagDef(sname, Q.SRC_STR_HASH, S.POOL_POS[sname],
+                   hydrate="ip6str"))
     # KnowledgeGraph universal tags, client (_0) / server (_1)
     for side in (0, 1):
         for j, kname in enumerate(S.KG_COLS):

@@ -29,6 +29,22 @@ def register_migration(from_version: int):
     return deco
 
 
+@register_migration(2)
+def _v2_to_v3(payload: Dict) -> Dict:
+    """v3 appended two pooled ip6 columns to the string block: pad the
+    per-column length tensor with zero rows (old rows have no ip6)."""
+    def pad(sl):
+        z = torch.zeros((2, sl.shape[1]), dtype=sl.dtype)
+        return torch.cat([sl, z], 0)
+    for st in payload.get("segments", []):
+        st["str_lens"] = pad(st["str_lens"])
+    for st in payload.get("cold", []):
+        st["str_lens"] = pad(st["str_lens"])
+        st["layout_version"] = 3
+    payload["layout_version"] = 3
+    return payload
+
+
 def _seg_state(seg) -> Dict:
     n = seg.n_rows
     out = {"n_rows": n, "capacity": seg.capacity,

@@ -6,7 +6,7 @@ these stay in sync."""
 # shape or packing changes; segments record it so mixed-layout windows are
 # rejected instead of misread (migration = drain + reingest, matching the
 # reference's at-most-once durability posture).
-LAYOUT_VERSION = 2  # v2 = packed attr pool + row string block
+LAYOUT_VERSION = 3  # v3 = pooled ip6 columns; v2 = packed attr pool
 
 U64_COLS = [
     "start_time", "end_time", "flow_id", "rrt", "syscall_trace_id_request",
@@ -31,7 +31,7 @@ STR_COLS = [
     "exception_desc", "response_result", "version", "trace_id", "span_id",
     "parent_span_id", "x_request_id_0", "x_request_id_1", "http_user_agent",
     "http_referer", "service_name", "process_kname_0", "process_kname_1",
-    "biz_code",
+    "biz_code", "ip6_0", "ip6_1",
 ]
 
 MAX_ATTRS = 16
@@ -67,7 +67,7 @@ POOL_COLS = [
         "exception_desc", "response_result", "trace_id", "span_id",
         "parent_span_id", "x_request_id_0", "x_request_id_1",
         "http_user_agent", "http_referer", "process_kname_0",
-        "process_kname_1", "biz_code",
+        "process_kname_1", "biz_code", "ip6_0", "ip6_1",
     ]
 ]
 

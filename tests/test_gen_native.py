@@ -90,3 +90,11 @@ def test_parallel_generator_identical(cpu_lib):
     assert buf.tobytes() == gen_span_payload(cfg)
     expected = framing.scan_record_offsets(buf.tobytes())
     assert [(int(o), int(l)) for o, l in zip(offs, lens)] == expected
+
+
+def test_span_payload_golden_ipv6(cpu_lib):
+    cfg = SpanGenConfig(n=48, seed=3, tag_cardinality=100, n_attrs=2,
+                        ip6_rate_pct=40)
+    py = gen_span_payload(cfg)
+    cc = _gen_native(cpu_lib, cfg, 0, 48)
+    assert py == cc
