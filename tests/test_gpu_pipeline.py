@@ -13,8 +13,11 @@ from deepflow_amd.store import l7_schema as S
 from deepflow_amd.store.kg import KnowledgeGraphTable, default_platform
 
 N = 5000
+# ip6_rate_pct > 0: the exact-equality tests below also cover the GPU
+# ip6 decode path (pooled 16-byte addresses)
 CFG = SpanGenConfig(n=N, seed=77, tag_cardinality=500, n_attrs=4,
-                    n_ips=256, n_services=16, n_resources=100)
+                    n_ips=256, n_services=16, n_resources=100,
+                    ip6_rate_pct=15)
 
 
 def _mk(device):
