@@ -22,6 +22,9 @@ class FlowGenConfig:
     n_agents: int = 8
     n_ips: int = 4096
     n_epcs: int = 16
+    acl_rate_pct: int = 0   # % of flows carrying a matched ACL gid
+    n_acls: int = 4
+    ip6_rate_pct: int = 0   # % of flows with IPv6 addresses
 
 
 def gen_flow_dict(cfg: FlowGenConfig, i: int) -> Dict:
@@ -35,14 +38,19 @@ def gen_flow_dict(cfg: FlowGenConfig, i: int) -> Dict:
     pkts_rx = 1 + rng.below(1000)
     bytes_tx = pkts_tx * (64 + rng.below(1400))
     bytes_rx = pkts_rx * (64 + rng.below(1400))
+    # non-rng decisions: stream stays stable when rates are 0
+    v6 = cfg.ip6_rate_pct > 0 and \
+        ((i * 2654435761 + cfg.seed) % 100) < cfg.ip6_rate_pct
+    acl = cfg.acl_rate_pct > 0 and \
+        ((i * 40503 + cfg.seed) % 100) < cfg.acl_rate_pct
     flow = {
         "flow_key": {
             "vtap_id": 1 + (r0 % cfg.n_agents),
             "tap_type": 3,
             "mac_src": r0 & 0xFFFFFFFFFFFF,
             "mac_dst": (r0 >> 8) & 0xFFFFFFFFFFFF,
-            "ip_src": ip_c,
-            "ip_dst": ip_s,
+            "ip_src": 0 if v6 else ip_c,
+            "ip_dst": 0 if v6 else ip_s,
             "port_src": 32768 + (r0 % 28000),
             "port_dst": 443,
             "proto": 6,
@@ -101,6 +109,13 @@ def gen_flow_dict(cfg: FlowGenConfig, i: int) -> Dict:
         "tap_side": 1,
         "direction_score": 255,
     }
+    if v6:
+        flow["flow_key"]["ip6_src"] = (b"\x20\x01\x0d\xb8" + b"\x00" * 8 +
+                                       ip_c.to_bytes(4, "big"))
+        flow["flow_key"]["ip6_dst"] = (b"\x20\x01\x0d\xb8" + b"\x00" * 8 +
+                                       ip_s.to_bytes(4, "big"))
+    if acl:
+        flow["acl_gids"] = [1 + (i % cfg.n_acls)]
     return {"flow": flow}
 
 

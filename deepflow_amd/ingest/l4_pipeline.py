@@ -15,7 +15,7 @@ from ..store import l4_schema as L4
 from ..store import l7_schema as S
 from ..store.segment import SegmentSet, L4Segment
 from ..store.kg import KnowledgeGraphTable
-from ..store.metrics import Net1sMetrics
+from ..store.metrics import L4_TABLES, RollupFamily
 from ..utils.stats import Counter
 
 
@@ -33,7 +33,10 @@ class L4IngestPipeline:
         self.device = device
         self.segments = SegmentSet(segment_rows, device, cls=L4Segment)
         self.kg = kg or KnowledgeGraphTable(device=device)
-        self.metrics = Net1sMetrics(time_base_s, device=device)
+        # flow_metrics network table family (network{,_map}.{1s,1m} +
+        # traffic_policy.1m)
+        self.rollups = RollupFamily(L4_TABLES, time_base_s, device=device)
+        self.metrics = self.rollups.get("network.1s")
         self.time_base_s = time_base_s
         self.stats = L4Stats()
         self.counter = counter or Counter("ingester.l4")
@@ -78,19 +81,18 @@ class L4IngestPipeline:
         ip1 = seg.u32[L4.U32_COLS.index("ip4_1"), base:base + n]
         gpu_ops.kg_probe_cols(epc0, ip0, epc1, ip1, n, self.kg.tkeys,
                               self.kg.tvals, seg.kg, seg.capacity, base)
-        # pool the request_domain strings
-        refs = sstr[0, :n]
-        lens64 = (refs & 0xFFFF).to(torch.int64)
-        cum = torch.cumsum(lens64, 0)
+        # pool the string columns (request_domain + ip6 pair)
+        pool_cols = torch.arange(L4.N_STR, dtype=torch.uint8, device=dev)
+        row_len = torch.zeros(n, dtype=torch.int32, device=dev)
+        gpu_ops.pool_lens(sstr, pool_cols, n, row_len)
+        cum = torch.cumsum(row_len.to(torch.int64), 0)
         total = int(cum[-1].item())
         seg.ensure_pool(total)
         if total:
-            starts = cum - lens64
-            pool_cols = torch.zeros(1, dtype=torch.uint8, device=dev)
+            starts = cum - row_len.to(torch.int64)
             gpu_ops.pool_gather(payload_t, seg, pool_cols, base, n, starts,
                                 seg.pool, seg.pool_len, sstr)
-        gpu_ops.agg_net1s(seg, base, n, self.time_base_s,
-                          self.metrics.tkeys, self.metrics.tvals)
+        self.rollups.update(seg, base, n)
         seg.pool_len += total
 
     def _ingest_cpu(self, payload, offs, lens, seg, base, n) -> None:
@@ -109,17 +111,15 @@ class L4IngestPipeline:
                 vals = info.as_list() if info else [0] * S.N_KG
                 for j, v in enumerate(vals):
                     seg.kg[side * S.N_KG + j, row] = v
-        row_len = torch.zeros(n, dtype=torch.int64)
-        for i in range(n):
-            row_len[i] = int(sstr[0, i].item()) & 0xFFFF
-        cum = torch.cumsum(row_len, 0)
+        pool_cols = list(range(L4.N_STR))
+        row_len = ref.pool_lens_ref(sstr, pool_cols, n)
+        cum = torch.cumsum(row_len.to(torch.int64), 0)
         total = int(cum[-1].item()) if n else 0
         seg.ensure_pool(total)
-        starts = cum - row_len
-        ref.pool_gather_ref(pbytes, seg, [0], base, n, starts, seg.pool_len,
-                            sstr)
-        ref_l4.agg_net1s_ref(seg, base, n, self.time_base_s,
-                             self.metrics.table)
+        starts = cum - row_len.to(torch.int64)
+        ref.pool_gather_ref(pbytes, seg, pool_cols, base, n, starts,
+                            seg.pool_len, sstr)
+        self.rollups.update(seg, base, n)
         seg.pool_len += total
 
     def ingest_frame_payload(self, payload: bytes) -> int:

@@ -24,7 +24,7 @@ from ..store import l7_schema as S
 from ..store.segment import SegmentSet, L7Segment
 from ..store.dictionary import TagDictionary
 from ..store.kg import KnowledgeGraphTable
-from ..store.metrics import App1sMetrics
+from ..store.metrics import L7_TABLES, RollupFamily
 from ..utils.stats import Counter
 
 
@@ -57,7 +57,9 @@ class L7IngestPipeline:
                                    max_bytes=window_bytes)
         self.kg = kg or KnowledgeGraphTable(device=device)
         self.dict = dictionary or TagDictionary(dict_capacity, device=device)
-        self.metrics = App1sMetrics(time_base_s, device=device)
+        # flow_metrics application table family (application{,_map}.{1s,1m})
+        self.rollups = RollupFamily(L7_TABLES, time_base_s, device=device)
+        self.metrics = self.rollups.get("application.1s")
         self.time_base_s = time_base_s
         self.stats = PipelineStats()
         self.counter = counter or Counter("ingester.l7")
@@ -166,8 +168,7 @@ class L7IngestPipeline:
         seg.ensure_pool(total)
         gpu_ops.pool_gather(payload_t, seg, self._pool_cols, base, n,
                             row_start, seg.pool, seg.pool_len, sstr)
-        gpu_ops.agg_app1s(seg, base, n, self.time_base_s,
-                          self.metrics.tkeys, self.metrics.tvals)
+        self.rollups.update(seg, base, n)
         new = self.dict.harvest(payload_host)  # syncs emit buffer
         naive = int((sstr[:, :n] & 0xFFFF).sum()) + \
             int((sattr[:, :n] & 0xFFFF).sum())
@@ -197,7 +198,7 @@ class L7IngestPipeline:
         seg.ensure_pool(total)
         ref.pool_gather_ref(pb, seg, S.POOL_COLS, base, n, row_start,
                             seg.pool_len, sstr)
-        ref.agg_app1s_ref(seg, base, n, self.time_base_s, self.metrics.table)
+        self.rollups.update(seg, base, n)
         naive = int((sstr[:, :n] & 0xFFFF).sum()) + \
             int((sattr[:, :n] & 0xFFFF).sum())
         self.stats.naive_str_bytes += naive

@@ -440,9 +440,25 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
                                 case 12: L4W8(L4_U8_PROTOCOL, v); break;
                                 default: break;
                             }
+                        } else if (w2 == 2 && (n2 == 8 || n2 == 9)) {
+                            // ip6_src/dst: raw 16 bytes into the str pool
+                            uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
+                            cols.strc[(uint64_t)(n2 == 8 ? L4_STR_IP6_0
+                                                         : L4_STR_IP6_1) *
+                                      cols.scratch_stride + rid] =
+                                STR_REF_PACK(p2, l3);
+                            p2 += l3;
                         } else {
                             skip_field(bs, p2, send, w2);
                         }
+                    }
+                    break;
+                }
+                case 24: {  // acl_gids (packed repeated u32): keep first
+                    uint32_t p2 = sub;
+                    if (p2 < send) {
+                        uint64_t v = rd_varint(bs, p2, send);
+                        L4W32(L4_U32_ACL_GID, v);
                     }
                     break;
                 }
@@ -576,39 +592,125 @@ __global__ void k_decode_l4(const uint8_t* __restrict__ payload,
     }
 }
 
-// network.1s rollup from L4 rows (reference: flow_metrics network table).
-// Key: (rel_s << 40) | (vtap & 0xFFF) << 28 | (epc_0 & 0xFFFF) << 12
-//      | (proto << 4) | 1
+// ----------------------------------------------------------------------
+// K5: flow_metrics table family (reference libs/flow-metrics/tag.go
+// 443-523: network{,_map}.{1s,1m}, application{,_map}.{1s,1m},
+// traffic_policy.1m — per-table Code bitmask selects tag columns).
+//
+// Exact keys: each table keeps three arrays — tkeys[cap] holds the 64-bit
+// mixed hash of the key tuple (the CAS claim word), traw[cap][RU_MAX_KEYS]
+// holds the raw tuple words for verification/harvest, tvals[cap][nv] the
+// accumulators. A reader that matches the hash verifies the raw words; if
+// the claimant has not published them yet it simply probes on and claims a
+// fresh slot with the same tuple (no spin — intra-wave spinning on another
+// lane's store can deadlock the wave on CDNA's exec-mask divergence
+// model); duplicate slots for one tuple are merged exactly at harvest.
+// This removes the round-1 masked-pack collisions (vtap&0xFFF etc.).
+// ----------------------------------------------------------------------
+
+#define RU_MAX_KEYS 8
+constexpr uint64_t RU_SENTINEL = 0xFFFFFFFFFFFFFFFFull;
+
+// host-built per-table key spec (the Code-bitmask analog)
+struct RuSpec {
+    uint32_t interval_s;            // time bucket width (1, 60, ...)
+    uint32_t n_keys;                // key sources after the time bucket
+    uint8_t fam[RU_MAX_KEYS];       // 0=u64 1=u32 2=u8
+    uint8_t idx[RU_MAX_KEYS];
+    uint8_t require_nonzero;        // 1-based key index that must be != 0
+                                    // (traffic_policy: acl_gid), 0 = off
+};
+
 enum { NAGG_BYTE_TX = 0, NAGG_BYTE_RX, NAGG_PKT_TX, NAGG_PKT_RX,
        NAGG_NEW_FLOW, NAGG_CLOSED_FLOW, NAGG_RTT_SUM, NAGG_RTT_CNT,
        NAGG_RTT_MAX, NAGG_RETRANS, NAGG_NVALS };
 
-__global__ void k_agg_net1s(const L4Cols cols, uint32_t n, uint64_t time_base_s,
-                            uint64_t* __restrict__ tkeys,
-                            unsigned long long* __restrict__ tvals,
-                            uint32_t cap_mask) {
-    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
-    uint64_t row = cols.base_row + i;
-    uint64_t t_s = cols.u64c[L4_U64_START_TIME * cols.stride + row] / 1000000000ull;
-    uint64_t rel = t_s > time_base_s ? t_s - time_base_s : 0;
-    uint32_t vtap = cols.u32c[L4_U32_VTAP_ID * cols.stride + row];
-    uint32_t epc = cols.u32c[L4_U32_EPC_0 * cols.stride + row];
-    uint8_t proto = cols.u8c[L4_U8_PROTOCOL * cols.stride + row];
-    uint64_t key = (rel << 40) | ((uint64_t)(vtap & 0xFFF) << 28) |
-                   ((uint64_t)(epc & 0xFFFF) << 12) |
-                   ((uint64_t)(proto & 0xFF) << 4) | 1ull;
-    uint32_t slot = (uint32_t)(mix64(key) & cap_mask);
+template <typename ColsT>
+DEV uint64_t ru_src(const ColsT& c, uint64_t row, uint8_t fam, uint8_t idx) {
+    switch (fam) {
+        case 0: return c.u64c[(uint64_t)idx * c.stride + row];
+        case 1: return c.u32c[(uint64_t)idx * c.stride + row];
+        default: return c.u8c[(uint64_t)idx * c.stride + row];
+    }
+}
+
+// claim a slot for the raw tuple kw[0..nw); returns slot or ~0u (table full)
+DEV uint32_t ru_claim(const uint64_t* kw, uint32_t nw,
+                      uint64_t* tkeys, uint64_t* traw, uint32_t cap_mask) {
+    uint64_t h = 0x9E3779B97F4A7C15ull;
+    for (uint32_t w = 0; w < nw; w++) h = mix64(h ^ kw[w]);
+    if (h == EMPTY_KEY) h = 1;
+    uint32_t slot = (uint32_t)(h & cap_mask);
     for (uint32_t probe = 0; probe <= cap_mask; probe++) {
         uint64_t cur = tkeys[slot];
-        if (cur == key) break;
         if (cur == EMPTY_KEY) {
             uint64_t old = atomicCAS((unsigned long long*)&tkeys[slot],
-                                     EMPTY_KEY, key);
-            if (old == EMPTY_KEY || old == key) break;
+                                     EMPTY_KEY, h);
+            if (old == EMPTY_KEY) {
+                uint64_t* r = &traw[(uint64_t)slot * RU_MAX_KEYS];
+                for (uint32_t w = nw; w-- > 1;)
+                    __hip_atomic_store(&r[w], kw[w], __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                // word 0 published last with release: a reader that sees
+                // it non-sentinel sees the full tuple
+                __hip_atomic_store(&r[0], kw[0], __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+                return slot;
+            }
+            cur = old;
+        }
+        if (cur == h) {
+            const uint64_t* r = &traw[(uint64_t)slot * RU_MAX_KEYS];
+            uint64_t w0 = __hip_atomic_load(&r[0], __ATOMIC_ACQUIRE,
+                                            __HIP_MEMORY_SCOPE_AGENT);
+            // w0 is a bucketed relative time — never the sentinel. If the
+            // claimant has not published yet, fall through and claim a
+            // duplicate slot (merged at harvest) instead of spinning.
+            if (w0 == kw[0]) {
+                bool match = true;
+                for (uint32_t w = 1; w < nw; w++)
+                    match = match &&
+                        __hip_atomic_load(&r[w], __ATOMIC_RELAXED,
+                                          __HIP_MEMORY_SCOPE_AGENT) == kw[w];
+                if (match) return slot;
+            }
         }
         slot = (slot + 1) & cap_mask;
     }
+    return 0xFFFFFFFFu;
+}
+
+template <typename ColsT>
+DEV uint32_t ru_key_claim(const ColsT& cols, uint64_t row, uint64_t time_base_s,
+                          const RuSpec& ru, uint64_t* tkeys, uint64_t* traw,
+                          uint32_t cap_mask, unsigned long long* drops) {
+    uint64_t t_s = cols.u64c[0 * cols.stride + row] / 1000000000ull;  // START_TIME
+    uint64_t rel = t_s > time_base_s ? t_s - time_base_s : 0;
+    if (ru.interval_s > 1) rel = (rel / ru.interval_s) * ru.interval_s;
+    uint64_t kw[RU_MAX_KEYS + 1];
+    kw[0] = rel;
+    for (uint32_t k = 0; k < ru.n_keys; k++)
+        kw[1 + k] = ru_src(cols, row, ru.fam[k], ru.idx[k]);
+    if (ru.require_nonzero && kw[ru.require_nonzero] == 0)
+        return 0xFFFFFFFEu;  // row not eligible for this table
+    uint32_t slot = ru_claim(kw, ru.n_keys + 1, tkeys, traw, cap_mask);
+    if (slot == 0xFFFFFFFFu) atomicAdd(drops, 1ull);
+    return slot;
+}
+
+__global__ void k_rollup_l4(const L4Cols cols, uint32_t n, uint64_t time_base_s,
+                            RuSpec ru,
+                            uint64_t* __restrict__ tkeys,
+                            uint64_t* __restrict__ traw,
+                            unsigned long long* __restrict__ tvals,
+                            uint32_t cap_mask,
+                            unsigned long long* __restrict__ drops) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t row = cols.base_row + i;
+    uint32_t slot = ru_key_claim(cols, row, time_base_s, ru, tkeys, traw,
+                                 cap_mask, drops);
+    if (slot >= 0xFFFFFFFEu) return;
     unsigned long long* acc = &tvals[(uint64_t)slot * NAGG_NVALS];
     atomicAdd(&acc[NAGG_BYTE_TX], cols.u64c[L4_U64_BYTE_TX * cols.stride + row]);
     atomicAdd(&acc[NAGG_BYTE_RX], cols.u64c[L4_U64_BYTE_RX * cols.stride + row]);
@@ -627,6 +729,31 @@ __global__ void k_agg_net1s(const L4Cols cols, uint32_t n, uint64_t time_base_s,
     uint64_t retrans = cols.u32c[L4_U32_RETRANS_TX * cols.stride + row] +
                        cols.u32c[L4_U32_RETRANS_RX * cols.stride + row];
     if (retrans) atomicAdd(&acc[NAGG_RETRANS], retrans);
+}
+
+// generic raw-tuple batch insert (agent Document ingest; ops: 0=sum 1=max)
+__global__ void k_rollup_insert(const uint64_t* __restrict__ kws,  // [n, nw]
+                                const unsigned long long* __restrict__ vals,  // [n, nv]
+                                const uint8_t* __restrict__ ops,   // [nv]
+                                uint32_t n, uint32_t nw, uint32_t nv,
+                                uint64_t* __restrict__ tkeys,
+                                uint64_t* __restrict__ traw,
+                                unsigned long long* __restrict__ tvals,
+                                uint32_t cap_mask,
+                                unsigned long long* __restrict__ drops) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t kw[RU_MAX_KEYS + 1];
+    for (uint32_t w = 0; w < nw; w++) kw[w] = kws[(uint64_t)i * nw + w];
+    uint32_t slot = ru_claim(kw, nw, tkeys, traw, cap_mask);
+    if (slot == 0xFFFFFFFFu) { atomicAdd(drops, 1ull); return; }
+    unsigned long long* acc = &tvals[(uint64_t)slot * nv];
+    for (uint32_t v = 0; v < nv; v++) {
+        unsigned long long x = vals[(uint64_t)i * nv + v];
+        if (!x) continue;
+        if (ops[v] == 0) atomicAdd(&acc[v], x);
+        else atomicMax(&acc[v], x);
+    }
 }
 
 // ----------------------------------------------------------------------
@@ -882,38 +1009,29 @@ __global__ void k_pool_gather(const uint8_t* __restrict__ payload,
 }
 
 // ----------------------------------------------------------------------
-// K5: 1s application-metric rollup (reference: agent QuadrupleGenerator +
-// flow_metrics unmarshaller -> application.1s). Key packs
-// (time_s, vtap, l7proto, status, server_port); values accumulated with
-// device-scope atomics.
+// K5 (application side): flow_metrics application{,_map} rollups from L7
+// rows (reference: flow_metrics unmarshaller -> application tables).
+// Exact keys via ru_key_claim (see the table-family section above).
 // ----------------------------------------------------------------------
 
 enum { AGG_REQ = 0, AGG_RESP, AGG_ERR_C, AGG_ERR_S, AGG_RRT_SUM, AGG_RRT_CNT, AGG_RRT_MAX, AGG_NVALS };
 
-__global__ void k_agg_app1s(const L7Cols cols, uint32_t n, uint64_t time_base_s,
+__global__ void k_rollup_l7(const L7Cols cols, uint32_t n, uint64_t time_base_s,
+                            RuSpec ru,
                             uint64_t* __restrict__ tkeys,
+                            uint64_t* __restrict__ traw,
                             unsigned long long* __restrict__ tvals,  // [cap, AGG_NVALS]
-                            uint32_t cap_mask) {
+                            uint32_t cap_mask,
+                            unsigned long long* __restrict__ drops) {
     uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t row = cols.base_row + i;
-    uint64_t t_s = cols.u64c[L7_U64_START_TIME * cols.stride + row] / 1000000000ull;
-    uint64_t rel = t_s > time_base_s ? t_s - time_base_s : 0;
-    uint32_t vtap = cols.u32c[L7_U32_VTAP_ID * cols.stride + row];
-    uint32_t port = cols.u32c[L7_U32_PORT_1 * cols.stride + row];
-    uint8_t l7p = cols.u8c[L7_U8_L7_PROTOCOL * cols.stride + row];
+    uint32_t slot = ru_key_claim(cols, row, time_base_s, ru, tkeys, traw,
+                                 cap_mask, drops);
+    if (slot >= 0xFFFFFFFEu) return;
     uint8_t status = cols.u8c[L7_U8_STATUS * cols.stride + row];
     uint8_t mtype = cols.u8c[L7_U8_MSG_TYPE * cols.stride + row];
     uint64_t rrt = cols.u64c[L7_U64_RRT * cols.stride + row];
-    uint64_t key = (rel << 42) | ((uint64_t)(vtap & 0xFFF) << 30) |
-                   ((uint64_t)l7p << 22) | ((uint64_t)(status & 0xF) << 18) |
-                   ((uint64_t)(port & 0xFFFF) << 2) | 1ull;
-    uint32_t slot = (uint32_t)(mix64(key) & cap_mask);
-    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
-        uint64_t old = atomicCAS((unsigned long long*)&tkeys[slot], EMPTY_KEY, key);
-        if (old == EMPTY_KEY || old == key) break;
-        slot = (slot + 1) & cap_mask;
-    }
     unsigned long long* acc = &tvals[(uint64_t)slot * AGG_NVALS];
     // msg_type: 0=request,1=response,2=session(both)
     if (mtype == 0 || mtype == 2) atomicAdd(&acc[AGG_REQ], 1ull);
@@ -1300,14 +1418,31 @@ int df_decode_l4(const void* payload, const void* offs, const void* lens,
     return (int)hipGetLastError();
 }
 
-int df_agg_net1s(void* u64c, void* u32c, void* u8c, uint64_t stride,
+int df_rollup_l4(void* u64c, void* u32c, void* u8c, uint64_t stride,
                  uint64_t base_row, uint32_t n, uint64_t time_base_s,
-                 void* tkeys, void* tvals, uint32_t cap, uint64_t stream) {
+                 const void* spec, void* tkeys, void* traw, void* tvals,
+                 uint32_t cap, void* drops, uint64_t stream) {
     L4Cols cols{(uint64_t*)u64c, (uint32_t*)u32c, (uint8_t*)u8c,
                 nullptr, stride, base_row};
-    hipLaunchKernelGGL(k_agg_net1s, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
-                       cols, n, time_base_s, (uint64_t*)tkeys,
-                       (unsigned long long*)tvals, cap - 1);
+    RuSpec ru;
+    __builtin_memcpy(&ru, spec, sizeof(RuSpec));
+    hipLaunchKernelGGL(k_rollup_l4, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
+                       cols, n, time_base_s, ru, (uint64_t*)tkeys,
+                       (uint64_t*)traw, (unsigned long long*)tvals, cap - 1,
+                       (unsigned long long*)drops);
+    return (int)hipGetLastError();
+}
+
+int df_rollup_insert(const void* kws, const void* vals, const void* ops,
+                     uint32_t n, uint32_t nw, uint32_t nv,
+                     void* tkeys, void* traw, void* tvals, uint32_t cap,
+                     void* drops, uint64_t stream) {
+    hipLaunchKernelGGL(k_rollup_insert, dim3(grid_for(n)), dim3(BLOCK), 0,
+                       STREAM(stream),
+                       (const uint64_t*)kws, (const unsigned long long*)vals,
+                       (const uint8_t*)ops, n, nw, nv, (uint64_t*)tkeys,
+                       (uint64_t*)traw, (unsigned long long*)tvals, cap - 1,
+                       (unsigned long long*)drops);
     return (int)hipGetLastError();
 }
 
@@ -1388,14 +1523,18 @@ int df_pool_gather(const void* payload, const void* strc, const void* pool_cols,
     return (int)hipGetLastError();
 }
 
-int df_agg_app1s(void* u64c, void* u32c, void* u8c, uint64_t stride,
+int df_rollup_l7(void* u64c, void* u32c, void* u8c, uint64_t stride,
                  uint64_t base_row, uint32_t n, uint64_t time_base_s,
-                 void* tkeys, void* tvals, uint32_t cap, uint64_t stream) {
+                 const void* spec, void* tkeys, void* traw, void* tvals,
+                 uint32_t cap, void* drops, uint64_t stream) {
     L7Cols cols{(uint64_t*)u64c, (uint32_t*)u32c, (uint8_t*)u8c,
                 nullptr, nullptr, nullptr, stride, base_row};
-    hipLaunchKernelGGL(k_agg_app1s, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
-                       cols, n, time_base_s, (uint64_t*)tkeys,
-                       (unsigned long long*)tvals, cap - 1);
+    RuSpec ru;
+    __builtin_memcpy(&ru, spec, sizeof(RuSpec));
+    hipLaunchKernelGGL(k_rollup_l7, dim3(grid_for(n)), dim3(BLOCK), 0, STREAM(stream),
+                       cols, n, time_base_s, ru, (uint64_t*)tkeys,
+                       (uint64_t*)traw, (unsigned long long*)tvals, cap - 1,
+                       (unsigned long long*)drops);
     return (int)hipGetLastError();
 }
 

@@ -73,6 +73,7 @@ enum {
     L4_U32_NAT_REAL_PORT_1,
     L4_U32_VLAN,
     L4_U32_ETH_TYPE,
+    L4_U32_ACL_GID,      // first matched ACL group id (traffic_policy key)
     L4_U32_N
 };
 
@@ -92,5 +93,7 @@ enum {
 
 enum {
     L4_STR_REQUEST_DOMAIN = 0,
+    L4_STR_IP6_0,        // 16-byte v6 address, pooled (empty for v4 rows)
+    L4_STR_IP6_1,
     L4_STR_N
 };

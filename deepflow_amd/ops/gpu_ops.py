@@ -42,14 +42,28 @@ def decode_l4(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
         scratch_str.shape[1], _stream()), "df_decode_l4")
 
 
-def agg_net1s(seg, base_row: int, n: int, time_base_s: int,
-              tkeys: torch.Tensor, tvals: torch.Tensor) -> None:
+def rollup(seg, base_row: int, n: int, time_base_s: int, table,
+           stream: int = 0) -> None:
+    """flow_metrics table rollup from segment rows (k_rollup_l4/l7)."""
     lib = native.gpu()
-    native.check(lib.df_agg_net1s(
+    fn = lib.df_rollup_l4 if table.td.source == "l4" else lib.df_rollup_l7
+    native.check(fn(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.capacity, base_row, n, time_base_s,
-        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(), _stream()),
-        "df_agg_net1s")
+        seg.capacity, base_row, n, time_base_s, table.spec_bytes(),
+        table.tkeys.data_ptr(), table.traw.data_ptr(),
+        table.tvals.data_ptr(), table.capacity, table.drops.data_ptr(),
+        stream or _stream()), "df_rollup")
+
+
+def rollup_insert(kws: torch.Tensor, vals: torch.Tensor, ops: torch.Tensor,
+                  table, stream: int = 0) -> None:
+    lib = native.gpu()
+    native.check(lib.df_rollup_insert(
+        kws.data_ptr(), vals.data_ptr(), ops.data_ptr(),
+        kws.shape[0], kws.shape[1], vals.shape[1],
+        table.tkeys.data_ptr(), table.traw.data_ptr(),
+        table.tvals.data_ptr(), table.capacity, table.drops.data_ptr(),
+        stream or _stream()), "df_rollup_insert")
 
 
 def kg_probe_cols(epc0, ip0, epc1, ip1, n: int, tkeys: torch.Tensor,
@@ -141,15 +155,6 @@ def pool_gather(payload: torch.Tensor, seg, pool_cols: torch.Tensor,
         seg.capacity, base_row, _stream()),
         "df_pool_gather")
 
-
-def agg_app1s(seg, base_row: int, n: int, time_base_s: int,
-              tkeys: torch.Tensor, tvals: torch.Tensor) -> None:
-    lib = native.gpu()
-    native.check(lib.df_agg_app1s(
-        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.capacity, base_row, n, time_base_s,
-        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(), _stream()),
-        "df_agg_app1s")
 
 
 def _opt_ptr(seg, attr: str) -> int:

@@ -87,8 +87,15 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_decode_l7.argtypes = [p, p, p, u32, p, p, p, p, p, p, u64, u64, u64, u64]
     lib.df_decode_l4.restype = ct.c_int
     lib.df_decode_l4.argtypes = [p, p, p, u32, p, p, p, p, u64, u64, u64, u64]
-    lib.df_agg_net1s.restype = ct.c_int
-    lib.df_agg_net1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
+    lib.df_rollup_l4.restype = ct.c_int
+    lib.df_rollup_l4.argtypes = [p, p, p, u64, u64, u32, u64, ct.c_char_p,
+                                 p, p, p, u32, p, u64]
+    lib.df_rollup_l7.restype = ct.c_int
+    lib.df_rollup_l7.argtypes = [p, p, p, u64, u64, u32, u64, ct.c_char_p,
+                                 p, p, p, u32, p, u64]
+    lib.df_rollup_insert.restype = ct.c_int
+    lib.df_rollup_insert.argtypes = [p, p, p, u32, u32, u32, p, p, p, u32,
+                                     p, u64]
     lib.df_kg_build.restype = ct.c_int
     lib.df_kg_build.argtypes = [p, p, u32, p, p, u32, u64]
     lib.df_kg_probe.restype = ct.c_int
@@ -103,8 +110,6 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_pool_lens.argtypes = [p, p, u32, u32, u64, u64, p, u64]
     lib.df_pool_gather.restype = ct.c_int
     lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, p, p, u64, u64, u64]
-    lib.df_agg_app1s.restype = ct.c_int
-    lib.df_agg_app1s.argtypes = [p, p, p, u64, u64, u32, u64, p, p, u32, u64]
     lib.df_query_agg.restype = ct.c_int
     lib.df_query_agg.argtypes = [p, p, p, p, p, p, p, p, p, p, p, u64, u64,
                                  p, u32, u64, p, p, p, u32, u64]

@@ -370,30 +370,3 @@ def pool_gather_ref(payload: bytes, seg, pool_cols, base_row: int, n: int,
 AGG_NVALS = 7  # req, resp, err_c, err_s, rrt_sum, rrt_cnt, rrt_max
 
 
-def agg_app1s_ref(seg, base_row: int, n: int, time_base_s: int,
-                  table: Dict[int, List[int]]) -> None:
-    for i in range(n):
-        row = base_row + i
-        t_s = (int(seg.u64[_U64_IDX["start_time"], row].item()) & M64) // 10**9
-        rel = max(t_s - time_base_s, 0)
-        vtap = int(seg.u32[_U32_IDX["vtap_id"], row].item()) & 0xFFFFFFFF
-        port = int(seg.u32[_U32_IDX["server_port"], row].item()) & 0xFFFFFFFF
-        l7p = int(seg.u8[_U8_IDX["l7_protocol"], row].item())
-        status = int(seg.u8[_U8_IDX["response_status"], row].item())
-        mtype = int(seg.u8[_U8_IDX["msg_type"], row].item())
-        rrt = int(seg.u64[_U64_IDX["rrt"], row].item()) & M64
-        key = (rel << 42) | ((vtap & 0xFFF) << 30) | (l7p << 22) | \
-              ((status & 0xF) << 18) | ((port & 0xFFFF) << 2) | 1
-        acc = table.setdefault(key, [0] * AGG_NVALS)
-        if mtype in (0, 2):
-            acc[0] += 1
-        if mtype in (1, 2):
-            acc[1] += 1
-        if status == 4:
-            acc[2] += 1
-        if status == 3:
-            acc[3] += 1
-        if rrt:
-            acc[4] += rrt
-            acc[5] += 1
-            acc[6] = max(acc[6], rrt)

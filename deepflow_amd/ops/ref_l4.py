@@ -20,8 +20,11 @@ _U32 = {c: i for i, c in enumerate(L4.U32_COLS)}
 _U8 = {c: i for i, c in enumerate(L4.U8_COLS)}
 
 
-def _find_request_domain_ref(mv: memoryview, pos: int, end: int):
-    """Byte range of Flow.request_domain (field path 1 -> 26)."""
+def _find_str_refs(mv: memoryview, pos: int, end: int) -> Dict[int, tuple]:
+    """Byte ranges of the pooled L4 strings: Flow.request_domain
+    (path 1 -> 26) and FlowKey.ip6_src/dst (path 1 -> 1 -> 8/9).
+    -> {str_col_idx: (off, len)}"""
+    out: Dict[int, tuple] = {}
     while pos < end:
         key, pos = read_varint(mv, pos)
         num, wt = key >> 3, key & 7
@@ -34,7 +37,25 @@ def _find_request_domain_ref(mv: memoryview, pos: int, end: int):
                 if w2 == 2:
                     l2, pos = read_varint(mv, pos)
                     if n2 == 26:
-                        return pos, l2
+                        out[L4.STR_COLS.index("request_domain")] = (pos, l2)
+                    elif n2 == 1:  # FlowKey
+                        p3, e3 = pos, pos + l2
+                        while p3 < e3:
+                            k3, p3 = read_varint(mv, p3)
+                            n3, w3 = k3 >> 3, k3 & 7
+                            if w3 == 2:
+                                l3, p3 = read_varint(mv, p3)
+                                if n3 == 8:
+                                    out[L4.STR_COLS.index("ip6_0")] = (p3, l3)
+                                elif n3 == 9:
+                                    out[L4.STR_COLS.index("ip6_1")] = (p3, l3)
+                                p3 += l3
+                            elif w3 == 0:
+                                _, p3 = read_varint(mv, p3)
+                            elif w3 == 1:
+                                p3 += 8
+                            else:
+                                p3 += 4
                     pos += l2
                 elif w2 == 0:
                     _, pos = read_varint(mv, pos)
@@ -42,7 +63,7 @@ def _find_request_domain_ref(mv: memoryview, pos: int, end: int):
                     pos += 8
                 else:
                     pos += 4
-            return None
+            return out
         if wt == 0:
             _, pos = read_varint(mv, pos)
         elif wt == 2:
@@ -52,7 +73,7 @@ def _find_request_domain_ref(mv: memoryview, pos: int, end: int):
             pos += 8
         else:
             pos += 4
-    return None
+    return out
 
 
 def decode_l4_ref(payload: bytes, offs, lens, seg, base_row: int,
@@ -138,6 +159,8 @@ def decode_l4_ref(payload: bytes, offs, lens, seg, base_row: int,
         w32("nat_real_port_1", dst.get("real_port", 0))
         w32("vlan", f.get("vlan", 0))
         w32("eth_type", f.get("eth_type", 0))
+        gids = f.get("acl_gids", [])
+        w32("acl_gid", gids[0] if gids else 0)
         w8("close_type", f.get("close_type", 0))
         w8("tap_side", f.get("tap_side", 0))
         w8("tap_type", fk.get("tap_type", 0))
@@ -148,42 +171,5 @@ def decode_l4_ref(payload: bytes, offs, lens, seg, base_row: int,
         w8("is_new_flow", f.get("is_new_flow", 0))
         w8("is_active_service", f.get("is_active_service", 0))
         w8("direction_score", f.get("direction_score", 0))
-        ref = _find_request_domain_ref(mv, off, off + ln)
-        if ref is not None:
-            sstr[0, rid] = S.str_ref_pack(ref[0], ref[1])
-
-
-NAGG_FIELDS = ["byte_tx", "byte_rx", "packet_tx", "packet_rx", "new_flow",
-               "closed_flow", "rtt_sum", "rtt_count", "rtt_max", "retrans"]
-NAGG_NVALS = len(NAGG_FIELDS)
-
-
-def agg_net1s_ref(seg, base_row: int, n: int, time_base_s: int,
-                  table: Dict[int, List[int]]) -> None:
-    for i in range(n):
-        row = base_row + i
-        t_s = (int(seg.u64[_U64["start_time"], row].item()) & M64) // 10**9
-        rel = max(t_s - time_base_s, 0)
-        vtap = int(seg.u32[_U32["vtap_id"], row].item()) & 0xFFFFFFFF
-        epc = int(seg.u32[_U32["l3_epc_id_0"], row].item()) & 0xFFFFFFFF
-        proto = int(seg.u8[_U8["protocol"], row].item())
-        key = (rel << 40) | ((vtap & 0xFFF) << 28) | ((epc & 0xFFFF) << 12) \
-            | ((proto & 0xFF) << 4) | 1
-        acc = table.setdefault(key, [0] * NAGG_NVALS)
-        acc[0] += int(seg.u64[_U64["byte_tx"], row].item())
-        acc[1] += int(seg.u64[_U64["byte_rx"], row].item())
-        acc[2] += int(seg.u64[_U64["packet_tx"], row].item())
-        acc[3] += int(seg.u64[_U64["packet_rx"], row].item())
-        if int(seg.u8[_U8["is_new_flow"], row].item()):
-            acc[4] += 1
-        if int(seg.u8[_U8["close_type"], row].item()):
-            acc[5] += 1
-        rtt = int(seg.u32[_U32["rtt"], row].item()) & 0xFFFFFFFF
-        if rtt:
-            acc[6] += rtt
-            acc[7] += 1
-            acc[8] = max(acc[8], rtt)
-        retrans = (int(seg.u32[_U32["retrans_tx"], row].item()) +
-                   int(seg.u32[_U32["retrans_rx"], row].item()))
-        if retrans:
-            acc[9] += retrans
+        for sc, (roff, rlen) in _find_str_refs(mv, off, off + ln).items():
+            sstr[sc, rid] = S.str_ref_pack(roff, rlen)

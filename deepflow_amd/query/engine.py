@@ -40,9 +40,7 @@ def rollup_rows(rows, bucket_s: int):
         t = (r["time"] // bucket_s) * bucket_s
         key_fields = tuple(sorted(
             (k, v) for k, v in r.items()
-            if isinstance(v, str) or k in ("vtap_id", "l7_protocol",
-                                           "response_status", "server_port",
-                                           "l3_epc_id", "protocol")))
+            if isinstance(v, str) or k in _ROLLUP_KEY_FIELDS))
         key = (t,) + key_fields
         acc = out.get(key)
         if acc is None:
@@ -55,10 +53,14 @@ def rollup_rows(rows, bucket_s: int):
                     continue
                 if k.endswith("_max"):
                     acc[k] = max(acc[k], v)
-                elif k not in ("vtap_id", "l7_protocol", "response_status",
-                               "server_port", "l3_epc_id", "protocol"):
+                elif k not in _ROLLUP_KEY_FIELDS:
                     acc[k] = acc[k] + v
     return sorted(out.values(), key=lambda r: r["time"])
+
+
+_ROLLUP_KEY_FIELDS = frozenset((
+    "vtap_id", "l7_protocol", "response_status", "server_port", "l3_epc_id",
+    "l3_epc_id_0", "l3_epc_id_1", "protocol", "tap_type", "acl_gid"))
 
 
 def _row_tags(fields: List[str]) -> Dict[str, TagDef]:
@@ -233,43 +235,38 @@ class QueryEngine:
         if table in row_tables:
             rows = getattr(self, row_tables[table], lambda: [])()
             return self._run_rows(sql, rows, time_base_s=0)
-        if table.startswith("application_map") or \
-                table.startswith("network_map"):
-            # _map series (per client/server pair): derived on demand from
-            # the columnar store by the GPU group-by kernel — at B-rows/s
-            # scan rates a persisted map rollup is unnecessary; the table
-            # surface matches the reference's flow_metrics *_map tables.
-            rows = self._map_rows("app" if table.startswith("application")
-                                  else "net",
-                                  self._table_interval(table)
-                                  if "." in table else 1)
-            return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s
-                                  if table.startswith("application")
-                                  else self.l4.time_base_s)
         if table == "application.agent":
             rows = getattr(self, "agent_app_rows", lambda: [])()
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
         if table == "network.agent":
             rows = getattr(self, "agent_net_rows", lambda: [])()
             return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
-        if table.startswith("application"):
-            rows = self.pipe.metrics.rows()
-            iv = self._table_interval(table)
-            if iv > 1:
-                rows = rollup_rows(rows, iv)
-            return self._run_rows(sql, rows, time_base_s=self.pipe.time_base_s)
         if table.startswith("deepflow_system") or \
                 table.startswith("deepflow_tenant"):
             rows = getattr(self, "system_rows", [])
             return self._run_rows(sql, rows, time_base_s=0)
-        if table.startswith("network"):
-            if self.l4 is None:
-                raise SqlError("network table not enabled")
-            rows = self.l4.metrics.rows()
-            iv = self._table_interval(table)
-            if iv > 1:
-                rows = rollup_rows(rows, iv)
-            return self._run_rows(sql, rows, time_base_s=self.l4.time_base_s)
+        # flow_metrics table family: native GPU rollup tables
+        # (network{,_map}.{1s,1m}, application{,_map}.{1s,1m},
+        # traffic_policy.1m), coarser datasources derived from the 1m rows
+        if table.split(".", 1)[0] in ("application", "application_map",
+                                      "network", "network_map",
+                                      "traffic_policy"):
+            pipe = self.pipe if table.startswith("application") else self.l4
+            if pipe is None:
+                raise SqlError(f"{table} needs the l4 pipeline")
+            base_name = table.split(".", 1)[0]
+            native = pipe.rollups.get(table if "." in table
+                                      else base_name + ".1s")
+            if native is not None:
+                rows = native.rows()
+            else:
+                iv = self._table_interval(table)
+                src = pipe.rollups.get(base_name + (".1m" if iv % 60 == 0
+                                                    else ".1s"))
+                if src is None:
+                    raise SqlError(f"unknown table {table!r}")
+                rows = rollup_rows(src.rows(), iv)
+            return self._run_rows(sql, rows, time_base_s=pipe.time_base_s)
         raise SqlError(f"unknown table {table!r}")
 
     # datasource intervals (reference ingester/datasource REST: 1h/1d MVs)
@@ -291,49 +288,6 @@ class QueryEngine:
         if not hasattr(self, "extra_datasources"):
             self.extra_datasources = {}
         self.extra_datasources[name] = interval_s
-
-    def _map_rows(self, kind: str, interval_s: int = 1) -> List[Dict]:
-        """(time, ip pair, server_port) series derived from the segment
-        store via the group-by kernel."""
-        plan = Q.Plan(time_base_s=self.pipe.time_base_s if kind == "app"
-                      else (self.l4.time_base_s if self.l4 else 0))
-        if kind == "app":
-            segs = self.pipe.segments.scan_list()
-            from .tags import L7_TAGS as T
-            dur = T["response_duration"]
-            blen = T["response_length"]
-        else:
-            if self.l4 is None:
-                raise SqlError("network_map needs the l4 pipeline")
-            segs = self.l4.segments.scan_list()
-            from .tags import L4_TAGS as T
-            dur = T["rtt"]
-            blen = T["byte_rx"]
-        plan.keys = [Q.Key(Q.SRC_TIME_BUCKET, 0, interval_s),
-                     Q.Key(T["ip4_0"].family, T["ip4_0"].idx),
-                     Q.Key(T["ip4_1"].family, T["ip4_1"].idx),
-                     Q.Key(T["server_port"].family, T["server_port"].idx)]
-        plan.aggs = [Q.Agg(Q.AGGOP_COUNT),
-                     Q.Agg(Q.AGGOP_SUM, dur.family, dur.idx),
-                     Q.Agg(Q.AGGOP_MAX, dur.family, dur.idx),
-                     Q.Agg(Q.AGGOP_SUM, blen.family, blen.idx)]
-        groups = execute(plan, segs, self.device)
-        base = plan.time_base_s
-        rows = []
-        import ipaddress
-        for g in groups:
-            t, ip0, ip1, port = g["key"]
-            rows.append({
-                "time": base + t,
-                "ip_0": str(ipaddress.IPv4Address(ip0 & 0xFFFFFFFF)),
-                "ip_1": str(ipaddress.IPv4Address(ip1 & 0xFFFFFFFF)),
-                "server_port": port,
-                "request": g["agg"][0],
-                "rrt_sum": g["agg"][1],
-                "rrt_max": g["agg"][2],
-                "byte": g["agg"][3],
-            })
-        return rows
 
     # ----------------------------------------------------------- show
     def _show(self, sql: str) -> Dict:
@@ -376,9 +330,11 @@ class QueryEngine:
                 vals.append([n, disp, unit, "counter"])
             return {"columns": cols, "values": vals}
         if what == "tables":
-            return {"columns": ["name"],
-                    "values": [["l7_flow_log"], ["l4_flow_log"],
-                               ["application.1s"], ["network.1s"]]}
+            names = ["l7_flow_log", "l4_flow_log"]
+            for pipe in (self.pipe, self.l4):
+                if pipe is not None and hasattr(pipe, "rollups"):
+                    names.extend(sorted(pipe.rollups.tables))
+            return {"columns": ["name"], "values": [[n] for n in names]}
         raise SqlError(f"unsupported show: {sql!r}")
 
     # --------------------------------------------------- distributed hook

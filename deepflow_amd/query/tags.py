@@ -170,6 +170,10 @@ def build_l4_tags() -> Dict[str, TagDef]:
             nm = f"{kname}_{side}"
             tags[nm] = TagDef(nm, Q.SRC_KG, side * S.N_KG + j)
     tags["agent_id"] = tags["vtap_id"]
+    # pooled v6 addresses: L4 str cols are pooled in schema order
+    for sname in ("ip6_0", "ip6_1"):
+        tags[sname] = TagDef(sname, Q.SRC_STR_HASH,
+                             L4.STR_COLS.index(sname), hydrate="ip6str")
     return tags
 
 

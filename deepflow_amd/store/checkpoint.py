@@ -154,8 +154,8 @@ def save_l7(pipeline, path: str) -> None:
             "host": dict(pipeline.kg.host),
             "version": pipeline.kg.version,
         },
-        "metrics": pipeline.metrics.state_dict()
-        if hasattr(pipeline.metrics, "state_dict") else None,
+        "rollups": pipeline.rollups.state_dict()
+        if hasattr(pipeline, "rollups") else None,
     }
     tmp = path + ".tmp"
     torch.save(payload, tmp)
@@ -211,7 +211,8 @@ def load_l7(pipeline, path: str) -> int:
     pipeline.kg.tvals.copy_(kg["tvals"].to(pipeline.kg.tvals.device))
     pipeline.kg.host.update(kg["host"])
     pipeline.kg.version = kg["version"]
-    if payload.get("metrics") and hasattr(pipeline.metrics,
-                                          "load_state_dict"):
+    if payload.get("rollups") and hasattr(pipeline, "rollups"):
+        pipeline.rollups.load_state_dict(payload["rollups"])
+    elif payload.get("metrics"):  # pre-table-family checkpoints
         pipeline.metrics.load_state_dict(payload["metrics"])
     return total

@@ -16,7 +16,7 @@ U32_COLS = [
     "retrans_total", "l7_request", "l7_response", "l7_rrt_count",
     "l7_rrt_max", "l7_err_client", "l7_err_server", "l7_err_timeout",
     "gprocess_id_0", "gprocess_id_1", "nat_real_ip_0", "nat_real_ip_1",
-    "nat_real_port_0", "nat_real_port_1", "vlan", "eth_type",
+    "nat_real_port_0", "nat_real_port_1", "vlan", "eth_type", "acl_gid",
 ]
 
 U8_COLS = [
@@ -25,7 +25,7 @@ U8_COLS = [
     "direction_score",
 ]
 
-STR_COLS = ["request_domain"]
+STR_COLS = ["request_domain", "ip6_0", "ip6_1"]
 
 KG_COLS = None  # shares l7_schema.KG_COLS
 

@@ -107,6 +107,11 @@ def test_pool_contents_match(pipes):
 def test_metrics_match(pipes):
     cpu, gpu = pipes
     assert cpu.metrics.rows() == gpu.metrics.rows()
+    for name in ("application.1s", "application.1m", "application_map.1s",
+                 "application_map.1m"):
+        ct, gt = cpu.rollups.get(name), gpu.rollups.get(name)
+        assert ct.rows() == gt.rows(), name
+        assert gt.drop_count() == 0, name
 
 
 @pytest.mark.gpu

@@ -10,7 +10,8 @@ from deepflow_amd.ingest.l4_pipeline import L4IngestPipeline
 from deepflow_amd.store.kg import KnowledgeGraphTable, KgInfo
 
 N = 3000
-CFG = FlowGenConfig(n=N, seed=61, n_ips=64, n_epcs=8)
+CFG = FlowGenConfig(n=N, seed=61, n_ips=64, n_epcs=8, acl_rate_pct=25,
+                    n_acls=3, ip6_rate_pct=20)
 
 
 def _mk(device):
@@ -58,3 +59,15 @@ def test_l4_pool_match(pipes):
 def test_l4_metrics_match(pipes):
     c, g = pipes
     assert c.metrics.rows() == g.metrics.rows()
+
+
+def test_l4_rollup_family_match(pipes):
+    """All network-family tables (incl. _map, 1m, traffic_policy) agree
+    with the CPU oracle exactly; no drops."""
+    c, g = pipes
+    for name in ("network.1s", "network.1m", "network_map.1s",
+                 "network_map.1m", "traffic_policy.1m"):
+        ct, gt = c.rollups.get(name), g.rollups.get(name)
+        assert ct.rows() == gt.rows(), name
+        assert gt.drop_count() == 0, name
+    assert g.rollups.get("traffic_policy.1m").rows()  # acl flows present
