@@ -634,14 +634,20 @@ DEV uint64_t ru_src(const ColsT& c, uint64_t row, uint8_t fam, uint8_t idx) {
     }
 }
 
-// claim a slot for the raw tuple kw[0..nw); returns slot or ~0u (table full)
+// claim a slot for the raw tuple kw[0..nw); returns slot or ~0u.
+// Probes are bounded: once a table region is saturated the row is DROPPED
+// and counted (drops tensor) instead of walking the whole table — a full
+// table must degrade to a watermark drop, not an O(cap) scan per row
+// (the reference throttles/evicts at capacity too).
+#define RU_MAX_PROBES 128u
 DEV uint32_t ru_claim(const uint64_t* kw, uint32_t nw,
                       uint64_t* tkeys, uint64_t* traw, uint32_t cap_mask) {
     uint64_t h = 0x9E3779B97F4A7C15ull;
     for (uint32_t w = 0; w < nw; w++) h = mix64(h ^ kw[w]);
     if (h == EMPTY_KEY) h = 1;
     uint32_t slot = (uint32_t)(h & cap_mask);
-    for (uint32_t probe = 0; probe <= cap_mask; probe++) {
+    uint32_t max_probes = cap_mask < RU_MAX_PROBES ? cap_mask : RU_MAX_PROBES;
+    for (uint32_t probe = 0; probe <= max_probes; probe++) {
         uint64_t cur = tkeys[slot];
         if (cur == EMPTY_KEY) {
             uint64_t old = atomicCAS((unsigned long long*)&tkeys[slot],

@@ -61,6 +61,9 @@ class TableDef:
     require_nonzero: Optional[str] = None  # key that must be != 0
     # output name per key (reference tag naming); None = same as key col
     out_names: Optional[Tuple[str, ...]] = None
+    # device table capacity hint (log2); _map tables carry endpoint pairs
+    # and need far more slots than the scalar-tag tables
+    cap_pow2: int = 18
 
 
 def _net(name, interval, map_):
@@ -72,7 +75,8 @@ def _net(name, interval, map_):
             CODE_VTAP | CODE_IP_PATH | CODE_L3_EPC | CODE_PROTOCOL |
             CODE_SERVER_PORT,
             out_names=("vtap_id", "ip_0", "ip_1", "l3_epc_id_0",
-                       "l3_epc_id_1", "protocol", "server_port"))
+                       "l3_epc_id_1", "protocol", "server_port"),
+            cap_pow2=21)
     return TableDef(
         name, "l4", interval,
         ("vtap_id", "l3_epc_id_0", "protocol", "server_port", "tap_type"),
@@ -91,7 +95,7 @@ def _app(name, interval, map_):
             "app",
             CODE_VTAP | CODE_IP_PATH | CODE_L7_PROTOCOL | CODE_SERVER_PORT,
             out_names=("vtap_id", "ip_0", "ip_1", "l7_protocol",
-                       "server_port"))
+                       "server_port"), cap_pow2=21)
     return TableDef(
         name, "l7", interval,
         ("vtap_id", "l7_protocol", "response_status", "server_port"), "app",
@@ -345,9 +349,9 @@ class RollupFamily:
     batch (one kernel launch per table)."""
 
     def __init__(self, defs: List[TableDef], time_base_s: int,
-                 device: str = "cpu", capacity_pow2: int = 1 << 18):
+                 device: str = "cpu"):
         self.tables: Dict[str, RollupTable] = {
-            td.name: RollupTable(td, time_base_s, device, capacity_pow2)
+            td.name: RollupTable(td, time_base_s, device, 1 << td.cap_pow2)
             for td in defs}
 
     def update(self, seg, base: int, n: int, stream: int = 0) -> None:
