@@ -39,6 +39,8 @@ def _lib():
         lib.dfa_packet.argtypes = [p, p, u32, u64]
         lib.dfa_packet_batch.restype = ct.c_int64
         lib.dfa_packet_batch.argtypes = [p, p, u64]
+        lib.dfa_syscall_batch.restype = ct.c_int64
+        lib.dfa_syscall_batch.argtypes = [p, p, u64]
         lib.dfa_tick.argtypes = [p, u64]
         lib.dfa_drain.restype = u64
         lib.dfa_drain.argtypes = [p, ct.c_int, p, u64]
@@ -104,6 +106,13 @@ class Agent:
         buf = np.frombuffer(blob, dtype=np.uint8)
         return int(self._lib.dfa_packet_batch(self._h, buf.ctypes.data,
                                               len(blob)))
+
+    def syscall_batch(self, blob: bytes) -> int:
+        """eBPF socket-trace events (ebpf/runtime.py wire format) into
+        the FlowMap + L7 parsers; one native call per drained batch."""
+        buf = np.frombuffer(blob, dtype=np.uint8)
+        return int(self._lib.dfa_syscall_batch(self._h, buf.ctypes.data,
+                                               len(blob)))
 
     def tick(self, now_ns: int) -> None:
         self._lib.dfa_tick(self._h, now_ns)

@@ -1,0 +1,237 @@
+"""Raw bpf(2) loader + tracepoint attach + perf-buffer reader.
+
+No libbpf, no BTF: maps are created with BPF_MAP_CREATE, programs loaded
+with BPF_PROG_LOAD (map refs patched to fds in the ld_imm64 stream), and
+attached to the stable raw_syscalls tracepoints via
+perf_event_open(PERF_TYPE_TRACEPOINT) + PERF_EVENT_IOC_SET_BPF.
+Reference counterpart: agent/src/ebpf/user/tracer.c + load/attach in
+user/socket.c — built on libbpf/BTF there.
+
+Degrades gracefully: `available()` probes for tracefs + CAP_BPF; in
+environments without them (this container) the same programs run under
+ebpf/vm.py instead.
+"""
+from __future__ import annotations
+
+import ctypes as ct
+import mmap
+import os
+import struct
+from typing import Callable, Dict, List, Optional
+
+from .insn import Asm
+
+SYS_bpf = 321           # x86_64
+SYS_perf_event_open = 298
+
+BPF_MAP_CREATE = 0
+BPF_MAP_LOOKUP_ELEM = 1
+BPF_MAP_UPDATE_ELEM = 2
+BPF_MAP_GET_NEXT_KEY = 4
+BPF_PROG_LOAD = 5
+
+BPF_PROG_TYPE_TRACEPOINT = 5
+BPF_PROG_TYPE_PERF_EVENT = 7
+
+PERF_TYPE_TRACEPOINT = 2
+PERF_TYPE_SOFTWARE = 1
+PERF_COUNT_SW_CPU_CLOCK = 0
+PERF_EVENT_IOC_ENABLE = 0x2400
+PERF_EVENT_IOC_SET_BPF = 0x40042408
+
+_libc = ct.CDLL(None, use_errno=True)
+
+
+def _bpf(cmd: int, attr: bytes) -> int:
+    buf = ct.create_string_buffer(attr, len(attr))
+    rc = _libc.syscall(SYS_bpf, cmd, buf, len(attr))
+    if rc < 0:
+        err = ct.get_errno()
+        raise OSError(err, f"bpf(cmd={cmd}): {os.strerror(err)}")
+    return rc
+
+
+def map_create(mtype: int, key_size: int, value_size: int,
+               max_entries: int) -> int:
+    if mtype == 4 and max_entries == 0:       # perf array: one per cpu
+        max_entries = os.cpu_count() or 1
+    attr = struct.pack("<IIIII", mtype, key_size, value_size, max_entries, 0)
+    return _bpf(BPF_MAP_CREATE, attr + b"\x00" * 48)
+
+def map_update(fd: int, key: bytes, value: bytes, flags: int = 0) -> None:
+    k = ct.create_string_buffer(key, len(key))
+    v = ct.create_string_buffer(value, len(value))
+    attr = struct.pack("<IIQQQ", fd, 0, ct.addressof(k), ct.addressof(v),
+                       flags)
+    _bpf(BPF_MAP_UPDATE_ELEM, attr)
+
+
+def map_lookup(fd: int, key: bytes, value_size: int) -> Optional[bytes]:
+    k = ct.create_string_buffer(key, len(key))
+    v = ct.create_string_buffer(value_size)
+    attr = struct.pack("<IIQQQ", fd, 0, ct.addressof(k), ct.addressof(v), 0)
+    try:
+        _bpf(BPF_MAP_LOOKUP_ELEM, attr)
+    except OSError:
+        return None
+    return v.raw
+
+
+def map_next_key(fd: int, key: Optional[bytes], key_size: int) -> Optional[bytes]:
+    k = ct.create_string_buffer(key or b"\x00" * key_size, key_size)
+    n = ct.create_string_buffer(key_size)
+    attr = struct.pack("<IIQQQ", fd, 0,
+                       ct.addressof(k) if key else 0, ct.addressof(n), 0)
+    try:
+        _bpf(BPF_MAP_GET_NEXT_KEY, attr)
+    except OSError:
+        return None
+    return n.raw
+
+
+def prog_load(prog_type: int, insns: bytes, license_: bytes = b"GPL",
+              log: bool = False) -> int:
+    lic = ct.create_string_buffer(license_, len(license_) + 1)
+    log_buf = ct.create_string_buffer(1 << 18) if log else None
+    attr = struct.pack(
+        "<IIQQIIQI4x", prog_type, len(insns) // 8,
+        ct.addressof(ct.create_string_buffer(insns, len(insns))),
+        ct.addressof(lic), 2 if log else 0,
+        len(log_buf) if log else 0,
+        ct.addressof(log_buf) if log else 0, 0)
+    try:
+        return _bpf(BPF_PROG_LOAD, attr + b"\x00" * 24)
+    except OSError:
+        if log and log_buf is not None:
+            raise OSError(f"prog_load failed; verifier log:\n"
+                          f"{log_buf.value.decode(errors='replace')}")
+        raise
+
+
+def _tracefs() -> Optional[str]:
+    for p in ("/sys/kernel/tracing", "/sys/kernel/debug/tracing"):
+        if os.path.isdir(os.path.join(p, "events")):
+            return p
+    return None
+
+
+def tracepoint_id(category: str, name: str) -> int:
+    base = _tracefs()
+    if base is None:
+        raise FileNotFoundError("tracefs not mounted")
+    with open(f"{base}/events/{category}/{name}/id") as f:
+        return int(f.read().strip())
+
+
+def perf_event_open(attr: bytes, pid: int, cpu: int, group_fd: int = -1,
+                    flags: int = 0) -> int:
+    buf = ct.create_string_buffer(attr, len(attr))
+    fd = _libc.syscall(SYS_perf_event_open, buf, pid, cpu, group_fd, flags)
+    if fd < 0:
+        err = ct.get_errno()
+        raise OSError(err, f"perf_event_open: {os.strerror(err)}")
+    return fd
+
+
+def attach_tracepoint(prog_fd: int, category: str, name: str) -> int:
+    tp_id = tracepoint_id(category, name)
+    # struct perf_event_attr: type, size, config, sample fields...
+    attr = struct.pack("<IIQQQ", PERF_TYPE_TRACEPOINT, 112, tp_id, 0, 0)
+    attr = attr.ljust(112, b"\x00")
+    fd = perf_event_open(attr, -1, 0)
+    import fcntl
+    fcntl.ioctl(fd, PERF_EVENT_IOC_SET_BPF, prog_fd)
+    fcntl.ioctl(fd, PERF_EVENT_IOC_ENABLE, 0)
+    return fd
+
+
+def available() -> bool:
+    """Can this process load + attach BPF programs here?"""
+    if _tracefs() is None:
+        return False
+    try:
+        fd = map_create(1, 8, 8, 4)
+        os.close(fd)
+        return True
+    except OSError:
+        return False
+
+
+class SocketTracer:
+    """Load the socket-trace programs into the kernel and stream events.
+
+    perf ring reading uses one mmap'd page-set per CPU
+    (PERF_EVENT_IOC_SET_BPF on the tracepoint event delivers
+    bpf_perf_event_output records into the same buffers)."""
+
+    def __init__(self):
+        from .progs import MAPS, build_sys_enter, build_sys_exit
+        self.map_fds: Dict[str, int] = {}
+        for name, spec in MAPS.items():
+            self.map_fds[name] = map_create(*spec)
+        self.enter_fd = prog_load(
+            BPF_PROG_TYPE_TRACEPOINT,
+            build_sys_enter().to_bytes(self.map_fds), log=True)
+        self.exit_fd = prog_load(
+            BPF_PROG_TYPE_TRACEPOINT,
+            build_sys_exit().to_bytes(self.map_fds), log=True)
+        self.tp_fds: List[int] = []
+        self.rings: List[tuple] = []
+
+    def attach(self) -> None:
+        self.tp_fds.append(attach_tracepoint(self.enter_fd, "raw_syscalls",
+                                             "sys_enter"))
+        self.tp_fds.append(attach_tracepoint(self.exit_fd, "raw_syscalls",
+                                             "sys_exit"))
+        self._open_rings()
+
+    def _open_rings(self, pages: int = 64) -> None:
+        page = mmap.PAGESIZE
+        n_cpu = os.cpu_count() or 1
+        for cpu in range(n_cpu):
+            # PERF_TYPE_SOFTWARE/BPF_OUTPUT per-cpu ring the program
+            # writes into via the events map
+            PERF_COUNT_SW_BPF_OUTPUT = 10
+            attr = struct.pack("<IIQQQ", PERF_TYPE_SOFTWARE, 112,
+                               PERF_COUNT_SW_BPF_OUTPUT, 1, 0)
+            # sample_type = PERF_SAMPLE_RAW (bit 10 = 0x400)
+            attr = attr[:24] + struct.pack("<Q", 0x400) + attr[32:]
+            attr = attr.ljust(112, b"\x00")
+            fd = perf_event_open(attr, -1, cpu)
+            buf = mmap.mmap(fd, page * (pages + 1))
+            map_update(self.map_fds["events"],
+                       struct.pack("<I", cpu), struct.pack("<I", fd))
+            import fcntl
+            fcntl.ioctl(fd, PERF_EVENT_IOC_ENABLE, 0)
+            self.rings.append((fd, buf, pages))
+
+    def poll(self, on_event: Callable[[bytes], None]) -> int:
+        """Drain all per-cpu rings; returns records delivered."""
+        page = mmap.PAGESIZE
+        n = 0
+        for fd, buf, pages in self.rings:
+            head = struct.unpack_from("<Q", buf, 1024)[0]   # data_head
+            tail = struct.unpack_from("<Q", buf, 1032)[0]   # data_tail
+            size = page * pages
+            while tail < head:
+                off = page + (tail % size)
+                etype, _misc, esize = struct.unpack_from("<IHH", buf, off)
+                rec = bytes(buf[off + 8: off + esize])
+                if etype == 1:  # PERF_RECORD_SAMPLE: u32 size + raw data
+                    raw_size = struct.unpack_from("<I", rec, 0)[0]
+                    on_event(rec[4:4 + raw_size])
+                    n += 1
+                tail += esize
+            struct.pack_into("<Q", buf, 1032, head)
+        return n
+
+    def close(self) -> None:
+        for fd, buf, _ in self.rings:
+            buf.close()
+            os.close(fd)
+        for fd in self.tp_fds:
+            os.close(fd)
+        for fd in (self.enter_fd, self.exit_fd):
+            os.close(fd)
+        for fd in self.map_fds.values():
+            os.close(fd)

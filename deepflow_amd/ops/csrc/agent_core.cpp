@@ -87,6 +87,12 @@ struct FlowNode {
     std::vector<uint8_t> h2_carry[2];  // cross-segment frame reassembly
     std::vector<uint32_t> acl_gids;    // fast-path cached ACL matches
     uint32_t acl_actions = 0;
+    // eBPF socket-trace provenance (signal_source 3 = SIGNAL_SOURCE_EBPF):
+    // syscall trace ids per direction (0 = request/client side) and the
+    // owning processes, carried into AppProtoLogsBaseInfo 25/26/29/30
+    uint8_t signal_source = 0;
+    uint64_t sc_trace[2] = {0, 0};
+    uint32_t sc_tgid[2] = {0, 0};
 };
 
 struct FlowKeyC {
@@ -1195,6 +1201,12 @@ void encode_l7_record(Agent& a, FlowNode& f, uint64_t req_ts, uint64_t resp_ts,
         dfpb::f_u(s, 18, f.port[0]);
         dfpb::f_u(s, 19, f.port[1]);
         dfpb::f_u(s, 20, f.proto);
+        if (f.signal_source) {  // eBPF-sourced: process + syscall joins
+            dfpb::f_u(s, 25, f.sc_tgid[0]);
+            dfpb::f_u(s, 26, f.sc_tgid[1]);
+            dfpb::f_u(s, 29, f.sc_trace[0]);
+            dfpb::f_u(s, 30, f.sc_trace[1]);
+        }
     });
     dfpb::f_i(b, 9, pend.req_len);
     dfpb::f_m<2048>(b, 11, [&](Buf& s) {  // req
@@ -1316,6 +1328,7 @@ void encode_l4_record(Agent& a, FlowNode& f) {
             });
         }
         dfpb::f_u(fl, 14, f.close_type ? f.close_type : 3 /* timeout */);
+        if (f.signal_source) dfpb::f_u(fl, 15, f.signal_source);
         dfpb::f_u(fl, 16, 1);
         dfpb::f_u(fl, 18, f.emitted_new ? 0 : 1);
         dfpb::f_u(fl, 19, 1);
@@ -2486,6 +2499,97 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
     }
     if (paylen > 0) handle_l7_payload(a, f, dir, payload, paylen, ts_ns);
     return 0;
+}
+
+// eBPF socket-trace event entry (reference: EbpfCollector turning
+// SK_BPF_DATA into MetaPacket and re-using FlowMap,
+// agent/src/ebpf_dispatcher.rs:460-520). The kernel program ships
+// (tgid, fd, direction, syscall trace id, payload); userspace resolves
+// the socket 4-tuple from /proc (ebpf/runtime.py) and calls here. The
+// payload runs through the SAME FlowMap + L7 parsers as the packet path,
+// so TLS-terminated/loopback traffic the NIC path cannot see still
+// yields flow logs.
+int dfa_syscall_event(void* h, uint64_t ts_ns, uint32_t tgid, int dir,
+                      uint64_t sc_trace_id,
+                      uint32_t ip_local, uint32_t ip_remote,
+                      uint16_t port_local, uint16_t port_remote,
+                      uint8_t proto, uint8_t l7_hint,
+                      const uint8_t* payload, uint32_t len) {
+    Agent& a = *(Agent*)h;
+    // orient by data movement: a read's payload was SENT by the remote
+    // peer, a write's by the local process — the first data sender
+    // becomes the flow's client, exactly as in the packet path
+    uint32_t ip_src = dir ? ip_remote : ip_local;
+    uint32_t ip_dst = dir ? ip_local : ip_remote;
+    uint16_t port_src = dir ? port_remote : port_local;
+    uint16_t port_dst = dir ? port_local : port_remote;
+    bool a_first = (ip_src < ip_dst) ||
+                   (ip_src == ip_dst && port_src <= port_dst);
+    FlowKeyC key{a_first ? ip_src : ip_dst, a_first ? ip_dst : ip_src,
+                 (uint16_t)(a_first ? port_src : port_dst),
+                 (uint16_t)(a_first ? port_dst : port_src), proto};
+    auto it = a.flows.find(key);
+    if (it == a.flows.end()) {
+        FlowNode f;
+        f.flow_id = a.next_flow_id++;
+        f.start_ns = f.last_ns = ts_ns;
+        f.ip[0] = ip_src; f.ip[1] = ip_dst;
+        f.port[0] = port_src; f.port[1] = port_dst;
+        f.proto = proto;
+        f.signal_source = 3;  // SIGNAL_SOURCE_EBPF
+        if (l7_hint) f.l7_protocol = l7_hint;
+        match_acls(a, f);
+        it = a.flows.emplace(key, std::move(f)).first;
+    }
+    FlowNode& f = it->second;
+    f.last_ns = ts_ns;
+    // data direction: which flow side sent these bytes
+    int data_dir = (ip_src == f.ip[0] && port_src == f.port[0]) ? 0 : 1;
+    // the local process sits on the side it writes from / reads toward
+    int local_side = dir ? (data_dir ^ 1) : data_dir;
+    f.sc_trace[data_dir] = sc_trace_id;
+    f.sc_tgid[local_side] = tgid;
+    PeerStats& ps = f.peer[data_dir];
+    ps.packets++; ps.total_packets++;
+    ps.bytes += len; ps.total_bytes += len;
+    ps.l4_bytes += len; ps.l3_bytes += len;
+    if (!ps.first_ns) ps.first_ns = ts_ns;
+    ps.last_ns = ts_ns;
+    if (len > 0) handle_l7_payload(a, f, data_dir, payload, len, ts_ns);
+    a.pkts++;
+    a.bytes += len;
+    return 0;
+}
+
+// batch entry: records framed as
+// [ts u64][tgid u32][dir u8][proto u8][l7_hint u8][pad u8]
+// [ip_src u32][ip_dst u32][port_src u16][port_dst u16]
+// [sc_trace u64][len u32][payload]
+int64_t dfa_syscall_batch(void* h, const uint8_t* buf, uint64_t total) {
+    uint64_t pos = 0;
+    int64_t n = 0;
+    while (pos + 40 <= total) {
+        uint64_t ts, trace;
+        uint32_t tgid, ips, ipd, len;
+        uint16_t psrc, pdst;
+        uint8_t dir, proto, hint;
+        memcpy(&ts, buf + pos, 8);
+        memcpy(&tgid, buf + pos + 8, 4);
+        dir = buf[pos + 12]; proto = buf[pos + 13]; hint = buf[pos + 14];
+        memcpy(&ips, buf + pos + 16, 4);
+        memcpy(&ipd, buf + pos + 20, 4);
+        memcpy(&psrc, buf + pos + 24, 2);
+        memcpy(&pdst, buf + pos + 26, 2);
+        memcpy(&trace, buf + pos + 28, 8);
+        memcpy(&len, buf + pos + 36, 4);
+        pos += 40;
+        if (pos + len > total) break;
+        dfa_syscall_event(h, ts, tgid, dir, trace, ips, ipd, psrc, pdst,
+                          proto, hint, buf + pos, len);
+        pos += len;
+        n++;
+    }
+    return n;
 }
 
 // Periodic tick: emit+drop closed/idle flows, roll meters into Documents.

@@ -1,0 +1,252 @@
+"""eBPF collection stack tests.
+
+The image has no BPF compiler or privileges, so the EXACT bytecode the
+loader would hand to the kernel runs under the userspace VM (ebpf/vm.py)
+against synthetic syscall streams; the resulting perf events flow through
+the real runtime (fd resolution -> agent_core FlowMap/L7 parse) into flow
+logs — the full path the reference exercises with SK_BPF_DATA
+(ebpf_dispatcher.rs:460-520)."""
+import struct
+
+import pytest
+
+from deepflow_amd.ebpf import insn as I
+from deepflow_amd.ebpf.inference import SPEC, infer
+from deepflow_amd.ebpf.progs import (EV_HDR, MAPS, SK_EVENT_FMT,
+                                     build_profiler, build_sys_enter,
+                                     build_sys_exit)
+from deepflow_amd.ebpf.runtime import (EbpfCollector, StaticResolver,
+                                       record_events)
+from deepflow_amd.ebpf.vm import SyscallSim, Vm
+
+CORPUS = [
+    (b"GET /api/v1/x HTTP/1.1\r\nHost: a\r\n\r\n", "http1"),
+    (b"POST /submit HTTP/1.1\r\n\r\n", "http1"),
+    (b"HTTP/1.1 200 OK\r\n\r\n", "http1"),
+    (b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n", "http2"),
+    (b"*2\r\n$3\r\nGET\r\n$1\r\nk\r\n", "redis"),
+    (b"+OK\r\n", "redis"),
+    (b"\x16\x03\x01\x00\xa5\x01\x00\x00", "tls"),
+    (b"\x21\x00\x00\x00\x03SELECT 1", "mysql"),
+    (b"Q\x00\x00\x00\x19SELECT * FROM t;\x00", "postgresql"),
+    (b"AMQP\x00\x00\x09\x01", "amqp"),
+    (b"PUB subj 5\r\nhello\r\n", "nats"),
+    (b"\x10\x20\x00\x04MQTT\x04\x02\x00\x3c", "mqtt"),
+    (b"\x00" * 16, None),
+    (b"randomgarbage!!", None),
+]
+
+
+def _name_of(proto_id):
+    for pid_, name, _ in SPEC:
+        if pid_ == proto_id:
+            return name
+    return None
+
+
+def test_inference_oracle_corpus():
+    for payload, want in CORPUS:
+        got = _name_of(infer(payload)) if infer(payload) else None
+        assert got == want, (payload, got, want)
+
+
+def test_assembler_encoding():
+    a = I.Asm()
+    a.mov64_imm(I.R0, 7)
+    a.ld_imm64(I.R1, 0x1122334455667788)
+    a.jmp_imm(I.BPF_JEQ, I.R0, 7, "out")
+    a.mov64_imm(I.R0, 0)
+    a.label("out")
+    a.exit()
+    raw = a.to_bytes(map_fds={})
+    assert len(raw) % 8 == 0
+    assert len(raw) == 8 * a.n_insns()
+    # run it: r0 must survive the taken jump
+    vm = Vm({})
+    assert vm.run(a, b"\x00" * 16) == 7
+
+
+def test_bpf_matcher_equals_python_oracle():
+    """The generated in-kernel matcher and the python twin agree on the
+    corpus AND on fuzzed payloads (one spec, two consumers)."""
+    import random
+    rng = random.Random(7)
+    payloads = [p for p, _ in CORPUS]
+    for _ in range(300):
+        n = rng.randrange(0, 64)
+        payloads.append(bytes(rng.randrange(256) for _ in range(n)))
+    for p in payloads:
+        sim = SyscallSim()
+        sim.syscall(10, 10, 1, 3, p)       # write on a fresh socket
+        evs = sim.events()
+        if not p:
+            continue
+        assert len(evs) == 1
+        hdr = struct.unpack(SK_EVENT_FMT, evs[0][:EV_HDR])
+        got_proto = hdr[7]
+        assert got_proto == infer(p[:192], len(p)), p
+
+
+def test_socket_tracer_event_fields():
+    sim = SyscallSim()
+    req = b"GET /x HTTP/1.1\r\n\r\n"
+    sim.syscall(2000, 2001, 0, 5, req)               # read (ingress)
+    sim.syscall(2000, 2001, 1, 5, b"HTTP/1.1 200 OK\r\n\r\n")
+    evs = sim.events()
+    assert len(evs) == 2
+    h_read = struct.unpack(SK_EVENT_FMT, evs[0][:EV_HDR])
+    h_write = struct.unpack(SK_EVENT_FMT, evs[1][:EV_HDR])
+    assert h_read[1] == 2000 and h_read[2] == 2001   # tgid/pid
+    assert h_read[3] == 5                            # fd
+    assert h_read[6] == 1 and h_write[6] == 0        # directions
+    assert h_read[9] == h_write[9] != 0              # trace id join
+    assert evs[0][EV_HDR:EV_HDR + len(req)] == req   # payload capture
+
+
+def test_syscall_events_to_flow_logs():
+    """VM events -> EbpfCollector -> agent_core parsers -> L7 flow log
+    with signal_source=EBPF + syscall_trace_ids (+ queryable)."""
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.wire import pb, flow_log, framing
+    sim = SyscallSim()
+    # server process 4000 on fd 9: HTTP request in, response out
+    sim.syscall(4000, 4000, 0, 9,
+                b"GET /api/orders HTTP/1.1\r\nHost: shop\r\n\r\n")
+    sim.syscall(4000, 4000, 1, 9,
+                b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+    # redis client 4001 on fd 4
+    sim.syscall(4001, 4001, 1, 4, b"*1\r\n$4\r\nPING\r\n")
+    sim.syscall(4001, 4001, 0, 4, b"+PONG\r\n")
+    agent = Agent(vtap_id=9)
+    resolver = StaticResolver({
+        # server: local 10.0.0.2:8080, remote client 10.0.0.1:51000
+        (4000, 9): (0x0A000002, 0x0A000001, 8080, 51000, 6),
+        (4001, 4): (0x0A000003, 0x0A000004, 52000, 6379, 6),
+    })
+    coll = EbpfCollector(agent, resolver)
+    blob = record_events(sim.events())
+    assert coll.replay(blob) == 4
+    assert coll.unresolved == 0
+    agent.tick(2_000_000_000_000_000_000)      # flush flows
+    out = agent.drain(1)                        # l7 records
+    recs = list(framing.iter_records(out))
+    assert len(recs) >= 2
+    by_proto = {}
+    for r in recs:
+        d = pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+        by_proto[d["base"]["head"]["proto"]] = d
+    http = by_proto[20]
+    assert http["req"]["resource"] == "/api/orders"
+    assert http["req"]["domain"] == "shop"
+    assert http["base"]["syscall_trace_id_request"] != 0
+    # the serving process (tgid 4000) lands on the server side
+    assert http["base"].get("process_id_1") == 4000
+    # request and response events carry the SAME minted trace id
+    assert http["base"]["syscall_trace_id_request"] == \
+        http["base"].get("syscall_trace_id_response")
+    redis = by_proto[80]
+    assert redis["req"]["req_type"] == "PING"
+    # l4 flow carries signal_source=EBPF
+    l4 = list(framing.iter_records(agent.drain(0)))
+    assert l4
+    srcs = {pb.decode(r, flow_log.TAGGED_FLOW)["flow"].get(
+        "signal_source", 0) for r in l4}
+    assert 3 in srcs
+
+
+def test_syscall_flow_logs_queryable():
+    """eBPF-sourced records run the normal GPU/CPU ingest + SQL path."""
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query.engine import QueryEngine
+    sim = SyscallSim()
+    for i in range(5):
+        sim.syscall(5000, 5000, 0, 3,
+                    b"GET /api/p%d HTTP/1.1\r\nHost: api\r\n\r\n" % i)
+        sim.syscall(5000, 5000, 1, 3, b"HTTP/1.1 200 OK\r\n\r\n")
+    agent = Agent(vtap_id=3)
+    coll = EbpfCollector(agent, StaticResolver(
+        {(5000, 3): (0x0A000005, 0x0A000006, 8080, 40000, 6)}))
+    coll.replay(record_events(sim.events()))
+    agent.tick(2_000_000_000_000_000_000)
+    payload = agent.drain(1)
+    pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 10,
+                            dict_capacity=1 << 10, time_base_s=0)
+    pipe.ingest_frame_payload(payload)
+    eng = QueryEngine(pipe, device="cpu")
+    r = eng.query("SELECT request_resource, Count(*) AS c FROM l7_flow_log "
+                  "WHERE request_domain = 'api' GROUP BY request_resource")
+    assert len(r["values"]) == 5
+
+
+def test_profiler_program_and_folding():
+    """The profiler bytecode counts (tgid, ustack, kstack) samples in the
+    VM; folding + symbolization produce flame-ready rows."""
+    from deepflow_amd.ebpf.progs import PROFILER_MAPS
+    from deepflow_amd.ebpf.profiler import CpuProfiler, DictSymbolizer, fold
+    from deepflow_amd.ingest.profile_pipeline import ProfilePipeline, \
+        build_flame
+    vm = Vm(PROFILER_MAPS)
+    prog = build_profiler()
+    vm.pid_tgid = (7777 << 32) | 7777
+    for _ in range(3):
+        vm.stackid_seq = {0x100: 1, 0: 11}   # same stacks every sample
+        vm.run(prog, b"\x00" * 8)
+    counts = vm.maps["counts"].data
+    assert len(counts) == 1
+    key, val = next(iter(counts.items()))
+    assert struct.unpack("<Q", bytes(val))[0] == 3
+    stacks = {1: [0x401000, 0x401500], 11: [0xFFFF0001]}
+    sym = DictSymbolizer({(7777, 0x401000): "handler",
+                          (7777, 0x401500): "main",
+                          (0, 0xFFFF0001): "tcp_sendmsg"})
+    rows = fold({bytes(k): struct.unpack("<Q", bytes(v))[0]
+                 for k, v in counts.items()}, stacks, sym,
+                comm_of=lambda t: "websrv")
+    assert rows == [(7777,
+                     b"websrv;main;handler;[k] tcp_sendmsg", 3)]
+    pipe = ProfilePipeline()
+    prof = CpuProfiler(pipe, symbolizer=sym)
+    n = prof.ingest_counts(
+        {bytes(k): 3 for k in counts}, stacks, 10**18,
+        comm_of=lambda t: "websrv")
+    assert n == 1 and pipe.store.rows
+    flame = build_flame(pipe.store.rows, pipe.store.id_to_loc)
+    names = str(flame)
+    assert "handler" in names and "tcp_sendmsg" in names
+
+
+def test_loader_bytecode_shape():
+    """Loadable bytes: map refs patch to fds, length is 8B-aligned, and
+    kernel availability degrades cleanly here."""
+    from deepflow_amd.ebpf import loader
+    fds = {name: 100 + i for i, name in enumerate(MAPS)}
+    for prog in (build_sys_enter(), build_sys_exit()):
+        raw = prog.to_bytes(fds)
+        assert len(raw) % 8 == 0 and len(raw) // 8 == prog.n_insns()
+    assert loader.available() in (False, True)  # no crash either way
+
+
+def test_inference_matches_packet_parsers():
+    """The shared SPEC agrees with the C++ packet parsers' protocol
+    detection for payload-identifiable protocols (one spec, two engines —
+    the reference keeps protocol_inference.h and check_payload in sync by
+    hand)."""
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import eth_ipv4_tcp, PSH_ACK
+    from deepflow_amd.wire import pb, flow_log, framing
+    cases = [(p, want) for p, want in CORPUS
+             if want in ("http1", "redis", "mysql", "postgresql", "kafka")
+             and not p.startswith(b"HTTP/") and not p.startswith(b"+")]
+    proto_ids = {name: pid_ for pid_, name, _ in SPEC}
+    for payload, want in cases:
+        agent = Agent(vtap_id=1)
+        pkt = eth_ipv4_tcp(0x0A000001, 0x0A000002, 40000, 9999,
+                           seq=1, flags=PSH_ACK, payload=payload)
+        agent.packet(pkt, 10**18)
+        agent.tick(3 * 10**18)
+        recs = list(framing.iter_records(agent.drain(1)))
+        if not recs:   # parser needs more context than one segment
+            continue
+        d = pb.decode(recs[0], flow_log.APP_PROTO_LOGS_DATA)
+        assert d["base"]["head"]["proto"] == proto_ids[want], payload
