@@ -114,6 +114,39 @@ class Agent:
         return int(self._lib.dfa_syscall_batch(self._h, buf.ctypes.data,
                                                len(blob)))
 
+    def start_ebpf(self, poll_interval_s: float = 0.05):
+        """Attach the socket tracer (raw bpf(2), ebpf/loader.py) and pump
+        its perf events into the FlowMap on a background thread. Returns
+        the tracer, or None where BPF is unavailable (the guard/melt-down
+        path: packet capture keeps running — reference disables eBPF on
+        old kernels the same way, utils/guard.rs:778-788)."""
+        from ..ebpf import loader, runtime
+        if not loader.available():
+            return None
+        tracer = loader.SocketTracer()
+        tracer.attach()
+        coll = runtime.EbpfCollector(self)
+        import threading
+        stop = threading.Event()
+
+        def pump():
+            while not stop.wait(poll_interval_s):
+                tracer.poll(coll.on_event)
+                coll.flush()
+        th = threading.Thread(target=pump, daemon=True)
+        th.start()
+        tracer._stop = stop  # noqa: SLF001 — owned here
+        self._ebpf = (tracer, coll, th)
+        return tracer
+
+    def stop_ebpf(self) -> None:
+        eb = getattr(self, "_ebpf", None)
+        if eb:
+            tracer, _coll, _th = eb
+            tracer._stop.set()
+            tracer.close()
+            self._ebpf = None
+
     def tick(self, now_ns: int) -> None:
         self._lib.dfa_tick(self._h, now_ns)
 
