@@ -75,12 +75,6 @@ class L4IngestPipeline:
         lens_t = torch.from_numpy(lens.view(np.int32)).to(dev, non_blocking=True)
         sstr = self._scratch(n, dev)
         gpu_ops.decode_l4(payload_t, offs_t, lens_t, seg, base, sstr)
-        epc0 = seg.u32[L4.U32_COLS.index("l3_epc_id_0"), base:base + n]
-        ip0 = seg.u32[L4.U32_COLS.index("ip4_0"), base:base + n]
-        epc1 = seg.u32[L4.U32_COLS.index("l3_epc_id_1"), base:base + n]
-        ip1 = seg.u32[L4.U32_COLS.index("ip4_1"), base:base + n]
-        gpu_ops.kg_probe_cols(epc0, ip0, epc1, ip1, n, self.kg.tkeys,
-                              self.kg.tvals, seg.kg, seg.capacity, base)
         # pool the string columns (request_domain + ip6 pair)
         pool_cols = torch.arange(L4.N_STR, dtype=torch.uint8, device=dev)
         row_len = torch.zeros(n, dtype=torch.int32, device=dev)
@@ -100,17 +94,6 @@ class L4IngestPipeline:
         pbytes = payload.tobytes()
         sstr = self._scratch(n, torch.device("cpu"))
         ref_l4.decode_l4_ref(pbytes, offs, lens, seg, base, sstr)
-        # KG join via the host mirror (independent oracle for the GPU probe)
-        for i in range(n):
-            row = base + i
-            for side, (ec, ic) in enumerate(
-                    [("l3_epc_id_0", "ip4_0"), ("l3_epc_id_1", "ip4_1")]):
-                epc = int(seg.u32[L4.U32_COLS.index(ec), row]) & 0xFFFFFFFF
-                ip = int(seg.u32[L4.U32_COLS.index(ic), row]) & 0xFFFFFFFF
-                info = self.kg.host.get((epc, ip))
-                vals = info.as_list() if info else [0] * S.N_KG
-                for j, v in enumerate(vals):
-                    seg.kg[side * S.N_KG + j, row] = v
         pool_cols = list(range(L4.N_STR))
         row_len = ref.pool_lens_ref(sstr, pool_cols, n)
         cum = torch.cumsum(row_len.to(torch.int64), 0)

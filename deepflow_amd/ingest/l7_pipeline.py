@@ -144,7 +144,6 @@ class L7IngestPipeline:
         dev = payload_t.device
         sstr, sattr = self._scratch(n, dev)
         gpu_ops.decode_l7(payload_t, offs_t, lens_t, seg, base, sstr, sattr)
-        gpu_ops.kg_probe(seg, base, n, self.kg.tkeys, self.kg.tvals)
         gpu_ops.intern_many(payload_t, sstr, self._ref_rows_scalar,
                             self._dom_scalar, 0, n, self.dict.tkeys,
                             self.dict.emit, self.dict.emit_ctr, seg.did, base)
@@ -184,7 +183,6 @@ class L7IngestPipeline:
         pb = payload.tobytes()
         sstr, sattr = self._scratch(n, torch.device("cpu"))
         ref.decode_l7_ref(pb, offs, lens, seg, base, sstr, sattr)
-        ref.kg_probe_ref(seg, base, n, self.kg.tkeys, self.kg.tvals)
         new = ref.intern_ref(pb, sstr, _SCALAR_DICT_REF_ROWS,
                              _SCALAR_DICT_DOMAINS, 0, n, self.dict.tkeys,
                              seg.did, base, dictionary=self.dict)

@@ -1095,7 +1095,13 @@ struct SegView {
     const uint32_t* u32c;
     const uint8_t* u8c;
     const uint32_t* didc;      // [L7_DID_N, stride]
-    const uint32_t* kgc;       // [2*KG_VALS_N, stride]
+    // KnowledgeGraph table for query-time join (SmartEncoding: the 24
+    // per-row KG id columns are NOT materialized — they are a pure
+    // function of the row's (epc, ip) key, resolved here by probing the
+    // GPU-resident platform table instead of spending 96 B/span of HBM)
+    const uint64_t* kg_tk;
+    const uint32_t* kg_tv;     // [cap, KG_VALS_N]
+    uint32_t kg_mask;
     const int32_t* attr_pool;  // variable attr-id pool
     const uint32_t* attr_start;  // [stride] row block offsets into attr_pool
     const uint8_t* attr_cnt;   // [stride]
@@ -1106,6 +1112,24 @@ struct SegView {
     uint64_t n_rows;
 };
 
+// epc/ip column indices are identical in the L7 and L4 u32 layouts
+// (vtap, ip4_0, ip4_1, epc_0, epc_1 — asserted in the layout headers)
+DEV uint64_t kg_join(const SegView& s, uint64_t row, uint16_t idx) {
+    if (s.kg_tk == nullptr) return 0;
+    uint32_t side = idx / KG_VALS_N, j = idx % KG_VALS_N;
+    uint32_t epc = s.u32c[(uint64_t)(3 + side) * s.stride + row];
+    uint32_t ip = s.u32c[(uint64_t)(1 + side) * s.stride + row];
+    uint64_t k = ((uint64_t)epc << 32) | ip;
+    uint32_t slot = (uint32_t)(mix64(k) & s.kg_mask);
+    for (uint32_t probe = 0; probe <= s.kg_mask; probe++) {
+        uint64_t tk = s.kg_tk[slot];
+        if (tk == k) return s.kg_tv[(uint64_t)slot * KG_VALS_N + j];
+        if (tk == EMPTY_KEY) return 0;
+        slot = (slot + 1) & s.kg_mask;
+    }
+    return 0;
+}
+
 DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
                        uint16_t idx, uint32_t bucket, uint64_t time_base_s) {
     switch (family) {
@@ -1113,7 +1137,7 @@ DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
         case SRC_U32: return s.u32c[(uint64_t)idx * s.stride + row];
         case SRC_U8:  return s.u8c[(uint64_t)idx * s.stride + row];
         case SRC_DID: return s.didc[(uint64_t)idx * s.stride + row];
-        case SRC_KG:  return s.kgc[(uint64_t)idx * s.stride + row];
+        case SRC_KG:  return kg_join(s, row, idx);
         case SRC_ATTR_VAL: {
             // idx is the attr slot (0..cnt-1); value ids follow name ids in
             // the row's attr-pool block
@@ -1545,7 +1569,9 @@ int df_rollup_l7(void* u64c, void* u32c, void* u8c, uint64_t stride,
 }
 
 int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
-                 const void* didc, const void* kgc, const void* attr_pool,
+                 const void* didc,
+                 const void* kg_tk, const void* kg_tv, uint32_t kg_cap,
+                 const void* attr_pool,
                  const void* attr_start, const void* attr_cnt,
                  const void* str_rowref, const void* str_lens,
                  const void* pool,
@@ -1555,7 +1581,9 @@ int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
                  void* gkeys, void* graw, void* gvals, uint32_t cap,
                  uint64_t stream) {
     SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
-              (const uint32_t*)didc, (const uint32_t*)kgc,
+              (const uint32_t*)didc,
+              (const uint64_t*)kg_tk, (const uint32_t*)kg_tv,
+              kg_cap ? kg_cap - 1 : 0,
               (const int32_t*)attr_pool, (const uint32_t*)attr_start,
               (const uint8_t*)attr_cnt, (const uint64_t*)str_rowref,
               (const int16_t*)str_lens,
@@ -1574,7 +1602,9 @@ int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
 }
 
 int df_query_select(const void* u64c, const void* u32c, const void* u8c,
-                    const void* didc, const void* kgc, const void* attr_pool,
+                    const void* didc,
+                    const void* kg_tk, const void* kg_tv, uint32_t kg_cap,
+                    const void* attr_pool,
                     const void* attr_start, const void* attr_cnt,
                     const void* str_rowref, const void* str_lens,
                     const void* pool,
@@ -1583,7 +1613,9 @@ int df_query_select(const void* u64c, const void* u32c, const void* u8c,
                     void* out_rows, void* out_ctr, uint32_t out_cap,
                     uint64_t stream) {
     SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
-              (const uint32_t*)didc, (const uint32_t*)kgc,
+              (const uint32_t*)didc,
+              (const uint64_t*)kg_tk, (const uint32_t*)kg_tv,
+              kg_cap ? kg_cap - 1 : 0,
               (const int32_t*)attr_pool, (const uint32_t*)attr_start,
               (const uint8_t*)attr_cnt, (const uint64_t*)str_rowref,
               (const int16_t*)str_lens,

@@ -66,17 +66,6 @@ def rollup_insert(kws: torch.Tensor, vals: torch.Tensor, ops: torch.Tensor,
         stream or _stream()), "df_rollup_insert")
 
 
-def kg_probe_cols(epc0, ip0, epc1, ip1, n: int, tkeys: torch.Tensor,
-                  tvals: torch.Tensor, out_kg: torch.Tensor,
-                  stride: int, base_row: int) -> None:
-    """Generic KG probe over explicit column row views."""
-    lib = native.gpu()
-    native.check(lib.df_kg_probe(
-        epc0.data_ptr(), ip0.data_ptr(), epc1.data_ptr(), ip1.data_ptr(), n,
-        tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(),
-        out_kg.data_ptr(), stride, base_row, _stream()), "df_kg_probe")
-
-
 def kg_build(keys: torch.Tensor, vals: torch.Tensor, tkeys: torch.Tensor,
              tvals: torch.Tensor) -> None:
     lib = native.gpu()
@@ -84,22 +73,6 @@ def kg_build(keys: torch.Tensor, vals: torch.Tensor, tkeys: torch.Tensor,
         keys.data_ptr(), vals.data_ptr(), keys.numel(),
         tkeys.data_ptr(), tvals.data_ptr(), tkeys.numel(), _stream()),
         "df_kg_build")
-
-
-def kg_probe(seg, base_row: int, n: int, tkeys: torch.Tensor,
-             tvals: torch.Tensor) -> None:
-    from ..store import l7_schema as S
-    lib = native.gpu()
-    u32 = seg.u32
-    cap = seg.capacity
-    epc0 = u32[S.U32_COLS.index("l3_epc_id_0")].data_ptr() + 4 * base_row
-    ip0 = u32[S.U32_COLS.index("ip4_0")].data_ptr() + 4 * base_row
-    epc1 = u32[S.U32_COLS.index("l3_epc_id_1")].data_ptr() + 4 * base_row
-    ip1 = u32[S.U32_COLS.index("ip4_1")].data_ptr() + 4 * base_row
-    native.check(lib.df_kg_probe(
-        epc0, ip0, epc1, ip1, n, tkeys.data_ptr(), tvals.data_ptr(),
-        tkeys.numel(), seg.kg.data_ptr(), cap, base_row, _stream()),
-        "df_kg_probe")
 
 
 def intern_many(payload: torch.Tensor, refs: torch.Tensor,
@@ -164,13 +137,14 @@ def _opt_ptr(seg, attr: str) -> int:
 
 def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
               gkeys: torch.Tensor, graw: torch.Tensor,
-              gvals: torch.Tensor) -> None:
+              gvals: torch.Tensor, kg=None) -> None:
     import ctypes
     lib = native.gpu()
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
+    ktk, ktv, kcap = _kg_args(kg)
     native.check(lib.df_query_agg(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_pool"),
+        _opt_ptr(seg, "did"), ktk, ktv, kcap, _opt_ptr(seg, "attr_pool"),
         _opt_ptr(seg, "attr_start"), _opt_ptr(seg, "attr_cnt"),
         seg.str_rowref.data_ptr(), seg.str_lens.data_ptr(),
         seg.pool.data_ptr(), seg.capacity, seg.n_rows,
@@ -179,14 +153,23 @@ def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
         _stream()), "df_query_agg")
 
 
+def _kg_args(kg):
+    """KnowledgeGraph table pointers for the query-time join."""
+    if kg is None:
+        return 0, 0, 0
+    return kg.tkeys.data_ptr(), kg.tvals.data_ptr(), kg.tkeys.numel()
+
+
 def query_select(seg, spec_bytes: bytes, base_row: int, n: int,
-                 out_rows: torch.Tensor, out_ctr: torch.Tensor) -> None:
+                 out_rows: torch.Tensor, out_ctr: torch.Tensor,
+                 kg=None) -> None:
     import ctypes
     lib = native.gpu()
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
+    ktk, ktv, kcap = _kg_args(kg)
     native.check(lib.df_query_select(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        _opt_ptr(seg, "did"), seg.kg.data_ptr(), _opt_ptr(seg, "attr_pool"),
+        _opt_ptr(seg, "did"), ktk, ktv, kcap, _opt_ptr(seg, "attr_pool"),
         _opt_ptr(seg, "attr_start"), _opt_ptr(seg, "attr_cnt"),
         seg.str_rowref.data_ptr(), seg.str_lens.data_ptr(),
         seg.pool.data_ptr(), seg.capacity, seg.n_rows,

@@ -98,8 +98,6 @@ def _decl_gpu(lib: ct.CDLL) -> None:
                                      p, u64]
     lib.df_kg_build.restype = ct.c_int
     lib.df_kg_build.argtypes = [p, p, u32, p, p, u32, u64]
-    lib.df_kg_probe.restype = ct.c_int
-    lib.df_kg_probe.argtypes = [p, p, p, p, u32, p, p, u32, p, u64, u64, u64]
     lib.df_intern_many.restype = ct.c_int
     lib.df_intern_many.argtypes = [p, p, p, p, u32, u32, u64, u64, p, u32,
                                    p, p, u32, p, u64, u64, u64]
@@ -111,11 +109,12 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_pool_gather.restype = ct.c_int
     lib.df_pool_gather.argtypes = [p, p, p, u32, u32, u64, u64, p, p, u64, p, p, u64, u64, u64]
     lib.df_query_agg.restype = ct.c_int
-    lib.df_query_agg.argtypes = [p, p, p, p, p, p, p, p, p, p, p, u64, u64,
-                                 p, u32, u64, p, p, p, u32, u64]
+    lib.df_query_agg.argtypes = [p, p, p, p, p, p, u32, p, p, p, p, p, p,
+                                 u64, u64, p, u32, u64, p, p, p, u32, u64]
     lib.df_query_select.restype = ct.c_int
-    lib.df_query_select.argtypes = [p, p, p, p, p, p, p, p, p, p, p, u64,
-                                    u64, p, u32, u64, p, p, u32, u64]
+    lib.df_query_select.argtypes = [p, p, p, p, p, p, u32, p, p, p, p, p,
+                                    p, u64, u64, p, u32, u64, p, p, u32,
+                                    u64]
     lib.df_spec_sizes.restype = ct.c_int
     lib.df_spec_sizes.argtypes = [p, p, p, p]
     lib.df_pack_bits.restype = ct.c_int

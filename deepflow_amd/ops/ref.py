@@ -218,37 +218,6 @@ def kg_build_ref(keys: torch.Tensor, vals: torch.Tensor,
             slot = (slot + 1) & cap_mask
 
 
-def kg_probe_ref(seg, base_row: int, n: int, tkeys: torch.Tensor,
-                 tvals: torch.Tensor) -> None:
-    cap_mask = tkeys.numel() - 1
-    tk = tkeys.numpy()
-    tv = tvals.numpy()
-    epc0 = seg.u32[_U32_IDX["l3_epc_id_0"]].numpy()
-    ip0 = seg.u32[_U32_IDX["ip4_0"]].numpy()
-    epc1 = seg.u32[_U32_IDX["l3_epc_id_1"]].numpy()
-    ip1 = seg.u32[_U32_IDX["ip4_1"]].numpy()
-    kg = seg.kg.numpy()
-    for i in range(n):
-        row = base_row + i
-        for side in range(2):
-            epc = int(epc0[row] if side == 0 else epc1[row]) & 0xFFFFFFFF
-            ip = int(ip0[row] if side == 0 else ip1[row]) & 0xFFFFFFFF
-            k = (epc << 32) | ip
-            slot = mix64(k) & cap_mask
-            found = False
-            for _ in range(cap_mask + 1):
-                cur = int(tk[slot]) & M64
-                if cur == k:
-                    found = True
-                    break
-                if cur == 0:
-                    break
-                slot = (slot + 1) & cap_mask
-            for j in range(S.N_KG):
-                kg[side * S.N_KG + j, row] = tv[slot, j] if found else 0
-
-
-# ------------------------------------------------------------- K3 intern
 
 def intern_ref(payload: bytes, refs: torch.Tensor, ref_rows, domains,
                ref_base_row: int, n: int, tkeys: torch.Tensor,

@@ -382,7 +382,7 @@ class QueryEngine:
             return {"kind": "rows",
                     "result": self._run_select(plan, segments, tags,
                                                str_cols)}
-        groups = execute(plan, segments, self.device)
+        groups = execute(plan, segments, self.device, kg=self.pipe.kg)
         key_rows = []
         aggs = []
         raw_keys = []
@@ -418,7 +418,8 @@ class QueryEngine:
         import math
         from .executor import execute_grouped_values
         uniq, per_meta = execute_grouped_values(plan, segments, q_metas,
-                                                self.device)
+                                                self.device,
+                                                kg=self.pipe.kg)
         index = {tuple(int(x) & ((1 << 64) - 1) for x in uniq[i].tolist()):
                  i for i in range(uniq.shape[0])}
         out = []
@@ -516,10 +517,12 @@ class QueryEngine:
         return self._qbucket_value(max(hist))
 
     # ----------------------------------------------------------- segments
-    def _run_segments(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:
+    def _run_segments(self, plan: Q.Plan, segments, tags, str_cols,
+                      kg=None) -> Dict:
+        kg = kg if kg is not None else self.pipe.kg
         if plan.select_rows:
-            return self._run_select(plan, segments, tags, str_cols)
-        groups = execute(plan, segments, self.device)
+            return self._run_select(plan, segments, tags, str_cols, kg=kg)
+        groups = execute(plan, segments, self.device, kg=kg)
         q_metas = [m for m in plan.agg_meta
                    if m["op"] in ("percentile", "apdex")]
         q_lookup = None
@@ -528,7 +531,8 @@ class QueryEngine:
             # cap): values sorted within each group for quantile math
             from .executor import execute_grouped_values
             uniq, per_meta = execute_grouped_values(plan, segments,
-                                                    q_metas, self.device)
+                                                    q_metas, self.device,
+                                                    kg=kg)
             key_index = {tuple(int(x) & ((1 << 64) - 1)
                                for x in uniq[i].tolist()): i
                          for i in range(uniq.shape[0])}
@@ -600,8 +604,10 @@ class QueryEngine:
         return [r for r in rows if tuple(r[i] for i in non_time) in keep]
 
     # ----------------------------------------------------------- select
-    def _run_select(self, plan: Q.Plan, segments, tags, str_cols) -> Dict:
-        hits = execute(plan, segments, self.device)
+    def _run_select(self, plan: Q.Plan, segments, tags, str_cols,
+                    kg=None) -> Dict:
+        hits = execute(plan, segments, self.device,
+                       kg=kg if kg is not None else self.pipe.kg)
         cols = plan.select_cols
         if cols == ["*"]:
             cols = ["start_time", "end_time", "flow_id", "l7_protocol",
@@ -653,7 +659,14 @@ class QueryEngine:
                 return self._hydrate(td.hydrate,
                                      int(seg.did[idx, row]) & 0xFFFFFFFF)
             if fam == Q.SRC_KG:
-                return self._hydrate(td.hydrate, int(seg.kg[idx, row]))
+                # query-time KG join (ids not materialized per row)
+                from ..store import l7_schema as S7
+                side, j = idx // S7.N_KG, idx % S7.N_KG
+                epc = int(seg.u32[3 + side, row]) & 0xFFFFFFFF
+                ip = int(seg.u32[1 + side, row]) & 0xFFFFFFFF
+                info = self.pipe.kg.host.get((epc, ip))
+                return self._hydrate(td.hydrate,
+                                     info.as_list()[j] if info else 0)
         if col in str_cols:
             # pooled string columns: row block ref + per-col u16 lens
             # (dict-encoded string tags were already handled via SRC_DID)

@@ -25,7 +25,7 @@ GROUP_CAP = 1 << 20
 
 
 def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
-            n: int) -> np.ndarray:
+            n: int, kg=None) -> np.ndarray:
     """Vectorized src_value over rows [0, n) as uint64 (CPU oracle)."""
     def u(t):
         return t[:, :n][idx].numpy()
@@ -38,7 +38,25 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
     if family == SRC_DID:
         return seg.did[idx, :n].numpy().view(np.uint32).astype(np.uint64)
     if family == SRC_KG:
-        return seg.kg[idx, :n].numpy().view(np.uint32).astype(np.uint64)
+        # query-time KnowledgeGraph join from the row's (epc, ip) key
+        # (KG ids are not materialized per row — SmartEncoding)
+        from ..store import l7_schema as S7
+        side, j = idx // S7.N_KG, idx % S7.N_KG
+        epc = seg.u32[3 + side, :n].numpy().view(np.uint32)
+        ip = seg.u32[1 + side, :n].numpy().view(np.uint32)
+        out = np.zeros(n, dtype=np.uint64)
+        if kg is not None:
+            host = kg.host
+            cache = {}
+            for i in range(n):
+                key = (int(epc[i]), int(ip[i]))
+                v = cache.get(key)
+                if v is None:
+                    info = host.get(key)
+                    v = info.as_list()[j] if info is not None else 0
+                    cache[key] = v
+                out[i] = v
+        return out
     if family == SRC_ATTR_VAL:
         starts = seg.attr_start[:n].numpy()
         cnts = seg.attr_cnt[:n].numpy()
@@ -69,7 +87,7 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
     return np.zeros(n, dtype=np.uint64)
 
 
-def _term_mask(seg, t, plan: Plan, n: int) -> np.ndarray:
+def _term_mask(seg, t, plan: Plan, n: int, kg=None) -> np.ndarray:
     if t.family == SRC_ATTR_MATCH:
         starts = seg.attr_start[:n].numpy()
         cnts = seg.attr_cnt[:n].numpy()
@@ -86,7 +104,7 @@ def _term_mask(seg, t, plan: Plan, n: int) -> np.ndarray:
         if t.op == OP_NE:
             out = ~out
         return out
-    v = _src_np(seg, t.family, t.idx, 0, plan.time_base_s, n)
+    v = _src_np(seg, t.family, t.idx, 0, plan.time_base_s, n, kg=kg)
     v0 = np.uint64(t.v0 & U64MAX)
     v1 = np.uint64(t.v1 & U64MAX)
     if t.op == OP_EQ:
@@ -106,11 +124,11 @@ def _term_mask(seg, t, plan: Plan, n: int) -> np.ndarray:
     return np.ones(n, dtype=bool)
 
 
-def _mask_np(seg, plan: Plan, n: int) -> np.ndarray:
+def _mask_np(seg, plan: Plan, n: int, kg=None) -> np.ndarray:
     mask = np.ones(n, dtype=bool)
     groups = {}
     for t in plan.terms:
-        tm = _term_mask(seg, t, plan, n)
+        tm = _term_mask(seg, t, plan, n, kg=kg)
         g = getattr(t, "group", 0)
         if g == 0:
             mask &= tm
@@ -121,16 +139,17 @@ def _mask_np(seg, plan: Plan, n: int) -> np.ndarray:
     return mask
 
 
-def execute_agg_cpu(plan: Plan, segments) -> List[Dict]:
+def execute_agg_cpu(plan: Plan, segments, kg=None) -> List[Dict]:
     groups: Dict[tuple, List[int]] = {}
     for seg in segments:
         n = seg.n_rows
         if n == 0:
             continue
-        mask = _mask_np(seg, plan, n)
+        mask = _mask_np(seg, plan, n, kg=kg)
         if not mask.any():
             continue
-        keys = [_src_np(seg, k.family, k.idx, k.bucket, plan.time_base_s, n)[mask]
+        keys = [_src_np(seg, k.family, k.idx, k.bucket, plan.time_base_s, n,
+                        kg=kg)[mask]
                 for k in plan.keys]
         aggvals = []
         for a in plan.aggs:
@@ -138,7 +157,7 @@ def execute_agg_cpu(plan: Plan, segments) -> List[Dict]:
                 aggvals.append(np.ones(int(mask.sum()), dtype=np.uint64))
             else:
                 aggvals.append(_src_np(seg, a.family, a.idx, 0,
-                                       plan.time_base_s, n)[mask])
+                                       plan.time_base_s, n, kg=kg)[mask])
         nk = len(keys)
         rows = int(mask.sum())
         key_tup = np.empty((rows, nk), dtype=np.uint64)
@@ -164,7 +183,7 @@ def execute_agg_cpu(plan: Plan, segments) -> List[Dict]:
     return [{"key": list(k), "agg": list(v)} for k, v in groups.items()]
 
 
-def execute_agg_gpu(plan: Plan, segments, device="cuda") -> List[Dict]:
+def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
     from ..ops import gpu_ops
     dev = torch.device(device)
     gkeys = torch.zeros(GROUP_CAP, dtype=torch.int64, device=dev)
@@ -177,7 +196,8 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda") -> List[Dict]:
     for seg in segments:
         if seg.n_rows == 0:
             continue
-        gpu_ops.query_agg(seg, spec, 0, seg.n_rows, gkeys, graw, gvals)
+        gpu_ops.query_agg(seg, spec, 0, seg.n_rows, gkeys, graw, gvals,
+                          kg=kg)
     torch.cuda.synchronize()
     mask = gkeys != 0
     raw = graw[mask].cpu().numpy().view(np.uint64)
@@ -190,14 +210,15 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda") -> List[Dict]:
     return out
 
 
-def execute_select_cpu(plan: Plan, segments, limit: int) -> List[int]:
+def execute_select_cpu(plan: Plan, segments, limit: int,
+                       kg=None) -> List[int]:
     """Returns (segment_idx, row) pairs encoded as global row ids."""
     out = []
     for si, seg in enumerate(segments):
         n = seg.n_rows
         if n == 0:
             continue
-        mask = _mask_np(seg, plan, n)
+        mask = _mask_np(seg, plan, n, kg=kg)
         rows = np.nonzero(mask)[0]
         for r in rows:
             out.append((si, int(r)))
@@ -207,7 +228,7 @@ def execute_select_cpu(plan: Plan, segments, limit: int) -> List[int]:
 
 
 def execute_select_gpu(plan: Plan, segments, limit: int,
-                       device="cuda") -> List[int]:
+                       device="cuda", kg=None) -> List[int]:
     from ..ops import gpu_ops
     dev = torch.device(device)
     out = []
@@ -218,7 +239,8 @@ def execute_select_gpu(plan: Plan, segments, limit: int,
         cap = min(limit * 4 + 1024, 1 << 22)
         out_rows = torch.zeros(cap, dtype=torch.int64, device=dev)
         out_ctr = torch.zeros(1, dtype=torch.int32, device=dev)
-        gpu_ops.query_select(seg, spec, 0, seg.n_rows, out_rows, out_ctr)
+        gpu_ops.query_select(seg, spec, 0, seg.n_rows, out_rows, out_ctr,
+                             kg=kg)
         torch.cuda.synchronize()
         cnt = min(int(out_ctr.item()), cap)
         rows = sorted(out_rows[:cnt].cpu().tolist())
@@ -229,17 +251,17 @@ def execute_select_gpu(plan: Plan, segments, limit: int,
     return out
 
 
-def execute(plan: Plan, segments, device: str = "cpu"):
+def execute(plan: Plan, segments, device: str = "cpu", kg=None):
     if plan.impossible:
         return []
     if plan.select_rows:
         limit = plan.limit or 100
         if device == "cpu":
-            return execute_select_cpu(plan, segments, limit)
-        return execute_select_gpu(plan, segments, limit, device)
+            return execute_select_cpu(plan, segments, limit, kg=kg)
+        return execute_select_gpu(plan, segments, limit, device, kg=kg)
     if device == "cpu":
-        return execute_agg_cpu(plan, segments)
-    return execute_agg_gpu(plan, segments, device)
+        return execute_agg_cpu(plan, segments, kg=kg)
+    return execute_agg_gpu(plan, segments, device, kg=kg)
 
 
 # ---------------------------------------------------------------------
@@ -251,7 +273,45 @@ def execute(plan: Plan, segments, device: str = "cpu"):
 _M32 = 0xFFFFFFFF
 
 
-def _col_torch(seg, family, idx, bucket, time_base_s, rows_t):
+def _kg_lookup_torch(kg, seg, rows_t, idx):
+    """Vectorized open-addressing probe of the KG table for the selected
+    rows (query-time join, device-resident)."""
+    from ..store import l7_schema as S7
+    side, j = idx // S7.N_KG, idx % S7.N_KG
+    epc = seg.u32[3 + side].index_select(0, rows_t).to(torch.int64) & _M32
+    ip = seg.u32[1 + side].index_select(0, rows_t).to(torch.int64) & _M32
+    key = (epc << 32) | ip
+    if kg is None:
+        return torch.zeros_like(key)
+    tk, tv = kg.tkeys, kg.tvals
+    if tk.device != key.device:
+        key = key.to(tk.device)
+    mask = tk.numel() - 1
+    # exact u64 mix64 via numpy (torch int64 shifts are arithmetic)
+    import numpy as np
+    kn = key.cpu().numpy().view(np.uint64)
+    zn = kn.copy()
+    zn = (zn ^ (zn >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+    zn = (zn ^ (zn >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    zn = zn ^ (zn >> np.uint64(31))
+    slot = torch.from_numpy((zn & np.uint64(mask)).astype(np.int64)).to(
+        tk.device)
+    out = torch.zeros(key.numel(), dtype=torch.int64, device=tk.device)
+    live = torch.ones(key.numel(), dtype=torch.bool, device=tk.device)
+    for _ in range(64):  # bounded probe chain
+        tkv = tk.index_select(0, slot)
+        hit = live & (tkv == key)
+        if hit.any():
+            out[hit] = tv.view(-1)[slot[hit] * tv.shape[1] + j].to(
+                torch.int64) & _M32
+        live = live & (tkv != key) & (tkv != 0)
+        if not bool(live.any()):
+            break
+        slot = torch.where(live, (slot + 1) & mask, slot)
+    return out.to(rows_t.device) if out.device != rows_t.device else out
+
+
+def _col_torch(seg, family, idx, bucket, time_base_s, rows_t, kg=None):
     if family == SRC_U64:
         return seg.u64[idx].index_select(0, rows_t)
     if family == SRC_U32:
@@ -261,7 +321,7 @@ def _col_torch(seg, family, idx, bucket, time_base_s, rows_t):
     if family == SRC_DID:
         return seg.did[idx].index_select(0, rows_t).to(torch.int64) & _M32
     if family == SRC_KG:
-        return seg.kg[idx].index_select(0, rows_t).to(torch.int64) & _M32
+        return _kg_lookup_torch(kg, seg, rows_t, idx)
     if family == SRC_TIME_BUCKET:
         t_s = seg.u64[0].index_select(0, rows_t) // 10**9
         rel = torch.clamp(t_s - time_base_s, min=0)
@@ -271,10 +331,10 @@ def _col_torch(seg, family, idx, bucket, time_base_s, rows_t):
     raise ValueError(f"unsupported tensor family {family}")
 
 
-def _matching_rows_t(plan: Plan, seg, device):
+def _matching_rows_t(plan: Plan, seg, device, kg=None):
     n = seg.n_rows
     if device == "cpu" or seg.u64.device.type == "cpu":
-        mask = _mask_np(seg, plan, n)
+        mask = _mask_np(seg, plan, n, kg=kg)
         return torch.from_numpy(np.nonzero(mask)[0].copy())
     from ..ops import gpu_ops
     dev = seg.u64.device
@@ -282,13 +342,14 @@ def _matching_rows_t(plan: Plan, seg, device):
     out_rows = torch.zeros(cap, dtype=torch.int64, device=dev)
     out_ctr = torch.zeros(1, dtype=torch.int32, device=dev)
     spec = plan.to_bytes()
-    gpu_ops.query_select(seg, spec, 0, n, out_rows, out_ctr)
+    gpu_ops.query_select(seg, spec, 0, n, out_rows, out_ctr, kg=kg)
     torch.cuda.synchronize()
     cnt = min(int(out_ctr.item()), cap)
     return out_rows[:cnt]
 
 
-def execute_grouped_values(plan: Plan, segments, metas, device: str):
+def execute_grouped_values(plan: Plan, segments, metas, device: str,
+                           kg=None):
     """-> (uniq_keys [g, nk] int64 cpu, per-meta dict
     {mi: (sorted_vals float64 cpu, starts, counts)}) for quantile
     finishing. Raises ValueError for unsupported key/metric families."""
@@ -300,18 +361,20 @@ def execute_grouped_values(plan: Plan, segments, metas, device: str):
     for seg in segments:
         if seg.n_rows == 0:
             continue
-        rows_t = _matching_rows_t(sub, seg, device)
+        rows_t = _matching_rows_t(sub, seg, device, kg=kg)
         if rows_t.numel() == 0:
             continue
         if rows_t.device != seg.u64.device:
             rows_t = rows_t.to(seg.u64.device)
         for ki, k in enumerate(plan.keys):
             key_cols[ki].append(_col_torch(seg, k.family, k.idx, k.bucket,
-                                           plan.time_base_s, rows_t).cpu())
+                                           plan.time_base_s, rows_t,
+                                           kg=kg).cpu())
         for mi, meta in enumerate(metas):
             val_cols[mi].append(_col_torch(seg, meta["family"],
                                            meta["idx"], 0,
-                                           plan.time_base_s, rows_t).cpu())
+                                           plan.time_base_s, rows_t,
+                                           kg=kg).cpu())
     if plan.keys and not key_cols[0]:
         return torch.zeros((0, len(plan.keys)), dtype=torch.int64), {}
     if plan.keys:

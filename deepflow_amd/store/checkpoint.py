@@ -29,6 +29,19 @@ def register_migration(from_version: int):
     return deco
 
 
+@register_migration(3)
+def _v3_to_v4(payload: Dict) -> Dict:
+    """v4 dropped the per-row KG block (query-time join): discard the
+    saved kg tensors; the KG table itself still restores."""
+    for st in payload.get("segments", []):
+        st.pop("kg", None)
+    for st in payload.get("cold", []):
+        st.pop("kg_cols", None)
+        st["layout_version"] = 4
+    payload["layout_version"] = 4
+    return payload
+
+
 @register_migration(2)
 def _v2_to_v3(payload: Dict) -> Dict:
     """v3 appended two pooled ip6 columns to the string block: pad the
@@ -50,7 +63,7 @@ def _seg_state(seg) -> Dict:
     out = {"n_rows": n, "capacity": seg.capacity,
            "pool_len": seg.pool_len,
            "attr_pool_len": getattr(seg, "attr_pool_len", 0)}
-    for name in ("u64", "u32", "u8", "did", "kg", "str_lens"):
+    for name in ("u64", "u32", "u8", "did", "str_lens"):
         t = getattr(seg, name, None)
         if t is not None:
             out[name] = t[..., :n].cpu().clone()
@@ -65,7 +78,7 @@ def _seg_state(seg) -> Dict:
 
 def _seg_restore(state: Dict, seg) -> None:
     n = state["n_rows"]
-    for name in ("u64", "u32", "u8", "did", "kg", "str_lens"):
+    for name in ("u64", "u32", "u8", "did", "str_lens"):
         if name in state:
             getattr(seg, name)[..., :n] = state[name].to(seg.device)
     seg.str_rowref[:n] = state["str_rowref"].to(seg.device)
@@ -93,7 +106,7 @@ def _cold_state(c) -> Dict:
     return {"n_rows": c.n_rows, "capacity": c.capacity,
             "layout_version": c.layout_version,
             "u64_cols": cols(c.u64_cols), "u32_cols": cols(c.u32_cols),
-            "did_cols": cols(c.did_cols), "kg_cols": cols(c.kg_cols),
+            "did_cols": cols(c.did_cols),
             "rowref_col": cols(c.rowref_col),
             "u8": c.u8.cpu().clone(), "str_lens": c.str_lens.cpu().clone(),
             "attr_start": c.attr_start.cpu().clone()
@@ -124,7 +137,6 @@ def _cold_restore(state: Dict, device: str):
     c.u64_cols = cols(state["u64_cols"])
     c.u32_cols = cols(state["u32_cols"])
     c.did_cols = cols(state["did_cols"])
-    c.kg_cols = cols(state["kg_cols"])
     c.rowref_col = cols(state["rowref_col"])
     for name in ("u8", "str_lens", "attr_start", "attr_cnt", "attr_pool",
                  "pool"):

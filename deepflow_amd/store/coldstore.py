@@ -192,7 +192,6 @@ class CompressedL7Segment:
         self.u32_cols = _compress_i32_matrix(seg.u32, n, stream)
         self.did_cols = _compress_i32_matrix(seg.did, n, stream) \
             if hasattr(seg, "did") else []
-        self.kg_cols = _compress_i32_matrix(seg.kg, n, stream)
         self.rowref_col = _compress_i32_matrix(seg.str_rowref.view(1, -1),
                                                n, stream)
         # small/raw blocks (u8 is already 1 B/row; pools are variable)
@@ -214,7 +213,7 @@ class CompressedL7Segment:
     def compressed_bytes(self) -> int:
         total = 0
         for group in (self.u64_cols, self.u32_cols, self.did_cols,
-                      self.kg_cols, self.rowref_col):
+                      self.rowref_col):
             total += sum(pc.nbytes() for pc in group)
         for t in (self.u8, self.str_lens, self.attr_start, self.attr_cnt,
                   self.attr_pool, self.pool):
@@ -247,8 +246,6 @@ class CompressedL7Segment:
         if self.did_cols:
             _restore_i32_matrix(self.did_cols, seg.did, n, stream,
                                 only=want(Q.SRC_DID))
-        _restore_i32_matrix(self.kg_cols, seg.kg, n, stream,
-                            only=want(Q.SRC_KG))
         u8_only = want(Q.SRC_U8)
         if u8_only is None:
             seg.u8[:, :n] = self.u8

@@ -6,7 +6,8 @@ these stay in sync."""
 # shape or packing changes; segments record it so mixed-layout windows are
 # rejected instead of misread (migration = drain + reingest, matching the
 # reference's at-most-once durability posture).
-LAYOUT_VERSION = 3  # v3 = pooled ip6 columns; v2 = packed attr pool
+LAYOUT_VERSION = 4  # v4 = KG joined at query time (no per-row kg block);
+                    # v3 = pooled ip6 columns; v2 = packed attr pool
 
 U64_COLS = [
     "start_time", "end_time", "flow_id", "rrt", "syscall_trace_id_request",

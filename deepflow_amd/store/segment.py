@@ -30,7 +30,9 @@ class L7Segment:
         self.str_rowref = z((capacity,), torch.int64)
         self.str_lens = z((S.N_POOL, capacity), torch.int16)
         self.did = torch.full((S.N_DID, capacity), -1, dtype=torch.int32, device=dev)
-        self.kg = z((2 * S.N_KG, capacity), torch.int32)
+        # KG ids are NOT stored per row: the query engine joins them from
+        # the (epc, ip) key against the GPU-resident platform table
+        # (SmartEncoding — saves 2*N_KG*4 = 96 B/span of HBM)
         self.attr_start = z((capacity,), torch.int32)
         self.attr_pool = torch.full((capacity * 8,), -1, dtype=torch.int32,
                                     device=dev)
@@ -74,7 +76,7 @@ class L7Segment:
         if self.n_rows == 0:
             return 0.0
         fixed = (S.N_U64 * 8 + S.N_U32 * 4 + S.N_U8 * 1 + S.N_DID * 4 +
-                 2 * S.N_KG * 4 + 4 + 1 + 8 + S.N_POOL * 2)
+                 4 + 1 + 8 + S.N_POOL * 2)
         return fixed + (self.pool_len + 4 * self.attr_pool_len) / self.n_rows
 
 
@@ -93,7 +95,6 @@ class L4Segment:
         self.u8 = z((L4.N_U8, capacity), torch.uint8)
         self.str_rowref = z((capacity,), torch.int64)
         self.str_lens = z((L4.N_STR, capacity), torch.int16)
-        self.kg = z((2 * S.N_KG, capacity), torch.int32)
         self.pool = z((pool_capacity or capacity * 24,), torch.uint8)
         self.pool_len = 0
         self.n_rows = 0
@@ -114,7 +115,7 @@ class L4Segment:
         from . import l4_schema as L4
         if self.n_rows == 0:
             return 0.0
-        fixed = (L4.N_U64 * 8 + L4.N_U32 * 4 + L4.N_U8 + 2 * S.N_KG * 4 +
+        fixed = (L4.N_U64 * 8 + L4.N_U32 * 4 + L4.N_U8 +
                  8 + L4.N_STR * 2)
         return fixed + self.pool_len / self.n_rows
 
@@ -139,7 +140,7 @@ class SegmentSet:
     def seg_alloc_bytes(seg) -> int:
         total = 0
         for name in ("u64", "u32", "u8", "str_rowref", "str_lens", "did",
-                     "kg", "attr_start", "attr_pool", "attr_cnt", "pool"):
+                     "attr_start", "attr_pool", "attr_cnt", "pool"):
             t = getattr(seg, name, None)
             if t is not None:
                 total += t.numel() * t.element_size()
@@ -181,7 +182,7 @@ class SegmentSet:
 
     @staticmethod
     def reset_segment(seg) -> None:
-        for name in ("u64", "u32", "u8", "str_rowref", "str_lens", "kg",
+        for name in ("u64", "u32", "u8", "str_rowref", "str_lens",
                      "attr_start"):
             t = getattr(seg, name, None)
             if t is not None:

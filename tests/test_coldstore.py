@@ -41,7 +41,7 @@ def test_segment_compress_materialize_identical():
     seg = pipe.segments.segments[0]
     n = seg.n_rows
     snap = {k: getattr(seg, k)[..., :n].clone()
-            for k in ("u64", "u32", "u8", "did", "kg", "str_lens",
+            for k in ("u64", "u32", "u8", "did", "str_lens",
                       "attr_start", "attr_cnt")}
     rowref = seg.str_rowref[:n].clone()
     pool = seg.pool[: seg.pool_len].clone()

@@ -55,10 +55,26 @@ def test_fixed_columns_match(pipes):
     assert torch.equal(a.attr_start[:n], b.attr_start[:n].cpu())
 
 
-def test_kg_columns_match(pipes):
+def test_kg_join_matches(pipes):
+    """GPU query-time KG join (in-kernel probe) == CPU oracle join."""
+    from deepflow_amd.query.executor import _src_np, _kg_lookup_torch
+    from deepflow_amd.query import spec as Q
     cpu, gpu = pipes
     a, b = cpu.segments.segments[0], gpu.segments.segments[0]
-    assert torch.equal(a.kg[:, :N], b.kg[:, :N].cpu())
+    rows = torch.arange(N, device="cuda")
+    for idx in (0, S.N_KG - 1, S.N_KG, 2 * S.N_KG - 1):
+        want = _src_np(a, Q.SRC_KG, idx, 0, 0, N, kg=cpu.kg)
+        got = _kg_lookup_torch(gpu.kg, b, rows, idx).cpu().numpy()
+        assert (want.astype("int64") == got).all(), idx
+    # and through the query kernel: group by a KG tag
+    from deepflow_amd.query.engine import QueryEngine
+    qc = QueryEngine(cpu, device="cpu").query(
+        "SELECT pod_id_1, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY pod_id_1 ORDER BY c DESC, pod_id_1 LIMIT 10")
+    qg = QueryEngine(gpu, device="cuda").query(
+        "SELECT pod_id_1, Count(*) AS c FROM l7_flow_log "
+        "GROUP BY pod_id_1 ORDER BY c DESC, pod_id_1 LIMIT 10")
+    assert qc == qg
 
 
 def test_dict_hydration_matches(pipes):

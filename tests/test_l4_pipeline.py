@@ -50,14 +50,19 @@ def test_columns(pipe):
 
 
 def test_kg(pipe):
+    """Query-time KG join over L4 rows (no per-row kg block)."""
+    from deepflow_amd.query.executor import _src_np
+    from deepflow_amd.query import spec as Q
     seg = pipe.segments.segments[0]
+    col = _src_np(seg, Q.SRC_KG, S.KG_COLS.index("pod_id"), 0, 0, N,
+                  kg=pipe.kg)
     for i in range(0, N, 13):
         t = gen_flow_dict(CFG, i)["flow"]
         ip = t["flow_key"]["ip_src"]
         epc = t["metrics_peer_src"]["l3_epc_id"]
         info = pipe.kg.host.get((epc, ip))
         want = info.pod_id if info else 0
-        assert int(seg.kg[S.KG_COLS.index("pod_id"), i]) == want
+        assert int(col[i]) == want
 
 
 def test_net1s(pipe):

@@ -102,17 +102,30 @@ def test_dict_dedup(pipe):
 
 
 def test_kg_join(pipe):
+    """KG ids resolve at query time from the (epc, ip) key — no per-row
+    kg block (SmartEncoding: 96 B/span not materialized)."""
+    from deepflow_amd.query.executor import _src_np
+    from deepflow_amd.query import spec as Q
     seg = pipe.segments.segments[0]
+    pod_col = _src_np(seg, Q.SRC_KG, S.KG_COLS.index("pod_id"), 0, 0, N,
+                      kg=pipe.kg)
+    svc_col = _src_np(seg, Q.SRC_KG, S.KG_COLS.index("service_id"), 0, 0, N,
+                      kg=pipe.kg)
+    pod1_col = _src_np(seg, Q.SRC_KG, S.N_KG + S.KG_COLS.index("pod_id"),
+                       0, 0, N, kg=pipe.kg)
     for i in range(0, N, 19):
         t = _truth(i)
-        ip = t["base"]["ip_src"]
-        info = pipe.kg.lookup(t["base"]["l3_epc_id_src"], ip)
-        assert int(seg.kg[S.KG_COLS.index("pod_id"), i]) == info.pod_id
-        assert int(seg.kg[S.KG_COLS.index("service_id"), i]) == info.service_id
-        # server side
-        ip1 = t["base"]["ip_dst"]
-        info1 = pipe.kg.lookup(t["base"]["l3_epc_id_dst"], ip1)
-        assert int(seg.kg[S.N_KG + S.KG_COLS.index("pod_id"), i]) == info1.pod_id
+        info = pipe.kg.lookup(t["base"]["l3_epc_id_src"], t["base"]["ip_src"])
+        assert int(pod_col[i]) == info.pod_id
+        assert int(svc_col[i]) == info.service_id
+        info1 = pipe.kg.lookup(t["base"]["l3_epc_id_dst"], t["base"]["ip_dst"])
+        assert int(pod1_col[i]) == info1.pod_id
+    # and through the SQL surface (group by a KG tag)
+    from deepflow_amd.query.engine import QueryEngine
+    eng = QueryEngine(pipe, device="cpu")
+    r = eng.query("SELECT pod_id_1, Count(*) AS c FROM l7_flow_log "
+                  "GROUP BY pod_id_1 ORDER BY c DESC LIMIT 3")
+    assert r["values"] and r["values"][0][1] > 0
 
 
 def test_pool_strings(pipe):
