@@ -117,6 +117,22 @@ DEV uint64_t str_hash(const uint8_t* s, uint32_t len, uint64_t seed) {
     return h ? h : 1ull;  // never return EMPTY_KEY
 }
 
+// parse 16 lowercase/uppercase hex chars at pos -> u64; false on non-hex
+DEV bool hex_parse64(ByteStream& bs, uint32_t pos, uint64_t& out) {
+    uint64_t v = 0;
+    for (uint32_t i = 0; i < 16; i++) {
+        uint8_t c = bs.get(pos + i);
+        uint8_t d;
+        if (c >= '0' && c <= '9') d = c - '0';
+        else if (c >= 'a' && c <= 'f') d = c - 'a' + 10;
+        else if (c >= 'A' && c <= 'F') d = c - 'A' + 10;
+        else return false;
+        v = (v << 4) | d;
+    }
+    out = v;
+    return true;
+}
+
 DEV uint64_t mix64(uint64_t z) {
     z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
     z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
@@ -289,16 +305,36 @@ __global__ void k_decode_l7(const uint8_t* __restrict__ payload,
                     break;
                 }
                 case 13: WSTR(L7_STR_VERSION, sub, ln); break;
-                case 14: {  // TraceInfo
+                case 14: {  // TraceInfo: hex ids transcode to binary
+                    // columns (SmartEncoding: a 32-hex trace id stores as
+                    // 16 B of u64s, not 48 B of pool+len); non-hex ids
+                    // fall back to the pool columns
                     uint32_t p2 = sub;
                     while (p2 < send) {
                         uint64_t k2 = rd_varint(bs, p2, send);
                         uint32_t n2 = (uint32_t)(k2 >> 3), w2 = (uint32_t)(k2 & 7);
                         if (w2 == 2) {
                             uint32_t l3 = (uint32_t)rd_varint(bs, p2, send);
-                            if (n2 == 1) WSTR(L7_STR_TRACE_ID, p2, l3);
-                            else if (n2 == 2) WSTR(L7_STR_SPAN_ID, p2, l3);
-                            else if (n2 == 3) WSTR(L7_STR_PARENT_SPAN_ID, p2, l3);
+                            if (n2 == 1) {
+                                uint64_t hi, lo;
+                                if (l3 == 32 &&
+                                    hex_parse64(bs, p2, hi) &&
+                                    hex_parse64(bs, p2 + 16, lo)) {
+                                    W64(L7_U64_TRACE_HI, hi);
+                                    W64(L7_U64_TRACE_LO, lo);
+                                } else {
+                                    WSTR(L7_STR_TRACE_ID, p2, l3);
+                                }
+                            } else if (n2 == 2) {
+                                uint64_t sv;
+                                if (l3 == 16 && hex_parse64(bs, p2, sv)) {
+                                    W64(L7_U64_SPAN_ID_B, sv);
+                                } else {
+                                    WSTR(L7_STR_SPAN_ID, p2, l3);
+                                }
+                            } else if (n2 == 3) {
+                                WSTR(L7_STR_PARENT_SPAN_ID, p2, l3);
+                            }
                             p2 += l3;
                         } else {
                             skip_field(bs, p2, send, w2);
@@ -1062,6 +1098,9 @@ enum {
     SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
     // filter-only: exists(slot): name_id == v0 && value_id == v1
     SRC_ATTR_MATCH,
+    // binary OTel ids (idx 0 = trace_id, 1 = span_id): mix64(hi)^lo of the
+    // binary columns, or the pooled-string hash for non-hex fallbacks
+    SRC_TRACE128,
 };
 // seed for SRC_STR_HASH terms (host twin: store/dictionary.py STR_FILTER_SEED)
 #define STR_FILTER_SEED 0x5157A15E5EEDull
@@ -1130,6 +1169,16 @@ DEV uint64_t kg_join(const SegView& s, uint64_t row, uint16_t idx) {
     return 0;
 }
 
+DEV uint64_t pool_str_hash(const SegView& s, uint64_t row, uint16_t idx) {
+    uint64_t rr = s.str_rowref[row];
+    uint64_t off = STR_REF_OFF(rr);
+    for (uint16_t c = 0; c < (uint16_t)idx; c++)
+        off += (uint16_t)s.str_lens[(uint64_t)c * s.stride + row];
+    uint32_t len = (uint16_t)s.str_lens[(uint64_t)idx * s.stride + row];
+    if (len == 0) return 0;
+    return str_hash(s.pool + off, len, STR_FILTER_SEED);
+}
+
 DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
                        uint16_t idx, uint32_t bucket, uint64_t time_base_s) {
     switch (family) {
@@ -1150,14 +1199,18 @@ DEV uint64_t src_value(const SegView& s, uint64_t row, uint8_t family,
             uint64_t rel = t_s > time_base_s ? t_s - time_base_s : 0;
             return bucket ? (rel / bucket) * bucket : rel;
         }
-        case SRC_STR_HASH: {
-            uint64_t rr = s.str_rowref[row];
-            uint64_t off = STR_REF_OFF(rr);
-            for (uint16_t c = 0; c < (uint16_t)idx; c++)
-                off += (uint16_t)s.str_lens[(uint64_t)c * s.stride + row];
-            uint32_t len = (uint16_t)s.str_lens[(uint64_t)idx * s.stride + row];
-            if (len == 0) return 0;
-            return str_hash(s.pool + off, len, STR_FILTER_SEED);
+        case SRC_STR_HASH:
+            return pool_str_hash(s, row, idx);
+        case SRC_TRACE128: {
+            if (idx == 0) {
+                uint64_t hi = s.u64c[L7_U64_TRACE_HI * s.stride + row];
+                uint64_t lo = s.u64c[L7_U64_TRACE_LO * s.stride + row];
+                if (hi | lo) return mix64(hi) ^ lo;
+                return pool_str_hash(s, row, L7_POOL_TRACE_ID);
+            }
+            uint64_t sv = s.u64c[L7_U64_SPAN_ID_B * s.stride + row];
+            if (sv) return sv;
+            return pool_str_hash(s, row, L7_POOL_SPAN_ID);
         }
         default: return 0;
     }

@@ -22,6 +22,12 @@ enum {
     L7_U64_RRT,             // us (head.rrt)
     L7_U64_SYSCALL_REQ,
     L7_U64_SYSCALL_RESP,
+    // OTel ids transcoded to binary at decode (hex strings on the wire):
+    // 16-byte trace id as two words, 8-byte span id. Zero = absent or
+    // non-hex (then the raw string falls back to the pool columns).
+    L7_U64_TRACE_HI,
+    L7_U64_TRACE_LO,
+    L7_U64_SPAN_ID_B,
     L7_U64_N
 };
 
@@ -135,6 +141,12 @@ enum {
     KG_GPROCESS_ID,
     KG_VALS_N
 };
+
+// pool positions of the trace-id fallback columns (the index of
+// trace_id/span_id within the pooled subset of the string columns —
+// python twin l7_schema.POOL_POS, sync-tested)
+#define L7_POOL_TRACE_ID 2
+#define L7_POOL_SPAN_ID 3
 
 #define DICT_ID_INVALID 0xFFFFFFFFu
 #define STR_REF_PACK(off, len) \

@@ -19,6 +19,7 @@ from ..store.dictionary import str_hash_py, domain_seed
 from ..wire.pb import read_varint
 
 M64 = (1 << 64) - 1
+_HEXSET = frozenset(b"0123456789abcdefABCDEF")
 
 
 def mix64(z: int) -> int:
@@ -174,6 +175,26 @@ def decode_l7_ref(payload: bytes, offs, lens, seg, base_row: int,
                                     sattr[S.MAX_ATTRS + n_vals, rid] = \
                                         S.str_ref_pack(p2, l3)
                                 n_vals += 1
+                            elif num == 14 and n2 in (1, 2):
+                                # hex trace/span ids -> binary u64 cols
+                                # (pool fallback for non-hex forms)
+                                raw = bytes(mv[p2:p2 + l3])
+                                hexok = all(c in _HEXSET for c in raw)
+                                done = False
+                                if hexok and n2 == 1 and l3 == 32:
+                                    v = int(raw, 16)
+                                    _w(seg, "u64", "trace_id_hi", row,
+                                       v >> 64)
+                                    _w(seg, "u64", "trace_id_lo", row,
+                                       v & M64)
+                                    done = True
+                                elif hexok and n2 == 2 and l3 == 16:
+                                    _w(seg, "u64", "span_id_b", row,
+                                       int(raw, 16))
+                                    done = True
+                                if not done:
+                                    sstr[_STR_IDX[strmap[n2]], rid] = \
+                                        S.str_ref_pack(p2, l3)
                             elif n2 in strmap:
                                 sstr[_STR_IDX[strmap[n2]], rid] = \
                                     S.str_ref_pack(p2, l3)

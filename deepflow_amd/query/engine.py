@@ -645,8 +645,25 @@ class QueryEngine:
                     return self.pipe.dict.hydrate(DICT_DOM_ATTR_VALUE,
                                                   [vid])[0]
             return None
+        if col in ("trace_id", "span_id") and hasattr(seg, "attr_pool"):
+            # binary-transcoded ids: reconstruct the hex form (pool holds
+            # only the non-hex fallbacks)
+            hi_i = S.U64_COLS.index("trace_id_hi")
+            lo_i = S.U64_COLS.index("trace_id_lo")
+            sp_i = S.U64_COLS.index("span_id_b")
+            M = (1 << 64) - 1
+            if col == "trace_id":
+                hi, lo = int(seg.u64[hi_i, row]) & M, \
+                    int(seg.u64[lo_i, row]) & M
+                if hi | lo:
+                    return f"{hi:016x}{lo:016x}"
+            else:
+                sv = int(seg.u64[sp_i, row]) & M
+                if sv:
+                    return f"{sv:016x}"
+            # fall through to the pooled fallback string
         td = tags.get(col)
-        if td is not None:
+        if td is not None and td.family != Q.SRC_TRACE128:
             fam, idx = td.family, td.idx
             if fam == Q.SRC_U64:
                 return int(seg.u64[idx, row])

@@ -15,7 +15,7 @@ import torch
 
 from .spec import (Plan, SRC_U64, SRC_U32, SRC_U8, SRC_DID, SRC_KG,
                    SRC_ATTR_VAL, SRC_TIME_BUCKET, SRC_CONST0, SRC_STR_HASH,
-                   SRC_ATTR_MATCH, STR_FILTER_SEED,
+                   SRC_ATTR_MATCH, SRC_TRACE128, STR_FILTER_SEED,
                    OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN,
                    AGGOP_COUNT, AGGOP_SUM, AGGOP_MIN, AGGOP_MAX,
                    QMAX_KEYS, QMAX_AGGS)
@@ -66,6 +66,23 @@ def _src_np(seg, family: int, idx: int, bucket: int, time_base_s: int,
             if idx < cnts[i]:
                 out[i] = np.uint32(pool[starts[i] + cnts[i] + idx])
         return out
+    if family == SRC_TRACE128:
+        from ..ops.ref import mix64
+        from ..store import l7_schema as S7
+        if idx == 0:
+            hi = seg.u64[S7.U64_COLS.index("trace_id_hi"), :n].numpy().view(np.uint64)
+            lo = seg.u64[S7.U64_COLS.index("trace_id_lo"), :n].numpy().view(np.uint64)
+            fb = _src_np(seg, SRC_STR_HASH, S7.POOL_POS["trace_id"], 0,
+                         time_base_s, n)
+            out = np.empty(n, dtype=np.uint64)
+            for i in range(n):
+                out[i] = (mix64(int(hi[i])) ^ int(lo[i])) \
+                    if (int(hi[i]) | int(lo[i])) else int(fb[i])
+            return out
+        sv = seg.u64[S7.U64_COLS.index("span_id_b"), :n].numpy().view(np.uint64)
+        fb = _src_np(seg, SRC_STR_HASH, S7.POOL_POS["span_id"], 0,
+                     time_base_s, n)
+        return np.where(sv != 0, sv, fb)
     if family == SRC_TIME_BUCKET:
         t_s = seg.u64[0, :n].numpy().view(np.uint64) // np.uint64(10**9)
         rel = np.maximum(t_s.astype(np.int64) - time_base_s, 0).astype(np.uint64)

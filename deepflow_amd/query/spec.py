@@ -24,6 +24,9 @@ SRC_TIME_BUCKET = 6
 SRC_CONST0 = 7
 SRC_STR_HASH = 8
 SRC_ATTR_MATCH = 9  # filter-only: attr name_id == v0 AND value_id == v1
+# binary-transcoded OTel ids: idx 0 = trace_id (mix64(hi)^lo), 1 = span_id;
+# zero binary cols fall back to the pooled-string hash (non-hex ids)
+SRC_TRACE128 = 10
 
 # seed for pooled-string filter hashing (twin: dfgpu.hip STR_FILTER_SEED)
 STR_FILTER_SEED = 0x5157A15E5EED

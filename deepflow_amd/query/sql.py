@@ -395,6 +395,25 @@ def _add_term(plan: Q.Plan, name: str, op: str, lit, dictionary,
         plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
                                  group=group))
         return
+    if td.hydrate == "tracebin":
+        # compile the literal exactly like the stored form: hex ids hash
+        # as mix64(hi)^lo of the binary value, other strings as the
+        # pooled-string hash (executor SRC_TRACE128 twin)
+        from ..ops.ref import mix64
+        from ..store.dictionary import str_hash_py
+        lit_s = lit[1]
+        hexok = len(lit_s) in (32, 16) and \
+            all(c in "0123456789abcdefABCDEF" for c in lit_s)
+        if hexok and len(lit_s) == 32 and td.idx == 0:
+            v128 = int(lit_s, 16)
+            v = mix64(v128 >> 64) ^ (v128 & ((1 << 64) - 1))
+        elif hexok and len(lit_s) == 16 and td.idx == 1:
+            v = int(lit_s, 16)
+        else:
+            v = str_hash_py(lit_s.encode(), Q.STR_FILTER_SEED)
+        plan.terms.append(Q.Term(td.family, td.idx, Q.OP_BY_NAME[op], v,
+                                 group=group))
+        return
     if td.hydrate == "ip6str":
         # stored value is the packed 16-byte address
         import ipaddress

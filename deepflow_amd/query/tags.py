@@ -83,10 +83,13 @@ def build_l7_tags() -> Dict[str, TagDef]:
     tags["app_service"] = tags["service_name"]
     # pooled string tags: filterable via GPU string-hash compare,
     # selectable from the segment pool
-    for sname in ["trace_id", "span_id", "parent_span_id", "x_request_id_0",
+    for sname in ["parent_span_id", "x_request_id_0",
                   "x_request_id_1", "http_user_agent", "biz_code"]:
         add(TagDef(sname, Q.SRC_STR_HASH, S.POOL_POS[sname],
                    hydrate="strhash"))
+    # binary-transcoded OTel ids: u64 columns when hex, pool fallback
+    add(TagDef("trace_id", Q.SRC_TRACE128, 0, hydrate="tracebin"))
+    add(TagDef("span_id", Q.SRC_TRACE128, 1, hydrate="tracebin"))
     # pooled 16-byte v6 addresses: equality filter hashes the packed
     # address bytes; select formats them back to text
     for sname in ["ip6_0", "ip6_1"]:

@@ -29,6 +29,25 @@ def register_migration(from_version: int):
     return deco
 
 
+@register_migration(4)
+def _v4_to_v5(payload: Dict) -> Dict:
+    """v5 added three binary trace/span id u64 columns: old rows keep the
+    ids in the string pool (the fallback form), so the new columns pad
+    with zeros (= 'use the pool fallback')."""
+    for st in payload.get("segments", []):
+        u = st["u64"]
+        z = torch.zeros((3, u.shape[1]), dtype=u.dtype)
+        st["u64"] = torch.cat([u, z], 0)
+    for st in payload.get("cold", []):
+        n = st["n_rows"]
+        for _ in range(3):
+            st["u64_cols"].append({"base": 0, "bits": 0, "data": None,
+                                   "raw": None, "n": n})
+        st["layout_version"] = 5
+    payload["layout_version"] = 5
+    return payload
+
+
 @register_migration(3)
 def _v3_to_v4(payload: Dict) -> Dict:
     """v4 dropped the per-row KG block (query-time join): discard the

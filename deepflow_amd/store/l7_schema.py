@@ -6,12 +6,13 @@ these stay in sync."""
 # shape or packing changes; segments record it so mixed-layout windows are
 # rejected instead of misread (migration = drain + reingest, matching the
 # reference's at-most-once durability posture).
-LAYOUT_VERSION = 4  # v4 = KG joined at query time (no per-row kg block);
+LAYOUT_VERSION = 5  # v5 = binary trace/span id columns;
+                    # v4 = KG joined at query time (no per-row kg block);
                     # v3 = pooled ip6 columns; v2 = packed attr pool
 
 U64_COLS = [
     "start_time", "end_time", "flow_id", "rrt", "syscall_trace_id_request",
-    "syscall_trace_id_response",
+    "syscall_trace_id_response", "trace_id_hi", "trace_id_lo", "span_id_b",
 ]
 
 U32_COLS = [
