@@ -76,6 +76,40 @@ class KnowledgeGraphTable:
         return self.host.get((epc, ip), KgInfo())
 
 
+def kg_display_name(col: str, ident: int) -> str:
+    """Synthetic tagrecorder inventory: the display name a resource id
+    hydrates to (k8s-realistic shapes; deterministic). This is the string
+    a naive ClickHouse schema would store PER ROW for each universal tag
+    — the SmartEncoding baseline measures the bytes of exactly these
+    names (reference: tagrecorder ch_* maps + dictGet hydration)."""
+    h = (ident * 0x9E3779B9) & 0xFFFFFFFFFF
+    if col == "pod_id":
+        return f"svc-{ident % 997:03d}-{h:010x}-{ident % 99999:05d}"
+    if col == "pod_node_id":
+        return f"node-us-east-{ident:04d}.prod.internal"
+    if col == "pod_ns_id":
+        return f"ns-team-{ident:03d}"
+    if col == "pod_group_id":
+        return f"svc-{ident % 997:03d}-deployment"
+    if col == "pod_cluster_id":
+        return f"prod-cluster-{ident:02d}"
+    if col == "l3_device_type":
+        return "pod" if ident == 14 else f"devtype-{ident}"
+    if col == "l3_device_id":
+        return f"vm-{h:010x}"
+    if col == "subnet_id":
+        return f"subnet-10-{ident:03d}-0-0"
+    if col == "host_id":
+        return f"host-{ident:04d}.dc1.example.com"
+    if col == "az_id":
+        return f"az-east-{ident}"
+    if col == "service_id":
+        return f"svc-{ident % 997:03d}.default.svc"
+    if col == "gprocess_id":
+        return f"proc-{h:08x}-{ident:05d}"
+    return f"{col}-{ident}"
+
+
 def default_platform(cfg) -> Dict[Tuple[int, int], KgInfo]:
     """Synthetic platform inventory matching gen.spans.SpanGenConfig: every
     (epc, ip) the generator can emit gets pod/node/service ids (the
