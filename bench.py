@@ -222,9 +222,16 @@ def main() -> None:
     pipe.segments.reserve(total_rows // (1 << 23) + 2)
 
     dict_sync = None
+    router = None
     if world > 1:
         from deepflow_amd.parallel.dict_sync import DictSync
+        from deepflow_amd.parallel.span_router import SpanRouter
         dict_sync = DictSync(pipe.dict)
+        # data-plane all-to-all: each rank's generated stream is routed to
+        # the shard owning each record's agent before local ingest
+        # (BASELINE #3: RCCL all-to-all over xGMI; the routing + exchange
+        # is INSIDE the timed step)
+        router = SpanRouter(device=device)
 
     # H2D prefetch pipeline: copy batch i+1 on a side stream while batch i's
     # kernels run on the main stream.
@@ -250,9 +257,20 @@ def main() -> None:
             for t in dev_batch:
                 t.record_stream(torch.cuda.current_stream())
             prefetch(i + 1)
-            pipe.ingest_device(*dev_batch, host_payload)
+            if router is not None:
+                buf, offs_h, lens_h = batches[i % n_distinct][:3]
+                pay, offs_t, lens_t = router.route_gpu(
+                    dev_batch[0], dev_batch[1], dev_batch[2],
+                    buf, offs_h, lens_h)
+                # harvest reads only the emitted slices from the device
+                # payload (no full-batch D2H on the routed path)
+                pipe.ingest_device(pay, offs_t, lens_t, pay)
+            else:
+                pipe.ingest_device(*dev_batch, host_payload)
         else:
             payload, offs, lens = batches[i % n_distinct][:3]
+            if router is not None:
+                payload, offs, lens = router.route_cpu(payload, offs, lens)
             pipe.ingest(payload, offs, lens)
         if dict_sync is not None:
             dict_sync.sync_step()

@@ -310,6 +310,112 @@ uint64_t df_scan_offsets(const uint8_t* payload, uint64_t len,
     return n;
 }
 
+// Shard routing peek: owner shard of each AppProtoLogsData record from its
+// base.vtap_id (field 1 -> 5) without a full decode — the data-plane
+// all-to-all key (reference hashes frames to queues by agent,
+// server/libs/receiver/receiver.go:519-566). OpenMP over records.
+static inline uint64_t route_mix(uint64_t z) {
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    return z ^ (z >> 31);
+}
+
+uint64_t df_route_spans(const uint8_t* payload, uint64_t total,
+                        const uint32_t* offs, const uint32_t* lens,
+                        uint64_t n, uint32_t world, uint32_t org_id,
+                        uint8_t* out_shard) {
+    (void)total;
+    uint64_t errs = 0;
+#pragma omp parallel for schedule(static) reduction(+ : errs)
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t pos = offs[i], end = offs[i] + lens[i];
+        uint32_t vtap = 0;
+        bool found = false;
+        // top level: find field 1 (base)
+        while (pos < end && !found) {
+            uint64_t key = 0;
+            int sh = 0;
+            while (pos < end) {
+                uint8_t b = payload[pos++];
+                key |= (uint64_t)(b & 0x7F) << sh;
+                if (!(b & 0x80)) break;
+                sh += 7;
+            }
+            uint32_t num = (uint32_t)(key >> 3), wt = (uint32_t)(key & 7);
+            if (wt == 2) {
+                uint64_t ln = 0;
+                sh = 0;
+                while (pos < end) {
+                    uint8_t b = payload[pos++];
+                    ln |= (uint64_t)(b & 0x7F) << sh;
+                    if (!(b & 0x80)) break;
+                    sh += 7;
+                }
+                if (num == 1) {
+                    uint32_t p2 = pos, e2 = pos + (uint32_t)ln;
+                    while (p2 < e2) {
+                        uint64_t k2 = 0;
+                        sh = 0;
+                        while (p2 < e2) {
+                            uint8_t b = payload[p2++];
+                            k2 |= (uint64_t)(b & 0x7F) << sh;
+                            if (!(b & 0x80)) break;
+                            sh += 7;
+                        }
+                        uint32_t n2 = (uint32_t)(k2 >> 3),
+                                 w2 = (uint32_t)(k2 & 7);
+                        if (w2 == 0) {
+                            uint64_t v = 0;
+                            sh = 0;
+                            while (p2 < e2) {
+                                uint8_t b = payload[p2++];
+                                v |= (uint64_t)(b & 0x7F) << sh;
+                                if (!(b & 0x80)) break;
+                                sh += 7;
+                            }
+                            if (n2 == 5) { vtap = (uint32_t)v; found = true; break; }
+                        } else if (w2 == 2) {
+                            uint64_t l2 = 0;
+                            sh = 0;
+                            while (p2 < e2) {
+                                uint8_t b = payload[p2++];
+                                l2 |= (uint64_t)(b & 0x7F) << sh;
+                                if (!(b & 0x80)) break;
+                                sh += 7;
+                            }
+                            p2 += (uint32_t)l2;
+                        } else if (w2 == 1) {
+                            p2 += 8;
+                        } else {
+                            p2 += 4;
+                        }
+                    }
+                    break;
+                }
+                pos += (uint32_t)ln;
+            } else if (wt == 0) {
+                uint64_t v = 0;
+                sh = 0;
+                while (pos < end) {
+                    uint8_t b = payload[pos++];
+                    v |= (uint64_t)(b & 0x7F) << sh;
+                    if (!(b & 0x80)) break;
+                    sh += 7;
+                }
+                (void)v;
+            } else if (wt == 1) {
+                pos += 8;
+            } else {
+                pos += 4;
+            }
+        }
+        if (!found) errs++;
+        out_shard[i] =
+            (uint8_t)(route_mix(((uint64_t)org_id << 32) | vtap) % world);
+    }
+    return errs;
+}
+
 // ---- zstd / lz4 codecs via dlopen (runtime libs are present without dev
 // headers; the trident frame encoder byte 3 = whole-payload zstd,
 // agent/src/trident.rs:416-431) ----

@@ -799,6 +799,29 @@ __global__ void k_rollup_insert(const uint64_t* __restrict__ kws,  // [n, nw]
 }
 
 // ----------------------------------------------------------------------
+// Data-plane shard routing: gather selected records into a packed
+// per-destination buffer for the RCCL all-to-all (one wave per record,
+// lanes copy bytes strided — records are 100-500 B, so a wave64 gets
+// coalesced 64 B segments).
+// ----------------------------------------------------------------------
+__global__ void k_gather_records(const uint8_t* __restrict__ src,
+                                 const uint32_t* __restrict__ offs,
+                                 const uint32_t* __restrict__ lens,
+                                 const uint32_t* __restrict__ sel,
+                                 const uint64_t* __restrict__ dst_off,
+                                 uint32_t m, uint8_t* __restrict__ out) {
+    uint32_t waves = blockDim.x / 64;
+    uint32_t rec = blockIdx.x * waves + (threadIdx.x / 64);
+    uint32_t lane = threadIdx.x % 64;
+    if (rec >= m) return;
+    uint32_t r = sel[rec];
+    const uint8_t* s = src + offs[r];
+    uint8_t* d = out + dst_off[rec];
+    uint32_t n = lens[r];
+    for (uint32_t b = lane; b < n; b += 64) d[b] = s[b];
+}
+
+// ----------------------------------------------------------------------
 // K2: KnowledgeGraph (epc,ip) -> resource-id join.
 //     Open-addressing table: keys u64 ((epc<<32)|ip), vals KG_VALS_N x u32.
 // ----------------------------------------------------------------------
@@ -1526,6 +1549,19 @@ int df_rollup_insert(const void* kws, const void* vals, const void* ops,
                        (const uint8_t*)ops, n, nw, nv, (uint64_t*)tkeys,
                        (uint64_t*)traw, (unsigned long long*)tvals, cap - 1,
                        (unsigned long long*)drops);
+    return (int)hipGetLastError();
+}
+
+int df_gather_records(const void* src, const void* offs, const void* lens,
+                      const void* sel, const void* dst_off, uint32_t m,
+                      void* out, uint64_t stream) {
+    uint32_t waves = BLOCK / 64;
+    uint32_t blocks = (m + waves - 1) / waves;
+    hipLaunchKernelGGL(k_gather_records, dim3(blocks), dim3(BLOCK), 0,
+                       STREAM(stream), (const uint8_t*)src,
+                       (const uint32_t*)offs, (const uint32_t*)lens,
+                       (const uint32_t*)sel, (const uint64_t*)dst_off, m,
+                       (uint8_t*)out);
     return (int)hipGetLastError();
 }
 

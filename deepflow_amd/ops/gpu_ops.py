@@ -66,6 +66,17 @@ def rollup_insert(kws: torch.Tensor, vals: torch.Tensor, ops: torch.Tensor,
         stream or _stream()), "df_rollup_insert")
 
 
+def gather_records(payload: torch.Tensor, offs: torch.Tensor,
+                   lens: torch.Tensor, sel: torch.Tensor,
+                   dst_off: torch.Tensor, out: torch.Tensor) -> None:
+    """Pack selected records contiguously (shard-routing all-to-all)."""
+    lib = native.gpu()
+    native.check(lib.df_gather_records(
+        payload.data_ptr(), offs.data_ptr(), lens.data_ptr(),
+        sel.data_ptr(), dst_off.data_ptr(), sel.numel(), out.data_ptr(),
+        _stream()), "df_gather_records")
+
+
 def kg_build(keys: torch.Tensor, vals: torch.Tensor, tkeys: torch.Tensor,
              tvals: torch.Tensor) -> None:
     lib = native.gpu()

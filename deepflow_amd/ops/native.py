@@ -67,6 +67,10 @@ def cpu() -> ct.CDLL:
         lib.df_otlp_to_l7.restype = ct.c_int64
         lib.df_otlp_to_l7.argtypes = [ct.c_void_p, ct.c_uint64,
                                       ct.c_void_p, ct.c_uint64]
+        lib.df_route_spans.restype = ct.c_uint64
+        lib.df_route_spans.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p,
+                                       ct.c_void_p, ct.c_uint64, ct.c_uint32,
+                                       ct.c_uint32, ct.c_void_p]
         lib.df_scan_offsets.restype = ct.c_uint64
         lib.df_scan_offsets.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p,
                                         ct.c_void_p, ct.c_uint64]
@@ -96,6 +100,8 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_rollup_insert.restype = ct.c_int
     lib.df_rollup_insert.argtypes = [p, p, p, u32, u32, u32, p, p, p, u32,
                                      p, u64]
+    lib.df_gather_records.restype = ct.c_int
+    lib.df_gather_records.argtypes = [p, p, p, p, p, u32, p, u64]
     lib.df_kg_build.restype = ct.c_int
     lib.df_kg_build.argtypes = [p, p, u32, p, p, u32, u64]
     lib.df_intern_many.restype = ct.c_int

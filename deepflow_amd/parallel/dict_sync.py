@@ -42,14 +42,23 @@ def _unpack(data: bytes) -> List[Tuple[int, int, bytes]]:
 
 
 class DictSync:
-    def __init__(self, dictionary, device: str | None = None):
+    def __init__(self, dictionary, device: str | None = None,
+                 max_remote_entries: int = 1 << 18):
         self.dict = dictionary
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
         self.device = device or ("cuda" if dist.get_backend() == "nccl" else "cpu")
-        # remote hydration maps: rank -> {(dom, id): bytes}
-        self.remote: Dict[int, Dict[Tuple[int, int], bytes]] = {
-            r: {} for r in range(self.world)}
+        # remote hydration maps: rank -> {(dom, id): bytes}. BOUNDED:
+        # with data-plane span routing each row lives on the shard that
+        # interned its strings, so remote hydration is a query-merge
+        # nicety, not a full mirror — FIFO-evict past the cap instead of
+        # duplicating 8 shards' dictionaries in host memory (round-1
+        # unbounded-growth defect).
+        from collections import OrderedDict
+        self.max_remote_entries = max_remote_entries
+        self.remote: Dict[int, "OrderedDict[Tuple[int, int], bytes]"] = {
+            r: OrderedDict() for r in range(self.world)}
+        self.remote_evicted = 0
         self.bytes_exchanged = 0
 
     def sync_step(self) -> int:
@@ -82,6 +91,9 @@ class DictSync:
             table = self.remote[r]
             for dom, ident, s in entries:
                 table[(dom, ident)] = s
+            while len(table) > self.max_remote_entries:
+                table.popitem(last=False)
+                self.remote_evicted += 1
             merged += len(entries)
             self.bytes_exchanged += size
         return merged

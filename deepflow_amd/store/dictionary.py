@@ -72,21 +72,27 @@ class TagDictionary:
     def n_entries(self) -> int:
         return len(self.id_to_str)
 
-    def harvest(self, payload_host: np.ndarray) -> int:
-        """Drain the emit buffer after a batch; payload_host is the batch's
-        raw payload bytes on the host (same bytes the kernels saw on GPU)."""
+    def harvest(self, payload) -> int:
+        """Drain the emit buffer after a batch. `payload` is the batch's
+        raw bytes — host numpy, or a DEVICE uint8 tensor (routed batches
+        have no host copy: only the emitted slices transfer, a few bytes
+        per NEW dictionary entry instead of a full-batch D2H)."""
         cnt = int(self.emit_ctr[0].item())
         if cnt == 0:
             return 0
         take = min(cnt, EMIT_CAP)
         rows = self.emit[:take].cpu().numpy()
+        is_dev = hasattr(payload, "device") and payload.device.type != "cpu"
         for tag, ref in rows:
             tag = int(tag) & 0xFFFFFFFFFFFFFFFF
             ref = int(ref) & 0xFFFFFFFFFFFFFFFF
             dom = tag >> 56
             slot = tag & 0xFFFFFFFF
             off, ln = ref >> 16, ref & 0xFFFF
-            sbytes = payload_host[off:off + ln].tobytes()
+            if is_dev:
+                sbytes = bytes(payload[off:off + ln].cpu().numpy())
+            else:
+                sbytes = payload[off:off + ln].tobytes()
             self.id_to_str[(dom, slot)] = sbytes
             self.str_to_id[(dom, sbytes)] = slot
             self.pending_sync.append((dom, slot, sbytes))
