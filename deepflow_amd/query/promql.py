@@ -100,18 +100,99 @@ def _parse_matchers(s: Optional[str]) -> List[Tuple[str, str, str]]:
 class PromQLEngine:
     DEFAULT_LOOKBACK = 300
 
-    def __init__(self, app_rows_fn, net_rows_fn=None, raw_sources=None):
+    # flow_log_* metrics execute on the GPU store through the DF-SQL
+    # engine (reference: PromQL -> SQL conversion, app/prometheus/service/
+    # converters.go:1701; BASELINE config #4 — the group-by runs
+    # k_query_agg, not a host row scan)
+    FLOW_LOG_METRICS = {
+        "count": "Count(*)",
+        "duration_sum": "Sum(response_duration)",
+        "duration_max": "Max(response_duration)",
+        "request_length_sum": "Sum(request_length)",
+        "response_length_sum": "Sum(response_length)",
+    }
+    # le bounds for flow_log_duration_bucket, seconds (prom convention)
+    LE_BOUNDS_S = [0.001, 0.005, 0.01, 0.025, 0.05, 0.1, 0.2, 0.3, 0.5,
+                   1.0, 5.0]
+
+    def __init__(self, app_rows_fn, net_rows_fn=None, raw_sources=None,
+                 sql_engine=None):
         """rows_fns return the 1s rollup dict-rows; raw_sources are
         callables (metric_name, matchers) -> series list (e.g. the
-        prometheus remote-write store)."""
+        prometheus remote-write store); sql_engine routes flow_log_*
+        metrics through the GPU query engine."""
         self.sources = {"application": app_rows_fn}
         if net_rows_fn is not None:
             self.sources["network"] = net_rows_fn
         self.raw_sources = list(raw_sources or [])
+        self.sql_engine = sql_engine
+        self._by_hint: List[str] = []
+
+    # ----------------------------------------------- flow_log SQL offload
+    def _flow_log_where(self, matchers) -> str:
+        conds = []
+        for lname, op, lval in matchers:
+            if lname == "le":
+                continue
+            if op == "=~":
+                raise PromError("regex matchers are not supported for "
+                                "flow_log_* metrics")
+            sqlop = "=" if op == "=" else "!="
+            lit = lval if lval.lstrip("-").isdigit() else f"'{lval}'"
+            conds.append(f"{lname} {sqlop} {lit}")
+        return (" WHERE " + " AND ".join(conds)) if conds else ""
+
+    def _sql_series(self, agg_expr: str, matchers, name: str,
+                    extra_where: str = "",
+                    extra_labels: Optional[Dict] = None) -> List[Dict]:
+        by = [b for b in self._by_hint if b != "le"]
+        cols = ", ".join(["time(1) AS time"] + by)
+        where = self._flow_log_where(matchers)
+        if extra_where:
+            where = (where + " AND " if where else " WHERE ") + extra_where
+        group = ", ".join(["time(1)"] + by)
+        sql = (f"SELECT {cols}, {agg_expr} AS v FROM l7_flow_log"
+               f"{where} GROUP BY {group}")
+        res = self.sql_engine.query(sql)
+        cols_out = res["columns"]
+        vi = cols_out.index("v")
+        ti = cols_out.index("time")
+        series: Dict[tuple, Dict] = {}
+        for row in res["values"]:
+            labels = {cols_out[i]: str(row[i])
+                      for i in range(len(cols_out)) if i not in (vi, ti)}
+            if extra_labels:
+                labels.update(extra_labels)
+            key = tuple(sorted(labels.items()))
+            s = series.setdefault(key, {"metric": dict(labels,
+                                                       __name__=name),
+                                        "samples": {}})
+            t = int(row[ti])
+            s["samples"][t] = s["samples"].get(t, 0) + (row[vi] or 0)
+        return list(series.values())
+
+    def _series_flow_log(self, name: str, matchers) -> List[Dict]:
+        field = name[len("flow_log_"):]
+        if field == "duration_bucket":
+            out = []
+            for le in self.LE_BOUNDS_S:
+                out.extend(self._sql_series(
+                    "Count(*)", matchers, name,
+                    extra_where=f"response_duration <= {int(le * 1e6)}",
+                    extra_labels={"le": repr(le)}))
+            out.extend(self._sql_series(
+                "Count(*)", matchers, name, extra_labels={"le": "+Inf"}))
+            return out
+        agg = self.FLOW_LOG_METRICS.get(field)
+        if agg is None:
+            raise PromError(f"unknown flow_log metric {field!r}")
+        return self._sql_series(agg, matchers, name)
 
     # -------------------------------------------------------- series fetch
     def _series(self, name: str,
                 matchers: List[Tuple[str, str, str]]) -> List[Dict]:
+        if name.startswith("flow_log_") and self.sql_engine is not None:
+            return self._series_flow_log(name, matchers)
         for prefix, fn in self.sources.items():
             if name.startswith(prefix + "_"):
                 field = name[len(prefix) + 1:]
@@ -265,8 +346,15 @@ class PromQLEngine:
         agg = _match_agg(expr)
         if agg:
             func, by, inner_expr = agg
-            inner = self._eval_at(inner_expr, t)
             by_labels = [x.strip() for x in by.split(",")] if by else []
+            # push the by-labels down so flow_log_* selectors GROUP BY
+            # exactly these dimensions on the GPU (converters.go analog)
+            saved_hint = self._by_hint
+            self._by_hint = by_labels
+            try:
+                inner = self._eval_at(inner_expr, t)
+            finally:
+                self._by_hint = saved_hint
             groups: Dict[tuple, List[float]] = {}
             metas: Dict[tuple, Dict] = {}
             for s in inner:

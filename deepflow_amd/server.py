@@ -65,7 +65,8 @@ class DeepflowServer:
                                self.prom.ingest_write_request(
                                    payload.tobytes()))
         self.promql = PromQLEngine(self.l7.metrics.rows, self.l4.metrics.rows,
-                                   raw_sources=[self.prom.series_for])
+                                   raw_sources=[self.prom.series_for],
+                                   sql_engine=self.engine)
         from .ingest.profile_pipeline import ProfilePipeline, ProfileApp
         self.profiles = ProfilePipeline()
         self.receiver.register(framing.MSG_PROFILE,
