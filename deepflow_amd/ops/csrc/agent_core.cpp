@@ -11,9 +11,11 @@
 // Document payload records, length-prefixed for trident framing).
 // Packet sources: callers feed raw frames (tests use synthetic packets;
 // AF_PACKET capture wiring is the host deployment's concern).
+#include <cerrno>
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
+#include <sys/socket.h>
 #include <map>
 #include <string>
 #include <unordered_map>
@@ -2590,6 +2592,60 @@ int64_t dfa_syscall_batch(void* h, const uint8_t* buf, uint64_t total) {
         n++;
     }
     return n;
+}
+
+// TPACKET_V3 block walker: one native call drains a whole kernel ring
+// block into the flow engine (reference: af_packet/tpacket.rs mmap ring;
+// round-1 used per-packet recvfrom — this is the ≥1 Mpps/core path).
+// Layout offsets are the stable kernel ABI (linux/if_packet.h):
+//   tpacket_block_desc: block_status @8, num_pkts @12, first_pkt @16
+//   tpacket3_hdr: next_off @0, sec @4, nsec @8, snaplen @12, mac @24
+int64_t dfa_ring_block(void* h, const uint8_t* block) {
+    uint32_t num = 0, first = 0;
+    memcpy(&num, block + 12, 4);
+    memcpy(&first, block + 16, 4);
+    const uint8_t* p = block + first;
+    for (uint32_t i = 0; i < num; i++) {
+        uint32_t next = 0, sec = 0, nsec = 0, snap = 0;
+        uint16_t mac = 0;
+        memcpy(&next, p, 4);
+        memcpy(&sec, p + 4, 4);
+        memcpy(&nsec, p + 8, 4);
+        memcpy(&snap, p + 12, 4);
+        memcpy(&mac, p + 24, 2);
+        dfa_packet(h, p + mac, snap,
+                   (uint64_t)sec * 1000000000ull + nsec);
+        if (next == 0) break;
+        p += next;
+    }
+    return num;
+}
+
+// loopback blaster for capture benchmarks: sendmmsg batches of the given
+// frame on a connected AF_PACKET fd; returns frames sent (or -errno)
+int64_t dfa_blast(int fd, const uint8_t* frame, uint32_t len,
+                  uint64_t count) {
+    constexpr int B = 64;
+    struct mmsghdr msgs[B];
+    struct iovec iovs[B];
+    memset(msgs, 0, sizeof msgs);
+    for (int i = 0; i < B; i++) {
+        iovs[i].iov_base = (void*)frame;
+        iovs[i].iov_len = len;
+        msgs[i].msg_hdr.msg_iov = &iovs[i];
+        msgs[i].msg_hdr.msg_iovlen = 1;
+    }
+    uint64_t sent = 0;
+    while (sent < count) {
+        int want = (int)((count - sent) < B ? (count - sent) : B);
+        int r = sendmmsg(fd, msgs, want, 0);
+        if (r < 0) {
+            if (errno == EINTR || errno == EAGAIN) continue;
+            return -(int64_t)errno;
+        }
+        sent += r;
+    }
+    return (int64_t)sent;
 }
 
 // Periodic tick: emit+drop closed/idle flows, roll meters into Documents.
