@@ -82,7 +82,9 @@ class TagDictionary:
             return 0
         take = min(cnt, EMIT_CAP)
         rows = self.emit[:take].cpu().numpy()
-        is_dev = hasattr(payload, "device") and payload.device.type != "cpu"
+        import torch
+        is_dev = isinstance(payload, torch.Tensor) and \
+            payload.device.type != "cpu"
         for tag, ref in rows:
             tag = int(tag) & 0xFFFFFFFFFFFFFFFF
             ref = int(ref) & 0xFFFFFFFFFFFFFFFF
