@@ -56,7 +56,23 @@ def build_gpu(force: bool = False) -> Path:
     return GPU_LIB
 
 
+PROF_DIR = OPS_DIR.parent / "profiler"
+PROF_LIB = PROF_DIR / "libdfprof.so"
+
+
+def build_prof(force: bool = False) -> Path:
+    """roctracer subscriber for the continuous GPU profiler."""
+    src = PROF_DIR / "csrc" / "gpuprof.cpp"
+    if force or not PROF_LIB.exists() or \
+            src.stat().st_mtime > PROF_LIB.stat().st_mtime:
+        _run([HIPCC, "-O2", "-std=c++17", "-shared", "-fPIC", str(src),
+              "-I/opt/rocm/include", "-L/opt/rocm/lib", "-lroctracer64",
+              "-o", str(PROF_LIB)])
+    return PROF_LIB
+
+
 def build_all(force: bool = False) -> None:
+    build_prof(force)
     build_cpu(force)
     build_gpu(force)
 

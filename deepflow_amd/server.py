@@ -167,8 +167,13 @@ class DeepflowServer:
             if self.device != "cuda":
                 return {"status": "skipped", "reason": "cpu device"}
             import time as _t
-            from .profiler import GpuProfiler
-            gp = GpuProfiler(self.profiles, process_name="deepflow-server")
+            from .profiler import GpuProfiler, NativeGpuProfiler
+            try:
+                gp = NativeGpuProfiler(self.profiles,
+                                       process_name="deepflow-server")
+            except Exception:   # roctracer unavailable -> kineto path
+                gp = GpuProfiler(self.profiles,
+                                 process_name="deepflow-server")
             with gp.capture():
                 _t.sleep(min(seconds, 10.0))
             return {"status": "ok",
