@@ -66,7 +66,7 @@ def build_prof(force: bool = False) -> Path:
     if force or not PROF_LIB.exists() or \
             src.stat().st_mtime > PROF_LIB.stat().st_mtime:
         _run([HIPCC, "-O2", "-std=c++17", "-shared", "-fPIC", str(src),
-              "-I/opt/rocm/include", "-L/opt/rocm/lib", "-lroctracer64",
+              "-I/opt/rocm/include", "-L/opt/rocm/lib", "-lrocprofiler-sdk",
               "-o", str(PROF_LIB)])
     return PROF_LIB
 

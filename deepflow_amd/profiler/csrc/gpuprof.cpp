@@ -1,25 +1,30 @@
-// dfprof — in-process roctracer subscriber for the continuous GPU
+// dfprof — in-process rocprofiler-sdk subscriber for the continuous GPU
 // profiler (BASELINE config #5). Round 1 wrapped torch.profiler/kineto,
-// paying ~0.7 s per capture window; this native subscriber turns a
-// window on/off in microseconds and aggregates kernel activity in-place,
-// so 1-window/60 s continuous mode costs <<0.5%.
+// paying ~0.7 s per capture window; here a window is
+// rocprofiler_start/stop_context (microseconds), with kernel dispatch
+// records aggregated natively, so 1-window/60 s continuous mode costs
+// <<0.5%. No kineto, no CUPTI, no torch.
 //
-// Design: ACTIVITY_DOMAIN_HIP_OPS pool records carry device begin/end
-// timestamps and the kernel name pointer for dispatch ops; the buffer
-// callback folds them into a (name -> count/total/max) table under a
-// mutex. dfp_drain serializes the table and resets it. No kineto, no
-// CUPTI, no torch — the records land in the same profile.in_process
-// store as the eBPF CPU profiler (ingest/profile_pipeline.py).
+// rocprofiler-sdk requires tool registration BEFORE the HIP runtime
+// initializes: dfp_register() (python: native_profiler.ensure_early())
+// must run before the first torch/HIP GPU touch — the server and bench
+// call it at process start. Two contexts:
+//   - code-object context: always on, maps kernel_id -> kernel name
+//   - dispatch context:    windowed, buffers kernel begin/end timestamps
 //
-// Build: hipcc -shared -fPIC gpuprof.cpp -lroctracer64
+// Build: hipcc -shared -fPIC gpuprof.cpp -lrocprofiler-sdk
 #include <cstdint>
 #include <cstring>
 #include <map>
 #include <mutex>
 #include <string>
 
-#include <roctracer/roctracer.h>
-#include <roctracer/roctracer_hip.h>
+#include <rocprofiler-sdk/buffer.h>
+#include <rocprofiler-sdk/buffer_tracing.h>
+#include <rocprofiler-sdk/callback_tracing.h>
+#include <rocprofiler-sdk/fwd.h>
+#include <rocprofiler-sdk/registration.h>
+#include <rocprofiler-sdk/rocprofiler.h>
 
 namespace {
 
@@ -31,68 +36,127 @@ struct Agg {
 
 std::mutex g_mu;
 std::map<std::string, Agg> g_table;
+std::map<uint64_t, std::string> g_kernel_names;
 uint64_t g_records = 0;
-uint64_t g_dropped = 0;
-bool g_pool_open = false;
+rocprofiler_context_id_t g_ctx_code = {0};
+rocprofiler_context_id_t g_ctx_disp = {0};
+rocprofiler_buffer_id_t g_buf = {0};
+int g_state = 0;  // 0 = unregistered, 1 = registered, 2 = initialized
 
-void activity_cb(const char* begin, const char* end, void* /*arg*/) {
-    const roctracer_record_t* rec = (const roctracer_record_t*)begin;
-    const roctracer_record_t* end_rec = (const roctracer_record_t*)end;
-    std::lock_guard<std::mutex> lk(g_mu);
-    while (rec < end_rec) {
-        if (rec->domain == ACTIVITY_DOMAIN_HIP_OPS ||
-            rec->domain == ACTIVITY_DOMAIN_HSA_OPS) {
-            uint64_t dur = rec->end_ns > rec->begin_ns
-                               ? rec->end_ns - rec->begin_ns
-                               : 0;
-            const char* nm = nullptr;
-            if (rec->op == HIP_OP_ID_DISPATCH && rec->kernel_name)
-                nm = rec->kernel_name;
-            if (nm == nullptr) {
-                // copies/barriers aggregate under the op label
-                nm = roctracer_op_string(rec->domain, rec->op, rec->kind);
-            }
-            if (nm != nullptr && dur > 0) {
-                Agg& a = g_table[nm];
-                a.count++;
-                a.total_ns += dur;
-                if (dur > a.max_ns) a.max_ns = dur;
-                g_records++;
-            }
-        }
-        if (roctracer_next_record(rec, &rec) != ROCTRACER_STATUS_SUCCESS)
-            break;
+void code_object_cb(rocprofiler_callback_tracing_record_t record,
+                    rocprofiler_user_data_t*, void*) {
+    if (record.kind != ROCPROFILER_CALLBACK_TRACING_CODE_OBJECT ||
+        record.operation !=
+            ROCPROFILER_CODE_OBJECT_DEVICE_KERNEL_SYMBOL_REGISTER ||
+        record.phase != ROCPROFILER_CALLBACK_PHASE_LOAD)
+        return;
+    auto* sym = (rocprofiler_callback_tracing_code_object_kernel_symbol_register_data_t*)
+        record.payload;
+    if (sym && sym->kernel_name) {
+        std::lock_guard<std::mutex> lk(g_mu);
+        g_kernel_names[sym->kernel_id] = sym->kernel_name;
     }
 }
+
+void buffer_cb(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
+               rocprofiler_record_header_t** headers, size_t n,
+               void*, uint64_t) {
+    std::lock_guard<std::mutex> lk(g_mu);
+    for (size_t i = 0; i < n; i++) {
+        rocprofiler_record_header_t* h = headers[i];
+        if (h->category != ROCPROFILER_BUFFER_CATEGORY_TRACING ||
+            h->kind != ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH)
+            continue;
+        auto* rec =
+            (rocprofiler_buffer_tracing_kernel_dispatch_record_t*)h->payload;
+        uint64_t dur = rec->end_timestamp > rec->start_timestamp
+                           ? rec->end_timestamp - rec->start_timestamp
+                           : 0;
+        if (dur == 0) continue;
+        auto it = g_kernel_names.find(rec->dispatch_info.kernel_id);
+        const std::string& nm =
+            it != g_kernel_names.end()
+                ? it->second
+                : (g_kernel_names[rec->dispatch_info.kernel_id] =
+                       "kernel_" +
+                       std::to_string(rec->dispatch_info.kernel_id));
+        Agg& a = g_table[nm];
+        a.count++;
+        a.total_ns += dur;
+        if (dur > a.max_ns) a.max_ns = dur;
+        g_records++;
+    }
+}
+
+int tool_init(rocprofiler_client_finalize_t, void*) {
+    if (rocprofiler_create_context(&g_ctx_code) != ROCPROFILER_STATUS_SUCCESS)
+        return -1;
+    if (rocprofiler_create_context(&g_ctx_disp) != ROCPROFILER_STATUS_SUCCESS)
+        return -1;
+    if (rocprofiler_configure_callback_tracing_service(
+            g_ctx_code, ROCPROFILER_CALLBACK_TRACING_CODE_OBJECT, nullptr, 0,
+            code_object_cb, nullptr) != ROCPROFILER_STATUS_SUCCESS)
+        return -1;
+    if (rocprofiler_create_buffer(g_ctx_disp, 1 << 22, 3 << 20,
+                                  ROCPROFILER_BUFFER_POLICY_LOSSLESS,
+                                  buffer_cb, nullptr,
+                                  &g_buf) != ROCPROFILER_STATUS_SUCCESS)
+        return -1;
+    if (rocprofiler_configure_buffer_tracing_service(
+            g_ctx_disp, ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH, nullptr,
+            0, g_buf) != ROCPROFILER_STATUS_SUCCESS)
+        return -1;
+    // names must be known for kernels loaded at any time
+    rocprofiler_start_context(g_ctx_code);
+    g_state = 2;
+    return 0;
+}
+
+void tool_fini(void*) {}
 
 }  // namespace
 
 extern "C" {
 
-// open the activity pool once; enable the async ops domain
-int dfp_start() {
-    if (!g_pool_open) {
-        roctracer_properties_t props{};
-        props.buffer_size = 1 << 20;
-        props.buffer_callback_fun = activity_cb;
-        if (roctracer_open_pool(&props) != ROCTRACER_STATUS_SUCCESS)
-            return -1;
-        g_pool_open = true;
-    }
-    if (roctracer_enable_domain_activity(ACTIVITY_DOMAIN_HIP_OPS) !=
-        ROCTRACER_STATUS_SUCCESS)
-        return -2;
+rocprofiler_tool_configure_result_t* dfp_configure(
+    uint32_t /*version*/, const char* /*runtime_version*/,
+    uint32_t /*priority*/, rocprofiler_client_id_t* id) {
+    id->name = "dfprof";
+    static rocprofiler_tool_configure_result_t cfg{
+        sizeof(rocprofiler_tool_configure_result_t), &tool_init, &tool_fini,
+        nullptr};
+    return &cfg;
+}
+
+// Must run BEFORE the HIP runtime initializes (ensure_early()).
+int dfp_register() {
+    if (g_state >= 1) return 0;
+    if (rocprofiler_force_configure(&dfp_configure) !=
+        ROCPROFILER_STATUS_SUCCESS)
+        return -1;
+    g_state = 1;
     return 0;
 }
 
+int dfp_ready() { return g_state; }
+
+int dfp_start() {
+    if (g_state != 2) return -10;  // not registered early enough
+    return rocprofiler_start_context(g_ctx_disp) ==
+                   ROCPROFILER_STATUS_SUCCESS
+               ? 0
+               : -1;
+}
+
 int dfp_stop() {
-    roctracer_disable_domain_activity(ACTIVITY_DOMAIN_HIP_OPS);
-    roctracer_flush_activity();
+    if (g_state != 2) return -10;
+    rocprofiler_stop_context(g_ctx_disp);
+    rocprofiler_flush_buffer(g_buf);
     return 0;
 }
 
 // [u32 name_len][name][u64 count][u64 total_ns][u64 max_ns] ...
-// returns bytes written (table is drained); 0-cap call sizes the buffer
+// returns bytes written (drains the table); 0-cap call sizes the buffer
 uint64_t dfp_drain(uint8_t* out, uint64_t cap) {
     std::lock_guard<std::mutex> lk(g_mu);
     uint64_t need = 0;

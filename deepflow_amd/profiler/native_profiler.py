@@ -31,6 +31,8 @@ def lib() -> ct.CDLL:
             from ..ops import build
             build.build_prof()
         h = ct.CDLL(str(path))
+        h.dfp_register.restype = ct.c_int
+        h.dfp_ready.restype = ct.c_int
         h.dfp_start.restype = ct.c_int
         h.dfp_stop.restype = ct.c_int
         h.dfp_drain.restype = ct.c_uint64
@@ -38,6 +40,22 @@ def lib() -> ct.CDLL:
         h.dfp_record_count.restype = ct.c_uint64
         _lib = h
     return _lib
+
+
+def ensure_early() -> bool:
+    """Register the rocprofiler-sdk tool. MUST run before the first HIP
+    runtime touch (i.e. before torch initializes the GPU) — call it at
+    process start (bench.py honors DF_GPU_PROF=1; the server calls it in
+    main()). Returns True when the subscriber is registered."""
+    return lib().dfp_register() == 0
+
+
+def available() -> bool:
+    """Registered early enough for capture windows to work?"""
+    try:
+        return lib().dfp_ready() == 2
+    except OSError:
+        return False
 
 
 def drain() -> List[Tuple[str, int, int, int]]:
@@ -79,6 +97,11 @@ class NativeGpuProfiler:
         t0 = time.perf_counter_ns()
         rc = h.dfp_start()
         t1 = time.perf_counter_ns()
+        if rc == -10:
+            raise RuntimeError(
+                "dfprof not registered before HIP init — call "
+                "native_profiler.ensure_early() at process start "
+                "(DF_GPU_PROF=1 for bench.py)")
         if rc != 0:
             raise RuntimeError(f"dfp_start failed rc={rc}")
         try:

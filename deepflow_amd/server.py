@@ -495,6 +495,12 @@ class DeepflowServer:
 
 def main() -> None:
     import argparse
+    # native GPU profiler registration must precede HIP init
+    try:
+        from .profiler.native_profiler import ensure_early
+        ensure_early()
+    except Exception:
+        pass
     import torch
     ap = argparse.ArgumentParser(description="deepflow-amd all-in-one server")
     ap.add_argument("--device", default=None)
