@@ -317,18 +317,30 @@ class QueryEngine:
                               for (d, s) in self.pipe.dict.str_to_id
                               if d == 7)
             return {"columns": ["value"], "values": [[v] for v in vals]}
-        if what == "tags":
-            cols = ["name", "display_name", "type"]
-            vals = [[n, n, t.hydrate] for n, t in sorted(tags.items())]
-            return {"columns": cols, "values": vals}
-        if what == "metrics":
-            from .tags import METRIC_UNITS
-            cols = ["name", "display_name", "unit", "type"]
-            vals = []
-            for n in sorted(mets):
-                unit, disp = METRIC_UNITS.get(n, ("count", n))
-                vals.append([n, disp, unit, "counter"])
-            return {"columns": cols, "values": vals}
+        if what in ("tags", "metrics"):
+            # db_descriptions-backed discovery (reference:
+            # querier/db_descriptions/clickhouse/{tag,metrics}/)
+            from .descriptions import table_descriptions
+            catalog = table_descriptions(self)
+            entry = catalog.get(table)
+            if entry is None and "." not in table:
+                entry = catalog.get(table + ".1s")
+            cols = ["name", "display_name", "unit", "type", "description"]
+            if entry is not None:
+                rows_ = entry[what]
+                return {"columns": cols,
+                        "values": [[d["name"], d["display_name"],
+                                    d["unit"], d["type"], d["description"]]
+                                   for d in rows_]}
+            # unknown table: fall back to the engine tag map
+            from .descriptions import describe
+            src = tags if what == "tags" else mets
+            return {"columns": cols,
+                    "values": [[d["name"], d["display_name"], d["unit"],
+                                d["type"], d["description"]]
+                               for d in (describe(n, "tag" if what == "tags"
+                                                  else "metric")
+                                         for n in sorted(src))]}
         if what == "tables":
             names = ["l7_flow_log", "l4_flow_log"]
             for pipe in (self.pipe, self.l4):

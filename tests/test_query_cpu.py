@@ -264,3 +264,26 @@ def test_select_attribute_column(eng):
         nm = dict(zip(t["ext_info"]["attribute_names"],
                       t["ext_info"]["attribute_values"]))
         assert (t["req"]["resource"], nm[name]) in got
+
+
+def test_show_descriptions(eng):
+    """db_descriptions parity: show tags/metrics carry display name,
+    unit and description, generated from the live tag maps for every
+    table (reference: querier/db_descriptions 203 data files)."""
+    r = eng.query("SHOW tags FROM l7_flow_log")
+    assert r["columns"] == ["name", "display_name", "unit", "type",
+                            "description"]
+    by_name = {v[0]: v for v in r["values"]}
+    assert by_name["request_domain"][1] == "Request Domain"
+    assert "Host" in by_name["request_domain"][4]
+    assert by_name["pod_id_1"][1] == "Server Pod"
+    r = eng.query("SHOW metrics FROM l7_flow_log")
+    mm = {v[0]: v for v in r["values"]}
+    assert mm["response_duration"][2] == "us"
+    # rollup tables are covered too
+    r = eng.query("SHOW metrics FROM application.1s")
+    mm = {v[0]: v for v in r["values"]}
+    assert "request" in mm and mm["rrt_max"][2] == "us"
+    r = eng.query("SHOW tags FROM application_map.1s")
+    names = {v[0] for v in r["values"]}
+    assert {"ip_0", "ip_1", "server_port"} <= names
