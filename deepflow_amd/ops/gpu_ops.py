@@ -66,6 +66,26 @@ def rollup_insert(kws: torch.Tensor, vals: torch.Tensor, ops: torch.Tensor,
         stream or _stream()), "df_rollup_insert")
 
 
+def sort_u64(data: torch.Tensor) -> None:
+    """In-place rocPRIM radix sort of a device int64 tensor (values
+    treated as u64)."""
+    import ctypes
+    lib = native.gpu()
+    n = data.numel()
+    if n == 0:
+        return
+    alt = torch.empty_like(data)
+    nbytes = ctypes.c_uint64(0)
+    native.check(lib.df_sort_u64(data.data_ptr(), alt.data_ptr(), n, None,
+                                 ctypes.byref(nbytes), _stream()),
+                 "df_sort_u64(size)")
+    temp = torch.empty(int(nbytes.value), dtype=torch.uint8,
+                       device=data.device)
+    native.check(lib.df_sort_u64(data.data_ptr(), alt.data_ptr(), n,
+                                 temp.data_ptr(), ctypes.byref(nbytes),
+                                 _stream()), "df_sort_u64")
+
+
 def gather_records(payload: torch.Tensor, offs: torch.Tensor,
                    lens: torch.Tensor, sel: torch.Tensor,
                    dst_off: torch.Tensor, out: torch.Tensor) -> None:

@@ -123,6 +123,8 @@ def _decl_gpu(lib: ct.CDLL) -> None:
                                     u64]
     lib.df_spec_sizes.restype = ct.c_int
     lib.df_spec_sizes.argtypes = [p, p, p, p]
+    lib.df_sort_u64.restype = ct.c_int
+    lib.df_sort_u64.argtypes = [p, p, u32, p, ct.POINTER(ct.c_uint64), u64]
     lib.df_pack_bits.restype = ct.c_int
     lib.df_pack_bits.argtypes = [p, u32, u32, u32, p, u32, u64]
     lib.df_unpack_bits.restype = ct.c_int

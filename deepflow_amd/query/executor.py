@@ -405,10 +405,26 @@ def execute_grouped_values(plan: Plan, segments, metas, device: str,
     counts = torch.bincount(inverse, minlength=g)
     starts = torch.cumsum(counts, 0) - counts
     out = {}
+    _VBITS = 44  # composite sort key: (group << 44) | value
+    _VMAX = (1 << _VBITS) - 1
     for mi in range(len(metas)):
-        vals = torch.cat(val_cols[mi]).to(torch.float64)
-        order1 = torch.argsort(vals)
-        inv1 = inverse[order1]
-        order2 = torch.argsort(inv1, stable=True)
-        out[mi] = (vals[order1[order2]], starts, counts)
+        vals64 = torch.cat(val_cols[mi])
+        if device != "cpu" and torch.cuda.is_available() and \
+                vals64.numel() and int(vals64.max()) <= _VMAX and \
+                g < (1 << (64 - _VBITS)):
+            # device path: ONE rocPRIM radix sort of the composite key
+            # replaces the two torch.argsort passes
+            from ..ops import gpu_ops
+            comp = (inverse.to("cuda") << _VBITS) | \
+                vals64.to("cuda").clamp_(min=0)
+            gpu_ops.sort_u64(comp)
+            torch.cuda.synchronize()
+            svals = (comp & _VMAX).cpu().to(torch.float64)
+        else:
+            vals = vals64.to(torch.float64)
+            order1 = torch.argsort(vals)
+            inv1 = inverse[order1]
+            order2 = torch.argsort(inv1, stable=True)
+            svals = vals[order1[order2]]
+        out[mi] = (svals, starts, counts)
     return uniq, out

@@ -129,3 +129,41 @@ def test_grouped_percentile_gpu(engines):
     want = engines["cpu"].query(q)
     got = engines["cuda"].query(q)
     assert want == got
+
+
+def test_grouped_percentile_rocprim_sort():
+    """Percentile group-gather uses the rocPRIM composite-key radix sort
+    on device; results must match the CPU oracle exactly."""
+    import torch
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query.engine import QueryEngine
+    cfg = SpanGenConfig(n=8000, seed=13, tag_cardinality=100, n_ips=64,
+                        n_services=8, n_resources=32)
+    pipes = {}
+    for dev in ("cuda", "cpu"):
+        p = L7IngestPipeline(device=dev, segment_rows=1 << 14,
+                             dict_capacity=1 << 14,
+                             time_base_s=cfg.base_time_ns // 10**9)
+        p.ingest_frame_payload(gen_span_payload(cfg))
+        pipes[dev] = p
+    torch.cuda.synchronize()
+    q = ("SELECT request_domain, Percentile(response_duration, 95) AS p95, "
+         "Apdex(response_duration, 100000) AS ap FROM l7_flow_log "
+         "GROUP BY request_domain ORDER BY request_domain")
+    got = QueryEngine(pipes["cuda"], device="cuda").query(q)
+    want = QueryEngine(pipes["cpu"], device="cpu").query(q)
+    assert got == want
+
+
+def test_sort_u64_kernel():
+    import torch
+    from deepflow_amd.ops import gpu_ops
+    g = torch.randint(0, 1 << 62, (100000,), dtype=torch.int64,
+                      device="cuda").abs()
+    want = g.cpu().numpy().copy()
+    want.sort()
+    gpu_ops.sort_u64(g)
+    torch.cuda.synchronize()
+    assert (g.cpu().numpy() == want).all()
