@@ -42,6 +42,7 @@ rocprofiler_context_id_t g_ctx_code = {0};
 rocprofiler_context_id_t g_ctx_disp = {0};
 rocprofiler_buffer_id_t g_buf = {0};
 int g_state = 0;  // 0 = unregistered, 1 = registered, 2 = initialized
+int g_err = 0;    // first failing step in tool_init (diagnostics)
 
 void code_object_cb(rocprofiler_callback_tracing_record_t record,
                     rocprofiler_user_data_t*, void*) {
@@ -89,25 +90,26 @@ void buffer_cb(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
 }
 
 int tool_init(rocprofiler_client_finalize_t, void*) {
-    if (rocprofiler_create_context(&g_ctx_code) != ROCPROFILER_STATUS_SUCCESS)
-        return -1;
-    if (rocprofiler_create_context(&g_ctx_disp) != ROCPROFILER_STATUS_SUCCESS)
-        return -1;
-    if (rocprofiler_configure_callback_tracing_service(
+    rocprofiler_status_t rc;
+    rc = rocprofiler_create_context(&g_ctx_code);
+    if (rc != ROCPROFILER_STATUS_SUCCESS) { g_err = 100 + (int)rc; return -1; }
+    rc = rocprofiler_create_context(&g_ctx_disp);
+    if (rc != ROCPROFILER_STATUS_SUCCESS) { g_err = 200 + (int)rc; return -1; }
+    rc = rocprofiler_configure_callback_tracing_service(
             g_ctx_code, ROCPROFILER_CALLBACK_TRACING_CODE_OBJECT, nullptr, 0,
-            code_object_cb, nullptr) != ROCPROFILER_STATUS_SUCCESS)
-        return -1;
-    if (rocprofiler_create_buffer(g_ctx_disp, 1 << 22, 3 << 20,
-                                  ROCPROFILER_BUFFER_POLICY_LOSSLESS,
-                                  buffer_cb, nullptr,
-                                  &g_buf) != ROCPROFILER_STATUS_SUCCESS)
-        return -1;
-    if (rocprofiler_configure_buffer_tracing_service(
+            code_object_cb, nullptr);
+    if (rc != ROCPROFILER_STATUS_SUCCESS) { g_err = 300 + (int)rc; return -1; }
+    rc = rocprofiler_create_buffer(g_ctx_disp, 1 << 22, 3 << 20,
+                                   ROCPROFILER_BUFFER_POLICY_LOSSLESS,
+                                   buffer_cb, nullptr, &g_buf);
+    if (rc != ROCPROFILER_STATUS_SUCCESS) { g_err = 400 + (int)rc; return -1; }
+    rc = rocprofiler_configure_buffer_tracing_service(
             g_ctx_disp, ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH, nullptr,
-            0, g_buf) != ROCPROFILER_STATUS_SUCCESS)
-        return -1;
+            0, g_buf);
+    if (rc != ROCPROFILER_STATUS_SUCCESS) { g_err = 500 + (int)rc; return -1; }
     // names must be known for kernels loaded at any time
-    rocprofiler_start_context(g_ctx_code);
+    rc = rocprofiler_start_context(g_ctx_code);
+    if (rc != ROCPROFILER_STATUS_SUCCESS) { g_err = 600 + (int)rc; return -1; }
     g_state = 2;
     return 0;
 }
@@ -118,27 +120,34 @@ void tool_fini(void*) {}
 
 extern "C" {
 
-rocprofiler_tool_configure_result_t* dfp_configure(
+// canonical tool entry: rocprofiler dlopens the libraries listed in
+// ROCP_TOOL_LIBRARIES at HIP runtime init and calls this export
+// (ensure_early() sets the env var before torch loads the runtime)
+rocprofiler_tool_configure_result_t* rocprofiler_configure(
     uint32_t /*version*/, const char* /*runtime_version*/,
     uint32_t /*priority*/, rocprofiler_client_id_t* id) {
     id->name = "dfprof";
+    if (g_state == 2) return nullptr;  // single client
+    g_state = 1;
     static rocprofiler_tool_configure_result_t cfg{
         sizeof(rocprofiler_tool_configure_result_t), &tool_init, &tool_fini,
         nullptr};
     return &cfg;
 }
 
-// Must run BEFORE the HIP runtime initializes (ensure_early()).
+// best-effort late registration (processes that skipped ensure_early)
 int dfp_register() {
     if (g_state >= 1) return 0;
-    if (rocprofiler_force_configure(&dfp_configure) !=
+    if (rocprofiler_force_configure(&rocprofiler_configure) !=
         ROCPROFILER_STATUS_SUCCESS)
         return -1;
-    g_state = 1;
+    if (g_state == 0) g_state = 1;
     return 0;
 }
 
 int dfp_ready() { return g_state; }
+
+int dfp_err() { return g_err; }
 
 int dfp_start() {
     if (g_state != 2) return -10;  // not registered early enough

@@ -43,11 +43,24 @@ def lib() -> ct.CDLL:
 
 
 def ensure_early() -> bool:
-    """Register the rocprofiler-sdk tool. MUST run before the first HIP
-    runtime touch (i.e. before torch initializes the GPU) — call it at
-    process start (bench.py honors DF_GPU_PROF=1; the server calls it in
-    main()). Returns True when the subscriber is registered."""
-    return lib().dfp_register() == 0
+    """Arrange tool registration. MUST run before the first HIP runtime
+    touch (before torch initializes the GPU): rocprofiler loads the
+    libraries named in ROCP_TOOL_LIBRARIES at runtime init and calls
+    their exported rocprofiler_configure — we add libdfprof.so there and
+    pre-load it so the python bindings share the same instance. Call at
+    process start (bench.py honors DF_GPU_PROF=1; server main() always
+    does)."""
+    import os
+    path = PROF_DIR / "libdfprof.so"
+    if not path.exists():
+        from ..ops import build
+        build.build_prof()
+    prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
+    if str(path) not in prev:
+        os.environ["ROCP_TOOL_LIBRARIES"] = \
+            f"{path}:{prev}" if prev else str(path)
+    lib()
+    return True
 
 
 def available() -> bool:

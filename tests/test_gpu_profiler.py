@@ -53,6 +53,7 @@ from deepflow_amd.profiler import native_profiler as npf
 assert npf.ensure_early()
 import torch
 assert torch.cuda.is_available()
+torch.zeros(1, device="cuda")   # full runtime init fires tool_init
 from deepflow_amd.gen import SpanGenConfig
 from deepflow_amd.gen.spans import gen_span_payload
 from deepflow_amd.ingest import L7IngestPipeline
