@@ -65,9 +65,12 @@ def build_prof(force: bool = False) -> Path:
     src = PROF_DIR / "csrc" / "gpuprof.cpp"
     if force or not PROF_LIB.exists() or \
             src.stat().st_mtime > PROF_LIB.stat().st_mtime:
-        _run([HIPCC, "-O2", "-std=c++17", "-shared", "-fPIC", str(src),
+        # plain g++: a rocprofiler tool library must NOT embed HIP
+        # runtime registration stubs (hipcc links them and the runtime
+        # then aborts in rocprofiler_set_api_table when the tool loads)
+        _run(["g++", "-O2", "-std=c++17", "-shared", "-fPIC", str(src),
               "-I/opt/rocm/include", "-L/opt/rocm/lib", "-lrocprofiler-sdk",
-              "-o", str(PROF_LIB)])
+              "-Wl,-rpath,/opt/rocm/lib", "-o", str(PROF_LIB)])
     return PROF_LIB
 
 
