@@ -68,7 +68,8 @@ def build_prof(force: bool = False) -> Path:
         # plain g++: a rocprofiler tool library must NOT embed HIP
         # runtime registration stubs (hipcc links them and the runtime
         # then aborts in rocprofiler_set_api_table when the tool loads)
-        _run(["g++", "-O2", "-std=c++17", "-shared", "-fPIC", str(src),
+        _run(["g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+              "-D__HIP_PLATFORM_AMD__", str(src),
               "-I/opt/rocm/include", "-L/opt/rocm/lib", "-lrocprofiler-sdk",
               "-Wl,-rpath,/opt/rocm/lib", "-o", str(PROF_LIB)])
     return PROF_LIB
