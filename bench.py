@@ -196,6 +196,7 @@ def e2e_main(args) -> None:
     if device == "cpu" and args.batch > 20000:
         args.batch = 2000
     cfg = SpanGenConfig(n=args.batch, seed=77, tag_cardinality=args.tag_card,
+                        dt_ns=1000,
                         n_ips=4096, n_services=256, n_resources=4096,
                         n_attrs=args.n_attrs)
     kg = KnowledgeGraphTable(capacity_pow2=1 << 14, device=device)
@@ -323,8 +324,12 @@ def main() -> None:
     if device == "cuda":
         torch.cuda.set_device(local_rank)
 
+    # dt_ns=1000: batch timestamps span batch/1e6 seconds — a ~1M
+    # spans/s/stream arrival rate (dt=1ms spread 2M spans over 2000 s of
+    # event time, inflating 1s-bucket cardinality 1000x beyond any real
+    # stream and saturating the map rollup tables)
     cfg = SpanGenConfig(n=args.batch, seed=1234,
-                        tag_cardinality=args.tag_card,
+                        tag_cardinality=args.tag_card, dt_ns=1000,
                         n_ips=4096, n_services=256, n_resources=4096,
                         n_attrs=args.n_attrs)
     n_distinct = min(args.steps + args.warmup, 4)

@@ -42,17 +42,25 @@ def decode_l4(payload: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor,
         scratch_str.shape[1], _stream()), "df_decode_l4")
 
 
-def rollup(seg, base_row: int, n: int, time_base_s: int, table,
-           stream: int = 0) -> None:
-    """flow_metrics table rollup from segment rows (k_rollup_l4/l7)."""
+def rollup_family(seg, base_row: int, n: int, time_base_s: int, tables,
+                  stream: int = 0) -> None:
+    """ONE fused launch updates every table of a family (k_rollup_l4/l7
+    iterate the per-table key specs over a single pass of the row)."""
+    import numpy as np
     lib = native.gpu()
-    fn = lib.df_rollup_l4 if table.td.source == "l4" else lib.df_rollup_l7
+    src = tables[0].td.source
+    fn = lib.df_rollup_l4 if src == "l4" else lib.df_rollup_l7
+    specs = b"".join(t.spec_bytes() for t in tables)
+    ptrs = np.array([x for t in tables
+                     for x in (t.tkeys.data_ptr(), t.traw.data_ptr(),
+                               t.tvals.data_ptr(), t.drops.data_ptr())],
+                    dtype=np.uint64)
+    caps = np.array([t.capacity for t in tables], dtype=np.uint32)
     native.check(fn(
         seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
-        seg.capacity, base_row, n, time_base_s, table.spec_bytes(),
-        table.tkeys.data_ptr(), table.traw.data_ptr(),
-        table.tvals.data_ptr(), table.capacity, table.drops.data_ptr(),
-        stream or _stream()), "df_rollup")
+        seg.capacity, base_row, n, time_base_s, specs, len(tables),
+        ptrs.ctypes.data, caps.ctypes.data,
+        stream or _stream()), "df_rollup_family")
 
 
 def rollup_insert(kws: torch.Tensor, vals: torch.Tensor, ops: torch.Tensor,

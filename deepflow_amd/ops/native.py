@@ -93,10 +93,10 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_decode_l4.argtypes = [p, p, p, u32, p, p, p, p, u64, u64, u64, u64]
     lib.df_rollup_l4.restype = ct.c_int
     lib.df_rollup_l4.argtypes = [p, p, p, u64, u64, u32, u64, ct.c_char_p,
-                                 p, p, p, u32, p, u64]
+                                 u32, p, p, u64]
     lib.df_rollup_l7.restype = ct.c_int
     lib.df_rollup_l7.argtypes = [p, p, p, u64, u64, u32, u64, ct.c_char_p,
-                                 p, p, p, u32, p, u64]
+                                 u32, p, p, u64]
     lib.df_rollup_insert.restype = ct.c_int
     lib.df_rollup_insert.argtypes = [p, p, p, u32, u32, u32, p, p, p, u32,
                                      p, u64]

@@ -183,8 +183,8 @@ class RollupTable:
             self._update_cpu(seg, base, n)
         else:
             from ..ops import gpu_ops
-            gpu_ops.rollup(seg, base, n, self.time_base_s, self,
-                           stream=stream)
+            gpu_ops.rollup_family(seg, base, n, self.time_base_s, [self],
+                                  stream=stream)
 
     def _key_tuple(self, seg, row: int) -> Optional[tuple]:
         t_s = (int(seg.u64[0, row].item()) & M64) // 10**9
@@ -355,8 +355,16 @@ class RollupFamily:
             for td in defs}
 
     def update(self, seg, base: int, n: int, stream: int = 0) -> None:
-        for t in self.tables.values():
-            t.update(seg, base, n, stream)
+        tables = list(self.tables.values())
+        if not tables or n == 0:
+            return
+        if tables[0].device == "cpu":
+            for t in tables:
+                t.update(seg, base, n, stream)
+            return
+        from ..ops import gpu_ops
+        gpu_ops.rollup_family(seg, base, n, tables[0].time_base_s, tables,
+                              stream=stream)
 
     def get(self, name: str) -> Optional[RollupTable]:
         return self.tables.get(name)
