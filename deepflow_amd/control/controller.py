@@ -218,10 +218,24 @@ class ControllerLite:
             })
         return out
 
-    def rebalance(self) -> Dict[int, int]:
-        """Re-assign agents across analyzers round-robin (reference
-        monitor/vtap/rebalance.go analog). Returns agent -> analyzer."""
-        assignment = {}
+    def rebalance(self, traffic: Optional[Dict[int, float]] = None
+                  ) -> Dict[int, int]:
+        """Re-assign agents across analyzers. With per-agent traffic
+        rates (bytes/s from receiver status or agent sync), bins are
+        packed greedily heaviest-first onto the least-loaded analyzer
+        (reference monitor/vtap/rebalance.go traffic-weighted mode);
+        without rates, round-robin."""
+        assignment: Dict[int, int] = {}
+        if traffic:
+            load = [0.0] * self.n_analyzers
+            order = sorted(self.agents,
+                           key=lambda a: -traffic.get(a, 0.0))
+            for agent_id in order:
+                tgt = min(range(self.n_analyzers), key=lambda i: load[i])
+                load[tgt] += traffic.get(agent_id, 0.0)
+                self.agents[agent_id].analyzer = tgt
+                assignment[agent_id] = tgt
+            return assignment
         for i, agent_id in enumerate(sorted(self.agents)):
             self.agents[agent_id].analyzer = i % self.n_analyzers
             assignment[agent_id] = self.agents[agent_id].analyzer
@@ -250,8 +264,15 @@ class ControllerLite:
             return {"status": "ok", "config_version": self.config_version}
 
         @app.post("/v1/rebalance/")
-        def rebalance():
-            return self.rebalance()
+        async def rebalance(request: Request):
+            body = {}
+            try:
+                body = await request.json()
+            except Exception:
+                pass
+            traffic = {int(k): float(v)
+                       for k, v in (body.get("traffic") or {}).items()}
+            return self.rebalance(traffic or None)
 
         @app.post("/v1/genesis/")
         async def genesis(request: Request):

@@ -129,6 +129,9 @@ class DeepflowServer:
         self.controller.register(self.app)
         from .export import OtlpExporter
         self.exporter = OtlpExporter(self.engine)
+        from .export.prom_exporter import PromExporter
+        self.prom_exporter = PromExporter(self.l7, self.l4)
+        self.prom_exporter.register(self.app)
         from .query.mcp import McpServer
         self.mcp = McpServer(self.engine, self.profiles)
         self.mcp.register(self.app)

@@ -68,3 +68,22 @@ def test_alert_policies():
     eng.alert_event_rows = lambda: events.alert_events
     r = eng.query("SELECT policy_name, level FROM alert_event LIMIT 5")
     assert ["too-many-errors", LEVEL_CRITICAL] in r["values"]
+
+
+def test_traffic_weighted_rebalance():
+    """Heavy agents spread across analyzers by load, not id order
+    (reference monitor/vtap/rebalance.go)."""
+    from deepflow_amd.control import ControllerLite
+    c = ControllerLite(n_analyzers=2)
+    for aid in range(1, 5):
+        c.sync(aid)
+    # agents 1+2 are heavy; round-robin would pair them on one analyzer
+    traffic = {1: 1000.0, 2: 900.0, 3: 10.0, 4: 5.0}
+    assign = c.rebalance(traffic)
+    assert assign[1] != assign[2]          # heaviest two split
+    loads = [sum(traffic[a] for a, t in assign.items() if t == i)
+             for i in range(2)]
+    assert abs(loads[0] - loads[1]) <= 100  # near-balanced
+    # no traffic -> round-robin still works
+    assign2 = c.rebalance()
+    assert sorted(assign2.values()) == [0, 0, 1, 1]

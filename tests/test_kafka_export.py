@@ -116,3 +116,29 @@ def test_kafka_export_query():
     vals = {r["request_resource"]: r["c"]
             for (_, r) in broker.received}
     assert sum(vals.values()) == cfg.n
+
+
+def test_prometheus_exporter():
+    """Prometheus text exposition of the rollup tables (reference
+    exporters/prometheus counterpart)."""
+    from fastapi.testclient import TestClient
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    from deepflow_amd.server import DeepflowServer
+    cfg = SpanGenConfig(n=300, seed=6, tag_cardinality=20, n_ips=16,
+                        n_services=4, n_resources=8)
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12,
+                         time_base_s=cfg.base_time_ns // 10**9)
+    srv.l7.ingest_frame_payload(gen_span_payload(cfg))
+    client = TestClient(srv.app)
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert "deepflow_application_request{" in body
+    assert 'l7_protocol="' in body
+    # values sum to the ingested span count
+    total = sum(float(line.rsplit(" ", 2)[1])
+                for line in body.splitlines()
+                if line.startswith("deepflow_application_request{"))
+    assert total == cfg.n
