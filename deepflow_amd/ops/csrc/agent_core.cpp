@@ -281,6 +281,68 @@ bool parse_redis_request(const uint8_t* p, uint32_t n, std::string& cmd,
     return true;
 }
 
+// SQL statement obfuscation (reference protocol_logs/sql/sql_obfuscate.rs):
+// literals collapse to '?' so stored request_resource strings carry no
+// payload data and dedupe under SmartEncoding — '...'/"..." strings
+// (with backslash escapes), numeric literals, and IN (...) lists.
+std::string obfuscate_sql(const std::string& in) {
+    std::string out;
+    out.reserve(in.size());
+    size_t i = 0;
+    auto is_num_start = [&](size_t k) {
+        if (!isdigit((unsigned char)in[k])) return false;
+        if (k == 0) return true;
+        unsigned char prev = in[k - 1];
+        return !(isalnum(prev) || prev == '_' || prev == '.');
+    };
+    while (i < in.size()) {
+        char c = in[i];
+        if (c == '\'' || c == '"') {
+            char q = c;
+            i++;
+            while (i < in.size()) {
+                if (in[i] == '\\' && i + 1 < in.size()) i += 2;
+                else if (in[i] == q) { i++; break; }
+                else i++;
+            }
+            out.push_back('?');
+        } else if (is_num_start(i)) {
+            while (i < in.size() &&
+                   (isdigit((unsigned char)in[i]) || in[i] == '.' ||
+                    in[i] == 'e' || in[i] == 'E' ||
+                    ((in[i] == '+' || in[i] == '-') && i > 0 &&
+                     (in[i - 1] == 'e' || in[i - 1] == 'E'))))
+                i++;
+            out.push_back('?');
+        } else {
+            out.push_back(c);
+            i++;
+        }
+    }
+    // collapse "IN (?, ?, ?)" shapes to "IN (?)"
+    std::string squeezed;
+    squeezed.reserve(out.size());
+    for (size_t k = 0; k < out.size();) {
+        if (out[k] == '?') {
+            size_t j = k;
+            bool list = false;
+            while (j < out.size()) {
+                if (out[j] == '?' || out[j] == ',' || out[j] == ' ') {
+                    if (out[j] == ',') list = true;
+                    j++;
+                } else break;
+            }
+            squeezed.push_back('?');
+            if (!list) { k++; continue; }
+            k = j;
+        } else {
+            squeezed.push_back(out[k]);
+            k++;
+        }
+    }
+    return squeezed;
+}
+
 // MySQL client command packet: [len3][seq1][cmd1][stmt...]
 bool parse_mysql_request(const uint8_t* p, uint32_t n, std::string& stmt) {
     if (n < 5) return false;
@@ -291,6 +353,7 @@ bool parse_mysql_request(const uint8_t* p, uint32_t n, std::string& stmt) {
         uint32_t sl = plen - 1;
         if (5 + sl > n) sl = n - 5;
         stmt.assign((const char*)p + 5, sl);
+        stmt = obfuscate_sql(stmt);
         return true;
     }
     if (cmd == 0x16 || cmd == 0x17 || cmd == 0x0e) {  // prepare/exec/ping
@@ -309,6 +372,7 @@ bool parse_pgsql_request(const uint8_t* p, uint32_t n, std::string& stmt) {
     if (5 + sl > n) sl = n - 5;
     while (sl && p[5 + sl - 1] == 0) sl--;
     stmt.assign((const char*)p + 5, sl);
+    stmt = obfuscate_sql(stmt);
     return true;
 }
 

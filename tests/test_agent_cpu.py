@@ -428,3 +428,32 @@ def test_iso8583():
     assert recs[0]["req"]["req_type"] == "0200"
     assert recs[0]["base"]["head"]["rrt"] == 7000
     a.close()
+
+
+def test_sql_obfuscation():
+    """MySQL/PG statement literals collapse to '?' in request_resource
+    (reference sql_obfuscate.rs) — payload data never stores, and
+    statements dedupe under SmartEncoding."""
+    import struct as _struct
+    import time as _t
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import eth_ipv4_tcp, PSH_ACK
+    from deepflow_amd.wire import pb, flow_log, framing
+    sql = b"SELECT * FROM users WHERE name = 'alice' AND id IN (1, 2, 3) LIMIT 10"
+    pkt_payload = _struct.pack("<I", len(sql) + 1)[:3] + b"\x00\x03" + sql
+    agent = Agent(vtap_id=1)
+    pkt = eth_ipv4_tcp(0x0A000001, 0x0A000002, 41000, 3306, seq=1,
+                       flags=PSH_ACK, payload=pkt_payload)
+    agent.packet(pkt, 10**18)
+    # server OK response
+    ok = b"\x07\x00\x00\x01\x00\x00\x00\x02\x00\x00\x00"
+    rpkt = eth_ipv4_tcp(0x0A000002, 0x0A000001, 3306, 41000, seq=1,
+                        flags=PSH_ACK, payload=ok)
+    agent.packet(rpkt, 10**18 + 10**6)
+    agent.tick(3 * 10**18)
+    recs = list(framing.iter_records(agent.drain(1)))
+    assert recs
+    d = pb.decode(recs[0], flow_log.APP_PROTO_LOGS_DATA)
+    res = d["req"]["resource"]
+    assert "alice" not in res and "10" not in res
+    assert res == "SELECT * FROM users WHERE name = ? AND id IN (?) LIMIT ?"
