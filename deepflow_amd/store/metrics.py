@@ -168,12 +168,16 @@ class RollupTable:
             self.drops = torch.zeros(1, dtype=torch.int64, device=dev)
 
     # RuSpec ABI twin (dfgpu.hip): {u32 interval; u32 n_keys; u8 fam[8];
-    # u8 idx[8]; u8 require_nonzero} padded to 4-byte alignment
+    # u8 idx[8]; u8 require_nonzero; u8 use_lds} padded to 4-byte alignment
     def spec_bytes(self) -> bytes:
         fam = [f for f, _ in self._srcs] + [0] * (RU_MAX_KEYS - len(self._srcs))
         idx = [i for _, i in self._srcs] + [0] * (RU_MAX_KEYS - len(self._srcs))
-        return struct.pack("<II8B8BB3x", self.td.interval_s, len(self._srcs),
-                           *fam, *idx, self._rnz)
+        # LDS pre-aggregation pays off on low-cardinality tables (hot
+        # global slots); the _map tables (cap >= 2^21) mostly miss the
+        # 256-slot block table and go direct
+        use_lds = 1 if self.td.cap_pow2 < 21 else 0
+        return struct.pack("<II8B8BBB2x", self.td.interval_s,
+                           len(self._srcs), *fam, *idx, self._rnz, use_lds)
 
     # ----------------------------------------------------------- update
     def update(self, seg, base: int, n: int, stream: int = 0) -> None:
