@@ -208,9 +208,25 @@ def e2e_main(args) -> None:
                           (1 << 23) + 2)
     n_distinct = min(args.steps + args.warmup, 4)
     batches = gen_batches(cfg, 0, n_distinct, args.batch, pinned=False)
-    frames = [framing.encode_frame(
-        framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
-        b.tobytes()) for (b, _, _, _, _, _) in batches]
+    # agents cap frame sizes (receiver MAX_FRAME = 64 MB): split each
+    # batch into <=32 MB frames on record boundaries
+    FRAME_CAP = 32 << 20
+    frames: list = []   # frames[i] = list of wire frames for batch i
+    for (b, offs, lens, *_rest) in batches:
+        fl = []
+        start = 0
+        i = 0
+        while i < len(offs):
+            j = i
+            while j < len(offs) and \
+                    (offs[j] + lens[j] - offs[i] + 4 * (j - i)) < FRAME_CAP:
+                j += 1
+            payload = b[offs[i] - 4: offs[j - 1] + lens[j - 1]].tobytes()
+            fl.append(framing.encode_frame(
+                framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG),
+                payload))
+            i = j
+        frames.append(fl)
 
     # pinned staging ring: receiver payload bytes -> pinned -> H2D async
     ring = [(torch.empty(max(len(b) for (b, *_ ) in batches),
@@ -251,13 +267,19 @@ def e2e_main(args) -> None:
         s = _socket.create_connection(("127.0.0.1", rx.tcp_port))
         s.setsockopt(_socket.IPPROTO_TCP, _socket.TCP_NODELAY, 1)
         for i in range(k):
-            s.sendall(frames[i % n_distinct])
+            for fr in frames[i % n_distinct]:
+                s.sendall(fr)
         s.close()
 
-    def wait_rows(target: int, timeout=600.0):
+    def wait_rows(target: int, timeout=120.0):
         t_end = time.time() + timeout
         while pipe.stats.spans_in < target and time.time() < t_end:
             time.sleep(0.002)
+        if pipe.stats.spans_in < target:
+            print(f"e2e stall: spans_in={pipe.stats.spans_in} "
+                  f"target={target} rx={rx.counter.snapshot()}",
+                  file=sys.stderr, flush=True)
+            raise SystemExit(3)
         if device == "cuda":
             torch.cuda.synchronize()
 
