@@ -85,28 +85,59 @@ class SpanRouter:
                              device=dev)
         gpu_ops.gather_records(payload_t, offs_t, lens_t, sel_t, dst_off_t,
                                packed)
-        # exchange split sizes: per-destination [count, bytes] pairs
-        send_sizes = torch.from_numpy(
-            np.stack([counts, byte_counts], axis=1).ravel().copy()).to(dev)
-        recv_sizes = torch.empty_like(send_sizes)
-        dist.all_to_all_single(recv_sizes, send_sizes)
-        rs = recv_sizes.view(self.world, 2).cpu().numpy()
-        recv_counts = rs[:, 0].tolist()
-        recv_bytes = rs[:, 1].tolist()
-        # payload bytes
-        out_payload = torch.empty(int(sum(recv_bytes)), dtype=torch.uint8,
-                                  device=dev)
-        dist.all_to_all_single(out_payload, packed,
-                               output_split_sizes=recv_bytes,
-                               input_split_sizes=byte_counts.tolist())
-        # record lens
-        sel_lens_t = torch.from_numpy(
-            sel_lens.astype(np.int32)).to(dev)
-        out_lens = torch.empty(int(sum(recv_counts)), dtype=torch.int32,
-                               device=dev)
-        dist.all_to_all_single(out_lens, sel_lens_t,
-                               output_split_sizes=recv_counts,
-                               input_split_sizes=counts.tolist())
+        from . import rccl
+        if rccl.enabled():
+            # direct librccl: all-gather the per-dest (count, bytes)
+            # matrix, then grouped pairwise send/recv for payload + lens
+            cm = rccl.comm()
+            meta = np.stack([counts, byte_counts], axis=1).ravel().copy()
+            meta_t = torch.from_numpy(meta.view(np.uint8)).to(dev)
+            meta_all = torch.empty(self.world * meta_t.numel(),
+                                   dtype=torch.uint8, device=dev)
+            cm.all_gather_u8(meta_t, meta_all)
+            am = meta_all.cpu().numpy().view(np.int64).reshape(
+                self.world, self.world, 2)
+            recv_counts = [int(am[src, self.rank, 0])
+                           for src in range(self.world)]
+            recv_bytes = [int(am[src, self.rank, 1])
+                          for src in range(self.world)]
+            out_payload = torch.empty(int(sum(recv_bytes)),
+                                      dtype=torch.uint8, device=dev)
+            cm.all_to_all_u8(packed, byte_counts.tolist(), out_payload,
+                             recv_bytes)
+            sel_lens_t = torch.from_numpy(
+                sel_lens.astype(np.int32)).to(dev)
+            out_lens = torch.empty(int(sum(recv_counts)),
+                                   dtype=torch.int32, device=dev)
+            cm.all_to_all_u8(sel_lens_t.view(torch.uint8),
+                             [c * 4 for c in counts.tolist()],
+                             out_lens.view(torch.uint8),
+                             [c * 4 for c in recv_counts])
+            torch.cuda.synchronize()
+        else:
+            # exchange split sizes: per-destination [count, bytes] pairs
+            send_sizes = torch.from_numpy(
+                np.stack([counts, byte_counts],
+                         axis=1).ravel().copy()).to(dev)
+            recv_sizes = torch.empty_like(send_sizes)
+            dist.all_to_all_single(recv_sizes, send_sizes)
+            rs = recv_sizes.view(self.world, 2).cpu().numpy()
+            recv_counts = rs[:, 0].tolist()
+            recv_bytes = rs[:, 1].tolist()
+            # payload bytes
+            out_payload = torch.empty(int(sum(recv_bytes)),
+                                      dtype=torch.uint8, device=dev)
+            dist.all_to_all_single(out_payload, packed,
+                                   output_split_sizes=recv_bytes,
+                                   input_split_sizes=byte_counts.tolist())
+            # record lens
+            sel_lens_t = torch.from_numpy(
+                sel_lens.astype(np.int32)).to(dev)
+            out_lens = torch.empty(int(sum(recv_counts)),
+                                   dtype=torch.int32, device=dev)
+            dist.all_to_all_single(out_lens, sel_lens_t,
+                                   output_split_sizes=recv_counts,
+                                   input_split_sizes=counts.tolist())
         # offsets: blocks arrive in source order, payload likewise — a
         # global exclusive cumsum of lens reconstructs offsets
         cum = torch.cumsum(out_lens.to(torch.int64), 0)

@@ -109,7 +109,8 @@ def test_route_exchange_gloo(world, port):
 
 
 @pytest.mark.gpu
-def test_route_gpu_world1():
+@pytest.mark.parametrize("transport", ["torchdist", "rccl_direct"])
+def test_route_gpu_world1(transport, monkeypatch):
     """GPU routing path on a single rank (RCCL pg, all_to_all_single with
     world=1): packed gather + exchange must reproduce the input records,
     and the routed batch must ingest identically to the unrouted one."""
@@ -118,6 +119,10 @@ def test_route_gpu_world1():
     assert torch.cuda.is_available()
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29741")
+    if transport == "rccl_direct":
+        monkeypatch.setenv("DF_RCCL_DIRECT", "1")
+    else:
+        monkeypatch.delenv("DF_RCCL_DIRECT", raising=False)
     dist.init_process_group("nccl", rank=0, world_size=1)
     try:
         from deepflow_amd.gen import SpanGenConfig
@@ -152,4 +157,8 @@ def test_route_gpu_world1():
         assert QueryEngine(p1, device="cuda").query(q) == \
             QueryEngine(p2, device="cuda").query(q)
     finally:
+        from deepflow_amd.parallel import rccl as _rc
+        if _rc._comm is not None:
+            _rc._comm.close()
+            _rc._comm = None
         dist.destroy_process_group()
