@@ -39,6 +39,38 @@ class PluginHost:
     def register(self, parse: ParseFn) -> None:
         self.plugins.append(parse)
 
+    def load_so(self, path: str) -> None:
+        """dlopen a shared-object plugin (include/df_plugin.h ABI —
+        the reference's plugin/shared_obj counterpart) and register it
+        behind the same parse contract as python plugins."""
+        import ctypes as ct
+
+        class _Info(ct.Structure):
+            _fields_ = [("req_type", ct.c_char * 32),
+                        ("domain", ct.c_char * 128),
+                        ("resource", ct.c_char * 256),
+                        ("endpoint", ct.c_char * 128),
+                        ("status", ct.c_int32), ("code", ct.c_int32)]
+
+        lib = ct.CDLL(path)
+        fn = lib.df_plugin_parse
+        fn.restype = ct.c_int
+        fn.argtypes = [ct.c_char_p, ct.c_uint32, ct.c_uint16,
+                       ct.POINTER(_Info)]
+
+        def parse(raw: bytes, port: int) -> Optional[L7PluginInfo]:
+            info = _Info()
+            if fn(raw, len(raw), port, ct.byref(info)) != 1:
+                return None
+            return L7PluginInfo(
+                req_type=info.req_type.decode("utf-8", "replace"),
+                domain=info.domain.decode("utf-8", "replace"),
+                resource=info.resource.decode("utf-8", "replace"),
+                endpoint=info.endpoint.decode("utf-8", "replace"),
+                status=info.status, code=info.code)
+
+        self.register(parse)
+
     def process_l7_payload(self, payload: bytes) -> bytes:
         """Rewrite custom-protocol (127) records via registered plugins;
         other records pass through untouched."""
