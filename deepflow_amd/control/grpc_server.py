@@ -183,6 +183,8 @@ class GrpcSyncServer:
         conn.settimeout(1.0)
         decoder = HpackDecoder()
         streams: Dict[int, Dict] = {}
+        # concurrent senders (Push/Upgrade stream threads) share the socket
+        send_lock = threading.Lock()
         try:
             pre = self._recv_exact(conn, len(PREFACE))
             if pre != PREFACE:
@@ -199,9 +201,11 @@ class GrpcSyncServer:
                 if payload is None:
                     return
                 if ftype == F_SETTINGS and not flags & FLAG_ACK:
-                    conn.sendall(frame(F_SETTINGS, FLAG_ACK, 0, b""))
+                    with send_lock:
+                        conn.sendall(frame(F_SETTINGS, FLAG_ACK, 0, b""))
                 elif ftype == F_PING and not flags & FLAG_ACK:
-                    conn.sendall(frame(F_PING, FLAG_ACK, 0, payload))
+                    with send_lock:
+                        conn.sendall(frame(F_PING, FLAG_ACK, 0, payload))
                 elif ftype == F_HEADERS:
                     off = 0
                     if flags & 0x08:  # padded
@@ -212,13 +216,13 @@ class GrpcSyncServer:
                     st = streams.setdefault(sid, {"data": b""})
                     st["headers"] = dict(headers)
                     if flags & FLAG_END_STREAM:
-                        self._dispatch(conn, sid, st)
+                        self._dispatch(conn, sid, st, send_lock)
                         streams.pop(sid, None)
                 elif ftype == F_DATA:
                     st = streams.setdefault(sid, {"data": b""})
                     st["data"] += payload
                     if flags & FLAG_END_STREAM:
-                        self._dispatch(conn, sid, st)
+                        self._dispatch(conn, sid, st, send_lock)
                         streams.pop(sid, None)
                 elif ftype == F_GOAWAY:
                     return
@@ -226,12 +230,22 @@ class GrpcSyncServer:
             decoder.close()
             conn.close()
 
-    def _dispatch(self, conn, sid: int, st: Dict) -> None:
+    @staticmethod
+    def _send(conn, lock, data: bytes) -> None:
+        with lock:
+            conn.sendall(data)
+
+    def _dispatch(self, conn, sid: int, st: Dict, lock) -> None:
+        # both the trident and the newer agent service names resolve to
+        # the same handlers (reference: trident.proto + agent.proto
+        # mirror each other, agent_service.go:57-118)
         path = st.get("headers", {}).get(":path", "")
+        path = path.replace("/agent.Synchronizer/", "/trident.Synchronizer/")
         data = st.get("data", b"")
         status = "0"
         body = b""
-        if path == "/trident.Synchronizer/Sync" and len(data) >= 5:
+        if path in ("/trident.Synchronizer/Sync",
+                    "/trident.Synchronizer/AnalyzerSync") and len(data) >= 5:
             mlen = struct.unpack(">I", data[1:5])[0]
             try:
                 req = pb.decode(data[5:5 + mlen], trident.SYNC_REQUEST)
@@ -239,11 +253,78 @@ class GrpcSyncServer:
                 body = grpc_message(pb.encode(resp, trident.SYNC_RESPONSE))
             except Exception:  # noqa: BLE001
                 status = "13"  # INTERNAL
+        elif path == "/trident.Synchronizer/Push" and len(data) >= 5:
+            # server-streaming: versioned SyncResponses pushed whenever
+            # config/platform move (reference trident.proto Push stream —
+            # agents stop polling; round-1 gap #3)
+            mlen = struct.unpack(">I", data[1:5])[0]
+            req = pb.decode(data[5:5 + mlen], trident.SYNC_REQUEST)
+            self._send(conn, lock, frame(F_HEADERS, FLAG_END_HEADERS, sid,
+                                   hpack_encode([
+                                       (":status", "200"),
+                                       ("content-type",
+                                        "application/grpc")])))
+            t = threading.Thread(target=self._push_loop,
+                                 args=(conn, lock, sid, req), daemon=True)
+            t.start()
+            self._threads.append(t)
+            return
+        elif path == "/trident.Synchronizer/Upgrade" and len(data) >= 5:
+            self._send(conn, lock, frame(F_HEADERS, FLAG_END_HEADERS, sid,
+                                   hpack_encode([
+                                       (":status", "200"),
+                                       ("content-type",
+                                        "application/grpc")])))
+            blob = getattr(self.controller, "upgrade_blob", b"")
+            # stream the package in chunks ({content, md5} messages;
+            # reference UpgradeResponse)
+            import hashlib
+            md5 = hashlib.md5(blob).hexdigest() if blob else ""
+            CHUNK = 64 << 10
+            total = (len(blob) + CHUNK - 1) // CHUNK if blob else 0
+            for i in range(total):
+                msg = pb.encode({"status": trident.STATUS_SUCCESS,
+                                 "content": blob[i * CHUNK:(i + 1) * CHUNK],
+                                 "md5": md5,
+                                 "pkt_count": total,
+                                 "total_len": len(blob)},
+                                trident.UPGRADE_RESPONSE)
+                self._send(conn, lock, frame(F_DATA, 0, sid, grpc_message(msg)))
+            self._send(conn, lock, frame(
+                F_HEADERS, FLAG_END_HEADERS | FLAG_END_STREAM, sid,
+                hpack_encode([("grpc-status",
+                               "0" if blob else "5")])))  # 5 = NOT_FOUND
+            return
         else:
             status = "12"  # UNIMPLEMENTED
-        conn.sendall(frame(F_HEADERS, FLAG_END_HEADERS, sid, hpack_encode([
-            (":status", "200"), ("content-type", "application/grpc")])))
+        self._send(conn, lock, frame(F_HEADERS, FLAG_END_HEADERS, sid,
+                               hpack_encode([
+                                   (":status", "200"),
+                                   ("content-type", "application/grpc")])))
         if body:
-            conn.sendall(frame(F_DATA, 0, sid, body))
-        conn.sendall(frame(F_HEADERS, FLAG_END_HEADERS | FLAG_END_STREAM,
-                           sid, hpack_encode([("grpc-status", status)])))
+            self._send(conn, lock, frame(F_DATA, 0, sid, body))
+        self._send(conn, lock, frame(F_HEADERS,
+                               FLAG_END_HEADERS | FLAG_END_STREAM, sid,
+                               hpack_encode([("grpc-status", status)])))
+
+    def _push_loop(self, conn, lock, sid: int, req: Dict) -> None:
+        """Version-gated push: one SyncResponse now, then one whenever
+        config/platform versions move."""
+        ctl = self.controller
+        last = (-1, -1)
+        try:
+            while not self._stop.is_set():
+                cur = (ctl.config_version, ctl.platform_version)
+                if cur != last:
+                    resp = self._handle_sync(dict(req))
+                    self._send(conn, lock, frame(
+                        F_DATA, 0, sid,
+                        grpc_message(pb.encode(resp,
+                                               trident.SYNC_RESPONSE))))
+                    last = cur
+                    # subsequent pushes must carry deltas: pretend the
+                    # agent acked the pushed versions
+                    req["version_platform_data"] = cur[1]
+                self._stop.wait(0.2)
+        except OSError:
+            return

@@ -96,3 +96,13 @@ SYNC_RESPONSE = {
 STATUS_SUCCESS = 0
 STATUS_FAILED = 1
 STATUS_HEARTBEAT = 2
+
+
+# Upgrade stream chunk (trident.proto UpgradeResponse)
+UPGRADE_RESPONSE = {
+    1: ("status", 'u'),
+    2: ("content", 'b'),
+    3: ("md5", 's'),
+    4: ("pkt_count", 'u'),
+    5: ("total_len", 'u'),
+}

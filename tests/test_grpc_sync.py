@@ -86,3 +86,69 @@ def test_grpc_unknown_method():
         s.close()
     finally:
         srv.stop()
+
+
+def test_grpc_push_stream():
+    """Push: the server streams a SyncResponse immediately and another
+    when the platform version moves (reference trident.proto Push —
+    agents stop polling)."""
+    import threading
+    import time
+    from deepflow_amd.agent.grpc_client import grpc_push
+    from deepflow_amd.store.kg import KgInfo
+    ctl = ControllerLite()
+    srv = GrpcSyncServer(ctl)
+    srv.start()
+    try:
+        got = []
+
+        def consume():
+            for msg in grpc_push("127.0.0.1", srv.port,
+                                 {"ctrl_ip": "10.0.0.9",
+                                  "ctrl_mac": "aa:bb"}, max_msgs=2):
+                got.append(msg)
+        t = threading.Thread(target=consume, daemon=True)
+        t.start()
+        time.sleep(0.6)
+        assert len(got) == 1                      # initial push
+        ctl.update_platform({(7, 0x0A000009): KgInfo(pod_id=42)})
+        t.join(timeout=8)
+        assert len(got) == 2                      # version-change push
+        assert "platform_data" in got[1]
+    finally:
+        srv.stop()
+
+
+def test_grpc_upgrade_stream():
+    import hashlib
+    from deepflow_amd.agent.grpc_client import grpc_upgrade
+    ctl = ControllerLite()
+    ctl.upgrade_blob = bytes(range(256)) * 1024   # 256 KiB "package"
+    srv = GrpcSyncServer(ctl)
+    srv.start()
+    try:
+        blob = grpc_upgrade("127.0.0.1", srv.port, {"ctrl_ip": "10.0.0.1"})
+        assert blob == ctl.upgrade_blob
+    finally:
+        srv.stop()
+
+
+def test_agent_service_alias():
+    """The newer agent.Synchronizer service name maps to the same
+    handlers (reference agent.proto mirrors trident.proto)."""
+    from deepflow_amd.agent.grpc_client import grpc_stream
+    from deepflow_amd.wire import trident as T
+    ctl = ControllerLite()
+    srv = GrpcSyncServer(ctl)
+    srv.start()
+    try:
+        msgs = list(grpc_stream("127.0.0.1", srv.port,
+                                "/agent.Synchronizer/Push",
+                                {"ctrl_ip": "1.2.3.4"},
+                                T.SYNC_REQUEST, T.SYNC_RESPONSE,
+                                max_msgs=1))
+        # proto3 zero-skip drops status=SUCCESS; the initial push always
+        # carries the config payload (agent versions start at 0)
+        assert msgs and "config" in msgs[0]
+    finally:
+        srv.stop()
