@@ -11,6 +11,7 @@
 // Document payload records, length-prefixed for trident framing).
 // Packet sources: callers feed raw frames (tests use synthetic packets;
 // AF_PACKET capture wiring is the host deployment's concern).
+#include <algorithm>
 #include <cerrno>
 #include <cstdint>
 #include <cstdio>
@@ -154,6 +155,142 @@ struct AclRule {
     uint32_t action = 0;     // bit0: pcap-capture tag
 };
 
+// DDBS-style first path (reference policy/first_path.rs): every ACL owns
+// a bit; each DIMENSION (proto, server-port interval, src/dst prefix)
+// precomputes which ACL bits it can satisfy, and a lookup ANDs the
+// per-dimension bitmaps — cost grows with dimension table sizes, not
+// with |acls| x |flows|. The LRU fast path (policy/fast_path.rs analog)
+// caches the final bitmap per (src, dst, proto, port) tuple.
+struct PolicyBitmap {
+    std::vector<uint64_t> w;
+    void ensure(size_t nbits) { w.assign((nbits + 63) / 64, 0); }
+    void set(size_t i) { w[i / 64] |= 1ull << (i % 64); }
+    static void and_into(std::vector<uint64_t>& acc,
+                         const std::vector<uint64_t>& other) {
+        for (size_t i = 0; i < acc.size(); i++) acc[i] &= other[i];
+    }
+};
+
+struct FirstPath {
+    size_t n = 0;
+    // proto dimension: exact byte (0 = any handled by baseline bitmap)
+    PolicyBitmap proto_any;
+    std::map<uint8_t, PolicyBitmap> proto_eq;
+    // port dimension: sorted interval boundaries -> bitmap per segment
+    std::vector<uint32_t> port_bounds;          // segment starts
+    std::vector<std::vector<uint64_t>> port_bm;  // per segment
+    // net dimensions: (net, mask) -> bit list (prefix count is small;
+    // the bitmap combine is the DDBS win)
+    struct NetEnt { uint32_t net, mask; PolicyBitmap bm; };
+    std::vector<NetEnt> src_nets, dst_nets;
+
+    void rebuild(const std::vector<AclRule>& acls) {
+        n = acls.size();
+        proto_any.ensure(n);
+        proto_eq.clear();
+        src_nets.clear();
+        dst_nets.clear();
+        // port segmentation from all rule boundaries
+        std::vector<uint32_t> bounds = {0, 65536};
+        for (const auto& r : acls) {
+            bounds.push_back(r.port_min);
+            bounds.push_back((uint32_t)r.port_max + 1);
+        }
+        std::sort(bounds.begin(), bounds.end());
+        bounds.erase(std::unique(bounds.begin(), bounds.end()),
+                     bounds.end());
+        port_bounds.assign(bounds.begin(), bounds.end() - 1);
+        port_bm.assign(port_bounds.size(),
+                       std::vector<uint64_t>((n + 63) / 64, 0));
+        auto net_bit = [&](std::vector<NetEnt>& v, uint32_t net,
+                           uint32_t mask, size_t bit) {
+            for (auto& e : v)
+                if (e.net == net && e.mask == mask) { e.bm.set(bit); return; }
+            v.push_back({net, mask, {}});
+            v.back().bm.ensure(n);
+            v.back().bm.set(bit);
+        };
+        for (size_t i = 0; i < n; i++) {
+            const AclRule& r = acls[i];
+            if (r.proto == 0) proto_any.set(i);
+            else {
+                auto& bm = proto_eq[r.proto];
+                if (bm.w.empty()) bm.ensure(n);
+                bm.set(i);
+            }
+            for (size_t s = 0; s < port_bounds.size(); s++) {
+                uint32_t lo = port_bounds[s];
+                if (lo >= r.port_min && lo <= r.port_max)
+                    port_bm[s][i / 64] |= 1ull << (i % 64);
+            }
+            net_bit(src_nets, r.src_net, r.src_mask, i);
+            net_bit(dst_nets, r.dst_net, r.dst_mask, i);
+        }
+    }
+
+    // matched ACL bitmap for one direction
+    std::vector<uint64_t> lookup(uint32_t src, uint32_t dst, uint8_t proto,
+                                 uint16_t port) const {
+        std::vector<uint64_t> acc((n + 63) / 64, 0);
+        if (n == 0) return acc;
+        // proto dim
+        std::vector<uint64_t> dim = proto_any.w;
+        auto it = proto_eq.find(proto);
+        if (it != proto_eq.end())
+            for (size_t i = 0; i < dim.size(); i++) dim[i] |= it->second.w[i];
+        acc = dim;
+        // port dim: binary search the segment
+        size_t lo = 0, hi = port_bounds.size();
+        while (lo + 1 < hi) {
+            size_t mid = (lo + hi) / 2;
+            if (port_bounds[mid] <= port) lo = mid; else hi = mid;
+        }
+        PolicyBitmap::and_into(acc, port_bm[lo]);
+        // net dims: union of matching prefixes per side, then AND
+        std::vector<uint64_t> sdim(acc.size(), 0), ddim(acc.size(), 0);
+        for (const auto& e : src_nets)
+            if ((src & e.mask) == e.net)
+                for (size_t i = 0; i < sdim.size(); i++) sdim[i] |= e.bm.w[i];
+        for (const auto& e : dst_nets)
+            if ((dst & e.mask) == e.net)
+                for (size_t i = 0; i < ddim.size(); i++) ddim[i] |= e.bm.w[i];
+        PolicyBitmap::and_into(acc, sdim);
+        PolicyBitmap::and_into(acc, ddim);
+        return acc;
+    }
+};
+
+// fast path: LRU of (src, dst, proto, port) -> matched bitmap
+struct FastPath {
+    struct Ent {
+        uint64_t key;
+        std::vector<uint64_t> bm;
+        uint64_t tick;
+    };
+    static constexpr size_t CAP = 1 << 14;
+    std::unordered_map<uint64_t, Ent> map;
+    uint64_t tick = 0;
+    uint64_t hits = 0, misses = 0;
+
+    const std::vector<uint64_t>* get(uint64_t key) {
+        auto it = map.find(key);
+        if (it == map.end()) { misses++; return nullptr; }
+        hits++;
+        it->second.tick = ++tick;
+        return &it->second.bm;
+    }
+    void put(uint64_t key, std::vector<uint64_t> bm) {
+        if (map.size() >= CAP) {  // evict oldest (approximate LRU sweep)
+            auto victim = map.begin();
+            for (auto it = map.begin(); it != map.end(); ++it)
+                if (it->second.tick < victim->second.tick) victim = it;
+            map.erase(victim);
+        }
+        map[key] = Ent{key, std::move(bm), ++tick};
+    }
+    void clear() { map.clear(); }
+};
+
 struct Agent {
     uint32_t vtap_id;
     uint64_t next_flow_id = 1;
@@ -161,6 +298,8 @@ struct Agent {
     std::vector<Cidr> cidrs;
     std::vector<uint16_t> custom_ports;  // port-rule custom protocols (127)
     std::vector<AclRule> acls;
+    FirstPath first_path;
+    FastPath fast_path;
     std::map<MeterKey, AppMeterAcc> meters;
     std::vector<uint8_t> out_l4, out_l7, out_doc, out_pcap;
     // stats
@@ -168,18 +307,28 @@ struct Agent {
              docs_emitted = 0, parse_errors = 0;
 };
 
-// first-path ACL match for a new flow (client = peer 0)
-void match_acls(const Agent& a, FlowNode& f) {
-    for (const auto& r : a.acls) {
-        bool fwd = (f.ip[0] & r.src_mask) == r.src_net &&
-                   (f.ip[1] & r.dst_mask) == r.dst_net;
-        bool rev = (f.ip[1] & r.src_mask) == r.src_net &&
-                   (f.ip[0] & r.dst_mask) == r.dst_net;
-        if (!fwd && !rev) continue;
-        if (r.proto && r.proto != f.proto) continue;
-        if (f.port[1] < r.port_min || f.port[1] > r.port_max) continue;
-        if (f.acl_gids.size() < 8) f.acl_gids.push_back(r.gid);
-        f.acl_actions |= r.action;
+// policy lookup for a new flow: fast-path LRU, then the DDBS first path
+// in both directions (reference Policy::lookup, policy/policy.rs:283)
+void match_acls(Agent& a, FlowNode& f) {
+    if (a.acls.empty()) return;
+    uint64_t key = ((uint64_t)(f.ip[0] ^ (f.ip[1] * 0x9E3779B9u)) << 32) |
+                   ((uint64_t)f.port[1] << 8) | f.proto;
+    const std::vector<uint64_t>* cached = a.fast_path.get(key);
+    std::vector<uint64_t> bm;
+    if (cached != nullptr) {
+        bm = *cached;
+    } else {
+        bm = a.first_path.lookup(f.ip[0], f.ip[1], f.proto, f.port[1]);
+        std::vector<uint64_t> rev =
+            a.first_path.lookup(f.ip[1], f.ip[0], f.proto, f.port[1]);
+        for (size_t i = 0; i < bm.size(); i++) bm[i] |= rev[i];
+        a.fast_path.put(key, bm);
+    }
+    for (size_t i = 0; i < a.acls.size(); i++) {
+        if (bm[i / 64] & (1ull << (i % 64))) {
+            if (f.acl_gids.size() < 8) f.acl_gids.push_back(a.acls[i].gid);
+            f.acl_actions |= a.acls[i].action;
+        }
     }
 }
 
@@ -2343,6 +2492,8 @@ void dfa_add_acl(void* h, uint32_t gid, uint32_t src_net,
     r.port_max = (uint16_t)port_max;
     r.action = action;
     a->acls.push_back(r);
+    a->first_path.rebuild(a->acls);   // DDBS dimension tables
+    a->fast_path.clear();             // cached verdicts are stale
 }
 
 void dfa_add_cidr(void* h, uint32_t net, uint32_t masklen, int32_t epc) {
