@@ -74,6 +74,12 @@ class ControllerLite:
         self.genesis_inventory: Dict[int, dict] = {}
         self._gpid_by_key: Dict[Tuple[int, int], int] = {}
         self._next_gpid = 1
+        # prometheus global label encoder (reference
+        # controller/prometheus: persistent metric/label-name/value ids
+        # served to ingesters via GetPrometheusLabelIDs; metadb-backed
+        # there, checkpoint-persisted here)
+        self.prom_ids: Dict[str, Dict[str, int]] = {
+            "metric": {}, "label_name": {}, "label_value": {}}
 
     # ------------------------------------------------------------ sync
     def sync(self, agent_id: int, hostname: str = "", ip: str = "",
@@ -162,6 +168,25 @@ class ControllerLite:
     def lookup_gpid(self, agent_id: int, pid: int) -> int:
         return self._gpid_by_key.get((agent_id, pid), 0)
 
+    # ------------------------------------------------------- prometheus
+    def alloc_prom_ids(self, batch: Dict[str, List[str]]
+                       ) -> Dict[str, Dict[str, int]]:
+        """GetPrometheusLabelIDs analog: allocate (or return) persistent
+        ids for metric names / label names / label values. Ids never
+        change once issued (they survive restarts via the checkpoint)."""
+        out: Dict[str, Dict[str, int]] = {}
+        for kind, strings in batch.items():
+            table = self.prom_ids[kind]
+            res = {}
+            for s in strings:
+                i = table.get(s)
+                if i is None:
+                    i = len(table) + 1      # 0 reserved
+                    table[s] = i
+                res[s] = i
+            out[kind] = res
+        return out
+
     # ------------------------------------------------------- persistence
     # (reference keeps this state in MySQL metadb; here the registry,
     # platform inventory, name maps and GPID allocations serialize into
@@ -180,6 +205,7 @@ class ControllerLite:
             "genesis_inventory": dict(self.genesis_inventory),
             "gpid_by_key": dict(self._gpid_by_key),
             "next_gpid": self._next_gpid,
+            "prom_ids": {k: dict(v) for k, v in self.prom_ids.items()},
         }
 
     def load_state_dict(self, state: dict) -> None:
@@ -200,6 +226,8 @@ class ControllerLite:
         self._gpid_by_key = {tuple(k): v
                              for k, v in state["gpid_by_key"].items()}
         self._next_gpid = state["next_gpid"]
+        for k, v in state.get("prom_ids", {}).items():
+            self.prom_ids[k].update(v)
 
     # ------------------------------------------------------------ monitor
     def agent_status(self, stale_after_s: float = 60.0) -> List[dict]:
@@ -262,6 +290,13 @@ class ControllerLite:
         async def set_config(group: str, request: Request):
             self.set_group_config(group, await request.json())
             return {"status": "ok", "config_version": self.config_version}
+
+        @app.post("/v1/prometheus/label-ids/")
+        async def prom_label_ids(request: Request):
+            body = await request.json()
+            return self.alloc_prom_ids(
+                {k: list(v) for k, v in body.items()
+                 if k in ("metric", "label_name", "label_value")})
 
         @app.post("/v1/rebalance/")
         async def rebalance(request: Request):

@@ -23,6 +23,12 @@ class Interner:
         self.to_id: Dict[str, int] = {}
         self.from_id: List[str] = []
 
+    def name(self, i: int) -> str:
+        return self.from_id[i]
+
+    def strings(self):
+        return iter(self.from_id)
+
     def intern(self, s: str) -> int:
         i = self.to_id.get(s)
         if i is None:
@@ -32,11 +38,46 @@ class Interner:
         return i
 
 
+class GlobalInterner(Interner):
+    """Interner backed by the controller\'s persistent id allocator
+    (reference: grpc_label_ids.go slow path -> GetPrometheusLabelIDs).
+    Local cache, controller miss-fill; ids survive restarts."""
+
+    def __init__(self, kind: str, alloc_fn):
+        super().__init__()
+        self.kind = kind
+        self.alloc_fn = alloc_fn
+        self.from_id_map: Dict[int, str] = {}
+
+    def name(self, i: int) -> str:
+        return self.from_id_map[i]
+
+    def strings(self):
+        return iter(self.from_id_map.values())
+
+    def intern(self, s: str) -> int:
+        i = self.to_id.get(s)
+        if i is None:
+            i = self.alloc_fn({self.kind: [s]})[self.kind][s]
+            self.to_id[s] = i
+            self.from_id_map[i] = s
+        return i
+
+
 class PromPipeline:
-    def __init__(self, counter: Optional[Counter] = None):
-        self.metric_names = Interner()
-        self.label_names = Interner()
-        self.label_values = Interner()
+    def __init__(self, counter: Optional[Counter] = None,
+                 id_allocator=None):
+        """id_allocator: ControllerLite.alloc_prom_ids-shaped callable —
+        when given, metric/label ids are controller-global + persistent
+        instead of per-shard volatile (round-1 weakness #48)."""
+        if id_allocator is not None:
+            self.metric_names = GlobalInterner("metric", id_allocator)
+            self.label_names = GlobalInterner("label_name", id_allocator)
+            self.label_values = GlobalInterner("label_value", id_allocator)
+        else:
+            self.metric_names = Interner()
+            self.label_names = Interner()
+            self.label_values = Interner()
         # series: (metric_id, ((lname_id, lval_id), ...)) -> series_id
         self.series: Dict[Tuple, int] = {}
         self.series_labels: List[Tuple] = []
@@ -110,8 +151,8 @@ class PromPipeline:
         for sid, smid in enumerate(self.series_metric):
             if smid != mid:
                 continue
-            labels = {self.label_names.from_id[ln]:
-                      self.label_values.from_id[lv]
+            labels = {self.label_names.name(ln):
+                      self.label_values.name(lv)
                       for ln, lv in self.series_labels[sid]}
             ok = True
             for lname, op, lval in matchers:
@@ -140,9 +181,9 @@ class PromPipeline:
     def stored_bytes(self) -> int:
         """SmartEncoding accounting: ID-encoded samples + dictionaries."""
         samples = len(self.s_series) * (4 + 8 + 8)
-        dicts = sum(len(s) for s in self.metric_names.from_id) + \
-            sum(len(s) for s in self.label_names.from_id) + \
-            sum(len(s) for s in self.label_values.from_id)
+        dicts = sum(len(s) for s in self.metric_names.strings()) + \
+            sum(len(s) for s in self.label_names.strings()) + \
+            sum(len(s) for s in self.label_values.strings())
         layout = sum(2 * 4 * len(t) for t in self.series_labels)
         return samples + dicts + layout
 
@@ -152,9 +193,9 @@ class PromPipeline:
         for i, sid in enumerate(self.s_series):
             row = 8 + 8
             mid = self.series_metric[sid]
-            row += len(self.metric_names.from_id[mid])
+            row += len(self.metric_names.name(mid))
             for ln, lv in self.series_labels[sid]:
-                row += len(self.label_names.from_id[ln]) + \
-                    len(self.label_values.from_id[lv])
+                row += len(self.label_names.name(ln)) + \
+                    len(self.label_values.name(lv))
             total += row
         return total

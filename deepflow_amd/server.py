@@ -108,6 +108,16 @@ class DeepflowServer:
         from .control import ControllerLite
         self.controller = ControllerLite(
             kg=self.kg, event_sink=self.events.add_resource_event)
+        # controller-global persistent prometheus label ids (reference
+        # GetPrometheusLabelIDs): rebuild the pipeline's interners over
+        # the allocator now that the controller exists
+        from .ingest.prom_pipeline import GlobalInterner
+        self.prom.metric_names = GlobalInterner(
+            "metric", self.controller.alloc_prom_ids)
+        self.prom.label_names = GlobalInterner(
+            "label_name", self.controller.alloc_prom_ids)
+        self.prom.label_values = GlobalInterner(
+            "label_value", self.controller.alloc_prom_ids)
         self.system_rows = []  # deepflow_system self-metrics store
         self.receiver.register(framing.MSG_DFSTATS, self._on_dfstats)
         self.engine.system_rows = self.system_rows

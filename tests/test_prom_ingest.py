@@ -55,3 +55,31 @@ def test_dfstats_self_loop(server):
         "sql": "SELECT table, spans_in FROM deepflow_system LIMIT 50"})
     body = r.json()
     assert body["OPT_STATUS"] == "SUCCESS", body
+
+
+def test_controller_global_label_ids_persist():
+    """Prometheus label ids are controller-allocated and survive a
+    controller restart (reference: persistent metadb ids served via
+    GetPrometheusLabelIDs; round-1 ids were per-shard volatile)."""
+    from deepflow_amd.control import ControllerLite
+    from deepflow_amd.ingest.prom_pipeline import PromPipeline
+    ctl = ControllerLite()
+    pipe = PromPipeline(id_allocator=ctl.alloc_prom_ids)
+    pipe.ingest_labeled_samples([
+        ("http_requests_total", {"job": "api", "code": "200"}, 1000, 5.0),
+        ("http_requests_total", {"job": "api", "code": "500"}, 1000, 1.0),
+    ])
+    mid = pipe.metric_names.to_id["http_requests_total"]
+    vid = pipe.label_values.to_id["500"]
+    assert mid >= 1 and vid >= 1
+    # restart: state roundtrip through the checkpoint dict
+    ctl2 = ControllerLite()
+    ctl2.load_state_dict(ctl.state_dict())
+    pipe2 = PromPipeline(id_allocator=ctl2.alloc_prom_ids)
+    pipe2.ingest_labeled_samples([
+        ("http_requests_total", {"job": "api", "code": "500"}, 2000, 2.0)])
+    assert pipe2.metric_names.to_id["http_requests_total"] == mid
+    assert pipe2.label_values.to_id["500"] == vid
+    # series still hydrate through the global interners
+    series = pipe2.series_for("http_requests_total", [])
+    assert series and series[0]["metric"]["job"] == "api"
