@@ -104,3 +104,55 @@ class CloudPoller:
         self._stop.set()
         if self._thread:
             self._thread.join(timeout=2)
+
+
+def filereader_source(path: str) -> Callable[[], Dict]:
+    """The reference's `filereader` cloud provider
+    (controller/cloud/filereader): the inventory lives in a YAML/JSON
+    file an operator (or external exporter) maintains; each poll reads
+    the file and the normal snapshot-diff machinery applies changes.
+    Supports the k8s-style snapshot shape plus a flat `hosts:` list of
+    {name, ip, epc, host_id, az}."""
+    import json
+    import os
+
+    def load() -> Dict:
+        with open(path) as f:
+            text = f.read()
+        if path.endswith((".yaml", ".yml")):
+            import yaml
+            snap = yaml.safe_load(text) or {}
+        else:
+            snap = json.loads(text or "{}")
+        return snap
+    return load
+
+
+def filereader_snapshot_to_platform(snap: Dict) -> Tuple[Dict, Dict]:
+    """Normalize a filereader inventory: k8s-shaped sections reuse the
+    k8s normalizer; a flat `hosts:` section maps to host/az entries."""
+    entries, names = k8s_snapshot_to_platform(snap)
+    names.setdefault("host", {})
+    names.setdefault("az", {})
+    az_ids: Dict[str, int] = {}
+    for i, h in enumerate(snap.get("hosts", []), start=1):
+        ip = int(ipaddress.IPv4Address(h["ip"]))
+        az = h.get("az", "")
+        az_id = az_ids.setdefault(az, len(az_ids) + 1) if az else 0
+        host_id = h.get("host_id", i)
+        entries[(h.get("epc", 0), ip)] = KgInfo(
+            host_id=host_id, az_id=az_id,
+            l3_device_id=h.get("device_id", host_id), l3_device_type=2)
+        names["host"][host_id] = h["name"]
+        if az:
+            names["az"][az_id] = az
+    return entries, names
+
+
+class FileReaderProvider(CloudPoller):
+    """CloudPoller over a local inventory file (reference filereader)."""
+
+    def __init__(self, controller, path: str, interval_s: float = 30.0):
+        super().__init__(controller, filereader_source(path),
+                         normalize=filereader_snapshot_to_platform,
+                         interval_s=interval_s)

@@ -108,3 +108,34 @@ def test_pod_name_hydration_and_filter():
         "SELECT Count(*) AS c FROM l7_flow_log "
         "WHERE pod_name_1 = 'no-such-pod'")
     assert r3["values"] in ([[0]], [])
+
+
+def test_filereader_cloud_provider(tmp_path):
+    """The filereader provider (reference controller/cloud/filereader):
+    operator-maintained inventory file -> snapshot diff -> platform/KG."""
+    import yaml
+    from deepflow_amd.control import ControllerLite
+    from deepflow_amd.control.cloud import FileReaderProvider
+    from deepflow_amd.store.kg import KnowledgeGraphTable
+    inv = tmp_path / "inventory.yaml"
+    inv.write_text(yaml.safe_dump({
+        "cluster_id": 5, "cluster_name": "edge-1",
+        "pods": [{"name": "web-0", "ip": "10.9.0.4", "epc": 2, "id": 31}],
+        "hosts": [{"name": "bm-01.dc", "ip": "10.9.1.1", "epc": 2,
+                   "host_id": 7, "az": "az-east-1"}],
+    }))
+    kg = KnowledgeGraphTable(device="cpu")
+    ctl = ControllerLite(kg=kg)
+    prov = FileReaderProvider(ctl, str(inv), interval_s=30)
+    assert prov.poll_once()
+    assert kg.lookup(2, 0x0A090004).pod_id == 31
+    assert kg.lookup(2, 0x0A090101).host_id == 7
+    assert ctl.lookup_name("host", 7) == "bm-01.dc"
+    # unchanged file -> no push; edited file -> diff push
+    assert not prov.poll_once()
+    data = yaml.safe_load(inv.read_text())
+    data["hosts"].append({"name": "bm-02.dc", "ip": "10.9.1.2", "epc": 2,
+                          "host_id": 8, "az": "az-east-1"})
+    inv.write_text(yaml.safe_dump(data))
+    assert prov.poll_once()
+    assert kg.lookup(2, 0x0A090102).host_id == 8
