@@ -24,6 +24,12 @@ ncclSuccess = 0
 ncclInt8 = 0  # ncclDataType_t
 
 
+class NcclUniqueId(ct.Structure):
+    # passed BY VALUE to ncclCommInitRank (a bare ctypes char array
+    # would decay to a pointer and the init fails with invalid argument)
+    _fields_ = [("internal", ct.c_char * NCCL_UNIQUE_ID_BYTES)]
+
+
 def _find_librccl() -> str:
     cand = os.path.join(os.path.dirname(torch.__file__), "lib",
                         "librccl.so")
@@ -42,7 +48,8 @@ def lib() -> ct.CDLL:
         h.ncclGetUniqueId.restype = ct.c_int
         h.ncclGetUniqueId.argtypes = [ct.c_void_p]
         h.ncclCommInitRank.restype = ct.c_int
-        h.ncclCommInitRank.argtypes = [ct.c_void_p, ct.c_int, ct.c_char * NCCL_UNIQUE_ID_BYTES, ct.c_int]
+        h.ncclCommInitRank.argtypes = [ct.c_void_p, ct.c_int,
+                                       NcclUniqueId, ct.c_int]
         h.ncclCommDestroy.restype = ct.c_int
         h.ncclCommDestroy.argtypes = [ct.c_void_p]
         h.ncclGroupStart.restype = ct.c_int
@@ -83,12 +90,12 @@ class RcclComm:
         h = lib()
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
-        uid = (ct.c_char * NCCL_UNIQUE_ID_BYTES)()
+        uid = NcclUniqueId()
         if self.rank == 0:
             _check(h.ncclGetUniqueId(ct.byref(uid)), "ncclGetUniqueId")
-        blob = [bytes(uid.raw)]
+        blob = [bytes(uid.internal)]
         dist.broadcast_object_list(blob, src=0)
-        uid = (ct.c_char * NCCL_UNIQUE_ID_BYTES).from_buffer_copy(blob[0])
+        uid = NcclUniqueId.from_buffer_copy(blob[0])
         self.comm = ct.c_void_p()
         _check(h.ncclCommInitRank(ct.byref(self.comm), self.world, uid,
                                   self.rank), "ncclCommInitRank")
