@@ -93,7 +93,8 @@ class RcclComm:
         uid = NcclUniqueId()
         if self.rank == 0:
             _check(h.ncclGetUniqueId(ct.byref(uid)), "ncclGetUniqueId")
-        blob = [bytes(uid.internal)]
+        # c_char fields NUL-truncate on attribute access; take raw bytes
+        blob = [ct.string_at(ct.byref(uid), NCCL_UNIQUE_ID_BYTES)]
         dist.broadcast_object_list(blob, src=0)
         uid = NcclUniqueId.from_buffer_copy(blob[0])
         self.comm = ct.c_void_p()
