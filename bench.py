@@ -263,13 +263,22 @@ def e2e_main(args) -> None:
     rx.start()
     import socket as _socket
 
+    n_streams = max(1, int(os.environ.get("DF_E2E_STREAMS", "2")))
+
     def send_frames(k: int) -> None:
-        s = _socket.create_connection(("127.0.0.1", rx.tcp_port))
-        s.setsockopt(_socket.IPPROTO_TCP, _socket.TCP_NODELAY, 1)
+        # N parallel sender connections (agents are many); each batch's
+        # sub-frames round-robin across streams
+        socks = []
+        for _ in range(n_streams):
+            s = _socket.create_connection(("127.0.0.1", rx.tcp_port))
+            s.setsockopt(_socket.IPPROTO_TCP, _socket.TCP_NODELAY, 1)
+            socks.append(s)
         for i in range(k):
-            for fr in frames[i % n_distinct]:
-                s.sendall(fr)
-        s.close()
+            fl = frames[i % n_distinct]
+            for j, fr in enumerate(fl):
+                socks[j % n_streams].sendall(fr)
+        for s in socks:
+            s.close()
 
     def wait_rows(target: int, timeout=120.0):
         t_end = time.time() + timeout

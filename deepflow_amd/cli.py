@@ -102,6 +102,55 @@ def agent_rebalance(ctx):
     click.echo(json.dumps(r.json()))
 
 
+@agent.command("run")
+@click.option("--server", "ingest_server", default="127.0.0.1:20033",
+              help="ingest host:port (trident framed protocol)")
+@click.option("--iface", default="lo", help="capture interface")
+@click.option("--vtap-id", default=1, type=int)
+@click.option("--ring/--no-ring", default=True,
+              help="TPACKET_V3 block ring vs per-packet AF_PACKET")
+@click.option("--ebpf/--no-ebpf", default=True,
+              help="attach the eBPF socket tracer when permitted")
+@click.option("--flush-interval", default=1.0, type=float)
+def agent_run(ingest_server, iface, vtap_id, ring, ebpf, flush_interval):
+    """Run the collection agent: capture -> FlowMap/L7 parse -> framed
+    sender (the deployment entry the manifests launch)."""
+    import time as _t
+    from deepflow_amd.agent import Agent
+    host, port = ingest_server.rsplit(":", 1)
+    a = Agent(vtap_id=vtap_id, server=(host, int(port)))
+    cap = None
+    try:
+        if ring:
+            from deepflow_amd.agent.capture import RingCapture
+            cap = RingCapture(a, iface=iface)
+        else:
+            from deepflow_amd.agent.capture import CaptureWorker
+            cap = CaptureWorker(a, iface=iface)
+        cap.start()
+        click.echo(f"capture on {iface} "
+                   f"({'tpacket_v3 ring' if ring else 'af_packet'})")
+    except (PermissionError, OSError) as e:
+        click.echo(f"capture unavailable ({e}); running without packets")
+    tracer = a.start_ebpf() if ebpf else None
+    click.echo(f"ebpf socket tracer: {'on' if tracer else 'unavailable'}")
+    click.echo(f"sending to {ingest_server}; ctrl-c to stop")
+    try:
+        while True:
+            _t.sleep(flush_interval)
+            try:
+                a.flush_to_server(_t.time_ns(), compress=True)
+            except OSError as e:
+                click.echo(f"send failed: {e}; retrying")
+    except KeyboardInterrupt:
+        pass
+    finally:
+        if cap is not None:
+            cap.stop()
+        a.stop_ebpf()
+        a.close()
+
+
 @cli.command()
 @click.pass_context
 def stats(ctx):

@@ -83,3 +83,54 @@ def test_cli_debug_bus(tmp_path):
         assert '"pong": true' in r.output
     finally:
         bus.stop()
+
+
+def test_agent_run_live(tmp_path):
+    """`cli agent run` end to end: subprocess agent captures loopback
+    traffic with the ring and ships it to a live server; the span shows
+    up via SQL."""
+    import socket
+    import subprocess
+    import sys
+    import time
+    import threading
+    from deepflow_amd.server import DeepflowServer
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12, time_base_s=0)
+    srv.start()
+    try:
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "deepflow_amd.cli", "agent", "run",
+             "--server", f"127.0.0.1:{srv.receiver.tcp_port}",
+             "--iface", "lo", "--vtap-id", "9", "--no-ebpf",
+             "--flush-interval", "0.3"],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        time.sleep(1.2)
+        # real HTTP round trip on loopback for the agent to capture
+        s = socket.socket()
+        s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        s.bind(("127.0.0.1", 0))
+        s.listen(1)
+        port = s.getsockname()[1]
+        cli_s = socket.create_connection(("127.0.0.1", port))
+        conn, _ = s.accept()
+        cli_s.sendall(b"GET /cli/run HTTP/1.1\r\nHost: cli.test\r\n\r\n")
+        conn.recv(4096)
+        conn.sendall(b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+        cli_s.recv(4096)
+        cli_s.close()
+        conn.close()
+        s.close()
+        deadline = time.time() + 15
+        found = False
+        while time.time() < deadline and not found:
+            time.sleep(0.5)
+            r = srv.engine.query(
+                "SELECT request_resource FROM l7_flow_log "
+                "WHERE request_domain = 'cli.test' LIMIT 5")
+            found = bool(r["values"])
+        proc.terminate()
+        out = proc.communicate(timeout=10)[0]
+        assert found, out
+    finally:
+        srv.stop()
