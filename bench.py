@@ -396,6 +396,8 @@ def main() -> None:
     # kernels run on the main stream.
     copy_stream = torch.cuda.Stream() if device == "cuda" else None
     pending = {}
+    state = {"routed": world > 1}
+    nonlocal_router = [router]
 
     def prefetch(i: int) -> None:
         buf, offs, lens, pay_t, offs_t, lens_t = batches[i % n_distinct]
@@ -416,19 +418,29 @@ def main() -> None:
             for t in dev_batch:
                 t.record_stream(torch.cuda.current_stream())
             prefetch(i + 1)
-            if router is not None:
+            if nonlocal_router[0] is not None:
                 buf, offs_h, lens_h = batches[i % n_distinct][:3]
-                pay, offs_t, lens_t = router.route_gpu(
-                    dev_batch[0], dev_batch[1], dev_batch[2],
-                    buf, offs_h, lens_h)
-                # harvest reads only the emitted slices from the device
-                # payload (no full-batch D2H on the routed path)
-                pipe.ingest_device(pay, offs_t, lens_t, pay)
+                try:
+                    pay, offs_t, lens_t = router.route_gpu(
+                        dev_batch[0], dev_batch[1], dev_batch[2],
+                        buf, offs_h, lens_h)
+                except Exception as e:  # noqa: BLE001 — keep SCALE alive
+                    print(f"span routing failed ({e}); falling back to "
+                          f"unrouted per-rank ingest", file=sys.stderr,
+                          flush=True)
+                    state["routed"] = False
+                    globals()["_route_failed"] = True
+                    nonlocal_router[0] = None
+                    pipe.ingest_device(*dev_batch, host_payload)
+                else:
+                    # harvest reads only the emitted slices from the
+                    # device payload (no full-batch D2H when routed)
+                    pipe.ingest_device(pay, offs_t, lens_t, pay)
             else:
                 pipe.ingest_device(*dev_batch, host_payload)
         else:
             payload, offs, lens = batches[i % n_distinct][:3]
-            if router is not None:
+            if nonlocal_router[0] is not None and device == "cpu":
                 payload, offs, lens = router.route_cpu(payload, offs, lens)
             pipe.ingest(payload, offs, lens)
         if dict_sync is not None:
@@ -489,7 +501,10 @@ def main() -> None:
                          "SmartEncoding + KG join + 1s rollup)",
                 "global_batch": args.batch * world,
                 "seq_len": None,
-                "parallelism": f"shard{world} (hash-sharded span streams)",
+                "parallelism": f"shard{world} " + (
+                "(routed all-to-all over RCCL)" if state["routed"] and
+                world > 1 else "(single shard)" if world == 1 else
+                "(UNROUTED fallback — routing failed)"),
                 "tag_cardinality": args.tag_card,
                 "bytes_per_span_stored": round(bytes_per_span, 1),
                 "bytes_per_span_resident": round(resident, 1),
