@@ -64,6 +64,11 @@ class TableDef:
     # device table capacity hint (log2); _map tables carry endpoint pairs
     # and need far more slots than the scalar-tag tables
     cap_pow2: int = 18
+    # harvest-time fold: this table's rows derive from another table's
+    # harvest (1m = fold(1s, 60)) instead of per-row kernel updates —
+    # the per-row path costs millions of random atomics, the fold costs
+    # thousands of group merges
+    derive_from: Optional[str] = None
 
 
 def _net(name, interval, map_):
@@ -102,19 +107,27 @@ def _app(name, interval, map_):
         CODE_VTAP | CODE_L7_PROTOCOL | CODE_STATUS | CODE_SERVER_PORT)
 
 
+import dataclasses as _dc
+
 L4_TABLES = [
     _net("network.1s", 1, False),
-    _net("network.1m", 60, False),
+    _dc.replace(_net("network.1m", 60, False),
+                derive_from="network.1s"),
     _net("network_map.1s", 1, True),
-    _net("network_map.1m", 60, True),
+    _dc.replace(_net("network_map.1m", 60, True),
+                derive_from="network_map.1s"),
+    # traffic_policy rows exist only for ACL-matched flows (tiny
+    # fraction) — the per-row path is already cheap there
     TableDef("traffic_policy.1m", "l4", 60, ("vtap_id", "acl_gid"), "net",
              CODE_VTAP | CODE_ACL_GID, require_nonzero="acl_gid"),
 ]
 L7_TABLES = [
     _app("application.1s", 1, False),
-    _app("application.1m", 60, False),
+    _dc.replace(_app("application.1m", 60, False),
+                derive_from="application.1s"),
     _app("application_map.1s", 1, True),
-    _app("application_map.1m", 60, True),
+    _dc.replace(_app("application_map.1m", 60, True),
+                derive_from="application_map.1s"),
 ]
 
 
@@ -155,6 +168,9 @@ class RollupTable:
                       else [_resolve(td.source, c) for c in td.keys])
         self._rnz = (1 + td.keys.index(td.require_nonzero)
                      if td.require_nonzero else 0)
+        self.source_table = None  # set by RollupFamily for derive_from
+        if td.derive_from is not None:
+            return
         if device == "cpu":
             self.table: Dict[tuple, List[int]] = {}
         else:
@@ -312,6 +328,8 @@ class RollupTable:
 
     def rows(self) -> List[Dict]:
         names = self.td.out_names or self.td.keys
+        if self.td.derive_from is not None:
+            return self._derived_rows()
         out = []
         for key, acc in self._items():
             row = {"time": self.time_base_s + key[0]}
@@ -323,13 +341,43 @@ class RollupTable:
             str(r[n]) for n in names))
         return out
 
+    def _derived_rows(self) -> List[Dict]:
+        """Fold the source table's harvested groups into this table's
+        coarser buckets (sums; *_max fields take max)."""
+        iv = self.td.interval_s
+        names = self.td.out_names or self.td.keys
+        folded: Dict[tuple, Dict] = {}
+        for r in self.source_table.rows():
+            t = ((r["time"] - self.time_base_s) // iv) * iv + \
+                self.time_base_s
+            key = (t,) + tuple(r[n] for n in names)
+            acc = folded.get(key)
+            if acc is None:
+                acc = dict(r)
+                acc["time"] = t
+                folded[key] = acc
+            else:
+                for f in self.fields:
+                    if f.endswith("_max"):
+                        acc[f] = max(acc[f], r[f])
+                    else:
+                        acc[f] += r[f]
+        out = list(folded.values())
+        out.sort(key=lambda r: (r["time"],) + tuple(
+            str(r[n]) for n in names))
+        return out
+
     def drop_count(self) -> int:
+        if self.td.derive_from is not None:
+            return self.source_table.drop_count()
         if self.device == "cpu":
             return 0
         return int(self.drops.item())
 
     # --------------------------------------------------------- checkpoint
     def state_dict(self):
+        if self.td.derive_from is not None:
+            return {}
         if self.device == "cpu":
             return {"table": {k: list(v) for k, v in self.table.items()}}
         return {"tkeys": self.tkeys.cpu().clone(),
@@ -337,6 +385,8 @@ class RollupTable:
                 "tvals": self.tvals.cpu().clone()}
 
     def load_state_dict(self, st):
+        if self.td.derive_from is not None:
+            return
         if self.device == "cpu":
             if "table" in st:
                 self.table.update(st["table"])
@@ -357,9 +407,13 @@ class RollupFamily:
         self.tables: Dict[str, RollupTable] = {
             td.name: RollupTable(td, time_base_s, device, 1 << td.cap_pow2)
             for td in defs}
+        for t in self.tables.values():
+            if t.td.derive_from is not None:
+                t.source_table = self.tables[t.td.derive_from]
 
     def update(self, seg, base: int, n: int, stream: int = 0) -> None:
-        tables = list(self.tables.values())
+        tables = [t for t in self.tables.values()
+                  if t.td.derive_from is None]
         if not tables or n == 0:
             return
         if tables[0].device == "cpu":

@@ -105,32 +105,47 @@ def test_agent_run_live(tmp_path):
              "--iface", "lo", "--vtap-id", "9", "--no-ebpf",
              "--flush-interval", "0.3"],
             stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
-        time.sleep(1.2)
-        # real HTTP round trip on loopback for the agent to capture
-        s = socket.socket()
-        s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
-        s.bind(("127.0.0.1", 0))
-        s.listen(1)
-        port = s.getsockname()[1]
-        cli_s = socket.create_connection(("127.0.0.1", port))
-        conn, _ = s.accept()
-        cli_s.sendall(b"GET /cli/run HTTP/1.1\r\nHost: cli.test\r\n\r\n")
-        conn.recv(4096)
-        conn.sendall(b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
-        cli_s.recv(4096)
-        cli_s.close()
-        conn.close()
-        s.close()
-        deadline = time.time() + 15
+        # wait for the agent subprocess (torch import takes seconds)
+        # to report its capture is attached before generating traffic
+        lines = []
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            line = proc.stdout.readline()
+            lines.append(line)
+            if "sending to" in line or not line:
+                break
+        assert any("capture on lo" in ln for ln in lines), lines
+
+        def http_roundtrip():
+            # real HTTP round trip on loopback for the agent to capture
+            s = socket.socket()
+            s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+            s.bind(("127.0.0.1", 0))
+            s.listen(1)
+            port = s.getsockname()[1]
+            cli_s = socket.create_connection(("127.0.0.1", port))
+            conn, _ = s.accept()
+            cli_s.sendall(
+                b"GET /cli/run HTTP/1.1\r\nHost: cli.test\r\n\r\n")
+            conn.recv(4096)
+            conn.sendall(
+                b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+            cli_s.recv(4096)
+            cli_s.close()
+            conn.close()
+            s.close()
+
+        deadline = time.time() + 20
         found = False
         while time.time() < deadline and not found:
-            time.sleep(0.5)
+            http_roundtrip()
+            time.sleep(0.6)
             r = srv.engine.query(
                 "SELECT request_resource FROM l7_flow_log "
                 "WHERE request_domain = 'cli.test' LIMIT 5")
             found = bool(r["values"])
         proc.terminate()
-        out = proc.communicate(timeout=10)[0]
+        out = "".join(lines) + proc.communicate(timeout=10)[0]
         assert found, out
     finally:
         srv.stop()
