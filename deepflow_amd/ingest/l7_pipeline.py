@@ -51,7 +51,11 @@ class L7IngestPipeline:
                  dictionary: Optional[TagDictionary] = None,
                  dict_capacity: int = 1 << 22,
                  window_bytes: Optional[int] = None,
-                 counter: Optional[Counter] = None):
+                 counter: Optional[Counter] = None,
+                 defer_harvest: bool = True):
+        # defer_harvest=False forces the per-batch dictionary harvest to
+        # resolve synchronously — required when the caller recycles the
+        # payload buffer as soon as ingest returns (native pump ring)
         self.device = device
         self.segments = SegmentSet(segment_rows, device,
                                    max_bytes=window_bytes)
@@ -80,6 +84,17 @@ class L7IngestPipeline:
         # persist into segments)
         self._scratch_str = None
         self._scratch_attr = None
+        # async bookkeeping (GPU mode): naive-bytes accounting accumulates
+        # on-device; dictionary harvest is deferred behind an event so the
+        # steady state (no new strings) costs zero host syncs
+        self.defer_harvest = defer_harvest
+        if device != "cpu":
+            self._naive_dev = torch.zeros(1, dtype=torch.int64, device=dev)
+            self._h_cnt = torch.zeros(1, dtype=torch.int32,
+                                      pin_memory=torch.cuda.is_available())
+            self._h_ev = torch.cuda.Event() if torch.cuda.is_available() \
+                else None
+            self._h_payload = None
 
     # ------------------------------------------------------------------
     def ingest(self, payload: np.ndarray, offs: np.ndarray,
@@ -142,15 +157,25 @@ class L7IngestPipeline:
                  seg: L7Segment, base: int, n: int) -> None:
         from ..ops import gpu_ops
         dev = payload_t.device
+        # resolve the PREVIOUS batch's deferred harvest first: its emit
+        # rows reference that batch's payload and must drain before this
+        # batch's intern kernels can append new rows
+        self._harvest_pending()
         sstr, sattr = self._scratch(n, dev)
         gpu_ops.decode_l7(payload_t, offs_t, lens_t, seg, base, sstr, sattr)
         gpu_ops.intern_many(payload_t, sstr, self._ref_rows_scalar,
                             self._dom_scalar, 0, n, self.dict.tkeys,
                             self.dict.emit, self.dict.emit_ctr, seg.did, base)
-        # variable attr-id pool: per-row block offsets from attr_cnt cumsum
+        # sizing: attr-pool cumsum (from decode's attr_cnt) and string-pool
+        # cumsum (from decode's sstr) both launch async, then ONE host sync
+        # reads both totals — the only sync in the batch
         cnts = seg.attr_cnt[base:base + n].to(torch.int64) * 2
         acum = torch.cumsum(cnts, 0)
-        attr_total = int(acum[-1].item())
+        row_len = torch.zeros(n, dtype=torch.int32, device=dev)
+        gpu_ops.pool_lens(sstr, self._pool_cols, n, row_len)
+        cum = torch.cumsum(row_len.to(torch.int64), 0)
+        totals = torch.stack((acum[-1], cum[-1])).cpu()
+        attr_total, total = int(totals[0]), int(totals[1])
         seg.ensure_attr_pool(attr_total)
         starts = (acum - cnts + seg.attr_pool_len).to(torch.int32)
         seg.attr_start[base:base + n] = starts
@@ -158,23 +183,45 @@ class L7IngestPipeline:
                              self.dict.emit, self.dict.emit_ctr, sattr,
                              starts)
         seg.attr_pool_len += attr_total
-        # pool sizing: lens kernel -> cumsum -> (sync) total
-        row_len = torch.zeros(n, dtype=torch.int32, device=dev)
-        gpu_ops.pool_lens(sstr, self._pool_cols, n, row_len)
-        cum = torch.cumsum(row_len.to(torch.int64), 0)
-        total = int(cum[-1].item())
         row_start = cum - row_len.to(torch.int64)
         seg.ensure_pool(total)
         gpu_ops.pool_gather(payload_t, seg, self._pool_cols, base, n,
                             row_start, seg.pool, seg.pool_len, sstr)
         self.rollups.update(seg, base, n)
-        new = self.dict.harvest(payload_host)  # syncs emit buffer
-        naive = int((sstr[:, :n] & 0xFFFF).sum()) + \
-            int((sattr[:, :n] & 0xFFFF).sum())
+        # naive-string accounting accumulates on-device (read via
+        # sync_stats); the old per-batch int(...) forced two extra syncs
+        self._naive_dev += (sstr[:, :n] & 0xFFFF).sum() + \
+            (sattr[:, :n] & 0xFFFF).sum()
+        self._harvest_defer(payload_host)
         seg.pool_len += total
-        self.stats.dict_new += new
         self.stats.pool_bytes += total
-        self.stats.naive_str_bytes += naive
+
+    # -------------------------------------------------- deferred harvest
+    def _harvest_defer(self, payload_host) -> None:
+        """Queue an async emit-counter readback; the actual harvest (if
+        any strings were new) happens at the start of the next batch or
+        at sync_stats()."""
+        if self._h_ev is None or not self.defer_harvest:
+            self.stats.dict_new += self.dict.harvest(payload_host)
+            return
+        self._h_cnt.copy_(self.dict.emit_ctr, non_blocking=True)
+        self._h_ev.record()
+        self._h_payload = payload_host
+
+    def _harvest_pending(self) -> None:
+        if self._h_payload is None:
+            return
+        self._h_ev.synchronize()  # waits only for the tiny D2H copy
+        if int(self._h_cnt[0]) > 0:
+            self.stats.dict_new += self.dict.harvest(self._h_payload)
+        self._h_payload = None
+
+    def sync_stats(self) -> PipelineStats:
+        """Flush deferred bookkeeping into `stats` (call after a sync)."""
+        if self.device != "cpu":
+            self._harvest_pending()
+            self.stats.naive_str_bytes = int(self._naive_dev.item())
+        return self.stats
 
     # ------------------------------------------------------------------
     def _ingest_cpu(self, payload, offs, lens, seg: L7Segment, base: int,

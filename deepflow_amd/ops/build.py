@@ -16,7 +16,8 @@ CSRC = OPS_DIR / "csrc"
 CPU_LIB = OPS_DIR / "libdfcpu.so"
 GPU_LIB = OPS_DIR / "libdfgpu.so"
 
-CPU_SOURCES = ["dfcpu.cpp", "agent_core.cpp", "otlp_conv.cpp"]
+CPU_SOURCES = ["dfcpu.cpp", "agent_core.cpp", "otlp_conv.cpp",
+               "recv_pump.cpp"]
 GPU_SOURCES = ["dfgpu.hip"]
 
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
@@ -41,7 +42,8 @@ def _run(cmd):
 def build_cpu(force: bool = False) -> Path:
     srcs = [str(CSRC / s) for s in CPU_SOURCES if (CSRC / s).exists()]
     if force or _needs_build(CPU_LIB, CPU_SOURCES):
-        _run(["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fopenmp", "-ldl",
+        _run(["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fopenmp",
+              "-pthread", "-ldl",
               *srcs, "-o", str(CPU_LIB)])
     return CPU_LIB
 

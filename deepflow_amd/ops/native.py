@@ -77,6 +77,21 @@ def cpu() -> ct.CDLL:
         lib.df_zstd_decompress.restype = ct.c_int64
         lib.df_zstd_decompress.argtypes = [ct.c_void_p, ct.c_uint64,
                                            ct.c_void_p, ct.c_uint64]
+        lib.df_pump_start.restype = ct.c_void_p
+        lib.df_pump_start.argtypes = [ct.c_int, ct.c_void_p, ct.c_uint64,
+                                      ct.c_int]
+        lib.df_pump_head.restype = ct.c_uint64
+        lib.df_pump_head.argtypes = [ct.c_void_p]
+        lib.df_pump_tail.restype = ct.c_uint64
+        lib.df_pump_tail.argtypes = [ct.c_void_p]
+        lib.df_pump_set_tail.restype = None
+        lib.df_pump_set_tail.argtypes = [ct.c_void_p, ct.c_uint64]
+        lib.df_pump_done.restype = ct.c_int
+        lib.df_pump_done.argtypes = [ct.c_void_p]
+        lib.df_pump_stats.restype = None
+        lib.df_pump_stats.argtypes = [ct.c_void_p] + [ct.c_void_p] * 4
+        lib.df_pump_free.restype = None
+        lib.df_pump_free.argtypes = [ct.c_void_p]
         lib.df_zstd_compress.restype = ct.c_int64
         lib.df_zstd_compress.argtypes = [ct.c_void_p, ct.c_uint64,
                                          ct.c_void_p, ct.c_uint64, ct.c_int]

@@ -168,6 +168,11 @@ class QueryEngine:
     def query(self, sql: str, _ctes: Optional[Dict[str, Dict]] = None) -> Dict:
         with self.lock:
             try:
+                # deferred ingest bookkeeping (async dictionary harvest)
+                # must land before hydration reads the host maps
+                for pipe in (self.pipe, self.l4):
+                    if pipe is not None and hasattr(pipe, "sync_stats"):
+                        pipe.sync_stats()
                 return self._query_locked(sql, _ctes)
             finally:
                 self._release_scratch()

@@ -1,0 +1,201 @@
+"""Native receiver pump (ops/csrc/recv_pump.cpp + ingest/native_pump.py):
+socket -> deframe -> zstd -> SPSC pinned ring, consumed zero-copy.
+"""
+import os
+import socket
+import time
+
+import numpy as np
+import pytest
+
+from deepflow_amd.ingest.native_pump import NativePump, PumpServer
+from deepflow_amd.wire import framing
+from deepflow_amd.ops import native
+
+
+def _frame(payload: bytes, zstd: bool = False,
+           msg_type: int = framing.MSG_PROTOCOLLOG) -> bytes:
+    if zstd:
+        import ctypes as ct
+        lib = native.cpu()
+        src = np.frombuffer(payload, dtype=np.uint8)
+        dst = np.zeros(len(payload) * 2 + 1024, dtype=np.uint8)
+        n = lib.df_zstd_compress(src.ctypes.data, len(src),
+                                 dst.ctypes.data, len(dst), 1)
+        assert n > 0
+        return framing.encode_frame(
+            framing.FrameHeader(msg_type=msg_type,
+                                encoder=framing.ENCODER_ZSTD),
+            dst[:n].tobytes())
+    return framing.encode_frame(framing.FrameHeader(msg_type=msg_type),
+                                payload)
+
+
+def _pump_pair(ring_bytes=1 << 20):
+    a, b = socket.socketpair()
+    p = NativePump(b, ring_bytes=ring_bytes, pin=False)
+    return a, p
+
+
+def _drain(p, n_expected, timeout=10.0):
+    out = []
+    t_end = time.time() + timeout
+    while len(out) < n_expected and time.time() < t_end:
+        v = p.poll()
+        if v is None:
+            time.sleep(0.001)
+            continue
+        out.append(bytes(v))
+        p.advance()
+    return out
+
+
+def test_pump_raw_and_zstd_frames():
+    a, p = _pump_pair()
+    try:
+        pay1 = b"hello-span-payload" * 10
+        pay2 = os.urandom(5000) + b"tail"
+        a.sendall(_frame(pay1) + _frame(pay2, zstd=True))
+        got = _drain(p, 2)
+        assert got == [pay1, pay2]
+        st = p.stats()
+        assert st["frames"] == 2 and st["bad_frames"] == 0
+        assert st["payload_bytes"] == len(pay1) + len(pay2)
+    finally:
+        a.close()
+        p.close()
+
+
+def test_pump_filters_msg_type():
+    a, p = _pump_pair()
+    try:
+        a.sendall(_frame(b"drop-me", msg_type=framing.MSG_SYSLOG) +
+                  _frame(b"keep-me"))
+        got = _drain(p, 1)
+        assert got == [b"keep-me"]
+    finally:
+        a.close()
+        p.close()
+
+
+def test_pump_ring_wrap_and_backpressure():
+    # ring far smaller than the stream: the producer must block on the
+    # consumer (backpressure) and wrap markers must be handled
+    a, p = _pump_pair(ring_bytes=1 << 14)  # 16 KB ring
+    payloads = [bytes([i % 251]) * (900 + 37 * i) for i in range(64)]
+    import threading
+
+    def send():
+        for pl in payloads:
+            a.sendall(_frame(pl))
+        a.shutdown(socket.SHUT_WR)
+
+    t = threading.Thread(target=send)
+    t.start()
+    try:
+        got = _drain(p, len(payloads), timeout=20)
+        assert got == payloads
+        assert p.stats()["frames"] == len(payloads)
+    finally:
+        t.join()
+        a.close()
+        p.close()
+
+
+def test_pump_desync_drops_connection():
+    a, p = _pump_pair()
+    try:
+        a.sendall(b"\xff\xff\xff\xff garbage that is not a frame")
+        t_end = time.time() + 5
+        while not p.lib.df_pump_done(p.h) and time.time() < t_end:
+            time.sleep(0.01)
+        assert p.stats()["bad_frames"] == 1
+    finally:
+        a.close()
+        p.close()
+
+
+def test_pump_server_end_to_end():
+    got = []
+    srv = PumpServer(lambda v: got.append(bytes(v)), pin=False,
+                     ring_bytes=1 << 20).start()
+    try:
+        payloads = [b"A" * 100, b"B" * 3000, os.urandom(999)]
+        conns = [socket.create_connection(("127.0.0.1", srv.port))
+                 for _ in range(2)]
+        conns[0].sendall(_frame(payloads[0]) + _frame(payloads[1],
+                                                      zstd=True))
+        conns[1].sendall(_frame(payloads[2]))
+        t_end = time.time() + 10
+        while len(got) < 3 and time.time() < t_end:
+            time.sleep(0.01)
+        assert sorted(got) == sorted(payloads)
+        st = srv.stats()
+        assert st["frames"] == 3 and st["connections"] == 2
+        for c in conns:
+            c.close()
+    finally:
+        srv.stop()
+
+
+def test_pump_throughput_near_wire_speed():
+    """The pump must consume at ~the raw socket rate (the Python
+    deframe loop loses ~half of it to interpreter work). Measured
+    against a same-host raw-recv baseline so the assertion tracks the
+    machine instead of a hard-coded number."""
+    import threading
+
+    n_frames, pay = 64, os.urandom(4 << 20)
+
+    def tcp_sender(port, blob, reps):
+        s = socket.create_connection(("127.0.0.1", port))
+        s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        for _ in range(reps):
+            s.sendall(blob)
+        s.shutdown(socket.SHUT_WR)
+        s.close()
+
+    # baseline: raw recv loop, no framing
+    srv = socket.socket()
+    srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+    t = threading.Thread(target=tcp_sender,
+                         args=(srv.getsockname()[1], pay, n_frames))
+    t.start()
+    conn, _ = srv.accept()
+    t0 = time.perf_counter()
+    while conn.recv(1 << 20):
+        pass
+    base_gbps = n_frames * len(pay) / (time.perf_counter() - t0) / 1e9
+    t.join()
+    conn.close()
+    srv.close()
+
+    # pump: same stream, framed
+    fr = _frame(pay)
+    srv = socket.socket()
+    srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+    t = threading.Thread(target=tcp_sender,
+                         args=(srv.getsockname()[1], fr, n_frames))
+    t.start()
+    conn, _ = srv.accept()
+    p = NativePump(conn, ring_bytes=256 << 20, pin=False)
+    t0 = time.perf_counter()
+    got = 0
+    while got < n_frames:
+        v = p.poll()
+        if v is None:
+            time.sleep(0.0002)  # yield the GIL to the test's sender
+            continue
+        got += 1
+        p.advance()
+    dt = time.perf_counter() - t0
+    t.join()
+    srv.close()
+    p.close()
+    gbps = n_frames * len(pay) / dt / 1e9
+    print(f"pump {gbps:.2f} GB/s vs raw recv {base_gbps:.2f} GB/s")
+    assert gbps > 0.5 * base_gbps
