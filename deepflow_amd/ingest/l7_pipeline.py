@@ -52,10 +52,12 @@ class L7IngestPipeline:
                  dict_capacity: int = 1 << 22,
                  window_bytes: Optional[int] = None,
                  counter: Optional[Counter] = None,
-                 defer_harvest: bool = True):
-        # defer_harvest=False forces the per-batch dictionary harvest to
-        # resolve synchronously — required when the caller recycles the
-        # payload buffer as soon as ingest returns (native pump ring)
+                 defer_harvest: bool = False):
+        # defer_harvest=True moves the per-batch dictionary harvest off
+        # the hot path (event-gated, resolves next batch / at query time
+        # via engine.sync_stats). Callers that read pipe.dict directly
+        # after ingest, or recycle the payload buffer immediately (native
+        # pump ring), need the synchronous default
         self.device = device
         self.segments = SegmentSet(segment_rows, device,
                                    max_bytes=window_bytes)
