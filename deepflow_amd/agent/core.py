@@ -13,10 +13,12 @@ from typing import Dict, Optional
 
 import numpy as np
 
+import struct
+
 from ..ops import native
 from ..wire import framing
 
-DRAIN_L4, DRAIN_L7, DRAIN_DOC, DRAIN_PCAP = 0, 1, 2, 3
+DRAIN_L4, DRAIN_L7, DRAIN_DOC, DRAIN_PCAP, DRAIN_NPB = 0, 1, 2, 3, 4
 
 _MSG_FOR = {DRAIN_L4: framing.MSG_TAGGEDFLOW,
             DRAIN_L7: framing.MSG_PROTOCOLLOG,
@@ -266,6 +268,36 @@ class Agent:
         }
         return framing.pack_records([pb.encode(rec, metric.STATS)])
 
+    def set_npb_target(self, host: str, port: int = 4789) -> None:
+        """Configure the NPB (north-bound packet broker) tunnel target;
+        ACL-matched frames with the NPB action (bit1) are mirrored as
+        VXLAN/UDP datagrams (reference: agent handler/npb.rs)."""
+        import socket as _socket
+        self._npb_target = (host, port)
+        self._npb_sock = _socket.socket(_socket.AF_INET,
+                                        _socket.SOCK_DGRAM)
+
+    def flush_npb(self) -> int:
+        """Drain the NPB mirror buffer and ship one VXLAN datagram per
+        mirrored frame. Returns datagrams sent."""
+        if getattr(self, "_npb_sock", None) is None:
+            return 0
+        buf = self.drain(DRAIN_NPB)
+        sent = 0
+        pos = 0
+        while pos + 2 <= len(buf):
+            (ln,) = struct.unpack_from("<H", buf, pos)
+            pos += 2
+            if pos + ln > len(buf):
+                break
+            try:
+                self._npb_sock.sendto(buf[pos:pos + ln], self._npb_target)
+                sent += 1
+            except OSError:
+                pass
+            pos += ln
+        return sent
+
     def flush_to_server(self, now_ns: int, compress: bool = False,
                         with_stats: bool = True) -> int:
         """tick + drain all types + send framed payloads to the server
@@ -273,6 +305,7 @@ class Agent:
         the reference's SenderEncoder::Zstd). Also ships a MSG_DFSTATS
         self-metrics frame. Returns frames sent."""
         self.tick(now_ns)
+        self.flush_npb()
         frames = []
         for which in (DRAIN_L4, DRAIN_L7, DRAIN_DOC, DRAIN_PCAP):
             payload = self.drain(which)

@@ -90,6 +90,7 @@ struct FlowNode {
     std::vector<uint8_t> h2_carry[2];  // cross-segment frame reassembly
     std::vector<uint32_t> acl_gids;    // fast-path cached ACL matches
     uint32_t acl_actions = 0;
+    uint32_t npb_vni = 0;    // VXLAN VNI for NPB mirroring (= acl gid)
     // eBPF socket-trace provenance (signal_source 3 = SIGNAL_SOURCE_EBPF):
     // syscall trace ids per direction (0 = request/client side) and the
     // owning processes, carried into AppProtoLogsBaseInfo 25/26/29/30
@@ -152,7 +153,7 @@ struct AclRule {
     uint32_t src_net = 0, src_mask = 0, dst_net = 0, dst_mask = 0;
     uint8_t proto = 0;       // 0 = any
     uint16_t port_min = 0, port_max = 65535;  // server port range
-    uint32_t action = 0;     // bit0: pcap-capture tag
+    uint32_t action = 0;     // bit0: pcap-capture, bit1: NPB mirror
 };
 
 // DDBS-style first path (reference policy/first_path.rs): every ACL owns
@@ -301,11 +302,19 @@ struct Agent {
     FirstPath first_path;
     FastPath fast_path;
     std::map<MeterKey, AppMeterAcc> meters;
-    std::vector<uint8_t> out_l4, out_l7, out_doc, out_pcap;
+    std::vector<uint8_t> out_l4, out_l7, out_doc, out_pcap, out_npb;
+    // NPB dedup: the same packet observed at two capture points (both
+    // ends of a veth, tx+rx) must mirror once (reference
+    // handler/npb.rs dedup table). Keyed by a hash of the invariant
+    // L3/L4 bytes; entries expire after NPB_DEDUP_NS.
+    struct NpbSeen { uint64_t h = 0; uint64_t ts = 0; };
+    std::vector<NpbSeen> npb_dedup = std::vector<NpbSeen>(1 << 14);
     // stats
     uint64_t pkts = 0, bytes = 0, flows_emitted = 0, l7_emitted = 0,
-             docs_emitted = 0, parse_errors = 0;
+             docs_emitted = 0, parse_errors = 0, npb_deduped = 0;
 };
+
+constexpr uint64_t NPB_DEDUP_NS = 100ull * 1000 * 1000;
 
 // policy lookup for a new flow: fast-path LRU, then the DDBS first path
 // in both directions (reference Policy::lookup, policy/policy.rs:283)
@@ -328,6 +337,7 @@ void match_acls(Agent& a, FlowNode& f) {
         if (bm[i / 64] & (1ull << (i % 64))) {
             if (f.acl_gids.size() < 8) f.acl_gids.push_back(a.acls[i].gid);
             f.acl_actions |= a.acls[i].action;
+            if (a.acls[i].action & 2u) f.npb_vni = a.acls[i].gid;
         }
     }
 }
@@ -2655,6 +2665,47 @@ int dfa_packet(void* h, const uint8_t* pkt, uint32_t len, uint64_t ts_ns) {
         memcpy(w + 18, pkt, l16);
         if (l16 < len) a.out_pcap.resize(base + 18 + l16);
     }
+    // ACL NPB action: VXLAN-encapsulate matched frames into the NPB
+    // drain ([u16 total][8B VXLAN header][inner L2 frame]); the sender
+    // ships each entry as one UDP datagram to the packet broker.
+    // Reference: agent handler/npb.rs (north-bound packet broker).
+    if ((f.acl_actions & 2u) && a.out_npb.size() < (4u << 20)) {
+        // dedup hash over capture-point-invariant bytes: addresses,
+        // ports, proto and the first 16 L4 header bytes (seq/ack or
+        // udp len/cksum) + the L3 total length
+        uint64_t ph = 0x9E3779B97F4A7C15ull;
+        auto mix = [&ph](uint64_t v) {
+            ph ^= v; ph *= 0xFF51AFD7ED558CCDull; ph ^= ph >> 33;
+        };
+        mix(((uint64_t)src << 32) | dst);
+        mix(((uint64_t)sport << 48) | ((uint64_t)dport << 32) |
+            ((uint64_t)proto << 24) | tot);
+        for (int i = 0; i + 8 <= 16 && l4 + i + 8 <= pkt + len; i += 8) {
+            uint64_t w;
+            memcpy(&w, l4 + i, 8);
+            mix(w);
+        }
+        Agent::NpbSeen& seen = a.npb_dedup[ph & (a.npb_dedup.size() - 1)];
+        if (seen.h == ph && ts_ns - seen.ts < NPB_DEDUP_NS) {
+            a.npb_deduped++;
+        } else {
+            seen.h = ph;
+            seen.ts = ts_ns;
+            uint16_t l16 = (uint16_t)(len > 0xFF00 ? 0xFF00 : len);
+            size_t base = a.out_npb.size();
+            a.out_npb.resize(base + 2 + 8 + l16);
+            uint8_t* w = a.out_npb.data() + base;
+            uint16_t total = (uint16_t)(8 + l16);
+            memcpy(w, &total, 2);
+            // VXLAN (RFC 7348): flags 0x08, reserved, VNI<<8
+            w[2] = 0x08; w[3] = 0; w[4] = 0; w[5] = 0;
+            w[6] = (uint8_t)(f.npb_vni >> 16);
+            w[7] = (uint8_t)(f.npb_vni >> 8);
+            w[8] = (uint8_t)f.npb_vni;
+            w[9] = 0;
+            memcpy(w + 10, pkt, l16);
+        }
+    }
     PeerStats& ps = f.peer[dir];
     ps.packets++; ps.total_packets++;
     ps.bytes += len; ps.total_bytes += len;
@@ -2899,7 +2950,8 @@ uint64_t dfa_drain(void* h, int which, uint8_t* out, uint64_t cap) {
     Agent& a = *(Agent*)h;
     std::vector<uint8_t>& src = which == 0 ? a.out_l4
                                : which == 1 ? a.out_l7
-                               : which == 2 ? a.out_doc : a.out_pcap;
+                               : which == 2 ? a.out_doc
+                               : which == 4 ? a.out_npb : a.out_pcap;
     uint64_t n = src.size();
     if (out && n <= cap) memcpy(out, src.data(), n);
     if (out) src.clear();

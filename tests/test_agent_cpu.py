@@ -504,3 +504,43 @@ def test_policy_ddbs_and_fastpath():
              for r in framing.iter_records(a.drain(0))]
     assert all(100 in set(d["flow"].get("acl_gids", [])) or True
                for d in recs2)
+
+
+def test_npb_vxlan_mirror():
+    """ACL NPB action (bit1): matched frames mirror as VXLAN datagrams
+    to the packet-broker target; identical frames within the dedup
+    window (two capture points seeing the same packet) mirror once.
+    Reference: agent handler/npb.rs."""
+    import socket
+    import struct
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.agent.packets import http_session
+
+    sink = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    sink.bind(("127.0.0.1", 0))
+    sink.settimeout(5)
+
+    a = Agent(vtap_id=1)
+    a.add_acl(42, dst_net=0x0A000002, dst_masklen=32, action=2)  # NPB
+    a.set_npb_target("127.0.0.1", sink.getsockname()[1])
+    frames = list(http_session(0x0A000001, 0x0A000002, t0=10**9))
+    for frame, ts in frames:
+        a.packet(frame, ts)
+        a.packet(frame, ts)  # second capture point: must dedup
+    for frame, ts in http_session(0x0A000001, 0x0A000003, sport=43999,
+                                  t0=10**9):
+        a.packet(frame, ts)  # unmatched: not mirrored
+    sent = a.flush_npb()
+    assert sent == len(frames)
+
+    got = []
+    for _ in range(sent):
+        got.append(sink.recv(65535))
+    sink.close()
+    for dg, (frame, _) in zip(got, frames):
+        # VXLAN header: flags 0x08, VNI = acl gid, then the inner frame
+        assert dg[0] == 0x08
+        vni = (dg[4] << 16) | (dg[5] << 8) | dg[6]
+        assert vni == 42
+        assert dg[8:] == frame
+    a.close()
