@@ -271,7 +271,7 @@ def e2e_main(args) -> None:
                                 pin_memory=have_gpu)) for _ in range(4)]
         si = [0]
 
-        def on_pump(view):
+        def on_pump(view, meta):
             offs_p, lens_p = scratch[si[0] % len(scratch)]
             si[0] += 1
             n = int(lib.df_scan_offsets(
@@ -437,29 +437,40 @@ def main() -> None:
         # is INSIDE the timed step)
         router = SpanRouter(device=device)
 
-    # H2D prefetch pipeline: copy batch i+1 on a side stream while batch i's
-    # kernels run on the main stream.
+    # H2D prefetch pipeline: copy batch i+1 while batch i's kernels run.
+    # The payload copy IS the step's critical path (727 MB/batch ~= the
+    # PCIe link: rocprofv3 runtime-trace shows 12.7 ms at 56.7 GB/s vs a
+    # 13.1 ms step) — split it across two streams so both SDMA engines
+    # pull concurrently.
     copy_stream = torch.cuda.Stream() if device == "cuda" else None
+    copy_stream2 = torch.cuda.Stream() if device == "cuda" else None
     pending = {}
     state = {"routed": world > 1}
     nonlocal_router = [router]
 
     def prefetch(i: int) -> None:
         buf, offs, lens, pay_t, offs_t, lens_t = batches[i % n_distinct]
+        half = pay_t.numel() // 2
+        dev_pay = torch.empty_like(pay_t, device="cuda")
         with torch.cuda.stream(copy_stream):
-            dev_batch = (pay_t.to("cuda", non_blocking=True),
-                         offs_t.to("cuda", non_blocking=True),
-                         lens_t.to("cuda", non_blocking=True))
+            dev_pay[:half].copy_(pay_t[:half], non_blocking=True)
+            dev_o = offs_t.to("cuda", non_blocking=True)
+            dev_l = lens_t.to("cuda", non_blocking=True)
             ev = torch.cuda.Event()
             ev.record(copy_stream)
-        pending[i] = (dev_batch, ev, buf)
+        with torch.cuda.stream(copy_stream2):
+            dev_pay[half:].copy_(pay_t[half:], non_blocking=True)
+            ev2 = torch.cuda.Event()
+            ev2.record(copy_stream2)
+        pending[i] = ((dev_pay, dev_o, dev_l), (ev, ev2), buf)
 
     def step(i: int) -> None:
         if device == "cuda":
             if i not in pending:
                 prefetch(i)
-            dev_batch, ev, host_payload = pending.pop(i)
-            torch.cuda.current_stream().wait_event(ev)
+            dev_batch, evs, host_payload = pending.pop(i)
+            for ev in evs:
+                torch.cuda.current_stream().wait_event(ev)
             for t in dev_batch:
                 t.record_stream(torch.cuda.current_stream())
             prefetch(i + 1)

@@ -60,18 +60,24 @@ class NativePump:
 
     def poll(self) -> Optional[np.ndarray]:
         """Next payload view, or None if the ring is empty.  The view
-        aliases pinned ring memory — valid until `advance()`."""
+        aliases pinned ring memory — valid until `advance()`; frame
+        metadata for it is in self.meta (msg_type, agent_id, org_id,
+        team_id)."""
         while True:
             head = self.lib.df_pump_head(self.h)
             if head == self.tail:
                 return None
             pos = self.tail % self.cap
-            ln = int(self.ring[pos:pos + 8].view(np.uint64)[0])
+            hdr = self.ring[pos:pos + 16].view(np.uint64)
+            ln = int(hdr[0])
             if ln == WRAP_MARK:
                 self.tail += self.cap - pos
                 continue
-            self._entry = 8 + ((ln + 7) & ~7)
-            return self.ring[pos + 8: pos + 8 + ln]
+            m = int(hdr[1])
+            self.meta = (m & 0xFF, (m >> 8) & 0xFFFF, (m >> 24) & 0xFFFF,
+                         (m >> 40) & 0xFFFFFFFF)
+            self._entry = 16 + ((ln + 7) & ~7)
+            return self.ring[pos + 16: pos + 16 + ln]
 
     def advance(self) -> None:
         self.tail += self._entry
@@ -99,8 +105,9 @@ class PumpServer:
     """Accept loop that spawns a NativePump per connection and runs a
     consumer thread draining every ring.
 
-    handler(payload_view) is called with each frame's decoded payload
-    (numpy view over pinned memory); it must finish consuming the bytes
+    handler(payload_view, meta) is called with each frame's decoded
+    payload (numpy view over pinned memory) and its frame metadata
+    (msg_type, agent_id, org_id, team_id); it must finish consuming the bytes
     (or enqueue an async copy and return an event-like with .query())
     before the view is recycled.  If the handler returns an object with
     a query() method, the pump defers the ring advance until it reports
@@ -108,7 +115,7 @@ class PumpServer:
     without racing the producer.
     """
 
-    def __init__(self, handler: Callable[[np.ndarray], object],
+    def __init__(self, handler: Callable[[np.ndarray, tuple], object],
                  host: str = "127.0.0.1", port: int = 0,
                  ring_bytes: int = 256 << 20,
                  accept_type: int = framing.MSG_PROTOCOLLOG,
@@ -171,7 +178,7 @@ class PumpServer:
                 if view is None:
                     continue
                 busy = True
-                ev = self.handler(view)
+                ev = self.handler(view, p.meta)
                 if ev is not None and hasattr(ev, "query"):
                     self._pending.append((p, ev))
                 else:

@@ -10,8 +10,9 @@
 // server/libs/receiver/receiver.go (flow per-connection goroutines).
 //
 // Ring entry format (8-byte aligned):
-//   [u64 payload_len][payload bytes][pad to 8]
-// A u64 of ~0 at the cursor is a wrap marker: skip to the next ring
+//   [u64 payload_len][u64 meta][payload bytes][pad to 8]
+//   meta = msg_type | (agent_id << 8) | (org_id << 24) | (team_id << 40)
+// A u64 of ~0 at the first word is a wrap marker: skip to the next ring
 // boundary.  Cursors are monotonic byte counts (position = cur % cap).
 
 #include <atomic>
@@ -99,13 +100,14 @@ struct Pump {
         }
     }
 
-    void publish(const uint8_t* payload, uint64_t n) {
-        uint64_t need = 8 + ((n + 7) & ~7ull);
+    void publish(const uint8_t* payload, uint64_t n, uint64_t meta) {
+        uint64_t need = 16 + ((n + 7) & ~7ull);
         uint64_t h = reserve(need);
         if (h == UINT64_MAX) return;
         uint64_t pos = h % cap;
         memcpy(ring + pos, &n, 8);
-        memcpy(ring + pos + 8, payload, n);
+        memcpy(ring + pos + 8, &meta, 8);
+        memcpy(ring + pos + 16, payload, n);
         head.store(h + need, std::memory_order_release);
         frames.fetch_add(1, std::memory_order_relaxed);
         payload_bytes.fetch_add(n, std::memory_order_relaxed);
@@ -133,6 +135,17 @@ struct Pump {
             // payload at frame[19] -> fbuf[15]
             int msg_type = fbuf[0];
             int encoder = fbuf[3];
+            // header (framing.py): team u32 @8, org u16 @12, agent u16
+            // @16 — all relative to the frame, fbuf starts at offset 4
+            uint32_t team = (uint32_t)fbuf[4] | ((uint32_t)fbuf[5] << 8) |
+                            ((uint32_t)fbuf[6] << 16) |
+                            ((uint32_t)fbuf[7] << 24);
+            uint16_t org = (uint16_t)(fbuf[8] | (fbuf[9] << 8));
+            uint16_t agent = (uint16_t)(fbuf[12] | (fbuf[13] << 8));
+            uint64_t meta = (uint64_t)(uint8_t)msg_type |
+                            ((uint64_t)agent << 8) |
+                            ((uint64_t)org << 24) |
+                            ((uint64_t)team << 40);
             const uint8_t* pay = fbuf.data() + (HEADER_LEN - 4);
             uint64_t pn = size - HEADER_LEN;
             if (accept_type >= 0 && msg_type != accept_type) continue;
@@ -151,7 +164,7 @@ struct Pump {
                     if (dbuf.size() < capg) dbuf.resize(capg);
                     size_t r = zdec(dbuf.data(), capg, pay, pn);
                     if (!ziserr(r)) {
-                        publish(dbuf.data(), r);
+                        publish(dbuf.data(), r, meta);
                         break;
                     }
                     capg *= 4;
@@ -162,7 +175,7 @@ struct Pump {
                     }
                 }
             } else if (encoder == ENCODER_RAW) {
-                publish(pay, pn);
+                publish(pay, pn, meta);
             } else {
                 bad_frames.fetch_add(1, std::memory_order_relaxed);
             }

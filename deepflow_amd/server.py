@@ -10,6 +10,8 @@ from __future__ import annotations
 import threading
 from typing import Optional
 
+import numpy as np
+
 from fastapi import Request
 
 from .gen import SpanGenConfig
@@ -27,7 +29,9 @@ class DeepflowServer:
                  segment_rows: int = 1 << 20,
                  dict_capacity: int = 1 << 20,
                  time_base_s: int = 1_700_000_000,
-                 platform_cfg: Optional[SpanGenConfig] = None):
+                 platform_cfg: Optional[SpanGenConfig] = None,
+                 native_pump: bool = False,
+                 pump_port: int = 0):
         self.device = device
         self.kg = KnowledgeGraphTable(device=device)
         if platform_cfg is not None:
@@ -42,6 +46,15 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_PROTOCOLLOG, self._on_l7)
         self.receiver.register(framing.MSG_TAGGEDFLOW, self._on_l4)
         self.engine = QueryEngine(self.l7, device=device, l4_pipeline=self.l4)
+        # native data-plane receiver (ops/csrc/recv_pump.cpp): C++
+        # deframe/zstd straight into a pinned ring, dispatched into the
+        # same per-type handlers. Started on its own port by start()
+        # when native_pump=True (agents point their data plane at it;
+        # the Python receiver keeps serving control-plane traffic).
+        self._use_pump = native_pump
+        self._pump_port = pump_port
+        self.pump = None
+        self.pump_port = None
         # multi-org isolation: each non-default org gets its own KG,
         # dictionary, segment sets and engine (reference: per-org
         # ClickHouse databases, org_id from the frame header / ORG_ID
@@ -407,6 +420,21 @@ class DeepflowServer:
                                     lens.ctypes.data_as(ct.c_void_p), max_n))
         return offs[:n].copy(), lens[:n].copy()
 
+    def _on_pump_frame(self, view, meta):
+        """PumpServer dispatch: frame metadata -> the registered
+        per-type handler. The pinned view is only valid until return,
+        so handlers that keep bytes get a copy."""
+        msg_type, agent_id, org_id, team_id = meta
+        handler = self.receiver.handlers.get(msg_type)
+        if handler is None:
+            self.receiver.counter.add("unhandled_type")
+            return None
+        hdr = framing.FrameHeader(msg_type=msg_type, agent_id=agent_id,
+                                  org_id=org_id or 1, team_id=team_id)
+        handler(hdr, np.array(view, copy=True))
+        self.receiver.counter.add("frames_in")
+        return None
+
     def _on_l7(self, hdr, payload) -> None:
         offs, lens = self._scan_records(payload)
         pipe = self.org_context(hdr.org_id).l7
@@ -488,10 +516,18 @@ class DeepflowServer:
 
     def start(self) -> None:
         self.receiver.start()
+        if self._use_pump:
+            from .ingest.native_pump import PumpServer
+            self.pump = PumpServer(self._on_pump_frame, port=self._pump_port,
+                                   accept_type=-1,
+                                   pin=self.device == "cuda").start()
+            self.pump_port = self.pump.port
         self.debug_bus.start()
 
     def stop(self) -> None:
         self.receiver.stop()
+        if self.pump is not None:
+            self.pump.stop()
         self.debug_bus.stop()
         if self._ckpt_thread is not None:
             self._stop_ckpt.set()
