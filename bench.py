@@ -303,26 +303,23 @@ def e2e_main(args) -> None:
 
     n_streams = max(1, int(os.environ.get("DF_E2E_STREAMS", "2")))
 
-    def _sender_proc(j: int, k: int) -> None:
-        s = _socket.create_connection(("127.0.0.1", rx.tcp_port))
-        s.setsockopt(_socket.IPPROTO_TCP, _socket.TCP_NODELAY, 1)
+    def send_frames(k: int) -> None:
+        # N parallel sender connections (agents are many); each batch's
+        # sub-frames round-robin across streams. One thread suffices:
+        # 32 MB sendall calls release the GIL, and forked sender
+        # processes measured 4x SLOWER (fork of a CUDA-context process
+        # with a multi-GB COW address space).
+        socks = []
+        for _ in range(n_streams):
+            s = _socket.create_connection(("127.0.0.1", rx.tcp_port))
+            s.setsockopt(_socket.IPPROTO_TCP, _socket.TCP_NODELAY, 1)
+            socks.append(s)
         for i in range(k):
             fl = frames[i % n_distinct]
-            for idx in range(j, len(fl), n_streams):
-                s.sendall(fl[idx])
-        s.close()
-
-    def send_frames(k: int) -> None:
-        # N sender PROCESSES (fork — frames inherit without pickling):
-        # agents are many independent hosts; a single GIL-bound sender
-        # thread caps the measurement at the harness, not the server
-        import multiprocessing as _mp
-        procs = [_mp.Process(target=_sender_proc, args=(j, k))
-                 for j in range(n_streams)]
-        for pr in procs:
-            pr.start()
-        for pr in procs:
-            pr.join()
+            for j, fr in enumerate(fl):
+                socks[j % n_streams].sendall(fr)
+        for s in socks:
+            s.close()
 
     def wait_rows(target: int, timeout=120.0):
         t_end = time.time() + timeout

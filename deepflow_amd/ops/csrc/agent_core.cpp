@@ -91,6 +91,10 @@ struct FlowNode {
     std::vector<uint32_t> acl_gids;    // fast-path cached ACL matches
     uint32_t acl_actions = 0;
     uint32_t npb_vni = 0;    // VXLAN VNI for NPB mirroring (= acl gid)
+    // SQL pipelining: pipelined requests queue FIFO and match responses
+    // in order (reference: perf-layer session queues). Bounded depth.
+    std::vector<std::pair<uint64_t, std::string>> sql_q;
+    bool sql_err_pending = false;
     // eBPF socket-trace provenance (signal_source 3 = SIGNAL_SOURCE_EBPF):
     // syscall trace ids per direction (0 = request/client side) and the
     // owning processes, carried into AppProtoLogsBaseInfo 25/26/29/30
@@ -1821,22 +1825,66 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
         }
     } else if (f.l7_protocol == 61) {  // PostgreSQL
         if (dir == 0) {
-            std::string stmt;
-            if (parse_pgsql_request(p, n, stmt)) {
+            // pipelined simple protocol: one client segment may carry
+            // SEVERAL 'Q' messages — queue each; responses pop in order
+            uint32_t pos = 0;
+            bool any = false;
+            while (pos + 5 <= n) {
+                uint8_t mtype = p[pos];
+                uint32_t mlen = (p[pos + 1] << 24) | (p[pos + 2] << 16) |
+                                (p[pos + 3] << 8) | p[pos + 4];
+                if (mlen < 4 || pos + 1 + mlen > n + 4) break;
+                if (mtype == 'Q' && mlen >= 5) {
+                    uint32_t sl = mlen - 5;
+                    if (pos + 5 + sl > n) sl = n - pos - 5;
+                    while (sl && p[pos + 5 + sl - 1] == 0) sl--;
+                    std::string stmt((const char*)p + pos + 5, sl);
+                    if (f.sql_q.size() < 8)
+                        f.sql_q.emplace_back(ts, obfuscate_sql(stmt));
+                    any = true;
+                }
+                if (pos + 1 + mlen > n) break;
+                pos += 1 + mlen;
+            }
+            if (any) {
                 f.l7.active = true;
                 f.l7.req_ts = ts;
                 f.l7.req_len = n;
                 f.l7.req_type = "Query";
-                f.l7.resource = stmt;
                 f.l7.endpoint = "";
                 f.l7.domain = "";
                 f.l7c.request_count++;
                 f.last_req_pkt_ts = ts;
             }
         } else if (dir == 1 && f.l7.active && n >= 1) {
-            bool err = p[0] == 'E';
-            encode_l7_record(a, f, f.l7.req_ts, ts, 0, err ? 3 : 0, f.l7, "");
-            f.l7.active = false;
+            // each query's reply ends with 'Z' (ReadyForQuery); an 'E'
+            // before it marks that query failed. One segment may close
+            // several pipelined queries.
+            uint32_t pos = 0;
+            while (pos + 5 <= n) {
+                uint8_t mtype = p[pos];
+                uint32_t mlen = (p[pos + 1] << 24) | (p[pos + 2] << 16) |
+                                (p[pos + 3] << 8) | p[pos + 4];
+                if (mlen < 4) break;
+                if (mtype == 'E') f.sql_err_pending = true;
+                if (mtype == 'Z' && !f.sql_q.empty()) {
+                    f.l7.resource = f.sql_q.front().second;
+                    encode_l7_record(a, f, f.sql_q.front().first, ts, 0,
+                                     f.sql_err_pending ? 3 : 0, f.l7, "");
+                    f.sql_q.erase(f.sql_q.begin());
+                    f.sql_err_pending = false;
+                }
+                if (pos + 1 + mlen > n) break;
+                pos += 1 + mlen;
+            }
+            // legacy single-shot path: truncated capture without a 'Z'
+            if (!f.sql_q.empty() && pos == 0 && n >= 1) {
+                f.l7.resource = f.sql_q.front().second;
+                encode_l7_record(a, f, f.sql_q.front().first, ts, 0,
+                                 p[0] == 'E' ? 3 : 0, f.l7, "");
+                f.sql_q.erase(f.sql_q.begin());
+            }
+            if (f.sql_q.empty()) f.l7.active = false;
         }
     } else if (f.l7_protocol == 100) {  // Kafka
         if (dir == 0) {
@@ -2451,6 +2499,11 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
         if (dir == 0) {
             std::string stmt;
             if (parse_mysql_request(p, n, stmt)) {
+                // pipelined request before the previous one's response:
+                // emit the outstanding record rather than overwrite it
+                if (f.l7.active)
+                    encode_l7_record(a, f, f.l7.req_ts, ts, 0, 0, f.l7,
+                                     "");
                 f.l7.active = true;
                 f.l7.req_ts = ts;
                 f.l7.req_len = n;
