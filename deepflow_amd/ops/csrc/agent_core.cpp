@@ -1861,6 +1861,7 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
             // before it marks that query failed. One segment may close
             // several pipelined queries.
             uint32_t pos = 0;
+            bool saw_z = false;
             while (pos + 5 <= n) {
                 uint8_t mtype = p[pos];
                 uint32_t mlen = (p[pos + 1] << 24) | (p[pos + 2] << 16) |
@@ -1868,6 +1869,7 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 if (mlen < 4) break;
                 if (mtype == 'E') f.sql_err_pending = true;
                 if (mtype == 'Z' && !f.sql_q.empty()) {
+                    saw_z = true;
                     f.l7.resource = f.sql_q.front().second;
                     encode_l7_record(a, f, f.sql_q.front().first, ts, 0,
                                      f.sql_err_pending ? 3 : 0, f.l7, "");
@@ -1877,12 +1879,15 @@ void handle_l7_payload(Agent& a, FlowNode& f, int dir, const uint8_t* p,
                 if (pos + 1 + mlen > n) break;
                 pos += 1 + mlen;
             }
-            // legacy single-shot path: truncated capture without a 'Z'
-            if (!f.sql_q.empty() && pos == 0 && n >= 1) {
+            // non-pipelined flow without a captured 'Z' (snaplen/segment
+            // truncation): emit on the first response segment, as before
+            if (!saw_z && f.sql_q.size() == 1) {
                 f.l7.resource = f.sql_q.front().second;
                 encode_l7_record(a, f, f.sql_q.front().first, ts, 0,
-                                 p[0] == 'E' ? 3 : 0, f.l7, "");
-                f.sql_q.erase(f.sql_q.begin());
+                                 (p[0] == 'E' || f.sql_err_pending) ? 3 : 0,
+                                 f.l7, "");
+                f.sql_q.clear();
+                f.sql_err_pending = false;
             }
             if (f.sql_q.empty()) f.l7.active = false;
         }

@@ -290,3 +290,65 @@ def test_compressed_sender():
     assert srv.receiver.handle_frame(framed)
     assert srv.l7.stats.spans_in == 10
     a.close()
+
+
+def test_pgsql_pipelined(agent):
+    """Two 'Q' messages in one segment (PG pipelining): each query gets
+    its own record, responses matched FIFO by ReadyForQuery ('Z'), with
+    a per-query error status from an 'E' before its 'Z'."""
+    import struct
+
+    def q(sql: bytes) -> bytes:
+        return b"Q" + struct.pack(">I", 4 + len(sql) + 1) + sql + b"\x00"
+
+    def m(t: bytes, body: bytes = b"") -> bytes:
+        return t + struct.pack(">I", 4 + len(body)) + body
+
+    req = q(b"SELECT 1") + q(b"INSERT INTO t VALUES (99)")
+    # reply: rows + Z (ok), then error + Z
+    resp = m(b"T", b"x") + m(b"Z", b"I") + m(b"E", b"boom") + m(b"Z", b"I")
+    _tcp_exchange(agent, 51010, 5432, req, resp)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 2
+    assert l7[0]["req"]["resource"] == "SELECT ?"
+    assert l7[0]["resp"].get("status", 0) == 0
+    assert l7[1]["req"]["resource"] == "INSERT INTO t VALUES (?)"
+    assert l7[1]["resp"]["status"] == 3
+
+
+def test_mysql_pipelined_requests(agent):
+    """A second COM_QUERY before the first response flushes the first
+    request as its own record instead of overwriting it."""
+    import struct
+    from deepflow_amd.agent.packets import eth_ipv4_tcp
+
+    def com_query(sql: bytes) -> bytes:
+        return struct.pack("<I", len(sql) + 1)[:3] + b"\x00\x03" + sql
+
+    t0 = 10**9
+    sport, dport = 51020, 3306
+    SYN, SYNACK, ACK, PSH_ACK = 0x02, 0x12, 0x10, 0x18
+    pkts = [
+        (eth_ipv4_tcp(CLIENT, SERVER, sport, dport, SYN, 0, 0), t0),
+        (eth_ipv4_tcp(SERVER, CLIENT, dport, sport, SYNACK, 0, 1),
+         t0 + 1_000_000),
+        (eth_ipv4_tcp(CLIENT, SERVER, sport, dport, ACK, 1, 1),
+         t0 + 1_100_000),
+        (eth_ipv4_tcp(CLIENT, SERVER, sport, dport, PSH_ACK, 1, 1,
+                      com_query(b"SELECT a FROM t1 WHERE x = 5")),
+         t0 + 2_000_000),
+        (eth_ipv4_tcp(CLIENT, SERVER, sport, dport, PSH_ACK, 40, 1,
+                      com_query(b"SELECT b FROM t2 WHERE y = 6")),
+         t0 + 3_000_000),
+        (eth_ipv4_tcp(SERVER, CLIENT, dport, sport, PSH_ACK, 1, 80,
+                      b"\x01\x00\x00\x01\x01"), t0 + 5_000_000),
+    ]
+    for frame, ts in pkts:
+        agent.packet(frame, ts)
+    agent.tick(10**9 * 100)
+    l7 = _decode(agent.drain(1), flow_log.APP_PROTO_LOGS_DATA)
+    assert len(l7) == 2
+    res = sorted(r["req"]["resource"] for r in l7)
+    assert res == ["SELECT a FROM t1 WHERE x = ?",
+                   "SELECT b FROM t2 WHERE y = ?"]
