@@ -113,69 +113,121 @@ struct Pump {
         payload_bytes.fetch_add(n, std::memory_order_relaxed);
     }
 
+    // Reserve ring space for an entry of `n` payload bytes and write
+    // its header; returns the payload write pointer (or null on stop).
+    // The caller fills the payload then calls commit(h, n).
+    uint8_t* begin_entry(uint64_t n, uint64_t meta, uint64_t& h) {
+        uint64_t need = 16 + ((n + 7) & ~7ull);
+        h = reserve(need);
+        if (h == UINT64_MAX) return nullptr;
+        uint64_t pos = h % cap;
+        memcpy(ring + pos, &n, 8);
+        memcpy(ring + pos + 8, &meta, 8);
+        return ring + pos + 16;
+    }
+
+    void commit(uint64_t h, uint64_t n) {
+        head.store(h + 16 + ((n + 7) & ~7ull), std::memory_order_release);
+        frames.fetch_add(1, std::memory_order_relaxed);
+        payload_bytes.fetch_add(n, std::memory_order_relaxed);
+    }
+
     void run() {
         static zstd_fn4 zdec = (zstd_fn4)zsym("ZSTD_decompress");
         static zstd_iserr_fn ziserr = (zstd_iserr_fn)zsym("ZSTD_isError");
         static zstd_size_fn zsize =
             (zstd_size_fn)zsym("ZSTD_getFrameContentSize");
-        uint8_t szb[4];
+        uint8_t hdr[HEADER_LEN];
         while (!stop.load(std::memory_order_relaxed)) {
-            if (!read_exact(szb, 4)) break;
-            uint64_t size = ((uint64_t)szb[0] << 24) |
-                            ((uint64_t)szb[1] << 16) |
-                            ((uint64_t)szb[2] << 8) | szb[3];
+            if (!read_exact(hdr, HEADER_LEN)) break;
+            uint64_t size = ((uint64_t)hdr[0] << 24) |
+                            ((uint64_t)hdr[1] << 16) |
+                            ((uint64_t)hdr[2] << 8) | hdr[3];
             if (size < (uint64_t)HEADER_LEN || size > MAX_FRAME) {
                 bad_frames.fetch_add(1, std::memory_order_relaxed);
                 break;  // stream is desynced; drop the connection
             }
-            if (fbuf.size() < size) fbuf.resize(size);
-            if (!read_exact(fbuf.data(), size - 4)) break;
             wire_bytes.fetch_add(size, std::memory_order_relaxed);
-            // frame[4]=msg_type -> fbuf[0]; frame[7]=encoder -> fbuf[3];
-            // payload at frame[19] -> fbuf[15]
-            int msg_type = fbuf[0];
-            int encoder = fbuf[3];
-            // header (framing.py): team u32 @8, org u16 @12, agent u16
-            // @16 — all relative to the frame, fbuf starts at offset 4
-            uint32_t team = (uint32_t)fbuf[4] | ((uint32_t)fbuf[5] << 8) |
-                            ((uint32_t)fbuf[6] << 16) |
-                            ((uint32_t)fbuf[7] << 24);
-            uint16_t org = (uint16_t)(fbuf[8] | (fbuf[9] << 8));
-            uint16_t agent = (uint16_t)(fbuf[12] | (fbuf[13] << 8));
+            int msg_type = hdr[4];
+            int encoder = hdr[7];
+            uint32_t team = (uint32_t)hdr[8] | ((uint32_t)hdr[9] << 8) |
+                            ((uint32_t)hdr[10] << 16) |
+                            ((uint32_t)hdr[11] << 24);
+            uint16_t org = (uint16_t)(hdr[12] | (hdr[13] << 8));
+            uint16_t agent = (uint16_t)(hdr[16] | (hdr[17] << 8));
             uint64_t meta = (uint64_t)(uint8_t)msg_type |
                             ((uint64_t)agent << 8) |
                             ((uint64_t)org << 24) |
                             ((uint64_t)team << 40);
-            const uint8_t* pay = fbuf.data() + (HEADER_LEN - 4);
             uint64_t pn = size - HEADER_LEN;
-            if (accept_type >= 0 && msg_type != accept_type) continue;
-            if (encoder == ENCODER_ZSTD) {
+            bool want_it = accept_type < 0 || msg_type == accept_type;
+            if (!want_it) {
+                // skip the payload without publishing
+                uint64_t left = pn;
+                uint8_t sink[4096];
+                bool ok = true;
+                while (left) {
+                    uint64_t c = left < sizeof(sink) ? left : sizeof(sink);
+                    if (!read_exact(sink, c)) { ok = false; break; }
+                    left -= c;
+                }
+                if (!ok) break;
+                continue;
+            }
+            if (encoder == ENCODER_RAW) {
+                // recv STRAIGHT into the reserved ring entry: the only
+                // copy between the socket and pinned memory
+                uint64_t h;
+                uint8_t* dst = begin_entry(pn, meta, h);
+                if (dst == nullptr) break;
+                if (!read_exact(dst, pn)) break;
+                commit(h, pn);
+            } else if (encoder == ENCODER_ZSTD) {
+                if (fbuf.size() < pn) fbuf.resize(pn);
+                if (!read_exact(fbuf.data(), pn)) break;
                 if (!zdec || !ziserr) {
                     bad_frames.fetch_add(1, std::memory_order_relaxed);
                     continue;
                 }
                 unsigned long long want =
-                    zsize ? zsize(pay, pn) : (unsigned long long)-1;
-                uint64_t capg = (want != (unsigned long long)-1 &&
-                                 want != (unsigned long long)-2 && want)
-                                    ? (uint64_t)want
-                                    : pn * 20 + (1u << 20);
-                for (;;) {
-                    if (dbuf.size() < capg) dbuf.resize(capg);
-                    size_t r = zdec(dbuf.data(), capg, pay, pn);
-                    if (!ziserr(r)) {
-                        publish(dbuf.data(), r, meta);
-                        break;
-                    }
-                    capg *= 4;
-                    if (capg > MAX_FRAME * 64) {
+                    zsize ? zsize(fbuf.data(), pn)
+                          : (unsigned long long)-1;
+                if (want != (unsigned long long)-1 &&
+                    want != (unsigned long long)-2 && want &&
+                    want + 64 < cap / 2) {
+                    // known content size: decompress straight into the
+                    // reserved ring entry
+                    uint64_t h;
+                    uint8_t* dst = begin_entry(want, meta, h);
+                    if (dst == nullptr) break;
+                    size_t r = zdec(dst, want, fbuf.data(), pn);
+                    if (ziserr(r)) {
+                        // commit a zero-length entry to keep cursors
+                        // consistent, count the failure
+                        commit(h, 0);
                         bad_frames.fetch_add(1,
                                              std::memory_order_relaxed);
-                        break;
+                    } else {
+                        commit(h, r);
+                    }
+                } else {
+                    uint64_t capg = pn * 20 + (1u << 20);
+                    for (;;) {
+                        if (dbuf.size() < capg) dbuf.resize(capg);
+                        size_t r = zdec(dbuf.data(), capg, fbuf.data(),
+                                        pn);
+                        if (!ziserr(r)) {
+                            publish(dbuf.data(), r, meta);
+                            break;
+                        }
+                        capg *= 4;
+                        if (capg > MAX_FRAME * 64) {
+                            bad_frames.fetch_add(
+                                1, std::memory_order_relaxed);
+                            break;
+                        }
                     }
                 }
-            } else if (encoder == ENCODER_RAW) {
-                publish(pay, pn, meta);
             } else {
                 bad_frames.fetch_add(1, std::memory_order_relaxed);
             }
