@@ -354,7 +354,7 @@ def e2e_main(args) -> None:
         rx.start()
     import socket as _socket
 
-    n_streams = max(1, int(os.environ.get("DF_E2E_STREAMS", "2")))
+    n_streams = max(1, int(os.environ.get("DF_E2E_STREAMS", "3")))
 
     def send_frames(k: int) -> None:
         # N parallel sender connections (agents are many); each batch's

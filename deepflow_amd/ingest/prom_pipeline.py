@@ -1,18 +1,22 @@
-"""Prometheus remote-write ingest with SmartEncoding label interning.
+"""Prometheus remote-write pipeline: SmartEncoding for metrics.
 
-Reference counterpart: server/ingester/prometheus (grpc_label_ids.go caches +
-fully ID-encoded prometheus.samples rows, prometheus_sample.go:106-122).
-Metric names, label names and label values are interned once; samples are
-stored with zero strings: (metric_id, series_id, ts, value) plus the
-series -> [(label_name_id, label_value_id)] layout.
+The reference encodes every metric/label/value string to a global int ID
+via the controller (prometheus encoder) and stores
+ID-encoded rows in ClickHouse (prometheus_sample schema). Here the IDs
+come from the same controller-global allocator (when wired) and the
+sample columns live in torch tensors on the server's device — on a GPU
+server the sample store is HBM-resident like the span store, and series
+extraction is a device mask/gather instead of a Python scan.
 
-This is the highest-cardinality dictionary stress case; the store is
-host-side (external metric volumes are far below span volumes), the
-encoding discipline is identical to the GPU span path.
+Reference: server/ingester/prometheus/decoder (label encoding via
+GetPrometheusLabelIDs), appendix D of SURVEY.md.
 """
 from __future__ import annotations
 
 from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
 
 from ..utils.stats import Counter
 from ..wire import pb, prompb
@@ -24,10 +28,10 @@ class Interner:
         self.from_id: List[str] = []
 
     def name(self, i: int) -> str:
-        return self.from_id[i]
+        return self.from_id[i] if 0 <= i < len(self.from_id) else ""
 
     def strings(self):
-        return iter(self.from_id)
+        return self.from_id
 
     def intern(self, s: str) -> int:
         i = self.to_id.get(s)
@@ -39,9 +43,8 @@ class Interner:
 
 
 class GlobalInterner(Interner):
-    """Interner backed by the controller\'s persistent id allocator
-    (reference: grpc_label_ids.go slow path -> GetPrometheusLabelIDs).
-    Local cache, controller miss-fill; ids survive restarts."""
+    """Controller-backed: ids are cluster-global and persistent
+    (GetPrometheusLabelIDs analog; round-1 weakness #48)."""
 
     def __init__(self, kind: str, alloc_fn):
         super().__init__()
@@ -50,10 +53,10 @@ class GlobalInterner(Interner):
         self.from_id_map: Dict[int, str] = {}
 
     def name(self, i: int) -> str:
-        return self.from_id_map[i]
+        return self.from_id_map.get(i, "")
 
     def strings(self):
-        return iter(self.from_id_map.values())
+        return list(self.from_id_map.values())
 
     def intern(self, s: str) -> int:
         i = self.to_id.get(s)
@@ -64,9 +67,91 @@ class GlobalInterner(Interner):
         return i
 
 
+class _SampleColumns:
+    """Append-only columnar sample store in torch tensors (series id,
+    timestamp ms, value). Appends stage in host lists and flush into
+    doubling device tensors — on a GPU server these columns are
+    HBM-resident and matcher scans are device ops."""
+
+    def __init__(self, device: str = "cpu"):
+        self.device = torch.device(device)
+        self.n = 0
+        cap = 1 << 12
+        self.series = torch.empty(cap, dtype=torch.int32,
+                                  device=self.device)
+        self.ts = torch.empty(cap, dtype=torch.int64, device=self.device)
+        self.value = torch.empty(cap, dtype=torch.float64,
+                                 device=self.device)
+        self._st_series: List[int] = []
+        self._st_ts: List[int] = []
+        self._st_value: List[float] = []
+
+    def append(self, sid: int, ts: int, value: float) -> None:
+        self._st_series.append(sid)
+        self._st_ts.append(ts)
+        self._st_value.append(value)
+
+    def _ensure(self, need: int) -> None:
+        cap = self.series.numel()
+        if need <= cap:
+            return
+        while cap < need:
+            cap *= 2
+        for name in ("series", "ts", "value"):
+            old = getattr(self, name)
+            new = torch.empty(cap, dtype=old.dtype, device=self.device)
+            new[: self.n] = old[: self.n]
+            setattr(self, name, new)
+
+    def flush(self) -> None:
+        k = len(self._st_series)
+        if k == 0:
+            return
+        self._ensure(self.n + k)
+        dev = self.device
+        self.series[self.n: self.n + k] = torch.from_numpy(
+            np.asarray(self._st_series, dtype=np.int32)).to(dev)
+        self.ts[self.n: self.n + k] = torch.from_numpy(
+            np.asarray(self._st_ts, dtype=np.int64)).to(dev)
+        self.value[self.n: self.n + k] = torch.from_numpy(
+            np.asarray(self._st_value, dtype=np.float64)).to(dev)
+        self.n += k
+        self._st_series.clear()
+        self._st_ts.clear()
+        self._st_value.clear()
+
+    def for_series(self, sids: List[int]):
+        """{sid: (ts_s ndarray, value ndarray)} via one device pass."""
+        self.flush()
+        if self.n == 0 or not sids:
+            return {}
+        col = self.series[: self.n]
+        want = torch.tensor(sids, dtype=torch.int32, device=self.device)
+        mask = torch.isin(col, want)
+        idx = mask.nonzero(as_tuple=True)[0]
+        sel_sid = col[idx].cpu().numpy()
+        sel_ts = (self.ts[: self.n][idx] // 1000).cpu().numpy()
+        sel_val = self.value[: self.n][idx].cpu().numpy()
+        out = {}
+        for sid in sids:
+            m = sel_sid == sid
+            out[sid] = (sel_ts[m], sel_val[m])
+        return out
+
+    def stored_bytes(self) -> int:
+        self.flush()
+        return self.n * (4 + 8 + 8)
+
+    def counts_per_series(self) -> np.ndarray:
+        self.flush()
+        if self.n == 0:
+            return np.zeros(0, dtype=np.int64)
+        return np.bincount(self.series[: self.n].cpu().numpy())
+
+
 class PromPipeline:
     def __init__(self, counter: Optional[Counter] = None,
-                 id_allocator=None):
+                 id_allocator=None, device: str = "cpu"):
         """id_allocator: ControllerLite.alloc_prom_ids-shaped callable —
         when given, metric/label ids are controller-global + persistent
         instead of per-shard volatile (round-1 weakness #48)."""
@@ -82,10 +167,7 @@ class PromPipeline:
         self.series: Dict[Tuple, int] = {}
         self.series_labels: List[Tuple] = []
         self.series_metric: List[int] = []
-        # columnar samples
-        self.s_series: List[int] = []
-        self.s_ts: List[int] = []      # ms
-        self.s_value: List[float] = []
+        self.samples = _SampleColumns(device)
         self.counter = counter or Counter("ingester.prometheus")
 
     def _series_id(self, metric: str, lab_ids) -> int:
@@ -109,9 +191,7 @@ class PromPipeline:
                         self.label_values.intern(str(v)))
                        for k, v in labels.items()]
             sid = self._series_id(metric, lab_ids)
-            self.s_series.append(sid)
-            self.s_ts.append(int(ts))
-            self.s_value.append(float(value))
+            self.samples.append(sid, int(ts), float(value))
             n += 1
         self.counter.add("samples_in", n)
         return n
@@ -132,9 +212,8 @@ class PromPipeline:
                                 self.label_values.intern(value)))
             sid = self._series_id(metric, lab_ids)
             for sm in ts.get("samples", []):
-                self.s_series.append(sid)
-                self.s_ts.append(int(sm.get("timestamp", 0)))
-                self.s_value.append(float(sm.get("value", 0.0)))
+                self.samples.append(sid, int(sm.get("timestamp", 0)),
+                                    float(sm.get("value", 0.0)))
                 n += 1
         self.counter.add("samples_in", n)
         return n
@@ -147,7 +226,7 @@ class PromPipeline:
         mid = self.metric_names.to_id.get(metric)
         if mid is None:
             return []
-        out = []
+        matched: List[Tuple[int, Dict[str, str]]] = []
         for sid, smid in enumerate(self.series_metric):
             if smid != mid:
                 continue
@@ -165,12 +244,13 @@ class PromPipeline:
                     ok = False
                 if not ok:
                     break
-            if not ok:
-                continue
-            samples: Dict[int, float] = {}
-            for i, s in enumerate(self.s_series):
-                if s == sid:
-                    samples[self.s_ts[i] // 1000] = self.s_value[i]
+            if ok:
+                matched.append((sid, labels))
+        per_sid = self.samples.for_series([sid for sid, _ in matched])
+        out = []
+        for sid, labels in matched:
+            ts_arr, val_arr = per_sid.get(sid, ((), ()))
+            samples = {int(t): float(v) for t, v in zip(ts_arr, val_arr)}
             out.append({"metric": dict(labels, __name__=metric),
                         "samples": samples,
                         # raw scraped series are cumulative counters/gauges,
@@ -180,7 +260,7 @@ class PromPipeline:
 
     def stored_bytes(self) -> int:
         """SmartEncoding accounting: ID-encoded samples + dictionaries."""
-        samples = len(self.s_series) * (4 + 8 + 8)
+        samples = self.samples.stored_bytes()
         dicts = sum(len(s) for s in self.metric_names.strings()) + \
             sum(len(s) for s in self.label_names.strings()) + \
             sum(len(s) for s in self.label_values.strings())
@@ -189,13 +269,16 @@ class PromPipeline:
 
     def naive_bytes(self) -> int:
         """What the same samples cost with string labels per row."""
+        counts = self.samples.counts_per_series()
         total = 0
-        for i, sid in enumerate(self.s_series):
+        for sid, cnt in enumerate(counts):
+            if cnt == 0:
+                continue
             row = 8 + 8
             mid = self.series_metric[sid]
             row += len(self.metric_names.name(mid))
             for ln, lv in self.series_labels[sid]:
                 row += len(self.label_names.name(ln)) + \
                     len(self.label_values.name(lv))
-            total += row
+            total += row * int(cnt)
         return total

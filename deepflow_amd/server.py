@@ -72,7 +72,7 @@ class DeepflowServer:
         self.receiver.register(framing.MSG_METRICS,
                                lambda hdr, payload:
                                self.docs.ingest_payload(payload.tobytes()))
-        self.prom = PromPipeline()
+        self.prom = PromPipeline(device=device)
         self.receiver.register(framing.MSG_PROMETHEUS,
                                lambda hdr, payload:
                                self.prom.ingest_write_request(

@@ -41,7 +41,7 @@ def test_otel_push(stack):
 def test_prometheus_push(stack):
     srv, client, ic = stack
     client.post("/api/v1/prometheus", content=mk_write_request())
-    assert _wait(lambda: len(srv.prom.s_series) >= 20)
+    assert _wait(lambda: srv.prom.samples.n + len(srv.prom.samples._st_series) >= 20)
 
 
 def test_profile_push(stack):

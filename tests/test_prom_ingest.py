@@ -32,7 +32,8 @@ def test_prom_ingest_and_query(server):
     hdr = framing.FrameHeader(msg_type=framing.MSG_PROMETHEUS, agent_id=2)
     frame = framing.encode_frame(hdr, mk_write_request())
     assert server.receiver.handle_frame(frame)
-    assert len(server.prom.s_series) == 20
+    server.prom.samples.flush()
+    assert server.prom.samples.n == 20
     # SmartEncoding: id-encoded storage beats naive strings
     assert server.prom.stored_bytes() < server.prom.naive_bytes()
     # PromQL over the ingested series
