@@ -84,3 +84,24 @@ def test_controller_global_label_ids_persist():
     # series still hydrate through the global interners
     series = pipe2.series_for("http_requests_total", [])
     assert series and series[0]["metric"]["job"] == "api"
+
+
+@pytest.mark.gpu
+def test_prom_device_store_gpu():
+    """Sample columns live in HBM on a GPU server; series extraction is
+    a device mask/gather and matches the CPU result exactly."""
+    from deepflow_amd.ingest.prom_pipeline import PromPipeline
+    rows = [("rpc_latency", {"svc": f"s{i % 3}"}, 1000 * i, float(i))
+            for i in range(500)]
+    gpu = PromPipeline(device="cuda")
+    cpu = PromPipeline(device="cpu")
+    gpu.ingest_labeled_samples(rows)
+    cpu.ingest_labeled_samples(rows)
+    assert gpu.samples.series.device.type == "cuda"
+    for matchers in ([], [("svc", "=", "s1")], [("svc", "=~", "s[02]")]):
+        a = gpu.series_for("rpc_latency", matchers)
+        b = cpu.series_for("rpc_latency", matchers)
+        assert [s["samples"] for s in a] == [s["samples"] for s in b]
+        assert [s["metric"] for s in a] == [s["metric"] for s in b]
+    assert gpu.stored_bytes() == cpu.stored_bytes()
+    assert gpu.naive_bytes() == cpu.naive_bytes()
