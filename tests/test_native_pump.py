@@ -227,3 +227,59 @@ def test_server_native_pump_data_plane():
         assert r["values"][0][0] == 50
     finally:
         srv.stop()
+
+
+@pytest.mark.gpu
+def test_pump_to_gpu_pipeline():
+    """Wire -> native pump (pinned ring) -> H2D -> GPU ingest: spans
+    land in the device store and are queryable (the bench --path e2e
+    fast path, minimally)."""
+    import ctypes as ct
+    import torch
+    from deepflow_amd.gen import SpanGenConfig
+    from deepflow_amd.gen.spans import gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.store.kg import (KnowledgeGraphTable,
+                                       default_platform)
+    from deepflow_amd.ops import native as nat
+
+    cfg = SpanGenConfig(n=5000, seed=11, tag_cardinality=300, n_ips=128)
+    kg = KnowledgeGraphTable(capacity_pow2=1 << 12, device="cuda")
+    kg.update(default_platform(cfg))
+    pipe = L7IngestPipeline(device="cuda", segment_rows=1 << 14, kg=kg,
+                            dict_capacity=1 << 15,
+                            time_base_s=cfg.base_time_ns // 10**9,
+                            defer_harvest=False)
+    lib = nat.cpu()
+    offs_p = torch.empty(5000, dtype=torch.int32, pin_memory=True)
+    lens_p = torch.empty(5000, dtype=torch.int32, pin_memory=True)
+
+    def on_frame(view, meta):
+        n = int(lib.df_scan_offsets(
+            ct.c_void_p(view.ctypes.data), len(view),
+            ct.c_void_p(offs_p.data_ptr()),
+            ct.c_void_p(lens_p.data_ptr()), 5000))
+        pay = torch.from_numpy(view).to("cuda", non_blocking=True)
+        pipe.ingest_device(pay, offs_p[:n].to("cuda", non_blocking=True),
+                           lens_p[:n].to("cuda", non_blocking=True),
+                           view)
+        ev = torch.cuda.Event()
+        ev.record()
+        return ev
+
+    srv = PumpServer(on_frame, pin=True, ring_bytes=64 << 20).start()
+    try:
+        payload = gen_span_payload(cfg)
+        s = socket.create_connection(("127.0.0.1", srv.port))
+        s.sendall(_frame(payload, zstd=True))
+        t_end = time.time() + 20
+        while pipe.stats.spans_in < 5000 and time.time() < t_end:
+            time.sleep(0.02)
+        s.close()
+        torch.cuda.synchronize()
+        assert pipe.stats.spans_in == 5000
+        assert srv.stats()["bad_frames"] == 0
+        rows = pipe.metrics.rows()
+        assert sum(r["request"] for r in rows) == 5000
+    finally:
+        srv.stop()
