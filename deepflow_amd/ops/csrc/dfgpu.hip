@@ -1442,104 +1442,84 @@ DEV uint32_t group_claim(uint64_t h, const uint64_t* kraw, uint32_t n_keys,
     return 0;
 }
 
-__global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_row,
-                            uint64_t* __restrict__ gkeys,
-                            uint64_t* __restrict__ graw,
-                            unsigned long long* __restrict__ gvals,
-                            uint32_t cap_mask) {
-    // Grid-strided scan with an LDS-resident group table: each workgroup
-    // aggregates its stripe into shared memory (~37 KB of the 160 KB/CU
-    // LDS: 512 slots x (key + global-slot + 8 accumulators)) and flushes
-    // once at the end — global atomics drop from per-row to
-    // per-(block x live slot). Groups are identified by the 64-bit mixed
-    // key hash (same convention as the global table); the LDS claimant
-    // registers the group globally immediately, so raw keys never stage
-    // in LDS. Rows whose key misses the table (very high per-block
-    // cardinality) fall back to per-lane global accumulation.
-    constexpr uint32_t NSLOT = 512;
-    __shared__ uint64_t lkey[NSLOT];
-    __shared__ uint32_t lgslot[NSLOT];
-    __shared__ unsigned long long lagg[NSLOT][QMAX_AGGS];
-    for (uint32_t sl = threadIdx.x; sl < NSLOT; sl += blockDim.x) {
+constexpr uint32_t QAGG_NSLOT = 512;
+
+// One row's LDS-table aggregation (shared by the direct and the
+// partitioned kernels): probe/claim the block-local table, accumulate;
+// on LDS saturation fall through to the global table directly.
+DEV void agg_row_lds(const SegView& s, const QuerySpec& q, uint64_t row,
+                     uint64_t* lkey, uint32_t* lgslot,
+                     unsigned long long (*lagg)[QMAX_AGGS],
+                     uint64_t* gkeys, uint64_t* graw,
+                     unsigned long long* gvals, uint32_t cap_mask) {
+    uint64_t kraw[QMAX_KEYS];
+    uint64_t h = 0x243F6A8885A308D3ull;
+    for (uint32_t k = 0; k < q.n_keys; k++) {
+        kraw[k] = src_value(s, row, q.keys[k].family, q.keys[k].idx,
+                            q.keys[k].bucket, q.time_base_s);
+        h = mix64(h ^ kraw[k] ^ ((uint64_t)k << 56));
+    }
+    if (h == EMPTY_KEY) h = 1;
+    uint64_t varr[QMAX_AGGS];
+    for (uint32_t a = 0; a < q.n_aggs; a++)
+        varr[a] = q.aggs[a].op == AGGOP_COUNT ? 1
+            : src_value(s, row, q.aggs[a].family, q.aggs[a].idx, 0,
+                        q.time_base_s);
+    uint32_t slot = (uint32_t)h & (QAGG_NSLOT - 1);
+    uint32_t gslot = 0xFFFFFFFFu;
+    for (uint32_t probe = 0; probe < 16; probe++) {
+        uint64_t cur = lkey[slot];
+        if (cur == EMPTY_KEY) {
+            uint64_t old = atomicCAS((unsigned long long*)&lkey[slot],
+                                     (unsigned long long)EMPTY_KEY,
+                                     (unsigned long long)h);
+            if (old == EMPTY_KEY) {
+                lgslot[slot] = group_claim(h, kraw, q.n_keys, gkeys, graw,
+                                           cap_mask);
+                __threadfence_block();
+                cur = h;
+            } else {
+                cur = old;
+            }
+        }
+        if (cur == h) {
+            gslot = slot;
+            break;
+        }
+        slot = (slot + 1) & (QAGG_NSLOT - 1);
+    }
+    unsigned long long* acc;
+    if (gslot != 0xFFFFFFFFu) {
+        acc = lagg[gslot];
+    } else {  // LDS table saturated for this key: go global directly
+        uint32_t g = group_claim(h, kraw, q.n_keys, gkeys, graw, cap_mask);
+        acc = &gvals[(uint64_t)g * QMAX_AGGS];
+    }
+    for (uint32_t a = 0; a < q.n_aggs; a++) {
+        uint32_t op = q.aggs[a].op;
+        if (op == AGGOP_COUNT || op == AGGOP_SUM)
+            atomicAdd(&acc[a], (unsigned long long)varr[a]);
+        else if (op == AGGOP_MIN)
+            atomicMin(&acc[a], (unsigned long long)varr[a]);
+        else
+            atomicMax(&acc[a], (unsigned long long)varr[a]);
+    }
+}
+
+DEV void agg_lds_init(const QuerySpec& q, uint64_t* lkey,
+                      unsigned long long (*lagg)[QMAX_AGGS]) {
+    for (uint32_t sl = threadIdx.x; sl < QAGG_NSLOT; sl += blockDim.x) {
         lkey[sl] = EMPTY_KEY;
         for (uint32_t a = 0; a < q.n_aggs; a++)
             lagg[sl][a] = q.aggs[a].op == AGGOP_MIN ? ~0ull : 0ull;
     }
-    __syncthreads();
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < n; i += (uint64_t)gridDim.x * blockDim.x) {
-        uint64_t row = base_row + i;
-        if (!eval_terms(s, row, q)) continue;
-        uint64_t kraw[QMAX_KEYS];
-        uint64_t h = 0x243F6A8885A308D3ull;
-        for (uint32_t k = 0; k < q.n_keys; k++) {
-            kraw[k] = src_value(s, row, q.keys[k].family, q.keys[k].idx,
-                                q.keys[k].bucket, q.time_base_s);
-            h = mix64(h ^ kraw[k] ^ ((uint64_t)k << 56));
-        }
-        if (h == EMPTY_KEY) h = 1;
-        uint64_t varr[QMAX_AGGS];
-        for (uint32_t a = 0; a < q.n_aggs; a++)
-            varr[a] = q.aggs[a].op == AGGOP_COUNT ? 1
-                : src_value(s, row, q.aggs[a].family, q.aggs[a].idx, 0,
-                            q.time_base_s);
-        // LDS probe; the claimant registers the group globally at claim
-        // time, so the block-local slot carries the global slot id and
-        // no raw keys need staging in LDS
-        uint32_t slot = (uint32_t)h & (NSLOT - 1);
-        uint32_t gslot = 0xFFFFFFFFu;
-        for (uint32_t probe = 0; probe < 16; probe++) {
-            uint64_t cur = lkey[slot];
-            if (cur == EMPTY_KEY) {
-                uint64_t old = atomicCAS(
-                    (unsigned long long*)&lkey[slot],
-                    (unsigned long long)EMPTY_KEY, (unsigned long long)h);
-                if (old == EMPTY_KEY) {
-                    lgslot[slot] = group_claim(h, kraw, q.n_keys, gkeys,
-                                               graw, cap_mask);
-                    __threadfence_block();
-                    cur = h;
-                } else {
-                    cur = old;
-                }
-            }
-            if (cur == h) {
-                gslot = slot;
-                break;
-            }
-            slot = (slot + 1) & (NSLOT - 1);
-        }
-        if (gslot != 0xFFFFFFFFu) {
-            for (uint32_t a = 0; a < q.n_aggs; a++) {
-                uint32_t op = q.aggs[a].op;
-                if (op == AGGOP_COUNT || op == AGGOP_SUM)
-                    atomicAdd(&lagg[gslot][a],
-                              (unsigned long long)varr[a]);
-                else if (op == AGGOP_MIN)
-                    atomicMin(&lagg[gslot][a],
-                              (unsigned long long)varr[a]);
-                else
-                    atomicMax(&lagg[gslot][a],
-                              (unsigned long long)varr[a]);
-            }
-        } else {  // LDS table saturated for this key: go global directly
-            uint32_t g = group_claim(h, kraw, q.n_keys, gkeys, graw,
-                                     cap_mask);
-            unsigned long long* acc = &gvals[(uint64_t)g * QMAX_AGGS];
-            for (uint32_t a = 0; a < q.n_aggs; a++) {
-                uint32_t op = q.aggs[a].op;
-                if (op == AGGOP_COUNT || op == AGGOP_SUM)
-                    atomicAdd(&acc[a], (unsigned long long)varr[a]);
-                else if (op == AGGOP_MIN)
-                    atomicMin(&acc[a], (unsigned long long)varr[a]);
-                else
-                    atomicMax(&acc[a], (unsigned long long)varr[a]);
-            }
-        }
-    }
-    __syncthreads();
-    // flush the block-local accumulators into the global group store
-    for (uint32_t sl = threadIdx.x; sl < NSLOT; sl += blockDim.x) {
+}
+
+DEV void agg_lds_flush(const QuerySpec& q, const uint64_t* lkey,
+                       const uint32_t* lgslot,
+                       unsigned long long (*lagg)[QMAX_AGGS],
+                       unsigned long long* gvals) {
+    for (uint32_t sl = threadIdx.x; sl < QAGG_NSLOT; sl += blockDim.x) {
         if (lkey[sl] == EMPTY_KEY) continue;
         unsigned long long* acc = &gvals[(uint64_t)lgslot[sl] * QMAX_AGGS];
         for (uint32_t a = 0; a < q.n_aggs; a++) {
@@ -1554,6 +1534,142 @@ __global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_ro
             }
         }
     }
+}
+
+__global__ void k_query_agg(SegView s, QuerySpec q, uint32_t n, uint64_t base_row,
+                            uint64_t* __restrict__ gkeys,
+                            uint64_t* __restrict__ graw,
+                            unsigned long long* __restrict__ gvals,
+                            uint32_t cap_mask) {
+    // Grid-strided scan with an LDS-resident group table: each workgroup
+    // aggregates its stripe into shared memory (~37 KB of the 160 KB/CU
+    // LDS: 512 slots x (key + global-slot + 8 accumulators)) and flushes
+    // once at the end — global atomics drop from per-row to
+    // per-(block x live slot). Groups are identified by the 64-bit mixed
+    // key hash (same convention as the global table); the LDS claimant
+    // registers the group globally immediately, so raw keys never stage
+    // in LDS. Rows whose key misses the table (very high per-block
+    // cardinality) fall back to per-lane global accumulation.
+    __shared__ uint64_t lkey[QAGG_NSLOT];
+    __shared__ uint32_t lgslot[QAGG_NSLOT];
+    __shared__ unsigned long long lagg[QAGG_NSLOT][QMAX_AGGS];
+    agg_lds_init(q, lkey, lagg);
+    __syncthreads();
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (uint64_t)gridDim.x * blockDim.x) {
+        uint64_t row = base_row + i;
+        if (!eval_terms(s, row, q)) continue;
+        agg_row_lds(s, q, row, lkey, lgslot, lagg, gkeys, graw, gvals,
+                    cap_mask);
+    }
+    __syncthreads();
+    agg_lds_flush(q, lkey, lgslot, lagg, gvals);
+}
+
+// ---- radix-partitioned group-by (high cardinality) ----
+// When the key cardinality is far above QAGG_NSLOT, the direct kernel
+// spills most rows to per-row global atomics. The partitioned path
+// buckets rows by 8 hash bits first, then aggregates one bucket per
+// workgroup group: each block then sees ~1/256th of the distinct keys,
+// so the LDS table holds them all (zero spill up to ~128k groups).
+constexpr uint32_t QPART_NB = 256;
+constexpr uint32_t QPART_SHIFT = 40;
+
+DEV uint64_t qpart_hash(const SegView& s, const QuerySpec& q,
+                        uint64_t row) {
+    uint64_t h = 0x243F6A8885A308D3ull;
+    for (uint32_t k = 0; k < q.n_keys; k++) {
+        uint64_t kv = src_value(s, row, q.keys[k].family, q.keys[k].idx,
+                                q.keys[k].bucket, q.time_base_s);
+        h = mix64(h ^ kv ^ ((uint64_t)k << 56));
+    }
+    return h == EMPTY_KEY ? 1 : h;
+}
+
+__global__ void k_qpart_count(SegView s, QuerySpec q, uint32_t n,
+                              uint64_t base_row,
+                              uint32_t* __restrict__ counts) {
+    __shared__ uint32_t lc[QPART_NB];
+    for (uint32_t i = threadIdx.x; i < QPART_NB; i += blockDim.x)
+        lc[i] = 0;
+    __syncthreads();
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (uint64_t)gridDim.x * blockDim.x) {
+        uint64_t row = base_row + i;
+        if (!eval_terms(s, row, q)) continue;
+        uint64_t h = qpart_hash(s, q, row);
+        atomicAdd(&lc[(h >> QPART_SHIFT) & (QPART_NB - 1)], 1u);
+    }
+    __syncthreads();
+    for (uint32_t i = threadIdx.x; i < QPART_NB; i += blockDim.x)
+        if (lc[i]) atomicAdd(&counts[i], lc[i]);
+}
+
+__global__ void k_qpart_scatter(SegView s, QuerySpec q, uint32_t n,
+                                uint64_t base_row,
+                                uint32_t* __restrict__ cursors,
+                                uint32_t* __restrict__ out_rows) {
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (uint64_t)gridDim.x * blockDim.x) {
+        uint64_t row = base_row + i;
+        if (!eval_terms(s, row, q)) continue;
+        uint64_t h = qpart_hash(s, q, row);
+        uint32_t b = (uint32_t)(h >> QPART_SHIFT) & (QPART_NB - 1);
+        uint32_t pos = atomicAdd(&cursors[b], 1u);
+        out_rows[pos] = (uint32_t)i;
+    }
+}
+
+// single-block exclusive scan over the QPART_NB counters:
+// counts[i] (in: per-bucket count) -> counts[i] = start offset (out);
+// cursors[i] = start offset too (the scatter pass advances cursors to
+// the bucket ends, which k_qpart_agg uses as `hi`... but scatter order
+// is nondeterministic, so ends must be start+count: write both).
+__global__ void k_qpart_prefix(uint32_t* counts, uint32_t* cursors) {
+    __shared__ uint32_t vals[QPART_NB];
+    uint32_t i = threadIdx.x;
+    vals[i] = counts[i];
+    __syncthreads();
+    // simple serial scan by lane 0 (256 values; latency-trivial)
+    if (i == 0) {
+        uint32_t acc = 0;
+        for (uint32_t k = 0; k < QPART_NB; k++) {
+            uint32_t c = vals[k];
+            counts[k] = acc;        // start
+            cursors[k] = acc;       // scatter cursor starts here
+            acc += c;
+            vals[k] = acc;          // end (reused below)
+        }
+    }
+    __syncthreads();
+}
+
+__global__ void k_qpart_agg(SegView s, QuerySpec q,
+                            const uint32_t* __restrict__ rows,
+                            const uint32_t* __restrict__ starts,
+                            const uint32_t* __restrict__ ends,
+                            uint64_t base_row,
+                            uint64_t* __restrict__ gkeys,
+                            uint64_t* __restrict__ graw,
+                            unsigned long long* __restrict__ gvals,
+                            uint32_t cap_mask,
+                            uint32_t blocks_per_bucket) {
+    __shared__ uint64_t lkey[QAGG_NSLOT];
+    __shared__ uint32_t lgslot[QAGG_NSLOT];
+    __shared__ unsigned long long lagg[QAGG_NSLOT][QMAX_AGGS];
+    agg_lds_init(q, lkey, lagg);
+    __syncthreads();
+    uint32_t bucket = blockIdx.x / blocks_per_bucket;
+    uint32_t sub = blockIdx.x % blocks_per_bucket;
+    uint32_t lo = starts[bucket], hi = ends[bucket];
+    // rows were pre-filtered by the scatter pass: no eval_terms here
+    for (uint64_t j = lo + (uint64_t)sub * blockDim.x + threadIdx.x;
+         j < hi; j += (uint64_t)blocks_per_bucket * blockDim.x) {
+        agg_row_lds(s, q, base_row + rows[j], lkey, lgslot, lagg, gkeys,
+                    graw, gvals, cap_mask);
+    }
+    __syncthreads();
+    agg_lds_flush(q, lkey, lgslot, lagg, gvals);
 }
 
 // non-aggregated SELECT: emit matching row ids (bounded)
@@ -1841,6 +1957,57 @@ int df_query_agg(const void* u64c, const void* u32c, const void* u8c,
     hipLaunchKernelGGL(k_query_agg, dim3(blocks), dim3(BLOCK), 0, STREAM(stream),
                        s, q, n, base_row, (uint64_t*)gkeys, (uint64_t*)graw,
                        (unsigned long long*)gvals, cap - 1);
+    return (int)hipGetLastError();
+}
+
+int df_qpart_agg(const void* u64c, const void* u32c, const void* u8c,
+                 const void* didc,
+                 const void* kg_tk, const void* kg_tv, uint32_t kg_cap,
+                 const void* attr_pool,
+                 const void* attr_start, const void* attr_cnt,
+                 const void* str_rowref, const void* str_lens,
+                 const void* pool,
+                 uint64_t stride, uint64_t n_rows,
+                 const void* spec, uint32_t n, uint64_t base_row,
+                 void* counts, void* cursors, void* row_scratch,
+                 void* gkeys, void* graw, void* gvals, uint32_t cap,
+                 uint64_t stream) {
+    // Three-pass radix-partitioned aggregation. counts/cursors are
+    // QPART_NB u32 device buffers; the caller pre-fills cursors with the
+    // exclusive prefix sums of counts AFTER the count pass via
+    // df_qpart_prefix (all device-side; no host sync).
+    SegView s{(const uint64_t*)u64c, (const uint32_t*)u32c, (const uint8_t*)u8c,
+              (const uint32_t*)didc,
+              (const uint64_t*)kg_tk, (const uint32_t*)kg_tv,
+              kg_cap ? kg_cap - 1 : 0,
+              (const int32_t*)attr_pool, (const uint32_t*)attr_start,
+              (const uint8_t*)attr_cnt, (const uint64_t*)str_rowref,
+              (const int16_t*)str_lens,
+              (const uint8_t*)pool, stride, n_rows};
+    QuerySpec q;
+    __builtin_memcpy(&q, spec, sizeof(QuerySpec));
+    uint32_t blocks = grid_for(n);
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL(k_qpart_count, dim3(blocks), dim3(BLOCK), 0,
+                       STREAM(stream), s, q, n, base_row,
+                       (uint32_t*)counts);
+    // device-side exclusive scan of 256 counters (one tiny block) —
+    // writes cursors = starts and counts = ends (inclusive scan)
+    hipLaunchKernelGGL(k_qpart_prefix, dim3(1), dim3(QPART_NB), 0,
+                       STREAM(stream), (uint32_t*)counts,
+                       (uint32_t*)cursors);
+    hipLaunchKernelGGL(k_qpart_scatter, dim3(blocks), dim3(BLOCK), 0,
+                       STREAM(stream), s, q, n, base_row,
+                       (uint32_t*)cursors, (uint32_t*)row_scratch);
+    // after scatter, cursors[b] == ends[b] (inclusive scan) and
+    // starts[b] = ends[b] - count — k_qpart_agg derives lo from the
+    // auxiliary starts written by k_qpart_prefix into counts
+    constexpr uint32_t BPB = 16;
+    hipLaunchKernelGGL(k_qpart_agg, dim3(QPART_NB * BPB), dim3(BLOCK), 0,
+                       STREAM(stream), s, q, (const uint32_t*)row_scratch,
+                       (const uint32_t*)counts, (const uint32_t*)cursors,
+                       base_row, (uint64_t*)gkeys, (uint64_t*)graw,
+                       (unsigned long long*)gvals, cap - 1, BPB);
     return (int)hipGetLastError();
 }
 

@@ -192,6 +192,31 @@ def query_agg(seg, spec_bytes: bytes, base_row: int, n: int,
         _stream()), "df_query_agg")
 
 
+def qpart_agg(seg, spec_bytes: bytes, base_row: int, n: int,
+              counts: torch.Tensor, cursors: torch.Tensor,
+              row_scratch: torch.Tensor,
+              gkeys: torch.Tensor, graw: torch.Tensor,
+              gvals: torch.Tensor, kg=None) -> None:
+    """Radix-partitioned group-by (high key cardinality): bucket rows by
+    8 hash bits, aggregate one bucket per workgroup set — each block's
+    LDS table then holds every key it sees (no per-row global atomics).
+    counts/cursors: zeroed u32[256]; row_scratch: u32[n]."""
+    import ctypes
+    lib = native.gpu()
+    buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))
+    ktk, ktv, kcap = _kg_args(kg)
+    native.check(lib.df_qpart_agg(
+        seg.u64.data_ptr(), seg.u32.data_ptr(), seg.u8.data_ptr(),
+        _opt_ptr(seg, "did"), ktk, ktv, kcap, _opt_ptr(seg, "attr_pool"),
+        _opt_ptr(seg, "attr_start"), _opt_ptr(seg, "attr_cnt"),
+        seg.str_rowref.data_ptr(), seg.str_lens.data_ptr(),
+        seg.pool.data_ptr(), seg.capacity, seg.n_rows,
+        ctypes.addressof(buf), n, base_row,
+        counts.data_ptr(), cursors.data_ptr(), row_scratch.data_ptr(),
+        gkeys.data_ptr(), graw.data_ptr(), gvals.data_ptr(),
+        gkeys.numel(), _stream()), "df_qpart_agg")
+
+
 def _kg_args(kg):
     """KnowledgeGraph table pointers for the query-time join."""
     if kg is None:

@@ -132,6 +132,10 @@ def _decl_gpu(lib: ct.CDLL) -> None:
     lib.df_query_agg.restype = ct.c_int
     lib.df_query_agg.argtypes = [p, p, p, p, p, p, u32, p, p, p, p, p, p,
                                  u64, u64, p, u32, u64, p, p, p, u32, u64]
+    lib.df_qpart_agg.restype = ct.c_int
+    lib.df_qpart_agg.argtypes = [p, p, p, p, p, p, u32, p, p, p, p, p, p,
+                                 u64, u64, p, u32, u64, p, p, p, p, p, p,
+                                 u32, u64]
     lib.df_query_select.restype = ct.c_int
     lib.df_query_select.argtypes = [p, p, p, p, p, p, u32, p, p, p, p, p,
                                     p, u64, u64, p, u32, u64, p, p, u32,

@@ -200,6 +200,14 @@ def execute_agg_cpu(plan: Plan, segments, kg=None) -> List[Dict]:
     return [{"key": list(k), "agg": list(v)} for k, v in groups.items()]
 
 
+# observed group-cardinality per key-set: repeated dashboard queries on a
+# high-cardinality GROUP BY switch to the radix-partitioned kernel after
+# the first run discovers the cardinality
+_CARDINALITY_CACHE: Dict[tuple, int] = {}
+_QPART_MIN_ROWS = 1 << 22
+_QPART_MIN_GROUPS = 1024
+
+
 def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
     from ..ops import gpu_ops
     dev = torch.device(device)
@@ -210,11 +218,26 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
         if a.op == AGGOP_MIN:
             gvals[:, ai] = -1  # 0xFFFF.. as int64
     spec = plan.to_bytes()
+    key_sig = tuple((k.family, k.idx, k.bucket) for k in plan.keys)
+    known_card = _CARDINALITY_CACHE.get(key_sig, 0)
+    scratch = None
     for seg in segments:
         if seg.n_rows == 0:
             continue
-        gpu_ops.query_agg(seg, spec, 0, seg.n_rows, gkeys, graw, gvals,
-                          kg=kg)
+        if known_card >= _QPART_MIN_GROUPS and \
+                seg.n_rows >= _QPART_MIN_ROWS and plan.keys:
+            if scratch is None or scratch[2].numel() < seg.n_rows:
+                scratch = (torch.empty(256, dtype=torch.int32, device=dev),
+                           torch.empty(256, dtype=torch.int32, device=dev),
+                           torch.empty(seg.n_rows, dtype=torch.int32,
+                                       device=dev))
+            scratch[0].zero_()
+            gpu_ops.qpart_agg(seg, spec, 0, seg.n_rows, scratch[0],
+                              scratch[1], scratch[2], gkeys, graw, gvals,
+                              kg=kg)
+        else:
+            gpu_ops.query_agg(seg, spec, 0, seg.n_rows, gkeys, graw,
+                              gvals, kg=kg)
     torch.cuda.synchronize()
     mask = gkeys != 0
     raw = graw[mask].cpu().numpy().view(np.uint64)
@@ -224,6 +247,8 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
     for r in range(raw.shape[0]):
         out.append({"key": [int(x) for x in raw[r, :nk]],
                     "agg": [int(x) for x in vals[r, :na]]})
+    if plan.keys:
+        _CARDINALITY_CACHE[key_sig] = len(out)
     return out
 
 
