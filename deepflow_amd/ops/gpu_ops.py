@@ -200,7 +200,7 @@ def qpart_agg(seg, spec_bytes: bytes, base_row: int, n: int,
     """Radix-partitioned group-by (high key cardinality): bucket rows by
     8 hash bits, aggregate one bucket per workgroup set — each block's
     LDS table then holds every key it sees (no per-row global atomics).
-    counts/cursors: zeroed u32[256]; row_scratch: u32[n]."""
+    counts/cursors: u32[256] (counts zeroed); row_scratch:\n    u64[n * (n_keys+n_aggs)] payload staging."""
     import ctypes
     lib = native.gpu()
     buf = ctypes.create_string_buffer(spec_bytes, len(spec_bytes))

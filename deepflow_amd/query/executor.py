@@ -226,10 +226,11 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
             continue
         if known_card >= _QPART_MIN_GROUPS and \
                 seg.n_rows >= _QPART_MIN_ROWS and plan.keys:
-            if scratch is None or scratch[2].numel() < seg.n_rows:
+            w = len(plan.keys) + len(plan.aggs)
+            if scratch is None or scratch[2].numel() < seg.n_rows * w:
                 scratch = (torch.empty(256, dtype=torch.int32, device=dev),
                            torch.empty(256, dtype=torch.int32, device=dev),
-                           torch.empty(seg.n_rows, dtype=torch.int32,
+                           torch.empty(seg.n_rows * w, dtype=torch.int64,
                                        device=dev))
             scratch[0].zero_()
             gpu_ops.qpart_agg(seg, spec, 0, seg.n_rows, scratch[0],
