@@ -557,10 +557,12 @@ class QueryEngine:
                         {id(m): mi for mi, m in enumerate(q_metas)})
         columns = plan.key_names + plan.agg_names
         rows: List[List] = []
+        # hoist hydration dispatch out of the per-group loop: closures
+        # bound to the host maps, no per-cell string matching
+        hyds = [self._hydrator(meta["hydrate"]) for meta in plan.key_meta]
         for g in groups:
-            row = []
-            for ki, meta in enumerate(plan.key_meta):
-                row.append(self._hydrate(meta["hydrate"], g["key"][ki]))
+            gk = g["key"]
+            row = [hyds[ki](gk[ki]) for ki in range(len(hyds))]
             ai = 0
             for meta in plan.agg_meta:
                 if meta["op"] == "avg":
@@ -819,6 +821,34 @@ class QueryEngine:
         return {"columns": columns, "values": out}
 
     # ----------------------------------------------------------- hydrate
+    def _hydrator(self, how: str):
+        """Column-level hydration closure (bound maps, no per-cell
+        dispatch) — the per-cell _hydrate path costs ~3us x cells on
+        many-group results."""
+        if how == "int":
+            return lambda v: v
+        if how == "time":
+            tb = self.pipe.time_base_s
+            return lambda v: tb + v
+        if how.startswith("dict:"):
+            dom = int(how.split(":")[1])
+            m = self.pipe.dict.id_to_str
+            from ..store import l7_schema as _S
+            inv = _S.DICT_ID_INVALID
+
+            def f(v, m=m, dom=dom, inv=inv):
+                if v < 0 or v == inv:
+                    return None
+                b = m.get((dom, v & 0xFFFFFFFF))
+                return b.decode("utf-8", "replace") if b is not None \
+                    else None
+            return f
+        if how == "l7proto":
+            return lambda v: L7_PROTOCOL_NAMES.get(v, str(v))
+        if how == "status":
+            return lambda v: STATUS_NAMES.get(v, str(v))
+        return lambda v, how=how: self._hydrate(how, v)
+
     def _hydrate(self, how: str, v: int):
         if how == "int":
             return v

@@ -243,11 +243,12 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
     mask = gkeys != 0
     raw = graw[mask].cpu().numpy().view(np.uint64)
     vals = gvals[mask].cpu().numpy().view(np.uint64)
-    out = []
     nk, na = len(plan.keys), len(plan.aggs)
-    for r in range(raw.shape[0]):
-        out.append({"key": [int(x) for x in raw[r, :nk]],
-                    "agg": [int(x) for x in vals[r, :na]]})
+    # .tolist() converts at C speed; per-element numpy indexing was the
+    # dominant wall cost on multi-thousand-group results
+    raw_l = raw[:, :nk].tolist()
+    vals_l = vals[:, :na].tolist()
+    out = [{"key": k, "agg": v} for k, v in zip(raw_l, vals_l)]
     if plan.keys:
         _CARDINALITY_CACHE[key_sig] = len(out)
     return out
