@@ -502,9 +502,18 @@ class QueryEngine:
                     row.append(agg[ai])
                     ai += 1
             rows.append(row)
+        if _qt:
+            import sys as _sys
+            print(f"[qtime] assemble {1e3*(_time.perf_counter()-_t0):.2f}"
+                  f"ms rows={len(rows)}", file=_sys.stderr)
+            _t0 = _time.perf_counter()
         if plan.slimit:
             rows = self._apply_slimit(plan, rows)
         rows = self._order_limit(plan, columns, rows)
+        if _qt:
+            import sys as _sys
+            print(f"[qtime] order {1e3*(_time.perf_counter()-_t0):.2f}ms",
+                  file=_sys.stderr)
         return {"columns": columns, "values": rows}
 
     def _finish_qdata(self, meta, q):
@@ -539,7 +548,15 @@ class QueryEngine:
         kg = kg if kg is not None else self.pipe.kg
         if plan.select_rows:
             return self._run_select(plan, segments, tags, str_cols, kg=kg)
+        import os as _os
+        import time as _time
+        _qt = _os.environ.get("DF_QTIME")
+        _t0 = _time.perf_counter()
         groups = execute(plan, segments, self.device, kg=kg)
+        if _qt:
+            import sys as _sys
+            print(f"[qtime] execute {1e3*(_time.perf_counter()-_t0):.2f}ms"
+                  f" groups={len(groups)}", file=_sys.stderr)
         q_metas = [m for m in plan.agg_meta
                    if m["op"] in ("percentile", "apdex")]
         q_lookup = None
@@ -557,6 +574,8 @@ class QueryEngine:
                         {id(m): mi for mi, m in enumerate(q_metas)})
         columns = plan.key_names + plan.agg_names
         rows: List[List] = []
+        if _qt:
+            _t0 = _time.perf_counter()
         # hoist hydration dispatch out of the per-group loop: closures
         # bound to the host maps, no per-cell string matching
         hyds = [self._hydrator(meta["hydrate"]) for meta in plan.key_meta]
@@ -588,9 +607,18 @@ class QueryEngine:
                     row.append(g["agg"][ai])
                     ai += 1
             rows.append(row)
+        if _qt:
+            import sys as _sys
+            print(f"[qtime] assemble {1e3*(_time.perf_counter()-_t0):.2f}"
+                  f"ms rows={len(rows)}", file=_sys.stderr)
+            _t0 = _time.perf_counter()
         if plan.slimit:
             rows = self._apply_slimit(plan, rows)
         rows = self._order_limit(plan, columns, rows)
+        if _qt:
+            import sys as _sys
+            print(f"[qtime] order {1e3*(_time.perf_counter()-_t0):.2f}ms",
+                  file=_sys.stderr)
         return {"columns": columns, "values": rows}
 
     @staticmethod

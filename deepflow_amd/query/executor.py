@@ -220,18 +220,13 @@ def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
     spec = plan.to_bytes()
     key_sig = tuple((k.family, k.idx, k.bucket) for k in plan.keys)
     known_card = _CARDINALITY_CACHE.get(key_sig, 0)
-    scratch = None
     for seg in segments:
         if seg.n_rows == 0:
             continue
         if known_card >= _QPART_MIN_GROUPS and \
                 seg.n_rows >= _QPART_MIN_ROWS and plan.keys:
             w = len(plan.keys) + len(plan.aggs)
-            if scratch is None or scratch[2].numel() < seg.n_rows * w:
-                scratch = (torch.empty(256, dtype=torch.int32, device=dev),
-                           torch.empty(256, dtype=torch.int32, device=dev),
-                           torch.empty(seg.n_rows * w, dtype=torch.int64,
-                                       device=dev))
+            scratch = _qpart_scratch(dev, seg.n_rows * w)
             scratch[0].zero_()
             gpu_ops.qpart_agg(seg, spec, 0, seg.n_rows, scratch[0],
                               scratch[1], scratch[2], gkeys, graw, gvals,
