@@ -206,6 +206,21 @@ def execute_agg_cpu(plan: Plan, segments, kg=None) -> List[Dict]:
 _CARDINALITY_CACHE: Dict[tuple, int] = {}
 _QPART_MIN_ROWS = 1 << 22
 _QPART_MIN_GROUPS = 1024
+# partition scratch persists across queries: a fresh ~134 MB hipMalloc
+# costs tens of ms and showed up as 20-80 ms query-latency spikes
+_QPART_SCRATCH: Dict[str, tuple] = {}
+
+
+def _qpart_scratch(dev, need_elems: int):
+    key = str(dev)
+    sc = _QPART_SCRATCH.get(key)
+    if sc is None or sc[2].numel() < need_elems:
+        sc = (torch.empty(256, dtype=torch.int32, device=dev),
+              torch.empty(256, dtype=torch.int32, device=dev),
+              torch.empty(max(need_elems, 1 << 20), dtype=torch.int64,
+                          device=dev))
+        _QPART_SCRATCH[key] = sc
+    return sc
 
 
 def execute_agg_gpu(plan: Plan, segments, device="cuda", kg=None) -> List[Dict]:
