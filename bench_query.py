@@ -92,6 +92,12 @@ def main() -> None:
                       "device": device}))
 
     eng = QueryEngine(pipe, device=device)
+    # the loaded store + dictionaries are long-lived: freeze them out of
+    # the cyclic GC's gen2 scans (a gen2 pass over ~1M host objects
+    # showed up as 60+ ms query-latency spikes)
+    import gc
+    gc.collect()
+    gc.freeze()
     for name, sql in QUERIES:
         eng.query(sql)  # warmup (plan + allocs)
         if device == "cuda":

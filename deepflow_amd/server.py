@@ -515,6 +515,12 @@ class DeepflowServer:
             self.system_rows.append(row)
 
     def start(self) -> None:
+        import gc
+        # long-lived server state (dictionaries, segments, name maps)
+        # stays out of gen2 GC scans — full collections over it showed
+        # up as 60+ ms query-latency spikes under load
+        gc.collect()
+        gc.freeze()
         self.receiver.start()
         if self._use_pump:
             from .ingest.native_pump import PumpServer
