@@ -288,4 +288,23 @@ void df_pump_free(void* h) {
     delete p;
 }
 
+// Benchmark sender: push `blob` (pre-framed wire bytes) `reps` times
+// through a connected socket. Lives here so e2e throughput tests are
+// not capped by a Python send loop (ctypes releases the GIL for the
+// whole call).
+int64_t df_tcp_blast(int fd, const uint8_t* blob, uint64_t n,
+                     uint32_t reps) {
+    int64_t total = 0;
+    for (uint32_t r = 0; r < reps; r++) {
+        uint64_t off = 0;
+        while (off < n) {
+            ssize_t w = send(fd, blob + off, n - off, 0);
+            if (w > 0) { off += (uint64_t)w; total += w; continue; }
+            if (errno == EINTR) continue;
+            return -1;
+        }
+    }
+    return total;
+}
+
 }  // extern "C"

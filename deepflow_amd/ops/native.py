@@ -90,6 +90,9 @@ def cpu() -> ct.CDLL:
         lib.df_pump_done.argtypes = [ct.c_void_p]
         lib.df_pump_stats.restype = None
         lib.df_pump_stats.argtypes = [ct.c_void_p] + [ct.c_void_p] * 4
+        lib.df_tcp_blast.restype = ct.c_int64
+        lib.df_tcp_blast.argtypes = [ct.c_int, ct.c_void_p, ct.c_uint64,
+                                     ct.c_uint32]
         lib.df_pump_free.restype = None
         lib.df_pump_free.argtypes = [ct.c_void_p]
         lib.df_zstd_compress.restype = ct.c_int64
