@@ -502,18 +502,9 @@ class QueryEngine:
                     row.append(agg[ai])
                     ai += 1
             rows.append(row)
-        if _qt:
-            import sys as _sys
-            print(f"[qtime] assemble {1e3*(_time.perf_counter()-_t0):.2f}"
-                  f"ms rows={len(rows)}", file=_sys.stderr)
-            _t0 = _time.perf_counter()
         if plan.slimit:
             rows = self._apply_slimit(plan, rows)
         rows = self._order_limit(plan, columns, rows)
-        if _qt:
-            import sys as _sys
-            print(f"[qtime] order {1e3*(_time.perf_counter()-_t0):.2f}ms",
-                  file=_sys.stderr)
         return {"columns": columns, "values": rows}
 
     def _finish_qdata(self, meta, q):
