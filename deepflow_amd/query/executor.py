@@ -296,7 +296,18 @@ def execute_select_gpu(plan: Plan, segments, limit: int,
         gpu_ops.query_select(seg, spec, 0, seg.n_rows, out_rows, out_ctr,
                              kg=kg)
         torch.cuda.synchronize()
-        cnt = min(int(out_ctr.item()), cap)
+        cnt = int(out_ctr.item())
+        if cnt > cap:
+            # the atomic-emit kernel counted every match but only the
+            # first `cap` landed (arrival order is nondeterministic);
+            # rerun with an exact-size buffer so LIMIT/ORDER BY sees
+            # the deterministic full match set
+            out_rows = torch.zeros(cnt, dtype=torch.int64, device=dev)
+            out_ctr.zero_()
+            gpu_ops.query_select(seg, spec, 0, seg.n_rows, out_rows,
+                                 out_ctr, kg=kg)
+            torch.cuda.synchronize()
+            cnt = min(int(out_ctr.item()), cnt)
         rows = sorted(out_rows[:cnt].cpu().tolist())
         for r in rows:
             out.append((si, int(r)))

@@ -187,3 +187,13 @@ def test_partitioned_groupby_matches_direct(engines, monkeypatch):
     assert first["values"] == cpu["values"]
     assert second["values"] == cpu["values"]
     assert len(cpu["values"]) > 30  # multi-group (fixture-sized)
+
+
+def test_select_overflow_retry_deterministic(engines):
+    """A filter matching far more rows than the emit buffer must rerun
+    with an exact-size buffer: LIMIT then sees the deterministic
+    earliest rows, identical to the CPU scan order."""
+    sql = ("SELECT request_resource FROM l7_flow_log LIMIT 7")
+    rc = engines["cpu"].query(sql)
+    rg = engines["cuda"].query(sql)
+    assert rc["values"] == rg["values"]
