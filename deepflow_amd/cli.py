@@ -151,6 +151,72 @@ def agent_run(ingest_server, iface, vtap_id, ring, ebpf, flush_interval):
         a.close()
 
 
+@agent.command("config")
+@click.argument("group", default="default")
+@click.option("--set", "kv", multiple=True,
+              help="key=value overrides to push")
+@click.pass_context
+def agent_config(ctx, group, kv):
+    """Show or update an agent group's config (reference:
+    deepflow-ctl agent-group-config)."""
+    base = f"{ctx.obj['server']}/v1/agent-group-config/{group}"
+    if kv:
+        body = {}
+        for item in kv:
+            k, _, v = item.partition("=")
+            body[k] = yaml_value(v)
+        r = requests.post(base, json=body, timeout=30)
+    else:
+        r = requests.get(base, timeout=30)
+    click.echo(json.dumps(r.json(), indent=2))
+
+
+def yaml_value(v: str):
+    for cast in (int, float):
+        try:
+            return cast(v)
+        except ValueError:
+            pass
+    return {"true": True, "false": False}.get(v.lower(), v)
+
+
+@cli.group()
+def domain():
+    """Cloud-platform domains (reference: deepflow-ctl domain)."""
+
+
+@domain.command("list")
+@click.pass_context
+def domain_list(ctx):
+    r = requests.get(f"{ctx.obj['server']}/v1/domains/", timeout=30)
+    rows = r.json()
+    if not rows:
+        click.echo("(no domains)")
+        return
+    _print_table({"columns": list(rows[0].keys()),
+                  "values": [list(d.values()) for d in rows]})
+
+
+@domain.command("add")
+@click.argument("name")
+@click.option("--type", "dtype", default="filereader")
+@click.option("--config", default="{}")
+@click.pass_context
+def domain_add(ctx, name, dtype, config):
+    r = requests.post(f"{ctx.obj['server']}/v1/domains/",
+                      json={"name": name, "type": dtype,
+                            "config": json.loads(config)}, timeout=30)
+    click.echo(json.dumps(r.json()))
+
+
+@cli.command()
+@click.pass_context
+def genesis(ctx):
+    """Dump the genesis (auto-discovered platform) inventory."""
+    r = requests.get(f"{ctx.obj['server']}/v1/genesis/", timeout=30)
+    click.echo(json.dumps(r.json(), indent=2))
+
+
 @cli.command()
 @click.pass_context
 def stats(ctx):

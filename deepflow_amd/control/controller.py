@@ -291,6 +291,13 @@ class ControllerLite:
             self.set_group_config(group, await request.json())
             return {"status": "ok", "config_version": self.config_version}
 
+        @app.get("/v1/agent-group-config/{group}")
+        def get_config(group: str):
+            return {"group": group,
+                    "config": self.group_configs.get(
+                        group, dict(DEFAULT_AGENT_CONFIG)),
+                    "config_version": self.config_version}
+
         @app.post("/v1/prometheus/label-ids/")
         async def prom_label_ids(request: Request):
             body = await request.json()
@@ -315,6 +322,11 @@ class ControllerLite:
             return self.genesis_report(int(body.get("agent_id", 0)),
                                        body.get("processes", []),
                                        body.get("sockets", []))
+
+        @app.get("/v1/genesis/")
+        def genesis_all():
+            return {str(aid): g for aid, g in
+                    self.genesis_inventory.items()}
 
         @app.get("/v1/genesis/{agent_id}")
         def genesis_get(agent_id: int):

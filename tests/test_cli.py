@@ -149,3 +149,20 @@ def test_agent_run_live(tmp_path):
         assert found, out
     finally:
         srv.stop()
+
+
+def test_cli_domain_config_genesis(live_server):
+    runner = CliRunner()
+    r = runner.invoke(cli, ["--server", live_server, "domain", "add",
+                            "mycloud", "--type", "filereader"])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(cli, ["--server", live_server, "domain", "list"])
+    assert "mycloud" in r.output
+    r = runner.invoke(cli, ["--server", live_server, "agent", "config",
+                            "default", "--set", "max_cpus=4"])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(cli, ["--server", live_server, "agent", "config",
+                            "default"])
+    assert '"max_cpus": 4' in r.output
+    r = runner.invoke(cli, ["--server", live_server, "genesis"])
+    assert r.exit_code == 0, r.output
