@@ -145,6 +145,103 @@ def attach_tracepoint(prog_fd: int, category: str, name: str) -> int:
     return fd
 
 
+def elf_sym_file_offset(path: str, name: str) -> int:
+    """Resolve `name` in an ELF shared object to its FILE offset (what
+    the uprobe PMU wants): st_value mapped through the containing
+    PT_LOAD segment. Reads .dynsym then .symtab. Raises KeyError."""
+    with open(path, "rb") as f:
+        data = f.read()
+    if data[:4] != b"\x7fELF" or data[4] != 2:
+        raise ValueError(f"{path}: not a 64-bit ELF")
+    e_phoff, = struct.unpack_from("<Q", data, 0x20)
+    e_shoff, = struct.unpack_from("<Q", data, 0x28)
+    e_phentsize, e_phnum = struct.unpack_from("<HH", data, 0x36)
+    e_shentsize, e_shnum = struct.unpack_from("<HH", data, 0x3A)
+    loads = []
+    for i in range(e_phnum):
+        off = e_phoff + i * e_phentsize
+        p_type, = struct.unpack_from("<I", data, off)
+        if p_type == 1:  # PT_LOAD
+            p_offset, p_vaddr = struct.unpack_from("<QQ", data, off + 8)
+            p_filesz, = struct.unpack_from("<Q", data, off + 0x20)
+            loads.append((p_vaddr, p_filesz, p_offset))
+    shdrs = []
+    for i in range(e_shnum):
+        off = e_shoff + i * e_shentsize
+        sh_type, = struct.unpack_from("<I", data, off + 4)
+        sh_offset, sh_size = struct.unpack_from("<QQ", data, off + 0x18)
+        sh_link, = struct.unpack_from("<I", data, off + 0x28)
+        sh_entsize, = struct.unpack_from("<Q", data, off + 0x38)
+        shdrs.append((sh_type, sh_offset, sh_size, sh_link, sh_entsize))
+    for sh_type in (11, 2):  # SHT_DYNSYM, SHT_SYMTAB
+        for st, soff, ssize, slink, sent in shdrs:
+            if st != sh_type or not sent:
+                continue
+            stroff = shdrs[slink][1]
+            for j in range(ssize // sent):
+                base = soff + j * sent
+                st_name, = struct.unpack_from("<I", data, base)
+                st_value, = struct.unpack_from("<Q", data, base + 8)
+                if not st_name or not st_value:
+                    continue
+                end = data.index(b"\x00", stroff + st_name)
+                if data[stroff + st_name:end].decode("latin1") == name:
+                    for v, sz, o in loads:
+                        if v <= st_value < v + sz:
+                            return st_value - v + o
+                    return st_value
+    raise KeyError(f"{name} not found in {path}")
+
+
+def _uprobe_pmu_type() -> int:
+    with open("/sys/bus/event_source/devices/uprobe/type") as f:
+        return int(f.read().strip())
+
+
+def _uprobe_retprobe_bit() -> int:
+    try:
+        with open("/sys/bus/event_source/devices/uprobe/format/retprobe"
+                  ) as f:
+            # "config:N"
+            return int(f.read().strip().split(":")[1])
+    except OSError:
+        return 0
+
+
+def attach_uprobe(prog_fd: int, path: str, offset: int,
+                  retprobe: bool = False) -> int:
+    """Attach a KPROBE-type program to a userspace probe point
+    (file + offset) via the uprobe PMU (no tracefs writes needed)."""
+    cfg = (1 << _uprobe_retprobe_bit()) if retprobe else 0
+    pathb = path.encode() + b"\x00"
+    pbuf = ct.create_string_buffer(pathb, len(pathb))
+    # perf_event_attr: type, size, config, sample_period, sample_type,
+    # read_format, flags, ..., config1 (uprobe_path), config2 (offset)
+    attr = bytearray(112)
+    struct.pack_into("<IIQ", attr, 0, _uprobe_pmu_type(), 112, cfg)
+    # perf_event_attr layout: type@0 size@4 config@8 sample_period@16
+    # sample_type@24 read_format@32 flags@40 wakeup@48 bp_type@52
+    # config1@56 (uprobe path ptr) config2@64 (file offset)
+    struct.pack_into("<Q", attr, 56, ct.addressof(pbuf))
+    struct.pack_into("<Q", attr, 64, offset)
+    fd = perf_event_open(bytes(attr), -1, 0)
+    import fcntl
+    fcntl.ioctl(fd, PERF_EVENT_IOC_SET_BPF, prog_fd)
+    fcntl.ioctl(fd, PERF_EVENT_IOC_ENABLE, 0)
+    return fd
+
+
+def find_libssl() -> Optional[str]:
+    import glob as _glob
+    for pat in ("/usr/lib/x86_64-linux-gnu/libssl.so*",
+                "/lib/x86_64-linux-gnu/libssl.so*",
+                "/usr/lib64/libssl.so*"):
+        hits = sorted(_glob.glob(pat))
+        if hits:
+            return hits[0]
+    return None
+
+
 def available() -> bool:
     """Can this process load + attach BPF programs here?"""
     if _tracefs() is None:
@@ -164,8 +261,10 @@ class SocketTracer:
     (PERF_EVENT_IOC_SET_BPF on the tracepoint event delivers
     bpf_perf_event_output records into the same buffers)."""
 
-    def __init__(self):
-        from .progs import MAPS, build_sys_enter, build_sys_exit
+    def __init__(self, with_tls: bool = True):
+        from .progs import (MAPS, SSL_MAPS, build_sys_enter,
+                            build_sys_exit, build_ssl_write,
+                            build_ssl_read_enter, build_ssl_read_exit)
         self.map_fds: Dict[str, int] = {}
         for name, spec in MAPS.items():
             self.map_fds[name] = map_create(*spec)
@@ -175,6 +274,17 @@ class SocketTracer:
         self.exit_fd = prog_load(
             BPF_PROG_TYPE_TRACEPOINT,
             build_sys_exit().to_bytes(self.map_fds), log=True)
+        self.ssl_fds: List[int] = []
+        self.libssl = find_libssl() if with_tls else None
+        if self.libssl is not None:
+            for name, spec in SSL_MAPS.items():
+                self.map_fds[name] = map_create(*spec)
+            BPF_PROG_TYPE_KPROBE = 2
+            for builder in (build_ssl_write, build_ssl_read_enter,
+                            build_ssl_read_exit):
+                self.ssl_fds.append(prog_load(
+                    BPF_PROG_TYPE_KPROBE,
+                    builder().to_bytes(self.map_fds), log=True))
         self.tp_fds: List[int] = []
         self.rings: List[tuple] = []
 
@@ -183,6 +293,21 @@ class SocketTracer:
                                              "sys_enter"))
         self.tp_fds.append(attach_tracepoint(self.exit_fd, "raw_syscalls",
                                              "sys_exit"))
+        if self.libssl is not None and self.ssl_fds:
+            # TLS plaintext capture: uprobes on OpenSSL entry points
+            # (reference kernel/openssl.bpf.c)
+            try:
+                w_off = elf_sym_file_offset(self.libssl, "SSL_write")
+                r_off = elf_sym_file_offset(self.libssl, "SSL_read")
+                self.tp_fds.append(attach_uprobe(self.ssl_fds[0],
+                                                 self.libssl, w_off))
+                self.tp_fds.append(attach_uprobe(self.ssl_fds[1],
+                                                 self.libssl, r_off))
+                self.tp_fds.append(attach_uprobe(self.ssl_fds[2],
+                                                 self.libssl, r_off,
+                                                 retprobe=True))
+            except (OSError, KeyError):
+                pass  # no uprobe PMU / stripped lib: syscalls still on
         self._open_rings()
 
     def _open_rings(self, pages: int = 64) -> None:

@@ -370,6 +370,146 @@ def build_sys_exit() -> Asm:
     return a
 
 
+# ---------------------------------------------------------------- TLS
+# OpenSSL uprobes (reference: kernel/openssl.bpf.c): plaintext capture
+# for TLS connections at SSL_write / SSL_read, where the syscall tracer
+# only sees ciphertext. Zero struct offsets are read from the SSL
+# object (version-proof); the socket 4-tuple is paired in userspace
+# with the thread's most recent socket syscall (runtime.record_events).
+# x86_64 pt_regs offsets (ptrace.h order):
+PT_RAX, PT_RDX, PT_RSI, PT_RDI = 80, 96, 104, 112
+SC_SSL_READ, SC_SSL_WRITE = 0xFFF0, 0xFFF1
+TLS_FD = 0xFFFFFFFF
+
+SSL_MAPS = {
+    # pid_tgid -> user buf ptr stashed at SSL_read entry
+    "ssl_args": (BPF_MAP_TYPE_HASH, 8, 8, 65536),
+}
+
+
+def _ssl_emit(a: Asm, syscall_marker: int, direction: int) -> None:
+    """Shared tail: expects fp-32 = user buf, fp-40 = data len.
+    Builds the event in the per-cpu scratch buffer and emits it."""
+    a.st_imm(BPF_W, R10, -44, 0)
+    a.ld_map_fd(R1, "scratch")
+    a.mov64(R2, R10)
+    a.alu64_imm(BPF_ADD, R2, -44)
+    a.call(H_MAP_LOOKUP)
+    a.jmp_imm(BPF_JNE, R0, 0, "have_buf")
+    a.mov64_imm(R0, 0)
+    a.exit()
+    a.label("have_buf")
+    a.mov64(R9, R0)
+    a.call(H_KTIME_GET_NS)
+    a.stx(BPF_DW, R9, EV_TS, R0)
+    a.call(H_GET_PID_TGID)
+    a.mov64(R1, R0)
+    a.alu64_imm(BPF_RSH, R1, 32)
+    a.stx(BPF_W, R9, EV_TGID, R1)
+    a.mov64(R2, R0)
+    a.alu64_imm(BPF_LSH, R2, 32)
+    a.alu64_imm(BPF_RSH, R2, 32)
+    a.stx(BPF_W, R9, EV_PID, R2)
+    a.ld_imm64(R2, TLS_FD)
+    a.stx(BPF_W, R9, EV_FD, R2)
+    a.st_imm(BPF_DW, R9, EV_SOCKKEY, 0)
+    a.st_imm(BPF_DW, R9, EV_TRACE, 0)
+    a.ldx(BPF_DW, R2, R10, -40)
+    a.stx(BPF_W, R9, EV_LEN, R2)
+    a.st_imm(BPF_H, R9, EV_SYSCALL, syscall_marker)
+    a.st_imm(BPF_B, R9, EV_DIR, direction)
+    # cap = min(len, CAP_LEN); copy plaintext
+    a.ldx(BPF_DW, R8, R10, -40)
+    a.jmp_imm(BPF_JLE, R8, CAP_LEN, "cap_ok")
+    a.mov64_imm(R8, CAP_LEN)
+    a.label("cap_ok")
+    a.stx(BPF_W, R9, EV_CAP, R8)
+    a.mov64(R1, R9)
+    a.alu64_imm(BPF_ADD, R1, EV_PAYLOAD)
+    a.mov64(R2, R8)
+    a.ldx(BPF_DW, R3, R10, -32)
+    a.call(H_PROBE_READ_USER)
+    # protocol inference over the captured plaintext
+    a.ldx(BPF_DW, R7, R10, -40)
+    _emit_inference(a)
+    a.stx(BPF_B, R9, EV_PROTO, R0)
+    a.mov64(R1, R6)
+    a.ld_map_fd(R2, "events")
+    a.ld_imm64(R3, 0xFFFFFFFF)            # BPF_F_CURRENT_CPU
+    a.mov64(R4, R9)
+    a.ldx(BPF_W, R5, R9, EV_CAP)
+    a.alu64_imm(BPF_ADD, R5, EV_HDR)
+    a.call(H_PERF_EVENT_OUTPUT)
+    a.mov64_imm(R0, 0)
+    a.exit()
+
+
+def build_ssl_write() -> Asm:
+    """uprobe SSL_write(ssl, buf, num): plaintext is in hand at entry."""
+    a = Asm()
+    a.mov64(R6, R1)                       # r6 = pt_regs ctx
+    a.ldx(BPF_DW, R2, R6, PT_RSI)         # buf
+    a.stx(BPF_DW, R10, -32, R2)
+    a.ldx(BPF_DW, R2, R6, PT_RDX)         # num
+    a.jmp_imm(BPF_JSGE, R2, 1, "len_ok")
+    a.mov64_imm(R0, 0)
+    a.exit()
+    a.label("len_ok")
+    a.stx(BPF_DW, R10, -40, R2)
+    _ssl_emit(a, SC_SSL_WRITE, 0)
+    return a
+
+
+def build_ssl_read_enter() -> Asm:
+    """uprobe SSL_read entry: stash the destination buffer per thread."""
+    a = Asm()
+    a.mov64(R6, R1)
+    a.call(H_GET_PID_TGID)
+    a.stx(BPF_DW, R10, -8, R0)
+    a.ldx(BPF_DW, R2, R6, PT_RSI)         # buf
+    a.stx(BPF_DW, R10, -16, R2)
+    a.ld_map_fd(R1, "ssl_args")
+    a.mov64(R2, R10)
+    a.alu64_imm(BPF_ADD, R2, -8)
+    a.mov64(R3, R10)
+    a.alu64_imm(BPF_ADD, R3, -16)
+    a.mov64_imm(R4, 0)
+    a.call(H_MAP_UPDATE)
+    a.mov64_imm(R0, 0)
+    a.exit()
+    return a
+
+
+def build_ssl_read_exit() -> Asm:
+    """uretprobe SSL_read: ret = plaintext length; emit the event."""
+    a = Asm()
+    a.mov64(R6, R1)
+    a.call(H_GET_PID_TGID)
+    a.stx(BPF_DW, R10, -8, R0)
+    a.ld_map_fd(R1, "ssl_args")
+    a.mov64(R2, R10)
+    a.alu64_imm(BPF_ADD, R2, -8)
+    a.call(H_MAP_LOOKUP)
+    a.jmp_imm(BPF_JNE, R0, 0, "have_args")
+    a.mov64_imm(R0, 0)
+    a.exit()
+    a.label("have_args")
+    a.ldx(BPF_DW, R2, R0, 0)              # stashed buf
+    a.stx(BPF_DW, R10, -32, R2)
+    a.ld_map_fd(R1, "ssl_args")
+    a.mov64(R2, R10)
+    a.alu64_imm(BPF_ADD, R2, -8)
+    a.call(H_MAP_DELETE)
+    a.ldx(BPF_DW, R2, R6, PT_RAX)         # return value
+    a.jmp_imm(BPF_JSGE, R2, 1, "ret_ok")
+    a.mov64_imm(R0, 0)
+    a.exit()
+    a.label("ret_ok")
+    a.stx(BPF_DW, R10, -40, R2)
+    _ssl_emit(a, SC_SSL_READ, 1)
+    return a
+
+
 def build_profiler() -> Asm:
     """On-CPU sampling profiler: perf_event program — count per
     (tgid, user stack id, kernel stack id). Reference:

@@ -90,6 +90,9 @@ _REC_FMT = "<QIBBBxIIHHQI"
 assert struct.calcsize(_REC_FMT) == 40
 
 
+TLS_FD = 0xFFFFFFFF
+
+
 class EbpfCollector:
     def __init__(self, agent, resolver=None):
         self.agent = agent
@@ -97,6 +100,11 @@ class EbpfCollector:
         self.events_in = 0
         self.unresolved = 0
         self._batch: List[bytes] = []
+        # TLS pairing: SSL_read/SSL_write uprobe events carry no fd (no
+        # portable way to read it from the SSL object without struct
+        # offsets) — pair them with the thread's most recent socket
+        # syscall, the reference's fallback when offsets are unknown
+        self._tid_last_fd: Dict[Tuple[int, int], int] = {}
 
     def on_event(self, ev: bytes) -> None:
         """One SkEvent (perf record payload) from any source."""
@@ -105,7 +113,13 @@ class EbpfCollector:
         (ts, tgid, pid, fd, ln, cap, direction, proto_hint, _sc, trace,
          _skey) = struct.unpack(SK_EVENT_FMT, ev[:EV_HDR])
         payload = ev[EV_HDR:EV_HDR + cap]
-        tup = self.resolver.resolve(tgid, fd)
+        if fd == TLS_FD:
+            fd = self._tid_last_fd.get((tgid, pid), -1)
+        else:
+            self._tid_last_fd[(tgid, pid)] = fd
+            if len(self._tid_last_fd) > (1 << 16):
+                self._tid_last_fd.clear()
+        tup = self.resolver.resolve(tgid, fd) if fd >= 0 else None
         self.events_in += 1
         if tup is None:
             self.unresolved += 1

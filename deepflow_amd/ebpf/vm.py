@@ -319,3 +319,54 @@ class SyscallSim:
 
     def events(self) -> List[bytes]:
         return self.vm.events
+
+
+def pt_regs_ctx(rdi: int = 0, rsi: int = 0, rdx: int = 0,
+                rax: int = 0) -> bytes:
+    """x86_64 pt_regs blob for uprobe programs (ptrace.h order)."""
+    regs = [0] * 21
+    regs[10] = rax & M64
+    regs[12] = rdx & M64
+    regs[13] = rsi & M64
+    regs[14] = rdi & M64
+    return struct.pack("<21Q", *regs)
+
+
+class SslSim:
+    """Drive the OpenSSL uprobe programs (TLS plaintext capture) with
+    synthetic SSL_read/SSL_write calls."""
+
+    def __init__(self):
+        from .progs import (MAPS, SSL_MAPS, build_ssl_write,
+                            build_ssl_read_enter, build_ssl_read_exit)
+        allmaps = dict(MAPS)
+        allmaps.update(SSL_MAPS)
+        self.vm = Vm(allmaps)
+        self.w = build_ssl_write()
+        self.r_enter = build_ssl_read_enter()
+        self.r_exit = build_ssl_read_exit()
+        self._next_user = USER_BASE
+
+    def _ubuf(self, payload: bytes) -> int:
+        ubuf = self._next_user
+        self._next_user += (len(payload) + 4095) & ~4095 or 4096
+        self.vm.user_mem[ubuf] = payload
+        return ubuf
+
+    def ssl_write(self, tgid: int, pid: int, payload: bytes) -> None:
+        self.vm.pid_tgid = ((tgid << 32) | pid) & M64
+        ubuf = self._ubuf(payload)
+        self.vm.run(self.w, pt_regs_ctx(rsi=ubuf, rdx=len(payload)))
+        self.vm.clock += 1000
+
+    def ssl_read(self, tgid: int, pid: int, payload: bytes,
+                 ret: Optional[int] = None) -> None:
+        self.vm.pid_tgid = ((tgid << 32) | pid) & M64
+        ubuf = self._ubuf(payload)
+        self.vm.run(self.r_enter, pt_regs_ctx(rsi=ubuf))
+        self.vm.run(self.r_exit, pt_regs_ctx(
+            rsi=ubuf, rax=len(payload) if ret is None else ret))
+        self.vm.clock += 1000
+
+    def events(self) -> List[bytes]:
+        return self.vm.events
