@@ -282,3 +282,49 @@ class PromPipeline:
                     len(self.label_values.name(lv))
             total += row * int(cnt)
         return total
+
+    def state_dict(self) -> dict:
+        self.samples.flush()
+        n = self.samples.n
+        return {
+            "interners": {
+                "metric": list(self.metric_names.strings())
+                if not isinstance(self.metric_names, GlobalInterner)
+                else None,
+                "to_id": {k: dict(v.to_id) for k, v in
+                          (("metric", self.metric_names),
+                           ("label_name", self.label_names),
+                           ("label_value", self.label_values))},
+            },
+            "series": {k: v for k, v in self.series.items()},
+            "series_labels": list(self.series_labels),
+            "series_metric": list(self.series_metric),
+            "samples": {
+                "series": self.samples.series[:n].cpu().clone(),
+                "ts": self.samples.ts[:n].cpu().clone(),
+                "value": self.samples.value[:n].cpu().clone(),
+            },
+        }
+
+    def load_state_dict(self, st: dict) -> None:
+        for kind, itn in (("metric", self.metric_names),
+                          ("label_name", self.label_names),
+                          ("label_value", self.label_values)):
+            for name, ident in st["interners"]["to_id"][kind].items():
+                itn.to_id[name] = ident
+                if isinstance(itn, GlobalInterner):
+                    itn.from_id_map[ident] = name
+                else:
+                    while len(itn.from_id) <= ident:
+                        itn.from_id.append("")
+                    itn.from_id[ident] = name
+        self.series = dict(st["series"])
+        self.series_labels = list(st["series_labels"])
+        self.series_metric = list(st["series_metric"])
+        n = st["samples"]["series"].numel()
+        self.samples._ensure(n)
+        dev = self.samples.device
+        self.samples.series[:n] = st["samples"]["series"].to(dev)
+        self.samples.ts[:n] = st["samples"]["ts"].to(dev)
+        self.samples.value[:n] = st["samples"]["value"].to(dev)
+        self.samples.n = n

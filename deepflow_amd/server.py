@@ -336,10 +336,14 @@ class DeepflowServer:
         os.makedirs(ckpt_dir, exist_ok=True)
         with self._lock:
             ck.save_l7(self.l7, os.path.join(ckpt_dir, "l7.ckpt"))
+            ck.save_l4(self.l4, os.path.join(ckpt_dir, "l4.ckpt"))
             import torch as _t
             tmp = os.path.join(ckpt_dir, "controller.ckpt.tmp")
             _t.save(self.controller.state_dict(), tmp)
             os.replace(tmp, os.path.join(ckpt_dir, "controller.ckpt"))
+            tmp = os.path.join(ckpt_dir, "prom.ckpt.tmp")
+            _t.save(self.prom.state_dict(), tmp)
+            os.replace(tmp, os.path.join(ckpt_dir, "prom.ckpt"))
         return {"status": "ok", "rows": self.l7.segments.n_rows,
                 "agents": len(self.controller.agents)}
 
@@ -351,6 +355,16 @@ class DeepflowServer:
         if os.path.exists(p7):
             with self._lock:
                 out["l7_rows"] = ck.load_l7(self.l7, p7)
+        p4 = os.path.join(ckpt_dir, "l4.ckpt")
+        if os.path.exists(p4):
+            with self._lock:
+                out["l4_rows"] = ck.load_l4(self.l4, p4)
+        pp = os.path.join(ckpt_dir, "prom.ckpt")
+        if os.path.exists(pp):
+            import torch as _t
+            self.prom.load_state_dict(
+                _t.load(pp, weights_only=False))
+            out["prom"] = True
         pc = os.path.join(ckpt_dir, "controller.ckpt")
         if os.path.exists(pc):
             import torch as _t

@@ -247,3 +247,65 @@ def load_l7(pipeline, path: str) -> int:
     elif payload.get("metrics"):  # pre-table-family checkpoints
         pipeline.metrics.load_state_dict(payload["metrics"])
     return total
+
+
+# ---------------------------------------------------------------- L4
+def save_l4(pipeline, path: str) -> None:
+    """Persist the L4 (flow) hot store. L4 segments carry no dictionary
+    ids or attr pools; the shared TagDictionary/KG are saved by save_l7
+    (the reference's durable-state analog is ClickHouse flow_log.l4)."""
+    from . import l4_schema as L4
+    segs = pipeline.segments
+
+    def seg_state(seg):
+        return {
+            "n_rows": seg.n_rows,
+            "capacity": seg.capacity,
+            "u64": seg.u64.cpu().clone(),
+            "u32": seg.u32.cpu().clone(),
+            "u8": seg.u8.cpu().clone(),
+            "str_rowref": seg.str_rowref.cpu().clone(),
+            "str_lens": seg.str_lens.cpu().clone(),
+            "pool": seg.pool[:seg.pool_len].cpu().clone(),
+            "pool_len": seg.pool_len,
+        }
+    payload = {
+        "layout_version": getattr(L4, "LAYOUT_VERSION", 1),
+        "time_base_s": pipeline.time_base_s,
+        "segment_rows": segs.segment_rows,
+        "segments": [seg_state(s) for s in segs.segments],
+        "rollups": pipeline.rollups.state_dict()
+        if hasattr(pipeline, "rollups") else None,
+    }
+    tmp = path + ".tmp"
+    torch.save(payload, tmp)
+    os.replace(tmp, path)
+
+
+def load_l4(pipeline, path: str) -> int:
+    from . import l4_schema as L4
+    state = torch.load(path, weights_only=False)
+    if state["layout_version"] != getattr(L4, "LAYOUT_VERSION", 1):
+        raise ValueError(
+            f"l4 checkpoint layout {state['layout_version']} != "
+            f"{getattr(L4, 'LAYOUT_VERSION', 1)} (no migration defined)")
+    if state["time_base_s"] != pipeline.time_base_s:
+        raise ValueError("time_base_s mismatch on l4 restore")
+    segs = pipeline.segments
+    dev = segs.device
+    segs.segments = []
+    total = 0
+    for st in state["segments"]:
+        seg = segs.cls(st["capacity"], device=dev)
+        for name in ("u64", "u32", "u8", "str_rowref", "str_lens"):
+            getattr(seg, name).copy_(st[name].to(dev))
+        seg.ensure_pool(st["pool_len"])
+        if st["pool_len"]:
+            seg.pool[:st["pool_len"]] = st["pool"].to(dev)
+        seg.pool_len = st["pool_len"]
+        seg.n_rows = st["n_rows"]
+        segs.segments.append(seg)
+        total += seg.n_rows
+    if hasattr(pipeline, "rollups") and state.get("rollups") is not None:
+        pipeline.rollups.load_state_dict(state["rollups"])
+    return total

@@ -193,3 +193,33 @@ def test_scratch_pool_released_after_cold_query():
     # after the query the scratch pool is empty and its segments are free
     assert not getattr(segs, "_scratch", [])
     assert segs._free
+
+
+def test_l4_and_prom_checkpoint_roundtrip(tmp_path):
+    """Server checkpoints now cover the L4 flow store and the prometheus
+    sample columns (reference durable state = ClickHouse tables)."""
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.gen.flows import FlowGenConfig, gen_flow_payload
+    from deepflow_amd.wire import framing
+
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12, time_base_s=0)
+    cfg = FlowGenConfig(n=40, seed=6)
+    srv._on_l4(framing.FrameHeader(msg_type=framing.MSG_TAGGEDFLOW),
+               __import__("numpy").frombuffer(gen_flow_payload(cfg),
+                                              dtype="uint8"))
+    srv.prom.ingest_labeled_samples([
+        ("up", {"job": "api"}, 1000, 1.0),
+        ("up", {"job": "db"}, 2000, 0.0)])
+    d = str(tmp_path / "ck")
+    srv.save_checkpoint(d)
+
+    srv2 = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                          dict_capacity=1 << 12, time_base_s=0)
+    out = srv2.load_checkpoint(d)
+    assert out["l4_rows"] == 40
+    assert out["prom"]
+    r = srv2.engine.query("SELECT COUNT(1) FROM l4_flow_log")
+    assert r["values"][0][0] == 40
+    series = srv2.prom.series_for("up", [("job", "=", "api")])
+    assert series and series[0]["samples"] == {1: 1.0}
