@@ -223,3 +223,27 @@ def test_l4_and_prom_checkpoint_roundtrip(tmp_path):
     assert r["values"][0][0] == 40
     series = srv2.prom.series_for("up", [("job", "=", "api")])
     assert series and series[0]["samples"] == {1: 1.0}
+
+
+@pytest.mark.gpu
+def test_l4_checkpoint_roundtrip_gpu(tmp_path):
+    from deepflow_amd.ingest.l4_pipeline import L4IngestPipeline
+    from deepflow_amd.store import checkpoint as ck
+    from deepflow_amd.gen.flows import FlowGenConfig, gen_flow_payload
+    import numpy as np
+
+    cfg = FlowGenConfig(n=500, seed=9)
+    pipe = L4IngestPipeline(device="cuda", segment_rows=1 << 12,
+                            time_base_s=0)
+    pipe.ingest_frame_payload(gen_flow_payload(cfg))
+    path = str(tmp_path / "l4.ckpt")
+    ck.save_l4(pipe, path)
+    pipe2 = L4IngestPipeline(device="cuda", segment_rows=1 << 12,
+                             time_base_s=0)
+    assert ck.load_l4(pipe2, path) == 500
+    a = pipe.segments.segments[0]
+    b = pipe2.segments.segments[0]
+    import torch
+    assert torch.equal(a.u64[:, :500], b.u64[:, :500])
+    assert torch.equal(a.u32[:, :500], b.u32[:, :500])
+    assert a.pool_len == b.pool_len
