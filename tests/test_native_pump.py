@@ -288,3 +288,30 @@ def test_pump_to_gpu_pipeline():
         assert sum(r["request"] for r in rows) == 5000
     finally:
         srv.stop()
+
+
+def test_server_pump_dispatches_all_types():
+    """accept_type=-1 pump: metrics/prometheus/flow frames all route to
+    their registered handlers with correct header metadata."""
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.gen.flows import FlowGenConfig, gen_flow_payload
+
+    srv = DeepflowServer(device="cpu", tcp_port=0, segment_rows=1 << 10,
+                         dict_capacity=1 << 12, native_pump=True)
+    srv.start()
+    try:
+        l4 = gen_flow_payload(FlowGenConfig(n=10, seed=2))
+        fr1 = framing.encode_frame(
+            framing.FrameHeader(msg_type=framing.MSG_TAGGEDFLOW,
+                                agent_id=42, org_id=1), l4)
+        s = socket.create_connection(("127.0.0.1", srv.pump_port))
+        s.sendall(fr1)
+        t_end = time.time() + 10
+        while srv.l4.stats.flows_in < 10 and time.time() < t_end:
+            time.sleep(0.05)
+        s.close()
+        assert srv.l4.stats.flows_in == 10
+        r = srv.engine.query("SELECT COUNT(1) FROM l4_flow_log")
+        assert r["values"][0][0] == 10
+    finally:
+        srv.stop()
