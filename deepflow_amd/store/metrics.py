@@ -169,6 +169,13 @@ class RollupTable:
         self._rnz = (1 + td.keys.index(td.require_nonzero)
                      if td.require_nonzero else 0)
         self.source_table = None  # set by RollupFamily for derive_from
+        # interval flush (reference: each 1s/1m interval is written out
+        # to ClickHouse and the in-memory aggregator moves on): the live
+        # device/CPU table archives-and-resets periodically so bounded
+        # caps never saturate on long runs. Archived groups merge with
+        # live groups at read time.
+        self.archive: List[Dict] = []
+        self.dropped_total = 0
         if td.derive_from is not None:
             return
         if device == "cpu":
@@ -326,10 +333,8 @@ class RollupTable:
             return v - (1 << 32) if v >= (1 << 31) else v
         return v
 
-    def rows(self) -> List[Dict]:
+    def _live_rows(self) -> List[Dict]:
         names = self.td.out_names or self.td.keys
-        if self.td.derive_from is not None:
-            return self._derived_rows()
         out = []
         for key, acc in self._items():
             row = {"time": self.time_base_s + key[0]}
@@ -337,17 +342,62 @@ class RollupTable:
                 row[name] = self._fmt_key(name, self.td.keys[ki], key[1 + ki])
             row.update({f: a for f, a in zip(self.fields, acc)})
             out.append(row)
+        return out
+
+    def _merge_rows(self, rows: List[Dict]) -> List[Dict]:
+        """Merge duplicate (time, key) groups (a group can span a
+        flush boundary: once in the archive, once live)."""
+        names = self.td.out_names or self.td.keys
+        merged: Dict[tuple, Dict] = {}
+        for r in rows:
+            key = (r["time"],) + tuple(r[n] for n in names)
+            acc = merged.get(key)
+            if acc is None:
+                merged[key] = dict(r)
+            else:
+                for f in self.fields:
+                    if f.endswith("_max"):
+                        acc[f] = max(acc[f], r[f])
+                    else:
+                        acc[f] += r[f]
+        out = list(merged.values())
         out.sort(key=lambda r: (r["time"],) + tuple(
             str(r[n]) for n in names))
         return out
 
-    def _derived_rows(self) -> List[Dict]:
-        """Fold the source table's harvested groups into this table's
+    def flush_live(self) -> int:
+        """Archive the live table's groups and reset it (the interval
+        flush). Derived tables archive their fold of the CURRENT live
+        source first — call on the whole family, not per table."""
+        if self.td.derive_from is not None:
+            rows = self._derived_live()
+            self.archive.extend(rows)
+            return len(rows)
+        rows = self._live_rows()
+        self.archive.extend(rows)
+        if self.device == "cpu":
+            self.table.clear()
+        else:
+            self.dropped_total += int(self.drops.item())
+            self.tkeys.zero_()
+            self.tvals.zero_()
+            self.drops.zero_()
+        return len(rows)
+
+    def rows(self) -> List[Dict]:
+        if self.td.derive_from is not None:
+            return self._merge_rows(self.archive + self._derived_live())
+        return self._merge_rows(self.archive + self._live_rows())
+
+    def _derived_live(self) -> List[Dict]:
+        """Fold the source table's LIVE groups into this table's
         coarser buckets (sums; *_max fields take max)."""
         iv = self.td.interval_s
         names = self.td.out_names or self.td.keys
         folded: Dict[tuple, Dict] = {}
-        for r in self.source_table.rows():
+        src = self.source_table
+        for r in (src._live_rows() if src.td.derive_from is None
+                  else src.rows()):
             t = ((r["time"] - self.time_base_s) // iv) * iv + \
                 self.time_base_s
             key = (t,) + tuple(r[n] for n in names)
@@ -371,20 +421,23 @@ class RollupTable:
         if self.td.derive_from is not None:
             return self.source_table.drop_count()
         if self.device == "cpu":
-            return 0
-        return int(self.drops.item())
+            return self.dropped_total
+        return self.dropped_total + int(self.drops.item())
 
     # --------------------------------------------------------- checkpoint
     def state_dict(self):
         if self.td.derive_from is not None:
-            return {}
+            return {"archive": list(self.archive)}
         if self.device == "cpu":
-            return {"table": {k: list(v) for k, v in self.table.items()}}
+            return {"table": {k: list(v) for k, v in self.table.items()},
+                    "archive": list(self.archive)}
         return {"tkeys": self.tkeys.cpu().clone(),
                 "traw": self.traw.cpu().clone(),
-                "tvals": self.tvals.cpu().clone()}
+                "tvals": self.tvals.cpu().clone(),
+                "archive": list(self.archive)}
 
     def load_state_dict(self, st):
+        self.archive = list(st.get("archive", []))
         if self.td.derive_from is not None:
             return
         if self.device == "cpu":
@@ -411,6 +464,11 @@ class RollupFamily:
             if t.td.derive_from is not None:
                 t.source_table = self.tables[t.td.derive_from]
 
+    # archive-and-reset the live tables after this many rows: the
+    # bounded device tables never saturate on long runs (the reference
+    # flushes every closed interval to ClickHouse)
+    FLUSH_EVERY_ROWS = 32_000_000
+
     def update(self, seg, base: int, n: int, stream: int = 0) -> None:
         tables = [t for t in self.tables.values()
                   if t.td.derive_from is None]
@@ -419,10 +477,24 @@ class RollupFamily:
         if tables[0].device == "cpu":
             for t in tables:
                 t.update(seg, base, n, stream)
-            return
-        from ..ops import gpu_ops
-        gpu_ops.rollup_family(seg, base, n, tables[0].time_base_s, tables,
-                              stream=stream)
+        else:
+            from ..ops import gpu_ops
+            gpu_ops.rollup_family(seg, base, n, tables[0].time_base_s,
+                                  tables, stream=stream)
+        self._rows_since_flush = getattr(self, "_rows_since_flush", 0) + n
+        if self._rows_since_flush >= self.FLUSH_EVERY_ROWS:
+            self.flush()
+
+    def flush(self) -> None:
+        """Interval flush: derived tables archive their fold of the
+        live source first, then the live tables archive-and-reset."""
+        for t in self.tables.values():
+            if t.td.derive_from is not None:
+                t.flush_live()
+        for t in self.tables.values():
+            if t.td.derive_from is None:
+                t.flush_live()
+        self._rows_since_flush = 0
 
     def get(self, name: str) -> Optional[RollupTable]:
         return self.tables.get(name)

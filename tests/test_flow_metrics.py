@@ -157,3 +157,31 @@ def test_doc_insert_merges_exactly():
     merged = [r for r in rows if r["vtap_id"] == 1][0]
     assert merged["request"] == 3 and merged["response"] == 1
     assert merged["rrt_sum"] == 40 and merged["rrt_max"] == 30  # max op
+
+
+def test_rollup_interval_flush_merges_exactly():
+    """Archive-and-reset flush: a group updated before AND after a flush
+    reads back as one exactly-merged row, identical to a never-flushed
+    table."""
+    from deepflow_amd.gen.spans import SpanGenConfig, gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+
+    cfg = SpanGenConfig(n=4000, seed=21, tag_cardinality=500, n_ips=64,
+                        n_services=8)
+    payload = gen_span_payload(cfg)
+    a = L7IngestPipeline(device="cpu", segment_rows=1 << 13,
+                         time_base_s=cfg.base_time_ns // 10**9)
+    b = L7IngestPipeline(device="cpu", segment_rows=1 << 13,
+                         time_base_s=cfg.base_time_ns // 10**9)
+    a.ingest_frame_payload(payload)
+    a.rollups.flush()           # everything moves to the archive
+    a.ingest_frame_payload(payload)
+    a.rollups.flush()
+    a.ingest_frame_payload(payload)
+    for _ in range(3):
+        b.ingest_frame_payload(payload)
+    for name in ("application.1s", "application.1m",
+                 "application_map.1s"):
+        ra = a.rollups.get(name).rows()
+        rb = b.rollups.get(name).rows()
+        assert ra == rb, name
