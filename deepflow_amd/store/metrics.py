@@ -19,6 +19,8 @@ the numerics oracle for the GPU tests.
 from __future__ import annotations
 
 import struct
+
+import numpy as np
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
@@ -149,6 +151,21 @@ M32 = 0xFFFFFFFF
 M64 = (1 << 64) - 1
 
 
+def _np_merge(keys: "np.ndarray", vals: "np.ndarray", ops):
+    """Merge duplicate key rows: sum columns with op 0, max otherwise."""
+    if keys.shape[0] == 0:
+        return keys, vals
+    uniq, inv = np.unique(keys, axis=0, return_inverse=True)
+    inv = inv.ravel()
+    out = np.zeros((uniq.shape[0], vals.shape[1]), dtype=np.uint64)
+    for v in range(vals.shape[1]):
+        if ops[v] == 0:
+            np.add.at(out[:, v], inv, vals[:, v])
+        else:
+            np.maximum.at(out[:, v], inv, vals[:, v])
+    return uniq, out
+
+
 class RollupTable:
     def __init__(self, td: TableDef, time_base_s: int, device: str = "cpu",
                  capacity_pow2: int = 1 << 18):
@@ -172,9 +189,11 @@ class RollupTable:
         # interval flush (reference: each 1s/1m interval is written out
         # to ClickHouse and the in-memory aggregator moves on): the live
         # device/CPU table archives-and-resets periodically so bounded
-        # caps never saturate on long runs. Archived groups merge with
-        # live groups at read time.
-        self.archive: List[Dict] = []
+        # caps never saturate on long runs. Archived groups are COMPACT
+        # numpy (keys, vals) chunks — a _map flush can carry millions of
+        # groups, far beyond what per-row python dicts can shuffle —
+        # and merge with live groups at read time.
+        self.archive: List[tuple] = []  # [(keys u64 [G,nw], vals [G,nv])]
         self.dropped_total = 0
         if td.derive_from is not None:
             return
@@ -305,6 +324,21 @@ class RollupTable:
         gpu_ops.rollup_insert(kws, vals, ops, self, stream=stream)
 
     # ------------------------------------------------------------ harvest
+    def _harvest_np(self):
+        """(keys u64 [G, nw], vals u64 [G, nv]) of the LIVE table with
+        duplicate-slot groups merged — all numpy, no per-row python."""
+        if self.device == "cpu":
+            if not self.table:
+                return (np.zeros((0, self.nw), dtype=np.uint64),
+                        np.zeros((0, self.nv), dtype=np.uint64))
+            keys = np.array(list(self.table.keys()), dtype=np.uint64)
+            vals = np.array(list(self.table.values()), dtype=np.uint64)
+            return keys.reshape(-1, self.nw), vals.reshape(-1, self.nv)
+        mask = self.tkeys != 0
+        raw = self.traw[mask][:, : self.nw].cpu().numpy().view(np.uint64)
+        vals = self.tvals[mask].cpu().numpy().view(np.uint64)
+        return _np_merge(raw, vals, self.ops)
+
     def _items(self) -> List[Tuple[tuple, List[int]]]:
         if self.device == "cpu":
             return [(k, list(v)) for k, v in self.table.items()]
@@ -333,48 +367,22 @@ class RollupTable:
             return v - (1 << 32) if v >= (1 << 31) else v
         return v
 
-    def _live_rows(self) -> List[Dict]:
-        names = self.td.out_names or self.td.keys
-        out = []
-        for key, acc in self._items():
-            row = {"time": self.time_base_s + key[0]}
-            for ki, name in enumerate(names):
-                row[name] = self._fmt_key(name, self.td.keys[ki], key[1 + ki])
-            row.update({f: a for f, a in zip(self.fields, acc)})
-            out.append(row)
-        return out
-
-    def _merge_rows(self, rows: List[Dict]) -> List[Dict]:
-        """Merge duplicate (time, key) groups (a group can span a
-        flush boundary: once in the archive, once live)."""
-        names = self.td.out_names or self.td.keys
-        merged: Dict[tuple, Dict] = {}
-        for r in rows:
-            key = (r["time"],) + tuple(r[n] for n in names)
-            acc = merged.get(key)
-            if acc is None:
-                merged[key] = dict(r)
-            else:
-                for f in self.fields:
-                    if f.endswith("_max"):
-                        acc[f] = max(acc[f], r[f])
-                    else:
-                        acc[f] += r[f]
-        out = list(merged.values())
-        out.sort(key=lambda r: (r["time"],) + tuple(
-            str(r[n]) for n in names))
-        return out
-
-    def flush_live(self) -> int:
-        """Archive the live table's groups and reset it (the interval
-        flush). Derived tables archive their fold of the CURRENT live
-        source first — call on the whole family, not per table."""
+    def flush_live(self, src_harvest=None) -> int:
+        """Archive the live table's groups (compact arrays) and reset
+        it (the interval flush). Derived tables archive their fold of
+        the CURRENT live source — call on the whole family, not per
+        table; the family passes the source harvest in so it is
+        computed once."""
         if self.td.derive_from is not None:
-            rows = self._derived_live()
-            self.archive.extend(rows)
-            return len(rows)
-        rows = self._live_rows()
-        self.archive.extend(rows)
+            keys, vals = src_harvest if src_harvest is not None \
+                else self.source_table._harvest_np()
+            keys, vals = self._fold_np(keys, vals)
+            if keys.shape[0]:
+                self.archive.append((keys, vals))
+            return keys.shape[0]
+        keys, vals = self._harvest_np()
+        if keys.shape[0]:
+            self.archive.append((keys, vals))
         if self.device == "cpu":
             self.table.clear()
         else:
@@ -382,37 +390,38 @@ class RollupTable:
             self.tkeys.zero_()
             self.tvals.zero_()
             self.drops.zero_()
-        return len(rows)
+        return keys.shape[0]
+
+    def _fold_np(self, keys, vals):
+        """Rebucket key word 0 (relative time) to this table's coarser
+        interval and merge."""
+        iv = self.td.interval_s
+        keys = keys.copy()
+        keys[:, 0] = keys[:, 0] // iv * iv
+        return _np_merge(keys, vals, self.ops)
+
+    def _all_groups_np(self):
+        """Merged (keys, vals) across the archive chunks + live."""
+        if self.td.derive_from is not None:
+            live = self._fold_np(*self.source_table._harvest_np())
+        else:
+            live = self._harvest_np()
+        chunks = self.archive + [live]
+        keys = np.concatenate([k for k, _ in chunks], axis=0)
+        vals = np.concatenate([v for _, v in chunks], axis=0)
+        return _np_merge(keys, vals, self.ops)
 
     def rows(self) -> List[Dict]:
-        if self.td.derive_from is not None:
-            return self._merge_rows(self.archive + self._derived_live())
-        return self._merge_rows(self.archive + self._live_rows())
-
-    def _derived_live(self) -> List[Dict]:
-        """Fold the source table's LIVE groups into this table's
-        coarser buckets (sums; *_max fields take max)."""
-        iv = self.td.interval_s
+        keys, vals = self._all_groups_np()
         names = self.td.out_names or self.td.keys
-        folded: Dict[tuple, Dict] = {}
-        src = self.source_table
-        for r in (src._live_rows() if src.td.derive_from is None
-                  else src.rows()):
-            t = ((r["time"] - self.time_base_s) // iv) * iv + \
-                self.time_base_s
-            key = (t,) + tuple(r[n] for n in names)
-            acc = folded.get(key)
-            if acc is None:
-                acc = dict(r)
-                acc["time"] = t
-                folded[key] = acc
-            else:
-                for f in self.fields:
-                    if f.endswith("_max"):
-                        acc[f] = max(acc[f], r[f])
-                    else:
-                        acc[f] += r[f]
-        out = list(folded.values())
+        out = []
+        for g in range(keys.shape[0]):
+            row = {"time": self.time_base_s + int(keys[g, 0])}
+            for ki, name in enumerate(names):
+                row[name] = self._fmt_key(name, self.td.keys[ki],
+                                          int(keys[g, 1 + ki]))
+            row.update({f: int(v) for f, v in zip(self.fields, vals[g])})
+            out.append(row)
         out.sort(key=lambda r: (r["time"],) + tuple(
             str(r[n]) for n in names))
         return out
@@ -486,14 +495,26 @@ class RollupFamily:
             self.flush()
 
     def flush(self) -> None:
-        """Interval flush: derived tables archive their fold of the
-        live source first, then the live tables archive-and-reset."""
+        """Interval flush: harvest each live table once, let derived
+        tables fold that harvest, then archive-and-reset the live."""
+        harvests = {name: t._harvest_np()
+                    for name, t in self.tables.items()
+                    if t.td.derive_from is None}
         for t in self.tables.values():
             if t.td.derive_from is not None:
-                t.flush_live()
-        for t in self.tables.values():
+                t.flush_live(src_harvest=harvests[t.td.derive_from])
+        for name, t in self.tables.items():
             if t.td.derive_from is None:
-                t.flush_live()
+                keys, vals = harvests[name]
+                if keys.shape[0]:
+                    t.archive.append((keys, vals))
+                if t.device == "cpu":
+                    t.table.clear()
+                else:
+                    t.dropped_total += int(t.drops.item())
+                    t.tkeys.zero_()
+                    t.tvals.zero_()
+                    t.drops.zero_()
         self._rows_since_flush = 0
 
     def get(self, name: str) -> Optional[RollupTable]:
