@@ -185,3 +185,24 @@ def test_rollup_interval_flush_merges_exactly():
         ra = a.rollups.get(name).rows()
         rb = b.rollups.get(name).rows()
         assert ra == rb, name
+
+
+def test_flow_metrics_sql_after_flush():
+    """DF-SQL over the flow_metrics family still sees every group after
+    interval flushes (archive + live merge feeds the engine)."""
+    from deepflow_amd.gen.spans import SpanGenConfig, gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query import QueryEngine
+
+    cfg = SpanGenConfig(n=3000, seed=31, tag_cardinality=200, n_ips=32)
+    pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 13,
+                            time_base_s=cfg.base_time_ns // 10**9)
+    payload = gen_span_payload(cfg)
+    pipe.ingest_frame_payload(payload)
+    pipe.rollups.flush()
+    pipe.ingest_frame_payload(payload)
+    eng = QueryEngine(pipe, device="cpu")
+    r = eng.query("SELECT Sum(request) AS s FROM application.1s")
+    assert r["values"][0][0] == 6000
+    r = eng.query("SELECT Sum(request) AS s FROM application.1m")
+    assert r["values"][0][0] == 6000
