@@ -270,82 +270,31 @@ def e2e_main(args) -> None:
                     torch.empty(args.batch, dtype=torch.int32,
                                 pin_memory=have_gpu)) for _ in range(16)]
         si = [0]
-        agg_lock = threading.Lock()
 
-        # frame aggregation: per-frame ingest (~90k spans) is kernel-
-        # launch bound; coalesce frames into one device buffer and run
-        # ONE ingest per ~1M spans. The dictionary harvest reads emitted
-        # slices from the device buffer (no host payload needed).
-        AGG_BYTES = 512 << 20
-        AGG_SPANS = 1_000_000
-        agg = {"buf": None, "offs": None, "lens": None,
-               "used": 0, "n": 0}
-
-        def _agg_reset():
-            if agg["buf"] is None and device == "cuda":
-                agg["buf"] = torch.empty(AGG_BYTES, dtype=torch.uint8,
-                                         device="cuda")
-                agg["offs"] = torch.empty(4 * args.batch,
-                                          dtype=torch.int32,
-                                          device="cuda")
-                agg["lens"] = torch.empty(4 * args.batch,
-                                          dtype=torch.int32,
-                                          device="cuda")
-            agg["used"] = 0
-            agg["n"] = 0
-
-        def _agg_flush():
-            if agg["n"] == 0:
-                return
-            pipe.ingest_device(agg["buf"][:agg["used"]],
-                               agg["offs"][:agg["n"]],
-                               agg["lens"][:agg["n"]],
-                               agg["buf"][:agg["used"]])
-            agg["used"] = 0
-            agg["n"] = 0
+        # GPU: the product coalescing feeder (~1M-span device ingests,
+        # ingest/native_pump.GpuL7Feeder — the same component the
+        # server's native_pump data plane uses); CPU: per-frame ingest
+        feeder = None
+        if device == "cuda":
+            from deepflow_amd.ingest.native_pump import GpuL7Feeder
+            feeder = GpuL7Feeder(pipe, max_records=args.batch)
 
         def on_pump(view, meta):
-            # agg state is shared with the wait_rows tail flush
-            with agg_lock:
-                return _on_pump_locked(view, meta)
-
-        def _on_pump_locked(view, meta):
+            if feeder is not None:
+                return feeder(view, meta)
             offs_p, lens_p = scratch[si[0] % len(scratch)]
             si[0] += 1
             n = int(lib.df_scan_offsets(
                 ct.c_void_p(view.ctypes.data), len(view),
                 ct.c_void_p(offs_p.data_ptr()),
                 ct.c_void_p(lens_p.data_ptr()), args.batch))
-            if device == "cuda":
-                if agg["buf"] is None:
-                    _agg_reset()
-                if agg["used"] + len(view) > AGG_BYTES or                         agg["n"] + n > 4 * args.batch:
-                    _agg_flush()
-                base = agg["used"]
-                pay_t = torch.from_numpy(view)  # aliases pinned ring
-                agg["buf"][base:base + len(view)].copy_(
-                    pay_t, non_blocking=True)
-                # record offsets are frame-relative; rebase into the
-                # aggregation buffer before the async copy
-                offs_np = offs_p.numpy()
-                offs_np[:n] += base
-                agg["offs"][agg["n"]:agg["n"] + n].copy_(
-                    offs_p[:n], non_blocking=True)
-                agg["lens"][agg["n"]:agg["n"] + n].copy_(
-                    lens_p[:n], non_blocking=True)
-                agg["used"] += (len(view) + 7) & ~7
-                agg["n"] += n
-                ev = torch.cuda.Event()
-                ev.record()
-                if agg["n"] >= AGG_SPANS:
-                    _agg_flush()
-                return ev
             pipe.ingest(view,
                         offs_p.numpy().view(np.uint32)[:n].copy(),
                         lens_p.numpy().view(np.uint32)[:n].copy())
             return None
 
-        rx = PumpServer(on_pump, ring_bytes=256 << 20, pin=have_gpu)
+        rx = PumpServer(on_pump, ring_bytes=256 << 20, pin=have_gpu,
+                        idle_handler=feeder.idle if feeder else None)
         rx.start()
         rx.tcp_port = rx.port
     else:
@@ -397,12 +346,7 @@ def e2e_main(args) -> None:
     def wait_rows(target: int, timeout=120.0):
         t_end = time.time() + timeout
         while pipe.stats.spans_in < target and time.time() < t_end:
-            if use_native and device == "cuda" and agg["n"] and \
-                    not rx._pending and all(
-                        p.pending() == 0 for p in rx.pumps):
-                # drained the wire but a partial aggregation remains
-                with agg_lock:
-                    _agg_flush()
+
             time.sleep(0.002)
         if pipe.stats.spans_in < target:
             rx_stats = rx.stats() if hasattr(rx, "stats") \

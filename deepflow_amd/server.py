@@ -436,9 +436,14 @@ class DeepflowServer:
 
     def _on_pump_frame(self, view, meta):
         """PumpServer dispatch: frame metadata -> the registered
-        per-type handler. The pinned view is only valid until return,
-        so handlers that keep bytes get a copy."""
+        per-type handler. L7 flow-log frames on a GPU server take the
+        zero-copy coalescing feeder; everything else copies out of the
+        ring and runs the normal handler."""
         msg_type, agent_id, org_id, team_id = meta
+        if msg_type == framing.MSG_PROTOCOLLOG and \
+                getattr(self, "_l7_feeder", None) is not None and \
+                (org_id or 1) == self.default_org:
+            return self._l7_feeder(view, meta)
         handler = self.receiver.handlers.get(msg_type)
         if handler is None:
             self.receiver.counter.add("unhandled_type")
@@ -537,10 +542,21 @@ class DeepflowServer:
         gc.freeze()
         self.receiver.start()
         if self._use_pump:
-            from .ingest.native_pump import PumpServer
+            from .ingest.native_pump import PumpServer, GpuL7Feeder
+            idle = None
+            if self.device == "cuda":
+                # GPU data plane: L7 frames bypass the per-frame python
+                # dispatch — pinned views coalesce into ~1M-span device
+                # ingests (the bench --path e2e fast path, as a product
+                # component)
+                self._l7_feeder = GpuL7Feeder(self.l7)
+                idle = self._l7_feeder.idle
+            else:
+                self._l7_feeder = None
             self.pump = PumpServer(self._on_pump_frame, port=self._pump_port,
                                    accept_type=-1,
-                                   pin=self.device == "cuda").start()
+                                   pin=self.device == "cuda",
+                                   idle_handler=idle).start()
             self.pump_port = self.pump.port
         self.debug_bus.start()
 

@@ -315,3 +315,36 @@ def test_server_pump_dispatches_all_types():
         assert r["values"][0][0] == 10
     finally:
         srv.stop()
+
+
+@pytest.mark.gpu
+def test_server_native_pump_gpu_feeder():
+    """GPU server + native_pump: L7 frames take the coalescing
+    GpuL7Feeder (zero-copy pinned ring -> device ingest) and land in
+    SQL; other frame types still dispatch through handlers."""
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.gen.spans import SpanGenConfig, gen_span_payload
+
+    srv = DeepflowServer(device="cuda", tcp_port=0, segment_rows=1 << 14,
+                         dict_capacity=1 << 15, native_pump=True,
+                         time_base_s=0)
+    srv.start()
+    try:
+        assert srv._l7_feeder is not None
+        cfg = SpanGenConfig(n=4000, seed=13, tag_cardinality=200,
+                            n_ips=64, base_time_ns=10**9)
+        records = gen_span_payload(cfg)
+        fr = _frame(records, zstd=True)
+        s = socket.create_connection(("127.0.0.1", srv.pump_port))
+        s.sendall(fr)
+        t_end = time.time() + 20
+        while srv.l7.stats.spans_in < 4000 and time.time() < t_end:
+            time.sleep(0.05)
+        s.close()
+        import torch
+        torch.cuda.synchronize()
+        assert srv.l7.stats.spans_in == 4000
+        r = srv.engine.query("SELECT COUNT(1) FROM l7_flow_log")
+        assert r["values"][0][0] == 4000
+    finally:
+        srv.stop()
