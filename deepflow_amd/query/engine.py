@@ -121,6 +121,29 @@ def _result_rows(res: Dict) -> List[Dict]:
     return [dict(zip(res["columns"], row)) for row in res["values"]]
 
 
+def _plan_time_range(plan):
+    """(lo_ns, hi_ns) from the plan's AND-level time predicates on
+    start_time (SRC_U64 col 0), or None — drives cold-segment pruning."""
+    lo, hi = None, None
+    for t in plan.terms:
+        if t.group != 0 or t.family != Q.SRC_U64 or t.idx != 0:
+            continue
+        if t.op == Q.OP_GE or t.op == Q.OP_GT:
+            lo = t.v0 if lo is None else max(lo, t.v0)
+        elif t.op == Q.OP_LE or t.op == Q.OP_LT:
+            hi = t.v0 if hi is None else min(hi, t.v0)
+        elif t.op == Q.OP_BETWEEN:
+            lo = t.v0 if lo is None else max(lo, t.v0)
+            hi = t.v1 if hi is None else min(hi, t.v1)
+        elif t.op == Q.OP_EQ:
+            lo = t.v0 if lo is None else max(lo, t.v0)
+            hi = t.v0 if hi is None else min(hi, t.v0)
+    if lo is None and hi is None:
+        return None
+    return (lo if lo is not None else 0,
+            hi if hi is not None else (1 << 63))
+
+
 def _plan_needed(plan) -> Optional[Dict]:
     """Column set a plan touches, by family — drives lazy cold-segment
     decompression. None = everything (row-fetch paths)."""
@@ -219,7 +242,9 @@ class QueryEngine:
                              name_maps=self.name_maps)
             return self._run_segments(
                 plan, self.pipe.segments.scan_list(
-                    needed=_plan_needed(plan)), L7_TAGS, S.STR_COLS)
+                    needed=_plan_needed(plan),
+                    time_range=_plan_time_range(plan)),
+                L7_TAGS, S.STR_COLS)
         if table == "l4_flow_log":
             if self.l4 is None:
                 raise SqlError("l4_flow_log table not enabled")
@@ -230,7 +255,9 @@ class QueryEngine:
             from ..store import l4_schema as L4S
             return self._run_segments(
                 plan, self.l4.segments.scan_list(
-                    needed=_plan_needed(plan)), L4_TAGS, L4S.STR_COLS)
+                    needed=_plan_needed(plan),
+                    time_range=_plan_time_range(plan)),
+                L4_TAGS, L4S.STR_COLS)
         row_tables = {
             "event": "event_rows", "perf_event": "perf_event_rows",
             "alert_event": "alert_event_rows",

@@ -122,7 +122,9 @@ def _cold_state(c) -> Dict:
                  else None,
                  "raw": pc.raw.cpu().clone() if pc.raw is not None
                  else None} for pc in lst]
-    return {"n_rows": c.n_rows, "capacity": c.capacity,
+    return {
+        "time_min": getattr(c, "time_min", 0),
+        "time_max": getattr(c, "time_max", None),"n_rows": c.n_rows, "capacity": c.capacity,
             "layout_version": c.layout_version,
             "u64_cols": cols(c.u64_cols), "u32_cols": cols(c.u32_cols),
             "did_cols": cols(c.did_cols),
@@ -163,6 +165,8 @@ def _cold_restore(state: Dict, device: str):
         setattr(c, name, t.to(device) if t is not None else None)
     c.attr_pool_len = state["attr_pool_len"]
     c.pool_len = state["pool_len"]
+    c.time_min = state.get("time_min", 0)
+    c.time_max = state.get("time_max", None)
     return c
 
 

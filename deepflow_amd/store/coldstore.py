@@ -209,6 +209,15 @@ class CompressedL7Segment:
             self.attr_pool_len = 0
         self.pool = seg.pool[: seg.pool_len].clone()
         self.pool_len = seg.pool_len
+        # time bounds of u64 col 0 (start_time, absolute ns): queries
+        # with a time predicate skip cold segments entirely outside it
+        if n:
+            t = seg.u64[0, :n]
+            self.time_min = int(t.min().item()) & ((1 << 64) - 1)
+            self.time_max = int(t.max().item()) & ((1 << 64) - 1)
+        else:
+            self.time_min = 0
+            self.time_max = 0
 
     def compressed_bytes(self) -> int:
         total = 0

@@ -253,15 +253,24 @@ class SegmentSet:
         self._free.append(seg)
         return True
 
-    def scan_list(self, stream: int = 0, needed=None) -> List:
+    def scan_list(self, stream: int = 0, needed=None,
+                  time_range=None) -> List:
         """All queryable segments: cold ones materialized into recycled
         scratch segments. `needed` restricts decompression to the columns
-        a query plan touches. Call release_scratch() when the query is
-        done — the scratch buffers go back to the free-list so they are
-        reused by ingest and counted by the watermark."""
+        a query plan touches; `time_range` = (lo_ns, hi_ns) prunes cold
+        segments whose recorded [time_min, time_max] lies entirely
+        outside the predicate (the reference's partition pruning). Call
+        release_scratch() when the query is done — the scratch buffers
+        go back to the free-list so they are reused by ingest and
+        counted by the watermark."""
         cold = getattr(self, "cold", [])
         if not cold:
             return self.segments
+        if time_range is not None:
+            lo, hi = time_range
+            cold = [c for c in cold
+                    if getattr(c, "time_max", None) is None or
+                    (c.time_max >= lo and c.time_min <= hi)]
         out = []
         if not hasattr(self, "_scratch"):
             self._scratch = []

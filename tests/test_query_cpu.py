@@ -287,3 +287,38 @@ def test_show_descriptions(eng):
     r = eng.query("SHOW tags FROM application_map.1s")
     names = {v[0] for v in r["values"]}
     assert {"ip_0", "ip_1", "server_port"} <= names
+
+
+def test_cold_segment_time_pruning():
+    """Queries with a time predicate skip cold segments entirely outside
+    it (partition pruning) and still return exact results."""
+    from deepflow_amd.gen.spans import SpanGenConfig, gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query import QueryEngine
+
+    base_s = 1_700_000_000
+    pipes = L7IngestPipeline(device="cpu", segment_rows=1 << 9,
+                             time_base_s=base_s)
+    # two 400-span batches an hour apart -> separate segments
+    for dt_h in (0, 1):
+        cfg = SpanGenConfig(n=400, seed=41 + dt_h, tag_cardinality=50,
+                            n_ips=16,
+                            base_time_ns=(base_s + dt_h * 3600) * 10**9)
+        pipes.ingest_frame_payload(gen_span_payload(cfg))
+    while pipes.segments.demote_oldest():
+        pass
+    cold = pipes.segments.cold
+    assert len(cold) >= 1   # hour-1 segment demoted; the tail stays hot
+    assert all(getattr(c, "time_max", 0) for c in cold)
+    # range covering only the second hour prunes the first segment
+    lo = (base_s + 3600) * 10**9
+    pruned = pipes.segments.scan_list(time_range=(lo, (1 << 63)))
+    full = pipes.segments.scan_list()
+    assert len(pruned) < len(full)
+    pipes.segments.release_scratch()
+    eng = QueryEngine(pipes, device="cpu")
+    r = eng.query("SELECT COUNT(1) FROM l7_flow_log "
+                  f"WHERE time >= {base_s + 3600}")
+    assert r["values"][0][0] == 400
+    r = eng.query("SELECT COUNT(1) FROM l7_flow_log")
+    assert r["values"][0][0] == 800
