@@ -120,6 +120,22 @@ class L4Segment:
         return fixed + self.pool_len / self.n_rows
 
 
+def _hot_overlaps(seg, lo: int, hi: int) -> bool:
+    """Hot-segment time pruning: per-segment [min,max] of u64 col 0
+    (start_time), cached until the row count changes. The tail segment
+    recomputes as it grows — one tiny device reduction per query."""
+    n = seg.n_rows
+    if n == 0:
+        return False
+    cache = getattr(seg, "_tb_cache", None)
+    if cache is None or cache[0] != n:
+        t = seg.u64[0, :n]
+        cache = (n, int(t.min().item()) & ((1 << 64) - 1),
+                 int(t.max().item()) & ((1 << 64) - 1))
+        seg._tb_cache = cache
+    return cache[2] >= lo and cache[1] <= hi
+
+
 class SegmentSet:
     """The shard-local hot window: ordered list of segments."""
 
@@ -264,13 +280,16 @@ class SegmentSet:
         go back to the free-list so they are reused by ingest and
         counted by the watermark."""
         cold = getattr(self, "cold", [])
-        if not cold:
-            return self.segments
+        hot = self.segments
         if time_range is not None:
             lo, hi = time_range
             cold = [c for c in cold
                     if getattr(c, "time_max", None) is None or
                     (c.time_max >= lo and c.time_min <= hi)]
+            hot = [g for g in hot
+                   if _hot_overlaps(g, lo, hi)]
+        if not cold:
+            return hot
         out = []
         if not hasattr(self, "_scratch"):
             self._scratch = []
@@ -283,7 +302,7 @@ class SegmentSet:
         for c, scratch in zip(cold, self._scratch):
             self.reset_segment(scratch)
             out.append(c.materialize(scratch, stream, needed=needed))
-        return out + self.segments
+        return out + hot
 
     def release_scratch(self) -> None:
         """Return cold-query scratch segments to the free-list (bounds the

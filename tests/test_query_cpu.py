@@ -322,3 +322,25 @@ def test_cold_segment_time_pruning():
     assert r["values"][0][0] == 400
     r = eng.query("SELECT COUNT(1) FROM l7_flow_log")
     assert r["values"][0][0] == 800
+
+
+def test_hot_segment_time_pruning():
+    from deepflow_amd.gen.spans import SpanGenConfig, gen_span_payload
+    from deepflow_amd.ingest import L7IngestPipeline
+    from deepflow_amd.query import QueryEngine
+
+    base_s = 1_700_000_000
+    pipe = L7IngestPipeline(device="cpu", segment_rows=1 << 9,
+                            time_base_s=base_s)
+    for dt_h in (0, 1):
+        cfg = SpanGenConfig(n=400, seed=51 + dt_h, tag_cardinality=50,
+                            n_ips=16,
+                            base_time_ns=(base_s + dt_h * 3600) * 10**9)
+        pipe.ingest_frame_payload(gen_span_payload(cfg))
+    assert len(pipe.segments.segments) == 2  # both hot
+    lo = (base_s + 3600) * 10**9
+    assert len(pipe.segments.scan_list(time_range=(lo, 1 << 63))) == 1
+    eng = QueryEngine(pipe, device="cpu")
+    r = eng.query("SELECT COUNT(1) FROM l7_flow_log "
+                  f"WHERE time >= {base_s + 3600}")
+    assert r["values"][0][0] == 400
