@@ -102,6 +102,10 @@ struct Pump {
 
     void publish(const uint8_t* payload, uint64_t n, uint64_t meta) {
         uint64_t need = 16 + ((n + 7) & ~7ull);
+        if (need + 16 > cap) {  // larger than the ring: count, drop
+            bad_frames.fetch_add(1, std::memory_order_relaxed);
+            return;
+        }
         uint64_t h = reserve(need);
         if (h == UINT64_MAX) return;
         uint64_t pos = h % cap;
@@ -118,6 +122,10 @@ struct Pump {
     // The caller fills the payload then calls commit(h, n).
     uint8_t* begin_entry(uint64_t n, uint64_t meta, uint64_t& h) {
         uint64_t need = 16 + ((n + 7) & ~7ull);
+        if (need + 16 > cap) {
+            h = UINT64_MAX - 1;  // caller drains the payload instead
+            return nullptr;
+        }
         h = reserve(need);
         if (h == UINT64_MAX) return nullptr;
         uint64_t pos = h % cap;
@@ -179,7 +187,22 @@ struct Pump {
                 // copy between the socket and pinned memory
                 uint64_t h;
                 uint8_t* dst = begin_entry(pn, meta, h);
-                if (dst == nullptr) break;
+                if (dst == nullptr) {
+                    if (h != UINT64_MAX - 1) break;  // stopping
+                    // frame larger than the ring: drain + count
+                    bad_frames.fetch_add(1, std::memory_order_relaxed);
+                    uint64_t left = pn;
+                    uint8_t sink[4096];
+                    bool ok = true;
+                    while (left) {
+                        uint64_t c =
+                            left < sizeof(sink) ? left : sizeof(sink);
+                        if (!read_exact(sink, c)) { ok = false; break; }
+                        left -= c;
+                    }
+                    if (!ok) break;
+                    continue;
+                }
                 if (!read_exact(dst, pn)) break;
                 commit(h, pn);
             } else if (encoder == ENCODER_ZSTD) {
@@ -199,7 +222,12 @@ struct Pump {
                     // reserved ring entry
                     uint64_t h;
                     uint8_t* dst = begin_entry(want, meta, h);
-                    if (dst == nullptr) break;
+                    if (dst == nullptr) {
+                        if (h != UINT64_MAX - 1) break;  // stopping
+                        bad_frames.fetch_add(1,
+                                             std::memory_order_relaxed);
+                        continue;  // decompressed size exceeds the ring
+                    }
                     size_t r = zdec(dst, want, fbuf.data(), pn);
                     if (ziserr(r)) {
                         // commit a zero-length entry to keep cursors

@@ -348,3 +348,18 @@ def test_server_native_pump_gpu_feeder():
         assert r["values"][0][0] == 4000
     finally:
         srv.stop()
+
+
+def test_pump_frame_larger_than_ring_dropped():
+    """A frame whose payload exceeds the ring capacity is drained and
+    counted as bad instead of deadlocking the producer."""
+    a, p = _pump_pair(ring_bytes=1 << 14)  # 16 KB ring
+    try:
+        big = b"Z" * (64 << 10)            # 64 KB payload
+        a.sendall(_frame(big) + _frame(b"after"))
+        got = _drain(p, 1, timeout=10)
+        assert got == [b"after"]
+        assert p.stats()["bad_frames"] == 1
+    finally:
+        a.close()
+        p.close()
