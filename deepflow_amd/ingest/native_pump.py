@@ -222,9 +222,13 @@ class GpuL7Feeder:
     `idle` flushes a partial aggregation once the wire goes quiet."""
 
     def __init__(self, pipe, max_records: int = 1 << 21,
-                 agg_spans: int = 1_000_000, agg_bytes: int = 512 << 20):
+                 agg_spans: int = 1_000_000, agg_bytes: int = 512 << 20,
+                 ingest_lock=None):
         import ctypes as ct
         self.pipe = pipe
+        # serializes the coalesced ingest against concurrent queries
+        # when the owner (the server) shares state with an engine
+        self.ingest_lock = ingest_lock
         self.max_records = max_records
         self.agg_spans = agg_spans
         self.agg_bytes = agg_bytes
@@ -250,10 +254,17 @@ class GpuL7Feeder:
     def _flush_locked(self) -> None:
         if self._n == 0:
             return
-        self.pipe.ingest_device(self._buf[: self._used],
-                                self._offs[: self._n],
-                                self._lens[: self._n],
-                                self._buf[: self._used])
+        if self.ingest_lock is not None:
+            with self.ingest_lock:
+                self.pipe.ingest_device(self._buf[: self._used],
+                                        self._offs[: self._n],
+                                        self._lens[: self._n],
+                                        self._buf[: self._used])
+        else:
+            self.pipe.ingest_device(self._buf[: self._used],
+                                    self._offs[: self._n],
+                                    self._lens[: self._n],
+                                    self._buf[: self._used])
         self._used = 0
         self._n = 0
 

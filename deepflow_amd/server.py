@@ -548,8 +548,10 @@ class DeepflowServer:
                 # GPU data plane: L7 frames bypass the per-frame python
                 # dispatch — pinned views coalesce into ~1M-span device
                 # ingests (the bench --path e2e fast path, as a product
-                # component)
-                self._l7_feeder = GpuL7Feeder(self.l7)
+                # component). The feeder's flushes serialize against
+                # queries through the server-wide ingest lock.
+                self._l7_feeder = GpuL7Feeder(self.l7,
+                                              ingest_lock=self._lock)
                 idle = self._l7_feeder.idle
             else:
                 self._l7_feeder = None
