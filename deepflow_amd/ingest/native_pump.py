@@ -234,11 +234,14 @@ class GpuL7Feeder:
         self.agg_bytes = agg_bytes
         self.lib = native.cpu()
         self._ct = ct
+        # 6-deep rotation: each pair's async H2D completes in us — deep
+        # enough to never overwrite an in-flight copy, shallow enough to
+        # keep pinned memory bounded (max_records=2^21 -> 96 MB)
         self._scratch = [(torch.empty(max_records, dtype=torch.int32,
                                       pin_memory=True),
                           torch.empty(max_records, dtype=torch.int32,
                                       pin_memory=True))
-                         for _ in range(16)]
+                         for _ in range(6)]
         self._si = 0
         self._buf = torch.empty(agg_bytes, dtype=torch.uint8,
                                 device="cuda")
