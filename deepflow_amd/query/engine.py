@@ -412,7 +412,9 @@ class QueryEngine:
                              time_base_s=self.pipe.time_base_s,
                              tags=L7_TAGS, metrics=L7_METRICS,
                              name_maps=self.name_maps)
-            segments, tags, str_cols = (self.pipe.segments.scan_list(), L7_TAGS,
+            segments, tags, str_cols = (self.pipe.segments.scan_list(
+                needed=_plan_needed(plan),
+                time_range=_plan_time_range(plan)), L7_TAGS,
                                         S.STR_COLS)
         else:
             from ..store import l4_schema as L4S
@@ -420,7 +422,9 @@ class QueryEngine:
                              time_base_s=self.l4.time_base_s,
                              tags=L4_TAGS, metrics=L4_METRICS,
                              name_maps=self.name_maps)
-            segments, tags, str_cols = (self.l4.segments.scan_list(), L4_TAGS,
+            segments, tags, str_cols = (self.l4.segments.scan_list(
+                needed=_plan_needed(plan),
+                time_range=_plan_time_range(plan)), L4_TAGS,
                                         L4S.STR_COLS)
         if plan.select_rows:
             return {"kind": "rows",

@@ -386,6 +386,13 @@ class RollupTable:
         keys, vals = self._harvest_np()
         if keys.shape[0]:
             self.archive.append((keys, vals))
+            if len(self.archive) > 8:
+                # compact: one lexsort merge bounds archive memory and
+                # read-time concat cost (groups recurring across flush
+                # windows collapse)
+                ak = np.concatenate([k for k, _ in self.archive], axis=0)
+                av = np.concatenate([v for _, v in self.archive], axis=0)
+                self.archive = [_np_merge(ak, av, self.ops)]
         if self.device == "cpu":
             self.table.clear()
         else:
