@@ -91,11 +91,18 @@ def map_next_key(fd: int, key: Optional[bytes], key_size: int) -> Optional[bytes
 
 def prog_load(prog_type: int, insns: bytes, license_: bytes = b"GPL",
               log: bool = False) -> int:
+    """Load a program; on failure re-load with the verifier log for a
+    useful error. NOTE: a verbose (level-2) log that overflows its
+    buffer FAILS the load with ENOSPC even for valid programs — the
+    buffer here is 8 MB, enough for ~500-insn programs."""
     lic = ct.create_string_buffer(license_, len(license_) + 1)
-    log_buf = ct.create_string_buffer(1 << 18) if log else None
+    log_buf = ct.create_string_buffer(1 << 23) if log else None
+    # the insns buffer must outlive the syscall (a temporary here once
+    # handed the kernel freed memory -> garbage verifier errors)
+    ibuf = ct.create_string_buffer(insns, len(insns))
     attr = struct.pack(
         "<IIQQIIQI4x", prog_type, len(insns) // 8,
-        ct.addressof(ct.create_string_buffer(insns, len(insns))),
+        ct.addressof(ibuf),
         ct.addressof(lic), 2 if log else 0,
         len(log_buf) if log else 0,
         ct.addressof(log_buf) if log else 0, 0)

@@ -363,6 +363,12 @@ def build_sys_exit() -> Asm:
     a.ld_imm64(R3, 0xFFFFFFFF)            # BPF_F_CURRENT_CPU
     a.mov64(R4, R9)
     a.ldx(BPF_W, R5, R9, EV_CAP)
+    # re-bound after the map round-trip: the verifier forgets the
+    # earlier clamp once the value passes through the per-cpu scratch
+    a.alu64_imm(BPF_AND, R5, 0xFF)
+    a.jmp_imm(BPF_JLE, R5, CAP_LEN, "evsz_ok")
+    a.mov64_imm(R5, CAP_LEN)
+    a.label("evsz_ok")
     a.alu64_imm(BPF_ADD, R5, EV_HDR)
     a.call(H_PERF_EVENT_OUTPUT)
     a.mov64_imm(R0, 0)
@@ -438,6 +444,10 @@ def _ssl_emit(a: Asm, syscall_marker: int, direction: int) -> None:
     a.ld_imm64(R3, 0xFFFFFFFF)            # BPF_F_CURRENT_CPU
     a.mov64(R4, R9)
     a.ldx(BPF_W, R5, R9, EV_CAP)
+    a.alu64_imm(BPF_AND, R5, 0xFF)
+    a.jmp_imm(BPF_JLE, R5, CAP_LEN, "evsz_ok")
+    a.mov64_imm(R5, CAP_LEN)
+    a.label("evsz_ok")
     a.alu64_imm(BPF_ADD, R5, EV_HDR)
     a.call(H_PERF_EVENT_OUTPUT)
     a.mov64_imm(R0, 0)
