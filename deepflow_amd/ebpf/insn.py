@@ -52,6 +52,7 @@ H_GET_COMM = 16
 H_PERF_EVENT_OUTPUT = 25
 H_GET_STACKID = 27
 H_PROBE_READ_USER = 112
+H_PROBE_READ_KERNEL = 113
 
 R0, R1, R2, R3, R4, R5, R6, R7, R8, R9, R10 = range(11)
 

@@ -115,6 +115,36 @@ def prog_load(prog_type: int, insns: bytes, license_: bytes = b"GPL",
         raise
 
 
+BPF_RAW_TRACEPOINT_OPEN = 17
+BPF_PROG_TYPE_RAW_TRACEPOINT = 17
+
+
+def raw_tracepoint_open(name: str, prog_fd: int) -> int:
+    """Attach a RAW_TRACEPOINT program by name — no tracefs needed
+    (works in unprivileged-ish containers where
+    /sys/kernel/tracing is not mounted)."""
+    nb = ct.create_string_buffer(name.encode() + b"\x00")
+    attr = struct.pack("<QI4x", ct.addressof(nb), prog_fd)
+    return _bpf(BPF_RAW_TRACEPOINT_OPEN, attr)
+
+
+def raw_tracepoint_available() -> bool:
+    """Can this process attach raw tracepoints here?"""
+    try:
+        # r0 = 0; exit
+        prog = (b"\xb7\x00\x00\x00\x00\x00\x00\x00"
+                b"\x95\x00\x00\x00\x00\x00\x00\x00")
+        pfd = prog_load(BPF_PROG_TYPE_RAW_TRACEPOINT, prog)
+        try:
+            tfd = raw_tracepoint_open("sys_enter", pfd)
+            os.close(tfd)
+            return True
+        finally:
+            os.close(pfd)
+    except OSError:
+        return False
+
+
 def _tracefs() -> Optional[str]:
     for p in ("/sys/kernel/tracing", "/sys/kernel/debug/tracing"):
         if os.path.isdir(os.path.join(p, "events")):
@@ -250,15 +280,14 @@ def find_libssl() -> Optional[str]:
 
 
 def available() -> bool:
-    """Can this process load + attach BPF programs here?"""
-    if _tracefs() is None:
-        return False
+    """Can this process load + attach BPF programs here? True with
+    tracefs OR raw-tracepoint attach (container-friendly)."""
     try:
         fd = map_create(1, 8, 8, 4)
         os.close(fd)
-        return True
     except OSError:
         return False
+    return _tracefs() is not None or raw_tracepoint_available()
 
 
 class SocketTracer:
@@ -268,19 +297,29 @@ class SocketTracer:
     (PERF_EVENT_IOC_SET_BPF on the tracepoint event delivers
     bpf_perf_event_output records into the same buffers)."""
 
-    def __init__(self, with_tls: bool = True):
+    def __init__(self, with_tls: bool = True, raw: Optional[bool] = None):
         from .progs import (MAPS, SSL_MAPS, build_sys_enter,
                             build_sys_exit, build_ssl_write,
                             build_ssl_read_enter, build_ssl_read_exit)
+        if raw is None:
+            # raw tracepoints need no tracefs and attach in containers
+            # where /sys/kernel/tracing is absent — prefer them
+            raw = _tracefs() is None
+        self.raw = raw
         self.map_fds: Dict[str, int] = {}
         for name, spec in MAPS.items():
-            self.map_fds[name] = map_create(*spec)
+            t_, k_, v_, n_ = spec
+            if n_ == 0:  # perf-event arrays size to the cpu count
+                n_ = os.cpu_count() or 1
+            self.map_fds[name] = map_create(t_, k_, v_, n_)
+        ptype = BPF_PROG_TYPE_RAW_TRACEPOINT if raw \
+            else BPF_PROG_TYPE_TRACEPOINT
         self.enter_fd = prog_load(
-            BPF_PROG_TYPE_TRACEPOINT,
-            build_sys_enter().to_bytes(self.map_fds), log=True)
+            ptype, build_sys_enter(raw=raw).to_bytes(self.map_fds),
+            log=True)
         self.exit_fd = prog_load(
-            BPF_PROG_TYPE_TRACEPOINT,
-            build_sys_exit().to_bytes(self.map_fds), log=True)
+            ptype, build_sys_exit(raw=raw).to_bytes(self.map_fds),
+            log=True)
         self.ssl_fds: List[int] = []
         self.libssl = find_libssl() if with_tls else None
         if self.libssl is not None:
@@ -296,10 +335,16 @@ class SocketTracer:
         self.rings: List[tuple] = []
 
     def attach(self) -> None:
-        self.tp_fds.append(attach_tracepoint(self.enter_fd, "raw_syscalls",
-                                             "sys_enter"))
-        self.tp_fds.append(attach_tracepoint(self.exit_fd, "raw_syscalls",
-                                             "sys_exit"))
+        if self.raw:
+            self.tp_fds.append(raw_tracepoint_open("sys_enter",
+                                                   self.enter_fd))
+            self.tp_fds.append(raw_tracepoint_open("sys_exit",
+                                                   self.exit_fd))
+        else:
+            self.tp_fds.append(attach_tracepoint(
+                self.enter_fd, "raw_syscalls", "sys_enter"))
+            self.tp_fds.append(attach_tracepoint(
+                self.exit_fd, "raw_syscalls", "sys_exit"))
         if self.libssl is not None and self.ssl_fds:
             # TLS plaintext capture: uprobes on OpenSSL entry points
             # (reference kernel/openssl.bpf.c)
@@ -347,9 +392,18 @@ class SocketTracer:
             size = page * pages
             while tail < head:
                 off = page + (tail % size)
-                etype, _misc, esize = struct.unpack_from("<IHH", buf, off)
-                rec = bytes(buf[off + 8: off + esize])
-                if etype == 1:  # PERF_RECORD_SAMPLE: u32 size + raw data
+                hdr = bytes(buf[off: off + 8]) if off + 8 <= page + size \
+                    else bytes(buf[off: page + size]) + \
+                    bytes(buf[page: page + 8 - (page + size - off)])
+                etype, _misc, esize = struct.unpack("<IHH", hdr)
+                body_off = (off + 8 - page) % size + page
+                end = body_off + (esize - 8)
+                if end <= page + size:
+                    rec = bytes(buf[body_off: end])
+                else:  # record wraps the ring
+                    rec = bytes(buf[body_off: page + size]) + \
+                        bytes(buf[page: page + end - (page + size)])
+                if etype == 9:  # PERF_RECORD_SAMPLE: u32 size + raw
                     raw_size = struct.unpack_from("<I", rec, 0)[0]
                     on_event(rec[4:4 + raw_size])
                     n += 1

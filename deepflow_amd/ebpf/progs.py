@@ -27,7 +27,8 @@ from .insn import (Asm, BPF_ADD, BPF_AND, BPF_B, BPF_DW, BPF_H, BPF_JEQ,
                    BPF_JSLE, BPF_LSH, BPF_MOV, BPF_OR, BPF_RSH, BPF_SUB,
                    BPF_W, H_GET_PID_TGID, H_GET_SMP_PROC_ID, H_GET_STACKID,
                    H_KTIME_GET_NS, H_MAP_DELETE, H_MAP_LOOKUP, H_MAP_UPDATE,
-                   H_PERF_EVENT_OUTPUT, H_PROBE_READ_USER, R0, R1, R2, R3,
+                   H_PERF_EVENT_OUTPUT, H_PROBE_READ_KERNEL,
+                   H_PROBE_READ_USER, R0, R1, R2, R3,
                    R4, R5, R6, R7, R8, R9, R10)
 from .inference import SPEC
 
@@ -85,10 +86,14 @@ PROFILER_MAPS = {
 }
 
 
-def build_sys_enter() -> Asm:
+def build_sys_enter(raw: bool = False) -> Asm:
+    """raw=True targets BPF_RAW_TRACEPOINT_OPEN (no tracefs needed):
+    ctx = {args[0]=struct pt_regs*, args[1]=syscall id}; syscall args
+    read from pt_regs with bpf_probe_read_kernel. raw=False is the
+    classic tracefs sys_enter layout (id@8, args@16)."""
     a = Asm()
     a.mov64(R6, R1)                       # r6 = ctx
-    a.ldx(BPF_DW, R8, R6, 8)              # r8 = syscall id
+    a.ldx(BPF_DW, R8, R6, 8)              # r8 = syscall id (both layouts)
     for sc in TRACED:
         a.jmp_imm(BPF_JEQ, R8, sc, "trace")
     a.mov64_imm(R0, 0)
@@ -99,10 +104,27 @@ def build_sys_enter() -> Asm:
     a.stx(BPF_DW, R10, -8, R7)            # key = pid_tgid
     # value {syscall, fd, buf} at fp-32
     a.stx(BPF_DW, R10, -32, R8)
-    a.ldx(BPF_DW, R2, R6, 16)             # args[0] = fd
-    a.stx(BPF_DW, R10, -24, R2)
-    a.ldx(BPF_DW, R2, R6, 24)             # args[1] = buf
-    a.stx(BPF_DW, R10, -16, R2)
+    if raw:
+        a.ldx(BPF_DW, R9, R6, 0)          # struct pt_regs *
+        # fd = regs->rdi -> fp-24
+        a.mov64(R1, R10)
+        a.alu64_imm(BPF_ADD, R1, -24)
+        a.mov64_imm(R2, 8)
+        a.mov64(R3, R9)
+        a.alu64_imm(BPF_ADD, R3, 112)     # offsetof(pt_regs, rdi)
+        a.call(H_PROBE_READ_KERNEL)
+        # buf = regs->rsi -> fp-16
+        a.mov64(R1, R10)
+        a.alu64_imm(BPF_ADD, R1, -16)
+        a.mov64_imm(R2, 8)
+        a.mov64(R3, R9)
+        a.alu64_imm(BPF_ADD, R3, 104)     # offsetof(pt_regs, rsi)
+        a.call(H_PROBE_READ_KERNEL)
+    else:
+        a.ldx(BPF_DW, R2, R6, 16)         # args[0] = fd
+        a.stx(BPF_DW, R10, -24, R2)
+        a.ldx(BPF_DW, R2, R6, 24)         # args[1] = buf
+        a.stx(BPF_DW, R10, -16, R2)
     a.ld_map_fd(R1, "active")
     a.mov64(R2, R10)
     a.alu64_imm(BPF_ADD, R2, -8)
@@ -205,7 +227,7 @@ def _emit_inference(a: Asm) -> None:
     a.label("infer_done")
 
 
-def build_sys_exit() -> Asm:
+def build_sys_exit(raw: bool = False) -> Asm:
     a = Asm()
     a.mov64(R6, R1)                       # r6 = ctx
     a.call(H_GET_PID_TGID)
@@ -231,8 +253,8 @@ def build_sys_exit() -> Asm:
     a.mov64(R2, R10)
     a.alu64_imm(BPF_ADD, R2, -8)
     a.call(H_MAP_DELETE)
-    # ret <= 0 -> done
-    a.ldx(BPF_DW, R2, R6, 16)
+    # ret <= 0 -> done (raw tp: {pt_regs*, ret}; tracefs tp: {id, ret})
+    a.ldx(BPF_DW, R2, R6, 8 if raw else 16)
     a.jmp_imm(BPF_JSGE, R2, 1, "have_ret")
     a.mov64_imm(R0, 0)
     a.exit()

@@ -2869,6 +2869,19 @@ int dfa_syscall_event(void* h, uint64_t ts_ns, uint32_t tgid, int dir,
     }
     FlowNode& f = it->second;
     f.last_ns = ts_ns;
+    // protocol upgrade: a flow created from pre-handshake noise (an
+    // unrelated read on a RECYCLED fd number racing the /proc resolver)
+    // locks to unknown with junk-derived orientation. When the first
+    // real in-kernel inference verdict arrives and the flow has emitted
+    // nothing, adopt the verdict AND re-seed the client side from this
+    // event's data sender (inference fires on request-shaped data).
+    if (f.l7_protocol == 0 && l7_hint &&
+        f.l7c.request_count == 0 && f.l7c.response_count == 0) {
+        f.l7_protocol = l7_hint;
+        f.ip[0] = ip_src; f.ip[1] = ip_dst;
+        f.port[0] = port_src; f.port[1] = port_dst;
+        f.l7.active = false;
+    }
     // data direction: which flow side sent these bytes
     int data_dir = (ip_src == f.ip[0] && port_src == f.port[0]) ? 0 : 1;
     // the local process sits on the side it writes from / reads toward

@@ -359,3 +359,78 @@ def test_all_programs_pass_kernel_verifier():
             (P.build_profiler, BPF_PROG_TYPE_PERF_EVENT, pm)):
         fd = loader.prog_load(ptype, builder().to_bytes(m), log=True)
         assert fd > 0, builder.__name__
+
+
+@pytest.mark.skipif(not __import__("deepflow_amd.ebpf.loader",
+                                   fromlist=["available"]).available(),
+                    reason="bpf attach not permitted here")
+def test_live_kernel_capture_to_flow_logs():
+    """The WHOLE eBPF reference capability, live against the real
+    kernel: assembled programs -> verifier -> raw-tracepoint attach ->
+    in-kernel protocol inference on our own loopback HTTP syscalls ->
+    perf rings -> /proc socket resolution -> agent FlowMap -> flow
+    logs via SQL. No tracefs needed (BPF_RAW_TRACEPOINT_OPEN)."""
+    import os
+    import socket
+    import struct
+    import time as _t
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.ebpf.loader import SocketTracer
+    from deepflow_amd.ebpf.runtime import EbpfCollector, ProcSocketResolver
+    from deepflow_amd.ebpf import progs as P
+    from deepflow_amd.wire import pb, flow_log, framing
+
+    t = SocketTracer(with_tls=False)
+    t.attach()
+    try:
+        srv = socket.socket()
+        srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        srv.bind(("127.0.0.1", 0))
+        srv.listen(1)
+        cli = socket.create_connection(("127.0.0.1",
+                                        srv.getsockname()[1]))
+        conn, _ = srv.accept()
+        # resolver snapshot while the sockets are open
+        resolver = ProcSocketResolver()
+        os.write(cli.fileno(), b"GET /live HTTP/1.1\r\n"
+                               b"Host: ci.test\r\n\r\n")
+        os.read(conn.fileno(), 4096)
+        os.write(conn.fileno(),
+                 b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+        os.read(cli.fileno(), 4096)
+        events = []
+        deadline = _t.time() + 5
+        while _t.time() < deadline:
+            t.poll(events.append)
+            if sum(1 for e in events
+                   if len(e) >= P.EV_HDR and
+                   struct.unpack_from(P.SK_EVENT_FMT, e)[1] ==
+                   os.getpid() and
+                   struct.unpack_from(P.SK_EVENT_FMT, e)[7]) >= 4:
+                break
+            _t.sleep(0.05)
+        me = os.getpid()
+        mine = [e for e in events
+                if len(e) >= P.EV_HDR and
+                struct.unpack_from(P.SK_EVENT_FMT, e)[1] == me]
+        assert len(mine) >= 4, f"captured {len(mine)} of our events"
+        agent = Agent(vtap_id=3)
+        coll = EbpfCollector(agent, resolver)
+        # resolve while the sockets are still open: (tgid, fd) ->
+        # 4-tuple reads the LIVE /proc fd link
+        for e in mine:
+            coll.on_event(e)
+        coll.flush()
+        cli.close()
+        conn.close()
+        srv.close()
+    finally:
+        t.close()
+    agent.tick(2_000_000_000_000_000_000)
+    recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+            for r in framing.iter_records(agent.drain(1))]
+    live = [r for r in recs if r["req"].get("resource") == "/live"]
+    assert live, [r.get("req") for r in recs]
+    assert live[0]["req"]["domain"] == "ci.test"
+    assert live[0]["base"]["head"]["proto"] == 20
+    agent.close()
