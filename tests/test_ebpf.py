@@ -434,3 +434,51 @@ def test_live_kernel_capture_to_flow_logs():
     assert live[0]["req"]["domain"] == "ci.test"
     assert live[0]["base"]["head"]["proto"] == 20
     agent.close()
+
+
+@pytest.mark.skipif(not __import__("deepflow_amd.ebpf.loader",
+                                   fromlist=["available"]).available(),
+                    reason="bpf attach not permitted here")
+def test_agent_start_ebpf_live():
+    """Agent.start_ebpf() — the production wiring — attaches the raw
+    tracer, pumps perf events on its thread, and our loopback HTTP
+    round trip lands as a flow log."""
+    import os
+    import socket
+    import time as _t
+    from deepflow_amd.agent import Agent
+    from deepflow_amd.wire import pb, flow_log, framing
+
+    a = Agent(vtap_id=5)
+    tr = a.start_ebpf()
+    assert tr is not None
+    try:
+        srv = socket.socket()
+        srv.bind(("127.0.0.1", 0))
+        srv.listen(1)
+        cli = socket.create_connection(("127.0.0.1",
+                                        srv.getsockname()[1]))
+        conn, _ = srv.accept()
+        os.write(cli.fileno(),
+                 b"GET /agent-live HTTP/1.1\r\nHost: a.test\r\n\r\n")
+        os.read(conn.fileno(), 4096)
+        os.write(conn.fileno(),
+                 b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+        os.read(cli.fileno(), 4096)
+        deadline = _t.time() + 5
+        hits = []
+        while _t.time() < deadline and not hits:
+            _t.sleep(0.3)
+            a.tick(2_000_000_000_000_000_000)
+            recs = [pb.decode(r, flow_log.APP_PROTO_LOGS_DATA)
+                    for r in framing.iter_records(a.drain(1))]
+            hits = [r for r in recs
+                    if r["req"].get("resource") == "/agent-live"]
+        cli.close()
+        conn.close()
+        srv.close()
+        assert hits
+        assert hits[0]["base"]["head"]["proto"] == 20
+    finally:
+        a.stop_ebpf()
+        a.close()
