@@ -268,6 +268,46 @@ def attach_uprobe(prog_fd: int, path: str, offset: int,
     return fd
 
 
+class ProfilerTracer:
+    """Continuous OnCPU profiler, live: a PERF_TYPE_SOFTWARE/CPU_CLOCK
+    sampling event per cpu with the assembled BPF program attached
+    (PERF_EVENT_IOC_SET_BPF) counting (tgid, ustack, kstack) into the
+    stack maps — no tracefs needed. Reference: perf_profiler.bpf.c."""
+
+    BPF_PROG_TYPE_PERF_EVENT = 7
+    PERF_COUNT_SW_CPU_CLOCK = 0
+
+    def __init__(self, sample_freq: int = 99):
+        from .progs import PROFILER_MAPS, build_profiler
+        self.sample_freq = sample_freq
+        self.map_fds = {n: map_create(*spec)
+                        for n, spec in PROFILER_MAPS.items()}
+        self.prog_fd = prog_load(
+            self.BPF_PROG_TYPE_PERF_EVENT,
+            build_profiler().to_bytes(self.map_fds), log=True)
+        self.fds: List[int] = []
+
+    def attach(self) -> None:
+        import fcntl
+        for cpu in range(os.cpu_count() or 1):
+            attr = bytearray(112)
+            struct.pack_into("<IIQQ", attr, 0, PERF_TYPE_SOFTWARE, 112,
+                             self.PERF_COUNT_SW_CPU_CLOCK,
+                             self.sample_freq)   # sample_freq (freq=1)
+            struct.pack_into("<Q", attr, 40, 1 << 10)  # flags: freq
+            fd = perf_event_open(bytes(attr), -1, cpu)
+            fcntl.ioctl(fd, PERF_EVENT_IOC_SET_BPF, self.prog_fd)
+            fcntl.ioctl(fd, PERF_EVENT_IOC_ENABLE, 0)
+            self.fds.append(fd)
+
+    def close(self) -> None:
+        for fd in self.fds:
+            os.close(fd)
+        os.close(self.prog_fd)
+        for fd in self.map_fds.values():
+            os.close(fd)
+
+
 def find_libssl() -> Optional[str]:
     import glob as _glob
     for pat in ("/usr/lib/x86_64-linux-gnu/libssl.so*",

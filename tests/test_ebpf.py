@@ -482,3 +482,35 @@ def test_agent_start_ebpf_live():
     finally:
         a.stop_ebpf()
         a.close()
+
+
+@pytest.mark.skipif(not __import__("deepflow_amd.ebpf.loader",
+                                   fromlist=["available"]).available(),
+                    reason="bpf attach not permitted here")
+def test_live_cpu_profiler():
+    """The continuous OnCPU profiler live: per-cpu CPU_CLOCK sampling
+    events with the assembled BPF program attached count (tgid, ustack,
+    kstack) in kernel maps; draining folds OUR busy loop's stacks into
+    the profile store."""
+    import os
+    import time as _t
+    import numpy as np
+    from deepflow_amd.ebpf.loader import ProfilerTracer
+    from deepflow_amd.ebpf.profiler import CpuProfiler, ProcSymbolizer
+    from deepflow_amd.ingest.profile_pipeline import ProfilePipeline
+
+    t = ProfilerTracer(sample_freq=199)
+    t.attach()
+    try:
+        x = np.random.rand(900, 900)
+        end = _t.time() + 1.2
+        while _t.time() < end:
+            x = x @ x / np.linalg.norm(x)   # C-level CPU burn
+        pipe = ProfilePipeline()
+        prof = CpuProfiler(pipe, symbolizer=ProcSymbolizer())
+        n = prof.drain_kernel(t.map_fds, int(_t.time() * 1e9))
+    finally:
+        t.close()
+    assert n > 0
+    mine = [r for r in pipe.store.rows if r.pid == os.getpid()]
+    assert mine, f"{n} rows drained, none for our pid"
