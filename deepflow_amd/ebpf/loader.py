@@ -310,9 +310,10 @@ class ProfilerTracer:
 
 def find_libssl() -> Optional[str]:
     import glob as _glob
-    for pat in ("/usr/lib/x86_64-linux-gnu/libssl.so*",
-                "/lib/x86_64-linux-gnu/libssl.so*",
-                "/usr/lib64/libssl.so*"):
+    for pat in ("/lib/x86_64-linux-gnu/libssl.so.*",
+                "/usr/lib/x86_64-linux-gnu/libssl.so.*",
+                "/usr/lib64/libssl.so.*",
+                "/usr/lib/x86_64-linux-gnu/libssl.so*"):
         hits = sorted(_glob.glob(pat))
         if hits:
             return hits[0]
@@ -366,11 +367,18 @@ class SocketTracer:
             for name, spec in SSL_MAPS.items():
                 self.map_fds[name] = map_create(*spec)
             BPF_PROG_TYPE_KPROBE = 2
-            for builder in (build_ssl_write, build_ssl_read_enter,
-                            build_ssl_read_exit):
-                self.ssl_fds.append(prog_load(
-                    BPF_PROG_TYPE_KPROBE,
-                    builder().to_bytes(self.map_fds), log=True))
+            # order: write, read_enter, read_exit, read_enter_ex,
+            # read_exit_ex — OpenSSL 3 callers (CPython included) go
+            # through SSL_write_ex/SSL_read_ex
+            for blob in (build_ssl_write().to_bytes(self.map_fds),
+                         build_ssl_read_enter().to_bytes(self.map_fds),
+                         build_ssl_read_exit().to_bytes(self.map_fds),
+                         build_ssl_read_enter(ex=True).to_bytes(
+                             self.map_fds),
+                         build_ssl_read_exit(ex=True).to_bytes(
+                             self.map_fds)):
+                self.ssl_fds.append(prog_load(BPF_PROG_TYPE_KPROBE,
+                                              blob, log=True))
         self.tp_fds: List[int] = []
         self.rings: List[tuple] = []
 
@@ -388,18 +396,19 @@ class SocketTracer:
         if self.libssl is not None and self.ssl_fds:
             # TLS plaintext capture: uprobes on OpenSSL entry points
             # (reference kernel/openssl.bpf.c)
-            try:
-                w_off = elf_sym_file_offset(self.libssl, "SSL_write")
-                r_off = elf_sym_file_offset(self.libssl, "SSL_read")
-                self.tp_fds.append(attach_uprobe(self.ssl_fds[0],
-                                                 self.libssl, w_off))
-                self.tp_fds.append(attach_uprobe(self.ssl_fds[1],
-                                                 self.libssl, r_off))
-                self.tp_fds.append(attach_uprobe(self.ssl_fds[2],
-                                                 self.libssl, r_off,
-                                                 retprobe=True))
-            except (OSError, KeyError):
-                pass  # no uprobe PMU / stripped lib: syscalls still on
+            for sym, prog_i, ret in (("SSL_write", 0, False),
+                                      ("SSL_write_ex", 0, False),
+                                      ("SSL_read", 1, False),
+                                      ("SSL_read", 2, True),
+                                      ("SSL_read_ex", 3, False),
+                                      ("SSL_read_ex", 4, True)):
+                try:
+                    off = elf_sym_file_offset(self.libssl, sym)
+                    self.tp_fds.append(attach_uprobe(
+                        self.ssl_fds[prog_i], self.libssl, off,
+                        retprobe=ret))
+                except (OSError, KeyError):
+                    pass  # symbol absent / no uprobe PMU: syscalls on
         self._open_rings()
 
     def _open_rings(self, pages: int = 64) -> None:

@@ -410,9 +410,10 @@ SC_SSL_READ, SC_SSL_WRITE = 0xFFF0, 0xFFF1
 TLS_FD = 0xFFFFFFFF
 
 SSL_MAPS = {
-    # pid_tgid -> user buf ptr stashed at SSL_read entry
-    "ssl_args": (BPF_MAP_TYPE_HASH, 8, 8, 65536),
+    # pid_tgid -> {u64 buf ptr, u64 out-len ptr (SSL_read_ex) or 0}
+    "ssl_args": (BPF_MAP_TYPE_HASH, 8, 16, 65536),
 }
+PT_RCX = 88
 
 
 def _ssl_emit(a: Asm, syscall_marker: int, direction: int) -> None:
@@ -492,19 +493,25 @@ def build_ssl_write() -> Asm:
     return a
 
 
-def build_ssl_read_enter() -> Asm:
-    """uprobe SSL_read entry: stash the destination buffer per thread."""
+def build_ssl_read_enter(ex: bool = False) -> Asm:
+    """uprobe SSL_read / SSL_read_ex entry: stash the destination
+    buffer (and for _ex the out-length pointer) per thread."""
     a = Asm()
     a.mov64(R6, R1)
     a.call(H_GET_PID_TGID)
     a.stx(BPF_DW, R10, -8, R0)
     a.ldx(BPF_DW, R2, R6, PT_RSI)         # buf
+    a.stx(BPF_DW, R10, -24, R2)
+    if ex:
+        a.ldx(BPF_DW, R2, R6, PT_RCX)     # size_t *readbytes
+    else:
+        a.mov64_imm(R2, 0)
     a.stx(BPF_DW, R10, -16, R2)
     a.ld_map_fd(R1, "ssl_args")
     a.mov64(R2, R10)
     a.alu64_imm(BPF_ADD, R2, -8)
     a.mov64(R3, R10)
-    a.alu64_imm(BPF_ADD, R3, -16)
+    a.alu64_imm(BPF_ADD, R3, -24)
     a.mov64_imm(R4, 0)
     a.call(H_MAP_UPDATE)
     a.mov64_imm(R0, 0)
@@ -512,8 +519,10 @@ def build_ssl_read_enter() -> Asm:
     return a
 
 
-def build_ssl_read_exit() -> Asm:
-    """uretprobe SSL_read: ret = plaintext length; emit the event."""
+def build_ssl_read_exit(ex: bool = False) -> Asm:
+    """uretprobe SSL_read / SSL_read_ex: plaintext length comes from
+    the return value (SSL_read) or the stashed out-pointer
+    (SSL_read_ex, where ret is just a success flag); emit the event."""
     a = Asm()
     a.mov64(R6, R1)
     a.call(H_GET_PID_TGID)
@@ -528,6 +537,8 @@ def build_ssl_read_exit() -> Asm:
     a.label("have_args")
     a.ldx(BPF_DW, R2, R0, 0)              # stashed buf
     a.stx(BPF_DW, R10, -32, R2)
+    a.ldx(BPF_DW, R2, R0, 8)              # stashed out-len ptr (_ex)
+    a.stx(BPF_DW, R10, -48, R2)
     a.ld_map_fd(R1, "ssl_args")
     a.mov64(R2, R10)
     a.alu64_imm(BPF_ADD, R2, -8)
@@ -537,7 +548,20 @@ def build_ssl_read_exit() -> Asm:
     a.mov64_imm(R0, 0)
     a.exit()
     a.label("ret_ok")
-    a.stx(BPF_DW, R10, -40, R2)
+    if ex:
+        # length lives behind the out-pointer: *readbytes
+        a.mov64(R1, R10)
+        a.alu64_imm(BPF_ADD, R1, -40)
+        a.mov64_imm(R2, 8)
+        a.ldx(BPF_DW, R3, R10, -48)
+        a.call(H_PROBE_READ_USER)
+        a.ldx(BPF_DW, R2, R10, -40)
+        a.jmp_imm(BPF_JSGE, R2, 1, "len_ok2")
+        a.mov64_imm(R0, 0)
+        a.exit()
+        a.label("len_ok2")
+    else:
+        a.stx(BPF_DW, R10, -40, R2)
     _ssl_emit(a, SC_SSL_READ, 1)
     return a
 

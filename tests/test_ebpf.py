@@ -514,3 +514,85 @@ def test_live_cpu_profiler():
     assert n > 0
     mine = [r for r in pipe.store.rows if r.pid == os.getpid()]
     assert mine, f"{n} rows drained, none for our pid"
+
+
+def _uprobe_pmu_present() -> bool:
+    import os
+    return os.path.exists("/sys/bus/event_source/devices/uprobe/type")
+
+
+@pytest.mark.skipif(not (__import__("deepflow_amd.ebpf.loader",
+                                    fromlist=["available"]).available()
+                         and _uprobe_pmu_present()),
+                    reason="bpf attach / uprobe PMU not available")
+def test_live_tls_plaintext_capture():
+    """OpenSSL uprobes LIVE: a real TLS 1.3 connection through
+    CPython's ssl module (OpenSSL 3, SSL_write_ex/SSL_read_ex) — the
+    uprobe programs capture the decrypted request/response plaintext
+    and the in-kernel inference tags it HTTP."""
+    import os
+    import socket
+    import ssl as _ssl
+    import struct
+    import subprocess
+    import tempfile
+    import threading
+    import time as _t
+    from deepflow_amd.ebpf.loader import SocketTracer
+    from deepflow_amd.ebpf import progs as P
+
+    t = SocketTracer(with_tls=True)
+    t.attach()
+    try:
+        d = tempfile.mkdtemp()
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+             "-keyout", f"{d}/k.pem", "-out", f"{d}/c.pem", "-days",
+             "1", "-nodes", "-subj", "/CN=localhost"],
+            capture_output=True, check=True)
+        ctx = _ssl.SSLContext(_ssl.PROTOCOL_TLS_SERVER)
+        ctx.load_cert_chain(f"{d}/c.pem", f"{d}/k.pem")
+        srv = socket.socket()
+        srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        srv.bind(("127.0.0.1", 0))
+        srv.listen(1)
+
+        def serve():
+            conn, _ = srv.accept()
+            tls = ctx.wrap_socket(conn, server_side=True)
+            tls.recv(4096)
+            tls.sendall(b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+            tls.close()
+
+        th = threading.Thread(target=serve)
+        th.start()
+        cctx = _ssl.SSLContext(_ssl.PROTOCOL_TLS_CLIENT)
+        cctx.check_hostname = False
+        cctx.verify_mode = _ssl.CERT_NONE
+        c = cctx.wrap_socket(
+            socket.create_connection(("127.0.0.1",
+                                      srv.getsockname()[1])))
+        c.sendall(b"GET /secret HTTP/1.1\r\nHost: tls.test\r\n\r\n")
+        c.recv(4096)
+        c.close()
+        th.join()
+        srv.close()
+        events = []
+        deadline = _t.time() + 4
+        while _t.time() < deadline:
+            t.poll(events.append)
+            _t.sleep(0.05)
+    finally:
+        t.close()
+    me = os.getpid()
+    tls_payloads = []
+    for e in events:
+        if len(e) < P.EV_HDR:
+            continue
+        f = struct.unpack_from(P.SK_EVENT_FMT, e)
+        if f[1] == me and f[3] == P.TLS_FD and f[7] == 20:
+            tls_payloads.append(bytes(e[P.EV_HDR:P.EV_HDR + f[5]]))
+    reqs = [p for p in tls_payloads if p.startswith(b"GET /secret")]
+    resps = [p for p in tls_payloads if p.startswith(b"HTTP/1.1 200")]
+    assert reqs, f"{len(tls_payloads)} TLS events, no request plaintext"
+    assert resps, f"{len(tls_payloads)} TLS events, no response plaintext"
