@@ -127,13 +127,20 @@ class Agent:
             return None
         tracer = loader.SocketTracer()
         tracer.attach()
-        coll = runtime.EbpfCollector(self)
+        coll = runtime.EbpfCollector(self, drop_stale_s=1.0)
         import threading
         stop = threading.Event()
 
         def pump():
             while not stop.wait(poll_interval_s):
-                tracer.poll(coll.on_event)
+                batch: list = []
+                tracer.poll(batch.append)
+                # per-cpu rings drain ring-by-ring: sort the batch by
+                # event timestamp so TLS-uprobe events see the pairing
+                # syscall first (EV_TS is the leading u64)
+                batch.sort(key=lambda e: int.from_bytes(e[:8], 'little'))
+                for ev in batch:
+                    coll.on_event(ev)
                 coll.flush()
         th = threading.Thread(target=pump, daemon=True)
         th.start()

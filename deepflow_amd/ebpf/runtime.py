@@ -94,9 +94,17 @@ TLS_FD = 0xFFFFFFFF
 
 
 class EbpfCollector:
-    def __init__(self, agent, resolver=None):
+    def __init__(self, agent, resolver=None,
+                 drop_stale_s: "Optional[float]" = None):
+        """drop_stale_s: in LIVE pumping (sub-100ms processing lag),
+        drop events older than this — the ring backlog from before the
+        workload would otherwise resolve against CURRENT /proc state
+        and attribute a recycled fd number's old traffic to the new
+        socket. Leave None for replay/batch feeding."""
         self.agent = agent
         self.resolver = resolver or ProcSocketResolver()
+        self.drop_stale_s = drop_stale_s
+        self.stale_dropped = 0
         self.events_in = 0
         self.unresolved = 0
         self._batch: List[bytes] = []
@@ -112,13 +120,24 @@ class EbpfCollector:
             return
         (ts, tgid, pid, fd, ln, cap, direction, proto_hint, _sc, trace,
          _skey) = struct.unpack(SK_EVENT_FMT, ev[:EV_HDR])
+        if self.drop_stale_s is not None:
+            import time as _time
+            now = _time.clock_gettime(_time.CLOCK_MONOTONIC) * 1e9
+            if ts < now - self.drop_stale_s * 1e9:
+                self.stale_dropped += 1
+                return
         payload = ev[EV_HDR:EV_HDR + cap]
         if fd == TLS_FD:
             fd = self._tid_last_fd.get((tgid, pid), -1)
         else:
-            self._tid_last_fd[(tgid, pid)] = fd
-            if len(self._tid_last_fd) > (1 << 16):
-                self._tid_last_fd.clear()
+            # pair TLS-uprobe events only with sockets whose syscalls
+            # carry TLS ciphertext (record header 0x14-0x17 0x03 0x0x) —
+            # "last fd of the thread" alone grabs unrelated chatty fds
+            if len(payload) >= 3 and 0x14 <= payload[0] <= 0x17 and \
+                    payload[1] == 0x03 and payload[2] <= 0x04:
+                self._tid_last_fd[(tgid, pid)] = fd
+                if len(self._tid_last_fd) > (1 << 16):
+                    self._tid_last_fd.clear()
         tup = self.resolver.resolve(tgid, fd) if fd >= 0 else None
         self.events_in += 1
         if tup is None:

@@ -2875,8 +2875,13 @@ int dfa_syscall_event(void* h, uint64_t ts_ns, uint32_t tgid, int dir,
     // real in-kernel inference verdict arrives and the flow has emitted
     // nothing, adopt the verdict AND re-seed the client side from this
     // event's data sender (inference fires on request-shaped data).
-    if (f.l7_protocol == 0 && l7_hint &&
-        f.l7c.request_count == 0 && f.l7c.response_count == 0) {
+    // 121 = TLS: ciphertext syscalls label the flow TLS (and may log
+    // the handshake); decrypted uprobe plaintext carries the REAL
+    // protocol — it supersedes TLS even after handshake records
+    bool fresh_unknown = f.l7_protocol == 0 &&
+        f.l7c.request_count == 0 && f.l7c.response_count == 0;
+    bool tls_to_plain = f.l7_protocol == 121 && l7_hint != 121;
+    if (l7_hint && (fresh_unknown || tls_to_plain)) {
         f.l7_protocol = l7_hint;
         f.ip[0] = ip_src; f.ip[1] = ip_dst;
         f.port[0] = port_src; f.port[1] = port_dst;
