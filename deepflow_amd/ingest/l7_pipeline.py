@@ -2,8 +2,10 @@
 intern -> string-pool gather -> metric rollup -> columnar segment.
 
 GPU mode launches the HIP kernels (K1/K2/K3/K4/K5) asynchronously on the
-current torch stream, with exactly two host syncs per batch (pool sizing
-cumsum + dictionary harvest). CPU mode runs the reference ops; both modes
+current torch stream with ONE host sync per batch (the combined
+attr/string pool sizing readback); the dictionary harvest defers behind
+a CUDA event (steady state = no sync) and the naive-bytes accounting
+accumulates on-device. CPU mode runs the reference ops; both modes
 produce identical logical contents (tests/test_pipeline_cpu.py,
 tests/test_gpu_pipeline.py).
 
