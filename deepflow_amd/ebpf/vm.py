@@ -54,6 +54,7 @@ class Vm:
         self.regions: List[Tuple[int, bytearray]] = []
         self._next_val = VAL_BASE
         self.user_mem: Dict[int, bytes] = {}   # user addr -> bytes
+        self.kernel_mem: Dict[int, bytes] = {}  # kernel addr -> bytes
         self.events: List[bytes] = []          # perf_event_output captures
         self.clock = 1_700_000_000_000_000_000
         self.pid_tgid = (1234 << 32) | 1234
@@ -138,9 +139,17 @@ class Vm:
             data = self.read_bytes(r[4], size)
             self.events.append(data)
             return 0
-        if hid == I.H_PROBE_READ_USER:
+        if hid in (I.H_PROBE_READ_USER, I.H_PROBE_READ_KERNEL):
             dst, size, src = r[1], r[2], r[3]
-            blob = self.user_mem.get(src)
+            pool = self.user_mem if hid == I.H_PROBE_READ_USER \
+                else self.kernel_mem
+            blob = pool.get(src)
+            if blob is None:
+                # offset into a registered base block (pt_regs fields)
+                for base, b in pool.items():
+                    if base <= src < base + len(b):
+                        blob = b[src - base:]
+                        break
             if blob is None:
                 return -14  # -EFAULT
             chunk = blob[:size].ljust(size, b"\x00")
@@ -319,6 +328,48 @@ class SyscallSim:
 
     def events(self) -> List[bytes]:
         return self.vm.events
+
+
+def raw_tp_sys_enter_ctx(vm: "Vm", syscall: int, fd: int,
+                         buf: int, count: int,
+                         regs_addr: int = 0x6B00_0000_0000) -> bytes:
+    """BPF_RAW_TRACEPOINT sys_enter ctx: {pt_regs*, id}; the pt_regs
+    block is registered as VM kernel memory for probe_read_kernel."""
+    regs = [0] * 21
+    regs[14] = fd & M64     # rdi
+    regs[13] = buf & M64    # rsi
+    regs[12] = count & M64  # rdx
+    vm.kernel_mem[regs_addr] = struct.pack("<21Q", *regs)
+    return struct.pack("<Qq", regs_addr, syscall)
+
+
+def raw_tp_sys_exit_ctx(ret: int) -> bytes:
+    return struct.pack("<Qq", 0, ret)
+
+
+class RawSyscallSim(SyscallSim):
+    """SyscallSim over the RAW-tracepoint program variants (the ones
+    BPF_RAW_TRACEPOINT_OPEN attaches in production)."""
+
+    def __init__(self):
+        from .progs import MAPS, build_sys_enter, build_sys_exit
+        self.vm = Vm(MAPS)
+        self.enter = build_sys_enter(raw=True)
+        self.exit = build_sys_exit(raw=True)
+        self._next_user = USER_BASE
+
+    def syscall(self, tgid: int, pid: int, syscall: int, fd: int,
+                payload: bytes, ret: Optional[int] = None) -> None:
+        self.vm.pid_tgid = ((tgid << 32) | pid) & M64
+        ubuf = self._next_user
+        self._next_user += (len(payload) + 4095) & ~4095 or 4096
+        self.vm.user_mem[ubuf] = payload
+        self.vm.run(self.enter, raw_tp_sys_enter_ctx(
+            self.vm, syscall, fd, ubuf, len(payload)))
+        self.vm.clock += 1000
+        self.vm.run(self.exit, raw_tp_sys_exit_ctx(
+            len(payload) if ret is None else ret))
+        self.vm.clock += 1000
 
 
 def pt_regs_ctx(rdi: int = 0, rsi: int = 0, rdx: int = 0,

@@ -596,3 +596,19 @@ def test_live_tls_plaintext_capture():
     resps = [p for p in tls_payloads if p.startswith(b"HTTP/1.1 200")]
     assert reqs, f"{len(tls_payloads)} TLS events, no request plaintext"
     assert resps, f"{len(tls_payloads)} TLS events, no response plaintext"
+
+
+def test_raw_tp_programs_match_tracefs_variants():
+    """The raw-tracepoint program variants (production attach mode)
+    produce byte-identical events to the classic tracefs variants in
+    the VM."""
+    from deepflow_amd.ebpf.vm import SyscallSim, RawSyscallSim
+    a, b = SyscallSim(), RawSyscallSim()
+    for sim in (a, b):
+        sim.syscall(900, 901, 1, 7,
+                    b"GET /raw HTTP/1.1\r\nHost: r\r\n\r\n")
+        sim.syscall(900, 901, 0, 7,
+                    b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n\r\nok")
+    assert len(a.events()) == len(b.events()) == 2
+    for ea, eb in zip(a.events(), b.events()):
+        assert ea == eb
