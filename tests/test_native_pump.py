@@ -363,3 +363,32 @@ def test_pump_frame_larger_than_ring_dropped():
     finally:
         a.close()
         p.close()
+
+
+@pytest.mark.gpu
+def test_multi_org_isolation_gpu():
+    """Per-org pipelines on the GPU: spans sent under different org ids
+    land in separate device stores and separate SQL scopes."""
+    from deepflow_amd.server import DeepflowServer
+    from deepflow_amd.gen.spans import SpanGenConfig, gen_span_payload
+    import numpy as np
+
+    srv = DeepflowServer(device="cuda", tcp_port=0, segment_rows=1 << 14,
+                         dict_capacity=1 << 15, time_base_s=0)
+    payload = np.frombuffer(
+        gen_span_payload(SpanGenConfig(n=300, seed=17, n_ips=32)),
+        dtype=np.uint8)
+    srv._on_l7(framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG,
+                                   org_id=1), payload)
+    srv._on_l7(framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG,
+                                   org_id=7), payload)
+    srv._on_l7(framing.FrameHeader(msg_type=framing.MSG_PROTOCOLLOG,
+                                   org_id=7), payload)
+    import torch
+    torch.cuda.synchronize()
+    r1 = srv.engine.query("SELECT COUNT(1) FROM l7_flow_log")
+    assert r1["values"][0][0] == 300
+    eng7 = srv.org_context(7).engine
+    r7 = eng7.query("SELECT COUNT(1) FROM l7_flow_log")
+    assert r7["values"][0][0] == 600
+    assert srv.org_context(7).l7.segments.device == "cuda"
